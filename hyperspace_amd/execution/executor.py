@@ -1,0 +1,498 @@
+"""Physical plan executor.
+
+Executes LogicalPlan trees over ColumnBatches.  On a GPU box all hot paths
+run the hand-written HIP kernels via hyperspace_amd.ops (which raises if
+the extension is missing — no silent eager fallback); on CPU the torch
+reference ops run instead (test/plumbing mode, BASELINE config 1).
+
+Operator -> data-plane mapping:
+  Scan            K1  parquet decode (host pyarrow path or device decode)
+  Filter          K7 / filter scan: native select_range + gather
+  Join            K4  per-bucket merge join (zero exchange when co-bucketed)
+  IndexScan       bucket-pruned index read (FilterIndexRule target)
+  BucketUnionNode K5  partition-aligned concat (+ per-bucket re-sort)
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from .columnar import ColumnBatch, StringColumn
+from .. import ops
+from ..config import IndexConstants
+from ..exceptions import HyperspaceException
+from ..plan.expr import (And, BinComp, Col, Expr, In, IsNotNull, Lit, Not,
+                         Or, extract_equi_join_keys)
+from ..plan.nodes import (BucketUnionNode, Filter, IndexScan, Join,
+                          LogicalPlan, Project, Scan)
+from ..sources.parquet_io import read_files_batch, bucket_id_of_file
+
+
+class PhysicalStats:
+    """Per-execution counters (telemetry + tests assert on these —
+    e.g. shuffle/exchange elimination in the indexed join path)."""
+
+    def __init__(self):
+        self.scanned_files = 0
+        self.scanned_bytes = 0
+        self.shuffles = 0          # on-the-fly repartitions (Exchange analog)
+        self.merge_joins = 0
+        self.hash_joins = 0
+        self.bucket_pruned_files = 0
+        self.operators: List[str] = []
+
+    def record(self, name: str):
+        self.operators.append(name)
+
+
+class Executor:
+    def __init__(self, session):
+        self.session = session
+        self.stats = PhysicalStats()
+
+    @property
+    def device(self):
+        return self.session.device
+
+    # ------------------------------------------------------------------
+    def execute(self, plan: LogicalPlan) -> ColumnBatch:
+        batch, _ = self._exec(plan)
+        return batch
+
+    def _exec(self, plan: LogicalPlan
+              ) -> Tuple[ColumnBatch, Optional[torch.Tensor]]:
+        """Returns (batch, bucket segment offsets or None).
+
+        The second element is set when the batch is bucket-partitioned
+        (IndexScan with use_bucket_spec, BucketUnion) — segment offsets
+        have length num_buckets+1.
+        """
+        if isinstance(plan, Scan):
+            return self._exec_scan(plan), None
+        if isinstance(plan, IndexScan):
+            return self._exec_index_scan(plan, eq_prune=None)
+        if isinstance(plan, Filter):
+            return self._exec_filter(plan)
+        if isinstance(plan, Project):
+            batch, seg = self._exec(plan.child)
+            self.stats.record("Project")
+            return batch.select(plan.columns), seg
+        if isinstance(plan, Join):
+            return self._exec_join(plan), None
+        if isinstance(plan, BucketUnionNode):
+            return self._exec_bucket_union(plan)
+        raise HyperspaceException(f"Cannot execute {type(plan).__name__}")
+
+    # ------------------------------------------------------------------
+    def _exec_scan(self, plan: Scan, file_subset: Optional[List[str]] = None,
+                   lineage_tracker=None) -> ColumnBatch:
+        self.stats.record("ParquetScan")
+        files = plan.relation.all_files()
+        if file_subset is not None:
+            subset = set(file_subset)
+            files = [f for f in files if f.name in subset]
+        paths = [f.name for f in files]
+        self.stats.scanned_files += len(paths)
+        self.stats.scanned_bytes += sum(f.size for f in files)
+        batch, row_counts = read_files_batch(paths)
+        if lineage_tracker is not None:
+            tracker = lineage_tracker
+            ids = []
+            for f, n in zip(files, row_counts):
+                fid = tracker.add_file(f.name, f.size, f.modifiedTime)
+                ids.append(torch.full((n,), fid, dtype=torch.int64))
+            lineage = (torch.cat(ids) if ids
+                       else torch.empty(0, dtype=torch.int64))
+            batch = batch.with_column(
+                IndexConstants.DATA_FILE_NAME_ID_COLUMN, lineage)
+        if self.device.type == "cuda":
+            batch = batch.to(self.device)
+        return batch
+
+    # ------------------------------------------------------------------
+    def _exec_index_scan(self, plan: IndexScan,
+                         eq_prune: Optional[Tuple[str, object]] = None
+                         ) -> Tuple[ColumnBatch, Optional[torch.Tensor]]:
+        """Read covering-index data.  When ``use_bucket_spec`` the result is
+        ordered by bucket with segment offsets returned; ``eq_prune`` =
+        (column, value) prunes to the single matching bucket file set."""
+        self.stats.record("IndexScan")
+        entry = plan.entry
+        index = entry.derivedDataset
+        num_buckets = index.num_buckets
+        all_files = (plan.version_files if plan.version_files is not None
+                     else [f for f in entry.content.os_files()
+                           if f.endswith(".parquet")])
+
+        # group files by bucket id from the filename contract
+        by_bucket: Dict[int, List[str]] = {}
+        for p in all_files:
+            b = bucket_id_of_file(p)
+            if b is None:
+                raise HyperspaceException(f"Index file without bucket id: {p}")
+            by_bucket.setdefault(b, []).append(p)
+
+        wanted_buckets = sorted(by_bucket)
+        if eq_prune is not None:
+            col_name, value = eq_prune
+            b = self._bucket_of_value(index, col_name, value, num_buckets)
+            pruned = [x for x in wanted_buckets if x != b]
+            self.stats.bucket_pruned_files += sum(
+                len(by_bucket[x]) for x in pruned)
+            wanted_buckets = [b] if b in by_bucket else []
+
+        read_cols = list(plan.columns)
+        if plan.excluded_source_file_ids:
+            lineage_col = IndexConstants.DATA_FILE_NAME_ID_COLUMN
+            if lineage_col not in read_cols:
+                read_cols = read_cols + [lineage_col]
+
+        batches: List[ColumnBatch] = []
+        seg_counts = torch.zeros(num_buckets + 1, dtype=torch.int64)
+        for b in wanted_buckets:
+            paths = sorted(by_bucket[b])
+            self.stats.scanned_files += len(paths)
+            sub, _ = read_files_batch(paths, columns=read_cols)
+            seg_counts[b + 1] = sub.num_rows
+            batches.append(sub)
+        if batches:
+            batch = ColumnBatch.concat(batches)
+        else:
+            batch = ColumnBatch({c: torch.empty(0) for c in read_cols})
+        if self.device.type == "cuda":
+            batch = batch.to(self.device)
+
+        # lineage delete filter (Hybrid Scan deletes, K7)
+        if plan.excluded_source_file_ids and batch.num_rows:
+            ids = torch.tensor(sorted(plan.excluded_source_file_ids),
+                               dtype=torch.int64, device=batch.device)
+            lineage = batch.tensor(IndexConstants.DATA_FILE_NAME_ID_COLUMN)
+            keep = ~ops.isin_sorted(lineage, ids)
+            kept_idx = torch.nonzero(keep, as_tuple=False).flatten()
+            # recompute per-bucket counts after the filter
+            if plan.use_bucket_spec:
+                bounds = torch.cumsum(seg_counts, 0)
+                new_counts = torch.zeros_like(seg_counts)
+                for b in wanted_buckets:
+                    lo, hi = int(bounds[b]), int(bounds[b + 1])
+                    new_counts[b + 1] = int(
+                        ((kept_idx >= lo) & (kept_idx < hi)).sum())
+                seg_counts = new_counts
+            batch = batch.gather(kept_idx)
+            self.stats.record("LineageFilter")
+        batch = batch.select(plan.columns)
+
+        if plan.use_bucket_spec:
+            seg = torch.cumsum(seg_counts, 0)
+            return batch, seg
+        return batch, None
+
+    def _bucket_of_value(self, index, col_name: str, value,
+                         num_buckets: int) -> int:
+        t = _value_tensor(index.schema.field_type(col_name), value)
+        return int(ops.cpu_ref.murmur3_bucket([t], num_buckets)[0])
+
+    # ------------------------------------------------------------------
+    def _exec_filter(self, plan: Filter
+                     ) -> Tuple[ColumnBatch, Optional[torch.Tensor]]:
+        # bucket pruning: Filter(eq on first indexed col, IndexScan)
+        child = plan.child
+        if isinstance(child, IndexScan) and child.use_bucket_spec:
+            eq = _single_equality(plan.condition)
+            index = child.entry.derivedDataset
+            if eq is not None and index.indexed_columns and \
+                    eq[0].lower() == index.indexed_columns[0].lower():
+                batch, seg = self._exec_index_scan(child, eq_prune=eq)
+                self.stats.record("Filter")
+                return self._apply_predicate(batch, plan.condition), None
+        batch, seg = self._exec(child)
+        self.stats.record("Filter")
+        return self._apply_predicate(batch, plan.condition), None
+
+    def _apply_predicate(self, batch: ColumnBatch, cond: Expr) -> ColumnBatch:
+        if batch.num_rows == 0:
+            return batch
+        idx = self._predicate_indices(batch, cond)
+        return batch.gather(idx)
+
+    def _predicate_indices(self, batch: ColumnBatch,
+                           cond: Expr) -> torch.Tensor:
+        """Native fast path for a single numeric comparison (select_range
+        kernel); general path composes boolean masks."""
+        rng = _as_range(cond, batch)
+        if rng is not None:
+            keys_u64, lo, hi, lo_incl, hi_incl = rng
+            return ops.select_range_u64(keys_u64, lo, hi, lo_incl, hi_incl)
+        mask = self._eval_mask(batch, cond)
+        return torch.nonzero(mask, as_tuple=False).flatten()
+
+    def _eval_mask(self, batch: ColumnBatch, e: Expr) -> torch.Tensor:
+        if isinstance(e, And):
+            return self._eval_mask(batch, e.left) & \
+                self._eval_mask(batch, e.right)
+        if isinstance(e, Or):
+            return self._eval_mask(batch, e.left) | \
+                self._eval_mask(batch, e.right)
+        if isinstance(e, Not):
+            return ~self._eval_mask(batch, e.child)
+        if isinstance(e, IsNotNull):
+            # nulls unsupported in v0: all rows pass
+            return torch.ones(batch.num_rows, dtype=torch.bool,
+                              device=batch.device)
+        if isinstance(e, In):
+            col = batch.column(e.col.name)
+            vals = e.values
+            if isinstance(col, StringColumn):
+                codes = torch.tensor(
+                    sorted(c for c in (col.code_of(v) for v in vals)
+                           if c >= 0),
+                    dtype=torch.int64, device=col.codes.device)
+                return ops.isin_sorted(col.codes.to(torch.int64), codes)
+            t = col.to(torch.int64)
+            vs = torch.tensor(sorted(int(v) for v in vals),
+                              dtype=torch.int64, device=t.device)
+            return ops.isin_sorted(t, vs)
+        if isinstance(e, BinComp):
+            return _compare(batch, e)
+        raise HyperspaceException(f"Cannot evaluate {e!r}")
+
+    # ------------------------------------------------------------------
+    def _exec_join(self, plan: Join) -> ColumnBatch:
+        pairs = extract_equi_join_keys(plan.condition)
+        if not pairs:
+            raise HyperspaceException(
+                "Only equi-joins are executable in v0")
+        lkeys_names = [p[0] for p in pairs]
+        rkeys_names = [p[1] for p in pairs]
+
+        left_bucketed = _bucketed_side(plan.left, lkeys_names)
+        right_bucketed = _bucketed_side(plan.right, rkeys_names)
+
+        lbatch, lseg = self._exec(plan.left)
+        rbatch, rseg = self._exec(plan.right)
+
+        if (left_bucketed and right_bucketed and lseg is not None
+                and rseg is not None and lseg.numel() == rseg.numel()):
+            # zero-shuffle co-bucketed sort-merge join (K4):
+            # both sides already hash-partitioned into the same buckets and
+            # sorted by the join key within each bucket at build time.
+            self.stats.merge_joins += 1
+            self.stats.record("SortMergeJoin(co-bucketed)")
+            lk = ops.normalize_key(lbatch.tensor(lkeys_names[0]))
+            rk = ops.normalize_key(rbatch.tensor(rkeys_names[0]))
+            lidx, ridx = ops.merge_join(lk, rk, lseg, rseg)
+        else:
+            # on-the-fly sort-merge join: shuffle-equivalent (counts as an
+            # Exchange in plan-diff tests)
+            self.stats.shuffles += 2
+            self.stats.hash_joins += 1
+            self.stats.record("SortMergeJoin(shuffled)")
+            lk = ops.normalize_key(lbatch.tensor(lkeys_names[0]))
+            rk = ops.normalize_key(rbatch.tensor(rkeys_names[0]))
+            lperm = ops.sort_perm(lk)
+            rperm = ops.sort_perm(rk)
+            lbatch = lbatch.gather(lperm)
+            rbatch = rbatch.gather(rperm)
+            lk, rk = lk[lperm], rk[rperm]
+            one_seg = torch.tensor([0, lk.numel()], dtype=torch.int64)
+            one_seg_r = torch.tensor([0, rk.numel()], dtype=torch.int64)
+            lidx, ridx = ops.merge_join(lk, rk, one_seg, one_seg_r)
+
+        # secondary key equality check for multi-key joins
+        if len(pairs) > 1 and lidx.numel():
+            keep = torch.ones(lidx.numel(), dtype=torch.bool,
+                              device=lidx.device)
+            for ln, rn in pairs[1:]:
+                lv = lbatch.tensor(ln)[lidx]
+                rv = rbatch.tensor(rn)[ridx]
+                keep &= (lv == rv)
+            sel = torch.nonzero(keep, as_tuple=False).flatten()
+            lidx, ridx = lidx[sel], ridx[sel]
+
+        lout = lbatch.gather(lidx)
+        rout = rbatch.gather(ridx)
+        cols: Dict[str, object] = {}
+        for k, v in lout.columns.items():
+            cols[k] = v
+        for k, v in rout.columns.items():
+            cols[k if k not in cols else f"{k}_r"] = v
+        return ColumnBatch(cols)
+
+    # ------------------------------------------------------------------
+    def _exec_bucket_union(self, plan: BucketUnionNode
+                           ) -> Tuple[ColumnBatch, Optional[torch.Tensor]]:
+        """Partition-aligned union: concatenate same-bucket segments from
+        each child, re-sorting each merged bucket by the bucket columns
+        (K5; appended data was shuffled in by K2/K6 beforehand)."""
+        self.stats.record("BucketUnion")
+        n = plan.num_buckets
+        child_parts: List[Tuple[ColumnBatch, torch.Tensor]] = []
+        for c in plan.children:
+            batch, seg = self._exec(c)
+            if seg is None:
+                # on-the-fly repartition of appended data (K2/K6)
+                self.stats.shuffles += 1
+                batch, seg = self._repartition(batch, plan.bucket_columns, n)
+            child_parts.append((batch, seg))
+
+        pieces: List[ColumnBatch] = []
+        counts = torch.zeros(n + 1, dtype=torch.int64)
+        for b in range(n):
+            merged = []
+            for batch, seg in child_parts:
+                lo, hi = int(seg[b]), int(seg[b + 1])
+                if hi > lo:
+                    merged.append(batch.slice(lo, hi))
+            if not merged:
+                continue
+            mb = ColumnBatch.concat(merged)
+            # re-sort bucket by first key for downstream merge join
+            key = ops.normalize_key(mb.tensor(plan.bucket_columns[0]))
+            perm = ops.sort_perm(key)
+            pieces.append(mb.gather(perm))
+            counts[b + 1] = mb.num_rows
+        out = (ColumnBatch.concat(pieces) if pieces
+               else child_parts[0][0].slice(0, 0))
+        return out, torch.cumsum(counts, 0)
+
+    def _repartition(self, batch: ColumnBatch, bucket_cols: List[str],
+                     num_buckets: int
+                     ) -> Tuple[ColumnBatch, torch.Tensor]:
+        """Hash-repartition + per-bucket sort (K2+K3 on the fly)."""
+        keys = [batch.tensor(c) for c in bucket_cols]
+        bucket_ids = ops.murmur3_bucket(keys, num_buckets)
+        sort_key = ops.normalize_key(batch.tensor(bucket_cols[0]))
+        perm = ops.sort_perm(sort_key)
+        b_sorted = bucket_ids.to(torch.int64)[perm]
+        perm2 = ops.sort_perm(ops.normalize_key(b_sorted))
+        perm = perm[perm2]
+        batch = batch.gather(perm)
+        counts = torch.bincount(bucket_ids.to(torch.int64).cpu(),
+                                minlength=num_buckets)
+        seg = torch.zeros(num_buckets + 1, dtype=torch.int64)
+        seg[1:] = torch.cumsum(counts, 0)
+        return batch, seg
+
+
+# ---------------------------------------------------------------------------
+# helpers
+# ---------------------------------------------------------------------------
+
+_SPARK_DTYPES = {"long": torch.int64, "integer": torch.int32,
+                 "double": torch.float64, "float": torch.float32}
+
+
+def _compare(batch: ColumnBatch, e: BinComp) -> torch.Tensor:
+    """General comparison mask (non-hot path; hot single comparisons lower
+    to select_range_u64 in _predicate_indices)."""
+    if not isinstance(e.left, Col):
+        raise HyperspaceException(f"Unsupported comparison {e!r}")
+    col = batch.column(e.left.name)
+    if isinstance(e.right, Col):
+        rhs = batch.column(e.right.name)
+        lv = col.codes if isinstance(col, StringColumn) else col
+        rv = rhs.codes if isinstance(rhs, StringColumn) else rhs
+    else:
+        value = e.right.value
+        if isinstance(col, StringColumn):
+            # dictionary is sorted, so code order == lexicographic order;
+            # an absent literal maps to its insertion point.
+            codes = col.codes
+            pos = col.searchsorted(str(value))
+            exact = (pos < len(col.values) and col.values[pos] == str(value))
+            op = e.op
+            if op == "=":
+                return (codes == pos if exact else
+                        torch.zeros(len(col), dtype=torch.bool,
+                                    device=codes.device))
+            if op == "!=":
+                return (codes != pos if exact else
+                        torch.ones(len(col), dtype=torch.bool,
+                                   device=codes.device))
+            if op == "<":
+                return codes < pos
+            if op == "<=":
+                return codes < pos + (1 if exact else 0)
+            if op == ">":
+                return codes >= pos + (1 if exact else 0)
+            return codes >= pos  # >=
+        else:
+            lv = col
+            rv = torch.tensor(value, dtype=col.dtype, device=col.device)
+    op = e.op
+    if op == "=":
+        return lv == rv
+    if op == "!=":
+        return lv != rv
+    if op == "<":
+        return lv < rv
+    if op == "<=":
+        return lv <= rv
+    if op == ">":
+        return lv > rv
+    return lv >= rv
+
+
+def _value_tensor(spark_type: Optional[str], value) -> torch.Tensor:
+    dt = _SPARK_DTYPES.get(spark_type or "", None)
+    if dt is None:
+        dt = torch.int64 if isinstance(value, int) else torch.float64
+    return torch.tensor([value], dtype=dt)
+
+
+def _single_equality(cond: Expr) -> Optional[Tuple[str, object]]:
+    if isinstance(cond, BinComp) and cond.op == "=" and \
+            isinstance(cond.left, Col) and isinstance(cond.right, Lit):
+        return cond.left.name, cond.right.value
+    return None
+
+
+def _as_range(cond: Expr, batch: ColumnBatch):
+    """Lower a single numeric comparison to the select_range_u64 kernel:
+    returns (keys_u64, lo, hi, lo_incl, hi_incl) or None."""
+    if not (isinstance(cond, BinComp) and isinstance(cond.left, Col)
+            and isinstance(cond.right, Lit)):
+        return None
+    col = batch.column(cond.left.name)
+    if isinstance(col, StringColumn):
+        return None
+    if col.dtype not in (torch.int8, torch.int16, torch.int32, torch.int64,
+                         torch.float32, torch.float64):
+        return None
+    value = cond.right.value
+    vt = torch.tensor([value], dtype=col.dtype)
+    v = int(ops.normalize_key(vt)[0])
+    keys = ops.normalize_key(col)
+    UMIN = -0x8000000000000000  # u64 0
+    UMAX = 0x7FFFFFFFFFFFFFFF   # u64 max (normalized encoding)
+    op = cond.op
+    if op == "=":
+        return keys, v, v, True, True
+    if op == "<":
+        return keys, UMIN, v, True, False
+    if op == "<=":
+        return keys, UMIN, v, True, True
+    if op == ">":
+        return keys, v, UMAX, False, True
+    if op == ">=":
+        return keys, v, UMAX, True, True
+    return None
+
+
+def _bucketed_side(plan: LogicalPlan, key_names: List[str]) -> bool:
+    """True if the subplan yields bucket-partitioned data keyed on
+    key_names (IndexScan/BucketUnion with matching indexed columns)."""
+    node: LogicalPlan = plan
+    while isinstance(node, Project):
+        node = node.child
+    if isinstance(node, IndexScan):
+        idx_cols = [c.lower() for c in node.entry.derivedDataset
+                    .indexed_columns]
+        return node.use_bucket_spec and \
+            idx_cols == [k.lower() for k in key_names]
+    if isinstance(node, BucketUnionNode):
+        return [c.lower() for c in node.bucket_columns] == \
+            [k.lower() for k in key_names]
+    return False

@@ -1,0 +1,92 @@
+"""CoveringIndexConfig: user-facing config + index-data creation.
+
+Reference: index/covering/CoveringIndexConfig.scala:38-62 (+ Builder),
+validation in CoveringIndexConfigTrait.scala:23-52, data creation in
+CoveringIndex.createIndexData (index/covering/CoveringIndex.scala:140-192:
+projection + lineage column via input_file_name() join against the
+broadcast file-id map — here the lineage ids are attached at scan time,
+K12 folded into the scan).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+from ..base import IndexConfigTrait, IndexerContext
+from .index import CoveringIndex
+from ...config import IndexConstants
+from ...exceptions import HyperspaceException
+from ...log.entry import Schema, SchemaField
+from ...utils.resolver import resolve_all
+
+
+class CoveringIndexConfig(IndexConfigTrait):
+    def __init__(self, index_name: str, indexed_columns: List[str],
+                 included_columns: Optional[List[str]] = None):
+        self._name = index_name
+        self.indexed_columns = list(indexed_columns)
+        self.included_columns = list(included_columns or [])
+        self._validate()
+
+    def _validate(self):
+        if not self._name:
+            raise HyperspaceException("Index name cannot be empty")
+        if not self.indexed_columns:
+            raise HyperspaceException("Indexed columns cannot be empty")
+        low_idx = [c.lower() for c in self.indexed_columns]
+        low_inc = [c.lower() for c in self.included_columns]
+        if len(set(low_idx)) != len(low_idx) or \
+                len(set(low_inc)) != len(low_inc):
+            raise HyperspaceException("Duplicate columns in index config")
+        if set(low_idx) & set(low_inc):
+            raise HyperspaceException(
+                "Indexed and included columns must be disjoint")
+
+    @property
+    def index_name(self) -> str:
+        return self._name
+
+    def referenced_columns(self) -> List[str]:
+        return self.indexed_columns + self.included_columns
+
+    def create_index(self, ctx: IndexerContext, df,
+                     properties: Dict[str, str]
+                     ) -> Tuple[CoveringIndex, object]:
+        from ...execution.executor import Executor
+        from ...plan.nodes import Scan
+
+        leaves = df.plan.collect_leaves()
+        if len(leaves) != 1 or not isinstance(leaves[0], Scan):
+            raise HyperspaceException(
+                "createIndex requires a single file-source relation plan")
+        scan: Scan = leaves[0]
+        source_schema = scan.relation.schema
+        indexed = resolve_all(source_schema.field_names(),
+                              self.indexed_columns)
+        included = resolve_all(source_schema.field_names(),
+                               self.included_columns)
+
+        lineage = (properties.get(IndexConstants.LINEAGE_PROPERTY, "false")
+                   .lower() == "true")
+
+        ex = Executor(ctx.session)
+        batch = ex._exec_scan(
+            scan, lineage_tracker=ctx.file_id_tracker if lineage else None)
+        cols = indexed + included
+        if lineage:
+            cols = cols + [IndexConstants.DATA_FILE_NAME_ID_COLUMN]
+        batch = batch.select(cols)
+
+        index_schema = Schema(
+            [f for f in source_schema.fields
+             if f.name.lower() in {c.lower() for c in indexed + included}])
+        if lineage:
+            index_schema = Schema(
+                index_schema.fields
+                + [SchemaField(IndexConstants.DATA_FILE_NAME_ID_COLUMN,
+                               "long", False)])
+
+        num_buckets = ctx.session.conf.num_buckets
+        index = CoveringIndex(indexed, included, index_schema, num_buckets,
+                              dict(properties))
+        return index, batch

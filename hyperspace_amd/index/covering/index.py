@@ -1,0 +1,239 @@
+"""CoveringIndex: bucketed + sorted vertical slice of the source.
+
+The flagship index kind (reference: index/covering/CoveringIndex.scala,
+CoveringIndexTrait.scala).  Index data = indexed + included columns
+[+ lineage column], hash-repartitioned into ``numBuckets`` by Spark-
+compatible Murmur3 of the indexed columns, sorted by the indexed columns
+within each bucket, stored as uncompressed PLAIN Parquet with the
+bucket-id-in-filename contract.
+
+The build pipeline is the device data plane: murmur3 bucketize (K2),
+stable radix sort by (bucket, indexed cols) (K3), segmented write.  With
+torch.distributed initialized, buckets are exchanged across GPUs via RCCL
+all-to-all before the sort (C1, parallel/exchange.py).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Any, Dict, List, Optional, Tuple
+
+import torch
+
+from ..base import Index, IndexerContext
+from ...config import IndexConstants
+from ...exceptions import HyperspaceException
+from ...log.entry import (Schema, register_derived_dataset,
+                          COVERING_INDEX_TYPE)
+from ... import ops
+from ...execution.columnar import ColumnBatch, StringColumn
+from ...sources.parquet_io import (bucket_file_name, write_batch_parquet,
+                                   read_files_batch, bucket_id_of_file)
+
+
+class CoveringIndex(Index):
+    def __init__(self, indexed_columns: List[str],
+                 included_columns: List[str], schema: Schema,
+                 num_buckets: int, properties: Dict[str, str]):
+        self.indexed_columns = list(indexed_columns)
+        self.included_columns = list(included_columns)
+        self.schema = schema
+        self.num_buckets = num_buckets
+        self._properties = dict(properties)
+
+    # -- identity ---------------------------------------------------------
+    @property
+    def kind(self) -> str:
+        return "CoveringIndex"
+
+    @property
+    def kind_abbr(self) -> str:
+        return "CI"
+
+    def indexed_columns_list(self) -> List[str]:
+        return list(self.indexed_columns)
+
+    def referenced_columns(self) -> List[str]:
+        return self.indexed_columns + self.included_columns
+
+    @property
+    def can_handle_deleted_files(self) -> bool:
+        return (self._properties.get(IndexConstants.LINEAGE_PROPERTY, "false")
+                .lower() == "true")
+
+    @property
+    def has_lineage(self) -> bool:
+        return self.can_handle_deleted_files
+
+    def with_new_properties(self, props: Dict[str, str]) -> "CoveringIndex":
+        return CoveringIndex(self.indexed_columns, self.included_columns,
+                             self.schema, self.num_buckets, props)
+
+    # -- json (wire-compatible with the reference) -------------------------
+    def to_json(self) -> Dict[str, Any]:
+        return {
+            "type": COVERING_INDEX_TYPE,
+            "indexedColumns": self.indexed_columns,
+            "includedColumns": self.included_columns,
+            "schema": self.schema.to_json(),
+            "numBuckets": self.num_buckets,
+            "properties": self._properties,
+        }
+
+    @staticmethod
+    def from_json(d: Dict[str, Any]) -> "CoveringIndex":
+        return CoveringIndex(
+            d["indexedColumns"], d["includedColumns"],
+            Schema.from_json(d["schema"]), d["numBuckets"],
+            d.get("properties", {}))
+
+    # -- build data plane ---------------------------------------------------
+    def write(self, ctx: IndexerContext, index_data: ColumnBatch
+              ) -> List[str]:
+        """Bucketize + per-bucket sort + bucketed parquet write (K2+K3).
+
+        With torch.distributed initialized this is the multi-GPU build:
+        each rank holds a source shard; bucket rows are exchanged via RCCL
+        all-to-all over xGMI so rank r owns buckets {b : b % world == r}.
+        """
+        os.makedirs(ctx.index_data_path, exist_ok=True)
+        batch = index_data
+        n = self.num_buckets
+
+        import torch.distributed as dist
+        distributed = dist.is_available() and dist.is_initialized()
+
+        keys = [batch.tensor(c) for c in self.indexed_columns]
+        bucket_ids = ops.murmur3_bucket(keys, n)
+
+        if distributed and dist.get_world_size() > 1:
+            from ...parallel.exchange import exchange_by_bucket
+            batch, bucket_ids = exchange_by_bucket(batch, bucket_ids, n)
+
+        batch, seg = sort_by_bucket_and_keys(
+            batch, bucket_ids, self.indexed_columns, n)
+
+        task_id = dist.get_rank() if distributed else 0
+        return write_bucketed(batch, seg, ctx.index_data_path, n, task_id)
+
+    def optimize(self, ctx: IndexerContext,
+                 files_to_optimize: List[str]) -> List[str]:
+        """Per-bucket compaction: merge this bucket's small files into one
+        (reference: CoveringIndexTrait.optimize + OptimizeAction —
+        files are already bucket-pure, so this is a read + re-sort +
+        single-file rewrite per bucket)."""
+        os.makedirs(ctx.index_data_path, exist_ok=True)
+        by_bucket: Dict[int, List[str]] = {}
+        for p in files_to_optimize:
+            b = bucket_id_of_file(p)
+            if b is None:
+                raise HyperspaceException(f"Not a bucketed index file: {p}")
+            by_bucket.setdefault(b, []).append(p)
+        written: List[str] = []
+        for b, paths in sorted(by_bucket.items()):
+            sub, _ = read_files_batch(sorted(paths))
+            if ctx.session.device.type == "cuda":
+                sub = sub.to(ctx.session.device)
+            perm = _multi_key_sort_perm(sub, self.indexed_columns)
+            sub = sub.gather(perm)
+            out = os.path.join(ctx.index_data_path, bucket_file_name(0, b))
+            write_batch_parquet(sub.to("cpu") if sub.device.type == "cuda"
+                                else sub, out)
+            written.append(out)
+        return written
+
+    def refresh_incremental(self, ctx: IndexerContext,
+                            appended_batch: Optional[ColumnBatch],
+                            deleted_file_ids: List[int],
+                            previous_files: List[str]
+                            ) -> Tuple[List[str], List[str]]:
+        """Index appended data into new files; rewrite old files dropping
+        rows whose lineage id is deleted
+        (reference: CoveringIndexTrait.scala:57-106).
+
+        Returns (new files written, kept previous files).
+        """
+        written: List[str] = []
+        if appended_batch is not None and appended_batch.num_rows:
+            written.extend(self.write(ctx, appended_batch))
+        kept = list(previous_files)
+        if deleted_file_ids:
+            if not self.has_lineage:
+                raise HyperspaceException(
+                    "Index has no lineage column; cannot handle deletes")
+            os.makedirs(ctx.index_data_path, exist_ok=True)
+            ids = torch.tensor(sorted(deleted_file_ids), dtype=torch.int64)
+            kept = []
+            lineage_col = IndexConstants.DATA_FILE_NAME_ID_COLUMN
+            for p in previous_files:
+                b = bucket_id_of_file(p)
+                sub, _ = read_files_batch([p])
+                lineage = sub.tensor(lineage_col)
+                keep_mask = ~ops.isin_sorted(lineage, ids)
+                if bool(keep_mask.all()):
+                    kept.append(p)
+                    continue
+                sub = sub.gather(
+                    torch.nonzero(keep_mask, as_tuple=False).flatten())
+                out = os.path.join(ctx.index_data_path,
+                                   bucket_file_name(1, b or 0))
+                write_batch_parquet(sub, out)
+                written.append(out)
+        return written, kept
+
+    def statistics(self) -> Dict[str, Any]:
+        return {"indexedColumns": self.indexed_columns,
+                "includedColumns": self.included_columns,
+                "numBuckets": self.num_buckets}
+
+
+# ---------------------------------------------------------------------------
+# build helpers (shared with zorder)
+# ---------------------------------------------------------------------------
+
+def _multi_key_sort_perm(batch: ColumnBatch, key_cols: List[str]
+                         ) -> torch.Tensor:
+    """Stable multi-column sort permutation: LSD over columns
+    (least-significant column first)."""
+    n = batch.num_rows
+    dev = batch.device
+    perm = torch.arange(n, dtype=torch.int64, device=dev)
+    for c in reversed(key_cols):
+        keys = ops.normalize_key(batch.tensor(c))[perm]
+        _, perm = ops.sort_pairs(keys, perm)
+    return perm
+
+
+def sort_by_bucket_and_keys(batch: ColumnBatch, bucket_ids: torch.Tensor,
+                            key_cols: List[str], num_buckets: int
+                            ) -> Tuple[ColumnBatch, torch.Tensor]:
+    """Stable sort rows by (bucket, key columns); returns the reordered
+    batch and per-bucket segment offsets (num_buckets+1)."""
+    perm = _multi_key_sort_perm(batch, key_cols)
+    b64 = bucket_ids.to(torch.int64)[perm]
+    _, perm2 = ops.sort_pairs(ops.normalize_key(b64), perm)
+    batch = batch.gather(perm2)
+    counts = torch.bincount(bucket_ids.to(torch.int64).cpu(),
+                            minlength=num_buckets)
+    seg = torch.zeros(num_buckets + 1, dtype=torch.int64)
+    seg[1:] = torch.cumsum(counts, 0)
+    return batch, seg
+
+
+def write_bucketed(batch: ColumnBatch, seg: torch.Tensor, out_dir: str,
+                   num_buckets: int, task_id: int = 0) -> List[str]:
+    """Write per-bucket parquet files honoring the bucket-id filename
+    contract.  Empty buckets produce no file (as in Spark)."""
+    host = batch.to("cpu") if batch.device.type == "cuda" else batch
+    written: List[str] = []
+    for b in range(num_buckets):
+        lo, hi = int(seg[b]), int(seg[b + 1])
+        if hi <= lo:
+            continue
+        out = os.path.join(out_dir, bucket_file_name(task_id, b))
+        write_batch_parquet(host.slice(lo, hi), out)
+        written.append(out)
+    return written
+
+
+register_derived_dataset(COVERING_INDEX_TYPE, CoveringIndex)

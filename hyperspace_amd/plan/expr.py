@@ -1,0 +1,218 @@
+"""Expression tree for filters and join conditions.
+
+The engine's analog of Catalyst expressions — only what the rule layer
+needs: column refs, literals, comparisons, boolean connectives, IN.
+"""
+
+from __future__ import annotations
+
+import re
+from typing import Any, List, Sequence, Set, Union
+
+from ..exceptions import HyperspaceException
+
+
+class Expr:
+    def references(self) -> Set[str]:
+        raise NotImplementedError
+
+    # sugar so users can write col("a") == 1
+    def __and__(self, other):
+        return And(self, _wrap(other))
+
+    def __or__(self, other):
+        return Or(self, _wrap(other))
+
+    def __invert__(self):
+        return Not(self)
+
+
+class Col(Expr):
+    def __init__(self, name: str):
+        self.name = name
+
+    def references(self):
+        return {self.name}
+
+    def __eq__(self, other):  # type: ignore[override]
+        return BinComp("=", self, _wrap(other))
+
+    def __ne__(self, other):  # type: ignore[override]
+        return BinComp("!=", self, _wrap(other))
+
+    def __lt__(self, other):
+        return BinComp("<", self, _wrap(other))
+
+    def __le__(self, other):
+        return BinComp("<=", self, _wrap(other))
+
+    def __gt__(self, other):
+        return BinComp(">", self, _wrap(other))
+
+    def __ge__(self, other):
+        return BinComp(">=", self, _wrap(other))
+
+    def isin(self, values: Sequence[Any]):
+        return In(self, list(values))
+
+    def is_not_null(self):
+        return IsNotNull(self)
+
+    def __hash__(self):
+        return hash(("Col", self.name.lower()))
+
+    def __repr__(self):
+        return self.name
+
+
+class Lit(Expr):
+    def __init__(self, value: Any):
+        self.value = value
+
+    def references(self):
+        return set()
+
+    def __repr__(self):
+        return repr(self.value)
+
+
+class BinComp(Expr):
+    """left <op> right with op in =, !=, <, <=, >, >=."""
+
+    OPS = ("=", "!=", "<", "<=", ">", ">=")
+
+    def __init__(self, op: str, left: Expr, right: Expr):
+        if op not in self.OPS:
+            raise HyperspaceException(f"Bad comparison op {op}")
+        self.op = op
+        self.left = left
+        self.right = right
+
+    def references(self):
+        return self.left.references() | self.right.references()
+
+    def __repr__(self):
+        return f"({self.left!r} {self.op} {self.right!r})"
+
+
+class And(Expr):
+    def __init__(self, left: Expr, right: Expr):
+        self.left, self.right = left, right
+
+    def references(self):
+        return self.left.references() | self.right.references()
+
+    def __repr__(self):
+        return f"({self.left!r} AND {self.right!r})"
+
+
+class Or(Expr):
+    def __init__(self, left: Expr, right: Expr):
+        self.left, self.right = left, right
+
+    def references(self):
+        return self.left.references() | self.right.references()
+
+    def __repr__(self):
+        return f"({self.left!r} OR {self.right!r})"
+
+
+class Not(Expr):
+    def __init__(self, child: Expr):
+        self.child = child
+
+    def references(self):
+        return self.child.references()
+
+    def __repr__(self):
+        return f"(NOT {self.child!r})"
+
+
+class In(Expr):
+    def __init__(self, col: Expr, values: List[Any]):
+        self.col = col
+        self.values = values
+
+    def references(self):
+        return self.col.references()
+
+    def __repr__(self):
+        return f"({self.col!r} IN {self.values!r})"
+
+
+class IsNotNull(Expr):
+    def __init__(self, col: Expr):
+        self.col = col
+
+    def references(self):
+        return self.col.references()
+
+    def __repr__(self):
+        return f"({self.col!r} IS NOT NULL)"
+
+
+def col(name: str) -> Col:
+    return Col(name)
+
+
+def lit(value: Any) -> Lit:
+    return Lit(value)
+
+
+def _wrap(v: Union[Expr, Any]) -> Expr:
+    return v if isinstance(v, Expr) else Lit(v)
+
+
+# ---------------------------------------------------------------------------
+# Helpers used by the rules
+# ---------------------------------------------------------------------------
+
+def split_conjunctive(e: Expr) -> List[Expr]:
+    """Flatten nested ANDs into a predicate list (CNF top level)."""
+    if isinstance(e, And):
+        return split_conjunctive(e.left) + split_conjunctive(e.right)
+    return [e]
+
+
+def extract_equi_join_keys(e: Expr) -> List[tuple]:
+    """For a join condition that is a CNF of Col = Col, return
+    [(left_name, right_name)] or [] if not an equi-join
+    (reference: JoinIndexRule CNF check, index/covering/JoinIndexRule.scala).
+    """
+    pairs = []
+    for p in split_conjunctive(e):
+        if isinstance(p, BinComp) and p.op == "=" and \
+                isinstance(p.left, Col) and isinstance(p.right, Col):
+            pairs.append((p.left.name, p.right.name))
+        else:
+            return []
+    return pairs
+
+
+_PRED_RE = re.compile(
+    r"^\s*([A-Za-z_][A-Za-z0-9_]*)\s*(=|==|!=|<=|>=|<|>)\s*(.+?)\s*$")
+
+
+def parse_predicate(s: str) -> Expr:
+    """Tiny predicate parser: 'col <op> literal' [AND ...]."""
+    parts = re.split(r"\s+(?:AND|and)\s+", s)
+    exprs: List[Expr] = []
+    for part in parts:
+        m = _PRED_RE.match(part)
+        if not m:
+            raise HyperspaceException(f"Cannot parse predicate: {part}")
+        name, op, val = m.groups()
+        if op == "==":
+            op = "="
+        if val.startswith("'") or val.startswith('"'):
+            value: Any = val.strip("'\"")
+        else:
+            try:
+                value = int(val)
+            except ValueError:
+                value = float(val)
+        exprs.append(BinComp(op, Col(name), Lit(value)))
+    out = exprs[0]
+    for e in exprs[1:]:
+        out = And(out, e)
+    return out

@@ -1,0 +1,76 @@
+"""Pluggable source-provider interfaces.
+
+Reference: index/sources/interfaces.scala:43-277 (FileBasedRelation /
+FileBasedSourceProvider / FileBasedRelationMetadata).
+"""
+
+from __future__ import annotations
+
+from abc import ABC, abstractmethod
+from typing import Dict, List, Optional
+
+from ..log.entry import FileInfo, Relation, Schema
+
+
+class FileBasedRelation(ABC):
+    """A file-based data source (immutable file set + schema)."""
+
+    @property
+    @abstractmethod
+    def root_paths(self) -> List[str]: ...
+
+    @property
+    @abstractmethod
+    def schema(self) -> Schema: ...
+
+    @property
+    @abstractmethod
+    def file_format(self) -> str: ...
+
+    @property
+    def options(self) -> Dict[str, str]:
+        return {}
+
+    @abstractmethod
+    def all_files(self) -> List[FileInfo]:
+        """Leaf data files as (abs path, size, mtime, UNKNOWN id)."""
+
+    @abstractmethod
+    def signature(self) -> str:
+        """Content fingerprint of the relation."""
+
+    def describe(self) -> str:
+        return f"{self.file_format}:{','.join(self.root_paths)}"
+
+    def create_relation_metadata(self, file_id_tracker) -> Relation:
+        """Build the log-entry Relation (with tracked file ids)."""
+        from ..log.entry import Content, Hdfs
+        files = []
+        for f in self.all_files():
+            fid = file_id_tracker.add_file(f.name, f.size, f.modifiedTime)
+            files.append((f.name, f.size, f.modifiedTime, fid))
+        return Relation(
+            rootPaths=list(self.root_paths),
+            data=Hdfs(Content.from_leaf_files(sorted(files))),
+            dataSchema=self.schema,
+            fileFormat=self.file_format,
+            options=dict(self.options))
+
+    def refreshed(self) -> "FileBasedRelation":
+        """Re-list the current state of the source (for refresh actions)."""
+        return self
+
+
+class FileBasedSourceProvider(ABC):
+    """Claims relations it supports and reconstructs them from metadata."""
+
+    @abstractmethod
+    def supports(self, relation: FileBasedRelation) -> bool: ...
+
+    @abstractmethod
+    def from_metadata(self, metadata: Relation) -> Optional[FileBasedRelation]:
+        """Rebuild a relation from a logged Relation (refresh path)."""
+
+    def enrich_index_properties(self, properties: Dict[str, str]
+                                ) -> Dict[str, str]:
+        return properties

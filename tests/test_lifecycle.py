@@ -1,0 +1,197 @@
+"""Index lifecycle (actions state machine) integration tests on CPU.
+
+Mirrors the reference's IndexManagerTest / RefreshIndexTest behaviors.
+"""
+
+import os
+
+import numpy as np
+import pyarrow as pa
+import pyarrow.parquet as pq
+import pytest
+
+import hyperspace_amd as hs
+from hyperspace_amd.exceptions import HyperspaceException
+from hyperspace_amd.log import IndexLogManager, States
+from hyperspace_amd.sources.parquet_io import bucket_id_of_file
+from hyperspace_amd.telemetry import (CreateActionEvent, DeleteActionEvent,
+                                      RecordingEventLogger)
+
+N_ROWS = 5000
+
+
+@pytest.fixture
+def env(tmp_path, monkeypatch):
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "indexes"))
+    data_dir = tmp_path / "data"
+    data_dir.mkdir()
+    rng = np.random.default_rng(7)
+    for i in range(3):
+        t = pa.table({
+            "key": rng.integers(0, 500, N_ROWS),
+            "val": rng.random(N_ROWS),
+        })
+        pq.write_table(t, str(data_dir / f"part-{i}.parquet"))
+    session = hs.HyperspaceSession(device="cpu")
+    session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 10)
+    session.event_logger = RecordingEventLogger()
+    h = hs.Hyperspace(session)
+    df = session.read_parquet(str(data_dir))
+    return session, h, df, data_dir, rng
+
+
+def _log_states(tmp_path_root, name):
+    mgr = IndexLogManager(os.path.join(tmp_path_root, name))
+    latest = mgr.get_latest_id()
+    return [mgr.get_log(i).state for i in range(latest + 1)]
+
+
+def test_create_lifecycle(env, tmp_path):
+    session, h, df, _, _ = env
+    h.create_index(df, hs.CoveringIndexConfig("ix", ["key"], ["val"]))
+    states = _log_states(str(tmp_path / "indexes"), "ix")
+    assert states == [States.CREATING, States.ACTIVE]
+    # bucket filename contract
+    entry = session.index_manager().get_index("ix")
+    for f in entry.content.os_files():
+        assert bucket_id_of_file(f) is not None
+    # events
+    evs = [e for e in session.event_logger.events
+           if isinstance(e, CreateActionEvent)]
+    assert len(evs) == 1
+
+
+def test_create_duplicate_fails(env):
+    _, h, df, _, _ = env
+    h.create_index(df, hs.CoveringIndexConfig("ix", ["key"], ["val"]))
+    with pytest.raises(HyperspaceException, match="already exists"):
+        h.create_index(df, hs.CoveringIndexConfig("ix", ["key"], ["val"]))
+
+
+def test_delete_restore_vacuum(env, tmp_path):
+    session, h, df, _, _ = env
+    h.create_index(df, hs.CoveringIndexConfig("ix", ["key"], ["val"]))
+    h.delete_index("ix")
+    assert session.index_manager().get_index("ix").state == States.DELETED
+    h.restore_index("ix")
+    assert session.index_manager().get_index("ix").state == States.ACTIVE
+    h.delete_index("ix")
+    h.vacuum_index("ix")
+    entry = session.index_manager().get_index("ix")
+    assert entry.state == States.DOESNOTEXIST
+    # data files are gone
+    idx_dir = tmp_path / "indexes" / "ix"
+    remaining = [p for p in os.listdir(idx_dir) if p != "_hyperspace_log"]
+    assert remaining == []
+    assert any(isinstance(e, DeleteActionEvent)
+               for e in session.event_logger.events)
+
+
+def test_delete_requires_active(env):
+    _, h, df, _, _ = env
+    h.create_index(df, hs.CoveringIndexConfig("ix", ["key"], ["val"]))
+    h.delete_index("ix")
+    with pytest.raises(HyperspaceException):
+        h.delete_index("ix")
+
+
+def test_refresh_full_new_version(env, tmp_path):
+    session, h, df, data_dir, rng = env
+    h.create_index(df, hs.CoveringIndexConfig("ix", ["key"], ["val"]))
+    # append a source file
+    t = pa.table({"key": rng.integers(0, 500, 1000),
+                  "val": rng.random(1000)})
+    pq.write_table(t, str(data_dir / "part-new.parquet"))
+    h.refresh_index("ix", "full")
+    entry = session.index_manager().get_index("ix")
+    assert entry.state == States.ACTIVE
+    assert all("v__=1" in f for f in entry.content.os_files())
+    assert len(entry.source_file_infos()) == 4
+
+
+def test_refresh_no_changes_is_noop(env):
+    session, h, df, _, _ = env
+    h.create_index(df, hs.CoveringIndexConfig("ix", ["key"], ["val"]))
+    h.refresh_index("ix", "full")  # no source change
+    entry = session.index_manager().get_index("ix")
+    assert entry.state == States.ACTIVE
+    assert all("v__=0" in f for f in entry.content.os_files())
+
+
+def test_refresh_incremental_appends(env, tmp_path):
+    session, h, df, data_dir, rng = env
+    h.create_index(df, hs.CoveringIndexConfig("ix", ["key"], ["val"]))
+    old_entry = session.index_manager().get_index("ix")
+    old_files = set(old_entry.content.os_files())
+    t = pa.table({"key": rng.integers(0, 500, 1000),
+                  "val": rng.random(1000)})
+    pq.write_table(t, str(data_dir / "part-new.parquet"))
+    h.refresh_index("ix", "incremental")
+    entry = session.index_manager().get_index("ix")
+    new_files = set(entry.content.os_files())
+    assert old_files < new_files  # old files kept, new files added
+    assert any("v__=1" in f for f in new_files - old_files)
+    assert len(entry.source_file_infos()) == 4
+
+
+def test_refresh_quick_records_update(env, data_dir=None):
+    session, h, df, data_dir, rng = env
+    h.create_index(df, hs.CoveringIndexConfig("ix", ["key"], ["val"]))
+    t = pa.table({"key": rng.integers(0, 500, 1000),
+                  "val": rng.random(1000)})
+    pq.write_table(t, str(data_dir / "part-new.parquet"))
+    h.refresh_index("ix", "quick")
+    entry = session.index_manager().get_index("ix")
+    assert len(entry.appended_files()) == 1
+    # index data untouched
+    assert all("v__=0" in f for f in entry.content.os_files())
+
+
+def test_optimize_compacts_buckets(env):
+    session, h, df, data_dir, rng = env
+    h.create_index(df, hs.CoveringIndexConfig("ix", ["key"], ["val"]))
+    # create second batch of files per bucket via incremental refresh
+    t = pa.table({"key": rng.integers(0, 500, 4000),
+                  "val": rng.random(4000)})
+    pq.write_table(t, str(data_dir / "part-new.parquet"))
+    h.refresh_index("ix", "incremental")
+    before = session.index_manager().get_index("ix")
+    n_before = len(before.content.os_files())
+    h.optimize_index("ix", "quick")
+    after = session.index_manager().get_index("ix")
+    n_after = len(after.content.os_files())
+    assert n_after < n_before
+    # every bucket now has exactly one file
+    buckets = [bucket_id_of_file(f) for f in after.content.os_files()]
+    assert len(buckets) == len(set(buckets))
+
+
+def test_optimize_noop_when_single_file_buckets(env):
+    session, h, df, _, _ = env
+    h.create_index(df, hs.CoveringIndexConfig("ix", ["key"], ["val"]))
+    h.optimize_index("ix", "quick")  # single file per bucket -> no-op
+    entry = session.index_manager().get_index("ix")
+    assert entry.state == States.ACTIVE
+
+
+def test_cancel_from_transient(env, tmp_path):
+    session, h, df, _, _ = env
+    h.create_index(df, hs.CoveringIndexConfig("ix", ["key"], ["val"]))
+    # manufacture a stuck transient state
+    mgr = IndexLogManager(str(tmp_path / "indexes" / "ix"))
+    entry = mgr.get_latest_log()
+    entry.state = States.REFRESHING
+    assert mgr.write_log(mgr.get_latest_id() + 1, entry)
+    h.cancel("ix")
+    latest = mgr.get_latest_log()
+    assert latest.state == States.ACTIVE
+
+
+def test_index_stats(env):
+    session, h, df, _, _ = env
+    h.create_index(df, hs.CoveringIndexConfig("ix", ["key"], ["val"]))
+    stats = h.index("ix")
+    assert stats["numBuckets"] == 10
+    assert stats["numSourceFiles"] == 3
+    assert stats["sizeOfIndexInBytes"] > 0
+    assert [d["name"] for d in h.indexes()] == ["ix"]
