@@ -206,7 +206,10 @@ class ColumnBatch:
                 if np_arr.dtype == np.dtype("datetime64[us]") or \
                         np_arr.dtype.kind == "M":
                     np_arr = np_arr.astype("int64")
-                cols[name] = torch.from_numpy(np.ascontiguousarray(np_arr))
+                np_arr = np.ascontiguousarray(np_arr)
+                if not np_arr.flags.writeable:
+                    np_arr = np_arr.copy()
+                cols[name] = torch.from_numpy(np_arr)
         return ColumnBatch(cols)
 
     def to_arrow(self):

@@ -26,7 +26,7 @@ from ..exceptions import HyperspaceException
 from ..plan.expr import (And, BinComp, Col, Expr, In, IsNotNull, Lit, Not,
                          Or, extract_equi_join_keys)
 from ..plan.nodes import (BucketUnionNode, Filter, IndexScan, Join,
-                          LogicalPlan, Project, Scan)
+                          LogicalPlan, Project, Scan, UnionNode)
 from ..sources.parquet_io import read_files_batch, bucket_id_of_file
 
 
@@ -83,6 +83,13 @@ class Executor:
             return self._exec_join(plan), None
         if isinstance(plan, BucketUnionNode):
             return self._exec_bucket_union(plan)
+        if isinstance(plan, UnionNode):
+            self.stats.record("Union")
+            parts = [self._exec(c)[0] for c in plan.children]
+            cols = plan.output_columns()
+            return ColumnBatch.concat(
+                [p.select(cols) for p in parts if p.num_rows > 0]
+                or [parts[0].select(cols)]), None
         raise HyperspaceException(f"Cannot execute {type(plan).__name__}")
 
     # ------------------------------------------------------------------

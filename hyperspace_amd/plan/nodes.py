@@ -177,6 +177,24 @@ class IndexScan(LogicalPlan):
                 + (", " + ",".join(flags) if flags else "") + ")")
 
 
+class UnionNode(LogicalPlan):
+    """Plain union (Hybrid Scan merge for the filter rule, where bucket
+    alignment is unnecessary — reference CoveringIndexRuleUtils uses
+    Union for FilterIndexRule, BucketUnion for JoinIndexRule)."""
+
+    def __init__(self, children: List[LogicalPlan]):
+        super().__init__(children)
+
+    def output_columns(self):
+        return self.children[0].output_columns()
+
+    def with_children(self, children):
+        return UnionNode(children)
+
+    def _node_str(self):
+        return "Union"
+
+
 class BucketUnionNode(LogicalPlan):
     """Partition-aligned union of index data and on-the-fly-bucketed
     appended source data (Hybrid Scan merge, K5/K6)."""

@@ -72,14 +72,14 @@ class ParquetRelation(FileBasedRelation):
         return self._schema
 
     def all_files(self) -> List[FileInfo]:
-        if self._files is None:
-            infos = []
-            for p in list_data_files(self._root_paths, suffix=".parquet"):
-                st = os.stat(p)
-                infos.append(FileInfo(p, st.st_size,
-                                      int(st.st_mtime * 1000)))
-            self._files = infos
-        return self._files
+        # re-listed on every call (like Spark's InMemoryFileIndex refresh
+        # semantics across queries): a stale cache would let a changed
+        # source pass the signature check
+        infos = []
+        for p in list_data_files(self._root_paths, suffix=".parquet"):
+            st = os.stat(p)
+            infos.append(FileInfo(p, st.st_size, int(st.st_mtime * 1000)))
+        return infos
 
     def signature(self) -> str:
         """md5 over sorted (path,size,mtime) — matches the reference's

@@ -1,0 +1,189 @@
+"""E2E query-acceleration tests (the reference's E2EHyperspaceRulesTest).
+
+Covers: filter rewrite to index-only scan, bucket pruning, join rewrite to
+the zero-shuffle co-bucketed merge join, result equivalence with the
+unindexed run, explain/whyNot output.
+"""
+
+import numpy as np
+import pyarrow as pa
+import pyarrow.parquet as pq
+import pytest
+import torch
+
+import hyperspace_amd as hs
+from hyperspace_amd.execution.executor import Executor
+from hyperspace_amd.plan.nodes import IndexScan, Join, Scan
+
+N = 20000
+
+
+@pytest.fixture
+def env(tmp_path, monkeypatch):
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "indexes"))
+    rng = np.random.default_rng(3)
+    left_dir = tmp_path / "left"
+    right_dir = tmp_path / "right"
+    left_dir.mkdir()
+    right_dir.mkdir()
+    for i in range(2):
+        t = pa.table({
+            "orderkey": rng.integers(0, 5000, N),
+            "qty": rng.integers(1, 50, N),
+            "price": rng.random(N) * 100,
+        })
+        pq.write_table(t, str(left_dir / f"part-{i}.parquet"))
+    t = pa.table({
+        "orderkey": np.arange(5000, dtype=np.int64),
+        "status": rng.integers(0, 3, 5000),
+    })
+    pq.write_table(t, str(right_dir / "part-0.parquet"))
+
+    session = hs.HyperspaceSession(device="cpu")
+    session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 16)
+    h = hs.Hyperspace(session)
+    left = session.read_parquet(str(left_dir))
+    right = session.read_parquet(str(right_dir))
+    return session, h, left, right
+
+
+def _sorted_rows(batch, cols):
+    arrs = batch.to_numpy()
+    rows = sorted(zip(*[arrs[c].tolist() for c in cols]))
+    return rows
+
+
+def test_filter_rewrite_and_equivalence(env):
+    session, h, left, _ = env
+    h.create_index(left, hs.CoveringIndexConfig("fidx", ["qty"], ["price"]))
+    q = left.filter("qty = 7").select("qty", "price")
+
+    baseline = q.collect()  # rules disabled by default
+    session.enable_hyperspace()
+    plan = q.optimized_plan()
+    leaves = plan.collect_leaves()
+    assert any(isinstance(l, IndexScan) for l in leaves), plan.pretty()
+    accel = q.collect()
+    assert _sorted_rows(accel, ["qty", "price"]) == \
+        _sorted_rows(baseline, ["qty", "price"])
+
+
+def test_filter_not_applied_without_first_indexed_col(env):
+    session, h, left, _ = env
+    h.create_index(left, hs.CoveringIndexConfig("fidx", ["qty"], ["price"]))
+    session.enable_hyperspace()
+    # filter on price (an included, not indexed, column) -> no rewrite
+    plan = left.filter("price >= 50.0").select("qty", "price") \
+        .optimized_plan()
+    assert not any(isinstance(l, IndexScan)
+                   for l in plan.collect_leaves())
+
+
+def test_filter_bucket_pruning(env):
+    session, h, left, _ = env
+    session.conf.set(hs.IndexConstants.INDEX_FILTER_RULE_USE_BUCKET_SPEC,
+                     True)
+    h.create_index(left, hs.CoveringIndexConfig("fidx", ["qty"], ["price"]))
+    session.enable_hyperspace()
+    q = left.filter("qty = 7").select("qty", "price")
+    ex = Executor(session)
+    out = ex.execute(q.optimized_plan())
+    assert ex.stats.bucket_pruned_files > 0
+    # correctness vs unindexed
+    session.disable_hyperspace()
+    base = q.collect()
+    assert _sorted_rows(out, ["qty", "price"]) == \
+        _sorted_rows(base, ["qty", "price"])
+
+
+def test_join_rewrite_zero_shuffle(env):
+    session, h, left, right = env
+    h.create_index(left, hs.CoveringIndexConfig(
+        "lidx", ["orderkey"], ["qty"]))
+    h.create_index(right, hs.CoveringIndexConfig(
+        "ridx", ["orderkey"], ["status"]))
+    q = left.select("orderkey", "qty").join(
+        right.select("orderkey", "status"), on="orderkey")
+
+    baseline = q.collect()
+    session.enable_hyperspace()
+    plan = q.optimized_plan()
+    # both leaves are bucketed index scans
+    leaves = plan.collect_leaves()
+    index_leaves = [l for l in leaves if isinstance(l, IndexScan)]
+    assert len(index_leaves) == 2, plan.pretty()
+    assert all(l.use_bucket_spec for l in index_leaves)
+
+    ex = Executor(session)
+    accel = ex.execute(plan)
+    assert ex.stats.shuffles == 0            # the whole point
+    assert ex.stats.merge_joins == 1
+    assert accel.num_rows == baseline.num_rows
+    cols = ["orderkey", "qty", "status"]
+    assert _sorted_rows(accel, cols) == _sorted_rows(baseline, cols)
+
+
+def test_join_no_rewrite_when_keys_not_indexed(env):
+    session, h, left, right = env
+    h.create_index(left, hs.CoveringIndexConfig("lidx", ["qty"], ["price"]))
+    session.enable_hyperspace()
+    plan = left.select("orderkey", "qty").join(
+        right.select("orderkey", "status"), on="orderkey").optimized_plan()
+    assert not any(isinstance(l, IndexScan)
+                   for l in plan.collect_leaves())
+
+
+def test_source_change_blocks_index(env, tmp_path):
+    session, h, left, _ = env
+    h.create_index(left, hs.CoveringIndexConfig("fidx", ["qty"], ["price"]))
+    # append a file -> signature mismatch, hybrid scan off -> no rewrite
+    rng = np.random.default_rng(9)
+    t = pa.table({"orderkey": rng.integers(0, 5000, 100),
+                  "qty": rng.integers(1, 50, 100),
+                  "price": rng.random(100)})
+    pq.write_table(t, str(tmp_path / "left" / "part-9.parquet"))
+    session.enable_hyperspace()
+    plan = left.filter("qty = 7").select("qty", "price").optimized_plan()
+    assert not any(isinstance(l, IndexScan)
+                   for l in plan.collect_leaves())
+
+
+def test_explain_output(env):
+    session, h, left, _ = env
+    h.create_index(left, hs.CoveringIndexConfig("fidx", ["qty"], ["price"]))
+    q = left.filter("qty = 7").select("qty", "price")
+    out = h.explain(q, verbose=True)
+    assert "Plan with indexes" in out
+    assert "fidx" in out
+    assert "IndexScan" in out
+
+
+def test_why_not_output(env):
+    session, h, left, right = env
+    h.create_index(left, hs.CoveringIndexConfig("fidx", ["qty"], ["price"]))
+    # query that cannot use the index
+    q = left.filter("price >= 50.0").select("price")
+    out = h.why_not(q, extended=True)
+    assert "fidx" in out
+    assert "NO_FIRST_INDEXED_COL_COND" in out
+    # and one that can
+    q2 = left.filter("qty = 7").select("qty", "price")
+    out2 = h.why_not(q2)
+    assert "APPLIED" in out2
+
+
+def test_score_prefers_join_over_filter(env):
+    # when both a filter index and join indexes could apply to a join
+    # query's sides, the join pair (70x2) beats per-side filter (50)
+    session, h, left, right = env
+    h.create_index(left, hs.CoveringIndexConfig(
+        "lidx", ["orderkey"], ["qty"]))
+    h.create_index(right, hs.CoveringIndexConfig(
+        "ridx", ["orderkey"], ["status"]))
+    session.enable_hyperspace()
+    q = left.select("orderkey", "qty").join(
+        right.select("orderkey", "status"), on="orderkey")
+    plan = q.optimized_plan()
+    scans = [l for l in plan.collect_leaves() if isinstance(l, IndexScan)]
+    assert {s.entry.name for s in scans} == {"lidx", "ridx"}
+    assert all(s.use_bucket_spec for s in scans)
