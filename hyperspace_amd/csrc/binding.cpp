@@ -1,0 +1,291 @@
+// Torch extension binding for the CDNA4 kernel suite (kernels.hip).
+// Written directly against the ROCm torch C++ surface (c10::hip) — no
+// CUDA-compat naming, no hipify.
+
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+
+#include <tuple>
+#include <vector>
+
+#include "kernels/kernels.h"
+
+namespace {
+
+hipStream_t current_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_cuda(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be a device tensor");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+int normalize_dtype_code(torch::ScalarType t) {
+  switch (t) {
+    case torch::kInt64: return 0;
+    case torch::kInt32: return 1;
+    case torch::kFloat64: return 2;
+    case torch::kFloat32: return 3;
+    case torch::kInt16: return 4;
+    case torch::kInt8: return 5;
+    case torch::kBool: return 6;
+    default:
+      TORCH_CHECK(false, "unsupported sort-key dtype");
+  }
+}
+
+torch::Tensor murmur3_bucket(std::vector<torch::Tensor> keys,
+                             int64_t num_buckets) {
+  TORCH_CHECK(!keys.empty(), "need at least one key column");
+  auto n = keys[0].numel();
+  auto dev = keys[0].device();
+  auto h = torch::empty({n}, torch::dtype(torch::kInt32).device(dev));
+  auto stream = current_stream();
+  bool first = true;
+  for (auto& k : keys) {
+    check_cuda(k, "key");
+    TORCH_CHECK(k.numel() == n, "key column length mismatch");
+    torch::Tensor col = k;
+    int kind;
+    switch (col.scalar_type()) {
+      case torch::kInt64:
+      case torch::kFloat64:
+        kind = 1;
+        break;
+      case torch::kInt32:
+      case torch::kFloat32:
+        kind = 0;
+        break;
+      case torch::kInt16:
+      case torch::kInt8:
+      case torch::kBool:
+        col = col.to(torch::kInt32);
+        kind = 0;
+        break;
+      default:
+        TORCH_CHECK(false, "unsupported hash-key dtype");
+    }
+    hsk::murmur3_column(col.data_ptr(), kind,
+                        (uint32_t*)h.data_ptr<int32_t>(), n, first,
+                        /*seed=*/42u, stream);
+    first = false;
+  }
+  auto out = torch::empty({n}, torch::dtype(torch::kInt32).device(dev));
+  hsk::pmod_buckets((const uint32_t*)h.data_ptr<int32_t>(),
+                    out.data_ptr<int32_t>(), n, (int32_t)num_buckets,
+                    stream);
+  return out;
+}
+
+torch::Tensor normalize_key(torch::Tensor vals) {
+  check_cuda(vals, "vals");
+  auto n = vals.numel();
+  auto out = torch::empty(
+      {n}, torch::dtype(torch::kInt64).device(vals.device()));
+  hsk::normalize_key(vals.data_ptr(),
+                     normalize_dtype_code(vals.scalar_type()),
+                     (uint64_t*)out.data_ptr<int64_t>(), n,
+                     current_stream());
+  return out;
+}
+
+std::tuple<torch::Tensor, torch::Tensor> radix_sort_pairs(
+    torch::Tensor keys, torch::Tensor payload) {
+  check_cuda(keys, "keys");
+  check_cuda(payload, "payload");
+  TORCH_CHECK(keys.scalar_type() == torch::kInt64, "keys must be int64");
+  TORCH_CHECK(payload.scalar_type() == torch::kInt64,
+              "payload must be int64");
+  auto n = keys.numel();
+  auto out_keys = keys.clone();
+  auto out_payload = payload.clone();
+  if (n <= 1) return {out_keys, out_payload};
+  auto tmp_keys = torch::empty_like(keys);
+  auto tmp_payload = torch::empty_like(payload);
+  auto hist = torch::empty({hsk::radix_sort_hist_size(n)},
+                           torch::dtype(torch::kInt32).device(keys.device()));
+  auto mask = torch::empty({1},
+                           torch::dtype(torch::kInt64).device(keys.device()));
+  hsk::radix_sort_pairs((uint64_t*)out_keys.data_ptr<int64_t>(),
+                        out_payload.data_ptr<int64_t>(),
+                        (uint64_t*)tmp_keys.data_ptr<int64_t>(),
+                        tmp_payload.data_ptr<int64_t>(),
+                        (uint32_t*)hist.data_ptr<int32_t>(),
+                        (uint64_t*)mask.data_ptr<int64_t>(), n,
+                        current_stream());
+  return {out_keys, out_payload};
+}
+
+std::tuple<torch::Tensor, torch::Tensor> merge_join(torch::Tensor lkeys,
+                                                    torch::Tensor rkeys,
+                                                    torch::Tensor lseg,
+                                                    torch::Tensor rseg) {
+  check_cuda(lkeys, "lkeys");
+  check_cuda(rkeys, "rkeys");
+  auto dev = lkeys.device();
+  auto opts = torch::dtype(torch::kInt64).device(dev);
+  // segment offsets may arrive on CPU
+  auto lseg_d = lseg.to(dev, torch::kInt64).contiguous();
+  auto rseg_d = rseg.to(dev, torch::kInt64).contiguous();
+  int64_t n_left = lkeys.numel();
+  int64_t n_seg = lseg_d.numel() - 1;
+  auto stream = current_stream();
+  if (n_left == 0 || rkeys.numel() == 0) {
+    return {torch::empty({0}, opts), torch::empty({0}, opts)};
+  }
+  auto counts = torch::empty({n_left}, opts);
+  auto starts = torch::empty({n_left}, opts);
+  auto seg_of = torch::empty({n_left}, opts);
+  hsk::merge_join_count((const uint64_t*)lkeys.data_ptr<int64_t>(),
+                        (const uint64_t*)rkeys.data_ptr<int64_t>(),
+                        lseg_d.data_ptr<int64_t>(),
+                        rseg_d.data_ptr<int64_t>(), n_left, n_seg,
+                        counts.data_ptr<int64_t>(),
+                        starts.data_ptr<int64_t>(),
+                        seg_of.data_ptr<int64_t>(), stream);
+  auto offsets = torch::empty({n_left}, opts);
+  auto total = torch::empty({1}, opts);
+  hsk::exclusive_scan_i64(counts.data_ptr<int64_t>(),
+                          offsets.data_ptr<int64_t>(), n_left,
+                          total.data_ptr<int64_t>(), stream);
+  int64_t n_out = total.cpu().item<int64_t>();
+  auto out_l = torch::empty({n_out}, opts);
+  auto out_r = torch::empty({n_out}, opts);
+  if (n_out > 0) {
+    hsk::merge_join_emit(offsets.data_ptr<int64_t>(),
+                         counts.data_ptr<int64_t>(),
+                         starts.data_ptr<int64_t>(),
+                         seg_of.data_ptr<int64_t>(),
+                         rseg_d.data_ptr<int64_t>(), n_left,
+                         out_l.data_ptr<int64_t>(),
+                         out_r.data_ptr<int64_t>(), stream);
+  }
+  return {out_l, out_r};
+}
+
+torch::Tensor isin_sorted(torch::Tensor vals, torch::Tensor sorted_set) {
+  check_cuda(vals, "vals");
+  auto n = vals.numel();
+  auto out = torch::empty(
+      {n}, torch::dtype(torch::kBool).device(vals.device()));
+  if (sorted_set.numel() == 0) {
+    out.zero_();
+    return out;
+  }
+  auto set_d = sorted_set.to(vals.device(), torch::kInt64).contiguous();
+  hsk::isin_sorted(vals.data_ptr<int64_t>(), n, set_d.data_ptr<int64_t>(),
+                   set_d.numel(), out.data_ptr<bool>(), current_stream());
+  return out;
+}
+
+std::tuple<torch::Tensor, torch::Tensor> segmented_minmax(
+    torch::Tensor vals, torch::Tensor seg_off) {
+  check_cuda(vals, "vals");
+  auto seg_d = seg_off.to(vals.device(), torch::kInt64).contiguous();
+  int64_t n_seg = seg_d.numel() - 1;
+  auto opts = torch::dtype(torch::kInt64).device(vals.device());
+  auto mins = torch::empty({n_seg}, opts);
+  auto maxs = torch::empty({n_seg}, opts);
+  hsk::segmented_minmax(vals.data_ptr<int64_t>(), seg_d.data_ptr<int64_t>(),
+                        n_seg, mins.data_ptr<int64_t>(),
+                        maxs.data_ptr<int64_t>(), current_stream());
+  return {mins, maxs};
+}
+
+torch::Tensor bloom_build(torch::Tensor vals, int64_t m_bits, int64_t k) {
+  check_cuda(vals, "vals");
+  int64_t words = (m_bits + 63) / 64;
+  auto out = torch::zeros(
+      {words}, torch::dtype(torch::kInt64).device(vals.device()));
+  hsk::bloom_build(vals.data_ptr<int64_t>(), vals.numel(),
+                   (uint64_t*)out.data_ptr<int64_t>(), m_bits, (int)k,
+                   current_stream());
+  return out;
+}
+
+torch::Tensor bloom_probe(torch::Tensor vals, torch::Tensor words,
+                          int64_t m_bits, int64_t k) {
+  check_cuda(vals, "vals");
+  auto words_d = words.to(vals.device(), torch::kInt64).contiguous();
+  auto out = torch::empty(
+      {vals.numel()}, torch::dtype(torch::kBool).device(vals.device()));
+  hsk::bloom_probe(vals.data_ptr<int64_t>(), vals.numel(),
+                   (const uint64_t*)words_d.data_ptr<int64_t>(), m_bits,
+                   (int)k, out.data_ptr<bool>(), current_stream());
+  return out;
+}
+
+torch::Tensor zorder_key(std::vector<torch::Tensor> cols,
+                         int64_t bits_per_col) {
+  TORCH_CHECK(!cols.empty() && cols.size() <= 8, "1..8 zorder columns");
+  auto n = cols[0].numel();
+  std::vector<const uint64_t*> ptrs;
+  for (auto& c : cols) {
+    check_cuda(c, "zorder col");
+    ptrs.push_back((const uint64_t*)c.data_ptr<int64_t>());
+  }
+  auto out = torch::empty(
+      {n}, torch::dtype(torch::kInt64).device(cols[0].device()));
+  hsk::zorder_key(ptrs.data(), (int)ptrs.size(), (int)bits_per_col, n,
+                  (uint64_t*)out.data_ptr<int64_t>(), current_stream());
+  return out;
+}
+
+torch::Tensor gather_rows(torch::Tensor values, torch::Tensor idx) {
+  check_cuda(values, "values");
+  check_cuda(idx, "idx");
+  TORCH_CHECK(idx.scalar_type() == torch::kInt64, "idx must be int64");
+  auto out = torch::empty({idx.numel()}, values.options());
+  hsk::gather(values.data_ptr(), idx.data_ptr<int64_t>(), out.data_ptr(),
+              idx.numel(), (int)values.element_size(), current_stream());
+  return out;
+}
+
+}  // namespace
+
+// out-of-namespace impl so the two-phase select is a single public symbol
+torch::Tensor select_range_u64_pub(torch::Tensor keys, int64_t lo,
+                                   int64_t hi, bool lo_incl, bool hi_incl) {
+  TORCH_CHECK(keys.is_cuda() && keys.is_contiguous(), "keys");
+  auto n = keys.numel();
+  auto opts = torch::dtype(torch::kInt64).device(keys.device());
+  if (n == 0) return torch::empty({0}, opts);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  int64_t nb = hsk::select_num_blocks(n);
+  auto bcounts = torch::empty({nb}, opts);
+  auto total = torch::empty({1}, opts);
+  // lo/hi arrive already u64-normalized (int64-encoded); the kernel
+  // compares raw u64 bits, so reinterpret without remapping
+  uint64_t ulo = (uint64_t)lo;
+  uint64_t uhi = (uint64_t)hi;
+  // phase 1: count + scan
+  hsk::select_range_count((const uint64_t*)keys.data_ptr<int64_t>(), n, ulo,
+                          uhi, lo_incl, hi_incl,
+                          bcounts.data_ptr<int64_t>(),
+                          total.data_ptr<int64_t>(), stream);
+  int64_t n_out = total.cpu().item<int64_t>();
+  auto out = torch::empty({n_out}, opts);
+  if (n_out > 0) {
+    hsk::select_range_emit((const uint64_t*)keys.data_ptr<int64_t>(), n, ulo,
+                           uhi, lo_incl, hi_incl,
+                           bcounts.data_ptr<int64_t>(),
+                           out.data_ptr<int64_t>(), stream);
+  }
+  return out;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("murmur3_bucket", &murmur3_bucket, "Spark-compatible murmur3 bucket");
+  m.def("normalize_key", &normalize_key, "order-preserving u64 key");
+  m.def("radix_sort_pairs", &radix_sort_pairs, "stable LSD radix sort");
+  m.def("merge_join", &merge_join, "segmented sorted merge join");
+  m.def("select_range_u64", &select_range_u64_pub, "range filter compaction");
+  m.def("isin_sorted", &isin_sorted, "sorted-set membership");
+  m.def("segmented_minmax", &segmented_minmax, "per-segment min/max");
+  m.def("bloom_build", &bloom_build, "bloom filter build");
+  m.def("bloom_probe", &bloom_probe, "bloom filter probe");
+  m.def("zorder_key", &zorder_key, "z-order bit interleave");
+  m.def("gather_rows", &gather_rows, "row gather by index");
+}

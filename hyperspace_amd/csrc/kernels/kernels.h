@@ -1,0 +1,84 @@
+// Host-side API of the CDNA4 kernel suite (pure HIP, gfx950).
+// Implemented in kernels.hip; called from the torch binding (binding.cpp).
+#pragma once
+
+#include <cstdint>
+#include <hip/hip_runtime.h>
+
+namespace hsk {
+
+// K2: Spark-compatible Murmur3 fold over one column into h (u32 in u32 buf).
+// kind: 0 = i32-hash (int32 input), 1 = i64-hash (int64 input).
+void murmur3_column(const void* vals, int kind, uint32_t* h, int64_t n,
+                    bool first, uint32_t seed, hipStream_t stream);
+void pmod_buckets(const uint32_t* h, int32_t* out, int64_t n,
+                  int32_t num_buckets, hipStream_t stream);
+
+// order-preserving u64 normalization (dtype codes: 0=i64,1=i32,2=f64,3=f32,
+// 4=i16, 5=i8, 6=bool/u8)
+void normalize_key(const void* vals, int dtype, uint64_t* out, int64_t n,
+                   hipStream_t stream);
+
+// K3: stable LSD radix sort of (u64 key, i64 payload), ascending u64 order.
+// keys/payload are sorted in place; tmp buffers same size provided by caller.
+void radix_sort_pairs(uint64_t* keys, int64_t* payload, uint64_t* tmp_keys,
+                      int64_t* tmp_payload, uint32_t* hist, uint64_t* d_mask,
+                      int64_t n, hipStream_t stream);
+// histogram buffer size requirement (u32 elements)
+int64_t radix_sort_hist_size(int64_t n);
+
+// K4: segmented sorted merge join. Phase 1: per-left-row match counts +
+// right start positions.  Phase 2 (after exclusive scan of counts into
+// offsets): emit pairs.
+void merge_join_count(const uint64_t* lkeys, const uint64_t* rkeys,
+                      const int64_t* lseg, const int64_t* rseg,
+                      int64_t n_left, int64_t n_seg, int64_t* counts,
+                      int64_t* starts, int64_t* seg_of_row,
+                      hipStream_t stream);
+void merge_join_emit(const int64_t* offsets, const int64_t* counts,
+                     const int64_t* starts, const int64_t* seg_of_row,
+                     const int64_t* rseg, int64_t n_left, int64_t* out_l,
+                     int64_t* out_r, hipStream_t stream);
+
+// generic exclusive scan over i64 (single pass, device-wide)
+void exclusive_scan_i64(const int64_t* in, int64_t* out, int64_t n,
+                        int64_t* total, hipStream_t stream);
+
+// filter scan: indices i (ascending) with lo <= key[i] <= hi (u64 order,
+// inclusive flags).  Two-phase: count (fills block_counts, scans them in
+// place, writes total) then emit (stable compaction using the scanned
+// block bases).  block_counts sized select_num_blocks(n).
+int64_t select_num_blocks(int64_t n);
+void select_range_count(const uint64_t* keys, int64_t n, uint64_t lo,
+                        uint64_t hi, bool lo_incl, bool hi_incl,
+                        int64_t* block_counts, int64_t* total,
+                        hipStream_t stream);
+void select_range_emit(const uint64_t* keys, int64_t n, uint64_t lo,
+                       uint64_t hi, bool lo_incl, bool hi_incl,
+                       const int64_t* block_bases, int64_t* out_idx,
+                       hipStream_t stream);
+
+// K7: membership of values in a sorted set -> bool mask
+void isin_sorted(const int64_t* vals, int64_t n, const int64_t* sorted_set,
+                 int64_t m, bool* out, hipStream_t stream);
+
+// K8: per-segment min/max of int64 values
+void segmented_minmax(const int64_t* vals, const int64_t* seg_off,
+                      int64_t n_seg, int64_t* mins, int64_t* maxs,
+                      hipStream_t stream);
+
+// K8: bloom filter build/probe (k hashes, m_bits bits over u64 words)
+void bloom_build(const int64_t* vals, int64_t n, uint64_t* words,
+                 int64_t m_bits, int k, hipStream_t stream);
+void bloom_probe(const int64_t* vals, int64_t n, const uint64_t* words,
+                 int64_t m_bits, int k, bool* out, hipStream_t stream);
+
+// K10: z-order bit interleave of up to 8 normalized u64 columns
+void zorder_key(const uint64_t* const* cols, int n_cols, int bits_per_col,
+                int64_t n, uint64_t* out, hipStream_t stream);
+
+// row gather: out[i] = in[idx[i]] for element sizes 4/8
+void gather(const void* in, const int64_t* idx, void* out, int64_t n,
+            int elem_size, hipStream_t stream);
+
+}  // namespace hsk

@@ -1,0 +1,896 @@
+// CDNA4 (gfx950 / MI355X) data-plane kernels for hyperspace_amd.
+//
+// Pure HIP, written for wave64 / 64-wide ballots / LDS-resident counters —
+// no CUDA-compat paths.  These implement the reference's Spark data-plane
+// operators:
+//   K2  murmur3 bucket hash (Spark Murmur3_x86_32-compatible, seed 42)
+//   K3  stable LSD radix sort (4-bit digits, ballot multi-split ranking,
+//       constant-nibble pass skipping)
+//   K4  segmented sorted merge join (count + emit)
+//   K7  sorted-set membership (lineage delete filter)
+//   K8  segmented min/max + bloom filter build/probe
+//   K10 z-order bit interleave
+// plus filter-scan compaction (select_range), device exclusive scan and
+// row gather.
+//
+// All kernels are memory-bound streaming ops: 256-thread blocks (4 waves),
+// grid-stride loops sized ≫ 256 workgroups to fill 8 XCDs; occupancy-based
+// latency hiding (no software pipelining needed per the CDNA guide's
+// regime rules for streaming ops).
+
+#include <hip/hip_runtime.h>
+
+#include <cassert>
+#include <cstdint>
+#include <cstdio>
+
+#include "kernels.h"
+
+#define HIP_CHECK(expr)                                              \
+  do {                                                               \
+    hipError_t _e = (expr);                                          \
+    if (_e != hipSuccess) {                                          \
+      fprintf(stderr, "HIP error %s at %s:%d\n",                     \
+              hipGetErrorString(_e), __FILE__, __LINE__);            \
+      abort();                                                       \
+    }                                                                \
+  } while (0)
+
+namespace hsk {
+
+constexpr int THREADS = 256;
+constexpr int WAVES = THREADS / 64;
+
+__host__ __device__ static inline int64_t cdiv(int64_t a, int64_t b) {
+  return (a + b - 1) / b;
+}
+
+static inline int grid_for(int64_t n, int64_t per_block = THREADS) {
+  int64_t blocks = cdiv(n, per_block);
+  if (blocks > 4096) blocks = 4096;  // grid-stride beyond
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
+
+// ---------------------------------------------------------------------------
+// Murmur3 (Spark Murmur3_x86_32, seed 42) — bit-exact with
+// ops/cpu_ref.py::murmur3_* (tested against Spark golden values).
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ uint32_t rotl32(uint32_t x, int r) {
+  return (x << r) | (x >> (32 - r));
+}
+
+__device__ __forceinline__ uint32_t mix_k1(uint32_t k1) {
+  k1 *= 0xCC9E2D51u;
+  k1 = rotl32(k1, 15);
+  k1 *= 0x1B873593u;
+  return k1;
+}
+
+__device__ __forceinline__ uint32_t mix_h1(uint32_t h1, uint32_t k1) {
+  h1 ^= k1;
+  h1 = rotl32(h1, 13);
+  h1 = h1 * 5 + 0xE6546B64u;
+  return h1;
+}
+
+__device__ __forceinline__ uint32_t fmix(uint32_t h1, uint32_t len) {
+  h1 ^= len;
+  h1 ^= h1 >> 16;
+  h1 *= 0x85EBCA6Bu;
+  h1 ^= h1 >> 13;
+  h1 *= 0xC2B2AE35u;
+  h1 ^= h1 >> 16;
+  return h1;
+}
+
+__device__ __forceinline__ uint32_t murmur_i32(uint32_t v, uint32_t seed) {
+  return fmix(mix_h1(seed, mix_k1(v)), 4);
+}
+
+__device__ __forceinline__ uint32_t murmur_i64(uint64_t v, uint32_t seed) {
+  uint32_t low = (uint32_t)v;
+  uint32_t high = (uint32_t)(v >> 32);
+  uint32_t h1 = mix_h1(seed, mix_k1(low));
+  h1 = mix_h1(h1, mix_k1(high));
+  return fmix(h1, 8);
+}
+
+template <typename T, bool IS64>
+__global__ void k_murmur(const T* __restrict__ vals, uint32_t* __restrict__ h,
+                         int64_t n, bool first, uint32_t seed) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    uint32_t s = first ? seed : h[i];
+    if (IS64)
+      h[i] = murmur_i64((uint64_t)vals[i], s);
+    else
+      h[i] = murmur_i32((uint32_t)vals[i], s);
+  }
+}
+
+__global__ void k_pmod(const uint32_t* __restrict__ h,
+                       int32_t* __restrict__ out, int64_t n, int32_t nb) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int32_t signed_h = (int32_t)h[i];
+    int32_t m = signed_h % nb;
+    out[i] = m < 0 ? m + nb : m;
+  }
+}
+
+void murmur3_column(const void* vals, int kind, uint32_t* h, int64_t n,
+                    bool first, uint32_t seed, hipStream_t stream) {
+  int g = grid_for(n);
+  if (kind == 1) {
+    hipLaunchKernelGGL((k_murmur<uint64_t, true>), dim3(g), dim3(THREADS), 0,
+                       stream, (const uint64_t*)vals, h, n, first, seed);
+  } else {
+    hipLaunchKernelGGL((k_murmur<uint32_t, false>), dim3(g), dim3(THREADS), 0,
+                       stream, (const uint32_t*)vals, h, n, first, seed);
+  }
+}
+
+void pmod_buckets(const uint32_t* h, int32_t* out, int64_t n,
+                  int32_t num_buckets, hipStream_t stream) {
+  hipLaunchKernelGGL(k_pmod, dim3(grid_for(n)), dim3(THREADS), 0, stream, h,
+                     out, n, num_buckets);
+}
+
+// ---------------------------------------------------------------------------
+// Order-preserving u64 key normalization
+// ---------------------------------------------------------------------------
+
+__global__ void k_norm_i64(const int64_t* in, uint64_t* out, int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    out[i] = (uint64_t)in[i] ^ 0x8000000000000000ull;
+}
+
+template <typename T>
+__global__ void k_norm_small_int(const T* in, uint64_t* out, int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    out[i] = (uint64_t)(int64_t)in[i] ^ 0x8000000000000000ull;
+}
+
+__global__ void k_norm_f64(const double* in, uint64_t* out, int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t bits = __double_as_longlong(in[i]);
+    uint64_t mask =
+        bits < 0 ? 0xFFFFFFFFFFFFFFFFull : 0x8000000000000000ull;
+    out[i] = (uint64_t)bits ^ mask;
+  }
+}
+
+__global__ void k_norm_f32(const float* in, uint64_t* out, int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    // match the CPU reference: widen to f64 first, then flip
+    int64_t bits = __double_as_longlong((double)in[i]);
+    uint64_t mask =
+        bits < 0 ? 0xFFFFFFFFFFFFFFFFull : 0x8000000000000000ull;
+    out[i] = (uint64_t)bits ^ mask;
+  }
+}
+
+void normalize_key(const void* vals, int dtype, uint64_t* out, int64_t n,
+                   hipStream_t stream) {
+  int g = grid_for(n);
+  switch (dtype) {
+    case 0:
+      hipLaunchKernelGGL(k_norm_i64, dim3(g), dim3(THREADS), 0, stream,
+                         (const int64_t*)vals, out, n);
+      break;
+    case 1:
+      hipLaunchKernelGGL(k_norm_small_int<int32_t>, dim3(g), dim3(THREADS), 0,
+                         stream, (const int32_t*)vals, out, n);
+      break;
+    case 2:
+      hipLaunchKernelGGL(k_norm_f64, dim3(g), dim3(THREADS), 0, stream,
+                         (const double*)vals, out, n);
+      break;
+    case 3:
+      hipLaunchKernelGGL(k_norm_f32, dim3(g), dim3(THREADS), 0, stream,
+                         (const float*)vals, out, n);
+      break;
+    case 4:
+      hipLaunchKernelGGL(k_norm_small_int<int16_t>, dim3(g), dim3(THREADS), 0,
+                         stream, (const int16_t*)vals, out, n);
+      break;
+    case 5:
+      hipLaunchKernelGGL(k_norm_small_int<int8_t>, dim3(g), dim3(THREADS), 0,
+                         stream, (const int8_t*)vals, out, n);
+      break;
+    default:
+      hipLaunchKernelGGL(k_norm_small_int<uint8_t>, dim3(g), dim3(THREADS),
+                         0, stream, (const uint8_t*)vals, out, n);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Stable LSD radix sort (4-bit digits, wave64 ballot multi-split)
+// ---------------------------------------------------------------------------
+
+constexpr int RS_RADIX = 16;
+constexpr int RS_MAX_BLOCKS = 1024;
+
+static inline int rs_num_blocks(int64_t n) {
+  int64_t tiles = cdiv(n, THREADS);
+  return (int)(tiles < RS_MAX_BLOCKS ? (tiles < 1 ? 1 : tiles)
+                                     : RS_MAX_BLOCKS);
+}
+
+int64_t radix_sort_hist_size(int64_t n) {
+  (void)n;
+  return (int64_t)RS_RADIX * RS_MAX_BLOCKS;
+}
+
+__global__ void k_xor_or_reduce(const uint64_t* __restrict__ keys, int64_t n,
+                                uint64_t* __restrict__ out_mask) {
+  __shared__ uint64_t lds[THREADS];
+  uint64_t k0 = keys[0];
+  uint64_t acc = 0;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    acc |= keys[i] ^ k0;
+  lds[threadIdx.x] = acc;
+  __syncthreads();
+  for (int off = THREADS / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) lds[threadIdx.x] |= lds[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0)
+    atomicOr((unsigned long long*)out_mask, (unsigned long long)lds[0]);
+}
+
+__global__ void k_rs_hist(const uint64_t* __restrict__ keys, int64_t n,
+                          int shift, uint32_t* __restrict__ hist, int nb) {
+  __shared__ uint32_t lh[RS_RADIX];
+  if (threadIdx.x < RS_RADIX) lh[threadIdx.x] = 0;
+  __syncthreads();
+  for (int64_t tile = blockIdx.x; tile * THREADS < n; tile += nb) {
+    int64_t i = tile * (int64_t)THREADS + threadIdx.x;
+    if (i < n) {
+      int d = (int)((keys[i] >> shift) & 15);
+      atomicAdd(&lh[d], 1u);
+    }
+  }
+  __syncthreads();
+  if (threadIdx.x < RS_RADIX)
+    hist[(int64_t)threadIdx.x * nb + blockIdx.x] = lh[threadIdx.x];
+}
+
+// single-block exclusive scan over u32 (hist is 16*nb <= 16384 elements)
+__global__ void k_scan_u32(uint32_t* __restrict__ data, int64_t n) {
+  constexpr int T = 1024;
+  __shared__ uint32_t sums[T];
+  int t = threadIdx.x;
+  int64_t chunk = cdiv(n, T);
+  int64_t start = t * chunk;
+  int64_t end = start + chunk < n ? start + chunk : n;
+  uint32_t s = 0;
+  for (int64_t i = start; i < end; i++) s += data[i];
+  sums[t] = s;
+  __syncthreads();
+  for (int off = 1; off < T; off <<= 1) {
+    uint32_t v = (t >= off) ? sums[t - off] : 0;
+    __syncthreads();
+    sums[t] += v;
+    __syncthreads();
+  }
+  uint32_t base = (t == 0) ? 0 : sums[t - 1];
+  for (int64_t i = start; i < end; i++) {
+    uint32_t v = data[i];
+    data[i] = base;
+    base += v;
+  }
+}
+
+__global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
+                             const int64_t* __restrict__ payload,
+                             uint64_t* __restrict__ okeys,
+                             int64_t* __restrict__ opayload, int64_t n,
+                             int shift, const uint32_t* __restrict__ hist,
+                             int nb) {
+  __shared__ uint32_t cur[RS_RADIX];
+  __shared__ uint32_t wave_cnt[WAVES][RS_RADIX];
+  __shared__ uint32_t wave_pref[WAVES][RS_RADIX];
+  __shared__ uint32_t tile_total[RS_RADIX];
+  if (threadIdx.x < RS_RADIX)
+    cur[threadIdx.x] = hist[(int64_t)threadIdx.x * nb + blockIdx.x];
+  int lane = threadIdx.x & 63;
+  int wave = threadIdx.x >> 6;
+  for (int64_t tile = blockIdx.x; tile * THREADS < n; tile += nb) {
+    int64_t i = tile * (int64_t)THREADS + threadIdx.x;
+    bool valid = i < n;
+    uint64_t key = valid ? keys[i] : 0;
+    int64_t pl = valid ? payload[i] : 0;
+    int d = valid ? (int)((key >> shift) & 15) : RS_RADIX;  // 16 = sentinel
+    if (threadIdx.x < WAVES * RS_RADIX)
+      ((uint32_t*)wave_cnt)[threadIdx.x] = 0;
+    __syncthreads();
+    // 5-bit ballot multi-split: lanes with equal digit (incl. sentinel)
+    unsigned long long eq = ~0ull;
+#pragma unroll
+    for (int b = 0; b < 5; b++) {
+      unsigned long long m = __ballot((d >> b) & 1);
+      eq &= ((d >> b) & 1) ? m : ~m;
+    }
+    int rank = __popcll(eq & ((1ull << lane) - 1ull));
+    if (valid && rank == 0) wave_cnt[wave][d] = (uint32_t)__popcll(eq);
+    __syncthreads();
+    if (threadIdx.x < RS_RADIX) {
+      uint32_t p = 0;
+      for (int w = 0; w < WAVES; w++) {
+        wave_pref[w][threadIdx.x] = p;
+        p += wave_cnt[w][threadIdx.x];
+      }
+      tile_total[threadIdx.x] = p;
+    }
+    __syncthreads();
+    if (valid) {
+      uint32_t pos = cur[d] + wave_pref[wave][d] + rank;
+      okeys[pos] = key;
+      opayload[pos] = pl;
+    }
+    __syncthreads();
+    if (threadIdx.x < RS_RADIX) cur[threadIdx.x] += tile_total[threadIdx.x];
+    __syncthreads();
+  }
+}
+
+void radix_sort_pairs(uint64_t* keys, int64_t* payload, uint64_t* tmp_keys,
+                      int64_t* tmp_payload, uint32_t* hist, uint64_t* d_mask,
+                      int64_t n, hipStream_t stream) {
+  if (n <= 1) return;
+  assert(n < (int64_t)UINT32_MAX);
+  int nb = rs_num_blocks(n);
+  int g = grid_for(n);
+
+  // which 4-bit digits actually vary? (constant-nibble pass skipping)
+  HIP_CHECK(hipMemsetAsync(d_mask, 0, sizeof(uint64_t), stream));
+  hipLaunchKernelGGL(k_xor_or_reduce, dim3(g), dim3(THREADS), 0, stream,
+                     keys, n, d_mask);
+  uint64_t mask = 0;
+  HIP_CHECK(hipMemcpyAsync(&mask, d_mask, sizeof(uint64_t),
+                           hipMemcpyDeviceToHost, stream));
+  HIP_CHECK(hipStreamSynchronize(stream));
+
+  uint64_t* ka = keys;
+  uint64_t* kb = tmp_keys;
+  int64_t* pa = payload;
+  int64_t* pb = tmp_payload;
+  for (int shift = 0; shift < 64; shift += 4) {
+    if (((mask >> shift) & 15ull) == 0) continue;  // constant digit
+    HIP_CHECK(hipMemsetAsync(hist, 0,
+                             (size_t)RS_RADIX * nb * sizeof(uint32_t),
+                             stream));
+    hipLaunchKernelGGL(k_rs_hist, dim3(nb), dim3(THREADS), 0, stream, ka, n,
+                       shift, hist, nb);
+    hipLaunchKernelGGL(k_scan_u32, dim3(1), dim3(1024), 0, stream, hist,
+                       (int64_t)RS_RADIX * nb);
+    hipLaunchKernelGGL(k_rs_scatter, dim3(nb), dim3(THREADS), 0, stream, ka,
+                       pa, kb, pb, n, shift, hist, nb);
+    uint64_t* tk = ka; ka = kb; kb = tk;
+    int64_t* tp = pa; pa = pb; pb = tp;
+  }
+  if (ka != keys) {
+    HIP_CHECK(hipMemcpyAsync(keys, ka, n * sizeof(uint64_t),
+                             hipMemcpyDeviceToDevice, stream));
+    HIP_CHECK(hipMemcpyAsync(payload, pa, n * sizeof(int64_t),
+                             hipMemcpyDeviceToDevice, stream));
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Device-wide exclusive scan (i64), 3-kernel two-level
+// ---------------------------------------------------------------------------
+
+constexpr int SCAN_MAX_BLOCKS = 2048;
+
+static inline int scan_num_blocks(int64_t n) {
+  int64_t per = cdiv(n, SCAN_MAX_BLOCKS);
+  if (per < THREADS) per = THREADS;
+  int64_t blocks = cdiv(n, per);
+  return (int)(blocks < 1 ? 1 : blocks);
+}
+
+// block-level exclusive scan of a contiguous range; emits block total
+__global__ void k_scan_partial(const int64_t* __restrict__ in,
+                               int64_t* __restrict__ out, int64_t n,
+                               int64_t per_block,
+                               int64_t* __restrict__ block_tot) {
+  __shared__ int64_t sums[THREADS];
+  int t = threadIdx.x;
+  int64_t b0 = (int64_t)blockIdx.x * per_block;
+  int64_t b1 = b0 + per_block < n ? b0 + per_block : n;
+  int64_t chunk = cdiv(b1 - b0, (int64_t)THREADS);
+  int64_t start = b0 + t * chunk;
+  int64_t end = start + chunk < b1 ? start + chunk : b1;
+  if (start > b1) start = b1;
+  int64_t s = 0;
+  for (int64_t i = start; i < end; i++) s += in[i];
+  sums[t] = s;
+  __syncthreads();
+  for (int off = 1; off < THREADS; off <<= 1) {
+    int64_t v = (t >= off) ? sums[t - off] : 0;
+    __syncthreads();
+    sums[t] += v;
+    __syncthreads();
+  }
+  if (t == THREADS - 1) block_tot[blockIdx.x] = sums[t];
+  int64_t base = (t == 0) ? 0 : sums[t - 1];
+  for (int64_t i = start; i < end; i++) {
+    int64_t v = in[i];
+    out[i] = base;
+    base += v;
+  }
+}
+
+__global__ void k_scan_block_tots(int64_t* __restrict__ tots, int nb,
+                                  int64_t* __restrict__ total) {
+  // single block: exclusive scan of up to SCAN_MAX_BLOCKS totals
+  __shared__ int64_t sums[1024];
+  int t = threadIdx.x;
+  int chunk = (nb + 1023) / 1024;
+  int start = t * chunk;
+  int end = start + chunk < nb ? start + chunk : nb;
+  if (start > nb) start = nb;
+  int64_t s = 0;
+  for (int i = start; i < end; i++) s += tots[i];
+  sums[t] = s;
+  __syncthreads();
+  for (int off = 1; off < 1024; off <<= 1) {
+    int64_t v = (t >= off) ? sums[t - off] : 0;
+    __syncthreads();
+    sums[t] += v;
+    __syncthreads();
+  }
+  if (t == 1023 && total != nullptr) total[0] = sums[t];
+  int64_t base = (t == 0) ? 0 : sums[t - 1];
+  for (int i = start; i < end; i++) {
+    int64_t v = tots[i];
+    tots[i] = base;
+    base += v;
+  }
+}
+
+__global__ void k_scan_add_base(int64_t* __restrict__ out, int64_t n,
+                                int64_t per_block,
+                                const int64_t* __restrict__ block_tot) {
+  int64_t base = block_tot[blockIdx.x];
+  if (base == 0) return;
+  int64_t b0 = (int64_t)blockIdx.x * per_block;
+  int64_t b1 = b0 + per_block < n ? b0 + per_block : n;
+  for (int64_t i = b0 + threadIdx.x; i < b1; i += blockDim.x) out[i] += base;
+}
+
+void exclusive_scan_i64(const int64_t* in, int64_t* out, int64_t n,
+                        int64_t* total, hipStream_t stream) {
+  if (n == 0) {
+    if (total) HIP_CHECK(hipMemsetAsync(total, 0, 8, stream));
+    return;
+  }
+  int nb = scan_num_blocks(n);
+  int64_t per_block = cdiv(n, nb);
+  // temp block totals: caller-invisible; small static buffer via hipMalloc
+  static thread_local int64_t* tots = nullptr;
+  if (tots == nullptr)
+    HIP_CHECK(hipMalloc(&tots, SCAN_MAX_BLOCKS * sizeof(int64_t)));
+  hipLaunchKernelGGL(k_scan_partial, dim3(nb), dim3(THREADS), 0, stream, in,
+                     out, n, per_block, tots);
+  hipLaunchKernelGGL(k_scan_block_tots, dim3(1), dim3(1024), 0, stream, tots,
+                     nb, total);
+  hipLaunchKernelGGL(k_scan_add_base, dim3(nb), dim3(THREADS), 0, stream,
+                     out, n, per_block, tots);
+}
+
+// ---------------------------------------------------------------------------
+// Segmented sorted merge join (K4)
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ int64_t lower_bound_u64(
+    const uint64_t* __restrict__ a, int64_t lo, int64_t hi, uint64_t v) {
+  while (lo < hi) {
+    int64_t mid = (lo + hi) >> 1;
+    if (a[mid] < v)
+      lo = mid + 1;
+    else
+      hi = mid;
+  }
+  return lo;
+}
+
+__device__ __forceinline__ int64_t upper_bound_u64(
+    const uint64_t* __restrict__ a, int64_t lo, int64_t hi, uint64_t v) {
+  while (lo < hi) {
+    int64_t mid = (lo + hi) >> 1;
+    if (a[mid] <= v)
+      lo = mid + 1;
+    else
+      hi = mid;
+  }
+  return lo;
+}
+
+__global__ void k_mj_count(const uint64_t* __restrict__ lkeys,
+                           const uint64_t* __restrict__ rkeys,
+                           const int64_t* __restrict__ lseg,
+                           const int64_t* __restrict__ rseg, int64_t n_left,
+                           int64_t n_seg, int64_t* __restrict__ counts,
+                           int64_t* __restrict__ starts,
+                           int64_t* __restrict__ seg_of_row) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n_left; i += stride) {
+    // segment of row i: binary search over lseg offsets
+    int64_t lo = 0, hi = n_seg;
+    while (lo < hi) {
+      int64_t mid = (lo + hi) >> 1;
+      if (lseg[mid + 1] <= i)
+        lo = mid + 1;
+      else
+        hi = mid;
+    }
+    int64_t s = lo;
+    uint64_t key = lkeys[i];
+    int64_t r0 = rseg[s], r1 = rseg[s + 1];
+    int64_t a = lower_bound_u64(rkeys, r0, r1, key);
+    int64_t b = upper_bound_u64(rkeys, r0, r1, key);
+    counts[i] = b - a;
+    starts[i] = a;
+    seg_of_row[i] = s;
+  }
+}
+
+__global__ void k_mj_emit(const int64_t* __restrict__ offsets,
+                          const int64_t* __restrict__ counts,
+                          const int64_t* __restrict__ starts,
+                          int64_t n_left, int64_t* __restrict__ out_l,
+                          int64_t* __restrict__ out_r) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n_left; i += stride) {
+    int64_t c = counts[i];
+    int64_t off = offsets[i];
+    int64_t a = starts[i];
+    for (int64_t j = 0; j < c; j++) {
+      out_l[off + j] = i;
+      out_r[off + j] = a + j;
+    }
+  }
+}
+
+void merge_join_count(const uint64_t* lkeys, const uint64_t* rkeys,
+                      const int64_t* lseg, const int64_t* rseg,
+                      int64_t n_left, int64_t n_seg, int64_t* counts,
+                      int64_t* starts, int64_t* seg_of_row,
+                      hipStream_t stream) {
+  if (n_left == 0) return;
+  hipLaunchKernelGGL(k_mj_count, dim3(grid_for(n_left)), dim3(THREADS), 0,
+                     stream, lkeys, rkeys, lseg, rseg, n_left, n_seg, counts,
+                     starts, seg_of_row);
+}
+
+void merge_join_emit(const int64_t* offsets, const int64_t* counts,
+                     const int64_t* starts, const int64_t* /*seg_of_row*/,
+                     const int64_t* /*rseg*/, int64_t n_left, int64_t* out_l,
+                     int64_t* out_r, hipStream_t stream) {
+  if (n_left == 0) return;
+  hipLaunchKernelGGL(k_mj_emit, dim3(grid_for(n_left)), dim3(THREADS), 0,
+                     stream, offsets, counts, starts, n_left, out_l, out_r);
+}
+
+// ---------------------------------------------------------------------------
+// Filter scan: stable compaction of rows in a u64 range (select_range)
+// ---------------------------------------------------------------------------
+
+int64_t select_num_blocks(int64_t n) { return rs_num_blocks(n); }
+
+__device__ __forceinline__ bool range_pred(uint64_t k, uint64_t lo,
+                                           uint64_t hi, bool lo_incl,
+                                           bool hi_incl) {
+  bool ge = lo_incl ? (k >= lo) : (k > lo);
+  bool le = hi_incl ? (k <= hi) : (k < hi);
+  return ge && le;
+}
+
+__global__ void k_sel_count(const uint64_t* __restrict__ keys, int64_t n,
+                            uint64_t lo, uint64_t hi, bool lo_incl,
+                            bool hi_incl, int64_t* __restrict__ bcounts,
+                            int nb) {
+  __shared__ int64_t lds[THREADS];
+  int64_t acc = 0;
+  for (int64_t tile = blockIdx.x; tile * THREADS < n; tile += nb) {
+    int64_t i = tile * (int64_t)THREADS + threadIdx.x;
+    if (i < n && range_pred(keys[i], lo, hi, lo_incl, hi_incl)) acc++;
+  }
+  lds[threadIdx.x] = acc;
+  __syncthreads();
+  for (int off = THREADS / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) lds[threadIdx.x] += lds[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) bcounts[blockIdx.x] = lds[0];
+}
+
+__global__ void k_sel_emit(const uint64_t* __restrict__ keys, int64_t n,
+                           uint64_t lo, uint64_t hi, bool lo_incl,
+                           bool hi_incl, const int64_t* __restrict__ bases,
+                           int64_t* __restrict__ out_idx, int nb) {
+  __shared__ int64_t cur;
+  __shared__ uint32_t wave_cnt[WAVES];
+  __shared__ uint32_t wave_pref[WAVES];
+  if (threadIdx.x == 0) cur = bases[blockIdx.x];
+  int lane = threadIdx.x & 63;
+  int wave = threadIdx.x >> 6;
+  for (int64_t tile = blockIdx.x; tile * THREADS < n; tile += nb) {
+    int64_t i = tile * (int64_t)THREADS + threadIdx.x;
+    bool pred =
+        i < n && range_pred(keys[i], lo, hi, lo_incl, hi_incl);
+    __syncthreads();
+    unsigned long long m = __ballot(pred);
+    int rank = __popcll(m & ((1ull << lane) - 1ull));
+    if (lane == 0) wave_cnt[wave] = (uint32_t)__popcll(m);
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      uint32_t p = 0;
+      for (int w = 0; w < WAVES; w++) {
+        wave_pref[w] = p;
+        p += wave_cnt[w];
+      }
+      wave_cnt[0] = p;  // tile total (reuse slot after prefix consumed)
+    }
+    __syncthreads();
+    if (pred) out_idx[cur + wave_pref[wave] + rank] = i;
+    __syncthreads();
+    if (threadIdx.x == 0) cur += wave_cnt[0];
+  }
+}
+
+void select_range_count(const uint64_t* keys, int64_t n, uint64_t lo,
+                        uint64_t hi, bool lo_incl, bool hi_incl,
+                        int64_t* block_counts, int64_t* total,
+                        hipStream_t stream) {
+  if (n == 0) {
+    if (total) HIP_CHECK(hipMemsetAsync(total, 0, 8, stream));
+    return;
+  }
+  int nb = (int)select_num_blocks(n);
+  hipLaunchKernelGGL(k_sel_count, dim3(nb), dim3(THREADS), 0, stream, keys,
+                     n, lo, hi, lo_incl, hi_incl, block_counts, nb);
+  hipLaunchKernelGGL(k_scan_block_tots, dim3(1), dim3(1024), 0, stream,
+                     block_counts, (int)nb, total);
+}
+
+void select_range_emit(const uint64_t* keys, int64_t n, uint64_t lo,
+                       uint64_t hi, bool lo_incl, bool hi_incl,
+                       const int64_t* block_bases, int64_t* out_idx,
+                       hipStream_t stream) {
+  if (n == 0) return;
+  int nb = (int)select_num_blocks(n);
+  hipLaunchKernelGGL(k_sel_emit, dim3(nb), dim3(THREADS), 0, stream, keys, n,
+                     lo, hi, lo_incl, hi_incl, block_bases, out_idx, nb);
+}
+
+// ---------------------------------------------------------------------------
+// Sorted-set membership (K7 lineage filter)
+// ---------------------------------------------------------------------------
+
+__global__ void k_isin(const int64_t* __restrict__ vals, int64_t n,
+                       const int64_t* __restrict__ set, int64_t m,
+                       bool* __restrict__ out) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t v = vals[i];
+    int64_t lo = 0, hi = m;
+    while (lo < hi) {
+      int64_t mid = (lo + hi) >> 1;
+      if (set[mid] < v)
+        lo = mid + 1;
+      else
+        hi = mid;
+    }
+    out[i] = (lo < m) && (set[lo] == v);
+  }
+}
+
+void isin_sorted(const int64_t* vals, int64_t n, const int64_t* sorted_set,
+                 int64_t m, bool* out, hipStream_t stream) {
+  if (n == 0) return;
+  hipLaunchKernelGGL(k_isin, dim3(grid_for(n)), dim3(THREADS), 0, stream,
+                     vals, n, sorted_set, m, out);
+}
+
+// ---------------------------------------------------------------------------
+// Segmented min/max (K8 MinMax sketch)
+// ---------------------------------------------------------------------------
+
+__global__ void k_seg_minmax(const int64_t* __restrict__ vals,
+                             const int64_t* __restrict__ seg_off,
+                             int64_t n_seg, int64_t* __restrict__ mins,
+                             int64_t* __restrict__ maxs) {
+  __shared__ int64_t lmin[THREADS];
+  __shared__ int64_t lmax[THREADS];
+  for (int64_t s = blockIdx.x; s < n_seg; s += gridDim.x) {
+    int64_t a = seg_off[s], b = seg_off[s + 1];
+    int64_t mn = INT64_MAX, mx = INT64_MIN;
+    for (int64_t i = a + threadIdx.x; i < b; i += blockDim.x) {
+      int64_t v = vals[i];
+      mn = v < mn ? v : mn;
+      mx = v > mx ? v : mx;
+    }
+    lmin[threadIdx.x] = mn;
+    lmax[threadIdx.x] = mx;
+    __syncthreads();
+    for (int off = THREADS / 2; off > 0; off >>= 1) {
+      if (threadIdx.x < off) {
+        if (lmin[threadIdx.x + off] < lmin[threadIdx.x])
+          lmin[threadIdx.x] = lmin[threadIdx.x + off];
+        if (lmax[threadIdx.x + off] > lmax[threadIdx.x])
+          lmax[threadIdx.x] = lmax[threadIdx.x + off];
+      }
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+      mins[s] = (b > a) ? lmin[0] : 0;
+      maxs[s] = (b > a) ? lmax[0] : 0;
+    }
+    __syncthreads();
+  }
+}
+
+void segmented_minmax(const int64_t* vals, const int64_t* seg_off,
+                      int64_t n_seg, int64_t* mins, int64_t* maxs,
+                      hipStream_t stream) {
+  if (n_seg == 0) return;
+  int g = (int)(n_seg < 4096 ? n_seg : 4096);
+  hipLaunchKernelGGL(k_seg_minmax, dim3(g), dim3(THREADS), 0, stream, vals,
+                     seg_off, n_seg, mins, maxs);
+}
+
+// ---------------------------------------------------------------------------
+// Bloom filter (K8), double hashing h1 + i*h2 over Murmur3 — matches
+// cpu_ref._bloom_hashes
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ void bloom_h12(int64_t v, uint32_t& h1,
+                                          uint32_t& h2) {
+  h1 = murmur_i64((uint64_t)v, 0);
+  h2 = murmur_i64((uint64_t)v, h1);
+}
+
+__global__ void k_bloom_build(const int64_t* __restrict__ vals, int64_t n,
+                              uint64_t* __restrict__ words, int64_t m_bits,
+                              int k) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    uint32_t h1, h2;
+    bloom_h12(vals[i], h1, h2);
+    for (int j = 0; j < k; j++) {
+      uint64_t combined =
+          ((uint64_t)h1 + (uint64_t)j * (uint64_t)h2) & 0x7FFFFFFFFFFFFFFFull;
+      uint64_t pos = combined % (uint64_t)m_bits;
+      atomicOr((unsigned long long*)&words[pos >> 6],
+               1ull << (pos & 63));
+    }
+  }
+}
+
+__global__ void k_bloom_probe(const int64_t* __restrict__ vals, int64_t n,
+                              const uint64_t* __restrict__ words,
+                              int64_t m_bits, int k, bool* __restrict__ out) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    uint32_t h1, h2;
+    bloom_h12(vals[i], h1, h2);
+    bool all = true;
+    for (int j = 0; j < k && all; j++) {
+      uint64_t combined =
+          ((uint64_t)h1 + (uint64_t)j * (uint64_t)h2) & 0x7FFFFFFFFFFFFFFFull;
+      uint64_t pos = combined % (uint64_t)m_bits;
+      all = (words[pos >> 6] >> (pos & 63)) & 1;
+    }
+    out[i] = all;
+  }
+}
+
+void bloom_build(const int64_t* vals, int64_t n, uint64_t* words,
+                 int64_t m_bits, int k, hipStream_t stream) {
+  if (n == 0) return;
+  hipLaunchKernelGGL(k_bloom_build, dim3(grid_for(n)), dim3(THREADS), 0,
+                     stream, vals, n, words, m_bits, k);
+}
+
+void bloom_probe(const int64_t* vals, int64_t n, const uint64_t* words,
+                 int64_t m_bits, int k, bool* out, hipStream_t stream) {
+  if (n == 0) return;
+  hipLaunchKernelGGL(k_bloom_probe, dim3(grid_for(n)), dim3(THREADS), 0,
+                     stream, vals, n, words, m_bits, k, out);
+}
+
+// ---------------------------------------------------------------------------
+// Z-order bit interleave (K10)
+// ---------------------------------------------------------------------------
+
+struct ZCols {
+  const uint64_t* ptr[8];
+};
+
+__global__ void k_zorder(ZCols cols, int n_cols, int bits_per_col, int64_t n,
+                         uint64_t* __restrict__ out) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    uint64_t z = 0;
+    for (int b = 0; b < bits_per_col; b++) {
+      for (int c = 0; c < n_cols; c++) {
+        uint64_t bit = (cols.ptr[c][i] >> (63 - b)) & 1ull;
+        int out_pos = 63 - (b * n_cols + c);
+        z |= bit << out_pos;
+      }
+    }
+    out[i] = z;
+  }
+}
+
+void zorder_key(const uint64_t* const* cols, int n_cols, int bits_per_col,
+                int64_t n, uint64_t* out, hipStream_t stream) {
+  if (n == 0) return;
+  ZCols z{};
+  for (int c = 0; c < n_cols && c < 8; c++) z.ptr[c] = cols[c];
+  hipLaunchKernelGGL(k_zorder, dim3(grid_for(n)), dim3(THREADS), 0, stream,
+                     z, n_cols, bits_per_col, n, out);
+}
+
+// ---------------------------------------------------------------------------
+// Row gather
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void k_gather(const T* __restrict__ in,
+                         const int64_t* __restrict__ idx,
+                         T* __restrict__ out, int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    out[i] = in[idx[i]];
+}
+
+void gather(const void* in, const int64_t* idx, void* out, int64_t n,
+            int elem_size, hipStream_t stream) {
+  if (n == 0) return;
+  int g = grid_for(n);
+  switch (elem_size) {
+    case 8:
+      hipLaunchKernelGGL(k_gather<uint64_t>, dim3(g), dim3(THREADS), 0,
+                         stream, (const uint64_t*)in, idx, (uint64_t*)out, n);
+      break;
+    case 4:
+      hipLaunchKernelGGL(k_gather<uint32_t>, dim3(g), dim3(THREADS), 0,
+                         stream, (const uint32_t*)in, idx, (uint32_t*)out, n);
+      break;
+    case 2:
+      hipLaunchKernelGGL(k_gather<uint16_t>, dim3(g), dim3(THREADS), 0,
+                         stream, (const uint16_t*)in, idx, (uint16_t*)out, n);
+      break;
+    default:
+      hipLaunchKernelGGL(k_gather<uint8_t>, dim3(g), dim3(THREADS), 0,
+                         stream, (const uint8_t*)in, idx, (uint8_t*)out, n);
+  }
+}
+
+}  // namespace hsk

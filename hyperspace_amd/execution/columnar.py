@@ -152,8 +152,12 @@ class ColumnBatch:
         return ColumnBatch(out)
 
     def gather(self, idx: torch.Tensor) -> "ColumnBatch":
+        from .. import ops
         return ColumnBatch({
-            k: (v.gather(idx) if isinstance(v, StringColumn) else v[idx])
+            k: (StringColumn(ops.gather_rows(v.codes, idx.to(
+                    v.codes.device)), v.values)
+                if isinstance(v, StringColumn)
+                else ops.gather_rows(v, idx.to(v.device)))
             for k, v in self.columns.items()})
 
     def slice(self, start: int, end: int) -> "ColumnBatch":

@@ -23,13 +23,26 @@ from typing import List, Tuple
 import torch
 
 from . import cpu_ref, native
-from .cpu_ref import normalize_key, MASK32, SPARK_HASH_SEED  # re-export
+from .cpu_ref import MASK32, SPARK_HASH_SEED  # re-export
 
 __all__ = [
     "murmur3_bucket", "normalize_key", "sort_pairs", "sort_perm",
     "merge_join", "select_range_u64", "isin_sorted", "segmented_minmax",
-    "bloom_build", "bloom_probe", "zorder_key", "native",
+    "bloom_build", "bloom_probe", "zorder_key", "gather_rows", "native",
 ]
+
+
+def normalize_key(col: torch.Tensor) -> torch.Tensor:
+    if col.is_cuda:
+        return native.ext().normalize_key(col.contiguous())
+    return cpu_ref.normalize_key(col)
+
+
+def gather_rows(values: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
+    if values.is_cuda:
+        return native.ext().gather_rows(values.contiguous(),
+                                        idx.contiguous())
+    return values[idx]
 
 
 def _is_cuda(*tensors: torch.Tensor) -> bool:
