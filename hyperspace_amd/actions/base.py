@@ -55,6 +55,44 @@ class Action(ABC):
 
     # -- template ----------------------------------------------------------
     def run(self) -> None:
+        from ..parallel import dist_context as dc
+        if dc.is_distributed() and dc.get_world_size() > 1:
+            self._run_distributed()
+            return
+        self._run_local()
+
+    def _run_distributed(self) -> None:
+        """Collective run: every rank executes the data-plane op();
+        rank 0 alone drives the metadata log transaction."""
+        from ..parallel import dist_context as dc
+        rank0 = dc.get_rank() == 0
+        dc.barrier()
+        if rank0:
+            self.validate()
+            base = self.log_manager.get_latest_id()
+            self.base_id = base if base is not None else -1
+            self.begin()
+        dc.barrier()
+        no_changes = False
+        try:
+            self.op()
+        except NoChangesException:
+            # all ranks see the same source state, so this is collective
+            no_changes = True
+        dc.barrier()
+        if rank0:
+            if no_changes:
+                self._write_or_fail(self.base_id + 2, self._current_stable())
+                self.log_manager.create_latest_stable_log(self.base_id + 2)
+            else:
+                self.end()
+            ev = self.event("Operation Succeeded." if not no_changes
+                            else "Operation needs no update.")
+            if ev is not None:
+                self._log_event(ev)
+        dc.barrier()
+
+    def _run_local(self) -> None:
         self.validate()
         base = self.log_manager.get_latest_id()
         self.base_id = base if base is not None else -1

@@ -254,11 +254,18 @@ __global__ void k_xor_or_reduce(const uint64_t* __restrict__ keys, int64_t n,
 }
 
 __global__ void k_rs_hist(const uint64_t* __restrict__ keys, int64_t n,
-                          int shift, uint32_t* __restrict__ hist, int nb) {
+                          int shift, uint32_t* __restrict__ hist, int nb,
+                          int64_t tiles_per_block) {
+  // CONTIGUOUS tile->block ranges: block-major order must equal element
+  // order or cross-pass LSD stability breaks (verified on hardware:
+  // round-robin tiling produced a correct per-pass grouping but a wrong
+  // multi-pass sort).
   __shared__ uint32_t lh[RS_RADIX];
   if (threadIdx.x < RS_RADIX) lh[threadIdx.x] = 0;
   __syncthreads();
-  for (int64_t tile = blockIdx.x; tile * THREADS < n; tile += nb) {
+  int64_t t0 = (int64_t)blockIdx.x * tiles_per_block;
+  int64_t t1 = t0 + tiles_per_block;
+  for (int64_t tile = t0; tile < t1 && tile * THREADS < n; tile++) {
     int64_t i = tile * (int64_t)THREADS + threadIdx.x;
     if (i < n) {
       int d = (int)((keys[i] >> shift) & 15);
@@ -301,7 +308,7 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
                              uint64_t* __restrict__ okeys,
                              int64_t* __restrict__ opayload, int64_t n,
                              int shift, const uint32_t* __restrict__ hist,
-                             int nb) {
+                             int nb, int64_t tiles_per_block) {
   __shared__ uint32_t cur[RS_RADIX];
   __shared__ uint32_t wave_cnt[WAVES][RS_RADIX];
   __shared__ uint32_t wave_pref[WAVES][RS_RADIX];
@@ -310,7 +317,9 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
     cur[threadIdx.x] = hist[(int64_t)threadIdx.x * nb + blockIdx.x];
   int lane = threadIdx.x & 63;
   int wave = threadIdx.x >> 6;
-  for (int64_t tile = blockIdx.x; tile * THREADS < n; tile += nb) {
+  int64_t t0 = (int64_t)blockIdx.x * tiles_per_block;
+  int64_t t1 = t0 + tiles_per_block;
+  for (int64_t tile = t0; tile < t1 && tile * THREADS < n; tile++) {
     int64_t i = tile * (int64_t)THREADS + threadIdx.x;
     bool valid = i < n;
     uint64_t key = valid ? keys[i] : 0;
@@ -355,6 +364,7 @@ void radix_sort_pairs(uint64_t* keys, int64_t* payload, uint64_t* tmp_keys,
   if (n <= 1) return;
   assert(n < (int64_t)UINT32_MAX);
   int nb = rs_num_blocks(n);
+  int64_t tpb = cdiv(cdiv(n, THREADS), nb);
   int g = grid_for(n);
 
   // which 4-bit digits actually vary? (constant-nibble pass skipping)
@@ -376,11 +386,11 @@ void radix_sort_pairs(uint64_t* keys, int64_t* payload, uint64_t* tmp_keys,
                              (size_t)RS_RADIX * nb * sizeof(uint32_t),
                              stream));
     hipLaunchKernelGGL(k_rs_hist, dim3(nb), dim3(THREADS), 0, stream, ka, n,
-                       shift, hist, nb);
+                       shift, hist, nb, tpb);
     hipLaunchKernelGGL(k_scan_u32, dim3(1), dim3(1024), 0, stream, hist,
                        (int64_t)RS_RADIX * nb);
     hipLaunchKernelGGL(k_rs_scatter, dim3(nb), dim3(THREADS), 0, stream, ka,
-                       pa, kb, pb, n, shift, hist, nb);
+                       pa, kb, pb, n, shift, hist, nb, tpb);
     uint64_t* tk = ka; ka = kb; kb = tk;
     int64_t* tp = pa; pa = pb; pb = tp;
   }
@@ -608,10 +618,12 @@ __device__ __forceinline__ bool range_pred(uint64_t k, uint64_t lo,
 __global__ void k_sel_count(const uint64_t* __restrict__ keys, int64_t n,
                             uint64_t lo, uint64_t hi, bool lo_incl,
                             bool hi_incl, int64_t* __restrict__ bcounts,
-                            int nb) {
+                            int nb, int64_t tiles_per_block) {
   __shared__ int64_t lds[THREADS];
   int64_t acc = 0;
-  for (int64_t tile = blockIdx.x; tile * THREADS < n; tile += nb) {
+  int64_t t0 = (int64_t)blockIdx.x * tiles_per_block;
+  int64_t t1 = t0 + tiles_per_block;
+  for (int64_t tile = t0; tile < t1 && tile * THREADS < n; tile++) {
     int64_t i = tile * (int64_t)THREADS + threadIdx.x;
     if (i < n && range_pred(keys[i], lo, hi, lo_incl, hi_incl)) acc++;
   }
@@ -627,14 +639,17 @@ __global__ void k_sel_count(const uint64_t* __restrict__ keys, int64_t n,
 __global__ void k_sel_emit(const uint64_t* __restrict__ keys, int64_t n,
                            uint64_t lo, uint64_t hi, bool lo_incl,
                            bool hi_incl, const int64_t* __restrict__ bases,
-                           int64_t* __restrict__ out_idx, int nb) {
+                           int64_t* __restrict__ out_idx, int nb,
+                           int64_t tiles_per_block) {
   __shared__ int64_t cur;
   __shared__ uint32_t wave_cnt[WAVES];
   __shared__ uint32_t wave_pref[WAVES];
   if (threadIdx.x == 0) cur = bases[blockIdx.x];
   int lane = threadIdx.x & 63;
   int wave = threadIdx.x >> 6;
-  for (int64_t tile = blockIdx.x; tile * THREADS < n; tile += nb) {
+  int64_t t0 = (int64_t)blockIdx.x * tiles_per_block;
+  int64_t t1 = t0 + tiles_per_block;
+  for (int64_t tile = t0; tile < t1 && tile * THREADS < n; tile++) {
     int64_t i = tile * (int64_t)THREADS + threadIdx.x;
     bool pred =
         i < n && range_pred(keys[i], lo, hi, lo_incl, hi_incl);
@@ -667,8 +682,9 @@ void select_range_count(const uint64_t* keys, int64_t n, uint64_t lo,
     return;
   }
   int nb = (int)select_num_blocks(n);
+  int64_t tpb = cdiv(cdiv(n, THREADS), nb);
   hipLaunchKernelGGL(k_sel_count, dim3(nb), dim3(THREADS), 0, stream, keys,
-                     n, lo, hi, lo_incl, hi_incl, block_counts, nb);
+                     n, lo, hi, lo_incl, hi_incl, block_counts, nb, tpb);
   hipLaunchKernelGGL(k_scan_block_tots, dim3(1), dim3(1024), 0, stream,
                      block_counts, (int)nb, total);
 }
@@ -679,8 +695,9 @@ void select_range_emit(const uint64_t* keys, int64_t n, uint64_t lo,
                        hipStream_t stream) {
   if (n == 0) return;
   int nb = (int)select_num_blocks(n);
+  int64_t tpb = cdiv(cdiv(n, THREADS), nb);
   hipLaunchKernelGGL(k_sel_emit, dim3(nb), dim3(THREADS), 0, stream, keys, n,
-                     lo, hi, lo_incl, hi_incl, block_bases, out_idx, nb);
+                     lo, hi, lo_incl, hi_incl, block_bases, out_idx, nb, tpb);
 }
 
 // ---------------------------------------------------------------------------

@@ -94,12 +94,17 @@ class Executor:
 
     # ------------------------------------------------------------------
     def _exec_scan(self, plan: Scan, file_subset: Optional[List[str]] = None,
-                   lineage_tracker=None) -> ColumnBatch:
+                   lineage_tracker=None, shard: bool = False) -> ColumnBatch:
         self.stats.record("ParquetScan")
         files = plan.relation.all_files()
         if file_subset is not None:
             subset = set(file_subset)
             files = [f for f in files if f.name in subset]
+        if shard:
+            # distributed build: round-robin file shard per rank
+            from ..parallel import dist_context as dc
+            if dc.is_distributed():
+                files = files[dc.get_rank()::dc.get_world_size()]
         paths = [f.name for f in files]
         self.stats.scanned_files += len(paths)
         self.stats.scanned_bytes += sum(f.size for f in files)

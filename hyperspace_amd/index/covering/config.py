@@ -69,9 +69,18 @@ class CoveringIndexConfig(IndexConfigTrait):
         lineage = (properties.get(IndexConstants.LINEAGE_PROPERTY, "false")
                    .lower() == "true")
 
+        # pre-assign file ids deterministically across ranks (sorted file
+        # order) so distributed shards agree on the lineage id space
+        for f in sorted(scan.relation.all_files(), key=lambda f: f.name):
+            ctx.file_id_tracker.add_file(f.name, f.size, f.modifiedTime)
+
+        from ...parallel import dist_context as dc
+        shard = dc.is_distributed() and dc.get_world_size() > 1
+
         ex = Executor(ctx.session)
         batch = ex._exec_scan(
-            scan, lineage_tracker=ctx.file_id_tracker if lineage else None)
+            scan, lineage_tracker=ctx.file_id_tracker if lineage else None,
+            shard=shard)
         cols = indexed + included
         if lineage:
             cols = cols + [IndexConstants.DATA_FILE_NAME_ID_COLUMN]
