@@ -46,6 +46,20 @@ def _fingerprint(relation) -> LogicalPlanFingerprint:
         [Signature(SIGNATURE_PROVIDER, relation.signature())])
 
 
+def _all_written_files(data_path: str, local_written: List[str]
+                       ) -> List[str]:
+    """In a distributed build every rank wrote its own bucket files;
+    after a barrier, enumerate the whole version directory so the log
+    entry covers all ranks' output."""
+    from ..parallel import dist_context as dc
+    if not (dc.is_distributed() and dc.get_world_size() > 1):
+        return local_written
+    dc.barrier()
+    return sorted(
+        os.path.join(data_path, f) for f in os.listdir(data_path)
+        if f.endswith(".parquet"))
+
+
 def _tracker_from_entry(entry: IndexLogEntry) -> FileIdTracker:
     tracker = FileIdTracker()
     tracker.add_file_info(entry.source_file_infos())
@@ -107,6 +121,7 @@ class CreateAction(Action):
         with self.session.with_rule_disabled():
             index, batch = self.config.create_index(ctx, self.df, properties)
             written = index.write(ctx, batch)
+        written = _all_written_files(data_path, written)
 
         from ..plan.nodes import Scan
         relation = self.df.plan.collect_leaves()[0].relation
@@ -234,6 +249,7 @@ class RefreshFullAction(RefreshActionBase):
             new_index, batch = config.create_index(
                 ctx, df, index.properties)
             written = new_index.write(ctx, batch)
+        written = _all_written_files(data_path, written)
         rel_meta = relation.create_relation_metadata(tracker)
         self._entry = IndexLogEntry.create(
             self.previous.name, new_index, _content_from_paths(written),

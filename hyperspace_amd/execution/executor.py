@@ -108,7 +108,10 @@ class Executor:
         paths = [f.name for f in files]
         self.stats.scanned_files += len(paths)
         self.stats.scanned_bytes += sum(f.size for f in files)
-        batch, row_counts = read_files_batch(paths)
+        if paths:
+            batch, row_counts = read_files_batch(paths)
+        else:
+            batch, row_counts = _empty_batch(plan.relation.schema), []
         if lineage_tracker is not None:
             tracker = lineage_tracker
             ids = []
@@ -147,6 +150,15 @@ class Executor:
             by_bucket.setdefault(b, []).append(p)
 
         wanted_buckets = sorted(by_bucket)
+        # distributed query: each rank serves its owned buckets
+        # (b % world == rank); results are per-rank partitions of the
+        # answer, co-located with the join's other side by construction
+        from ..parallel import dist_context as dc
+        if dc.is_distributed() and dc.get_world_size() > 1 and \
+                plan.use_bucket_spec:
+            rank, world = dc.get_rank(), dc.get_world_size()
+            wanted_buckets = [b for b in wanted_buckets
+                              if b % world == rank]
         if eq_prune is not None:
             col_name, value = eq_prune
             b = self._bucket_of_value(index, col_name, value, num_buckets)
@@ -394,6 +406,20 @@ class Executor:
 
 _SPARK_DTYPES = {"long": torch.int64, "integer": torch.int32,
                  "double": torch.float64, "float": torch.float32}
+
+
+def _empty_batch(schema) -> ColumnBatch:
+    """Zero-row batch carrying the relation's column structure (empty
+    shards in a distributed scan must keep the schema)."""
+    cols: Dict[str, object] = {}
+    for f in schema.fields:
+        if f.type == "string":
+            cols[f.name] = StringColumn(
+                torch.empty(0, dtype=torch.int32), [])
+        else:
+            dt = _SPARK_DTYPES.get(f.type, torch.int64)
+            cols[f.name] = torch.empty(0, dtype=dt)
+    return ColumnBatch(cols)
 
 
 def _compare(batch: ColumnBatch, e: BinComp) -> torch.Tensor:
