@@ -159,34 +159,53 @@ class Executor:
             rank, world = dc.get_rank(), dc.get_world_size()
             wanted_buckets = [b for b in wanted_buckets
                               if b % world == rank]
-        if eq_prune is not None:
-            col_name, value = eq_prune
-            b = self._bucket_of_value(index, col_name, value, num_buckets)
-            pruned = [x for x in wanted_buckets if x != b]
-            self.stats.bucket_pruned_files += sum(
-                len(by_bucket[x]) for x in pruned)
-            wanted_buckets = [b] if b in by_bucket else []
-
         read_cols = list(plan.columns)
         if plan.excluded_source_file_ids:
             lineage_col = IndexConstants.DATA_FILE_NAME_ID_COLUMN
             if lineage_col not in read_cols:
                 read_cols = read_cols + [lineage_col]
 
-        batches: List[ColumnBatch] = []
-        seg_counts = torch.zeros(num_buckets + 1, dtype=torch.int64)
-        for b in wanted_buckets:
-            paths = sorted(by_bucket[b])
-            self.stats.scanned_files += len(paths)
-            sub, _ = read_files_batch(paths, columns=read_cols)
-            seg_counts[b + 1] = sub.num_rows
-            batches.append(sub)
-        if batches:
-            batch = ColumnBatch.concat(batches)
+        # -- device-resident load, cached (288 GB HBM keeps indexes hot) --
+        cache = self.session.index_data_cache()
+        cache_files = [p for b in wanted_buckets
+                       for p in sorted(by_bucket[b])]
+        key = cache.key(entry, cache_files, read_cols) if cache else None
+        cached = cache.get(key) if cache else None
+        if cached is not None:
+            batch, seg = cached
+            self.stats.record("IndexScan(cached)")
         else:
-            batch = ColumnBatch({c: torch.empty(0) for c in read_cols})
-        if self.device.type == "cuda":
-            batch = batch.to(self.device)
+            batches: List[ColumnBatch] = []
+            seg_counts = torch.zeros(num_buckets + 1, dtype=torch.int64)
+            for b in wanted_buckets:
+                paths = sorted(by_bucket[b])
+                self.stats.scanned_files += len(paths)
+                sub, _ = read_files_batch(paths, columns=read_cols)
+                seg_counts[b + 1] = sub.num_rows
+                batches.append(sub)
+            if batches:
+                batch = ColumnBatch.concat(batches)
+            else:
+                batch = ColumnBatch({c: torch.empty(0) for c in read_cols})
+            if self.device.type == "cuda":
+                batch = batch.to(self.device)
+            seg = torch.cumsum(seg_counts, 0)
+            if cache:
+                cache.put(key, batch, seg)
+
+        # equality bucket pruning: slice the single matching bucket range
+        if eq_prune is not None:
+            col_name, value = eq_prune
+            b = self._bucket_of_value(index, col_name, value, num_buckets)
+            self.stats.bucket_pruned_files += sum(
+                len(by_bucket[x]) for x in wanted_buckets if x != b)
+            if b in wanted_buckets:
+                batch = batch.slice(int(seg[b]), int(seg[b + 1]))
+            else:
+                batch = batch.slice(0, 0)
+            seg_local = None
+        else:
+            seg_local = seg
 
         # lineage delete filter (Hybrid Scan deletes, K7)
         if plan.excluded_source_file_ids and batch.num_rows:
@@ -196,21 +215,19 @@ class Executor:
             keep = ~ops.isin_sorted(lineage, ids)
             kept_idx = torch.nonzero(keep, as_tuple=False).flatten()
             # recompute per-bucket counts after the filter
-            if plan.use_bucket_spec:
-                bounds = torch.cumsum(seg_counts, 0)
-                new_counts = torch.zeros_like(seg_counts)
+            if seg_local is not None:
+                new_counts = torch.zeros(num_buckets + 1, dtype=torch.int64)
+                kc = kept_idx.cpu()
                 for b in wanted_buckets:
-                    lo, hi = int(bounds[b]), int(bounds[b + 1])
-                    new_counts[b + 1] = int(
-                        ((kept_idx >= lo) & (kept_idx < hi)).sum())
-                seg_counts = new_counts
+                    lo, hi = int(seg_local[b]), int(seg_local[b + 1])
+                    new_counts[b + 1] = int(((kc >= lo) & (kc < hi)).sum())
+                seg_local = torch.cumsum(new_counts, 0)
             batch = batch.gather(kept_idx)
             self.stats.record("LineageFilter")
         batch = batch.select(plan.columns)
 
-        if plan.use_bucket_spec:
-            seg = torch.cumsum(seg_counts, 0)
-            return batch, seg
+        if plan.use_bucket_spec and eq_prune is None:
+            return batch, seg_local
         return batch, None
 
     def _bucket_of_value(self, index, col_name: str, value,

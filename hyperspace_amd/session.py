@@ -70,6 +70,19 @@ class HyperspaceSession:
             self._index_manager = CachingIndexCollectionManager(self)
         return self._index_manager
 
+    def index_data_cache(self):
+        """Device-resident index data cache (None when disabled)."""
+        from .config import IndexConstants
+        if not self.conf.get(
+                "spark.hyperspace.index.dataCache.enabled", True):
+            return None
+        if not hasattr(self, "_index_data_cache"):
+            from .execution.index_cache import IndexDataCache
+            budget = int(self.conf.get(
+                "spark.hyperspace.index.dataCache.budgetBytes", 64 << 30))
+            self._index_data_cache = IndexDataCache(budget)
+        return self._index_data_cache
+
     @property
     def event_logger(self):
         if self._event_logger is None:
