@@ -22,6 +22,10 @@ from .dataframe import DataFrame
 from .exceptions import HyperspaceException, KernelUnavailableError
 from .hyperspace import Hyperspace
 from .index.covering import CoveringIndex, CoveringIndexConfig
+from .index.dataskipping import (BloomFilterSketch, DataSkippingIndex,
+                                 DataSkippingIndexConfig, MinMaxSketch,
+                                 PartitionSketch)
+from .index.zorder import ZOrderCoveringIndex, ZOrderCoveringIndexConfig
 from .plan.expr import col, lit
 from .session import HyperspaceSession, get_session, set_session
 
@@ -31,5 +35,8 @@ __all__ = [
     "Conf", "IndexConstants", "DataFrame", "Hyperspace",
     "HyperspaceSession", "HyperspaceException", "KernelUnavailableError",
     "CoveringIndex", "CoveringIndexConfig", "col", "lit",
+    "DataSkippingIndex", "DataSkippingIndexConfig", "MinMaxSketch",
+    "BloomFilterSketch", "PartitionSketch",
+    "ZOrderCoveringIndex", "ZOrderCoveringIndexConfig",
     "get_session", "set_session",
 ]

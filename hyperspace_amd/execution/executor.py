@@ -97,6 +97,8 @@ class Executor:
                    lineage_tracker=None, shard: bool = False) -> ColumnBatch:
         self.stats.record("ParquetScan")
         files = plan.relation.all_files()
+        if file_subset is None and plan.file_subset is not None:
+            file_subset = plan.file_subset
         if file_subset is not None:
             subset = set(file_subset)
             files = [f for f in files if f.name in subset]
@@ -136,7 +138,9 @@ class Executor:
         self.stats.record("IndexScan")
         entry = plan.entry
         index = entry.derivedDataset
-        num_buckets = index.num_buckets
+        # z-order indexes have chunk ids, not hash buckets: size the
+        # segment array to the max id seen
+        num_buckets = getattr(index, "num_buckets", 0)
         all_files = (plan.version_files if plan.version_files is not None
                      else [f for f in entry.content.os_files()
                            if f.endswith(".parquet")])
@@ -149,6 +153,8 @@ class Executor:
                 raise HyperspaceException(f"Index file without bucket id: {p}")
             by_bucket.setdefault(b, []).append(p)
 
+        if by_bucket:
+            num_buckets = max(num_buckets, max(by_bucket) + 1)
         wanted_buckets = sorted(by_bucket)
         # distributed query: each rank serves its owned buckets
         # (b % world == rank); results are per-rank partitions of the
@@ -230,6 +236,29 @@ class Executor:
             return batch, seg_local
         return batch, None
 
+    def _zorder_prune(self, cond: Expr, scan: IndexScan):
+        """Prune a z-order IndexScan's file list using Parquet column
+        stats for every comparison conjunct on an indexed column."""
+        from ..plan.expr import split_conjunctive
+        index = scan.entry.derivedDataset
+        files = [f for f in scan.entry.content.os_files()
+                 if f.endswith(".parquet")]
+        indexed = {c.lower() for c in index.indexed_columns}
+        total_skipped = 0
+        for conj in split_conjunctive(cond):
+            if isinstance(conj, BinComp) and isinstance(conj.left, Col) \
+                    and isinstance(conj.right, Lit) and \
+                    conj.left.name.lower() in indexed:
+                files, skipped = index.prune_files_by_stats(
+                    files, conj.left.name, conj.op, conj.right.value)
+                total_skipped += skipped
+        if total_skipped == 0:
+            return None
+        self.stats.bucket_pruned_files += total_skipped
+        return IndexScan(scan.entry, scan.columns, False,
+                         scan.excluded_source_file_ids,
+                         version_files=files)
+
     def _bucket_of_value(self, index, col_name: str, value,
                          num_buckets: int) -> int:
         t = _value_tensor(index.schema.field_type(col_name), value)
@@ -238,8 +267,17 @@ class Executor:
     # ------------------------------------------------------------------
     def _exec_filter(self, plan: Filter
                      ) -> Tuple[ColumnBatch, Optional[torch.Tensor]]:
-        # bucket pruning: Filter(eq on first indexed col, IndexScan)
         child = plan.child
+        # z-order path: prune index files via their Parquet column stats
+        if isinstance(child, IndexScan) and \
+                child.entry.derivedDataset.kind == "ZOrderCoveringIndex" \
+                and child.version_files is None:
+            pruned = self._zorder_prune(plan.condition, child)
+            if pruned is not None:
+                batch, _ = self._exec_index_scan(pruned)
+                self.stats.record("Filter")
+                return self._apply_predicate(batch, plan.condition), None
+        # bucket pruning: Filter(eq on first indexed col, IndexScan)
         if isinstance(child, IndexScan) and child.use_bucket_spec:
             eq = _single_equality(plan.condition)
             index = child.entry.derivedDataset

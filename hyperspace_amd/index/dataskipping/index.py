@@ -1,0 +1,181 @@
+"""DataSkippingIndex: one row per source file of sketch aggregates.
+
+Reference: index/dataskipping/DataSkippingIndex.scala —
+createIndexData (:291-317, groupBy input_file_name -> per-file sketch
+aggregation = K8 segmented reductions + device bloom build),
+translateFilterCondition (:143-185), DataSkippingFileIndex
+(execution/DataSkippingFileIndex.scala:32-74: prunes the source file
+list with the translated predicate; unconvertible files are KEPT).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Any, Dict, List, Optional, Tuple
+
+import torch
+
+from ..base import Index, IndexerContext
+from ...exceptions import HyperspaceException
+from ...log.entry import (Schema, register_derived_dataset,
+                          DATASKIPPING_INDEX_TYPE)
+from ...plan.expr import And, Expr, Not, Or
+from .sketches import Sketch, sketch_from_json
+
+
+class DataSkippingIndex(Index):
+    def __init__(self, sketches: List[Sketch], schema: Schema,
+                 properties: Dict[str, str]):
+        self.sketches = list(sketches)
+        self.schema = schema  # source column schema for sketched columns
+        self._properties = dict(properties)
+
+    @property
+    def kind(self) -> str:
+        return "DataSkippingIndex"
+
+    @property
+    def kind_abbr(self) -> str:
+        return "DS"
+
+    def indexed_columns_list(self) -> List[str]:
+        return sorted({s.expr for s in self.sketches})
+
+    def referenced_columns(self) -> List[str]:
+        return self.indexed_columns_list()
+
+    def with_new_properties(self, props):
+        return DataSkippingIndex(self.sketches, self.schema, props)
+
+    @property
+    def can_handle_deleted_files(self) -> bool:
+        # per-file rows: deleted files simply drop out of the join
+        return True
+
+    def to_json(self) -> Dict[str, Any]:
+        return {"type": DATASKIPPING_INDEX_TYPE,
+                "sketches": [s.to_json() for s in self.sketches],
+                "schema": self.schema.to_json(),
+                "properties": self._properties}
+
+    @staticmethod
+    def from_json(d) -> "DataSkippingIndex":
+        return DataSkippingIndex(
+            [sketch_from_json(s) for s in d.get("sketches", [])],
+            Schema.from_json(d.get("schema", {"fields": []})),
+            d.get("properties", {}))
+
+    # -- build -------------------------------------------------------------
+    def write(self, ctx: IndexerContext, index_data) -> List[str]:
+        """index_data: dict with '_data_file_id' per-file ids and the
+        sketch aggregate tensors (built by config.create_index)."""
+        os.makedirs(ctx.index_data_path, exist_ok=True)
+        out = os.path.join(ctx.index_data_path,
+                           "part-00000-sketches_00000.c000.parquet")
+        import pyarrow as pa
+        import pyarrow.parquet as pq
+        arrays = {}
+        for name, t in index_data.items():
+            if t.dim() == 2:  # bloom word matrix -> fixed-size list column
+                arrays[name] = pa.array(list(t.numpy()))
+            else:
+                arrays[name] = pa.array(t.numpy())
+        pq.write_table(pa.table(arrays), out, compression="NONE",
+                       use_dictionary=False)
+        return [out]
+
+    def refresh_full(self, ctx, df):
+        raise NotImplementedError
+
+    # -- query -------------------------------------------------------------
+    def load_sketch_data(self, entry):
+        import pyarrow.parquet as pq
+        import numpy as np
+        from ...execution.columnar import ColumnBatch
+        paths = [p for p in entry.content.os_files()
+                 if p.endswith(".parquet")]
+        t = pq.read_table(paths)
+        cols = {}
+        for name, col in zip(t.column_names, t.columns):
+            arr = col.to_numpy(zero_copy_only=False)
+            if arr.dtype == object:  # list column (bloom words)
+                arr = np.stack([np.asarray(x, dtype=np.int64)
+                                for x in arr])
+                cols[name] = torch.from_numpy(arr)
+            else:
+                arr = np.ascontiguousarray(arr)
+                if not arr.flags.writeable:
+                    arr = arr.copy()
+                cols[name] = torch.from_numpy(arr)
+        return _SketchData(cols)
+
+    def translate_filter(self, pred: Expr, sketch_data
+                         ) -> Optional[torch.Tensor]:
+        """Filter predicate -> per-file "may contain" mask, or None when
+        nothing is convertible (reference translateFilterCondition:
+        AND: convertible side suffices; OR: both sides must convert;
+        NOT: not convertible — conservative)."""
+        if isinstance(pred, And):
+            left = self.translate_filter(pred.left, sketch_data)
+            right = self.translate_filter(pred.right, sketch_data)
+            if left is None:
+                return right
+            if right is None:
+                return left
+            return left & right
+        if isinstance(pred, Or):
+            left = self.translate_filter(pred.left, sketch_data)
+            right = self.translate_filter(pred.right, sketch_data)
+            if left is None or right is None:
+                return None
+            return left | right
+        if isinstance(pred, Not):
+            return None
+        refs = pred.references()
+        for sketch in self.sketches:
+            if {r.lower() for r in refs} == {sketch.expr.lower()}:
+                dtype_name = self.schema.field_type(sketch.expr) or "long"
+                mask = sketch.convert_predicate(pred, sketch_data,
+                                                dtype_name)
+                if mask is not None:
+                    return mask
+        return None
+
+    def prune_files(self, entry, pred: Expr, source_files: List[str]
+                    ) -> Tuple[List[str], int]:
+        """Returns (files to scan, number skipped).  Files absent from the
+        sketch table (e.g. appended after build) are kept."""
+        data = self.load_sketch_data(entry)
+        mask = self.translate_filter(pred, data)
+        if mask is None:
+            return source_files, 0
+        file_ids = data.tensor("_data_file_id")
+        id_to_keep = {int(fid): bool(m)
+                      for fid, m in zip(file_ids, mask)}
+        # map file path -> logged id
+        path_to_id = {f.name: f.id for f in entry.source_file_infos()}
+        kept, skipped = [], 0
+        for p in source_files:
+            fid = path_to_id.get(p)
+            if fid is None or id_to_keep.get(fid, True):
+                kept.append(p)
+            else:
+                skipped += 1
+        return kept, skipped
+
+    def statistics(self) -> Dict[str, Any]:
+        return {"sketches": [f"{s.kind}({s.expr})" for s in self.sketches]}
+
+
+class _SketchData:
+    def __init__(self, cols: Dict[str, torch.Tensor]):
+        self.cols = cols
+
+    def tensor(self, name: str) -> torch.Tensor:
+        for k, v in self.cols.items():
+            if k.lower() == name.lower():
+                return v
+        raise HyperspaceException(f"No sketch column {name}")
+
+
+register_derived_dataset(DATASKIPPING_INDEX_TYPE, DataSkippingIndex)

@@ -1,0 +1,27 @@
+"""ZOrderCoveringIndexConfig (reference:
+index/zordercovering/ZOrderCoveringIndexConfig — same validation as the
+covering config)."""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+from ..base import IndexerContext
+from ..covering.config import CoveringIndexConfig
+from .index import ZOrderCoveringIndex
+from ...config import IndexConstants
+from ...log.entry import Schema, SchemaField
+from ...utils.resolver import resolve_all
+
+
+class ZOrderCoveringIndexConfig(CoveringIndexConfig):
+    def create_index(self, ctx: IndexerContext, df,
+                     properties: Dict[str, str]
+                     ) -> Tuple[ZOrderCoveringIndex, object]:
+        # reuse the covering projection/lineage pipeline, then wrap the
+        # result in a z-order index
+        cov_index, batch = super().create_index(ctx, df, properties)
+        index = ZOrderCoveringIndex(
+            cov_index.indexed_columns, cov_index.included_columns,
+            cov_index.schema, dict(properties))
+        return index, batch

@@ -56,12 +56,21 @@ class LogicalPlan:
 
 
 class Scan(LogicalPlan):
-    """Leaf: scan of a file-based source relation."""
+    """Leaf: scan of a file-based source relation.
 
-    def __init__(self, relation, options: Optional[Dict[str, Any]] = None):
+    ``file_subset`` restricts the scan to specific files — the rewrite
+    target of ApplyDataSkippingIndex (the reference swaps the FileIndex
+    for a DataSkippingFileIndex; here the pruned list is explicit).
+    """
+
+    def __init__(self, relation, options: Optional[Dict[str, Any]] = None,
+                 file_subset: Optional[List[str]] = None,
+                 skipped_files: int = 0):
         super().__init__([])
         self.relation = relation  # sources.FileBasedRelation
         self.options = options or {}
+        self.file_subset = file_subset
+        self.skipped_files = skipped_files
 
     def output_columns(self):
         return self.relation.schema.field_names()
@@ -71,7 +80,9 @@ class Scan(LogicalPlan):
         return self
 
     def _node_str(self):
-        return f"Scan({self.relation.describe()})"
+        extra = (f", dataskipping:-{self.skipped_files}files"
+                 if self.file_subset is not None else "")
+        return f"Scan({self.relation.describe()}{extra})"
 
 
 class Filter(LogicalPlan):

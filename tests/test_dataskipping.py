@@ -1,0 +1,140 @@
+"""DataSkipping index tests (reference: index/dataskipping/* suites)."""
+
+import numpy as np
+import pyarrow as pa
+import pyarrow.parquet as pq
+import pytest
+import torch
+
+import hyperspace_amd as hs
+from hyperspace_amd.execution.executor import Executor
+from hyperspace_amd.exceptions import HyperspaceException
+from hyperspace_amd.plan.nodes import Scan
+
+
+@pytest.fixture
+def env(tmp_path, monkeypatch):
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "indexes"))
+    data = tmp_path / "data"
+    data.mkdir()
+    rng = np.random.default_rng(51)
+    # files with DISJOINT key ranges so min/max skipping bites
+    for i in range(8):
+        t = pa.table({
+            "key": rng.integers(i * 1000, (i + 1) * 1000, 5000),
+            "cat": rng.integers(0, 10, 5000),
+            "val": rng.random(5000),
+        })
+        pq.write_table(t, str(data / f"part-{i}.parquet"))
+    session = hs.HyperspaceSession(device="cpu")
+    h = hs.Hyperspace(session)
+    df = session.read_parquet(str(data))
+    return session, h, df, data, rng
+
+
+def _rows(batch, cols):
+    arrs = batch.to_numpy()
+    return sorted(zip(*[arrs[c].tolist() for c in cols]))
+
+
+def test_config_validation():
+    with pytest.raises(HyperspaceException):
+        hs.DataSkippingIndexConfig("x")  # no sketches
+    with pytest.raises(HyperspaceException):
+        hs.DataSkippingIndexConfig(
+            "x", hs.MinMaxSketch("a"), hs.MinMaxSketch("a"))  # dup
+
+
+def test_minmax_skips_files(env):
+    session, h, df, _, _ = env
+    h.create_index(df, hs.DataSkippingIndexConfig(
+        "ds1", hs.MinMaxSketch("key")))
+    session.enable_hyperspace()
+    q = df.filter("key = 2500").select("key", "val")
+    plan = q.optimized_plan()
+    scans = [l for l in plan.collect_leaves() if isinstance(l, Scan)]
+    assert scans and scans[0].file_subset is not None, plan.pretty()
+    assert scans[0].skipped_files == 7  # only file 2 can contain 2500
+    ex = Executor(session)
+    out = ex.execute(plan)
+    assert ex.stats.scanned_files == 1
+    session.disable_hyperspace()
+    base = q.collect()
+    assert _rows(out, ["key", "val"]) == _rows(base, ["key", "val"])
+
+
+def test_minmax_range_predicates(env):
+    session, h, df, _, _ = env
+    h.create_index(df, hs.DataSkippingIndexConfig(
+        "ds1", hs.MinMaxSketch("key")))
+    session.enable_hyperspace()
+    for pred, max_files in [("key < 1500", 2), ("key >= 6500", 2),
+                            ("key <= 999", 1)]:
+        q = df.filter(pred).select("key")
+        ex = Executor(session)
+        out = ex.execute(q.optimized_plan())
+        assert ex.stats.scanned_files <= max_files, pred
+        session.disable_hyperspace()
+        assert out.num_rows == q.collect().num_rows, pred
+        session.enable_hyperspace()
+
+
+def test_bloom_skips_files(env):
+    session, h, df, data, rng = env
+    # add a column where equality has no range structure: use cat values
+    # that only exist in one file
+    t = pa.table({"key": rng.integers(0, 8000, 1000),
+                  "cat": np.full(1000, 777, dtype=np.int64),
+                  "val": rng.random(1000)})
+    pq.write_table(t, str(data / "part-special.parquet"))
+    h.create_index(df, hs.DataSkippingIndexConfig(
+        "dsb", hs.BloomFilterSketch("cat", 0.01, 1000)))
+    session.enable_hyperspace()
+    q = df.filter("cat = 777").select("cat", "val")
+    ex = Executor(session)
+    out = ex.execute(q.optimized_plan())
+    assert ex.stats.scanned_files <= 2  # the special file (+ fp allowance)
+    session.disable_hyperspace()
+    assert out.num_rows == q.collect().num_rows == 1000
+
+
+def test_combined_and_or_translation(env):
+    session, h, df, _, _ = env
+    h.create_index(df, hs.DataSkippingIndexConfig(
+        "ds2", hs.MinMaxSketch("key"), hs.MinMaxSketch("cat")))
+    session.enable_hyperspace()
+    # AND: either side prunes; OR: both must convert
+    q = df.filter("key <= 999 AND cat <= 5").select("key", "cat")
+    ex = Executor(session)
+    out = ex.execute(q.optimized_plan())
+    assert ex.stats.scanned_files == 1
+    session.disable_hyperspace()
+    assert out.num_rows == q.collect().num_rows
+
+
+def test_covering_index_beats_dataskipping(env):
+    session, h, df, _, _ = env
+    h.create_index(df, hs.DataSkippingIndexConfig(
+        "ds1", hs.MinMaxSketch("key")))
+    h.create_index(df, hs.CoveringIndexConfig("ci1", ["key"], ["val"]))
+    session.enable_hyperspace()
+    q = df.filter("key = 2500").select("key", "val")
+    plan = q.optimized_plan()
+    from hyperspace_amd.plan.nodes import IndexScan
+    # covering index (score 50) wins over data skipping (score 1)
+    assert any(isinstance(l, IndexScan) for l in plan.collect_leaves())
+
+
+def test_dataskipping_json_roundtrip(env):
+    session, h, df, _, _ = env
+    h.create_index(df, hs.DataSkippingIndexConfig(
+        "dsr", hs.MinMaxSketch("key"),
+        hs.BloomFilterSketch("cat", 0.05, 500)))
+    entry = session.index_manager().get_index("dsr")
+    d = entry.to_json()
+    assert d["derivedDataset"]["type"].endswith("DataSkippingIndex")
+    from hyperspace_amd.log.entry import IndexLogEntry
+    back = IndexLogEntry.from_json(d)
+    kinds = [s.kind for s in back.derivedDataset.sketches]
+    assert kinds == ["MinMax", "BloomFilter"]
+    assert back.derivedDataset.sketches[1].fpp == 0.05
