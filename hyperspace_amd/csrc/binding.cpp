@@ -233,6 +233,19 @@ torch::Tensor zorder_key(std::vector<torch::Tensor> cols,
   return out;
 }
 
+void copy_unaligned(torch::Tensor src_u8, int64_t src_off,
+                    torch::Tensor dst, int64_t dst_byte_off,
+                    int64_t nbytes) {
+  TORCH_CHECK(src_u8.is_cuda() && dst.is_cuda(), "device tensors required");
+  TORCH_CHECK(src_u8.scalar_type() == torch::kUInt8, "src must be u8");
+  TORCH_CHECK(src_off + nbytes + 4 <= src_u8.numel(),
+              "src buffer must extend 4 bytes past the payload");
+  TORCH_CHECK(dst_byte_off % 4 == 0 && nbytes % 4 == 0, "4B granularity");
+  hsk::copy_unaligned(src_u8.data_ptr<uint8_t>(), src_off,
+                      (uint8_t*)dst.data_ptr(), dst_byte_off, nbytes,
+                      current_stream());
+}
+
 torch::Tensor gather_rows(torch::Tensor values, torch::Tensor idx) {
   check_cuda(values, "values");
   check_cuda(idx, "idx");
@@ -288,4 +301,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bloom_probe", &bloom_probe, "bloom filter probe");
   m.def("zorder_key", &zorder_key, "z-order bit interleave");
   m.def("gather_rows", &gather_rows, "row gather by index");
+  m.def("copy_unaligned", &copy_unaligned,
+        "device parquet page decode (unaligned copy)");
 }

@@ -81,4 +81,11 @@ void zorder_key(const uint64_t* const* cols, int n_cols, int bits_per_col,
 void gather(const void* in, const int64_t* idx, void* out, int64_t n,
             int elem_size, hipStream_t stream);
 
+// K1 parquet page decode: copy nbytes from (src + src_off) — any byte
+// alignment — to 4-byte-aligned dst+dst_off.  nbytes % 4 == 0.  The src
+// buffer must extend >= 4 bytes past src_off+nbytes (parquet footers
+// guarantee this for page payloads).
+void copy_unaligned(const uint8_t* src, int64_t src_off, uint8_t* dst,
+                    int64_t dst_off, int64_t nbytes, hipStream_t stream);
+
 }  // namespace hsk

@@ -910,4 +910,47 @@ void gather(const void* in, const int64_t* idx, void* out, int64_t n,
   }
 }
 
+// ---------------------------------------------------------------------------
+// K1: parquet PLAIN page decode = unaligned-source device copy
+// ---------------------------------------------------------------------------
+
+__global__ void k_copy_aligned32(const uint32_t* __restrict__ src,
+                                 uint32_t* __restrict__ dst, int64_t n32) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n32;
+       i += stride)
+    dst[i] = src[i];
+}
+
+__global__ void k_copy_unaligned32(const uint32_t* __restrict__ src_base,
+                                   int shift_bits,
+                                   uint32_t* __restrict__ dst, int64_t n32) {
+  // dst word j = bytes src_base[j..j+1] funnel-shifted by the source
+  // misalignment (shift_bits = 8 * (src_off & 3))
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n32;
+       i += stride) {
+    uint32_t lo = src_base[i];
+    uint32_t hi = src_base[i + 1];
+    dst[i] = (lo >> shift_bits) | (hi << (32 - shift_bits));
+  }
+}
+
+void copy_unaligned(const uint8_t* src, int64_t src_off, uint8_t* dst,
+                    int64_t dst_off, int64_t nbytes, hipStream_t stream) {
+  if (nbytes == 0) return;
+  uint32_t* d = (uint32_t*)(dst + dst_off);
+  int64_t n32 = nbytes / 4;
+  int a = (int)(src_off & 3);
+  const uint32_t* s = (const uint32_t*)(src + (src_off & ~3ll));
+  int g = grid_for(n32);
+  if (a == 0) {
+    hipLaunchKernelGGL(k_copy_aligned32, dim3(g), dim3(THREADS), 0, stream,
+                       s, d, n32);
+  } else {
+    hipLaunchKernelGGL(k_copy_unaligned32, dim3(g), dim3(THREADS), 0,
+                       stream, s, 8 * a, d, n32);
+  }
+}
+
 }  // namespace hsk

@@ -27,7 +27,9 @@ from ..plan.expr import (And, BinComp, Col, Expr, In, IsNotNull, Lit, Not,
                          Or, extract_equi_join_keys)
 from ..plan.nodes import (BucketUnionNode, Filter, IndexScan, Join,
                           LogicalPlan, Project, Scan, UnionNode)
-from ..sources.parquet_io import read_files_batch, bucket_id_of_file
+from ..sources.parquet_io import (read_files_batch,
+                                  read_files_batch_device,
+                                  bucket_id_of_file)
 
 
 class PhysicalStats:
@@ -110,7 +112,10 @@ class Executor:
         paths = [f.name for f in files]
         self.stats.scanned_files += len(paths)
         self.stats.scanned_bytes += sum(f.size for f in files)
-        if paths:
+        if paths and self.device.type == "cuda":
+            batch, row_counts = read_files_batch_device(
+                paths, self.device)
+        elif paths:
             batch, row_counts = read_files_batch(paths)
         else:
             batch, row_counts = _empty_batch(plan.relation.schema), []
@@ -123,7 +128,8 @@ class Executor:
             lineage = (torch.cat(ids) if ids
                        else torch.empty(0, dtype=torch.int64))
             batch = batch.with_column(
-                IndexConstants.DATA_FILE_NAME_ID_COLUMN, lineage)
+                IndexConstants.DATA_FILE_NAME_ID_COLUMN,
+                lineage.to(batch.device))
         if self.device.type == "cuda":
             batch = batch.to(self.device)
         return batch
@@ -186,7 +192,11 @@ class Executor:
             for b in wanted_buckets:
                 paths = sorted(by_bucket[b])
                 self.stats.scanned_files += len(paths)
-                sub, _ = read_files_batch(paths, columns=read_cols)
+                if self.device.type == "cuda":
+                    sub, _ = read_files_batch_device(
+                        paths, self.device, columns=read_cols)
+                else:
+                    sub, _ = read_files_batch(paths, columns=read_cols)
                 seg_counts[b + 1] = sub.num_rows
                 batches.append(sub)
             if batches:

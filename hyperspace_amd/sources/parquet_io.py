@@ -19,6 +19,7 @@ import uuid
 from typing import Dict, List, Optional, Tuple
 
 import numpy as np
+import torch
 
 _BUCKET_RE = re.compile(r".*_(\d+)(?:\.\w+)*\.parquet$")
 
@@ -37,8 +38,16 @@ def write_batch_parquet(batch, path: str, compression: Optional[str] = None,
                         use_dictionary: bool = False) -> Tuple[int, int]:
     """Write a ColumnBatch to one parquet file.  Returns (size, mtime_ms).
 
-    Defaults to uncompressed PLAIN pages (device-decodable).
+    All-numeric batches use the native writer (uncompressed PLAIN pages
+    assembled directly from the column buffers — K3 encode side); string
+    columns fall back to pyarrow.
     """
+    from ..execution.columnar import StringColumn
+    if compression is None and not any(
+            isinstance(c, StringColumn) for c in batch.columns.values()):
+        from .native_parquet import write_parquet_native
+        cols = {name: t.numpy() for name, t in batch.columns.items()}
+        return write_parquet_native(cols, path)
     import pyarrow.parquet as pq
     table = batch.to_arrow()
     pq.write_table(table, path, compression=compression or "NONE",
@@ -51,10 +60,37 @@ def write_batch_parquet(batch, path: str, compression: Optional[str] = None,
 
 def read_files_batch(paths: List[str], columns: Optional[List[str]] = None):
     """Read parquet files into a single host ColumnBatch, plus per-file row
-    counts (for lineage / per-file segmentation)."""
+    counts (for lineage / per-file segmentation).  Native-layout files
+    (uncompressed PLAIN numeric) bypass pyarrow."""
+    import numpy as np
+    from ..execution.columnar import ColumnBatch
+    from .native_parquet import read_native_host
+
+    per_file = []
+    row_counts = []
+    native_ok = True
+    for p in paths:
+        cols = read_native_host(p, columns)
+        if cols is None:
+            native_ok = False
+            break
+        per_file.append(cols)
+        n = len(next(iter(cols.values()))) if cols else 0
+        row_counts.append(n)
+    if native_ok and paths:
+        names = list(per_file[0].keys())
+        if columns is not None:
+            order = {c.lower(): i for i, c in enumerate(columns)}
+            names.sort(key=lambda n: order.get(n.lower(), 99))
+        merged = {}
+        for name in names:
+            arrs = [f[name] for f in per_file]
+            merged[name] = torch.from_numpy(
+                np.concatenate(arrs) if len(arrs) > 1 else arrs[0])
+        return ColumnBatch(merged), row_counts
+
     import pyarrow.parquet as pq
     import pyarrow as pa
-    from ..execution.columnar import ColumnBatch
     tables = []
     row_counts = []
     for p in paths:
@@ -65,3 +101,61 @@ def read_files_batch(paths: List[str], columns: Optional[List[str]] = None):
         return ColumnBatch({}), []
     table = pa.concat_tables(tables, promote_options="default")
     return ColumnBatch.from_arrow(table), row_counts
+
+
+def read_files_batch_device(paths: List[str], device,
+                            columns: Optional[List[str]] = None):
+    """Device-decoded parquet read (K1): raw file bytes -> HBM -> the
+    unaligned-copy decode kernel assembles each column.  Falls back to the
+    host path + H2D for non-native files.  Returns (ColumnBatch on
+    ``device``, per-file row counts)."""
+    import numpy as np
+    import torch as _torch
+    from ..execution.columnar import ColumnBatch
+    from ..ops import native as native_ext
+    from .native_parquet import read_native_layout
+
+    layouts = []
+    for p in paths:
+        lay = read_native_layout(p, columns)
+        if lay is None:
+            batch, row_counts = read_files_batch(paths, columns)
+            return batch.to(device), row_counts
+        layouts.append(lay)
+
+    ext = native_ext.ext()
+    # column structure from the first file
+    names = [c.name for c in layouts[0][1]]
+    if columns is not None:
+        order = {c.lower(): i for i, c in enumerate(columns)}
+        names.sort(key=lambda n: order.get(n.lower(), 99))
+    dtypes = {c.name: c.np_dtype for c in layouts[0][1]}
+    totals = {n: 0 for n in names}
+    row_counts = []
+    for _, chunks in layouts:
+        nrows = chunks[0].num_values if chunks else 0
+        row_counts.append(nrows)
+        for c in chunks:
+            totals[c.name] += c.num_values
+
+    np_to_torch = {np.dtype("int64"): _torch.int64,
+                   np.dtype("int32"): _torch.int32,
+                   np.dtype("float64"): _torch.float64,
+                   np.dtype("float32"): _torch.float32}
+    out = {n: _torch.empty(totals[n], dtype=np_to_torch[dtypes[n]],
+                           device=device) for n in names}
+    written = {n: 0 for n in names}
+    for data, chunks in layouts:
+        # upload raw bytes once per file (+4B slack for the decode kernel)
+        buf = _torch.empty(len(data) + 4, dtype=_torch.uint8)
+        buf[:len(data)] = _torch.frombuffer(bytearray(data),
+                                            dtype=_torch.uint8)
+        dev_bytes = buf.to(device, non_blocking=True)
+        for c in chunks:
+            itemsize = c.np_dtype.itemsize
+            for off, nv in c.pages:
+                ext.copy_unaligned(dev_bytes, off, out[c.name],
+                                   written[c.name] * itemsize,
+                                   nv * itemsize)
+                written[c.name] += nv
+    return ColumnBatch(out), row_counts
