@@ -179,6 +179,41 @@ class Executor:
 
         # -- device-resident load, cached (288 GB HBM keeps indexes hot) --
         cache = self.session.index_data_cache()
+
+        # equality bucket pruning, cold-path aware: if the full index is
+        # already cached we slice it; otherwise we only LOAD the single
+        # matching bucket (first-query latency = one bucket, not the
+        # whole index)
+        prune_bucket = None
+        if eq_prune is not None:
+            col_name, value = eq_prune
+            prune_bucket = self._bucket_of_value(index, col_name, value,
+                                                 num_buckets)
+            full_files = [p for b in wanted_buckets
+                          for p in sorted(by_bucket[b])]
+            full_key = (cache.key(entry, full_files, read_cols)
+                        if cache else None)
+            full_cached = cache.get(full_key) if cache else None
+            if full_cached is not None:
+                batch, seg = full_cached
+                self.stats.record("IndexScan(cached)")
+                self.stats.bucket_pruned_files += sum(
+                    len(by_bucket[x]) for x in wanted_buckets
+                    if x != prune_bucket)
+                b = prune_bucket
+                if b in wanted_buckets:
+                    batch = batch.slice(int(seg[b]), int(seg[b + 1]))
+                else:
+                    batch = batch.slice(0, 0)
+                batch = self._apply_excluded(plan, batch)
+                return batch.select(plan.columns), None
+            # cold: restrict the load to the matching bucket
+            self.stats.bucket_pruned_files += sum(
+                len(by_bucket[x]) for x in wanted_buckets
+                if x != prune_bucket)
+            wanted_buckets = ([prune_bucket]
+                              if prune_bucket in by_bucket else [])
+
         cache_files = [p for b in wanted_buckets
                        for p in sorted(by_bucket[b])]
         key = cache.key(entry, cache_files, read_cols) if cache else None
@@ -209,19 +244,7 @@ class Executor:
             if cache:
                 cache.put(key, batch, seg)
 
-        # equality bucket pruning: slice the single matching bucket range
-        if eq_prune is not None:
-            col_name, value = eq_prune
-            b = self._bucket_of_value(index, col_name, value, num_buckets)
-            self.stats.bucket_pruned_files += sum(
-                len(by_bucket[x]) for x in wanted_buckets if x != b)
-            if b in wanted_buckets:
-                batch = batch.slice(int(seg[b]), int(seg[b + 1]))
-            else:
-                batch = batch.slice(0, 0)
-            seg_local = None
-        else:
-            seg_local = seg
+        seg_local = None if eq_prune is not None else seg
 
         # lineage delete filter (Hybrid Scan deletes, K7)
         if plan.excluded_source_file_ids and batch.num_rows:
@@ -245,6 +268,18 @@ class Executor:
         if plan.use_bucket_spec and eq_prune is None:
             return batch, seg_local
         return batch, None
+
+    def _apply_excluded(self, plan: IndexScan, batch: ColumnBatch
+                        ) -> ColumnBatch:
+        """Lineage delete filter for the cached-slice fast path."""
+        if not plan.excluded_source_file_ids or batch.num_rows == 0:
+            return batch
+        ids = torch.tensor(sorted(plan.excluded_source_file_ids),
+                           dtype=torch.int64, device=batch.device)
+        lineage = batch.tensor(IndexConstants.DATA_FILE_NAME_ID_COLUMN)
+        keep = ~ops.isin_sorted(lineage, ids)
+        self.stats.record("LineageFilter")
+        return batch.gather(torch.nonzero(keep, as_tuple=False).flatten())
 
     def _zorder_prune(self, cond: Expr, scan: IndexScan):
         """Prune a z-order IndexScan's file list using Parquet column

@@ -328,12 +328,14 @@ class ColumnChunkLayout:
 
 
 def read_native_layout(path: str,
-                       columns: Optional[List[str]] = None
+                       columns: Optional[List[str]] = None,
+                       data: Optional[bytes] = None
                        ) -> Optional[Tuple[bytes, List[ColumnChunkLayout]]]:
     """If ``path`` decodes natively (uncompressed PLAIN pages of numeric
     columns — our writer's files and pyarrow's NONE/PLAIN files), return
     (raw file bytes, per-column page layouts); else None (caller falls
-    back to pyarrow)."""
+    back to pyarrow).  ``data`` can supply pre-read file bytes (any
+    buffer protocol object) to avoid a second disk read."""
     import pyarrow.parquet as pq
     try:
         md = pq.ParquetFile(path).metadata
@@ -341,8 +343,9 @@ def read_native_layout(path: str,
         return None
     if md.num_row_groups != 1:
         return None
-    with open(path, "rb") as f:
-        data = f.read()
+    if data is None:
+        with open(path, "rb") as f:
+            data = f.read()
     rg = md.row_group(0)
     out: List[ColumnChunkLayout] = []
     want = {c.lower() for c in columns} if columns is not None else None

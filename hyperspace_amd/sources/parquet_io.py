@@ -115,24 +115,33 @@ def read_files_batch_device(paths: List[str], device,
     from ..ops import native as native_ext
     from .native_parquet import read_native_layout
 
+    # read each file ONCE into a pinned buffer (+4B slack for the decode
+    # kernel), parse the layout from the same bytes, upload async
     layouts = []
+    bufs = []
     for p in paths:
-        lay = read_native_layout(p, columns)
+        size = os.path.getsize(p)
+        buf = _torch.empty(size + 4, dtype=_torch.uint8, pin_memory=True)
+        view = memoryview(buf.numpy())
+        with open(p, "rb", buffering=0) as f:
+            f.readinto(view[:size])
+        lay = read_native_layout(p, columns, data=view[:size])
         if lay is None:
             batch, row_counts = read_files_batch(paths, columns)
             return batch.to(device), row_counts
-        layouts.append(lay)
+        layouts.append(lay[1])
+        bufs.append(buf)
 
     ext = native_ext.ext()
     # column structure from the first file
-    names = [c.name for c in layouts[0][1]]
+    names = [c.name for c in layouts[0]]
     if columns is not None:
         order = {c.lower(): i for i, c in enumerate(columns)}
         names.sort(key=lambda n: order.get(n.lower(), 99))
-    dtypes = {c.name: c.np_dtype for c in layouts[0][1]}
+    dtypes = {c.name: c.np_dtype for c in layouts[0]}
     totals = {n: 0 for n in names}
     row_counts = []
-    for _, chunks in layouts:
+    for chunks in layouts:
         nrows = chunks[0].num_values if chunks else 0
         row_counts.append(nrows)
         for c in chunks:
@@ -145,11 +154,7 @@ def read_files_batch_device(paths: List[str], device,
     out = {n: _torch.empty(totals[n], dtype=np_to_torch[dtypes[n]],
                            device=device) for n in names}
     written = {n: 0 for n in names}
-    for data, chunks in layouts:
-        # upload raw bytes once per file (+4B slack for the decode kernel)
-        buf = _torch.empty(len(data) + 4, dtype=_torch.uint8)
-        buf[:len(data)] = _torch.frombuffer(bytearray(data),
-                                            dtype=_torch.uint8)
+    for buf, chunks in zip(bufs, layouts):
         dev_bytes = buf.to(device, non_blocking=True)
         for c in chunks:
             itemsize = c.np_dtype.itemsize
