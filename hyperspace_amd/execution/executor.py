@@ -485,19 +485,11 @@ class Executor:
                      num_buckets: int
                      ) -> Tuple[ColumnBatch, torch.Tensor]:
         """Hash-repartition + per-bucket sort (K2+K3 on the fly)."""
+        from ..index.covering.index import sort_by_bucket_and_keys
         keys = [batch.tensor(c) for c in bucket_cols]
         bucket_ids = ops.murmur3_bucket(keys, num_buckets)
-        sort_key = ops.normalize_key(batch.tensor(bucket_cols[0]))
-        perm = ops.sort_perm(sort_key)
-        b_sorted = bucket_ids.to(torch.int64)[perm]
-        perm2 = ops.sort_perm(ops.normalize_key(b_sorted))
-        perm = perm[perm2]
-        batch = batch.gather(perm)
-        counts = torch.bincount(bucket_ids.to(torch.int64).cpu(),
-                                minlength=num_buckets)
-        seg = torch.zeros(num_buckets + 1, dtype=torch.int64)
-        seg[1:] = torch.cumsum(counts, 0)
-        return batch, seg
+        return sort_by_bucket_and_keys(batch, bucket_ids, bucket_cols,
+                                       num_buckets)
 
 
 # ---------------------------------------------------------------------------

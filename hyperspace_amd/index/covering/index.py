@@ -210,13 +210,17 @@ def sort_by_bucket_and_keys(batch: ColumnBatch, bucket_ids: torch.Tensor,
     """Stable sort rows by (bucket, key columns); returns the reordered
     batch and per-bucket segment offsets (num_buckets+1)."""
     perm = _multi_key_sort_perm(batch, key_cols)
-    b64 = bucket_ids.to(torch.int64)[perm]
-    _, perm2 = ops.sort_pairs(ops.normalize_key(b64), perm)
+    b64 = ops.gather_rows(bucket_ids.to(torch.int64), perm)
+    sorted_bkeys, perm2 = ops.sort_pairs(ops.normalize_key(b64), perm)
     batch = batch.gather(perm2)
-    counts = torch.bincount(bucket_ids.to(torch.int64).cpu(),
-                            minlength=num_buckets)
-    seg = torch.zeros(num_buckets + 1, dtype=torch.int64)
-    seg[1:] = torch.cumsum(counts, 0)
+    # segment offsets via searchsorted on the sorted bucket keys (device
+    # side; avoids a full-column D2H)
+    probes = ops.cpu_ref.normalize_key(
+        torch.arange(num_buckets + 1, dtype=torch.int64)).to(
+            sorted_bkeys.device)
+    sortable = sorted_bkeys ^ (-0x8000000000000000)
+    sortable_probes = probes ^ (-0x8000000000000000)
+    seg = torch.searchsorted(sortable, sortable_probes).cpu()
     return batch, seg
 
 
