@@ -123,6 +123,15 @@ class CreateAction(Action):
             written = index.write(ctx, batch)
         written = _all_written_files(data_path, written)
 
+        # provider property enrichment (delta tables record the
+        # index->table version history, reference CreateActionBase ->
+        # FileBasedRelationMetadata.enrichIndexProperties)
+        relation0 = self.df.plan.collect_leaves()[0].relation
+        if hasattr(relation0, "enrich_index_properties"):
+            index = index.with_new_properties(
+                relation0.enrich_index_properties(
+                    index.properties, (self.base_id or 0) + 2))
+
         from ..plan.nodes import Scan
         relation = self.df.plan.collect_leaves()[0].relation
         rel_meta = relation.create_relation_metadata(tracker)

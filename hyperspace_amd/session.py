@@ -64,6 +64,15 @@ class HyperspaceSession:
         from .sources.parquet_source import ParquetRelation
         return DataFrame(self, Scan(ParquetRelation(list(paths))))
 
+    def read_delta(self, path: str, version_as_of=None):
+        """Read a delta-style transactional table (time travel via
+        ``version_as_of``)."""
+        from .dataframe import DataFrame
+        from .plan.nodes import Scan
+        from .sources.delta_source import DeltaTableRelation
+        return DataFrame(self, Scan(DeltaTableRelation(path,
+                                                       version_as_of)))
+
     def index_manager(self):
         from .index_management import CachingIndexCollectionManager
         if not hasattr(self, "_index_manager"):

@@ -1,0 +1,121 @@
+"""Delta-style table source tests (reference: DeltaLakeIntegrationTest
+behaviors: version signatures, time travel, refresh after commits,
+deltaVersions history property)."""
+
+import numpy as np
+import pyarrow as pa
+import pyarrow.parquet as pq
+import pytest
+import torch
+
+import hyperspace_amd as hs
+from hyperspace_amd.execution.columnar import ColumnBatch
+from hyperspace_amd.exceptions import HyperspaceException
+from hyperspace_amd.plan.nodes import IndexScan
+from hyperspace_amd.sources.delta_source import (DeltaTable,
+                                                 DeltaTableRelation)
+
+
+def _batch(rng, n=5000, key_hi=500):
+    return ColumnBatch({
+        "key": torch.from_numpy(rng.integers(0, key_hi, n)),
+        "val": torch.from_numpy(rng.random(n))})
+
+
+@pytest.fixture
+def env(tmp_path, monkeypatch):
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "indexes"))
+    rng = np.random.default_rng(81)
+    table = DeltaTable.create(str(tmp_path / "dtable"))
+    table.append_batch(_batch(rng))
+    table.append_batch(_batch(rng))
+    session = hs.HyperspaceSession(device="cpu")
+    session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 4)
+    return session, hs.Hyperspace(session), table, rng
+
+
+def test_log_and_time_travel(env, tmp_path):
+    session, h, table, rng = env
+    assert table.version == 2
+    v2_files = table.files_at(2)
+    assert len(v2_files) == 2
+    table.append_batch(_batch(rng))
+    assert table.version == 3
+    assert len(table.files_at()) == 3
+    assert len(table.files_at(2)) == 2
+    # remove
+    table.remove_files([v2_files[0].name])
+    assert len(table.files_at()) == 2
+    assert len(table.files_at(3)) == 3  # history intact
+
+
+def test_signature_changes_per_version(env):
+    session, h, table, rng = env
+    rel = DeltaTableRelation(table.path)
+    s1 = rel.signature()
+    table.append_batch(_batch(rng))
+    assert DeltaTableRelation(table.path).signature() != s1
+    # pinned snapshot keeps its signature
+    pinned = DeltaTableRelation(table.path, version_as_of=2)
+    assert pinned.signature() == s1
+
+
+def test_index_on_delta_and_query(env):
+    session, h, table, rng = env
+    df = session.read_delta(table.path)
+    h.create_index(df, hs.CoveringIndexConfig("dix", ["key"], ["val"]))
+    entry = session.index_manager().get_index("dix")
+    assert entry.relations[0].fileFormat == "delta"
+    # deltaVersions history property recorded
+    assert "deltaVersions" in entry.properties
+    assert entry.properties["deltaVersions"].endswith(":2")
+
+    session.enable_hyperspace()
+    q = df.filter("key = 42").select("key", "val")
+    plan = q.optimized_plan()
+    assert any(isinstance(l, IndexScan) for l in plan.collect_leaves())
+    accel = q.collect()
+    session.disable_hyperspace()
+    assert accel.num_rows == q.collect().num_rows
+
+
+def test_commit_invalidates_index(env):
+    session, h, table, rng = env
+    df = session.read_delta(table.path)
+    h.create_index(df, hs.CoveringIndexConfig("dix", ["key"], ["val"]))
+    table.append_batch(_batch(rng))
+    session.enable_hyperspace()
+    plan = df.filter("key = 42").select("key", "val").optimized_plan()
+    # version changed, hybrid scan off -> no rewrite
+    assert not any(isinstance(l, IndexScan)
+                   for l in plan.collect_leaves())
+
+
+def test_refresh_after_commit(env):
+    session, h, table, rng = env
+    df = session.read_delta(table.path)
+    h.create_index(df, hs.CoveringIndexConfig("dix", ["key"], ["val"]))
+    table.append_batch(_batch(rng))
+    h.refresh_index("dix", "full")
+    entry = session.index_manager().get_index("dix")
+    assert len(entry.source_file_infos()) == 3
+    session.enable_hyperspace()
+    plan = df.filter("key = 42").select("key", "val").optimized_plan()
+    assert any(isinstance(l, IndexScan) for l in plan.collect_leaves())
+
+
+def test_time_travel_query_uses_pinned_snapshot(env):
+    session, h, table, rng = env
+    df_v2 = session.read_delta(table.path, version_as_of=2)
+    n_v2 = df_v2.count()
+    table.append_batch(_batch(rng))
+    assert df_v2.count() == n_v2  # pinned
+    assert session.read_delta(table.path).count() > n_v2
+
+
+def test_concurrent_commit_collision(env, tmp_path):
+    session, h, table, rng = env
+    v = table.version
+    table._commit(v + 1, [], [])
+    with pytest.raises(HyperspaceException, match="lost race"):
+        table._commit(v + 1, [], [])
