@@ -467,6 +467,7 @@ class _StateFlipAction(Action):
 
 class DeleteAction(_StateFlipAction):
     """Soft delete: ACTIVE -> DELETED (reference: actions/DeleteAction)."""
+    collective_op = False
     transient_state = States.DELETING
     final_state = States.DELETED
 
@@ -481,6 +482,7 @@ class DeleteAction(_StateFlipAction):
 
 class RestoreAction(_StateFlipAction):
     """DELETED -> ACTIVE (reference: actions/RestoreAction)."""
+    collective_op = False
     transient_state = States.RESTORING
     final_state = States.ACTIVE
 
@@ -496,6 +498,7 @@ class RestoreAction(_StateFlipAction):
 class VacuumAction(_StateFlipAction):
     """Hard delete all files of a DELETED index -> DOESNOTEXIST
     (reference: actions/VacuumAction)."""
+    collective_op = False
     transient_state = States.VACUUMING
     final_state = States.DOESNOTEXIST
 
@@ -525,6 +528,7 @@ class VacuumAction(_StateFlipAction):
 class VacuumOutdatedAction(_StateFlipAction):
     """GC data versions not referenced by the latest content; keep index
     ACTIVE (reference: actions/VacuumOutdatedAction.scala:34-144)."""
+    collective_op = False
     transient_state = States.VACUUMINGOUTDATED
     final_state = States.ACTIVE
 

@@ -31,6 +31,11 @@ class Action(ABC):
         self.log_manager = log_manager
         self.base_id: Optional[int] = None
 
+    # data-plane ops (create/refresh/optimize builds) are collective —
+    # every rank participates; metadata/file-deletion ops run on rank 0
+    # only in distributed mode
+    collective_op = True
+
     # -- abstract pieces ---------------------------------------------------
     @property
     @abstractmethod
@@ -74,11 +79,21 @@ class Action(ABC):
             self.begin()
         dc.barrier()
         no_changes = False
-        try:
-            self.op()
-        except NoChangesException:
-            # all ranks see the same source state, so this is collective
-            no_changes = True
+        if self.collective_op or rank0:
+            try:
+                self.op()
+            except NoChangesException:
+                # all ranks see the same source state -> collective signal
+                no_changes = True
+        if self.collective_op:
+            dc.barrier()
+        else:
+            # non-collective: broadcast the no-op outcome from rank 0
+            import torch
+            import torch.distributed as dist
+            t = torch.tensor([1 if no_changes else 0])
+            dist.broadcast(t, src=0)
+            no_changes = bool(t[0])
         dc.barrier()
         if rank0:
             if no_changes:
