@@ -227,17 +227,31 @@ def sort_by_bucket_and_keys(batch: ColumnBatch, bucket_ids: torch.Tensor,
 def write_bucketed(batch: ColumnBatch, seg: torch.Tensor, out_dir: str,
                    num_buckets: int, task_id: int = 0) -> List[str]:
     """Write per-bucket parquet files honoring the bucket-id filename
-    contract.  Empty buckets produce no file (as in Spark)."""
+    contract.  Empty buckets produce no file (as in Spark).
+
+    Files are written by a thread pool: the native encoder's byte
+    assembly (numpy tobytes) and os.write both release the GIL, so the
+    200-file write overlaps to page-cache speed.
+    """
+    from concurrent.futures import ThreadPoolExecutor
     host = batch.to("cpu") if batch.device.type == "cuda" else batch
-    written: List[str] = []
+    jobs = []
     for b in range(num_buckets):
         lo, hi = int(seg[b]), int(seg[b + 1])
         if hi <= lo:
             continue
-        out = os.path.join(out_dir, bucket_file_name(task_id, b))
+        jobs.append((b, lo, hi,
+                     os.path.join(out_dir, bucket_file_name(task_id, b))))
+
+    def write_one(job):
+        b, lo, hi, out = job
         write_batch_parquet(host.slice(lo, hi), out)
-        written.append(out)
-    return written
+        return out
+
+    if len(jobs) <= 2:
+        return [write_one(j) for j in jobs]
+    with ThreadPoolExecutor(max_workers=16) as pool:
+        return list(pool.map(write_one, jobs))
 
 
 register_derived_dataset(COVERING_INDEX_TYPE, CoveringIndex)

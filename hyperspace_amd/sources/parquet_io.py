@@ -116,21 +116,29 @@ def read_files_batch_device(paths: List[str], device,
     from .native_parquet import read_native_layout
 
     # read each file ONCE into a pinned buffer (+4B slack for the decode
-    # kernel), parse the layout from the same bytes, upload async
-    layouts = []
-    bufs = []
-    for p in paths:
+    # kernel), parse the layout from the same bytes, upload async; file
+    # reads overlap on a thread pool (readinto releases the GIL)
+    from concurrent.futures import ThreadPoolExecutor
+
+    def load_one(p):
         size = os.path.getsize(p)
         buf = _torch.empty(size + 4, dtype=_torch.uint8, pin_memory=True)
         view = memoryview(buf.numpy())
         with open(p, "rb", buffering=0) as f:
             f.readinto(view[:size])
         lay = read_native_layout(p, columns, data=view[:size])
-        if lay is None:
-            batch, row_counts = read_files_batch(paths, columns)
-            return batch.to(device), row_counts
-        layouts.append(lay[1])
-        bufs.append(buf)
+        return buf, (lay[1] if lay is not None else None)
+
+    if len(paths) > 2:
+        with ThreadPoolExecutor(max_workers=16) as pool:
+            loaded = list(pool.map(load_one, paths))
+    else:
+        loaded = [load_one(p) for p in paths]
+    if any(lay is None for _, lay in loaded):
+        batch, row_counts = read_files_batch(paths, columns)
+        return batch.to(device), row_counts
+    bufs = [b for b, _ in loaded]
+    layouts = [lay for _, lay in loaded]
 
     ext = native_ext.ext()
     # column structure from the first file
