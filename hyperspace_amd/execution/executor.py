@@ -207,12 +207,14 @@ class Executor:
                     batch = batch.slice(0, 0)
                 batch = self._apply_excluded(plan, batch)
                 return batch.select(plan.columns), None
-            # cold: restrict the load to the matching bucket
+            # cold: restrict the load to the matching bucket (membership
+            # checked against the OWNERSHIP-filtered set so each rank
+            # serves only its own buckets at N>1)
             self.stats.bucket_pruned_files += sum(
                 len(by_bucket[x]) for x in wanted_buckets
                 if x != prune_bucket)
             wanted_buckets = ([prune_bucket]
-                              if prune_bucket in by_bucket else [])
+                              if prune_bucket in wanted_buckets else [])
 
         cache_files = [p for b in wanted_buckets
                        for p in sorted(by_bucket[b])]

@@ -45,6 +45,14 @@ def _worker(rank, world, tmpdir, rdv_file, results):
         files = entry.content.os_files()
 
         session.enable_hyperspace()
+        session.conf.set(
+            hs.IndexConstants.INDEX_FILTER_RULE_USE_BUCKET_SPEC, True)
+        fq = fact.filter("key = 777").select("key", "val")
+        fex = Executor(session)
+        fout = fex.execute(fq.optimized_plan())
+        t_f = torch.tensor([fout.num_rows])
+        dist.all_reduce(t_f)
+
         q = fact.select("key", "val").join(dim.select("key", "status"),
                                            on="key")
         plan = q.optimized_plan()
@@ -64,6 +72,8 @@ def _worker(rank, world, tmpdir, rdv_file, results):
             "shuffles": ex.stats.shuffles,
             "local_rows": local_rows,
             "total_rows": int(t[0]),
+            "filter_total": int(t_f[0]),
+            "filter_local": fout.num_rows,
         }
     finally:
         dist.destroy_process_group()
@@ -120,6 +130,13 @@ def test_distributed_build_and_join(dist_env, tmp_path):
     assert r0["total_rows"] == expected
     assert r0["local_rows"] + r1["local_rows"] == expected
     assert 0 < r0["local_rows"] < expected
+    # bucket-pruned equality filter: exactly one rank owns key 777's
+    # bucket; the global count matches an unindexed count
+    t = pq.read_table(str(tmp_path / "fact"), columns=["key"])
+    expected_f = int((t.column("key").to_numpy() == 777).sum())
+    assert r0["filter_total"] == expected_f
+    assert sorted([r0["filter_local"], r1["filter_local"]]) == \
+        sorted([0, expected_f]) or expected_f == 0
 
 
 def test_distributed_index_readable_locally(dist_env, tmp_path):
