@@ -73,6 +73,15 @@ class HyperspaceSession:
         return DataFrame(self, Scan(DeltaTableRelation(path,
                                                        version_as_of)))
 
+    def read_iceberg(self, path: str, snapshot_id=None):
+        """Read an iceberg-style snapshot table (time travel via
+        ``snapshot_id``)."""
+        from .dataframe import DataFrame
+        from .plan.nodes import Scan
+        from .sources.iceberg_source import IcebergTableRelation
+        return DataFrame(self, Scan(IcebergTableRelation(path,
+                                                         snapshot_id)))
+
     def index_manager(self):
         from .index_management import CachingIndexCollectionManager
         if not hasattr(self, "_index_manager"):

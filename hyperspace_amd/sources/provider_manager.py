@@ -20,12 +20,11 @@ class FileBasedSourceProviderManager:
                  = None):
         self.providers: List[FileBasedSourceProvider] = providers or [
             ParquetSourceProvider()]
-        # delta/iceberg-style providers register here
-        try:
-            from .delta_source import DeltaTableSourceProvider
-            self.providers.append(DeltaTableSourceProvider())
-        except ImportError:
-            pass
+        # table-format providers (delta-style log, iceberg-style snapshots)
+        from .delta_source import DeltaTableSourceProvider
+        from .iceberg_source import IcebergTableSourceProvider
+        self.providers.append(DeltaTableSourceProvider())
+        self.providers.append(IcebergTableSourceProvider())
 
     def register(self, provider: FileBasedSourceProvider) -> None:
         self.providers.append(provider)

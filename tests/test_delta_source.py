@@ -119,3 +119,68 @@ def test_concurrent_commit_collision(env, tmp_path):
     table._commit(v + 1, [], [])
     with pytest.raises(HyperspaceException, match="lost race"):
         table._commit(v + 1, [], [])
+
+
+# ---------------------------------------------------------------------------
+# Iceberg-style snapshot source
+# ---------------------------------------------------------------------------
+
+from hyperspace_amd.sources.iceberg_source import (IcebergTable,
+                                                   IcebergTableRelation)
+
+
+@pytest.fixture
+def ice_env(tmp_path, monkeypatch):
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "indexes"))
+    rng = np.random.default_rng(91)
+    table = IcebergTable.create(str(tmp_path / "itable"))
+    table.append_batch(_batch(rng))
+    table.append_batch(_batch(rng))
+    session = hs.HyperspaceSession(device="cpu")
+    session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 4)
+    return session, hs.Hyperspace(session), table, rng
+
+
+def test_iceberg_snapshots_and_signature(ice_env):
+    session, h, table, rng = ice_env
+    rel = IcebergTableRelation(table.path)
+    s1 = rel.signature()
+    snap1 = table.snapshot_id
+    assert len(rel.all_files()) == 2
+    table.append_batch(_batch(rng))
+    assert IcebergTableRelation(table.path).signature() != s1
+    # pinned snapshot time travel
+    pinned = IcebergTableRelation(table.path, snapshot_id=snap1)
+    assert len(pinned.all_files()) == 2
+    assert pinned.signature() == s1
+
+
+def test_index_on_iceberg_and_query(ice_env):
+    session, h, table, rng = ice_env
+    df = session.read_iceberg(table.path)
+    h.create_index(df, hs.CoveringIndexConfig("iix", ["key"], ["val"]))
+    entry = session.index_manager().get_index("iix")
+    assert entry.relations[0].fileFormat == "iceberg"
+    session.enable_hyperspace()
+    q = df.filter("key = 42").select("key", "val")
+    plan = q.optimized_plan()
+    assert any(isinstance(l, IndexScan) for l in plan.collect_leaves())
+    accel = q.collect()
+    session.disable_hyperspace()
+    assert accel.num_rows == q.collect().num_rows
+    # a new snapshot invalidates the index signature
+    table.append_batch(_batch(rng))
+    session.enable_hyperspace()
+    plan = df.filter("key = 42").select("key", "val").optimized_plan()
+    assert not any(isinstance(l, IndexScan)
+                   for l in plan.collect_leaves())
+
+
+def test_iceberg_refresh_after_snapshot(ice_env):
+    session, h, table, rng = ice_env
+    df = session.read_iceberg(table.path)
+    h.create_index(df, hs.CoveringIndexConfig("iix", ["key"], ["val"]))
+    table.append_batch(_batch(rng))
+    h.refresh_index("iix", "incremental")
+    entry = session.index_manager().get_index("iix")
+    assert len(entry.source_file_infos()) == 3
