@@ -40,6 +40,29 @@ def _operator_counts(plan: LogicalPlan) -> Dict[str, int]:
     return counts
 
 
+class DisplayMode:
+    """Rendering modes for explain output (reference:
+    index/plananalysis/ BufferStream/DisplayMode — console ANSI,
+    plaintext markers, HTML tags; selected by conf
+    spark.hyperspace.explain.displayMode)."""
+
+    def __init__(self, highlight_open: str, highlight_close: str,
+                 newline: str = "\n"):
+        self.highlight_open = highlight_open
+        self.highlight_close = highlight_close
+        self.newline = newline
+
+    @staticmethod
+    def for_conf(conf) -> "DisplayMode":
+        from ..config import IndexConstants
+        mode = (conf.get(IndexConstants.DISPLAY_MODE) or "plaintext").lower()
+        if mode == "console":
+            return DisplayMode("\x1b[92m", "\x1b[0m")
+        if mode == "html":
+            return DisplayMode("<b>", "</b>", newline="<br>")
+        return DisplayMode("<----", "---->")
+
+
 class PlanAnalyzer:
     def __init__(self, session):
         self.session = session
@@ -54,17 +77,33 @@ class PlanAnalyzer:
         finally:
             self.session._hyperspace_enabled = was_enabled
 
+        mode = DisplayMode.for_conf(self.session.conf)
+        orig_lines = set(original.pretty().splitlines())
+        new_lines = set(rewritten.pretty().splitlines())
+
+        def highlight(plan_str, other):
+            out = []
+            for ln in plan_str.splitlines():
+                if ln not in other:
+                    out.append(f"{mode.highlight_open}{ln.lstrip()}"
+                               f"{mode.highlight_close}".rjust(
+                                   len(ln) + len(mode.highlight_open)
+                                   + len(mode.highlight_close)))
+                else:
+                    out.append(ln)
+            return "\n".join(out)
+
         lines: List[str] = []
         bar = "=" * 64
         lines.append(bar)
         lines.append("Plan with indexes:")
         lines.append(bar)
-        lines.append(rewritten.pretty())
+        lines.append(highlight(rewritten.pretty(), orig_lines))
         lines.append("")
         lines.append(bar)
         lines.append("Plan without indexes:")
         lines.append(bar)
-        lines.append(original.pretty())
+        lines.append(highlight(original.pretty(), new_lines))
         lines.append("")
         lines.append(bar)
         lines.append("Indexes used:")
@@ -78,6 +117,9 @@ class PlanAnalyzer:
                 loc = os.path.dirname(infos[0].name)
             lines.append(f"{entry.name}:{loc}")
         lines.append("")
+        out = "\n".join(lines)
+        tail: List[str] = []
+        lines = tail  # verbose section appended below
         if verbose:
             lines.append(bar)
             lines.append("Physical operator stats:")
@@ -91,4 +133,7 @@ class PlanAnalyzer:
                 w = with_counts.get(op, 0)
                 wo = without_counts.get(op, 0)
                 lines.append(f"{op:<28}{w:>6}{wo:>9}{w - wo:>6}")
-        return "\n".join(lines)
+        result = out + ("\n" + "\n".join(tail) if tail else "")
+        if mode.newline != "\n":
+            result = result.replace("\n", mode.newline)
+        return result

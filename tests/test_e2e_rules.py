@@ -187,3 +187,18 @@ def test_score_prefers_join_over_filter(env):
     scans = [l for l in plan.collect_leaves() if isinstance(l, IndexScan)]
     assert {s.entry.name for s in scans} == {"lidx", "ridx"}
     assert all(s.use_bucket_spec for s in scans)
+
+
+def test_explain_display_modes(env):
+    session, h, left, _ = env
+    h.create_index(left, hs.CoveringIndexConfig("fidx", ["qty"], ["price"]))
+    q = left.filter("qty = 7").select("qty", "price")
+    session.conf.set(hs.IndexConstants.DISPLAY_MODE, "html")
+    out = h.explain(q)
+    assert "<b>" in out and "<br>" in out
+    session.conf.set(hs.IndexConstants.DISPLAY_MODE, "console")
+    out = h.explain(q)
+    assert "\x1b[92m" in out
+    session.conf.set(hs.IndexConstants.DISPLAY_MODE, "plaintext")
+    out = h.explain(q)
+    assert "<----" in out and "IndexScan" in out

@@ -125,3 +125,17 @@ def test_copy_with_update():
     assert [f.id for f in updated.appended_files()] == [9]
     assert entry.appended_files() == []  # original untouched
     assert updated.signature.value == "v2"
+
+
+def test_cache_with_transform():
+    from hyperspace_amd.utils.cache import CacheWithTransform
+    calls = []
+    state = {"k": 1}
+    c = CacheWithTransform(lambda: state["k"], lambda k: calls.append(k)
+                           or k * 10)
+    assert c.load() == 10
+    assert c.load() == 10
+    assert calls == [1]       # memoized
+    state["k"] = 2
+    assert c.load() == 20     # conf change invalidates
+    assert calls == [1, 2]
