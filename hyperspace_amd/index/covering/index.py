@@ -234,7 +234,7 @@ def write_bucketed(batch: ColumnBatch, seg: torch.Tensor, out_dir: str,
     200-file write overlaps to page-cache speed.
     """
     from concurrent.futures import ThreadPoolExecutor
-    host = batch.to("cpu") if batch.device.type == "cuda" else batch
+    on_device = batch.device.type == "cuda"
     jobs = []
     for b in range(num_buckets):
         lo, hi = int(seg[b]), int(seg[b + 1])
@@ -245,7 +245,12 @@ def write_bucketed(batch: ColumnBatch, seg: torch.Tensor, out_dir: str,
 
     def write_one(job):
         b, lo, hi, out = job
-        write_batch_parquet(host.slice(lo, hi), out)
+        piece = batch.slice(lo, hi)
+        if on_device:
+            # per-bucket D2H inside the worker: copies overlap other
+            # workers' os.write calls (both release the GIL)
+            piece = piece.to("cpu")
+        write_batch_parquet(piece, out)
         return out
 
     if len(jobs) <= 2:
