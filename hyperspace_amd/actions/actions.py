@@ -66,6 +66,24 @@ def _tracker_from_entry(entry: IndexLogEntry) -> FileIdTracker:
     return tracker
 
 
+def _rebuild_config(name: str, index):
+    """Reconstruct the matching config for a full rebuild of any index
+    kind (refresh full must not assume a covering index)."""
+    kind = index.kind
+    if kind == "CoveringIndex":
+        from ..index.covering.config import CoveringIndexConfig
+        return CoveringIndexConfig(name, index.indexed_columns,
+                                   index.included_columns)
+    if kind == "ZOrderCoveringIndex":
+        from ..index.zorder.config import ZOrderCoveringIndexConfig
+        return ZOrderCoveringIndexConfig(name, index.indexed_columns,
+                                         index.included_columns)
+    if kind == "DataSkippingIndex":
+        from ..index.dataskipping.config import DataSkippingIndexConfig
+        return DataSkippingIndexConfig(name, *index.sketches)
+    raise HyperspaceException(f"Cannot rebuild index kind {kind}")
+
+
 # ---------------------------------------------------------------------------
 # Create
 # ---------------------------------------------------------------------------
@@ -250,10 +268,7 @@ class RefreshFullAction(RefreshActionBase):
         from ..plan.nodes import Scan
         df = DataFrame(self.session, Scan(relation))
         index = self.previous.derivedDataset
-        from ..index.covering.config import CoveringIndexConfig
-        config = CoveringIndexConfig(
-            self.previous.name, index.indexed_columns,
-            index.included_columns)
+        config = _rebuild_config(self.previous.name, index)
         with self.session.with_rule_disabled():
             new_index, batch = config.create_index(
                 ctx, df, index.properties)
@@ -294,27 +309,38 @@ class RefreshIncrementalAction(RefreshActionBase):
         from ..index.base import IndexerContext
         ctx = IndexerContext(self.session, tracker, data_path)
 
-        appended_batch = None
-        if appended:
-            relation = self.source_relation()
-            from ..plan.nodes import Scan
-            from ..execution.executor import Executor
-            scan = Scan(relation)
-            ex = Executor(self.session)
-            with self.session.with_rule_disabled():
-                batch = ex._exec_scan(
-                    scan, file_subset=[f.name for f in appended],
-                    lineage_tracker=tracker if index.has_lineage else None)
-                cols = index.indexed_columns + index.included_columns
-                if index.has_lineage:
-                    cols = cols + [IndexConstants.DATA_FILE_NAME_ID_COLUMN]
-                appended_batch = batch.select(cols)
-
         previous_files = [p for p in self.previous.content.os_files()
                           if p.endswith(".parquet")]
-        with self.session.with_rule_disabled():
-            written, kept = index.refresh_incremental(
-                ctx, appended_batch, deleted_ids, previous_files)
+
+        if index.kind == "DataSkippingIndex":
+            # per-file sketch rows: appended files are sketched directly,
+            # deleted files drop their rows
+            with self.session.with_rule_disabled():
+                written, kept = index.refresh_incremental_files(
+                    ctx, [f.name for f in appended], deleted_ids,
+                    previous_files)
+        else:
+            appended_batch = None
+            if appended:
+                relation = self.source_relation()
+                from ..plan.nodes import Scan
+                from ..execution.executor import Executor
+                scan = Scan(relation)
+                ex = Executor(self.session)
+                with self.session.with_rule_disabled():
+                    batch = ex._exec_scan(
+                        scan, file_subset=[f.name for f in appended],
+                        lineage_tracker=tracker if index.has_lineage
+                        else None)
+                    cols = index.indexed_columns + index.included_columns
+                    if index.has_lineage:
+                        cols = cols + [
+                            IndexConstants.DATA_FILE_NAME_ID_COLUMN]
+                    appended_batch = batch.select(cols)
+
+            with self.session.with_rule_disabled():
+                written, kept = index.refresh_incremental(
+                    ctx, appended_batch, deleted_ids, previous_files)
 
         relation = self.source_relation()
         rel_meta = relation.create_relation_metadata(tracker)
@@ -455,6 +481,15 @@ class _StateFlipAction(Action):
             raise HyperspaceException("Index does not exist")
         self.previous = prev
 
+    def validate(self):
+        # re-read at run time: a concurrent action may have completed
+        # between construction and run (validate-then-claim; the log-slot
+        # claim still arbitrates true ties)
+        prev = self.log_manager.get_latest_stable_log()
+        if prev is None:
+            raise HyperspaceException("Index does not exist")
+        self.previous = prev
+
     def op(self):
         pass
 
@@ -472,6 +507,7 @@ class DeleteAction(_StateFlipAction):
     final_state = States.DELETED
 
     def validate(self):
+        _StateFlipAction.validate(self)
         if self.previous.state != States.ACTIVE:
             raise HyperspaceException("Delete requires ACTIVE index")
 
@@ -487,6 +523,7 @@ class RestoreAction(_StateFlipAction):
     final_state = States.ACTIVE
 
     def validate(self):
+        _StateFlipAction.validate(self)
         if self.previous.state != States.DELETED:
             raise HyperspaceException("Restore requires DELETED index")
 
@@ -507,6 +544,7 @@ class VacuumAction(_StateFlipAction):
         self.index_path = index_path
 
     def validate(self):
+        _StateFlipAction.validate(self)
         if self.previous.state != States.DELETED:
             raise HyperspaceException("Vacuum requires DELETED index")
 
@@ -537,6 +575,7 @@ class VacuumOutdatedAction(_StateFlipAction):
         self.data_manager = data_manager
 
     def validate(self):
+        _StateFlipAction.validate(self)
         if self.previous.state != States.ACTIVE:
             raise HyperspaceException(
                 "VacuumOutdated requires ACTIVE index")

@@ -595,10 +595,12 @@ def _as_range(cond: Expr, batch: ColumnBatch):
         return None
     value = cond.right.value
     vt = torch.tensor([value], dtype=col.dtype)
-    v = int(ops.normalize_key(vt)[0])
+    v = int(ops.cpu_ref.normalize_key(vt)[0])
     keys = ops.normalize_key(col)
-    UMIN = -0x8000000000000000  # u64 0
-    UMAX = 0x7FFFFFFFFFFFFFFF   # u64 max (normalized encoding)
+    # u64-order sentinels in the int64-encoded u64 space:
+    # u64 0 encodes as int64 0; u64 max (0xFFFF..FF) encodes as int64 -1
+    UMIN = 0
+    UMAX = -1
     op = cond.op
     if op == "=":
         return keys, v, v, True, True

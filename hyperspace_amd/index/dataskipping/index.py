@@ -87,6 +87,75 @@ class DataSkippingIndex(Index):
     def refresh_full(self, ctx, df):
         raise NotImplementedError
 
+    def refresh_incremental_files(self, ctx, appended_files: List[str],
+                                  deleted_file_ids: List[int],
+                                  previous_files: List[str]):
+        """Sketch the appended files into a new index data file; rewrite
+        old sketch files dropping rows of deleted source files
+        (the per-file-row analog of the covering incremental refresh)."""
+        import uuid as _uuid
+        from ...sources.parquet_io import read_files_batch
+        from ...utils.resolver import resolve_all
+        os.makedirs(ctx.index_data_path, exist_ok=True)
+        written: List[str] = []
+        if appended_files:
+            batch, row_counts = read_files_batch(
+                sorted(appended_files), columns=self.referenced_columns())
+            if ctx.session.device.type == "cuda":
+                batch = batch.to(ctx.session.device)
+            seg = torch.zeros(len(appended_files) + 1, dtype=torch.int64)
+            seg[1:] = torch.cumsum(
+                torch.tensor(row_counts, dtype=torch.int64), 0)
+            file_ids = torch.tensor(
+                [ctx.file_id_tracker.add_file(
+                    p, os.path.getsize(p),
+                    int(os.stat(p).st_mtime * 1000))
+                 for p in sorted(appended_files)], dtype=torch.int64)
+            from ...config import IndexConstants as IC
+            data: Dict[str, torch.Tensor] = {
+                IC.DATA_FILE_NAME_ID_COLUMN: file_ids}
+            for sketch in self.sketches:
+                values = batch.tensor(sketch.expr)
+                dtype_name = self.schema.field_type(sketch.expr) or "long"
+                data.update(sketch.aggregate(values, seg, dtype_name))
+            out = os.path.join(
+                ctx.index_data_path,
+                f"part-00001-{_uuid.uuid4().hex[:8]}_00000.c000.parquet")
+            import pyarrow as pa
+            import pyarrow.parquet as pq
+            arrays = {}
+            for name, t in data.items():
+                arrays[name] = (pa.array(list(t.numpy())) if t.dim() == 2
+                                else pa.array(t.numpy()))
+            pq.write_table(pa.table(arrays), out, compression="NONE",
+                           use_dictionary=False)
+            written.append(out)
+
+        kept = list(previous_files)
+        if deleted_file_ids:
+            import pyarrow.parquet as pq
+            import pyarrow as pa
+            gone = set(deleted_file_ids)
+            kept = []
+            for p in previous_files:
+                t = pq.read_table(p)
+                ids = t.column("_data_file_id").to_pylist()
+                keep_rows = [i for i, fid in enumerate(ids)
+                             if fid not in gone]
+                if len(keep_rows) == len(ids):
+                    kept.append(p)
+                    continue
+                if keep_rows:
+                    out = os.path.join(
+                        ctx.index_data_path,
+                        f"part-00002-{_uuid.uuid4().hex[:8]}_00000"
+                        ".c000.parquet")
+                    pq.write_table(t.take(keep_rows), out,
+                                   compression="NONE",
+                                   use_dictionary=False)
+                    written.append(out)
+        return written, kept
+
     # -- query -------------------------------------------------------------
     def load_sketch_data(self, entry):
         import pyarrow.parquet as pq

@@ -138,3 +138,69 @@ def test_dataskipping_json_roundtrip(env):
     kinds = [s.kind for s in back.derivedDataset.sketches]
     assert kinds == ["MinMax", "BloomFilter"]
     assert back.derivedDataset.sketches[1].fpp == 0.05
+
+
+def test_dataskipping_refresh_incremental(env, tmp_path):
+    session, h, df, data, rng = env
+    h.create_index(df, hs.DataSkippingIndexConfig(
+        "dsr2", hs.MinMaxSketch("key")))
+    # append a file with a disjoint key range
+    t = pa.table({"key": rng.integers(50_000, 51_000, 2000),
+                  "cat": rng.integers(0, 10, 2000),
+                  "val": rng.random(2000)})
+    pq.write_table(t, str(data / "part-new.parquet"))
+    h.refresh_index("dsr2", "incremental")
+    entry = session.index_manager().get_index("dsr2")
+    assert len(entry.source_file_infos()) == 9
+    session.enable_hyperspace()
+    # a query on the NEW range must skip the 8 old files
+    q = df.filter("key >= 50000").select("key", "val")
+    ex = Executor(session)
+    out = ex.execute(q.optimized_plan())
+    assert ex.stats.scanned_files == 1
+    session.disable_hyperspace()
+    assert out.num_rows == q.collect().num_rows == 2000
+
+
+def test_dataskipping_refresh_incremental_deletes(env, tmp_path):
+    import os as _os
+    session, h, df, data, rng = env
+    h.create_index(df, hs.DataSkippingIndexConfig(
+        "dsr3", hs.MinMaxSketch("key")))
+    _os.unlink(str(data / "part-7.parquet"))
+    h.refresh_index("dsr3", "incremental")
+    entry = session.index_manager().get_index("dsr3")
+    assert len(entry.source_file_infos()) == 7
+    session.enable_hyperspace()
+    q = df.filter("key = 2500").select("key", "val")
+    accel = q.collect()
+    session.disable_hyperspace()
+    assert accel.num_rows == q.collect().num_rows
+
+
+def test_dataskipping_refresh_full(env, tmp_path):
+    session, h, df, data, rng = env
+    h.create_index(df, hs.DataSkippingIndexConfig(
+        "dsr4", hs.MinMaxSketch("key")))
+    t = pa.table({"key": rng.integers(60_000, 61_000, 1000),
+                  "cat": rng.integers(0, 10, 1000),
+                  "val": rng.random(1000)})
+    pq.write_table(t, str(data / "part-new2.parquet"))
+    h.refresh_index("dsr4", "full")
+    entry = session.index_manager().get_index("dsr4")
+    assert entry.derivedDataset.kind == "DataSkippingIndex"
+    assert len(entry.source_file_infos()) == 9
+
+
+def test_zorder_refresh_full(env, tmp_path):
+    session, h, df, data, rng = env
+    h.create_index(df, hs.ZOrderCoveringIndexConfig(
+        "zr4", ["key", "cat"], ["val"]))
+    t = pa.table({"key": rng.integers(0, 8000, 1000),
+                  "cat": rng.integers(0, 10, 1000),
+                  "val": rng.random(1000)})
+    pq.write_table(t, str(data / "part-newz.parquet"))
+    h.refresh_index("zr4", "full")
+    entry = session.index_manager().get_index("zr4")
+    assert entry.derivedDataset.kind == "ZOrderCoveringIndex"
+    assert len(entry.source_file_infos()) == 9
