@@ -213,3 +213,15 @@ def test_device_parquet_decode_pyarrow_file(tmp_path):
     dev, _ = read_files_batch_device([p], "cuda")
     for name in cols:
         assert torch.equal(host.tensor(name), dev.tensor(name).cpu()), name
+
+
+def test_select_range_one_sided_sentinels():
+    """Regression: one-sided ranges (>=, <) with u64-order sentinels —
+    u64 0 encodes as int64 0 and u64 max as int64 -1."""
+    vals = _rand_i64(500_000, lo=-10_000, hi=10_000, seed=42)
+    keys = cpu_ref.normalize_key(vals)
+    v = int(cpu_ref.normalize_key(torch.tensor([5000]))[0])
+    got = ops.select_range_u64(keys.cuda(), v, -1, True, True).cpu()
+    assert got.numel() == int((vals >= 5000).sum())
+    got = ops.select_range_u64(keys.cuda(), 0, v, True, False).cpu()
+    assert got.numel() == int((vals < 5000).sum())
