@@ -308,6 +308,9 @@ class Executor:
 
     def _bucket_of_value(self, index, col_name: str, value,
                          num_buckets: int) -> int:
+        if index.schema.field_type(col_name) == "string":
+            from ..ops.string_hash import bucket_of_string_value
+            return bucket_of_string_value(str(value), num_buckets)
         t = _value_tensor(index.schema.field_type(col_name), value)
         return int(ops.cpu_ref.murmur3_bucket([t], num_buckets)[0])
 
@@ -488,7 +491,8 @@ class Executor:
                      ) -> Tuple[ColumnBatch, torch.Tensor]:
         """Hash-repartition + per-bucket sort (K2+K3 on the fly)."""
         from ..index.covering.index import sort_by_bucket_and_keys
-        keys = [batch.tensor(c) for c in bucket_cols]
+        from ..ops.string_hash import bucket_hash_keys
+        keys = bucket_hash_keys(batch, bucket_cols)
         bucket_ids = ops.murmur3_bucket(keys, num_buckets)
         return sort_by_bucket_and_keys(batch, bucket_ids, bucket_cols,
                                        num_buckets)

@@ -103,7 +103,8 @@ class CoveringIndex(Index):
         import torch.distributed as dist
         distributed = dist.is_available() and dist.is_initialized()
 
-        keys = [batch.tensor(c) for c in self.indexed_columns]
+        from ...ops.string_hash import bucket_hash_keys
+        keys = bucket_hash_keys(batch, self.indexed_columns)
         bucket_ids = ops.murmur3_bucket(keys, n)
 
         if distributed and dist.get_world_size() > 1:

@@ -202,3 +202,61 @@ def test_explain_display_modes(env):
     session.conf.set(hs.IndexConstants.DISPLAY_MODE, "plaintext")
     out = h.explain(q)
     assert "<----" in out and "IndexScan" in out
+
+
+def test_string_indexed_column(env, tmp_path):
+    """Covering index on a string column: bucket hashing must depend on
+    the VALUE (per-dictionary murmur3 LUT), not per-batch codes."""
+    session, h, _, _ = env
+    rng = np.random.default_rng(17)
+    sdir = tmp_path / "sdata"
+    sdir.mkdir()
+    cities = np.array(["oslo", "lima", "pune", "kiel", "bern",
+                       "reno", "kobe", "baku"])
+    for i in range(2):
+        # different value distributions per file -> different dicts
+        vals = cities[rng.integers(i * 2, 4 + i * 4, 5000)]
+        t = pa.table({"city": vals, "pop": rng.integers(0, 10**6, 5000)})
+        pq.write_table(t, str(sdir / f"part-{i}.parquet"))
+    df = session.read_parquet(str(sdir))
+    h.create_index(df, hs.CoveringIndexConfig("six", ["city"], ["pop"]))
+    session.enable_hyperspace()
+    session.conf.set(hs.IndexConstants.INDEX_FILTER_RULE_USE_BUCKET_SPEC,
+                     True)
+    import pyarrow.compute as pc
+    full = pq.read_table(str(sdir))
+    for city in cities:
+        expected = int(pc.sum(
+            pc.equal(full.column("city"), str(city))).as_py() or 0)
+        q = df.filter(f"city = '{city}'").select("city", "pop")
+        plan = q.optimized_plan()
+        assert any(isinstance(l, IndexScan)
+                   for l in plan.collect_leaves()), city
+        ex = Executor(session)
+        out = ex.execute(plan)
+        assert out.num_rows == expected, city
+
+
+def test_string_join_keys_not_rewritten(env, tmp_path):
+    session, h, _, _ = env
+    rng = np.random.default_rng(18)
+    adir = tmp_path / "sa"
+    bdir = tmp_path / "sb"
+    adir.mkdir()
+    bdir.mkdir()
+    names = np.array(["a", "b", "c", "d"])
+    pq.write_table(pa.table({"name": names[rng.integers(0, 4, 1000)],
+                             "v": rng.random(1000)}),
+                   str(adir / "part-0.parquet"))
+    pq.write_table(pa.table({"name": names, "s": np.arange(4)}),
+                   str(bdir / "part-0.parquet"))
+    left = session.read_parquet(str(adir))
+    right = session.read_parquet(str(bdir))
+    h.create_index(left, hs.CoveringIndexConfig("sjl", ["name"], ["v"]))
+    h.create_index(right, hs.CoveringIndexConfig("sjr", ["name"], ["s"]))
+    session.enable_hyperspace()
+    plan = left.select("name", "v").join(right.select("name", "s"),
+                                         on="name").optimized_plan()
+    # string join keys: rule declines, plan keeps source scans
+    assert not any(isinstance(l, IndexScan)
+                   for l in plan.collect_leaves())
