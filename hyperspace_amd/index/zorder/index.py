@@ -84,7 +84,7 @@ class ZOrderCoveringIndex(Index):
         if n == 0:
             return []
 
-        z = self._zaddress(batch)
+        z = self._zaddress(ctx, batch)
         perm = ops.sort_perm(z)
         batch = batch.gather(perm)
 
@@ -108,27 +108,34 @@ class ZOrderCoveringIndex(Index):
             chunk += 1
         return written
 
-    def _zaddress(self, batch: ColumnBatch) -> torch.Tensor:
-        """Min/max scale each indexed column into its bit budget, then
-        interleave (K11 + K10)."""
+    def _zaddress(self, ctx: IndexerContext,
+                  batch: ColumnBatch) -> torch.Tensor:
+        """Scale each indexed column into its bit budget — min/max by
+        default, sampled quantile ranks when zorder.quantile.enabled (the
+        reference's percentile-bucketed ZOrderField for skewed data,
+        ZOrderField.scala) — then interleave (K11 + K10)."""
+        from ...config import IndexConstants
         n_cols = len(self.indexed_columns)
         bits = 64 // n_cols
+        quantile = bool(ctx.session.conf.get(
+            IndexConstants.ZORDER_QUANTILE_ENABLED))
         cols_u64 = []
         for c in self.indexed_columns:
             norm = ops.normalize_key(batch.tensor(c))
-            # global min/max (device reduce); distributed: all-reduce
             s = ops.cpu_ref._as_unsigned_sortable(norm)
-            lo64, hi64 = int(s.min()), int(s.max())
-            from ...parallel import dist_context as dc
-            if dc.is_distributed() and dc.get_world_size() > 1:
-                import torch.distributed as dist
-                t = torch.tensor([lo64, -hi64])
-                dist.all_reduce(t, op=dist.ReduceOp.MIN)
-                lo64, hi64 = int(t[0]), -int(t[1])
-            span = max(1, hi64 - lo64)
-            # scale into [0, 2^bits) then shift to the top bits
-            scaled = ((s - lo64).to(torch.float64) / span
-                      * float((1 << bits) - 1)).to(torch.int64)
+            if quantile:
+                scaled = _quantile_rank(s, bits)
+            else:
+                lo64, hi64 = int(s.min()), int(s.max())
+                from ...parallel import dist_context as dc
+                if dc.is_distributed() and dc.get_world_size() > 1:
+                    import torch.distributed as dist
+                    t = torch.tensor([lo64, -hi64])
+                    dist.all_reduce(t, op=dist.ReduceOp.MIN)
+                    lo64, hi64 = int(t[0]), -int(t[1])
+                span = max(1, hi64 - lo64)
+                scaled = ((s - lo64).to(torch.float64) / span
+                          * float((1 << bits) - 1)).to(torch.int64)
             cols_u64.append(scaled << (64 - bits))
         return ops.zorder_key(cols_u64, bits)
 
@@ -217,6 +224,28 @@ class ZOrderCoveringIndex(Index):
         return {"indexedColumns": self.indexed_columns,
                 "includedColumns": self.included_columns,
                 "zorder": True}
+
+
+def _quantile_rank(s: torch.Tensor, bits: int) -> torch.Tensor:
+    """Sampled-quantile scaling: rank each value against 2^bits - 1
+    sampled quantile boundaries (equi-depth z-cells for skewed columns).
+    Sampling + sort happen on the device via the radix kernel."""
+    n = s.numel()
+    n_bounds = (1 << bits) - 1
+    sample_n = min(n, max(4096, n_bounds * 8))
+    if n > sample_n:
+        step = max(1, n // sample_n)
+        sample = s[::step].contiguous()
+    else:
+        sample = s.contiguous()
+    norm_sample = sample ^ (-0x8000000000000000)
+    perm = ops.sort_perm(norm_sample)
+    sample_sorted = ops.gather_rows(sample, perm)
+    m = sample_sorted.numel()
+    idx = (torch.arange(1, n_bounds + 1, dtype=torch.float64)
+           * m / (n_bounds + 1)).to(torch.int64).clamp(max=m - 1)
+    bounds = sample_sorted[idx.to(sample_sorted.device)]
+    return torch.searchsorted(bounds.contiguous(), s)
 
 
 register_derived_dataset(ZORDER_INDEX_TYPE, ZOrderCoveringIndex)

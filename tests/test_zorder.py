@@ -117,3 +117,42 @@ def test_zorder_json_roundtrip(env):
     back = IndexLogEntry.from_json(entry.to_json())
     assert back.derivedDataset.indexed_columns == ["x", "y"]
     assert back.derivedDataset.kind == "ZOrderCoveringIndex"
+
+
+def test_zorder_quantile_mode_on_skewed_data(tmp_path, monkeypatch):
+    """Quantile scaling spreads a heavily skewed column across z-cells:
+    a hot-range box query scans fewer files than min/max scaling."""
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "idx"))
+    rng = np.random.default_rng(71)
+    data = tmp_path / "d"
+    data.mkdir()
+    n = 50_000
+    # x: 99% of mass in [0,100], tail to 10^9 (minmax scaling squashes
+    # the hot range into one cell)
+    x = np.where(rng.random(n) < 0.99,
+                 rng.integers(0, 100, n),
+                 rng.integers(0, 10**9, n))
+    t = pa.table({"x": x, "y": rng.integers(0, 100, n),
+                  "val": rng.random(n)})
+    pq.write_table(t, str(data / "part-0.parquet"))
+
+    def scanned(quantile):
+        session = hs.HyperspaceSession(device="cpu")
+        session.conf.set(
+            IndexConstants.ZORDER_TARGET_SOURCE_BYTES_PER_PARTITION,
+            64 * 1024)
+        session.conf.set(IndexConstants.ZORDER_QUANTILE_ENABLED, quantile)
+        h = hs.Hyperspace(session)
+        df = session.read_parquet(str(data))
+        name = f"zq_{int(quantile)}"
+        h.create_index(df, hs.ZOrderCoveringIndexConfig(
+            name, ["x", "y"], ["val"]))
+        session.enable_hyperspace()
+        q = df.filter("x <= 50 AND y <= 50").select("x", "y", "val")
+        ex = Executor(session)
+        out = ex.execute(q.optimized_plan())
+        session.disable_hyperspace()
+        assert out.num_rows == q.collect().num_rows
+        return ex.stats.scanned_files
+
+    assert scanned(True) <= scanned(False)
