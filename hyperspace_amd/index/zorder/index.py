@@ -227,11 +227,14 @@ class ZOrderCoveringIndex(Index):
 
 
 def _quantile_rank(s: torch.Tensor, bits: int) -> torch.Tensor:
-    """Sampled-quantile scaling: rank each value against 2^bits - 1
-    sampled quantile boundaries (equi-depth z-cells for skewed columns).
-    Sampling + sort happen on the device via the radix kernel."""
+    """Sampled-quantile scaling: rank each value against equi-depth
+    quantile boundaries (z-cells for skewed columns).  Boundary
+    resolution is capped at 2^12 cells — ranks are shifted up into the
+    full per-column bit budget.  Sampling + sort run on the device via
+    the radix kernel."""
     n = s.numel()
-    n_bounds = (1 << bits) - 1
+    cell_bits = min(bits, 12)
+    n_bounds = (1 << cell_bits) - 1
     sample_n = min(n, max(4096, n_bounds * 8))
     if n > sample_n:
         step = max(1, n // sample_n)
@@ -245,7 +248,8 @@ def _quantile_rank(s: torch.Tensor, bits: int) -> torch.Tensor:
     idx = (torch.arange(1, n_bounds + 1, dtype=torch.float64)
            * m / (n_bounds + 1)).to(torch.int64).clamp(max=m - 1)
     bounds = sample_sorted[idx.to(sample_sorted.device)]
-    return torch.searchsorted(bounds.contiguous(), s)
+    rank = torch.searchsorted(bounds.contiguous(), s)
+    return rank << (bits - cell_bits)
 
 
 register_derived_dataset(ZORDER_INDEX_TYPE, ZOrderCoveringIndex)
