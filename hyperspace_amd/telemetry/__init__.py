@@ -85,6 +85,33 @@ class NoOpEventLogger(EventLogger):
     pass
 
 
+class JsonlEventLogger(EventLogger):
+    """Structured event sink: one JSON object per line, appended to the
+    file named by env HYPERSPACE_EVENT_LOG (default
+    ~/.hyperspace/events.jsonl).  Loadable via conf
+    spark.hyperspace.eventLoggerClass =
+    "hyperspace_amd.telemetry.JsonlEventLogger"."""
+
+    def __init__(self, path: str = ""):
+        import os
+        self.path = path or os.environ.get(
+            "HYPERSPACE_EVENT_LOG",
+            os.path.join(os.path.expanduser("~"), ".hyperspace",
+                         "events.jsonl"))
+
+    def log_event(self, event: HyperspaceEvent) -> None:
+        import json
+        import os
+        os.makedirs(os.path.dirname(self.path), exist_ok=True)
+        record = {"event": event.name, "timestamp": event.timestamp,
+                  "message": event.message}
+        for attr in ("index_name", "index_names"):
+            if getattr(event, attr, None):
+                record[attr] = getattr(event, attr)
+        with open(self.path, "a") as f:
+            f.write(json.dumps(record) + "\n")
+
+
 class RecordingEventLogger(EventLogger):
     """Test sink (reference: MockEventLogger, TestUtils.scala:93-110)."""
 

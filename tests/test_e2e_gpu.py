@@ -112,3 +112,39 @@ def test_gpu_sorted_within_buckets(env, tmp_path):
         keys = torch.from_numpy(t.column("key").to_numpy())
         assert torch.all(keys[1:] >= keys[:-1]), f
         assert (cpu_ref.murmur3_bucket([keys], 8) == b).all(), f
+
+
+def test_gpu_hybrid_scan(env, tmp_path):
+    """Hybrid scan on device: appended delta merged via Union, deleted
+    file rows excluded via the lineage kernel."""
+    import pyarrow as pa
+    data, _ = env
+    gpu = hs.HyperspaceSession(device="cuda")
+    gpu.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 16)
+    gpu.conf.set(hs.IndexConstants.INDEX_LINEAGE_ENABLED, True)
+    h = hs.Hyperspace(gpu)
+    df = gpu.read_parquet(str(data))
+    h.create_index(df, hs.CoveringIndexConfig("hgx", ["key"], ["val"]))
+    # append a small delta
+    rng = np.random.default_rng(77)
+    t = pa.table({"key": rng.integers(0, 50_000, 20_000),
+                  "val": rng.random(20_000)})
+    pq.write_table(t, str(data / "part-delta.parquet"))
+    gpu.conf.set(hs.IndexConstants.INDEX_HYBRID_SCAN_ENABLED, True)
+    gpu.enable_hyperspace()
+    q = df.filter("key = 4242").select("key", "val")
+    plan = q.optimized_plan()
+    from hyperspace_amd.plan.nodes import UnionNode, BucketUnionNode
+    found = []
+
+    def walk(n):
+        if isinstance(n, (UnionNode, BucketUnionNode)):
+            found.append(n)
+        for c in n.children:
+            walk(c)
+    walk(plan)
+    assert found, plan.pretty()
+    out = q.collect()
+    gpu.disable_hyperspace()
+    base = q.collect()
+    assert out.num_rows == base.num_rows

@@ -195,3 +195,20 @@ def test_index_stats(env):
     assert stats["numSourceFiles"] == 3
     assert stats["sizeOfIndexInBytes"] > 0
     assert [d["name"] for d in h.indexes()] == ["ix"]
+
+
+def test_jsonl_event_logger(env, tmp_path, monkeypatch):
+    import json
+    session, h, df, _, _ = env
+    log_path = str(tmp_path / "events.jsonl")
+    monkeypatch.setenv("HYPERSPACE_EVENT_LOG", log_path)
+    session.conf.set(hs.IndexConstants.EVENT_LOGGER_CLASS,
+                     "hyperspace_amd.telemetry.JsonlEventLogger")
+    session.event_logger = None  # force re-resolution from conf
+    h.create_index(df, hs.CoveringIndexConfig("jx", ["key"], ["val"]))
+    h.delete_index("jx")
+    with open(log_path) as f:
+        events = [json.loads(l) for l in f]
+    assert [e["event"] for e in events] == ["CreateActionEvent",
+                                           "DeleteActionEvent"]
+    assert events[0]["index_name"] == "jx"
