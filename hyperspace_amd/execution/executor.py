@@ -224,6 +224,8 @@ class Executor:
             batch, seg = cached
             self.stats.record("IndexScan(cached)")
         else:
+            sort_col = (index.indexed_columns[0]
+                        if getattr(index, "indexed_columns", None) else None)
             batches: List[ColumnBatch] = []
             seg_counts = torch.zeros(num_buckets + 1, dtype=torch.int64)
             for b in wanted_buckets:
@@ -234,6 +236,16 @@ class Executor:
                         paths, self.device, columns=read_cols)
                 else:
                     sub, _ = read_files_batch(paths, columns=read_cols)
+                if len(paths) > 1 and plan.use_bucket_spec and \
+                        sort_col is not None and sub.num_rows:
+                    # each file is sorted, their concatenation is not:
+                    # re-sort the merged bucket so downstream merge joins
+                    # keep their sorted-segment contract (Spark re-sorts
+                    # multi-file buckets inside SortMergeJoinExec the
+                    # same way)
+                    perm = ops.sort_perm(
+                        ops.normalize_key(sub.tensor(sort_col)))
+                    sub = sub.gather(perm)
                 seg_counts[b + 1] = sub.num_rows
                 batches.append(sub)
             if batches:

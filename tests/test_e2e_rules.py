@@ -260,3 +260,40 @@ def test_string_join_keys_not_rewritten(env, tmp_path):
     # string join keys: rule declines, plan keeps source scans
     assert not any(isinstance(l, IndexScan)
                    for l in plan.collect_leaves())
+
+
+def test_join_correct_after_incremental_refresh(env, tmp_path):
+    """Regression: incremental refresh leaves multiple sorted files per
+    bucket; the bucket must be re-sorted at scan time or the merge join
+    returns wrong results."""
+    session, h, left, right = env
+    h.create_index(left, hs.CoveringIndexConfig(
+        "lidx", ["orderkey"], ["qty"]))
+    h.create_index(right, hs.CoveringIndexConfig(
+        "ridx", ["orderkey"], ["status"]))
+    # append to the RIGHT source (the searched side of the merge join —
+    # the side whose sortedness the join depends on) and refresh
+    # incrementally -> second file per bucket in ridx
+    rng = np.random.default_rng(23)
+    t = pa.table({"orderkey": rng.integers(0, 5000, 3000),
+                  "status": rng.integers(0, 3, 3000)})
+    pq.write_table(t, str(tmp_path / "right" / "part-extra.parquet"))
+    h.refresh_index("ridx", "incremental")
+    entry = session.index_manager().get_index("ridx")
+    from collections import Counter
+    from hyperspace_amd.sources.parquet_io import bucket_id_of_file
+    per_bucket = Counter(bucket_id_of_file(f)
+                         for f in entry.content.os_files())
+    assert max(per_bucket.values()) > 1  # multi-file buckets exist
+
+    q = left.select("orderkey", "qty").join(
+        right.select("orderkey", "status"), on="orderkey")
+    baseline = q.collect()
+    session.enable_hyperspace()
+    from hyperspace_amd.execution.executor import Executor
+    ex = Executor(session)
+    accel = ex.execute(q.optimized_plan())
+    assert ex.stats.merge_joins == 1
+    assert accel.num_rows == baseline.num_rows
+    cols = ["orderkey", "qty", "status"]
+    assert _sorted_rows(accel, cols) == _sorted_rows(baseline, cols)
