@@ -71,20 +71,27 @@ class CoveringIndexConfig(IndexConfigTrait):
 
         # pre-assign file ids deterministically across ranks (sorted file
         # order) so distributed shards agree on the lineage id space
-        for f in sorted(scan.relation.all_files(), key=lambda f: f.name):
+        all_files = sorted(scan.relation.all_files(), key=lambda f: f.name)
+        for f in all_files:
             ctx.file_id_tracker.add_file(f.name, f.size, f.modifiedTime)
 
         from ...parallel import dist_context as dc
-        shard = dc.is_distributed() and dc.get_world_size() > 1
+        files = all_files
+        if dc.is_distributed() and dc.get_world_size() > 1:
+            files = files[dc.get_rank()::dc.get_world_size()]
 
-        ex = Executor(ctx.session)
-        batch = ex._exec_scan(
-            scan, lineage_tracker=ctx.file_id_tracker if lineage else None,
-            shard=shard)
+        # the build consumes a ScanStream: byte-bounded file groups with
+        # prefetch so disk/PCIe overlap the bucketize+sort+write pipeline
+        # (group size 0 disables streaming -> one materialized batch)
+        from ...execution.scan_stream import ScanStream
+        group_bytes = int(ctx.session.conf.get(
+            "spark.hyperspace.index.build.groupBytes", 1 << 30))
         cols = indexed + included
-        if lineage:
-            cols = cols + [IndexConstants.DATA_FILE_NAME_ID_COLUMN]
-        batch = batch.select(cols)
+        stream = ScanStream(
+            files, cols, ctx.session.device,
+            lineage_tracker=ctx.file_id_tracker if lineage else None,
+            group_bytes=group_bytes if group_bytes > 0 else (1 << 62))
+        batch = stream
 
         index_schema = Schema(
             [f for f in source_schema.fields

@@ -88,34 +88,77 @@ class CoveringIndex(Index):
             d.get("properties", {}))
 
     # -- build data plane ---------------------------------------------------
-    def write(self, ctx: IndexerContext, index_data: ColumnBatch
-              ) -> List[str]:
+    def write(self, ctx: IndexerContext, index_data) -> List[str]:
         """Bucketize + per-bucket sort + bucketed parquet write (K2+K3).
 
+        ``index_data`` is a ColumnBatch or a ScanStream.  Streams are
+        processed group by group — read/decode of group g+1 prefetches
+        while g bucketizes/sorts and g-1's files flush on write threads —
+        bounding memory to O(group) and overlapping disk, PCIe and
+        kernels (out-of-core builds of sources larger than HBM).
+
         With torch.distributed initialized this is the multi-GPU build:
-        each rank holds a source shard; bucket rows are exchanged via RCCL
-        all-to-all over xGMI so rank r owns buckets {b : b % world == r}.
+        each rank holds a source shard; bucket rows are exchanged via
+        RCCL all-to-all over xGMI so rank r owns buckets
+        {b : b % world == r}.
         """
+        from ...execution.scan_stream import ScanStream
         os.makedirs(ctx.index_data_path, exist_ok=True)
-        batch = index_data
-        n = self.num_buckets
 
         import torch.distributed as dist
-        distributed = dist.is_available() and dist.is_initialized()
+        distributed = dist.is_available() and dist.is_initialized() and \
+            dist.get_world_size() > 1
+        task_id = dist.get_rank() if distributed else 0
 
+        if isinstance(index_data, ScanStream):
+            batches = index_data.batches()
+            n_groups = len(index_data.file_groups())
+        else:
+            batches = iter([index_data])
+            n_groups = 1
+
+        # collective iteration count: every rank must join every
+        # all-to-all, so ranks with fewer groups contribute empty batches
+        if distributed:
+            t = torch.tensor([n_groups])
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            n_groups = int(t[0])
+
+        written: List[str] = []
+        for g in range(n_groups):
+            try:
+                batch = next(batches)
+            except StopIteration:
+                batch = self._empty_batch(ctx)
+            written.extend(self._write_group(ctx, batch, task_id,
+                                             distributed))
+        return written
+
+    def _write_group(self, ctx, batch: ColumnBatch, task_id: int,
+                     distributed: bool) -> List[str]:
+        n = self.num_buckets
         from ...ops.string_hash import bucket_hash_keys
         keys = bucket_hash_keys(batch, self.indexed_columns)
         bucket_ids = ops.murmur3_bucket(keys, n)
-
-        if distributed and dist.get_world_size() > 1:
+        if distributed:
             from ...parallel.exchange import exchange_by_bucket
             batch, bucket_ids = exchange_by_bucket(batch, bucket_ids, n)
-
+        if batch.num_rows == 0:
+            return []
         batch, seg = sort_by_bucket_and_keys(
             batch, bucket_ids, self.indexed_columns, n)
-
-        task_id = dist.get_rank() if distributed else 0
         return write_bucketed(batch, seg, ctx.index_data_path, n, task_id)
+
+    def _empty_batch(self, ctx) -> ColumnBatch:
+        import torch as _torch
+        dts = {"long": _torch.int64, "integer": _torch.int32,
+               "double": _torch.float64, "float": _torch.float32}
+        cols = {}
+        for f in self.schema.fields:
+            cols[f.name] = _torch.empty(
+                0, dtype=dts.get(f.type, _torch.int64),
+                device=ctx.session.device)
+        return ColumnBatch(cols)
 
     def optimize(self, ctx: IndexerContext,
                  files_to_optimize: List[str]) -> List[str]:

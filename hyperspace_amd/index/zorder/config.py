@@ -19,8 +19,12 @@ class ZOrderCoveringIndexConfig(CoveringIndexConfig):
                      properties: Dict[str, str]
                      ) -> Tuple[ZOrderCoveringIndex, object]:
         # reuse the covering projection/lineage pipeline, then wrap the
-        # result in a z-order index
+        # result in a z-order index; the global z-sort needs the whole
+        # batch, so a streaming scan is materialized here
         cov_index, batch = super().create_index(ctx, df, properties)
+        from ...execution.scan_stream import ScanStream
+        if isinstance(batch, ScanStream):
+            batch = batch.materialize()
         index = ZOrderCoveringIndex(
             cov_index.indexed_columns, cov_index.included_columns,
             cov_index.schema, dict(properties))

@@ -212,3 +212,45 @@ def test_jsonl_event_logger(env, tmp_path, monkeypatch):
     assert [e["event"] for e in events] == ["CreateActionEvent",
                                            "DeleteActionEvent"]
     assert events[0]["index_name"] == "jx"
+
+
+def test_streaming_build_multi_group(env, tmp_path):
+    """Small group size forces a multi-group streaming build: several
+    sorted files per bucket, queries still exact."""
+    session, h, df, data_dir, rng = env
+    # each source file ~78KB; 100KB groups -> 3 groups
+    session.conf.set("spark.hyperspace.index.build.groupBytes", 100_000)
+    h.create_index(df, hs.CoveringIndexConfig("sg", ["key"], ["val"]))
+    entry = session.index_manager().get_index("sg")
+    from collections import Counter
+    per_bucket = Counter(bucket_id_of_file(f)
+                         for f in entry.content.os_files())
+    assert max(per_bucket.values()) > 1  # one file per (group, bucket)
+    session.enable_hyperspace()
+    session.conf.set(hs.IndexConstants.INDEX_FILTER_RULE_USE_BUCKET_SPEC,
+                     True)
+    q = df.filter("key = 77").select("key", "val")
+    accel = q.collect()
+    session.disable_hyperspace()
+    base = q.collect()
+    a = sorted(zip(accel.to_numpy()["key"].tolist(),
+                   accel.to_numpy()["val"].tolist()))
+    b = sorted(zip(base.to_numpy()["key"].tolist(),
+                   base.to_numpy()["val"].tolist()))
+    assert a == b and len(a) > 0
+
+
+def test_streaming_vs_materialized_same_rows(env, tmp_path):
+    session, h, df, _, _ = env
+    session.conf.set("spark.hyperspace.index.build.groupBytes", 100_000)
+    h.create_index(df, hs.CoveringIndexConfig("s1", ["key"], ["val"]))
+    session.conf.set("spark.hyperspace.index.build.groupBytes", 0)
+    h.create_index(df, hs.CoveringIndexConfig("s2", ["key"], ["val"]))
+    import pyarrow.parquet as pq2
+    e1 = session.index_manager().get_index("s1")
+    e2 = session.index_manager().get_index("s2")
+    t1 = pq2.read_table(e1.content.os_files()).sort_by(
+        [("key", "ascending"), ("val", "ascending")])
+    t2 = pq2.read_table(e2.content.os_files()).sort_by(
+        [("key", "ascending"), ("val", "ascending")])
+    assert t1.equals(t2)
