@@ -85,7 +85,7 @@ class CoveringIndexConfig(IndexConfigTrait):
         # (group size 0 disables streaming -> one materialized batch)
         from ...execution.scan_stream import ScanStream
         group_bytes = int(ctx.session.conf.get(
-            "spark.hyperspace.index.build.groupBytes", 1 << 30))
+            "spark.hyperspace.index.build.groupBytes", 32 << 30))
         cols = indexed + included
         stream = ScanStream(
             files, cols, ctx.session.device,
