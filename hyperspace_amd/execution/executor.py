@@ -112,11 +112,9 @@ class Executor:
         paths = [f.name for f in files]
         self.stats.scanned_files += len(paths)
         self.stats.scanned_bytes += sum(f.size for f in files)
-        if paths and self.device.type == "cuda":
-            batch, row_counts = read_files_batch_device(
-                paths, self.device)
-        elif paths:
-            batch, row_counts = read_files_batch(paths)
+        if paths:
+            batch, row_counts = plan.relation.read_files(
+                paths, None, self.device)
         else:
             batch, row_counts = _empty_batch(plan.relation.schema), []
         if lineage_tracker is not None:

@@ -21,13 +21,17 @@ from ..config import IndexConstants
 
 class ScanStream:
     def __init__(self, files, columns: List[str], device,
-                 lineage_tracker=None, group_bytes: int = 1 << 30):
-        """``files``: FileInfo list (already sharded for this rank)."""
+                 lineage_tracker=None, group_bytes: int = 1 << 30,
+                 reader=None):
+        """``files``: FileInfo list (already sharded for this rank);
+        ``reader(paths, columns, device) -> (batch, row_counts)``
+        defaults to the parquet path."""
         self.files = list(files)
         self.columns = list(columns)
         self.device = device
         self.tracker = lineage_tracker
         self.group_bytes = max(1, group_bytes)
+        self.reader = reader
 
     @property
     def total_bytes(self) -> int:
@@ -49,15 +53,19 @@ class ScanStream:
         return groups
 
     def _load_group(self, group) -> ColumnBatch:
-        from ..sources.parquet_io import (read_files_batch,
-                                          read_files_batch_device)
         paths = [f.name for f in group]
-        if self.device.type == "cuda":
-            batch, row_counts = read_files_batch_device(
-                paths, self.device, columns=self.columns or None)
+        if self.reader is not None:
+            batch, row_counts = self.reader(paths, self.columns or None,
+                                            self.device)
         else:
-            batch, row_counts = read_files_batch(
-                paths, columns=self.columns or None)
+            from ..sources.parquet_io import (read_files_batch,
+                                              read_files_batch_device)
+            if self.device.type == "cuda":
+                batch, row_counts = read_files_batch_device(
+                    paths, self.device, columns=self.columns or None)
+            else:
+                batch, row_counts = read_files_batch(
+                    paths, columns=self.columns or None)
         if self.tracker is not None:
             ids = []
             for f, n in zip(group, row_counts):

@@ -90,7 +90,8 @@ class CoveringIndexConfig(IndexConfigTrait):
         stream = ScanStream(
             files, cols, ctx.session.device,
             lineage_tracker=ctx.file_id_tracker if lineage else None,
-            group_bytes=group_bytes if group_bytes > 0 else (1 << 62))
+            group_bytes=group_bytes if group_bytes > 0 else (1 << 62),
+            reader=scan.relation.read_files)
         batch = stream
 
         index_schema = Schema(

@@ -73,6 +73,19 @@ class HyperspaceSession:
         return DataFrame(self, Scan(DeltaTableRelation(path,
                                                        version_as_of)))
 
+    def read_csv(self, *paths: str):
+        from .dataframe import DataFrame
+        from .plan.nodes import Scan
+        from .sources.text_source import TextFormatRelation
+        return DataFrame(self, Scan(TextFormatRelation("csv", list(paths))))
+
+    def read_json(self, *paths: str):
+        from .dataframe import DataFrame
+        from .plan.nodes import Scan
+        from .sources.text_source import TextFormatRelation
+        return DataFrame(self, Scan(TextFormatRelation("json",
+                                                       list(paths))))
+
     def read_iceberg(self, path: str, snapshot_id=None):
         """Read an iceberg-style snapshot table (time travel via
         ``snapshot_id``)."""

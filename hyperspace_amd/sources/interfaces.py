@@ -39,6 +39,15 @@ class FileBasedRelation(ABC):
     def signature(self) -> str:
         """Content fingerprint of the relation."""
 
+    def read_files(self, paths: List[str], columns, device):
+        """Read data files into a (ColumnBatch on ``device``, per-file row
+        counts).  Default: the native/pyarrow Parquet path with device
+        decode; text formats override with their host readers."""
+        from .parquet_io import read_files_batch, read_files_batch_device
+        if getattr(device, "type", "cpu") == "cuda":
+            return read_files_batch_device(paths, device, columns)
+        return read_files_batch(paths, columns)
+
     def describe(self) -> str:
         return f"{self.file_format}:{','.join(self.root_paths)}"
 

@@ -23,8 +23,10 @@ class FileBasedSourceProviderManager:
         # table-format providers (delta-style log, iceberg-style snapshots)
         from .delta_source import DeltaTableSourceProvider
         from .iceberg_source import IcebergTableSourceProvider
+        from .text_source import TextFormatSourceProvider
         self.providers.append(DeltaTableSourceProvider())
         self.providers.append(IcebergTableSourceProvider())
+        self.providers.append(TextFormatSourceProvider())
 
     def register(self, provider: FileBasedSourceProvider) -> None:
         self.providers.append(provider)
