@@ -24,6 +24,10 @@ void normalize_key(const void* vals, int dtype, uint64_t* out, int64_t n,
 void radix_sort_pairs(uint64_t* keys, int64_t* payload, uint64_t* tmp_keys,
                       int64_t* tmp_payload, uint32_t* hist, uint64_t* d_mask,
                       int64_t n, hipStream_t stream);
+void radix_sort_pairs32(uint64_t* keys, int32_t* payload,
+                        uint64_t* tmp_keys, int32_t* tmp_payload,
+                        uint32_t* hist, uint64_t* d_mask, int64_t n,
+                        hipStream_t stream);
 // histogram buffer size requirement (u32 elements)
 int64_t radix_sort_hist_size(int64_t n);
 

@@ -220,7 +220,7 @@ void normalize_key(const void* vals, int dtype, uint64_t* out, int64_t n,
 // Stable LSD radix sort (4-bit digits, wave64 ballot multi-split)
 // ---------------------------------------------------------------------------
 
-constexpr int RS_RADIX = 16;
+constexpr int RS_RADIX = 256;      // 8-bit digits
 constexpr int RS_MAX_BLOCKS = 1024;
 
 static inline int rs_num_blocks(int64_t n) {
@@ -257,27 +257,25 @@ __global__ void k_rs_hist(const uint64_t* __restrict__ keys, int64_t n,
                           int shift, uint32_t* __restrict__ hist, int nb,
                           int64_t tiles_per_block) {
   // CONTIGUOUS tile->block ranges: block-major order must equal element
-  // order or cross-pass LSD stability breaks (verified on hardware:
-  // round-robin tiling produced a correct per-pass grouping but a wrong
-  // multi-pass sort).
+  // order or cross-pass LSD stability breaks.
   __shared__ uint32_t lh[RS_RADIX];
-  if (threadIdx.x < RS_RADIX) lh[threadIdx.x] = 0;
+  for (int d = threadIdx.x; d < RS_RADIX; d += blockDim.x) lh[d] = 0;
   __syncthreads();
   int64_t t0 = (int64_t)blockIdx.x * tiles_per_block;
   int64_t t1 = t0 + tiles_per_block;
   for (int64_t tile = t0; tile < t1 && tile * THREADS < n; tile++) {
     int64_t i = tile * (int64_t)THREADS + threadIdx.x;
     if (i < n) {
-      int d = (int)((keys[i] >> shift) & 15);
+      int d = (int)((keys[i] >> shift) & 255);
       atomicAdd(&lh[d], 1u);
     }
   }
   __syncthreads();
-  if (threadIdx.x < RS_RADIX)
-    hist[(int64_t)threadIdx.x * nb + blockIdx.x] = lh[threadIdx.x];
+  for (int d = threadIdx.x; d < RS_RADIX; d += blockDim.x)
+    hist[(int64_t)d * nb + blockIdx.x] = lh[d];
 }
 
-// single-block exclusive scan over u32 (hist is 16*nb <= 16384 elements)
+// single-block exclusive scan over u32 (hist is 256*nb <= 262144 elements)
 __global__ void k_scan_u32(uint32_t* __restrict__ data, int64_t n) {
   constexpr int T = 1024;
   __shared__ uint32_t sums[T];
@@ -303,18 +301,19 @@ __global__ void k_scan_u32(uint32_t* __restrict__ data, int64_t n) {
   }
 }
 
+template <typename P>
 __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
-                             const int64_t* __restrict__ payload,
+                             const P* __restrict__ payload,
                              uint64_t* __restrict__ okeys,
-                             int64_t* __restrict__ opayload, int64_t n,
+                             P* __restrict__ opayload, int64_t n,
                              int shift, const uint32_t* __restrict__ hist,
                              int nb, int64_t tiles_per_block) {
   __shared__ uint32_t cur[RS_RADIX];
   __shared__ uint32_t wave_cnt[WAVES][RS_RADIX];
   __shared__ uint32_t wave_pref[WAVES][RS_RADIX];
   __shared__ uint32_t tile_total[RS_RADIX];
-  if (threadIdx.x < RS_RADIX)
-    cur[threadIdx.x] = hist[(int64_t)threadIdx.x * nb + blockIdx.x];
+  for (int d = threadIdx.x; d < RS_RADIX; d += blockDim.x)
+    cur[d] = hist[(int64_t)d * nb + blockIdx.x];
   int lane = threadIdx.x & 63;
   int wave = threadIdx.x >> 6;
   int64_t t0 = (int64_t)blockIdx.x * tiles_per_block;
@@ -323,28 +322,28 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
     int64_t i = tile * (int64_t)THREADS + threadIdx.x;
     bool valid = i < n;
     uint64_t key = valid ? keys[i] : 0;
-    int64_t pl = valid ? payload[i] : 0;
-    int d = valid ? (int)((key >> shift) & 15) : RS_RADIX;  // 16 = sentinel
-    if (threadIdx.x < WAVES * RS_RADIX)
-      ((uint32_t*)wave_cnt)[threadIdx.x] = 0;
+    P pl = valid ? payload[i] : P(0);
+    int d = valid ? (int)((key >> shift) & 255) : RS_RADIX;  // sentinel
+    for (int x = threadIdx.x; x < WAVES * RS_RADIX; x += blockDim.x)
+      ((uint32_t*)wave_cnt)[x] = 0;
     __syncthreads();
-    // 5-bit ballot multi-split: lanes with equal digit (incl. sentinel)
+    // 9-bit ballot multi-split: lanes with equal digit (incl. sentinel)
     unsigned long long eq = ~0ull;
 #pragma unroll
-    for (int b = 0; b < 5; b++) {
+    for (int b = 0; b < 9; b++) {
       unsigned long long m = __ballot((d >> b) & 1);
       eq &= ((d >> b) & 1) ? m : ~m;
     }
     int rank = __popcll(eq & ((1ull << lane) - 1ull));
     if (valid && rank == 0) wave_cnt[wave][d] = (uint32_t)__popcll(eq);
     __syncthreads();
-    if (threadIdx.x < RS_RADIX) {
+    for (int dd = threadIdx.x; dd < RS_RADIX; dd += blockDim.x) {
       uint32_t p = 0;
       for (int w = 0; w < WAVES; w++) {
-        wave_pref[w][threadIdx.x] = p;
-        p += wave_cnt[w][threadIdx.x];
+        wave_pref[w][dd] = p;
+        p += wave_cnt[w][dd];
       }
-      tile_total[threadIdx.x] = p;
+      tile_total[dd] = p;
     }
     __syncthreads();
     if (valid) {
@@ -353,21 +352,24 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
       opayload[pos] = pl;
     }
     __syncthreads();
-    if (threadIdx.x < RS_RADIX) cur[threadIdx.x] += tile_total[threadIdx.x];
+    for (int dd = threadIdx.x; dd < RS_RADIX; dd += blockDim.x)
+      cur[dd] += tile_total[dd];
     __syncthreads();
   }
 }
 
-void radix_sort_pairs(uint64_t* keys, int64_t* payload, uint64_t* tmp_keys,
-                      int64_t* tmp_payload, uint32_t* hist, uint64_t* d_mask,
-                      int64_t n, hipStream_t stream) {
+template <typename P>
+static void radix_sort_impl(uint64_t* keys, P* payload, uint64_t* tmp_keys,
+                            P* tmp_payload, uint32_t* hist,
+                            uint64_t* d_mask, int64_t n,
+                            hipStream_t stream) {
   if (n <= 1) return;
   assert(n < (int64_t)UINT32_MAX);
   int nb = rs_num_blocks(n);
   int64_t tpb = cdiv(cdiv(n, THREADS), nb);
   int g = grid_for(n);
 
-  // which 4-bit digits actually vary? (constant-nibble pass skipping)
+  // which bytes actually vary? (constant-byte pass skipping)
   HIP_CHECK(hipMemsetAsync(d_mask, 0, sizeof(uint64_t), stream));
   hipLaunchKernelGGL(k_xor_or_reduce, dim3(g), dim3(THREADS), 0, stream,
                      keys, n, d_mask);
@@ -378,10 +380,10 @@ void radix_sort_pairs(uint64_t* keys, int64_t* payload, uint64_t* tmp_keys,
 
   uint64_t* ka = keys;
   uint64_t* kb = tmp_keys;
-  int64_t* pa = payload;
-  int64_t* pb = tmp_payload;
-  for (int shift = 0; shift < 64; shift += 4) {
-    if (((mask >> shift) & 15ull) == 0) continue;  // constant digit
+  P* pa = payload;
+  P* pb = tmp_payload;
+  for (int shift = 0; shift < 64; shift += 8) {
+    if (((mask >> shift) & 255ull) == 0) continue;  // constant byte
     HIP_CHECK(hipMemsetAsync(hist, 0,
                              (size_t)RS_RADIX * nb * sizeof(uint32_t),
                              stream));
@@ -389,17 +391,32 @@ void radix_sort_pairs(uint64_t* keys, int64_t* payload, uint64_t* tmp_keys,
                        shift, hist, nb, tpb);
     hipLaunchKernelGGL(k_scan_u32, dim3(1), dim3(1024), 0, stream, hist,
                        (int64_t)RS_RADIX * nb);
-    hipLaunchKernelGGL(k_rs_scatter, dim3(nb), dim3(THREADS), 0, stream, ka,
-                       pa, kb, pb, n, shift, hist, nb, tpb);
+    hipLaunchKernelGGL(k_rs_scatter<P>, dim3(nb), dim3(THREADS), 0, stream,
+                       ka, pa, kb, pb, n, shift, hist, nb, tpb);
     uint64_t* tk = ka; ka = kb; kb = tk;
-    int64_t* tp = pa; pa = pb; pb = tp;
+    P* tp = pa; pa = pb; pb = tp;
   }
   if (ka != keys) {
     HIP_CHECK(hipMemcpyAsync(keys, ka, n * sizeof(uint64_t),
                              hipMemcpyDeviceToDevice, stream));
-    HIP_CHECK(hipMemcpyAsync(payload, pa, n * sizeof(int64_t),
+    HIP_CHECK(hipMemcpyAsync(payload, pa, n * sizeof(P),
                              hipMemcpyDeviceToDevice, stream));
   }
+}
+
+void radix_sort_pairs(uint64_t* keys, int64_t* payload, uint64_t* tmp_keys,
+                      int64_t* tmp_payload, uint32_t* hist, uint64_t* d_mask,
+                      int64_t n, hipStream_t stream) {
+  radix_sort_impl<int64_t>(keys, payload, tmp_keys, tmp_payload, hist,
+                           d_mask, n, stream);
+}
+
+void radix_sort_pairs32(uint64_t* keys, int32_t* payload,
+                        uint64_t* tmp_keys, int32_t* tmp_payload,
+                        uint32_t* hist, uint64_t* d_mask, int64_t n,
+                        hipStream_t stream) {
+  radix_sort_impl<int32_t>(keys, payload, tmp_keys, tmp_payload, hist,
+                           d_mask, n, stream);
 }
 
 // ---------------------------------------------------------------------------
