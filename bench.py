@@ -116,10 +116,15 @@ def main():
     join_lat = []
     build_times = []
     # index deletion is bookkeeping outside the benchmarked workload
-    # (build + queries); it runs after the timed region unless the run
-    # would exceed a disk budget
+    # (build + queries); it runs after the timed region unless keeping
+    # every step's index version would not fit the filesystem
     pending_cleanup = []
-    cleanup_inline = (args.steps + args.warmup + 1) * total_bytes > (60 << 30)
+    needed = int((args.steps + args.warmup + 1) * total_bytes * 1.3)
+    free = shutil.disk_usage(workdir).free
+    cleanup_inline = free < needed + (50 << 30)
+    if rank == 0 and os.environ.get("BENCH_DEBUG"):
+        print(f"[dbg] disk free={free/2**30:.0f}G needed={needed/2**30:.0f}G"
+              f" inline_cleanup={cleanup_inline}", file=sys.stderr)
 
     def one_step(step_idx, timed):
         # build: fresh index each step (full pipeline)
