@@ -62,8 +62,17 @@ def main():
 
     device = args.device or ("cuda" if on_gpu else "cpu")
 
-    workdir = args.workdir or os.environ.get(
-        "BENCH_WORKDIR", "/tmp/hyperspace_bench")
+    workdir = args.workdir or os.environ.get("BENCH_WORKDIR")
+    if workdir is None:
+        # prefer RAM-backed storage: GPU-box /tmp overlays are small
+        # (~79G) while hosts carry TBs of RAM — /dev/shm holds the
+        # synthetic source + index versions without disk limits
+        needed_est = int(args.gb_per_gpu * (1 << 30)) * world * 4
+        shm_free = shutil.disk_usage("/dev/shm").free \
+            if os.path.isdir("/dev/shm") else 0
+        workdir = ("/dev/shm/hyperspace_bench"
+                   if shm_free > needed_est + (8 << 30)
+                   else "/tmp/hyperspace_bench")
     data_dir = os.path.join(workdir, "fact")
     dim_dir = os.path.join(workdir, "dim")
     index_root = os.path.join(workdir, "indexes")

@@ -66,6 +66,22 @@ def _tracker_from_entry(entry: IndexLogEntry) -> FileIdTracker:
     return tracker
 
 
+def _agree_data_version(data_manager) -> int:
+    """Next index data version, agreed across ranks BEFORE any rank
+    creates the directory (rank 0 lists and broadcasts — a faster rank's
+    makedirs must not race a slower rank's listing)."""
+    latest = data_manager.get_latest_version_id()
+    version = 0 if latest is None else latest + 1
+    from ..parallel import dist_context as dc
+    if dc.is_distributed() and dc.get_world_size() > 1:
+        import torch
+        import torch.distributed as dist
+        t = torch.tensor([version])
+        dist.broadcast(t, src=0)
+        version = int(t[0])
+    return version
+
+
 def _rebuild_config(name: str, index):
     """Reconstruct the matching config for a full rebuild of any index
     kind (refresh full must not assume a covering index)."""
@@ -126,8 +142,7 @@ class CreateAction(Action):
     def op(self):
         from ..index.base import IndexerContext
 
-        latest_v = self.data_manager.get_latest_version_id()
-        version = 0 if latest_v is None else latest_v + 1
+        version = _agree_data_version(self.data_manager)
         data_path = self.data_manager.get_path(version)
         tracker = FileIdTracker()
         ctx = IndexerContext(self.session, tracker, data_path)
@@ -258,8 +273,7 @@ class RefreshFullAction(RefreshActionBase):
         if not appended and not deleted:
             raise NoChangesException("Refresh full: no source changes")
         relation = self.source_relation()
-        latest_v = self.data_manager.get_latest_version_id()
-        version = 0 if latest_v is None else latest_v + 1
+        version = _agree_data_version(self.data_manager)
         data_path = self.data_manager.get_path(version)
         tracker = FileIdTracker()
         from ..index.base import IndexerContext
@@ -303,8 +317,7 @@ class RefreshIncrementalAction(RefreshActionBase):
                        if (f.name, f.size, f.modifiedTime) in
                        {(d.name, d.size, d.modifiedTime) for d in deleted}]
 
-        latest_v = self.data_manager.get_latest_version_id()
-        version = 0 if latest_v is None else latest_v + 1
+        version = _agree_data_version(self.data_manager)
         data_path = self.data_manager.get_path(version)
         from ..index.base import IndexerContext
         ctx = IndexerContext(self.session, tracker, data_path)
@@ -441,8 +454,7 @@ class OptimizeAction(Action):
         to_optimize, self._ignored = self._select_files()
         if not to_optimize:
             raise NoChangesException("Optimize: no files to compact")
-        latest_v = self.data_manager.get_latest_version_id()
-        version = 0 if latest_v is None else latest_v + 1
+        version = _agree_data_version(self.data_manager)
         data_path = self.data_manager.get_path(version)
         from ..index.base import IndexerContext
         ctx = IndexerContext(self.session, _tracker_from_entry(self.previous),

@@ -518,7 +518,11 @@ class IndexLogEntry:
         }
 
     def to_json_str(self) -> str:
-        return json.dumps(self.to_json(), indent=2, sort_keys=False)
+        # compact separators: entries with hundreds of file nodes are
+        # serialized on every action write, and pretty-printing tripled
+        # maintenance latency
+        return json.dumps(self.to_json(), separators=(",", ":"),
+                          sort_keys=False)
 
     @staticmethod
     def from_json(d: Dict[str, Any]) -> "IndexLogEntry":
