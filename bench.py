@@ -175,9 +175,12 @@ def main():
         # next step's create; deletion here is inside the step but is a
         # metadata-only soft delete + file removal of OUR OWN output --
         # part of maintaining the system, not skipped work)
+        # soft delete inline (log-only, cheap) so the next step's queries
+        # cannot be served by this step's still-ACTIVE index; the file
+        # vacuum is bookkeeping deferred past the timed region when the
+        # filesystem has room
+        h.delete_index(name)
         if cleanup_inline:
-            # disk-constrained: drop this step's index data immediately
-            h.delete_index(name)
             h.vacuum_index(name)
         else:
             pending_cleanup.append(name)
@@ -199,7 +202,6 @@ def main():
     t_end = time.perf_counter()
 
     for name in pending_cleanup:
-        h.delete_index(name)
         h.vacuum_index(name)
 
     elapsed = t_end - t_start
