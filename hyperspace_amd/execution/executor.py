@@ -224,35 +224,48 @@ class Executor:
         else:
             sort_col = (index.indexed_columns[0]
                         if getattr(index, "indexed_columns", None) else None)
-            batches: List[ColumnBatch] = []
-            seg_counts = torch.zeros(num_buckets + 1, dtype=torch.int64)
+            # one batched read of every wanted file (bucket-major order) —
+            # the reader's thread pool overlaps disk/PCIe/decode across
+            # files; per-bucket segments come from the per-file row counts
+            ordered_paths: List[str] = []
+            files_per_bucket: List[Tuple[int, int]] = []
             for b in wanted_buckets:
                 paths = sorted(by_bucket[b])
-                self.stats.scanned_files += len(paths)
-                if self.device.type == "cuda":
-                    sub, _ = read_files_batch_device(
-                        paths, self.device, columns=read_cols)
-                else:
-                    sub, _ = read_files_batch(paths, columns=read_cols)
-                if len(paths) > 1 and plan.use_bucket_spec and \
-                        sort_col is not None and sub.num_rows:
-                    # each file is sorted, their concatenation is not:
-                    # re-sort the merged bucket so downstream merge joins
-                    # keep their sorted-segment contract (Spark re-sorts
-                    # multi-file buckets inside SortMergeJoinExec the
-                    # same way)
-                    perm = ops.sort_perm(
-                        ops.normalize_key(sub.tensor(sort_col)))
-                    sub = sub.gather(perm)
-                seg_counts[b + 1] = sub.num_rows
-                batches.append(sub)
-            if batches:
-                batch = ColumnBatch.concat(batches)
-            else:
+                files_per_bucket.append((b, len(paths)))
+                ordered_paths.extend(paths)
+            self.stats.scanned_files += len(ordered_paths)
+            if not ordered_paths:
                 batch = ColumnBatch({c: torch.empty(0) for c in read_cols})
-            if self.device.type == "cuda":
-                batch = batch.to(self.device)
+                row_counts = []
+            elif self.device.type == "cuda":
+                batch, row_counts = read_files_batch_device(
+                    ordered_paths, self.device, columns=read_cols)
+            else:
+                batch, row_counts = read_files_batch(
+                    ordered_paths, columns=read_cols)
+            seg_counts = torch.zeros(num_buckets + 1, dtype=torch.int64)
+            fi = 0
+            for b, nfiles in files_per_bucket:
+                seg_counts[b + 1] = sum(row_counts[fi:fi + nfiles])
+                fi += nfiles
             seg = torch.cumsum(seg_counts, 0)
+            # multi-file buckets: each file is sorted but their
+            # concatenation is not — re-sort merged buckets so merge
+            # joins keep the sorted-segment contract (Spark re-sorts
+            # multi-file buckets inside SortMergeJoinExec the same way)
+            if plan.use_bucket_spec and sort_col is not None and \
+                    any(n > 1 for _, n in files_per_bucket) and \
+                    batch.num_rows:
+                pieces: List[ColumnBatch] = []
+                for b, nfiles in files_per_bucket:
+                    lo, hi = int(seg[b]), int(seg[b + 1])
+                    sub = batch.slice(lo, hi)
+                    if nfiles > 1 and sub.num_rows:
+                        perm = ops.sort_perm(
+                            ops.normalize_key(sub.tensor(sort_col)))
+                        sub = sub.gather(perm)
+                    pieces.append(sub)
+                batch = ColumnBatch.concat(pieces)
             if cache:
                 cache.put(key, batch, seg)
 
