@@ -208,6 +208,8 @@ def main():
     # max over ranks
     if distributed:
         t = torch.tensor([elapsed])
+        if on_gpu:
+            t = t.cuda()
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t[0])
 

@@ -74,9 +74,8 @@ def _agree_data_version(data_manager) -> int:
     version = 0 if latest is None else latest + 1
     from ..parallel import dist_context as dc
     if dc.is_distributed() and dc.get_world_size() > 1:
-        import torch
         import torch.distributed as dist
-        t = torch.tensor([version])
+        t = dc.collective_tensor([version])
         dist.broadcast(t, src=0)
         version = int(t[0])
     return version

@@ -120,7 +120,8 @@ class CoveringIndex(Index):
         # collective iteration count: every rank must join every
         # all-to-all, so ranks with fewer groups contribute empty batches
         if distributed:
-            t = torch.tensor([n_groups])
+            from ...parallel import dist_context as dc
+            t = dc.collective_tensor([n_groups])
             dist.all_reduce(t, op=dist.ReduceOp.MAX)
             n_groups = int(t[0])
 

@@ -130,7 +130,7 @@ class ZOrderCoveringIndex(Index):
                 from ...parallel import dist_context as dc
                 if dc.is_distributed() and dc.get_world_size() > 1:
                     import torch.distributed as dist
-                    t = torch.tensor([lo64, -hi64])
+                    t = dc.collective_tensor([lo64, -hi64])
                     dist.all_reduce(t, op=dist.ReduceOp.MIN)
                     lo64, hi64 = int(t[0]), -int(t[1])
                 span = max(1, hi64 - lo64)

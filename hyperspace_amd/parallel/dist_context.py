@@ -32,6 +32,18 @@ def barrier() -> None:
         dist.barrier()
 
 
+def collective_tensor(values):
+    """Small int/float tensor placed correctly for the active backend:
+    NCCL/RCCL collectives need device tensors (gloo-only CPU tests never
+    catch a CPU tensor here — it fails on real multi-GPU runs)."""
+    import torch
+    t = torch.tensor(values)
+    if is_distributed() and dist.get_backend() == "nccl" and \
+            torch.cuda.is_available():
+        t = t.cuda()
+    return t
+
+
 def bucket_owner(bucket: int, world: int) -> int:
     return bucket % world
 
