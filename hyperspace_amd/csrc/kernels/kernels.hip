@@ -557,26 +557,67 @@ __global__ void k_mj_count(const uint64_t* __restrict__ lkeys,
                            int64_t n_seg, int64_t* __restrict__ counts,
                            int64_t* __restrict__ starts,
                            int64_t* __restrict__ seg_of_row) {
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < n_left; i += stride) {
-    // segment of row i: binary search over lseg offsets
-    int64_t lo = 0, hi = n_seg;
-    while (lo < hi) {
-      int64_t mid = (lo + hi) >> 1;
-      if (lseg[mid + 1] <= i)
-        lo = mid + 1;
-      else
-        hi = mid;
+  // Tile-narrowed search: both sides are sorted within a segment, so a
+  // 256-row left tile maps into a contiguous right range.  Two
+  // cooperative searches bound that range; per-row searches then probe
+  // only the (cache-resident) narrowed window instead of walking the
+  // whole multi-MB segment from HBM per row.
+  __shared__ int64_t tile_lo, tile_hi, tile_seg;
+  int64_t n_tiles = cdiv(n_left, (int64_t)blockDim.x);
+  for (int64_t tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
+    int64_t base = tile * (int64_t)blockDim.x;
+    int64_t last = base + blockDim.x - 1 < n_left - 1
+                       ? base + blockDim.x - 1
+                       : n_left - 1;
+    if (threadIdx.x == 0) {
+      // segment of the tile's first row
+      int64_t lo = 0, hi = n_seg;
+      while (lo < hi) {
+        int64_t mid = (lo + hi) >> 1;
+        if (lseg[mid + 1] <= base)
+          lo = mid + 1;
+        else
+          hi = mid;
+      }
+      tile_seg = lo;
+      // does the tile span a segment boundary?
+      if (last < lseg[lo + 1]) {
+        int64_t r0 = rseg[lo], r1 = rseg[lo + 1];
+        tile_lo = lower_bound_u64(rkeys, r0, r1, lkeys[base]);
+        tile_hi = upper_bound_u64(rkeys, r0, r1, lkeys[last]);
+      } else {
+        tile_seg = -1;  // mixed-segment tile: per-row fallback
+      }
     }
-    int64_t s = lo;
-    uint64_t key = lkeys[i];
-    int64_t r0 = rseg[s], r1 = rseg[s + 1];
-    int64_t a = lower_bound_u64(rkeys, r0, r1, key);
-    int64_t b = upper_bound_u64(rkeys, r0, r1, key);
-    counts[i] = b - a;
-    starts[i] = a;
-    seg_of_row[i] = s;
+    __syncthreads();
+    int64_t i = base + threadIdx.x;
+    if (i < n_left) {
+      int64_t s, r0, r1;
+      if (tile_seg >= 0) {
+        s = tile_seg;
+        r0 = tile_lo;
+        r1 = tile_hi;
+      } else {
+        int64_t lo = 0, hi = n_seg;
+        while (lo < hi) {
+          int64_t mid = (lo + hi) >> 1;
+          if (lseg[mid + 1] <= i)
+            lo = mid + 1;
+          else
+            hi = mid;
+        }
+        s = lo;
+        r0 = rseg[s];
+        r1 = rseg[s + 1];
+      }
+      uint64_t key = lkeys[i];
+      int64_t a = lower_bound_u64(rkeys, r0, r1, key);
+      int64_t b = upper_bound_u64(rkeys, r0, r1, key);
+      counts[i] = b - a;
+      starts[i] = a;
+      seg_of_row[i] = s;
+    }
+    __syncthreads();
   }
 }
 
