@@ -489,25 +489,30 @@ class Executor:
                 batch, seg = self._repartition(batch, plan.bucket_columns, n)
             child_parts.append((batch, seg))
 
-        pieces: List[ColumnBatch] = []
-        counts = torch.zeros(n + 1, dtype=torch.int64)
-        for b in range(n):
-            merged = []
-            for batch, seg in child_parts:
-                lo, hi = int(seg[b]), int(seg[b + 1])
-                if hi > lo:
-                    merged.append(batch.slice(lo, hi))
-            if not merged:
+        # vectorized union: concatenate children, rebuild per-row bucket
+        # ids from the segment offsets, then ONE device-wide
+        # sort-by-(bucket, key) pass restores the merged bucketed+sorted
+        # layout (no per-bucket Python loop)
+        from ..index.covering.index import sort_by_bucket_and_keys
+        cols = child_parts[0][0].names
+        batches = []
+        bucket_ids = []
+        for batch, seg in child_parts:
+            if batch.num_rows == 0:
                 continue
-            mb = ColumnBatch.concat(merged)
-            # re-sort bucket by first key for downstream merge join
-            key = ops.normalize_key(mb.tensor(plan.bucket_columns[0]))
-            perm = ops.sort_perm(key)
-            pieces.append(mb.gather(perm))
-            counts[b + 1] = mb.num_rows
-        out = (ColumnBatch.concat(pieces) if pieces
-               else child_parts[0][0].slice(0, 0))
-        return out, torch.cumsum(counts, 0)
+            batches.append(batch.select(cols))
+            seg_d = seg.to(batch.device)
+            counts = (seg_d[1:] - seg_d[:-1])
+            bucket_ids.append(torch.repeat_interleave(
+                torch.arange(n, dtype=torch.int64, device=batch.device),
+                counts))
+        if not batches:
+            return child_parts[0][0].slice(0, 0), torch.zeros(
+                n + 1, dtype=torch.int64)
+        merged = ColumnBatch.concat(batches)
+        all_buckets = torch.cat(bucket_ids).to(torch.int32)
+        return sort_by_bucket_and_keys(merged, all_buckets,
+                                       [plan.bucket_columns[0]], n)
 
     def _repartition(self, batch: ColumnBatch, bucket_cols: List[str],
                      num_buckets: int
