@@ -174,8 +174,10 @@ class CoveringIndex(Index):
             if b is None:
                 raise HyperspaceException(f"Not a bucketed index file: {p}")
             by_bucket.setdefault(b, []).append(p)
-        written: List[str] = []
-        for b, paths in sorted(by_bucket.items()):
+        from concurrent.futures import ThreadPoolExecutor
+
+        def compact_one(item):
+            b, paths = item
             sub, _ = read_files_batch(sorted(paths))
             if ctx.session.device.type == "cuda":
                 sub = sub.to(ctx.session.device)
@@ -184,8 +186,13 @@ class CoveringIndex(Index):
             out = os.path.join(ctx.index_data_path, bucket_file_name(0, b))
             write_batch_parquet(sub.to("cpu") if sub.device.type == "cuda"
                                 else sub, out)
-            written.append(out)
-        return written
+            return out
+
+        items = sorted(by_bucket.items())
+        if len(items) <= 2:
+            return [compact_one(i) for i in items]
+        with ThreadPoolExecutor(max_workers=8) as pool:
+            return list(pool.map(compact_one, items))
 
     def refresh_incremental(self, ctx: IndexerContext,
                             appended_batch: Optional[ColumnBatch],

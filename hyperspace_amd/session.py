@@ -86,6 +86,12 @@ class HyperspaceSession:
         return DataFrame(self, Scan(TextFormatRelation("json",
                                                        list(paths))))
 
+    def read_orc(self, *paths: str):
+        from .dataframe import DataFrame
+        from .plan.nodes import Scan
+        from .sources.text_source import TextFormatRelation
+        return DataFrame(self, Scan(TextFormatRelation("orc", list(paths))))
+
     def read_iceberg(self, path: str, snapshot_id=None):
         """Read an iceberg-style snapshot table (time travel via
         ``snapshot_id``)."""

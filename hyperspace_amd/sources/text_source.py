@@ -1,4 +1,4 @@
-"""CSV / JSON-lines file sources.
+"""CSV / JSON-lines / ORC file sources.
 
 Reference: the default source supports avro,csv,json,orc,parquet,text
 (util/HyperspaceConf.scala:110-115).  These formats have no device
@@ -24,7 +24,7 @@ class TextFormatRelation(FileBasedRelation):
 
     def __init__(self, fmt: str, root_paths: List[str],
                  options: Optional[Dict[str, str]] = None):
-        assert fmt in ("csv", "json")
+        assert fmt in ("csv", "json", "orc")
         self._fmt = fmt
         self._root_paths = [os.path.abspath(p) for p in root_paths]
         self._options = dict(options or {})
@@ -53,10 +53,12 @@ class TextFormatRelation(FileBasedRelation):
         return infos
 
     def read_table(self, path: str):
-        import pyarrow as pa
         if self._fmt == "csv":
             from pyarrow import csv
             return csv.read_csv(path)
+        if self._fmt == "orc":
+            from pyarrow import orc
+            return orc.read_table(path)
         from pyarrow import json as pa_json
         return pa_json.read_json(path)
 
@@ -106,7 +108,7 @@ class TextFormatSourceProvider(FileBasedSourceProvider):
         return isinstance(relation, TextFormatRelation)
 
     def from_metadata(self, metadata: Relation):
-        if metadata.fileFormat not in ("csv", "json"):
+        if metadata.fileFormat not in ("csv", "json", "orc"):
             return None
         return TextFormatRelation(metadata.fileFormat, metadata.rootPaths,
                                   metadata.options)

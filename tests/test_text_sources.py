@@ -82,3 +82,53 @@ def test_csv_refresh_incremental(env):
     accel = q.collect()
     session.disable_hyperspace()
     assert accel.num_rows == q.collect().num_rows
+
+
+def test_orc_index_build_and_query(tmp_path, monkeypatch):
+    import pyarrow as pa
+    from pyarrow import orc
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "idx"))
+    d = tmp_path / "orcdata"
+    d.mkdir()
+    rng = np.random.default_rng(7)
+    t = pa.table({"key": rng.integers(0, 100, 3000),
+                  "val": rng.random(3000)})
+    orc.write_table(t, str(d / "part-0.orc"))
+    session = hs.HyperspaceSession(device="cpu")
+    session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 4)
+    h = hs.Hyperspace(session)
+    df = session.read_orc(str(d))
+    assert df.count() == 3000
+    h.create_index(df, hs.CoveringIndexConfig("oix", ["key"], ["val"]))
+    session.enable_hyperspace()
+    q = df.filter("key = 42").select("key", "val")
+    plan = q.optimized_plan()
+    assert any(isinstance(l, IndexScan) for l in plan.collect_leaves())
+    accel = q.collect()
+    session.disable_hyperspace()
+    assert accel.num_rows == q.collect().num_rows
+
+
+def test_camelcase_api_aliases(tmp_path, monkeypatch):
+    """Reference python binding naming (py4j mirror) works verbatim."""
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "idx"))
+    d = tmp_path / "d"
+    d.mkdir()
+    rng = np.random.default_rng(5)
+    import pyarrow.parquet as pq2
+    import pyarrow as pa
+    pq2.write_table(pa.table({"key": rng.integers(0, 50, 1000),
+                              "val": rng.random(1000)}),
+                    str(d / "p.parquet"))
+    session = hs.HyperspaceSession(device="cpu")
+    session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 2)
+    h = hs.Hyperspace(session)
+    df = session.read_parquet(str(d))
+    h.createIndex(df, hs.CoveringIndexConfig("ax", ["key"], ["val"]))
+    h.refreshIndex("ax", "full")
+    assert h.index("ax")["name"] == "ax"
+    assert h.whyNot(df.filter("key = 1")) is not None
+    h.deleteIndex("ax")
+    h.restoreIndex("ax")
+    h.deleteIndex("ax")
+    h.vacuumIndex("ax")
