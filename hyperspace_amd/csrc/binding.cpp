@@ -254,6 +254,83 @@ torch::Tensor zorder_key(std::vector<torch::Tensor> cols,
   return out;
 }
 
+std::vector<torch::Tensor> parse_rle_runs(torch::Tensor bytes,
+                                           int64_t start, int64_t end,
+                                           int64_t bit_width,
+                                           int64_t num_values) {
+  TORCH_CHECK(!bytes.is_cuda() && bytes.scalar_type() == torch::kUInt8,
+              "bytes must be a cpu u8 tensor");
+  const uint8_t* d = bytes.data_ptr<uint8_t>();
+  std::vector<int64_t> kind, out_off, len, bitoff, value;
+  int64_t pos = start, out = 0;
+  int64_t bw_bytes = (bit_width + 7) / 8;
+  while (out < num_values && pos < end) {
+    // varint header
+    uint64_t h = 0;
+    int shift = 0;
+    while (true) {
+      TORCH_CHECK(pos < end, "rle: truncated varint");
+      uint8_t b = d[pos++];
+      h |= (uint64_t)(b & 0x7F) << shift;
+      if (!(b & 0x80)) break;
+      shift += 7;
+    }
+    if (h & 1) {
+      int64_t groups = (int64_t)(h >> 1);
+      int64_t count = groups * 8;
+      if (count > num_values - out) count = num_values - out;
+      kind.push_back(1);
+      out_off.push_back(out);
+      len.push_back(count);
+      bitoff.push_back(pos * 8);
+      value.push_back(0);
+      pos += groups * bit_width;  // payload bytes = groups * bit_width
+      out += count;
+    } else {
+      int64_t run = (int64_t)(h >> 1);
+      if (run > num_values - out) run = num_values - out;
+      int64_t v = 0;
+      for (int64_t i = 0; i < bw_bytes; i++) {
+        TORCH_CHECK(pos < end, "rle: truncated repeat value");
+        v |= (int64_t)d[pos++] << (8 * i);
+      }
+      kind.push_back(0);
+      out_off.push_back(out);
+      len.push_back(run);
+      bitoff.push_back(0);
+      value.push_back(v);
+      out += run;
+    }
+  }
+  TORCH_CHECK(out == num_values, "rle: produced ", out, " of ",
+              num_values, " values");
+  auto opts = torch::dtype(torch::kInt64);
+  auto mk = [&](std::vector<int64_t>& v) {
+    return torch::tensor(v, opts);
+  };
+  return {mk(kind), mk(out_off), mk(len), mk(bitoff), mk(value)};
+}
+
+torch::Tensor rle_decode(torch::Tensor src_u8_dev, torch::Tensor kind,
+                         torch::Tensor out_off, torch::Tensor len,
+                         torch::Tensor bitoff, torch::Tensor value,
+                         int64_t bit_width, int64_t n_out) {
+  check_cuda(src_u8_dev, "src");
+  auto dev = src_u8_dev.device();
+  auto k = kind.to(dev), o = out_off.to(dev), l = len.to(dev),
+       b = bitoff.to(dev), v = value.to(dev);
+  auto out = torch::empty({n_out},
+                          torch::dtype(torch::kInt32).device(dev));
+  hsk::rle_decode_indices(src_u8_dev.data_ptr<uint8_t>(),
+                          k.data_ptr<int64_t>(), o.data_ptr<int64_t>(),
+                          l.data_ptr<int64_t>(), b.data_ptr<int64_t>(),
+                          v.data_ptr<int64_t>(), kind.numel(),
+                          (int)bit_width,
+                          (uint32_t*)out.data_ptr<int32_t>(), n_out,
+                          current_stream());
+  return out;
+}
+
 void copy_unaligned(torch::Tensor src_u8, int64_t src_off,
                     torch::Tensor dst, int64_t dst_byte_off,
                     int64_t nbytes) {
@@ -324,4 +401,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gather_rows", &gather_rows, "row gather by index");
   m.def("copy_unaligned", &copy_unaligned,
         "device parquet page decode (unaligned copy)");
+  m.def("parse_rle_runs", &parse_rle_runs,
+        "host parse of an RLE/bit-packed hybrid run table");
+  m.def("rle_decode", &rle_decode,
+        "device RLE/bit-packed dictionary-index decode");
 }

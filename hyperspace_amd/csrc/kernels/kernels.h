@@ -85,6 +85,20 @@ void zorder_key(const uint64_t* const* cols, int n_cols, int bits_per_col,
 void gather(const void* in, const int64_t* idx, void* out, int64_t n,
             int elem_size, hipStream_t stream);
 
+// K1 parquet dictionary-page decode: expand an RLE/bit-packed hybrid
+// run table (host-parsed) into u32 dictionary indices.
+// runs arrays (one entry per run):
+//   kind: 0 = repeated (value in run_value), 1 = bit-packed literal
+//         (bits start at src bit offset run_payload_bitoff)
+//   out_off: first output index of the run; len: output count
+void rle_decode_indices(const uint8_t* src, const int64_t* run_kind,
+                        const int64_t* run_out_off,
+                        const int64_t* run_len,
+                        const int64_t* run_payload_bitoff,
+                        const int64_t* run_value, int64_t n_runs,
+                        int bit_width, uint32_t* out, int64_t n_out,
+                        hipStream_t stream);
+
 // K1 parquet page decode: copy nbytes from (src + src_off) — any byte
 // alignment — to 4-byte-aligned dst+dst_off.  nbytes % 4 == 0.  The src
 // buffer must extend >= 4 bytes past src_off+nbytes (parquet footers

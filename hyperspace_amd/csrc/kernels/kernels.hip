@@ -969,6 +969,67 @@ void gather(const void* in, const int64_t* idx, void* out, int64_t n,
 }
 
 // ---------------------------------------------------------------------------
+// K1: parquet RLE/bit-packed hybrid decode (dictionary indices)
+// ---------------------------------------------------------------------------
+
+__global__ void k_rle_decode(const uint8_t* __restrict__ src,
+                             const int64_t* __restrict__ run_kind,
+                             const int64_t* __restrict__ run_out_off,
+                             const int64_t* __restrict__ run_len,
+                             const int64_t* __restrict__ run_bitoff,
+                             const int64_t* __restrict__ run_value,
+                             int64_t n_runs, int bit_width,
+                             uint32_t* __restrict__ out, int64_t n_out) {
+  // one WAVE per run: lanes cooperate on literal bit-unpacking and
+  // repeated fills (runs are short — parquet literal groups are 8
+  // values, writers emit up to ~512-value runs)
+  int64_t waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  int64_t wid = (((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6);
+  int lane = threadIdx.x & 63;
+  uint32_t mask = bit_width >= 32 ? 0xFFFFFFFFu
+                                  : ((1u << bit_width) - 1u);
+  for (int64_t r = wid; r < n_runs; r += waves) {
+    int64_t off = run_out_off[r];
+    int64_t len = run_len[r];
+    if (run_kind[r] == 0) {
+      uint32_t v = (uint32_t)run_value[r];
+      for (int64_t i = lane; i < len; i += 64) out[off + i] = v;
+    } else {
+      int64_t bit0 = run_bitoff[r];
+      for (int64_t i = lane; i < len; i += 64) {
+        int64_t bit = bit0 + i * bit_width;
+        int64_t byte = bit >> 3;
+        int shift = (int)(bit & 7);
+        // little-endian bit order; value spans at most 5 bytes
+        uint64_t word = (uint64_t)src[byte] |
+                        ((uint64_t)src[byte + 1] << 8) |
+                        ((uint64_t)src[byte + 2] << 16) |
+                        ((uint64_t)src[byte + 3] << 24) |
+                        ((uint64_t)src[byte + 4] << 32);
+        out[off + i] = (uint32_t)(word >> shift) & mask;
+      }
+    }
+  }
+}
+
+void rle_decode_indices(const uint8_t* src, const int64_t* run_kind,
+                        const int64_t* run_out_off,
+                        const int64_t* run_len,
+                        const int64_t* run_payload_bitoff,
+                        const int64_t* run_value, int64_t n_runs,
+                        int bit_width, uint32_t* out, int64_t n_out,
+                        hipStream_t stream) {
+  if (n_runs == 0 || n_out == 0) return;
+  int64_t waves_wanted = n_runs < 16384 ? n_runs : 16384;
+  int g = (int)cdiv(waves_wanted * 64, THREADS);
+  if (g < 1) g = 1;
+  if (g > 4096) g = 4096;
+  hipLaunchKernelGGL(k_rle_decode, dim3(g), dim3(THREADS), 0, stream, src,
+                     run_kind, run_out_off, run_len, run_payload_bitoff,
+                     run_value, n_runs, bit_width, out, n_out);
+}
+
+// ---------------------------------------------------------------------------
 // K1: parquet PLAIN page decode = unaligned-source device copy
 // ---------------------------------------------------------------------------
 

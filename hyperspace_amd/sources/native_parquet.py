@@ -317,14 +317,24 @@ def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str
 # ---------------------------------------------------------------------------
 
 class ColumnChunkLayout:
-    """Per-column page list: [(values byte offset, num values), ...]."""
-    __slots__ = ("name", "np_dtype", "pages", "num_values")
+    """Per-column page layout.
 
-    def __init__(self, name, np_dtype, pages, num_values):
+    encoding "plain": pages = [(values byte offset, num values), ...]
+    encoding "dict":  pages = [(payload start, payload end, num values,
+                      bit width), ...] of RLE/bit-packed index payloads,
+                      plus dict_page = (values byte offset, dict size)
+    """
+    __slots__ = ("name", "np_dtype", "pages", "num_values", "encoding",
+                 "dict_page")
+
+    def __init__(self, name, np_dtype, pages, num_values,
+                 encoding="plain", dict_page=None):
         self.name = name
         self.np_dtype = np_dtype
         self.pages = pages
         self.num_values = num_values
+        self.encoding = encoding
+        self.dict_page = dict_page
 
 
 def read_native_layout(path: str,
@@ -346,55 +356,105 @@ def read_native_layout(path: str,
     if data is None:
         with open(path, "rb") as f:
             data = f.read()
+    try:
+        pf_schema = __import__("pyarrow.parquet", fromlist=["x"]) \
+            .ParquetFile(path).schema
+    except Exception:  # noqa: BLE001
+        return None
     rg = md.row_group(0)
     out: List[ColumnChunkLayout] = []
     want = {c.lower() for c in columns} if columns is not None else None
+    col_index = -1
     for i in range(rg.num_columns):
         col = rg.column(i)
+        col_index += 1
         name = col.path_in_schema
         if want is not None and name.lower() not in want:
             continue
         if col.compression.upper() != "UNCOMPRESSED":
             return None
-        if set(col.encodings) - {"PLAIN", "RLE", "BIT_PACKED"}:
+        encs = set(col.encodings)
+        is_dict = bool(encs & {"PLAIN_DICTIONARY", "RLE_DICTIONARY"})
+        if encs - {"PLAIN", "RLE", "BIT_PACKED", "PLAIN_DICTIONARY",
+                   "RLE_DICTIONARY"}:
             return None
         ptype = {"INT64": T_INT64, "INT32": T_INT32, "FLOAT": T_FLOAT,
                  "DOUBLE": T_DOUBLE}.get(col.physical_type)
         if ptype is None:
             return None
         np_dtype = _PARQUET_TO_NP[ptype]
-        # walk pages until the chunk's num_values are covered
+        # nulls unsupported: skip the def-level prefix only when levels
+        # exist; reject columns that actually contain nulls
+        has_levels = pf_schema.column(col_index).max_definition_level > 0
+        st = col.statistics
+        if has_levels and (st is None or st.null_count not in (0, None)):
+            if st is None or st.null_count != 0:
+                return None
+
+        dict_page = None
         pos = col.data_page_offset
+        if is_dict:
+            dict_off = col.dictionary_page_offset
+            if dict_off is None:
+                return None
+            r = TReader(data, dict_off)
+            try:
+                hdr = r.read_struct()
+            except Exception:  # noqa: BLE001
+                return None
+            if hdr.get(1) != 2:  # DICTIONARY_PAGE
+                return None
+            dict_n = hdr.get(7, {}).get(1)
+            if dict_n is None:
+                return None
+            dict_page = (r.pos, dict_n)
+            if dict_off >= pos:
+                # dictionary physically precedes data pages
+                pos = r.pos + hdr.get(3)
+
         seen = 0
-        pages: List[Tuple[int, int]] = []
+        pages: List[Tuple] = []
         while seen < col.num_values:
             r = TReader(data, pos)
             try:
                 hdr = r.read_struct()
             except Exception:  # noqa: BLE001
                 return None
+            page_bytes = hdr.get(3)
+            if hdr.get(1) == 2:  # stray dictionary page: skip
+                pos = r.pos + page_bytes
+                continue
             if hdr.get(1) != PAGE_DATA:
                 return None
-            page_bytes = hdr.get(3)
             dph = hdr.get(5, {})
             num_values = dph.get(1)
             if num_values is None:
                 return None
             values_off = r.pos
-            expected = num_values * np_dtype.itemsize
-            if page_bytes != expected:
-                # v1 page with def levels: 4-byte len prefix + RLE run
+            page_end = r.pos + page_bytes
+            if has_levels:
                 lvl_len = struct.unpack_from("<I", data, values_off)[0]
-                if 4 + lvl_len + expected != page_bytes:
-                    return None
                 values_off += 4 + lvl_len
-            pages.append((values_off, num_values))
+            page_enc = dph.get(2)
+            if is_dict and page_enc in (2, 8):  # PLAIN_DICT / RLE_DICT
+                bit_width = data[values_off]
+                pages.append((values_off + 1, page_end, num_values,
+                              bit_width))
+            elif page_enc == ENC_PLAIN:
+                expected = num_values * np_dtype.itemsize
+                if values_off + expected > page_end:
+                    return None
+                pages.append((values_off, num_values))
+            else:
+                return None
             seen += num_values
-            pos = r.pos + page_bytes
+            pos = page_end
         if seen != col.num_values:
             return None
+        enc_kind = "dict" if (is_dict and dict_page and
+                              len(pages[0]) == 4) else "plain"
         out.append(ColumnChunkLayout(name, np_dtype, pages,
-                                     col.num_values))
+                                     col.num_values, enc_kind, dict_page))
     return data, out
 
 
@@ -405,6 +465,8 @@ def read_native_host(path: str, columns: Optional[List[str]] = None
     if layout is None:
         return None
     data, chunks = layout
+    if any(c.encoding != "plain" for c in chunks):
+        return None  # dictionary decode is the device path; host->pyarrow
     out = {}
     for c in chunks:
         parts = [np.frombuffer(data, dtype=c.np_dtype, count=nv,

@@ -184,10 +184,30 @@ def read_files_batch_device(paths: List[str], device,
         for c in lay[1]:
             itemsize = c.np_dtype.itemsize
             written = int(file_base[i])
-            for off, nv in c.pages:
-                ext.copy_unaligned(dev_bytes, off, out[c.name],
-                                   written * itemsize, nv * itemsize)
-                written += nv
+            if c.encoding == "dict":
+                # K1 dictionary path: decode the dictionary page with the
+                # unaligned copy, expand RLE/bit-packed index runs on
+                # device, then gather values through the dictionary
+                dict_off, dict_n = c.dict_page
+                dict_vals = _torch.empty(
+                    dict_n + 1,  # +1 slack for the 4B-overread
+                    dtype=out[c.name].dtype, device=device)
+                ext.copy_unaligned(dev_bytes, dict_off, dict_vals, 0,
+                                   dict_n * itemsize)
+                idx_parts = []
+                for p_start, p_end, nv, bw in c.pages:
+                    runs = ext.parse_rle_runs(buf, p_start, p_end, bw, nv)
+                    idx_parts.append(ext.rle_decode(
+                        dev_bytes, *runs, bw, nv))
+                idx = (_torch.cat(idx_parts) if len(idx_parts) > 1
+                       else idx_parts[0]).to(_torch.int64)
+                out[c.name][written:written + c.num_values] = \
+                    ext.gather_rows(dict_vals[:dict_n].contiguous(), idx)
+            else:
+                for off, nv in c.pages:
+                    ext.copy_unaligned(dev_bytes, off, out[c.name],
+                                       written * itemsize, nv * itemsize)
+                    written += nv
         # pinned buffer must stay alive until the stream drains; the
         # caller-side synchronize below holds them via `bufs`
         return buf

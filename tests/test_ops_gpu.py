@@ -225,3 +225,41 @@ def test_select_range_one_sided_sentinels():
     assert got.numel() == int((vals >= 5000).sum())
     got = ops.select_range_u64(keys.cuda(), 0, v, True, False).cpu()
     assert got.numel() == int((vals < 5000).sum())
+
+
+def test_device_dictionary_decode(tmp_path):
+    """K1 dictionary path: pyarrow's default dictionary-encoded pages
+    decode on device (RLE/bit-packed runs + dictionary gather)."""
+    import numpy as np
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from hyperspace_amd.sources.parquet_io import read_files_batch_device
+    rng = np.random.default_rng(33)
+    cols = {
+        "low_card": rng.integers(0, 7, 300_000),          # bw ~3
+        "mid_card": rng.integers(0, 40_000, 300_000),     # bw ~16
+        "fval": np.round(rng.random(300_000) * 100, 1),   # float dict
+    }
+    p = str(tmp_path / "dict.parquet")
+    pq.write_table(pa.table(cols), p, compression="NONE")  # dict default
+    md = pq.ParquetFile(p).metadata.row_group(0)
+    assert any("DICTIONARY" in e for e in md.column(0).encodings)
+    dev, counts = read_files_batch_device([p], "cuda")
+    ref = pq.read_table(p)
+    for name in cols:
+        got = dev.tensor(name).cpu().numpy()
+        exp = ref.column(name).to_numpy()
+        assert np.array_equal(got, exp), name
+
+
+def test_device_dictionary_decode_repeat_runs(tmp_path):
+    """Long repeated runs (sorted low-cardinality data -> RLE repeats)."""
+    import numpy as np
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from hyperspace_amd.sources.parquet_io import read_files_batch_device
+    vals = np.repeat(np.arange(20, dtype=np.int64), 50_000)  # 1M rows
+    p = str(tmp_path / "runs.parquet")
+    pq.write_table(pa.table({"v": vals}), p, compression="NONE")
+    dev, _ = read_files_batch_device([p], "cuda")
+    assert np.array_equal(dev.tensor("v").cpu().numpy(), vals)
