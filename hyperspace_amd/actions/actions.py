@@ -178,27 +178,14 @@ class CreateAction(Action):
 
     def log_entry_for_begin(self) -> IndexLogEntry:
         # content not yet known at begin(): record config-level metadata
-        from ..plan.nodes import Scan
         from ..log.entry import Directory
         relation = self.df.plan.collect_leaves()[0].relation
         rel_meta = relation.create_relation_metadata(FileIdTracker())
-        from ..index.covering.config import CoveringIndexConfig
-        placeholder = self._placeholder_index(relation)
+        placeholder = self.config.placeholder_index(
+            relation, self.session.conf)
         return IndexLogEntry.create(
             self.config.index_name, placeholder, Content(Directory("")),
             Source(SourcePlan([rel_meta], _fingerprint(relation))), {})
-
-    def _placeholder_index(self, relation):
-        from ..index.covering.index import CoveringIndex
-        from ..utils.resolver import resolve_all
-        schema = relation.schema
-        indexed = resolve_all(schema.field_names(),
-                              getattr(self.config, "indexed_columns",
-                                      self.config.referenced_columns()))
-        included = [c for c in self.config.referenced_columns()
-                    if c not in indexed]
-        return CoveringIndex(indexed, included, schema.select(
-            indexed + included), self.session.conf.num_buckets, {})
 
     def event(self, message):
         return CreateActionEvent(index_name=self.config.index_name,

@@ -88,3 +88,8 @@ class IndexConfigTrait(ABC):
     def create_index(self, ctx: IndexerContext, df,
                      properties: Dict[str, str]) -> Tuple[Index, Any]:
         """Returns (Index instance, index data ColumnBatch/DataFrame)."""
+
+    def placeholder_index(self, relation, conf) -> Index:
+        """Kind-appropriate metadata-only Index for the begin() log entry
+        (content is unknown until op() completes)."""
+        raise NotImplementedError

@@ -107,3 +107,12 @@ class CoveringIndexConfig(IndexConfigTrait):
         index = CoveringIndex(indexed, included, index_schema, num_buckets,
                               dict(properties))
         return index, batch
+
+    def placeholder_index(self, relation, conf):
+        from ...utils.resolver import resolve_all
+        schema = relation.schema
+        indexed = resolve_all(schema.field_names(), self.indexed_columns)
+        included = resolve_all(schema.field_names(), self.included_columns)
+        return CoveringIndex(indexed, included,
+                             schema.select(indexed + included),
+                             conf.num_buckets, {})

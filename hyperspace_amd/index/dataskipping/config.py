@@ -88,3 +88,9 @@ class DataSkippingIndexConfig(IndexConfigTrait):
         index = DataSkippingIndex(self.sketches, index_schema,
                                   dict(properties))
         return index, index_data
+
+    def placeholder_index(self, relation, conf):
+        cols = resolve_all(relation.schema.field_names(),
+                           self.referenced_columns())
+        return DataSkippingIndex(
+            self.sketches, relation.schema.select(cols), {})

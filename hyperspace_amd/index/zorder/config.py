@@ -29,3 +29,11 @@ class ZOrderCoveringIndexConfig(CoveringIndexConfig):
             cov_index.indexed_columns, cov_index.included_columns,
             cov_index.schema, dict(properties))
         return index, batch
+
+    def placeholder_index(self, relation, conf):
+        from ...utils.resolver import resolve_all
+        schema = relation.schema
+        indexed = resolve_all(schema.field_names(), self.indexed_columns)
+        included = resolve_all(schema.field_names(), self.included_columns)
+        return ZOrderCoveringIndex(indexed, included,
+                                   schema.select(indexed + included), {})
