@@ -348,22 +348,30 @@ def read_native_layout(path: str,
     buffer protocol object) to avoid a second disk read."""
     import pyarrow.parquet as pq
     try:
-        md = pq.ParquetFile(path).metadata
+        pf = pq.ParquetFile(path)
+        md = pf.metadata
+        pf_schema = pf.schema
     except Exception:  # noqa: BLE001
-        return None
-    if md.num_row_groups != 1:
         return None
     if data is None:
         with open(path, "rb") as f:
             data = f.read()
-    try:
-        pf_schema = __import__("pyarrow.parquet", fromlist=["x"]) \
-            .ParquetFile(path).schema
-    except Exception:  # noqa: BLE001
-        return None
-    rg = md.row_group(0)
     out: List[ColumnChunkLayout] = []
     want = {c.lower() for c in columns} if columns is not None else None
+    # row-group-major walk: large files carry many row groups; each
+    # column chunk (and its dictionary) is per-row-group
+    for rg_i in range(md.num_row_groups):
+        rg = md.row_group(rg_i)
+        lay = _walk_row_group(rg, pf_schema, data, want)
+        if lay is None:
+            return None
+        out.extend(lay)
+    return data, out
+
+
+def _walk_row_group(rg, pf_schema, data, want
+                    ) -> Optional[List[ColumnChunkLayout]]:
+    out: List[ColumnChunkLayout] = []
     col_index = -1
     for i in range(rg.num_columns):
         col = rg.column(i)
@@ -455,7 +463,7 @@ def read_native_layout(path: str,
                               len(pages[0]) == 4) else "plain"
         out.append(ColumnChunkLayout(name, np_dtype, pages,
                                      col.num_values, enc_kind, dict_page))
-    return data, out
+    return out
 
 
 def read_native_host(path: str, columns: Optional[List[str]] = None
@@ -467,10 +475,11 @@ def read_native_host(path: str, columns: Optional[List[str]] = None
     data, chunks = layout
     if any(c.encoding != "plain" for c in chunks):
         return None  # dictionary decode is the device path; host->pyarrow
-    out = {}
-    for c in chunks:
+    acc: Dict[str, list] = {}
+    for c in chunks:  # row-group-major order
         parts = [np.frombuffer(data, dtype=c.np_dtype, count=nv,
                                offset=off) for off, nv in c.pages]
-        out[c.name] = (np.concatenate(parts) if len(parts) > 1
-                       else parts[0].copy())
-    return out
+        acc.setdefault(c.name, []).extend(parts)
+    return {name: (np.concatenate(parts) if len(parts) > 1
+                   else parts[0].copy())
+            for name, parts in acc.items()}

@@ -137,8 +137,6 @@ def read_files_batch_device(paths: List[str], device,
             md = pq.ParquetFile(p).metadata
         except Exception:  # noqa: BLE001
             return fallback()
-        if md.num_row_groups != 1:
-            return fallback()
         metas.append(md)
 
     want = {c.lower() for c in columns} if columns is not None else None
@@ -158,7 +156,7 @@ def read_files_batch_device(paths: List[str], device,
         order = {c.lower(): i for i, c in enumerate(columns)}
         names.sort(key=lambda n: order.get(n.lower(), 99))
 
-    row_counts = [md.row_group(0).num_rows for md in metas]
+    row_counts = [md.num_rows for md in metas]
     file_base = np.concatenate([[0], np.cumsum(row_counts)[:-1]])
     total_rows = int(sum(row_counts))
 
@@ -181,9 +179,10 @@ def read_files_batch_device(paths: List[str], device,
         if lay is None:
             return None
         dev_bytes = buf.to(device, non_blocking=True)
+        cursors = {n: int(file_base[i]) for n in names}
         for c in lay[1]:
             itemsize = c.np_dtype.itemsize
-            written = int(file_base[i])
+            written = cursors[c.name]
             if c.encoding == "dict":
                 # K1 dictionary path: decode the dictionary page with the
                 # unaligned copy, expand RLE/bit-packed index runs on
@@ -203,11 +202,13 @@ def read_files_batch_device(paths: List[str], device,
                        else idx_parts[0]).to(_torch.int64)
                 out[c.name][written:written + c.num_values] = \
                     ext.gather_rows(dict_vals[:dict_n].contiguous(), idx)
+                written += c.num_values
             else:
                 for off, nv in c.pages:
                     ext.copy_unaligned(dev_bytes, off, out[c.name],
                                        written * itemsize, nv * itemsize)
                     written += nv
+            cursors[c.name] = written
         # pinned buffer must stay alive until the stream drains; the
         # caller-side synchronize below holds them via `bufs`
         return buf

@@ -88,3 +88,20 @@ def test_empty_and_single_row(tmp_path):
     assert pq.read_table(p).column("a").to_pylist() == [7]
     back = read_native_host(p)
     assert back["a"].tolist() == [7]
+
+
+def test_multi_rowgroup_native_read(tmp_path):
+    rng = np.random.default_rng(9)
+    cols = {"key": rng.integers(-(10**9), 10**9, 1_200_000),
+            "val": rng.random(1_200_000)}
+    p = str(tmp_path / "mrg.parquet")
+    pq.write_table(pa.table(cols), p, compression="NONE",
+                   use_dictionary=False, data_page_version="1.0",
+                   row_group_size=250_000)
+    assert pq.ParquetFile(p).metadata.num_row_groups > 1
+    back = read_native_host(p)
+    assert back is not None
+    for k, v in cols.items():
+        assert np.array_equal(back[k], v), k
+    batch, counts = read_files_batch([p])
+    assert counts == [1_200_000]

@@ -263,3 +263,22 @@ def test_device_dictionary_decode_repeat_runs(tmp_path):
     pq.write_table(pa.table({"v": vals}), p, compression="NONE")
     dev, _ = read_files_batch_device([p], "cuda")
     assert np.array_equal(dev.tensor("v").cpu().numpy(), vals)
+
+
+def test_device_decode_multi_rowgroup_dict(tmp_path):
+    import numpy as np
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from hyperspace_amd.sources.parquet_io import read_files_batch_device
+    rng = np.random.default_rng(44)
+    cols = {"key": rng.integers(0, 5000, 1_500_000),
+            "val": rng.random(1_500_000)}
+    p = str(tmp_path / "mrgd.parquet")
+    pq.write_table(pa.table(cols), p, compression="NONE",
+                   row_group_size=400_000)  # dict + 4 row groups
+    dev, counts = read_files_batch_device([p], "cuda")
+    assert counts == [1_500_000]
+    ref = pq.read_table(p)
+    for name in cols:
+        assert np.array_equal(dev.tensor(name).cpu().numpy(),
+                              ref.column(name).to_numpy()), name
