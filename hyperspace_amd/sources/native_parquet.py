@@ -446,21 +446,23 @@ def _walk_row_group(rg, pf_schema, data, want
             page_enc = dph.get(2)
             if is_dict and page_enc in (2, 8):  # PLAIN_DICT / RLE_DICT
                 bit_width = data[values_off]
-                pages.append((values_off + 1, page_end, num_values,
-                              bit_width))
+                pages.append(("dict", values_off + 1, page_end,
+                              num_values, bit_width))
             elif page_enc == ENC_PLAIN:
+                # also reached as the writer's mid-chunk fallback when a
+                # dictionary overflows: later pages switch to PLAIN
                 expected = num_values * np_dtype.itemsize
                 if values_off + expected > page_end:
                     return None
-                pages.append((values_off, num_values))
+                pages.append(("plain", values_off, num_values))
             else:
                 return None
             seen += num_values
             pos = page_end
         if seen != col.num_values:
             return None
-        enc_kind = "dict" if (is_dict and dict_page and
-                              len(pages[0]) == 4) else "plain"
+        enc_kind = ("dict" if any(pg[0] == "dict" for pg in pages)
+                    else "plain")
         out.append(ColumnChunkLayout(name, np_dtype, pages,
                                      col.num_values, enc_kind, dict_page))
     return out
@@ -478,7 +480,7 @@ def read_native_host(path: str, columns: Optional[List[str]] = None
     acc: Dict[str, list] = {}
     for c in chunks:  # row-group-major order
         parts = [np.frombuffer(data, dtype=c.np_dtype, count=nv,
-                               offset=off) for off, nv in c.pages]
+                               offset=off) for _, off, nv in c.pages]
         acc.setdefault(c.name, []).extend(parts)
     return {name: (np.concatenate(parts) if len(parts) > 1
                    else parts[0].copy())

@@ -183,31 +183,32 @@ def read_files_batch_device(paths: List[str], device,
         for c in lay[1]:
             itemsize = c.np_dtype.itemsize
             written = cursors[c.name]
+            dict_vals = None
             if c.encoding == "dict":
-                # K1 dictionary path: decode the dictionary page with the
-                # unaligned copy, expand RLE/bit-packed index runs on
-                # device, then gather values through the dictionary
+                # K1 dictionary path: decode the dictionary page once;
+                # pages gather through it (a chunk can also carry PLAIN
+                # pages — the writer's mid-chunk dictionary-overflow
+                # fallback)
                 dict_off, dict_n = c.dict_page
                 dict_vals = _torch.empty(
                     dict_n + 1,  # +1 slack for the 4B-overread
                     dtype=out[c.name].dtype, device=device)
                 ext.copy_unaligned(dev_bytes, dict_off, dict_vals, 0,
                                    dict_n * itemsize)
-                idx_parts = []
-                for p_start, p_end, nv, bw in c.pages:
+                dict_vals = dict_vals[:dict_n].contiguous()
+            for page in c.pages:
+                if page[0] == "dict":
+                    _, p_start, p_end, nv, bw = page
                     runs = ext.parse_rle_runs(buf, p_start, p_end, bw, nv)
-                    idx_parts.append(ext.rle_decode(
-                        dev_bytes, *runs, bw, nv))
-                idx = (_torch.cat(idx_parts) if len(idx_parts) > 1
-                       else idx_parts[0]).to(_torch.int64)
-                out[c.name][written:written + c.num_values] = \
-                    ext.gather_rows(dict_vals[:dict_n].contiguous(), idx)
-                written += c.num_values
-            else:
-                for off, nv in c.pages:
+                    idx = ext.rle_decode(dev_bytes, *runs, bw,
+                                         nv).to(_torch.int64)
+                    out[c.name][written:written + nv] = \
+                        ext.gather_rows(dict_vals, idx)
+                else:
+                    _, off, nv = page
                     ext.copy_unaligned(dev_bytes, off, out[c.name],
                                        written * itemsize, nv * itemsize)
-                    written += nv
+                written += nv
             cursors[c.name] = written
         # pinned buffer must stay alive until the stream drains; the
         # caller-side synchronize below holds them via `bufs`
