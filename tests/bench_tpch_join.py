@@ -1,0 +1,107 @@
+"""TPC-H-shaped join benchmark (BASELINE config 4's workload at reduced
+scale): lineitem ⋈ orders on co-partitioned covering indexes, bucketed
+sort-merge join with zero exchange.  Run on a GPU box."""
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import numpy as np
+import torch
+
+import hyperspace_amd as hs
+from hyperspace_amd.execution.executor import Executor
+from hyperspace_amd.plan.nodes import IndexScan
+from hyperspace_amd.sources.native_parquet import write_parquet_native
+
+SF = float(os.environ.get("TPCH_SF", 30))
+GB = 1 << 30
+
+
+def gen(workdir):
+    rng = np.random.default_rng(0)
+    li_dir = os.path.join(workdir, "lineitem")
+    o_dir = os.path.join(workdir, "orders")
+    os.makedirs(li_dir, exist_ok=True)
+    os.makedirs(o_dir, exist_ok=True)
+    n_orders = int(1_500_000 * SF)
+    n_li = int(6_000_000 * SF)
+    per_file = 8_000_000
+    for i in range(0, n_li, per_file):
+        n = min(per_file, n_li - i)
+        write_parquet_native({
+            "l_orderkey": rng.integers(0, n_orders, n),
+            "l_partkey": rng.integers(0, 200_000 * SF, n).astype(np.int64),
+            "l_quantity": rng.integers(1, 51, n).astype(np.float64),
+            "l_extendedprice": rng.random(n) * 100_000,
+        }, os.path.join(li_dir, f"part-{i//per_file:05d}.parquet"))
+    for i in range(0, n_orders, per_file):
+        n = min(per_file, n_orders - i)
+        write_parquet_native({
+            "o_orderkey": np.arange(i, i + n, dtype=np.int64),
+            "o_custkey": rng.integers(0, 150_000 * SF, n).astype(np.int64),
+            "o_totalprice": rng.random(n) * 500_000,
+        }, os.path.join(o_dir, f"part-{i//per_file:05d}.parquet"))
+    li_b = sum(os.path.getsize(os.path.join(li_dir, f))
+               for f in os.listdir(li_dir))
+    o_b = sum(os.path.getsize(os.path.join(o_dir, f))
+              for f in os.listdir(o_dir))
+    return li_dir, o_dir, li_b, o_b, n_li, n_orders
+
+
+def t_sync():
+    torch.cuda.synchronize()
+    return time.perf_counter()
+
+
+def main():
+    work = "/dev/shm/tpch_bench" if os.path.isdir("/dev/shm") \
+        else "/tmp/tpch_bench"
+    os.environ["HYPERSPACE_SYSTEM_PATH"] = os.path.join(work, "indexes")
+    t0 = time.perf_counter()
+    li_dir, o_dir, li_b, o_b, n_li, n_orders = gen(work)
+    print(f"gen SF{SF:g}: lineitem {n_li/1e6:.0f}M rows {li_b/GB:.1f}G, "
+          f"orders {n_orders/1e6:.0f}M rows {o_b/GB:.1f}G "
+          f"in {time.perf_counter()-t0:.1f}s")
+
+    session = hs.HyperspaceSession(device="cuda")
+    session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 200)
+    h = hs.Hyperspace(session)
+    li = session.read_parquet(li_dir)
+    orders = session.read_parquet(o_dir)
+
+    t0 = time.perf_counter()
+    h.create_index(li, hs.CoveringIndexConfig(
+        "li_ix", ["l_orderkey"], ["l_quantity", "l_extendedprice"]))
+    t1 = t_sync()
+    h.create_index(orders, hs.CoveringIndexConfig(
+        "o_ix", ["o_orderkey"], ["o_totalprice"]))
+    t2 = t_sync()
+    print(f"build lineitem index: {t1-t0:.2f}s = {li_b/GB/(t1-t0):.2f} "
+          f"GB/s; orders index: {t2-t1:.2f}s")
+
+    session.enable_hyperspace()
+    q = li.select("l_orderkey", "l_quantity", "l_extendedprice").join(
+        orders.select("o_orderkey", "o_totalprice"),
+        on=(hs.col("l_orderkey") == hs.col("o_orderkey")))
+    plan = q.optimized_plan()
+    assert sum(isinstance(l, IndexScan)
+               for l in plan.collect_leaves()) == 2, plan.pretty()
+    ex = Executor(session)
+    t3 = t_sync()
+    out = ex.execute(plan)
+    t4 = t_sync()
+    assert ex.stats.shuffles == 0
+    for i in range(3):
+        ex = Executor(session)
+        ta = t_sync()
+        out = ex.execute(plan)
+        tb = t_sync()
+        print(f"join warm {i}: {tb-ta:.3f}s  rows={out.num_rows/1e6:.1f}M "
+              f"(zero-shuffle co-bucketed)")
+    print(f"join cold: {t4-t3:.2f}s")
+
+
+if __name__ == "__main__":
+    main()
