@@ -37,14 +37,18 @@ int normalize_dtype_code(torch::ScalarType t) {
 }
 
 torch::Tensor murmur3_bucket(std::vector<torch::Tensor> keys,
-                             int64_t num_buckets) {
+                             int64_t num_buckets,
+                             std::vector<torch::Tensor> masks = {}) {
   TORCH_CHECK(!keys.empty(), "need at least one key column");
+  TORCH_CHECK(masks.empty() || masks.size() == keys.size(),
+              "masks must be empty or parallel to keys");
   auto n = keys[0].numel();
   auto dev = keys[0].device();
   auto h = torch::empty({n}, torch::dtype(torch::kInt32).device(dev));
   auto stream = current_stream();
   bool first = true;
-  for (auto& k : keys) {
+  for (size_t ci = 0; ci < keys.size(); ++ci) {
+    auto& k = keys[ci];
     check_cuda(k, "key");
     TORCH_CHECK(k.numel() == n, "key column length mismatch");
     torch::Tensor col = k;
@@ -67,7 +71,19 @@ torch::Tensor murmur3_bucket(std::vector<torch::Tensor> keys,
       default:
         TORCH_CHECK(false, "unsupported hash-key dtype");
     }
-    hsk::murmur3_column(col.data_ptr(), kind,
+    // an empty mask tensor means "column has no nulls"
+    const uint8_t* valid = nullptr;
+    torch::Tensor mcol;
+    if (!masks.empty() && masks[ci].numel() > 0) {
+      mcol = masks[ci];
+      check_cuda(mcol, "mask");
+      TORCH_CHECK(mcol.numel() == n, "mask length mismatch");
+      TORCH_CHECK(mcol.scalar_type() == torch::kBool ||
+                      mcol.scalar_type() == torch::kUInt8,
+                  "mask must be bool/uint8");
+      valid = (const uint8_t*)mcol.data_ptr();
+    }
+    hsk::murmur3_column(col.data_ptr(), kind, valid,
                         (uint32_t*)h.data_ptr<int32_t>(), n, first,
                         /*seed=*/42u, stream);
     first = false;
@@ -388,7 +404,9 @@ torch::Tensor select_range_u64_pub(torch::Tensor keys, int64_t lo,
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("murmur3_bucket", &murmur3_bucket, "Spark-compatible murmur3 bucket");
+  m.def("murmur3_bucket", &murmur3_bucket, "Spark-compatible murmur3 bucket",
+        py::arg("keys"), py::arg("num_buckets"),
+        py::arg("masks") = std::vector<torch::Tensor>{});
   m.def("normalize_key", &normalize_key, "order-preserving u64 key");
   m.def("radix_sort_pairs", &radix_sort_pairs, "stable LSD radix sort");
   m.def("merge_join", &merge_join, "segmented sorted merge join");

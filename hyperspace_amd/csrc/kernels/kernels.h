@@ -9,8 +9,11 @@ namespace hsk {
 
 // K2: Spark-compatible Murmur3 fold over one column into h (u32 in u32 buf).
 // kind: 0 = i32-hash (int32 input), 1 = i64-hash (int64 input).
-void murmur3_column(const void* vals, int kind, uint32_t* h, int64_t n,
-                    bool first, uint32_t seed, hipStream_t stream);
+// valid: optional per-row validity bytes — a null row leaves h unchanged
+// (Spark Murmur3Hash skips null children); nullptr = all valid.
+void murmur3_column(const void* vals, int kind, const uint8_t* valid,
+                    uint32_t* h, int64_t n, bool first, uint32_t seed,
+                    hipStream_t stream);
 void pmod_buckets(const uint32_t* h, int32_t* out, int64_t n,
                   int32_t num_buckets, hipStream_t stream);
 

@@ -98,12 +98,20 @@ __device__ __forceinline__ uint32_t murmur_i64(uint64_t v, uint32_t seed) {
 }
 
 template <typename T, bool IS64>
-__global__ void k_murmur(const T* __restrict__ vals, uint32_t* __restrict__ h,
+__global__ void k_murmur(const T* __restrict__ vals,
+                         const uint8_t* __restrict__ valid,
+                         uint32_t* __restrict__ h,
                          int64_t n, bool first, uint32_t seed) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
     uint32_t s = first ? seed : h[i];
+    if (valid != nullptr && !valid[i]) {
+      // Spark HashPartitioning: a null child leaves the running hash
+      // unchanged (Murmur3Hash.eval skips null inputs)
+      h[i] = s;
+      continue;
+    }
     if (IS64)
       h[i] = murmur_i64((uint64_t)vals[i], s);
     else
@@ -122,15 +130,18 @@ __global__ void k_pmod(const uint32_t* __restrict__ h,
   }
 }
 
-void murmur3_column(const void* vals, int kind, uint32_t* h, int64_t n,
-                    bool first, uint32_t seed, hipStream_t stream) {
+void murmur3_column(const void* vals, int kind, const uint8_t* valid,
+                    uint32_t* h, int64_t n, bool first, uint32_t seed,
+                    hipStream_t stream) {
   int g = grid_for(n);
   if (kind == 1) {
     hipLaunchKernelGGL((k_murmur<uint64_t, true>), dim3(g), dim3(THREADS), 0,
-                       stream, (const uint64_t*)vals, h, n, first, seed);
+                       stream, (const uint64_t*)vals, valid, h, n, first,
+                       seed);
   } else {
     hipLaunchKernelGGL((k_murmur<uint32_t, false>), dim3(g), dim3(THREADS), 0,
-                       stream, (const uint32_t*)vals, h, n, first, seed);
+                       stream, (const uint32_t*)vals, valid, h, n, first,
+                       seed);
   }
 }
 

@@ -7,8 +7,13 @@ iterators; ours is columnar tensors sized for 288 GB HBM3E per GPU).
 
 Strings are dictionary-encoded: a ``StringColumn`` holds int32 codes plus a
 sorted value dictionary, so comparisons and sorts on codes are
-order-correct.  Nulls are not supported in v0 (synthetic + TPC-H-shaped data
-is non-null); validity masks are a planned extension.
+order-correct.
+
+Nullable columns carry a per-column validity mask (bool tensor, True =
+valid) in ``ColumnBatch.masks``; a column with no mask entry is entirely
+non-null, so null-free data pays nothing.  Null semantics follow Spark:
+comparisons/IN never match null, inner joins drop null keys, sorts are
+NULLS FIRST, hash partitioning passes the seed through null keys.
 """
 
 from __future__ import annotations
@@ -78,10 +83,16 @@ Column = Union[torch.Tensor, StringColumn]
 
 
 class ColumnBatch:
-    """Ordered name -> column map; all columns share the same row count."""
+    """Ordered name -> column map; all columns share the same row count.
 
-    def __init__(self, columns: "Dict[str, Column]"):
+    ``masks`` maps column name -> bool validity tensor (True = valid);
+    columns absent from ``masks`` have no nulls.
+    """
+
+    def __init__(self, columns: "Dict[str, Column]",
+                 masks: "Optional[Dict[str, torch.Tensor]]" = None):
         self.columns: Dict[str, Column] = dict(columns)
+        self.masks: Dict[str, torch.Tensor] = dict(masks or {})
         n = None
         for name, col in self.columns.items():
             cn = len(col) if isinstance(col, StringColumn) else col.numel()
@@ -91,6 +102,10 @@ class ColumnBatch:
                 raise HyperspaceException(
                     f"Column {name} length {cn} != {n}")
         self._num_rows = n or 0
+        for name, m in self.masks.items():
+            if m.numel() != self._num_rows:
+                raise HyperspaceException(
+                    f"Mask {name} length {m.numel()} != {self._num_rows}")
 
     # -- basic accessors --------------------------------------------------
     @property
@@ -114,6 +129,24 @@ class ColumnBatch:
 
     def has_column(self, name: str) -> bool:
         return any(k.lower() == name.lower() for k in self.columns)
+
+    def mask(self, name: str) -> Optional[torch.Tensor]:
+        """Validity mask (True = valid) for a column, or None if the
+        column has no nulls."""
+        for k, v in self.masks.items():
+            if k.lower() == name.lower():
+                return v
+        return None
+
+    def has_nulls(self, name: str) -> bool:
+        m = self.mask(name)
+        return m is not None and not bool(m.all())
+
+    def _stored_key(self, name: str) -> str:
+        for k in self.columns:
+            if k.lower() == name.lower():
+                return k
+        raise HyperspaceException(f"No column {name}")
 
     @property
     def device(self) -> torch.device:
@@ -139,36 +172,47 @@ class ColumnBatch:
                     break
             else:
                 raise HyperspaceException(f"No column {n}")
-        return ColumnBatch(out)
+        masks = {k: m for k, m in self.masks.items() if k in out}
+        return ColumnBatch(out, masks)
 
-    def with_column(self, name: str, col: Column) -> "ColumnBatch":
+    def with_column(self, name: str, col: Column,
+                    mask: Optional[torch.Tensor] = None) -> "ColumnBatch":
         out = dict(self.columns)
         out[name] = col
-        return ColumnBatch(out)
+        masks = {k: m for k, m in self.masks.items()
+                 if k.lower() != name.lower()}
+        if mask is not None:
+            masks[name] = mask
+        return ColumnBatch(out, masks)
 
     def drop(self, name: str) -> "ColumnBatch":
         out = {k: v for k, v in self.columns.items()
                if k.lower() != name.lower()}
-        return ColumnBatch(out)
+        masks = {k: m for k, m in self.masks.items()
+                 if k.lower() != name.lower()}
+        return ColumnBatch(out, masks)
 
     def gather(self, idx: torch.Tensor) -> "ColumnBatch":
         from .. import ops
-        return ColumnBatch({
-            k: (StringColumn(ops.gather_rows(v.codes, idx.to(
-                    v.codes.device)), v.values)
-                if isinstance(v, StringColumn)
-                else ops.gather_rows(v, idx.to(v.device)))
-            for k, v in self.columns.items()})
+        return ColumnBatch(
+            {k: (StringColumn(ops.gather_rows(v.codes, idx.to(
+                     v.codes.device)), v.values)
+                 if isinstance(v, StringColumn)
+                 else ops.gather_rows(v, idx.to(v.device)))
+             for k, v in self.columns.items()},
+            {k: m[idx.to(m.device)] for k, m in self.masks.items()})
 
     def slice(self, start: int, end: int) -> "ColumnBatch":
-        return ColumnBatch({
-            k: (StringColumn(v.codes[start:end], v.values)
-                if isinstance(v, StringColumn) else v[start:end])
-            for k, v in self.columns.items()})
+        return ColumnBatch(
+            {k: (StringColumn(v.codes[start:end], v.values)
+                 if isinstance(v, StringColumn) else v[start:end])
+             for k, v in self.columns.items()},
+            {k: m[start:end] for k, m in self.masks.items()})
 
     def to(self, device) -> "ColumnBatch":
-        return ColumnBatch({
-            k: v.to(device) for k, v in self.columns.items()})
+        return ColumnBatch(
+            {k: v.to(device) for k, v in self.columns.items()},
+            {k: m.to(device) for k, m in self.masks.items()})
 
     @staticmethod
     def concat(batches: "List[ColumnBatch]") -> "ColumnBatch":
@@ -177,6 +221,7 @@ class ColumnBatch:
             return ColumnBatch({})
         names = batches[0].names
         out: Dict[str, Column] = {}
+        masks: Dict[str, torch.Tensor] = {}
         for n in names:
             cols = [b.column(n) for b in batches]
             if isinstance(cols[0], StringColumn):
@@ -192,16 +237,36 @@ class ColumnBatch:
                 out[n] = StringColumn(torch.cat(remap_codes), merged)
             else:
                 out[n] = torch.cat(cols)
-        return ColumnBatch(out)
+            if any(b.mask(n) is not None for b in batches):
+                parts = []
+                for b in batches:
+                    m = b.mask(n)
+                    if m is None:
+                        m = torch.ones(b.num_rows, dtype=torch.bool,
+                                       device=b.device)
+                    parts.append(m)
+                masks[n] = torch.cat(parts)
+        return ColumnBatch(out, masks)
 
     # -- conversion -------------------------------------------------------
     @staticmethod
     def from_arrow(table) -> "ColumnBatch":
-        """pyarrow.Table -> ColumnBatch (host tensors)."""
+        """pyarrow.Table -> ColumnBatch (host tensors).  Null slots become
+        validity-mask entries; the value buffer holds a fill (0 / "")."""
         import pyarrow as pa
         cols: Dict[str, Column] = {}
+        masks: Dict[str, torch.Tensor] = {}
         for name, col in zip(table.column_names, table.columns):
             col = col.combine_chunks()
+            if col.null_count:
+                masks[name] = torch.from_numpy(
+                    np.ascontiguousarray(col.is_valid().to_numpy(
+                        zero_copy_only=False)))
+                if pa.types.is_string(col.type) or \
+                        pa.types.is_large_string(col.type):
+                    col = col.fill_null("")
+                else:
+                    col = col.fill_null(0)
             if pa.types.is_string(col.type) or pa.types.is_large_string(
                     col.type):
                 cols[name] = StringColumn.from_strings(col.to_pylist())
@@ -214,16 +279,18 @@ class ColumnBatch:
                 if not np_arr.flags.writeable:
                     np_arr = np_arr.copy()
                 cols[name] = torch.from_numpy(np_arr)
-        return ColumnBatch(cols)
+        return ColumnBatch(cols, masks)
 
     def to_arrow(self):
         import pyarrow as pa
         arrays = {}
         for k, v in self.columns.items():
+            m = self.mask(k)
+            null_mask = (~m.cpu()).numpy() if m is not None else None
             if isinstance(v, StringColumn):
-                arrays[k] = pa.array(v.to_numpy())
+                arrays[k] = pa.array(v.to_numpy(), mask=null_mask)
             else:
-                arrays[k] = pa.array(v.cpu().numpy())
+                arrays[k] = pa.array(v.cpu().numpy(), mask=null_mask)
         return pa.table(arrays)
 
     def to_numpy(self) -> Dict[str, np.ndarray]:

@@ -23,8 +23,8 @@ from .columnar import ColumnBatch, StringColumn
 from .. import ops
 from ..config import IndexConstants
 from ..exceptions import HyperspaceException
-from ..plan.expr import (And, BinComp, Col, Expr, In, IsNotNull, Lit, Not,
-                         Or, extract_equi_join_keys)
+from ..plan.expr import (And, BinComp, Col, Expr, In, IsNotNull, IsNull,
+                         Lit, Not, Or, extract_equi_join_keys)
 from ..plan.nodes import (BucketUnionNode, Filter, IndexScan, Join,
                           LogicalPlan, Project, Scan, UnionNode)
 from ..sources.parquet_io import (read_files_batch,
@@ -374,11 +374,26 @@ class Executor:
         """Native fast path for a single numeric comparison (select_range
         kernel); general path composes boolean masks."""
         rng = _as_range(cond, batch)
-        if rng is not None:
+        if rng is not None and not any(
+                batch.has_column(c) and batch.has_nulls(c)
+                for c in cond.references()):
             keys_u64, lo, hi, lo_incl, hi_incl = rng
             return ops.select_range_u64(keys_u64, lo, hi, lo_incl, hi_incl)
         mask = self._eval_mask(batch, cond)
         return torch.nonzero(mask, as_tuple=False).flatten()
+
+    def _valid_of(self, batch: ColumnBatch, e: Expr
+                  ) -> Optional[torch.Tensor]:
+        """Conjunction of validity masks over e's referenced columns, or
+        None when none of them are nullable."""
+        out = None
+        for c in e.references():
+            if not batch.has_column(c):
+                continue
+            m = batch.mask(c)
+            if m is not None:
+                out = m if out is None else out & m
+        return out
 
     def _eval_mask(self, batch: ColumnBatch, e: Expr) -> torch.Tensor:
         if isinstance(e, And):
@@ -388,11 +403,25 @@ class Executor:
             return self._eval_mask(batch, e.left) | \
                 self._eval_mask(batch, e.right)
         if isinstance(e, Not):
-            return ~self._eval_mask(batch, e.child)
+            # SQL three-valued logic: NOT(null-involving predicate) is
+            # null -> the row is excluded, so AND the child's validity
+            m = ~self._eval_mask(batch, e.child)
+            v = self._valid_of(batch, e.child)
+            return m & v if v is not None else m
         if isinstance(e, IsNotNull):
-            # nulls unsupported in v0: all rows pass
+            m = batch.mask(e.col.name) if batch.has_column(e.col.name) \
+                else None
+            if m is not None:
+                return m.clone()
             return torch.ones(batch.num_rows, dtype=torch.bool,
                               device=batch.device)
+        if isinstance(e, IsNull):
+            m = batch.mask(e.col.name) if batch.has_column(e.col.name) \
+                else None
+            if m is not None:
+                return ~m
+            return torch.zeros(batch.num_rows, dtype=torch.bool,
+                               device=batch.device)
         if isinstance(e, In):
             col = batch.column(e.col.name)
             vals = e.values
@@ -401,13 +430,18 @@ class Executor:
                     sorted(c for c in (col.code_of(v) for v in vals)
                            if c >= 0),
                     dtype=torch.int64, device=col.codes.device)
-                return ops.isin_sorted(col.codes.to(torch.int64), codes)
-            t = col.to(torch.int64)
-            vs = torch.tensor(sorted(int(v) for v in vals),
-                              dtype=torch.int64, device=t.device)
-            return ops.isin_sorted(t, vs)
+                m = ops.isin_sorted(col.codes.to(torch.int64), codes)
+            else:
+                t = col.to(torch.int64)
+                vs = torch.tensor(sorted(int(v) for v in vals),
+                                  dtype=torch.int64, device=t.device)
+                m = ops.isin_sorted(t, vs)
+            v = self._valid_of(batch, e)
+            return m & v if v is not None else m
         if isinstance(e, BinComp):
-            return _compare(batch, e)
+            m = _compare(batch, e)
+            v = self._valid_of(batch, e)
+            return m & v if v is not None else m
         raise HyperspaceException(f"Cannot evaluate {e!r}")
 
     # ------------------------------------------------------------------
@@ -424,6 +458,12 @@ class Executor:
 
         lbatch, lseg = self._exec(plan.left)
         rbatch, rseg = self._exec(plan.right)
+
+        # inner-join SQL semantics: null join keys never match — drop
+        # them up front (bucketed layouts keep NULLS FIRST per bucket,
+        # so the segment offsets shift by the per-prefix null count)
+        lbatch, lseg = _drop_null_keys(lbatch, lkeys_names[0], lseg)
+        rbatch, rseg = _drop_null_keys(rbatch, rkeys_names[0], rseg)
 
         if (left_bucketed and right_bucketed and lseg is not None
                 and rseg is not None and lseg.numel() == rseg.numel()):
@@ -452,7 +492,8 @@ class Executor:
             one_seg_r = torch.tensor([0, rk.numel()], dtype=torch.int64)
             lidx, ridx = ops.merge_join(lk, rk, one_seg, one_seg_r)
 
-        # secondary key equality check for multi-key joins
+        # secondary key equality check for multi-key joins (null
+        # secondary keys never match)
         if len(pairs) > 1 and lidx.numel():
             keep = torch.ones(lidx.numel(), dtype=torch.bool,
                               device=lidx.device)
@@ -460,17 +501,28 @@ class Executor:
                 lv = lbatch.tensor(ln)[lidx]
                 rv = rbatch.tensor(rn)[ridx]
                 keep &= (lv == rv)
+                lm, rm = lbatch.mask(ln), rbatch.mask(rn)
+                if lm is not None:
+                    keep &= lm[lidx]
+                if rm is not None:
+                    keep &= rm[ridx]
             sel = torch.nonzero(keep, as_tuple=False).flatten()
             lidx, ridx = lidx[sel], ridx[sel]
 
         lout = lbatch.gather(lidx)
         rout = rbatch.gather(ridx)
         cols: Dict[str, object] = {}
+        masks: Dict[str, torch.Tensor] = {}
         for k, v in lout.columns.items():
             cols[k] = v
+            if lout.mask(k) is not None:
+                masks[k] = lout.mask(k)
         for k, v in rout.columns.items():
-            cols[k if k not in cols else f"{k}_r"] = v
-        return ColumnBatch(cols)
+            name = k if k not in cols else f"{k}_r"
+            cols[name] = v
+            if rout.mask(k) is not None:
+                masks[name] = rout.mask(k)
+        return ColumnBatch(cols, masks)
 
     # ------------------------------------------------------------------
     def _exec_bucket_union(self, plan: BucketUnionNode
@@ -521,7 +573,10 @@ class Executor:
         from ..index.covering.index import sort_by_bucket_and_keys
         from ..ops.string_hash import bucket_hash_keys
         keys = bucket_hash_keys(batch, bucket_cols)
-        bucket_ids = ops.murmur3_bucket(keys, num_buckets)
+        key_masks = [batch.mask(c) for c in bucket_cols]
+        bucket_ids = ops.murmur3_bucket(
+            keys, num_buckets,
+            key_masks if any(m is not None for m in key_masks) else None)
         return sort_by_bucket_and_keys(batch, bucket_ids, bucket_cols,
                                        num_buckets)
 
@@ -532,6 +587,27 @@ class Executor:
 
 _SPARK_DTYPES = {"long": torch.int64, "integer": torch.int32,
                  "double": torch.float64, "float": torch.float32}
+
+
+def _drop_null_keys(batch: ColumnBatch, key_name: str,
+                    seg: Optional[torch.Tensor]
+                    ) -> Tuple[ColumnBatch, Optional[torch.Tensor]]:
+    """Remove rows whose join key is null (inner-join semantics).
+
+    When per-bucket segment offsets accompany the batch, they stay valid
+    because every bucketed layout in the engine sorts NULLS FIRST within
+    each bucket: each offset shifts down by the number of nulls before
+    it."""
+    if not batch.has_nulls(key_name):
+        return batch, seg
+    m = batch.mask(key_name)
+    keep = torch.nonzero(m, as_tuple=False).flatten()
+    if seg is not None:
+        nullcum = torch.cat([
+            torch.zeros(1, dtype=torch.int64, device=m.device),
+            torch.cumsum((~m).to(torch.int64), 0)])
+        seg = (seg.to(m.device) - nullcum[seg.to(m.device)]).cpu()
+    return batch.gather(keep), seg
 
 
 def _empty_batch(schema) -> ColumnBatch:

@@ -140,7 +140,10 @@ class CoveringIndex(Index):
         n = self.num_buckets
         from ...ops.string_hash import bucket_hash_keys
         keys = bucket_hash_keys(batch, self.indexed_columns)
-        bucket_ids = ops.murmur3_bucket(keys, n)
+        key_masks = [batch.mask(c) for c in self.indexed_columns]
+        bucket_ids = ops.murmur3_bucket(
+            keys, n,
+            key_masks if any(m is not None for m in key_masks) else None)
         if distributed:
             from ...parallel.exchange import exchange_by_bucket
             batch, bucket_ids = exchange_by_bucket(batch, bucket_ids, n)
@@ -246,13 +249,22 @@ class CoveringIndex(Index):
 def _multi_key_sort_perm(batch: ColumnBatch, key_cols: List[str]
                          ) -> torch.Tensor:
     """Stable multi-column sort permutation: LSD over columns
-    (least-significant column first)."""
+    (least-significant column first).  Nullable columns order ASC NULLS
+    FIRST (Spark's default): a stable partition moves this column's null
+    rows ahead after its key sort, giving (null-flag, key) lexicographic
+    order per column."""
     n = batch.num_rows
     dev = batch.device
     perm = torch.arange(n, dtype=torch.int64, device=dev)
     for c in reversed(key_cols):
         keys = ops.normalize_key(batch.tensor(c))[perm]
         _, perm = ops.sort_pairs(keys, perm)
+        m = batch.mask(c)
+        if m is not None and not bool(m.all()):
+            f = m[perm]
+            perm = perm[torch.cat([
+                torch.nonzero(~f, as_tuple=False).flatten(),
+                torch.nonzero(f, as_tuple=False).flatten()])]
     return perm
 
 

@@ -80,7 +80,8 @@ class DataSkippingIndexConfig(IndexConfigTrait):
             sketch.expr = resolved
             values = batch.tensor(resolved)
             dtype_name = schema.field_type(resolved) or "long"
-            index_data.update(sketch.aggregate(values, seg, dtype_name))
+            index_data.update(sketch.aggregate(values, seg, dtype_name,
+                                               batch.mask(resolved)))
 
         index_schema = Schema([f for f in schema.fields
                                if f.name.lower() in

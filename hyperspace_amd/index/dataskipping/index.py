@@ -117,7 +117,8 @@ class DataSkippingIndex(Index):
             for sketch in self.sketches:
                 values = batch.tensor(sketch.expr)
                 dtype_name = self.schema.field_type(sketch.expr) or "long"
-                data.update(sketch.aggregate(values, seg, dtype_name))
+                data.update(sketch.aggregate(values, seg, dtype_name,
+                                             batch.mask(sketch.expr)))
             out = os.path.join(
                 ctx.index_data_path,
                 f"part-00001-{_uuid.uuid4().hex[:8]}_00000.c000.parquet")

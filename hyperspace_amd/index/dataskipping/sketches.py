@@ -45,9 +45,13 @@ class Sketch:
         raise NotImplementedError
 
     def aggregate(self, values: torch.Tensor, seg_off: torch.Tensor,
-                  dtype_name: str) -> Dict[str, torch.Tensor]:
+                  dtype_name: str,
+                  mask: Optional[torch.Tensor] = None
+                  ) -> Dict[str, torch.Tensor]:
         """Per-file aggregates; ``values`` is the source column (device ok),
-        ``seg_off`` the per-file row offsets."""
+        ``seg_off`` the per-file row offsets, ``mask`` the optional
+        validity mask (nulls are ignored, as in Spark's min/max/first
+        aggregates)."""
         raise NotImplementedError
 
     def convert_predicate(self, pred: Expr, sketch_data,
@@ -86,10 +90,21 @@ class MinMaxSketch(Sketch):
     def out_columns(self):
         return [f"MinMax_{self.expr}__min", f"MinMax_{self.expr}__max"]
 
-    def aggregate(self, values, seg_off, dtype_name):
+    def aggregate(self, values, seg_off, dtype_name, mask=None):
         norm = ops.normalize_key(values)
-        mins, maxs = ops.segmented_minmax(norm, seg_off.to(norm.device)
-                                          if norm.is_cuda else seg_off)
+        seg = seg_off.to(norm.device) if norm.is_cuda else seg_off
+        if mask is not None and not bool(mask.all()):
+            # normalized keys compare as signed int64, so the
+            # null-neutral sentinels are the signed extremes
+            m = mask.to(norm.device)
+            mins, _ = ops.segmented_minmax(
+                torch.where(m, norm,
+                            torch.full_like(norm, 2**63 - 1)), seg)
+            _, maxs = ops.segmented_minmax(
+                torch.where(m, norm,
+                            torch.full_like(norm, -2**63)), seg)
+        else:
+            mins, maxs = ops.segmented_minmax(norm, seg)
         return {self.out_columns()[0]: mins.cpu(),
                 self.out_columns()[1]: maxs.cpu()}
 
@@ -155,7 +170,7 @@ class BloomFilterSketch(Sketch):
     def out_columns(self):
         return [f"BloomFilter_{self.expr}__bf"]
 
-    def aggregate(self, values, seg_off, dtype_name):
+    def aggregate(self, values, seg_off, dtype_name, mask=None):
         vals = values
         if vals.dtype == torch.float64:
             vals = vals.view(torch.int64)
@@ -164,10 +179,15 @@ class BloomFilterSketch(Sketch):
             vals = vals.to(torch.float64).view(torch.int64)
         elif vals.dtype != torch.int64:
             vals = vals.to(torch.int64)
+        if mask is not None:
+            mask = mask.to(vals.device)
         words_per_file = []
         for s in range(seg_off.numel() - 1):
             a, b = int(seg_off[s]), int(seg_off[s + 1])
-            words = ops.bloom_build(vals[a:b], self.m_bits, self.k)
+            part = vals[a:b]
+            if mask is not None:
+                part = part[mask[a:b]]  # nulls never enter the filter
+            words = ops.bloom_build(part, self.m_bits, self.k)
             words_per_file.append(words.cpu())
         return {self.out_columns()[0]: torch.stack(words_per_file)
                 if words_per_file
@@ -230,11 +250,16 @@ class PartitionSketch(Sketch):
     def out_columns(self):
         return [f"Partition_{self.expr}__first"]
 
-    def aggregate(self, values, seg_off, dtype_name):
+    def aggregate(self, values, seg_off, dtype_name, mask=None):
         firsts = []
         vals = values.cpu()
+        mk = mask.cpu() if mask is not None else None
         for s in range(seg_off.numel() - 1):
             a, b = int(seg_off[s]), int(seg_off[s + 1])
+            if mk is not None and b > a:
+                valid_pos = torch.nonzero(mk[a:b],
+                                          as_tuple=False).flatten()
+                a = a + int(valid_pos[0]) if valid_pos.numel() else b
             firsts.append(vals[a] if b > a else torch.tensor(
                 0, dtype=vals.dtype))
         return {self.out_columns()[0]: torch.stack(firsts) if firsts

@@ -49,11 +49,21 @@ def _is_cuda(*tensors: torch.Tensor) -> bool:
     return any(t.is_cuda for t in tensors)
 
 
-def murmur3_bucket(keys: List[torch.Tensor], num_buckets: int
-                   ) -> torch.Tensor:
+def murmur3_bucket(keys: List[torch.Tensor], num_buckets: int,
+                   masks=None) -> torch.Tensor:
+    """masks: optional list parallel to keys; entries are bool validity
+    tensors or None (no nulls in that key column)."""
     if _is_cuda(*keys):
+        if masks is not None and any(m is not None for m in masks):
+            mlist = [
+                (m.contiguous() if m is not None
+                 else torch.empty(0, dtype=torch.bool,
+                                  device=keys[0].device))
+                for m in masks]
+            return native.ext().murmur3_bucket(list(keys), num_buckets,
+                                               mlist)
         return native.ext().murmur3_bucket(list(keys), num_buckets)
-    return cpu_ref.murmur3_bucket(keys, num_buckets)
+    return cpu_ref.murmur3_bucket(keys, num_buckets, masks)
 
 
 def sort_pairs(keys_u64: torch.Tensor, payload: torch.Tensor

@@ -69,20 +69,31 @@ def murmur3_hash_int64(x: torch.Tensor, seed) -> torch.Tensor:
     return _fmix(h1, 8)
 
 
-def murmur3_bucket(keys: List[torch.Tensor], num_buckets: int) -> torch.Tensor:
+def murmur3_bucket(keys: List[torch.Tensor], num_buckets: int,
+                   masks: Optional[List[Optional[torch.Tensor]]] = None
+                   ) -> torch.Tensor:
     """Bucket id per row: Spark HashPartitioning semantics —
-    h = seed 42 folded over columns, bucket = pmod(h, n).  Returns int32."""
+    h = seed 42 folded over columns, bucket = pmod(h, n).  A null key
+    column leaves the running hash unchanged (Murmur3Hash skips null
+    children, so an all-null single-key row lands in pmod(42, n)).
+    ``masks[i]`` is a bool validity tensor or None.  Returns int32."""
     h: Optional[torch.Tensor] = None
-    for k in keys:
+    for ci, k in enumerate(keys):
         seed = SPARK_HASH_SEED if h is None else h
         if k.dtype in (torch.int64, torch.float64):
             if k.dtype == torch.float64:
                 k = k.view(torch.int64)
-            h = murmur3_hash_int64(k, seed)
+            nh = murmur3_hash_int64(k, seed)
         else:
             if k.dtype == torch.float32:
                 k = k.view(torch.int32)
-            h = murmur3_hash_int32(k, seed)
+            nh = murmur3_hash_int32(k, seed)
+        mask = masks[ci] if masks is not None else None
+        if mask is not None and mask.numel():
+            seed_t = (torch.full_like(nh, SPARK_HASH_SEED)
+                      if h is None else h)
+            nh = torch.where(mask.to(torch.bool), nh, seed_t)
+        h = nh
     assert h is not None
     signed = h.to(torch.int64)
     signed = torch.where(signed > 0x7FFFFFFF, signed - (1 << 32), signed)

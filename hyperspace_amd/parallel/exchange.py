@@ -66,11 +66,16 @@ def exchange_by_bucket(batch: ColumnBatch, bucket_ids: torch.Tensor,
         return out
 
     out_cols = {}
+    out_masks = {}
     for name, col in batch.columns.items():
         if isinstance(col, StringColumn):
             raise HyperspaceException(
                 "string columns not yet supported in the distributed "
                 "bucket exchange; dictionary-merge exchange is planned")
         out_cols[name] = a2a(col)
+        m = batch.mask(name)
+        if m is not None:
+            # validity masks ride along as uint8 splits on the same links
+            out_masks[name] = a2a(m.to(torch.uint8)).to(torch.bool)
     out_buckets = a2a(bucket_sorted)
-    return ColumnBatch(out_cols), out_buckets.to(torch.int32)
+    return ColumnBatch(out_cols, out_masks), out_buckets.to(torch.int32)

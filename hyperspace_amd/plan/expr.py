@@ -58,6 +58,9 @@ class Col(Expr):
     def is_not_null(self):
         return IsNotNull(self)
 
+    def is_null(self):
+        return IsNull(self)
+
     def __hash__(self):
         return hash(("Col", self.name.lower()))
 
@@ -149,6 +152,17 @@ class IsNotNull(Expr):
 
     def __repr__(self):
         return f"({self.col!r} IS NOT NULL)"
+
+
+class IsNull(Expr):
+    def __init__(self, col: Expr):
+        self.col = col
+
+    def references(self):
+        return self.col.references()
+
+    def __repr__(self):
+        return f"({self.col!r} IS NULL)"
 
 
 def col(name: str) -> Col:

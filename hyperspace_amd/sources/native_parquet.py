@@ -40,9 +40,11 @@ _NP_TO_PARQUET = {
 _PARQUET_TO_NP = {v: k for k, v in _NP_TO_PARQUET.items()}
 
 ENC_PLAIN = 0
+ENC_RLE = 3
 CODEC_UNCOMPRESSED = 0
 PAGE_DATA = 0
 REP_REQUIRED = 0
+REP_OPTIONAL = 1
 
 
 # ---------------------------------------------------------------------------
@@ -201,7 +203,8 @@ class TReader:
 # Writer
 # ---------------------------------------------------------------------------
 
-def _page_header(num_values: int, nbytes: int) -> bytes:
+def _page_header(num_values: int, nbytes: int,
+                 has_def_levels: bool = False) -> bytes:
     w = TWriter()
     w.struct_begin()
     w.field_i32(1, PAGE_DATA)
@@ -210,7 +213,8 @@ def _page_header(num_values: int, nbytes: int) -> bytes:
     w.field_struct_begin(5)  # DataPageHeader
     w.field_i32(1, num_values)
     w.field_i32(2, ENC_PLAIN)
-    w.field_i32(3, ENC_PLAIN)  # definition_level_encoding (unused: REQUIRED)
+    # definition_level_encoding: RLE for OPTIONAL columns
+    w.field_i32(3, ENC_RLE if has_def_levels else ENC_PLAIN)
     w.field_i32(4, ENC_PLAIN)  # repetition_level_encoding
     w.struct_end()
     w.struct_end()
@@ -222,27 +226,52 @@ def _statistics(arr: np.ndarray) -> Tuple[bytes, bytes]:
     return arr.min().tobytes(), arr.max().tobytes()
 
 
-def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str
+def _def_levels_payload(mask: np.ndarray) -> bytes:
+    """4-byte-length-prefixed RLE-hybrid encoding of max_def=1 levels:
+    one bit-packed run (groups of 8, LSB-first) covering all rows."""
+    groups = (len(mask) + 7) // 8
+    packed = np.packbits(mask.astype(np.uint8), bitorder="little")
+    body = _varint((groups << 1) | 1) + packed.tobytes()
+    return struct.pack("<I", len(body)) + body
+
+
+def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str,
+                         masks: "Optional[Dict[str, np.ndarray]]" = None
                          ) -> Tuple[int, int]:
     """Write numeric columns as one row group, one PLAIN page per column.
+    ``masks``: optional name -> bool validity array; masked columns are
+    written OPTIONAL with RLE def-levels and compacted values.
     Returns (size, mtime_ms)."""
     names = list(columns.keys())
     arrays = [np.ascontiguousarray(columns[n]) for n in names]
     num_rows = len(arrays[0]) if arrays else 0
-    for a in arrays:
+    col_masks: List[Optional[np.ndarray]] = []
+    for n, a in zip(names, arrays):
         if a.dtype not in _NP_TO_PARQUET:
             raise ValueError(f"dtype {a.dtype} not supported natively")
         assert len(a) == num_rows
+        m = (masks or {}).get(n)
+        if m is not None and bool(m.all()):
+            m = None  # all-valid mask: write the cheaper REQUIRED form
+        col_masks.append(m)
 
     chunks: List[bytes] = []
-    col_meta: List[Tuple[str, int, int, int, int, bytes, bytes]] = []
+    col_meta: List[Tuple] = []
     offset = 4  # after magic
-    for name, arr in zip(names, arrays):
-        payload = arr.tobytes()  # PLAIN little-endian
-        header = _page_header(num_rows, len(payload))
-        mn, mx = _statistics(arr) if num_rows else (b"", b"")
+    for name, arr, mask in zip(names, arrays, col_masks):
+        if mask is None:
+            payload = arr.tobytes()  # PLAIN little-endian
+            null_count = 0
+            valid = arr
+        else:
+            valid = arr[mask]
+            payload = _def_levels_payload(mask) + valid.tobytes()
+            null_count = int(num_rows - len(valid))
+        header = _page_header(num_rows, len(payload), mask is not None)
+        mn, mx = _statistics(valid) if len(valid) else (b"", b"")
         col_meta.append((name, _NP_TO_PARQUET[arr.dtype], offset,
-                         len(header) + len(payload), num_rows, mn, mx))
+                         len(header) + len(payload), num_rows, mn, mx,
+                         null_count, mask is not None))
         chunks.append(header)
         chunks.append(payload)
         offset += len(header) + len(payload)
@@ -257,10 +286,11 @@ def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str
     w.field_string(4, "schema")
     w.field_i32(5, len(names))
     w.struct_end()
-    for name, ptype, *_ in col_meta:
+    for name, ptype, *rest in col_meta:
+        optional = rest[-1]
         w.struct_elem_begin()
         w.field_i32(1, ptype)
-        w.field_i32(3, REP_REQUIRED)
+        w.field_i32(3, REP_OPTIONAL if optional else REP_REQUIRED)
         w.field_string(4, name)
         w.struct_end()
     w.field_i64(3, num_rows)
@@ -269,13 +299,19 @@ def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str
     w.struct_elem_begin()
     total_bytes = sum(m[3] for m in col_meta)
     w.field_list_begin(1, CT_STRUCT, len(col_meta))
-    for name, ptype, off, nbytes, nvals, mn, mx in col_meta:
+    for name, ptype, off, nbytes, nvals, mn, mx, null_count, optional \
+            in col_meta:
         w.struct_elem_begin()  # ColumnChunk
         w.field_i64(2, off)  # file_offset
         w.field_struct_begin(3)  # ColumnMetaData
         w.field_i32(1, ptype)
-        w.field_list_begin(2, CT_I32, 1)
-        w.i32_elem(ENC_PLAIN)
+        if optional:
+            w.field_list_begin(2, CT_I32, 2)
+            w.i32_elem(ENC_PLAIN)
+            w.i32_elem(ENC_RLE)
+        else:
+            w.field_list_begin(2, CT_I32, 1)
+            w.i32_elem(ENC_PLAIN)
         w.field_list_begin(3, CT_BINARY, 1)
         w.buf += _varint(len(name.encode()))
         w.buf += name.encode()
@@ -284,13 +320,15 @@ def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str
         w.field_i64(6, nbytes)
         w.field_i64(7, nbytes)
         w.field_i64(9, off)  # data_page_offset
-        if mn:
+        if mn or null_count:
             w.field_struct_begin(12)  # Statistics
-            w.field_binary(1, mx)  # max (legacy)
-            w.field_binary(2, mn)  # min (legacy)
-            w.field_i64(3, 0)      # null_count
-            w.field_binary(5, mx)  # max_value
-            w.field_binary(6, mn)  # min_value
+            if mn:
+                w.field_binary(1, mx)  # max (legacy)
+                w.field_binary(2, mn)  # min (legacy)
+            w.field_i64(3, null_count)
+            if mn:
+                w.field_binary(5, mx)  # max_value
+                w.field_binary(6, mn)  # min_value
             w.struct_end()
         w.struct_end()
         w.struct_end()
@@ -319,22 +357,70 @@ def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str
 class ColumnChunkLayout:
     """Per-column page layout.
 
-    encoding "plain": pages = [(values byte offset, num values), ...]
-    encoding "dict":  pages = [(payload start, payload end, num values,
-                      bit width), ...] of RLE/bit-packed index payloads,
-                      plus dict_page = (values byte offset, dict size)
+    encoding "plain": pages = [("plain", values byte offset, num values)]
+    encoding "dict":  pages = [("dict", payload start, payload end,
+                      num values, bit width), ...] of RLE/bit-packed index
+                      payloads, plus dict_page = (values offset, dict size)
+
+    ``page_masks`` parallels ``pages``: a bool validity array per page for
+    OPTIONAL columns with nulls in that page, else None.  A page's stored
+    values are compacted (non-null only); num values counts rows.
     """
     __slots__ = ("name", "np_dtype", "pages", "num_values", "encoding",
-                 "dict_page")
+                 "dict_page", "page_masks")
 
     def __init__(self, name, np_dtype, pages, num_values,
-                 encoding="plain", dict_page=None):
+                 encoding="plain", dict_page=None, page_masks=None):
         self.name = name
         self.np_dtype = np_dtype
         self.pages = pages
         self.num_values = num_values
         self.encoding = encoding
         self.dict_page = dict_page
+        self.page_masks = page_masks or [None] * len(pages)
+
+    @property
+    def has_nulls(self) -> bool:
+        return any(m is not None for m in self.page_masks)
+
+
+def _decode_defs(data, off: int, length: int, n: int
+                 ) -> Optional[np.ndarray]:
+    """Decode RLE-hybrid definition levels (max_def=1) into a bool
+    validity array, or None when every row is valid."""
+    end = off + length
+    out = np.empty(n, dtype=np.uint8)
+    pos = off
+    filled = 0
+    while filled < n and pos < end:
+        header = 0
+        shift = 0
+        while True:
+            b = data[pos]
+            pos += 1
+            header |= (b & 0x7F) << shift
+            if not b & 0x80:
+                break
+            shift += 7
+        if header & 1:  # bit-packed run: (header>>1) groups of 8 levels
+            groups = header >> 1
+            cnt = min(groups * 8, n - filled)
+            packed = np.frombuffer(data, dtype=np.uint8, count=groups,
+                                   offset=pos)
+            out[filled:filled + cnt] = np.unpackbits(
+                packed, bitorder="little")[:cnt]
+            pos += groups
+            filled += cnt
+        else:  # RLE run: one value byte (bit_width 1 -> 1 byte)
+            run = min(header >> 1, n - filled)
+            v = data[pos]
+            pos += 1
+            out[filled:filled + run] = v
+            filled += run
+    if filled < n:
+        out[filled:] = 1
+    mask = out.astype(bool)
+    return None if bool(mask.all()) else mask
 
 
 def read_native_layout(path: str,
@@ -391,13 +477,12 @@ def _walk_row_group(rg, pf_schema, data, want
         if ptype is None:
             return None
         np_dtype = _PARQUET_TO_NP[ptype]
-        # nulls unsupported: skip the def-level prefix only when levels
-        # exist; reject columns that actually contain nulls
+        # OPTIONAL columns carry a def-level prefix per page; when the
+        # chunk statistics prove null_count == 0 the levels are skipped
+        # without decoding, else each page's levels become a validity mask
         has_levels = pf_schema.column(col_index).max_definition_level > 0
         st = col.statistics
-        if has_levels and (st is None or st.null_count not in (0, None)):
-            if st is None or st.null_count != 0:
-                return None
+        chunk_all_valid = st is not None and st.null_count == 0
 
         dict_page = None
         pos = col.data_page_offset
@@ -422,6 +507,7 @@ def _walk_row_group(rg, pf_schema, data, want
 
         seen = 0
         pages: List[Tuple] = []
+        page_masks: List[Optional[np.ndarray]] = []
         while seen < col.num_values:
             r = TReader(data, pos)
             try:
@@ -440,9 +526,14 @@ def _walk_row_group(rg, pf_schema, data, want
                 return None
             values_off = r.pos
             page_end = r.pos + page_bytes
+            mask = None
             if has_levels:
                 lvl_len = struct.unpack_from("<I", data, values_off)[0]
+                if not chunk_all_valid:
+                    mask = _decode_defs(data, values_off + 4, lvl_len,
+                                        num_values)
                 values_off += 4 + lvl_len
+            n_valid = int(mask.sum()) if mask is not None else num_values
             page_enc = dph.get(2)
             if is_dict and page_enc in (2, 8):  # PLAIN_DICT / RLE_DICT
                 bit_width = data[values_off]
@@ -451,12 +542,13 @@ def _walk_row_group(rg, pf_schema, data, want
             elif page_enc == ENC_PLAIN:
                 # also reached as the writer's mid-chunk fallback when a
                 # dictionary overflows: later pages switch to PLAIN
-                expected = num_values * np_dtype.itemsize
+                expected = n_valid * np_dtype.itemsize
                 if values_off + expected > page_end:
                     return None
                 pages.append(("plain", values_off, num_values))
             else:
                 return None
+            page_masks.append(mask)
             seen += num_values
             pos = page_end
         if seen != col.num_values:
@@ -464,13 +556,17 @@ def _walk_row_group(rg, pf_schema, data, want
         enc_kind = ("dict" if any(pg[0] == "dict" for pg in pages)
                     else "plain")
         out.append(ColumnChunkLayout(name, np_dtype, pages,
-                                     col.num_values, enc_kind, dict_page))
+                                     col.num_values, enc_kind, dict_page,
+                                     page_masks))
     return out
 
 
 def read_native_host(path: str, columns: Optional[List[str]] = None
-                     ) -> Optional[Dict[str, np.ndarray]]:
-    """Host-side decode of a native-layout file (testing / CPU path)."""
+                     ) -> Optional[Tuple[Dict[str, np.ndarray],
+                                         Dict[str, np.ndarray]]]:
+    """Host-side decode of a native-layout file (testing / CPU path).
+    Returns (columns, validity masks); masks holds entries only for
+    columns containing nulls (null slots in the value array are 0)."""
     layout = read_native_layout(path, columns)
     if layout is None:
         return None
@@ -478,10 +574,28 @@ def read_native_host(path: str, columns: Optional[List[str]] = None
     if any(c.encoding != "plain" for c in chunks):
         return None  # dictionary decode is the device path; host->pyarrow
     acc: Dict[str, list] = {}
+    macc: Dict[str, list] = {}
+    any_null: Dict[str, bool] = {}
     for c in chunks:  # row-group-major order
-        parts = [np.frombuffer(data, dtype=c.np_dtype, count=nv,
-                               offset=off) for _, off, nv in c.pages]
-        acc.setdefault(c.name, []).extend(parts)
-    return {name: (np.concatenate(parts) if len(parts) > 1
-                   else parts[0].copy())
+        for (_, off, nv), mask in zip(c.pages, c.page_masks):
+            if mask is None:
+                part = np.frombuffer(data, dtype=c.np_dtype, count=nv,
+                                     offset=off)
+                pm = np.ones(nv, dtype=bool)
+            else:
+                vals = np.frombuffer(data, dtype=c.np_dtype,
+                                     count=int(mask.sum()), offset=off)
+                part = np.zeros(nv, dtype=c.np_dtype)
+                part[mask] = vals
+                pm = mask
+                any_null[c.name] = True
+            acc.setdefault(c.name, []).append(part)
+            macc.setdefault(c.name, []).append(pm)
+    cols = {name: (np.concatenate(parts) if len(parts) > 1
+                   else (parts[0] if parts[0].flags.writeable
+                         else parts[0].copy()))
             for name, parts in acc.items()}
+    masks = {name: np.concatenate(macc[name]) if len(macc[name]) > 1
+             else macc[name][0]
+             for name in acc if any_null.get(name)}
+    return cols, masks

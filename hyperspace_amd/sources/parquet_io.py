@@ -47,7 +47,9 @@ def write_batch_parquet(batch, path: str, compression: Optional[str] = None,
             isinstance(c, StringColumn) for c in batch.columns.values()):
         from .native_parquet import write_parquet_native
         cols = {name: t.numpy() for name, t in batch.columns.items()}
-        return write_parquet_native(cols, path)
+        masks = {name: m.numpy() for name, m in batch.masks.items()} \
+            if batch.masks else None
+        return write_parquet_native(cols, path, masks)
     import pyarrow.parquet as pq
     table = batch.to_arrow()
     pq.write_table(table, path, compression=compression or "NONE",
@@ -67,14 +69,17 @@ def read_files_batch(paths: List[str], columns: Optional[List[str]] = None):
     from .native_parquet import read_native_host
 
     per_file = []
+    per_file_masks = []
     row_counts = []
     native_ok = True
     for p in paths:
-        cols = read_native_host(p, columns)
-        if cols is None:
+        res = read_native_host(p, columns)
+        if res is None:
             native_ok = False
             break
+        cols, fmasks = res
         per_file.append(cols)
+        per_file_masks.append(fmasks)
         n = len(next(iter(cols.values()))) if cols else 0
         row_counts.append(n)
     if native_ok and paths:
@@ -83,11 +88,18 @@ def read_files_batch(paths: List[str], columns: Optional[List[str]] = None):
             order = {c.lower(): i for i, c in enumerate(columns)}
             names.sort(key=lambda n: order.get(n.lower(), 99))
         merged = {}
+        merged_masks = {}
         for name in names:
             arrs = [f[name] for f in per_file]
             merged[name] = torch.from_numpy(
                 np.concatenate(arrs) if len(arrs) > 1 else arrs[0])
-        return ColumnBatch(merged), row_counts
+            if any(name in fm for fm in per_file_masks):
+                mparts = [fm.get(name, np.ones(rc, dtype=bool))
+                          for fm, rc in zip(per_file_masks, row_counts)]
+                merged_masks[name] = torch.from_numpy(
+                    np.concatenate(mparts) if len(mparts) > 1
+                    else mparts[0])
+        return ColumnBatch(merged, merged_masks), row_counts
 
     import pyarrow.parquet as pq
     import pyarrow as pa
@@ -132,12 +144,30 @@ def read_files_batch_device(paths: List[str], device,
         return batch.to(device), rc
 
     metas = []
+    schemas = []
     for p in paths:
         try:
-            md = pq.ParquetFile(p).metadata
+            pf = pq.ParquetFile(p)
         except Exception:  # noqa: BLE001
             return fallback()
-        metas.append(md)
+        metas.append(pf.metadata)
+        schemas.append(pf.schema)
+
+    # column nullability from chunk statistics (sizes the preallocated
+    # masks); an OPTIONAL chunk without statistics could hide nulls ->
+    # the host path decides instead
+    nullable_cols = set()
+    for md, sch in zip(metas, schemas):
+        for rg_i in range(md.num_row_groups):
+            rg = md.row_group(rg_i)
+            for i in range(rg.num_columns):
+                col = rg.column(i)
+                st = col.statistics
+                if st is None or st.null_count is None:
+                    if sch.column(i).max_definition_level > 0:
+                        return fallback()
+                elif st.null_count > 0:
+                    nullable_cols.add(col.path_in_schema)
 
     want = {c.lower() for c in columns} if columns is not None else None
     rg0 = metas[0].row_group(0)
@@ -166,6 +196,9 @@ def read_files_batch_device(paths: List[str], device,
                    np.dtype("float32"): _torch.float32}
     out = {n: _torch.empty(total_rows, dtype=np_to_torch[dtypes[n]],
                            device=device) for n in names}
+    out_masks = {n: _torch.ones(total_rows, dtype=_torch.bool,
+                                device=device)
+                 for n in names if n in nullable_cols}
     ext = native_ext.ext()
 
     def load_decode(i):
@@ -196,18 +229,46 @@ def read_files_batch_device(paths: List[str], device,
                 ext.copy_unaligned(dev_bytes, dict_off, dict_vals, 0,
                                    dict_n * itemsize)
                 dict_vals = dict_vals[:dict_n].contiguous()
-            for page in c.pages:
-                if page[0] == "dict":
-                    _, p_start, p_end, nv, bw = page
-                    runs = ext.parse_rle_runs(buf, p_start, p_end, bw, nv)
-                    idx = ext.rle_decode(dev_bytes, *runs, bw,
-                                         nv).to(_torch.int64)
-                    out[c.name][written:written + nv] = \
-                        ext.gather_rows(dict_vals, idx)
+            for page, pmask in zip(c.pages, c.page_masks):
+                nv = page[3] if page[0] == "dict" else page[2]
+                if pmask is None:
+                    if page[0] == "dict":
+                        _, p_start, p_end, _, bw = page
+                        runs = ext.parse_rle_runs(buf, p_start, p_end, bw,
+                                                  nv)
+                        idx = ext.rle_decode(dev_bytes, *runs, bw,
+                                             nv).to(_torch.int64)
+                        out[c.name][written:written + nv] = \
+                            ext.gather_rows(dict_vals, idx)
+                    else:
+                        _, off, _ = page
+                        ext.copy_unaligned(dev_bytes, off, out[c.name],
+                                           written * itemsize,
+                                           nv * itemsize)
                 else:
-                    _, off, nv = page
-                    ext.copy_unaligned(dev_bytes, off, out[c.name],
-                                       written * itemsize, nv * itemsize)
+                    # nullable page: compacted values scatter to the
+                    # valid slots; null slots hold 0
+                    n_valid = int(pmask.sum())
+                    mask_dev = _torch.from_numpy(pmask).to(device)
+                    if page[0] == "dict":
+                        _, p_start, p_end, _, bw = page
+                        runs = ext.parse_rle_runs(buf, p_start, p_end, bw,
+                                                  n_valid)
+                        idx = ext.rle_decode(dev_bytes, *runs, bw,
+                                             n_valid).to(_torch.int64)
+                        vals = ext.gather_rows(dict_vals, idx)
+                    else:
+                        _, off, _ = page
+                        vals = _torch.empty(
+                            n_valid + 1, dtype=out[c.name].dtype,
+                            device=device)
+                        ext.copy_unaligned(dev_bytes, off, vals, 0,
+                                           n_valid * itemsize)
+                        vals = vals[:n_valid]
+                    dst = out[c.name][written:written + nv]
+                    dst.zero_()
+                    dst[mask_dev] = vals
+                    out_masks[c.name][written:written + nv] = mask_dev
                 written += nv
             cursors[c.name] = written
         # pinned buffer must stay alive until the stream drains; the
@@ -223,4 +284,4 @@ def read_files_batch_device(paths: List[str], device,
         return fallback()
     _torch.cuda.current_stream().synchronize()
     del bufs
-    return ColumnBatch(out), row_counts
+    return ColumnBatch(out, out_masks), row_counts

@@ -44,10 +44,11 @@ def test_pyarrow_reads_native_files(tmp_path, cols):
 def test_native_reads_own_files(tmp_path, cols):
     p = str(tmp_path / "t.parquet")
     write_parquet_native(cols, p)
-    back = read_native_host(p)
+    back, masks = read_native_host(p)
+    assert masks == {}
     for name, arr in cols.items():
         assert np.array_equal(back[name], arr), name
-    sub = read_native_host(p, columns=["key"])
+    sub, _ = read_native_host(p, columns=["key"])
     assert list(sub.keys()) == ["key"]
 
 
@@ -56,8 +57,9 @@ def test_native_reads_pyarrow_files_multipage(tmp_path, cols):
     p = str(tmp_path / "t.parquet")
     pq.write_table(pa.table(cols), p, compression="NONE",
                    use_dictionary=False, data_page_version="1.0")
-    back = read_native_host(p)
-    assert back is not None
+    res = read_native_host(p)
+    assert res is not None
+    back, _ = res
     for name, arr in cols.items():
         assert np.array_equal(back[name], arr), name
 
@@ -86,7 +88,7 @@ def test_empty_and_single_row(tmp_path):
     p = str(tmp_path / "one.parquet")
     write_parquet_native({"a": np.array([7], dtype=np.int64)}, p)
     assert pq.read_table(p).column("a").to_pylist() == [7]
-    back = read_native_host(p)
+    back, _ = read_native_host(p)
     assert back["a"].tolist() == [7]
 
 
@@ -99,8 +101,9 @@ def test_multi_rowgroup_native_read(tmp_path):
                    use_dictionary=False, data_page_version="1.0",
                    row_group_size=250_000)
     assert pq.ParquetFile(p).metadata.num_row_groups > 1
-    back = read_native_host(p)
-    assert back is not None
+    res = read_native_host(p)
+    assert res is not None
+    back, _ = res
     for k, v in cols.items():
         assert np.array_equal(back[k], v), k
     batch, counts = read_files_batch([p])
