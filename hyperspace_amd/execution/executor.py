@@ -472,8 +472,10 @@ class Executor:
             # sorted by the join key within each bucket at build time.
             self.stats.merge_joins += 1
             self.stats.record("SortMergeJoin(co-bucketed)")
-            lk = ops.normalize_key(lbatch.tensor(lkeys_names[0]))
-            rk = ops.normalize_key(rbatch.tensor(rkeys_names[0]))
+            lk_t, rk_t = _join_key_tensors(lbatch, rbatch,
+                                           lkeys_names[0], rkeys_names[0])
+            lk = ops.normalize_key(lk_t)
+            rk = ops.normalize_key(rk_t)
             lidx, ridx = ops.merge_join(lk, rk, lseg, rseg)
         else:
             # on-the-fly sort-merge join: shuffle-equivalent (counts as an
@@ -481,8 +483,10 @@ class Executor:
             self.stats.shuffles += 2
             self.stats.hash_joins += 1
             self.stats.record("SortMergeJoin(shuffled)")
-            lk = ops.normalize_key(lbatch.tensor(lkeys_names[0]))
-            rk = ops.normalize_key(rbatch.tensor(rkeys_names[0]))
+            lk_t, rk_t = _join_key_tensors(lbatch, rbatch,
+                                           lkeys_names[0], rkeys_names[0])
+            lk = ops.normalize_key(lk_t)
+            rk = ops.normalize_key(rk_t)
             lperm = ops.sort_perm(lk)
             rperm = ops.sort_perm(rk)
             lbatch = lbatch.gather(lperm)
@@ -498,9 +502,8 @@ class Executor:
             keep = torch.ones(lidx.numel(), dtype=torch.bool,
                               device=lidx.device)
             for ln, rn in pairs[1:]:
-                lv = lbatch.tensor(ln)[lidx]
-                rv = rbatch.tensor(rn)[ridx]
-                keep &= (lv == rv)
+                lv_t, rv_t = _join_key_tensors(lbatch, rbatch, ln, rn)
+                keep &= (lv_t[lidx] == rv_t[ridx])
                 lm, rm = lbatch.mask(ln), rbatch.mask(rn)
                 if lm is not None:
                     keep &= lm[lidx]
@@ -587,6 +590,40 @@ class Executor:
 
 _SPARK_DTYPES = {"long": torch.int64, "integer": torch.int32,
                  "double": torch.float64, "float": torch.float32}
+
+
+def _join_key_tensors(lbatch: ColumnBatch, rbatch: ColumnBatch,
+                      ln: str, rn: str
+                      ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Comparable key tensors for one join-key pair.
+
+    String keys from independent indexes carry independent dictionaries,
+    so both sides remap onto their merged sorted dictionary.  The remap
+    is monotone (sorted dict -> code order == lexicographic order), so
+    per-bucket sortedness from the build survives and the merge join
+    stays valid (reference joins strings via Spark's UTF8String
+    comparisons; here the dictionary merge happens once per join and the
+    remap is a device gather)."""
+    lcol = lbatch.column(ln)
+    rcol = rbatch.column(rn)
+    lstr = isinstance(lcol, StringColumn)
+    rstr = isinstance(rcol, StringColumn)
+    if lstr != rstr:
+        raise HyperspaceException(
+            f"join key type mismatch: {ln} vs {rn}")
+    if not lstr:
+        return lbatch.tensor(ln), rbatch.tensor(rn)
+    merged = sorted(set(lcol.values) | set(rcol.values))
+    vi = {v: i for i, v in enumerate(merged)}
+    llut = torch.tensor([vi[v] for v in lcol.values], dtype=torch.int64,
+                        device=lcol.codes.device)
+    rlut = torch.tensor([vi[v] for v in rcol.values], dtype=torch.int64,
+                        device=rcol.codes.device)
+    lk = llut[lcol.codes.long()] if len(lcol.values) else \
+        lcol.codes.long()
+    rk = rlut[rcol.codes.long()] if len(rcol.values) else \
+        rcol.codes.long()
+    return lk, rk
 
 
 def _drop_null_keys(batch: ColumnBatch, key_name: str,

@@ -236,13 +236,6 @@ class JoinIndexRule(HyperspaceRule):
             index = cand.index
             if index.kind != "CoveringIndex":
                 continue
-            # string join keys are not merge-joinable across independent
-            # dictionaries yet
-            if any(index.schema.field_type(k) == "string" for k in keys):
-                self.reasons.add(cand.name, plan, FilterReason(
-                    FilterReasons.NOT_ELIGIBLE_JOIN,
-                    {"reason": "string join keys unsupported"}))
-                continue
             # indexed columns must be exactly the join keys (set equality;
             # order compatibility across the pair is checked by the ranker)
             if set(c.lower() for c in index.indexed_columns) != \
