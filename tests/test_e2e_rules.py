@@ -255,11 +255,15 @@ def test_string_join_keys_not_rewritten(env, tmp_path):
     h.create_index(left, hs.CoveringIndexConfig("sjl", ["name"], ["v"]))
     h.create_index(right, hs.CoveringIndexConfig("sjr", ["name"], ["s"]))
     session.enable_hyperspace()
-    plan = left.select("name", "v").join(right.select("name", "s"),
-                                         on="name").optimized_plan()
-    # string join keys: rule declines, plan keeps source scans
-    assert not any(isinstance(l, IndexScan)
-                   for l in plan.collect_leaves())
+    q = left.select("name", "v").join(right.select("name", "s"),
+                                      on="name")
+    plan = q.optimized_plan()
+    # string join keys rewrite via the merged-dictionary remap
+    assert all(isinstance(l, IndexScan) for l in plan.collect_leaves())
+    ex = Executor(session)
+    out = ex.execute(plan)
+    assert ex.stats.merge_joins == 1 and ex.stats.shuffles == 0
+    assert out.num_rows == 1000
 
 
 def test_join_correct_after_incremental_refresh(env, tmp_path):
