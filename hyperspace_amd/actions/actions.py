@@ -247,6 +247,18 @@ class RefreshActionBase(Action):
         assert self._entry is not None
         return self._entry
 
+    def _enrich_index(self, index, relation):
+        """Provider property enrichment on refresh (reference
+        RefreshActionBase via FileBasedRelationMetadata
+        .enrichIndexProperties — delta appends the new
+        end-entry-id:table-version pair to the 'deltaVersions' history
+        consumed by closestIndex time travel)."""
+        if hasattr(relation, "enrich_index_properties"):
+            return index.with_new_properties(
+                relation.enrich_index_properties(
+                    index.properties, (self.base_id or 0) + 2))
+        return index
+
 
 class RefreshFullAction(RefreshActionBase):
     """Full rebuild into a new data version."""
@@ -274,6 +286,7 @@ class RefreshFullAction(RefreshActionBase):
                 ctx, df, index.properties)
             written = new_index.write(ctx, batch)
         written = _all_written_files(data_path, written)
+        new_index = self._enrich_index(new_index, relation)
         rel_meta = relation.create_relation_metadata(tracker)
         self._entry = IndexLogEntry.create(
             self.previous.name, new_index, _content_from_paths(written),
@@ -342,6 +355,7 @@ class RefreshIncrementalAction(RefreshActionBase):
                     ctx, appended_batch, deleted_ids, previous_files)
 
         relation = self.source_relation()
+        index = self._enrich_index(index, relation)
         rel_meta = relation.create_relation_metadata(tracker)
         all_files = sorted(set(written) | set(kept))
         self._entry = IndexLogEntry.create(

@@ -34,6 +34,11 @@ class IndexCollectionManager:
         path = self.path_resolver.get_index_path(name)
         return path, IndexLogManager(path), IndexDataManager(path)
 
+    def log_manager(self, name: str) -> IndexLogManager:
+        """Log manager for one index (closestIndex time travel reads
+        retained historical entries through it)."""
+        return self._managers(name)[1]
+
     # -- mutations ---------------------------------------------------------
     def create(self, df, config) -> None:
         path, log_mgr, data_mgr = self._managers(config.index_name)

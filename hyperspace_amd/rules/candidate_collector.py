@@ -111,6 +111,24 @@ class CandidateIndexCollector:
             cand.tags[TAG_COMMON_SOURCE_SIZE] = entry.source_files_size()
             return cand
 
+        # Delta time travel: swap in the closest retained index version
+        # (reference FileSignatureFilter delegates closestIndex to the
+        # relation, delta/DeltaLakeRelation.scala:179-251)
+        closest_fn = getattr(scan.relation, "closest_index_log_entry",
+                             None)
+        if closest_fn is not None:
+            import os as _os
+            lm = self.session.index_manager().log_manager(entry.name)
+            closest = closest_fn(entry, lm)
+            if closest is not None and closest.signature is not None and \
+                    closest.signature.value == current_sig and \
+                    all(_os.path.exists(p)
+                        for p in closest.content.os_files()[:1]):
+                c2 = Candidate(closest)
+                c2.tags[TAG_COMMON_SOURCE_SIZE] = \
+                    closest.source_files_size()
+                return c2
+
         # signature mismatch -> hybrid scan file-level overlap
         if not self.session.conf.hybrid_scan_enabled and \
                 not entry.has_source_update():

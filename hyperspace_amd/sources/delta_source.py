@@ -180,6 +180,35 @@ class DeltaTableRelation(FileBasedRelation):
             else ""
         return f"delta:{self.table.path}{v}"
 
+    def closest_index_log_entry(self, entry, log_manager):
+        """Time-travel index selection (reference
+        delta/DeltaLakeRelation.scala:179-251, closestIndex): for a query
+        pinned at table version Q, pick the retained index log version
+        whose recorded delta version is nearest to Q — preferring the
+        newest version at or below Q (its data is a subset-superset
+        closest in bytes) — and load that log entry."""
+        if self.version_as_of is None:
+            return None
+        hist = (entry.properties or {}).get(
+            "deltaVersions",
+            entry.derivedDataset.properties.get("deltaVersions", ""))
+        pairs = []
+        for tok in hist.split(","):
+            if ":" in tok:
+                k, v = tok.split(":", 1)
+                pairs.append((int(k), int(v)))
+        if not pairs:
+            return None
+        q = self.snapshot_version
+        below = [(k, v) for k, v in pairs if v <= q]
+        log_id, _ = (max(below, key=lambda p: (p[1], p[0])) if below
+                     else min(pairs, key=lambda p: (p[1], -p[0])))
+        from ..log.constants import States
+        e = log_manager.get_log(log_id)
+        if e is None or e.state != States.ACTIVE:
+            return None
+        return e
+
     def enrich_index_properties(self, properties: Dict[str, str],
                                 index_log_version: int) -> Dict[str, str]:
         """Maintain the index->table version history (reference

