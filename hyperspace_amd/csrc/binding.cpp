@@ -172,32 +172,30 @@ std::tuple<torch::Tensor, torch::Tensor> merge_join(torch::Tensor lkeys,
   if (n_left == 0 || rkeys.numel() == 0) {
     return {torch::empty({0}, opts), torch::empty({0}, opts)};
   }
-  auto counts = torch::empty({n_left}, opts);
-  auto starts = torch::empty({n_left}, opts);
-  auto seg_of = torch::empty({n_left}, opts);
-  hsk::merge_join_count((const uint64_t*)lkeys.data_ptr<int64_t>(),
-                        (const uint64_t*)rkeys.data_ptr<int64_t>(),
-                        lseg_d.data_ptr<int64_t>(),
-                        rseg_d.data_ptr<int64_t>(), n_left, n_seg,
-                        counts.data_ptr<int64_t>(),
-                        starts.data_ptr<int64_t>(),
-                        seg_of.data_ptr<int64_t>(), stream);
-  auto offsets = torch::empty({n_left}, opts);
+  int64_t tile = hsk::merge_join_tile_size();
+  int64_t n_tiles = (n_left + tile - 1) / tile;
+  auto tile_counts = torch::empty({n_tiles}, opts);
+  hsk::merge_join_tile_count((const uint64_t*)lkeys.data_ptr<int64_t>(),
+                             (const uint64_t*)rkeys.data_ptr<int64_t>(),
+                             lseg_d.data_ptr<int64_t>(),
+                             rseg_d.data_ptr<int64_t>(), n_left, n_seg,
+                             tile_counts.data_ptr<int64_t>(), stream);
+  auto tile_offsets = torch::empty({n_tiles}, opts);
   auto total = torch::empty({1}, opts);
-  hsk::exclusive_scan_i64(counts.data_ptr<int64_t>(),
-                          offsets.data_ptr<int64_t>(), n_left,
+  hsk::exclusive_scan_i64(tile_counts.data_ptr<int64_t>(),
+                          tile_offsets.data_ptr<int64_t>(), n_tiles,
                           total.data_ptr<int64_t>(), stream);
   int64_t n_out = total.cpu().item<int64_t>();
   auto out_l = torch::empty({n_out}, opts);
   auto out_r = torch::empty({n_out}, opts);
   if (n_out > 0) {
-    hsk::merge_join_emit(offsets.data_ptr<int64_t>(),
-                         counts.data_ptr<int64_t>(),
-                         starts.data_ptr<int64_t>(),
-                         seg_of.data_ptr<int64_t>(),
-                         rseg_d.data_ptr<int64_t>(), n_left,
-                         out_l.data_ptr<int64_t>(),
-                         out_r.data_ptr<int64_t>(), stream);
+    hsk::merge_join_tile_emit((const uint64_t*)lkeys.data_ptr<int64_t>(),
+                              (const uint64_t*)rkeys.data_ptr<int64_t>(),
+                              lseg_d.data_ptr<int64_t>(),
+                              rseg_d.data_ptr<int64_t>(), n_left, n_seg,
+                              tile_offsets.data_ptr<int64_t>(),
+                              out_l.data_ptr<int64_t>(),
+                              out_r.data_ptr<int64_t>(), stream);
   }
   return {out_l, out_r};
 }

@@ -34,18 +34,21 @@ void radix_sort_pairs32(uint64_t* keys, int32_t* payload,
 // histogram buffer size requirement (u32 elements)
 int64_t radix_sort_hist_size(int64_t n);
 
-// K4: segmented sorted merge join. Phase 1: per-left-row match counts +
-// right start positions.  Phase 2 (after exclusive scan of counts into
-// offsets): emit pairs.
-void merge_join_count(const uint64_t* lkeys, const uint64_t* rkeys,
-                      const int64_t* lseg, const int64_t* rseg,
-                      int64_t n_left, int64_t n_seg, int64_t* counts,
-                      int64_t* starts, int64_t* seg_of_row,
-                      hipStream_t stream);
-void merge_join_emit(const int64_t* offsets, const int64_t* counts,
-                     const int64_t* starts, const int64_t* seg_of_row,
-                     const int64_t* rseg, int64_t n_left, int64_t* out_l,
-                     int64_t* out_r, hipStream_t stream);
+// K4: segmented sorted merge join, two-phase per-tile form.  Phase 1
+// writes one pair-count per 256-row left tile; after an exclusive scan
+// of tile counts, phase 2 recomputes the tile-narrowed searches and
+// emits pairs at tile_offset + intra-tile LDS prefix (no per-row
+// metadata arrays touch HBM).
+int64_t merge_join_tile_size();
+void merge_join_tile_count(const uint64_t* lkeys, const uint64_t* rkeys,
+                           const int64_t* lseg, const int64_t* rseg,
+                           int64_t n_left, int64_t n_seg,
+                           int64_t* tile_counts, hipStream_t stream);
+void merge_join_tile_emit(const uint64_t* lkeys, const uint64_t* rkeys,
+                          const int64_t* lseg, const int64_t* rseg,
+                          int64_t n_left, int64_t n_seg,
+                          const int64_t* tile_offsets, int64_t* out_l,
+                          int64_t* out_r, hipStream_t stream);
 
 // generic exclusive scan over i64 (single pass, device-wide)
 void exclusive_scan_i64(const int64_t* in, int64_t* out, int64_t n,

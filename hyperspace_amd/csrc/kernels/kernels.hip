@@ -561,113 +561,192 @@ __device__ __forceinline__ int64_t upper_bound_u64(
   return lo;
 }
 
-__global__ void k_mj_count(const uint64_t* __restrict__ lkeys,
-                           const uint64_t* __restrict__ rkeys,
-                           const int64_t* __restrict__ lseg,
-                           const int64_t* __restrict__ rseg, int64_t n_left,
-                           int64_t n_seg, int64_t* __restrict__ counts,
-                           int64_t* __restrict__ starts,
-                           int64_t* __restrict__ seg_of_row) {
-  // Tile-narrowed search: both sides are sorted within a segment, so a
-  // 256-row left tile maps into a contiguous right range.  Two
-  // cooperative searches bound that range; per-row searches then probe
-  // only the (cache-resident) narrowed window instead of walking the
-  // whole multi-MB segment from HBM per row.
-  __shared__ int64_t tile_lo, tile_hi, tile_seg;
-  int64_t n_tiles = cdiv(n_left, (int64_t)blockDim.x);
+// Two-phase tile merge join.  The old design wrote three per-left-row
+// i64 arrays (counts/starts/segment) then scanned and re-read them —
+// ~19 GB of HBM traffic for a 268M-row probe side.  Both phases here
+// recompute the (L2-resident) narrowed searches instead and only per-
+// TILE totals touch HBM: phase 1 block-reduces each 256-row tile's pair
+// count (n/256 i64 writes), an exclusive scan over tiles sizes the
+// output, and phase 2 redoes the searches, prefix-sums the counts in
+// LDS and writes the pairs at tile_offset + intra-tile prefix — output
+// ordered by left row, bit-identical to the one-phase result.
+//
+// find_range: tile-narrowed search shared by both phases.  Both sides
+// are sorted within a segment, so a left tile maps into a contiguous
+// right window; two cooperative searches bound it and per-row probes
+// stay inside the cache-resident window.
+// rows per thread: amortizes the serial tile-narrow searches over 4x
+// bigger tiles; consecutive rows per thread keep output left-row order
+#define MJ_RPT 4
+#define MJ_TILE ((int64_t)THREADS * MJ_RPT)
+
+__device__ __forceinline__ int64_t mj_seg_of(
+    const int64_t* __restrict__ lseg, int64_t n_seg, int64_t row) {
+  int64_t lo = 0, hi = n_seg;
+  while (lo < hi) {
+    int64_t mid = (lo + hi) >> 1;
+    if (lseg[mid + 1] <= row)
+      lo = mid + 1;
+    else
+      hi = mid;
+  }
+  return lo;
+}
+
+// threads 0/1 compute the two window bounds in parallel (each bound is
+// a serial dependent-load chain; splitting halves the critical path)
+__device__ __forceinline__ void mj_tile_narrow(
+    const uint64_t* __restrict__ lkeys, const uint64_t* __restrict__ rkeys,
+    const int64_t* __restrict__ lseg, const int64_t* __restrict__ rseg,
+    int64_t n_left, int64_t n_seg, int64_t base, int64_t* tile_seg,
+    int64_t* tile_lo, int64_t* tile_hi) {
+  int64_t last = base + MJ_TILE - 1 < n_left - 1 ? base + MJ_TILE - 1
+                                                 : n_left - 1;
+  int64_t s = mj_seg_of(lseg, n_seg, base);
+  if (threadIdx.x == 0) {
+    if (last < lseg[s + 1]) {
+      *tile_seg = s;
+      *tile_lo = lower_bound_u64(rkeys, rseg[s], rseg[s + 1], lkeys[base]);
+    } else {
+      *tile_seg = -1;  // tile spans a segment boundary: per-row fallback
+    }
+  } else {
+    if (last < lseg[s + 1])
+      *tile_hi = upper_bound_u64(rkeys, rseg[s], rseg[s + 1], lkeys[last]);
+  }
+}
+
+// fill this thread's MJ_RPT consecutive rows: one binary search for the
+// first row, forward-monotone searches for the rest (keys ascend, so
+// each row's range starts at or after the previous row's)
+__device__ __forceinline__ int64_t mj_thread_rows(
+    const uint64_t* __restrict__ lkeys, const uint64_t* __restrict__ rkeys,
+    const int64_t* __restrict__ lseg, const int64_t* __restrict__ rseg,
+    int64_t n_left, int64_t n_seg, int64_t first, int64_t tile_seg,
+    int64_t tile_lo, int64_t tile_hi, int64_t* a_out, int64_t* c_out) {
+  int64_t total = 0;
+  int64_t a_prev = tile_lo;
+  int64_t r1 = tile_hi;
+  for (int k = 0; k < MJ_RPT; k++) {
+    int64_t i = first + k;
+    if (i >= n_left || (tile_seg >= 0 && i >= lseg[tile_seg + 1])) {
+      c_out[k] = 0;
+      continue;
+    }
+    if (tile_seg < 0) {
+      // boundary tile: resolve the segment per row; reset the monotone
+      // cursor on the first row (tile_lo is unset) and on segment change
+      int64_t s = mj_seg_of(lseg, n_seg, i);
+      int64_t s_start = rseg[s];
+      if (k == 0 || s_start > a_prev) a_prev = s_start;
+      r1 = rseg[s + 1];
+    }
+    uint64_t key = lkeys[i];
+    int64_t a = lower_bound_u64(rkeys, a_prev, r1, key);
+    int64_t b = upper_bound_u64(rkeys, a, r1, key);
+    a_out[k] = a;
+    c_out[k] = b - a;
+    total += b - a;
+    a_prev = a;
+  }
+  return total;
+}
+
+__global__ void k_mj_tile_count(const uint64_t* __restrict__ lkeys,
+                                const uint64_t* __restrict__ rkeys,
+                                const int64_t* __restrict__ lseg,
+                                const int64_t* __restrict__ rseg,
+                                int64_t n_left, int64_t n_seg,
+                                int64_t* __restrict__ tile_counts) {
+  __shared__ int64_t tile_seg, tile_lo, tile_hi;
+  __shared__ int64_t red[THREADS];
+  int64_t n_tiles = cdiv(n_left, MJ_TILE);
   for (int64_t tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
-    int64_t base = tile * (int64_t)blockDim.x;
-    int64_t last = base + blockDim.x - 1 < n_left - 1
-                       ? base + blockDim.x - 1
-                       : n_left - 1;
-    if (threadIdx.x == 0) {
-      // segment of the tile's first row
-      int64_t lo = 0, hi = n_seg;
-      while (lo < hi) {
-        int64_t mid = (lo + hi) >> 1;
-        if (lseg[mid + 1] <= base)
-          lo = mid + 1;
-        else
-          hi = mid;
-      }
-      tile_seg = lo;
-      // does the tile span a segment boundary?
-      if (last < lseg[lo + 1]) {
-        int64_t r0 = rseg[lo], r1 = rseg[lo + 1];
-        tile_lo = lower_bound_u64(rkeys, r0, r1, lkeys[base]);
-        tile_hi = upper_bound_u64(rkeys, r0, r1, lkeys[last]);
-      } else {
-        tile_seg = -1;  // mixed-segment tile: per-row fallback
-      }
-    }
+    int64_t base = tile * MJ_TILE;
+    if (threadIdx.x < 2)
+      mj_tile_narrow(lkeys, rkeys, lseg, rseg, n_left, n_seg, base,
+                     &tile_seg, &tile_lo, &tile_hi);
     __syncthreads();
-    int64_t i = base + threadIdx.x;
-    if (i < n_left) {
-      int64_t s, r0, r1;
-      if (tile_seg >= 0) {
-        s = tile_seg;
-        r0 = tile_lo;
-        r1 = tile_hi;
-      } else {
-        int64_t lo = 0, hi = n_seg;
-        while (lo < hi) {
-          int64_t mid = (lo + hi) >> 1;
-          if (lseg[mid + 1] <= i)
-            lo = mid + 1;
-          else
-            hi = mid;
-        }
-        s = lo;
-        r0 = rseg[s];
-        r1 = rseg[s + 1];
+    int64_t a_arr[MJ_RPT], c_arr[MJ_RPT];
+    red[threadIdx.x] = mj_thread_rows(
+        lkeys, rkeys, lseg, rseg, n_left, n_seg,
+        base + (int64_t)threadIdx.x * MJ_RPT, tile_seg, tile_lo, tile_hi,
+        a_arr, c_arr);
+    __syncthreads();
+    for (int off = blockDim.x / 2; off > 0; off >>= 1) {
+      if ((int)threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) tile_counts[tile] = red[0];
+    __syncthreads();
+  }
+}
+
+__global__ void k_mj_tile_emit(const uint64_t* __restrict__ lkeys,
+                               const uint64_t* __restrict__ rkeys,
+                               const int64_t* __restrict__ lseg,
+                               const int64_t* __restrict__ rseg,
+                               int64_t n_left, int64_t n_seg,
+                               const int64_t* __restrict__ tile_offsets,
+                               int64_t* __restrict__ out_l,
+                               int64_t* __restrict__ out_r) {
+  __shared__ int64_t tile_seg, tile_lo, tile_hi;
+  __shared__ int64_t pfx[THREADS];
+  int64_t n_tiles = cdiv(n_left, MJ_TILE);
+  for (int64_t tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
+    int64_t base = tile * MJ_TILE;
+    if (threadIdx.x < 2)
+      mj_tile_narrow(lkeys, rkeys, lseg, rseg, n_left, n_seg, base,
+                     &tile_seg, &tile_lo, &tile_hi);
+    __syncthreads();
+    int64_t a_arr[MJ_RPT], c_arr[MJ_RPT];
+    int64_t my_total = mj_thread_rows(
+        lkeys, rkeys, lseg, rseg, n_left, n_seg,
+        base + (int64_t)threadIdx.x * MJ_RPT, tile_seg, tile_lo, tile_hi,
+        a_arr, c_arr);
+    // exclusive intra-tile prefix over per-thread totals (Hillis-Steele)
+    pfx[threadIdx.x] = my_total;
+    __syncthreads();
+    for (int off = 1; off < (int)blockDim.x; off <<= 1) {
+      int64_t v = (int)threadIdx.x >= off ? pfx[threadIdx.x - off] : 0;
+      __syncthreads();
+      pfx[threadIdx.x] += v;
+      __syncthreads();
+    }
+    int64_t off_out = tile_offsets[tile] + pfx[threadIdx.x] - my_total;
+    for (int k = 0; k < MJ_RPT; k++) {
+      int64_t i = base + (int64_t)threadIdx.x * MJ_RPT + k;
+      for (int64_t j = 0; j < c_arr[k]; j++) {
+        out_l[off_out] = i;
+        out_r[off_out] = a_arr[k] + j;
+        off_out++;
       }
-      uint64_t key = lkeys[i];
-      int64_t a = lower_bound_u64(rkeys, r0, r1, key);
-      int64_t b = upper_bound_u64(rkeys, r0, r1, key);
-      counts[i] = b - a;
-      starts[i] = a;
-      seg_of_row[i] = s;
     }
     __syncthreads();
   }
 }
 
-__global__ void k_mj_emit(const int64_t* __restrict__ offsets,
-                          const int64_t* __restrict__ counts,
-                          const int64_t* __restrict__ starts,
-                          int64_t n_left, int64_t* __restrict__ out_l,
-                          int64_t* __restrict__ out_r) {
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       i < n_left; i += stride) {
-    int64_t c = counts[i];
-    int64_t off = offsets[i];
-    int64_t a = starts[i];
-    for (int64_t j = 0; j < c; j++) {
-      out_l[off + j] = i;
-      out_r[off + j] = a + j;
-    }
-  }
+int64_t merge_join_tile_size() { return MJ_TILE; }
+
+void merge_join_tile_count(const uint64_t* lkeys, const uint64_t* rkeys,
+                           const int64_t* lseg, const int64_t* rseg,
+                           int64_t n_left, int64_t n_seg,
+                           int64_t* tile_counts, hipStream_t stream) {
+  if (n_left == 0) return;
+  hipLaunchKernelGGL(k_mj_tile_count, dim3(grid_for(n_left)), dim3(THREADS),
+                     0, stream, lkeys, rkeys, lseg, rseg, n_left, n_seg,
+                     tile_counts);
 }
 
-void merge_join_count(const uint64_t* lkeys, const uint64_t* rkeys,
-                      const int64_t* lseg, const int64_t* rseg,
-                      int64_t n_left, int64_t n_seg, int64_t* counts,
-                      int64_t* starts, int64_t* seg_of_row,
-                      hipStream_t stream) {
+void merge_join_tile_emit(const uint64_t* lkeys, const uint64_t* rkeys,
+                          const int64_t* lseg, const int64_t* rseg,
+                          int64_t n_left, int64_t n_seg,
+                          const int64_t* tile_offsets, int64_t* out_l,
+                          int64_t* out_r, hipStream_t stream) {
   if (n_left == 0) return;
-  hipLaunchKernelGGL(k_mj_count, dim3(grid_for(n_left)), dim3(THREADS), 0,
-                     stream, lkeys, rkeys, lseg, rseg, n_left, n_seg, counts,
-                     starts, seg_of_row);
-}
-
-void merge_join_emit(const int64_t* offsets, const int64_t* counts,
-                     const int64_t* starts, const int64_t* /*seg_of_row*/,
-                     const int64_t* /*rseg*/, int64_t n_left, int64_t* out_l,
-                     int64_t* out_r, hipStream_t stream) {
-  if (n_left == 0) return;
-  hipLaunchKernelGGL(k_mj_emit, dim3(grid_for(n_left)), dim3(THREADS), 0,
-                     stream, offsets, counts, starts, n_left, out_l, out_r);
+  hipLaunchKernelGGL(k_mj_tile_emit, dim3(grid_for(n_left)), dim3(THREADS),
+                     0, stream, lkeys, rkeys, lseg, rseg, n_left, n_seg,
+                     tile_offsets, out_l, out_r);
 }
 
 // ---------------------------------------------------------------------------
