@@ -577,7 +577,7 @@ __device__ __forceinline__ int64_t upper_bound_u64(
 // stay inside the cache-resident window.
 // rows per thread: amortizes the serial tile-narrow searches over 4x
 // bigger tiles; consecutive rows per thread keep output left-row order
-#define MJ_RPT 4
+#define MJ_RPT 8
 #define MJ_TILE ((int64_t)THREADS * MJ_RPT)
 
 __device__ __forceinline__ int64_t mj_seg_of(
