@@ -252,12 +252,31 @@ class ColumnBatch:
     @staticmethod
     def from_arrow(table) -> "ColumnBatch":
         """pyarrow.Table -> ColumnBatch (host tensors).  Null slots become
-        validity-mask entries; the value buffer holds a fill (0 / "")."""
+        validity-mask entries; the value buffer holds a fill (0 / "").
+        Struct columns flatten into dotted leaf columns ("a.b.c") with
+        parent-null propagation — the engine's analog of the reference's
+        ``__hs_nested.`` flat aliases (util/ResolverUtils.scala)."""
         import pyarrow as pa
+        import pyarrow.compute as pc
         cols: Dict[str, Column] = {}
         masks: Dict[str, torch.Tensor] = {}
+        pairs = []
         for name, col in zip(table.column_names, table.columns):
             col = col.combine_chunks()
+            if pa.types.is_struct(col.type):
+                stack = [(name, col)]
+                while stack:
+                    pname, pcol = stack.pop()
+                    for i, f in enumerate(pcol.type):
+                        child = pc.struct_field(pcol, [i])
+                        leaf = f"{pname}.{f.name}"
+                        if pa.types.is_struct(child.type):
+                            stack.append((leaf, child))
+                        else:
+                            pairs.append((leaf, child))
+            else:
+                pairs.append((name, col))
+        for name, col in pairs:
             if col.null_count:
                 masks[name] = torch.from_numpy(
                     np.ascontiguousarray(col.is_valid().to_numpy(

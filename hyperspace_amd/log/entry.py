@@ -82,10 +82,25 @@ class Schema:
 
     @staticmethod
     def from_arrow(arrow_schema) -> "Schema":
+        """Struct fields flatten into dotted leaf fields ("a.b.c"),
+        matching the batch layer's nested flattening (the reference
+        resolves nested leaves as ``__hs_nested.``-prefixed flat columns,
+        util/ResolverUtils.scala)."""
+        import pyarrow as pa
         fields = []
+
+        def walk(f, prefix, nullable):
+            if pa.types.is_struct(f.type):
+                for child in f.type:
+                    walk(child, f"{prefix}{f.name}.",
+                         nullable or f.nullable)
+            else:
+                t = _PYARROW_TO_SPARK.get(str(f.type), str(f.type))
+                fields.append(SchemaField(prefix + f.name, t,
+                                          nullable or f.nullable))
+
         for f in arrow_schema:
-            t = _PYARROW_TO_SPARK.get(str(f.type), str(f.type))
-            fields.append(SchemaField(f.name, t, f.nullable))
+            walk(f, "", False)
         return Schema(fields)
 
     def select(self, names: Iterable[str]) -> "Schema":

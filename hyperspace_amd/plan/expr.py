@@ -204,7 +204,7 @@ def extract_equi_join_keys(e: Expr) -> List[tuple]:
 
 
 _PRED_RE = re.compile(
-    r"^\s*([A-Za-z_][A-Za-z0-9_]*)\s*(=|==|!=|<=|>=|<|>)\s*(.+?)\s*$")
+    r"^\s*([A-Za-z_][A-Za-z0-9_.]*)\s*(=|==|!=|<=|>=|<|>)\s*(.+?)\s*$")
 
 
 def parse_predicate(s: str) -> Expr:

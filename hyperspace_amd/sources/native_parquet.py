@@ -479,8 +479,14 @@ def _walk_row_group(rg, pf_schema, data, want
         np_dtype = _PARQUET_TO_NP[ptype]
         # OPTIONAL columns carry a def-level prefix per page; when the
         # chunk statistics prove null_count == 0 the levels are skipped
-        # without decoding, else each page's levels become a validity mask
-        has_levels = pf_schema.column(col_index).max_definition_level > 0
+        # without decoding, else each page's levels become a validity
+        # mask.  Nested leaves (max_def > 1 or repeated) use wider level
+        # encodings: pyarrow fallback.
+        col_schema = pf_schema.column(col_index)
+        if col_schema.max_definition_level > 1 or \
+                col_schema.max_repetition_level > 0:
+            return None
+        has_levels = col_schema.max_definition_level > 0
         st = col.statistics
         chunk_all_valid = st is not None and st.null_count == 0
 

@@ -106,13 +106,34 @@ def read_files_batch(paths: List[str], columns: Optional[List[str]] = None):
     tables = []
     row_counts = []
     for p in paths:
-        t = pq.read_table(p, columns=columns)
+        read_cols = columns
+        if columns is not None:
+            # a dotted request ("a.b.c") may be a nested leaf: read the
+            # top-level struct and let from_arrow flatten it.  Flat
+            # columns whose NAME contains dots (our index data for
+            # nested leaves) match the file schema directly.
+            top = {n.lower(): n for n in pq.ParquetFile(p)
+                   .schema_arrow.names}
+            read_cols = []
+            for c in columns:
+                if c.lower() in top:
+                    read_cols.append(top[c.lower()])
+                elif "." in c and c.split(".")[0].lower() in top:
+                    root = top[c.split(".")[0].lower()]
+                    if root not in read_cols:
+                        read_cols.append(root)
+                else:
+                    read_cols.append(c)
+        t = pq.read_table(p, columns=read_cols)
         tables.append(t)
         row_counts.append(t.num_rows)
     if not tables:
         return ColumnBatch({}), []
     table = pa.concat_tables(tables, promote_options="default")
-    return ColumnBatch.from_arrow(table), row_counts
+    batch = ColumnBatch.from_arrow(table)
+    if columns is not None:
+        batch = batch.select(columns)
+    return batch, row_counts
 
 
 def read_files_batch_device(paths: List[str], device,

@@ -1,7 +1,9 @@
 """Case-insensitive column resolution.
 
-Reference: util/ResolverUtils.scala:44-104 (nested-column ``__hs_nested.``
-prefixing is not supported in v0 — flat columns only).
+Reference: util/ResolverUtils.scala:44-104.  Nested struct leaves are
+flattened to dotted names ("a.b.c") at the Arrow boundary; the
+reference's ``__hs_nested.a.b.c`` spelling resolves as an alias of the
+dotted form, so configs written for the reference keep working.
 """
 
 from __future__ import annotations
@@ -10,12 +12,21 @@ from typing import Iterable, List, Optional
 
 from ..exceptions import HyperspaceException
 
+NESTED_PREFIX = "__hs_nested."
+
+
+def _canon(name: str) -> str:
+    n = name.lower()
+    return n[len(NESTED_PREFIX):] if n.startswith(NESTED_PREFIX) else n
+
 
 def resolve(available: Iterable[str], requested: str) -> Optional[str]:
     """Return the canonical-cased name from ``available`` matching
-    ``requested`` case-insensitively, or None."""
+    ``requested`` case-insensitively (``__hs_nested.`` prefixes are
+    ignored on both sides), or None."""
+    req = _canon(requested)
     for a in available:
-        if a.lower() == requested.lower():
+        if _canon(a) == req:
             return a
     return None
 
