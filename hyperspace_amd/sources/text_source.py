@@ -1,4 +1,4 @@
-"""CSV / JSON-lines / ORC file sources.
+"""CSV / JSON-lines / ORC / Avro file sources.
 
 Reference: the default source supports avro,csv,json,orc,parquet,text
 (util/HyperspaceConf.scala:110-115).  These formats have no device
@@ -20,11 +20,11 @@ from ..utils.hashing import md5_hex
 
 
 class TextFormatRelation(FileBasedRelation):
-    """Shared implementation for csv and json-lines directories."""
+    """Shared implementation for csv/json/orc/avro directories."""
 
     def __init__(self, fmt: str, root_paths: List[str],
                  options: Optional[Dict[str, str]] = None):
-        assert fmt in ("csv", "json", "orc")
+        assert fmt in ("csv", "json", "orc", "avro")
         self._fmt = fmt
         self._root_paths = [os.path.abspath(p) for p in root_paths]
         self._options = dict(options or {})
@@ -59,6 +59,9 @@ class TextFormatRelation(FileBasedRelation):
         if self._fmt == "orc":
             from pyarrow import orc
             return orc.read_table(path)
+        if self._fmt == "avro":
+            from .avro_io import read_avro
+            return read_avro(path)
         from pyarrow import json as pa_json
         return pa_json.read_json(path)
 
@@ -108,7 +111,8 @@ class TextFormatSourceProvider(FileBasedSourceProvider):
         return isinstance(relation, TextFormatRelation)
 
     def from_metadata(self, metadata: Relation):
-        if metadata.fileFormat not in ("csv", "json", "orc"):
+        if metadata.fileFormat not in ("csv", "json", "orc",
+                                       "avro"):
             return None
         return TextFormatRelation(metadata.fileFormat, metadata.rootPaths,
                                   metadata.options)

@@ -132,3 +132,86 @@ def test_camelcase_api_aliases(tmp_path, monkeypatch):
     h.restoreIndex("ax")
     h.deleteIndex("ax")
     h.vacuumIndex("ax")
+
+
+# ---------------------------------------------------------------------------
+# Avro (built-in container-format reader, sources/avro_io.py)
+# ---------------------------------------------------------------------------
+
+def test_avro_roundtrip(tmp_path):
+    import numpy as np
+    from hyperspace_amd.sources.avro_io import read_avro, write_avro
+    rng = np.random.default_rng(3)
+    vals = rng.integers(-(10**9), 10**9, 2000)
+    d = rng.random(2000)
+    names = np.array(["x", "yy", "zzz"])[rng.integers(0, 3, 2000)]
+    mask = rng.random(2000) > 0.2
+    p = str(tmp_path / "t.avro")
+    write_avro({"k": vals, "v": d, "name": names.tolist()}, p,
+               masks={"v": mask})
+    t = read_avro(p)
+    assert t.num_rows == 2000
+    assert t.column("k").to_pylist() == vals.tolist()
+    assert t.column("name").to_pylist() == names.tolist()
+    got_v = t.column("v")
+    assert got_v.null_count == int((~mask).sum())
+    import pyarrow.compute as pc
+    assert np.allclose(np.asarray(got_v.drop_null()), d[mask])
+
+
+def test_avro_deflate_codec(tmp_path):
+    # hand-build a deflate-codec file to cover the codec branch
+    import json as _json
+    import os
+    import struct
+    import zlib
+    from hyperspace_amd.sources.avro_io import (MAGIC, _write_long,
+                                                read_avro)
+    schema = {"type": "record", "name": "r",
+              "fields": [{"name": "a", "type": "long"}]}
+    rows = [7, -3, 1 << 40]
+    body = b"".join(_write_long(v) for v in rows)
+    comp = zlib.compress(body)[2:-4]  # raw deflate
+    sync = os.urandom(16)
+    p = str(tmp_path / "d.avro")
+    with open(p, "wb") as f:
+        f.write(MAGIC)
+        f.write(_write_long(2))
+        for k, v in (("avro.schema", _json.dumps(schema).encode()),
+                     ("avro.codec", b"deflate")):
+            kb = k.encode()
+            f.write(_write_long(len(kb)) + kb)
+            f.write(_write_long(len(v)) + v)
+        f.write(_write_long(0))
+        f.write(sync)
+        f.write(_write_long(len(rows)))
+        f.write(_write_long(len(comp)))
+        f.write(comp)
+        f.write(sync)
+    t = read_avro(p)
+    assert t.column("a").to_pylist() == rows
+
+
+def test_index_on_avro_source(tmp_path, monkeypatch):
+    import numpy as np
+    import hyperspace_amd as hs
+    from hyperspace_amd.plan.nodes import IndexScan
+    from hyperspace_amd.sources.avro_io import write_avro
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "idx"))
+    d = tmp_path / "av"
+    d.mkdir()
+    rng = np.random.default_rng(4)
+    key = rng.integers(0, 100, 5000)
+    val = rng.random(5000)
+    write_avro({"key": key, "val": val}, str(d / "part-0.avro"))
+    session = hs.HyperspaceSession(device="cpu")
+    session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 4)
+    h = hs.Hyperspace(session)
+    df = session.read_avro(str(d))
+    h.create_index(df, hs.CoveringIndexConfig("avx", ["key"], ["val"]))
+    session.enable_hyperspace()
+    q = df.filter("key = 42").select("key", "val")
+    assert any(isinstance(l, IndexScan)
+               for l in q.optimized_plan().collect_leaves())
+    out = q.collect()
+    assert out.num_rows == int((key == 42).sum())
