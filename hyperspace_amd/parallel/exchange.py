@@ -69,10 +69,25 @@ def exchange_by_bucket(batch: ColumnBatch, bucket_ids: torch.Tensor,
     out_masks = {}
     for name, col in batch.columns.items():
         if isinstance(col, StringColumn):
-            raise HyperspaceException(
-                "string columns not yet supported in the distributed "
-                "bucket exchange; dictionary-merge exchange is planned")
-        out_cols[name] = a2a(col)
+            # dictionary-merge exchange: every rank gathers all local
+            # dictionaries (small, host-side pickle collective), remaps
+            # its codes onto the merged sorted dictionary, and the i32
+            # codes travel like any other column.  The merged dictionary
+            # is identical on every rank, so received codes are
+            # comparable without further translation.
+            local_values = [list(col.values)]
+            gathered: List[object] = [None] * world
+            dist.all_gather_object(gathered, local_values)
+            merged = sorted(set().union(
+                *[set(v[0]) for v in gathered]))  # type: ignore[index]
+            val_index = {v: i for i, v in enumerate(merged)}
+            lut = torch.tensor([val_index[v] for v in col.values],
+                               dtype=torch.int32, device=col.codes.device)
+            codes = lut[col.codes.long()] if len(col.values) else \
+                col.codes
+            out_cols[name] = StringColumn(a2a(codes), merged)
+        else:
+            out_cols[name] = a2a(col)
         m = batch.mask(name)
         if m is not None:
             # validity masks ride along as uint8 splits on the same links
