@@ -23,6 +23,36 @@ import torch
 
 _BUCKET_RE = re.compile(r".*_(\d+)(?:\.\w+)*\.parquet$")
 
+# Pinned host staging buffers are expensive to allocate (~60 ms/GB —
+# comparable to the copy itself), so the device read path recycles them
+# through a size-classed pool (power-of-two classes, bounded total).
+_PINNED_POOL: Dict[int, List["torch.Tensor"]] = {}
+_PINNED_POOL_BYTES = 0
+_PINNED_POOL_MAX = 16 << 30
+import threading as _threading
+
+_pool_lock = _threading.Lock()
+
+
+def _pinned_get(nbytes: int) -> "torch.Tensor":
+    global _PINNED_POOL_BYTES
+    size = 1 << max(12, (nbytes - 1).bit_length())
+    with _pool_lock:
+        lst = _PINNED_POOL.get(size)
+        if lst:
+            _PINNED_POOL_BYTES -= size
+            return lst.pop()
+    return torch.empty(size, dtype=torch.uint8, pin_memory=True)
+
+
+def _pinned_put(buf: "torch.Tensor") -> None:
+    global _PINNED_POOL_BYTES
+    size = buf.numel()
+    with _pool_lock:
+        if _PINNED_POOL_BYTES + size <= _PINNED_POOL_MAX:
+            _PINNED_POOL.setdefault(size, []).append(buf)
+            _PINNED_POOL_BYTES += size
+
 
 def bucket_id_of_file(path: str) -> Optional[int]:
     m = _BUCKET_RE.match(os.path.basename(path))
@@ -225,14 +255,17 @@ def read_files_batch_device(paths: List[str], device,
     def load_decode(i):
         p = paths[i]
         size = os.path.getsize(p)
-        buf = _torch.empty(size + 4, dtype=_torch.uint8, pin_memory=True)
+        buf = _pinned_get(size + 4)
         view = memoryview(buf.numpy())
         with open(p, "rb", buffering=0) as f:
             f.readinto(view[:size])
         lay = read_native_layout(p, columns, data=view[:size])
         if lay is None:
+            _pinned_put(buf)
             return None
-        dev_bytes = buf.to(device, non_blocking=True)
+        # upload only the file's bytes (+4B decode slack), not the whole
+        # pooled size class
+        dev_bytes = buf[:size + 4].to(device, non_blocking=True)
         cursors = {n: int(file_base[i]) for n in names}
         for c in lay[1]:
             itemsize = c.np_dtype.itemsize
@@ -302,7 +335,12 @@ def read_files_batch_device(paths: List[str], device,
     else:
         bufs = [load_decode(i) for i in range(len(paths))]
     if not all(b is not None and b is not False for b in bufs):
+        _torch.cuda.current_stream().synchronize()
+        for b in bufs:
+            if b is not None and b is not False:
+                _pinned_put(b)
         return fallback()
     _torch.cuda.current_stream().synchronize()
-    del bufs
+    for b in bufs:
+        _pinned_put(b)
     return ColumnBatch(out, out_masks), row_counts
