@@ -298,6 +298,8 @@ def write_bucketed(batch: ColumnBatch, seg: torch.Tensor, out_dir: str,
     200-file write overlaps to page-cache speed.
     """
     from concurrent.futures import ThreadPoolExecutor
+    from ...execution.columnar import StringColumn
+    from ...sources.parquet_io import _pinned_get, _pinned_put
     on_device = batch.device.type == "cuda"
     jobs = []
     for b in range(num_buckets):
@@ -307,20 +309,45 @@ def write_bucketed(batch: ColumnBatch, seg: torch.Tensor, out_dir: str,
         jobs.append((b, lo, hi,
                      os.path.join(out_dir, bucket_file_name(task_id, b))))
 
-    def write_one(job):
-        b, lo, hi, out = job
+    # per-worker HIP streams + pooled pinned staging: pageable D2H runs
+    # at ~13 GB/s, pinned at ~53 GB/s, and per-bucket copies on separate
+    # streams overlap other workers' os.write calls
+    streams = ([torch.cuda.Stream(device=batch.device)
+                for _ in range(min(8, max(1, len(jobs))))]
+               if on_device else [])
+
+    def write_one(job_i):
+        b, lo, hi, out = jobs[job_i]
         piece = batch.slice(lo, hi)
+        held = []
         if on_device:
-            # per-bucket D2H inside the worker: copies overlap other
-            # workers' os.write calls (both release the GIL)
-            piece = piece.to("cpu")
+            s = streams[job_i % len(streams)]
+            cols = {}
+            masks = {}
+            with torch.cuda.stream(s):
+                for name, col in piece.columns.items():
+                    if isinstance(col, StringColumn):
+                        cols[name] = col.to("cpu")
+                        continue
+                    nb = col.numel() * col.element_size()
+                    bufp = _pinned_get(nb)
+                    host = bufp[:nb].view(col.dtype)
+                    host.copy_(col, non_blocking=True)
+                    cols[name] = host
+                    held.append(bufp)
+                for name, m in piece.masks.items():
+                    masks[name] = m.to("cpu")
+            s.synchronize()
+            piece = ColumnBatch(cols, masks)
         write_batch_parquet(piece, out)
+        for bufp in held:
+            _pinned_put(bufp)
         return out
 
     if len(jobs) <= 2:
-        return [write_one(j) for j in jobs]
+        return [write_one(i) for i in range(len(jobs))]
     with ThreadPoolExecutor(max_workers=16) as pool:
-        return list(pool.map(write_one, jobs))
+        return list(pool.map(write_one, range(len(jobs))))
 
 
 register_derived_dataset(COVERING_INDEX_TYPE, CoveringIndex)

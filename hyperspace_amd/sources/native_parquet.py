@@ -23,7 +23,7 @@ from __future__ import annotations
 
 import os
 import struct
-from typing import Dict, List, Optional, Tuple
+from typing import Any, Dict, List, Optional, Tuple
 
 import numpy as np
 
@@ -255,26 +255,29 @@ def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str,
             m = None  # all-valid mask: write the cheaper REQUIRED form
         col_masks.append(m)
 
-    chunks: List[bytes] = []
+    chunks: List[Any] = []  # bytes | memoryview (os.write takes both)
     col_meta: List[Tuple] = []
     offset = 4  # after magic
     for name, arr, mask in zip(names, arrays, col_masks):
         if mask is None:
-            payload = arr.tobytes()  # PLAIN little-endian
+            # zero-copy: the PLAIN payload IS the little-endian column
+            # buffer; write straight from the array's memory
+            payload = arr.data.cast("B")
             null_count = 0
             valid = arr
         else:
             valid = arr[mask]
             payload = _def_levels_payload(mask) + valid.tobytes()
             null_count = int(num_rows - len(valid))
-        header = _page_header(num_rows, len(payload), mask is not None)
+        nbytes = len(payload)
+        header = _page_header(num_rows, nbytes, mask is not None)
         mn, mx = _statistics(valid) if len(valid) else (b"", b"")
         col_meta.append((name, _NP_TO_PARQUET[arr.dtype], offset,
-                         len(header) + len(payload), num_rows, mn, mx,
+                         len(header) + nbytes, num_rows, mn, mx,
                          null_count, mask is not None))
         chunks.append(header)
         chunks.append(payload)
-        offset += len(header) + len(payload)
+        offset += len(header) + nbytes
 
     # FileMetaData
     w = TWriter()
