@@ -235,6 +235,40 @@ def _def_levels_payload(mask: np.ndarray) -> bytes:
     return struct.pack("<I", len(body)) + body
 
 
+# Layout cache: (path, size, mtime_ns) -> (num_rows, [ColumnChunkLayout]).
+# The native writer knows the page layout it just wrote, so the first
+# device read of a freshly built index skips the footer + page-header
+# parse entirely (the reference relies on Spark's FileIndex/footer
+# caching for the same effect).  Bounded FIFO.
+_LAYOUT_CACHE: "Dict[str, Tuple[Tuple[int, int], int, list]]" = {}
+_LAYOUT_CACHE_MAX = 16384
+
+
+def layout_cache_get(path: str):
+    """(num_rows, layouts) if the cached entry matches the file's current
+    size+mtime, else None."""
+    ent = _LAYOUT_CACHE.get(path)
+    if ent is None:
+        return None
+    try:
+        st = os.stat(path)
+    except OSError:
+        return None
+    if ent[0] != (st.st_size, st.st_mtime_ns):
+        _LAYOUT_CACHE.pop(path, None)
+        return None
+    return ent[1], ent[2]
+
+
+def _layout_cache_put(path: str, num_rows: int, layouts: list) -> None:
+    if len(_LAYOUT_CACHE) >= _LAYOUT_CACHE_MAX:
+        # FIFO evict ~1/8th
+        for k in list(_LAYOUT_CACHE)[:_LAYOUT_CACHE_MAX // 8]:
+            _LAYOUT_CACHE.pop(k, None)
+    st = os.stat(path)
+    _LAYOUT_CACHE[path] = ((st.st_size, st.st_mtime_ns), num_rows, layouts)
+
+
 def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str,
                          masks: "Optional[Dict[str, np.ndarray]]" = None
                          ) -> Tuple[int, int]:
@@ -257,8 +291,10 @@ def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str,
 
     chunks: List[Any] = []  # bytes | memoryview (os.write takes both)
     col_meta: List[Tuple] = []
+    layouts: List[ColumnChunkLayout] = []
     offset = 4  # after magic
     for name, arr, mask in zip(names, arrays, col_masks):
+        lvl_len = 0
         if mask is None:
             # zero-copy: the PLAIN payload IS the little-endian column
             # buffer; write straight from the array's memory
@@ -267,7 +303,9 @@ def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str,
             valid = arr
         else:
             valid = arr[mask]
-            payload = _def_levels_payload(mask) + valid.tobytes()
+            levels = _def_levels_payload(mask)
+            lvl_len = len(levels)
+            payload = levels + valid.tobytes()
             null_count = int(num_rows - len(valid))
         nbytes = len(payload)
         header = _page_header(num_rows, nbytes, mask is not None)
@@ -275,6 +313,10 @@ def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str,
         col_meta.append((name, _NP_TO_PARQUET[arr.dtype], offset,
                          len(header) + nbytes, num_rows, mn, mx,
                          null_count, mask is not None))
+        layouts.append(ColumnChunkLayout(
+            name, arr.dtype,
+            [("plain", offset + len(header) + lvl_len, num_rows)],
+            num_rows, "plain", None, [mask]))
         chunks.append(header)
         chunks.append(payload)
         offset += len(header) + nbytes
@@ -349,6 +391,7 @@ def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str,
         f.write(footer)
         f.write(struct.pack("<I", len(footer)))
         f.write(MAGIC)
+    _layout_cache_put(path, num_rows, layouts)
     st = os.stat(path)
     return st.st_size, int(st.st_mtime * 1000)
 
@@ -428,20 +471,26 @@ def _decode_defs(data, off: int, length: int, n: int
 
 def read_native_layout(path: str,
                        columns: Optional[List[str]] = None,
-                       data: Optional[bytes] = None
+                       data: Optional[bytes] = None,
+                       meta=None, pf_schema=None
                        ) -> Optional[Tuple[bytes, List[ColumnChunkLayout]]]:
     """If ``path`` decodes natively (uncompressed PLAIN pages of numeric
     columns — our writer's files and pyarrow's NONE/PLAIN files), return
     (raw file bytes, per-column page layouts); else None (caller falls
     back to pyarrow).  ``data`` can supply pre-read file bytes (any
-    buffer protocol object) to avoid a second disk read."""
-    import pyarrow.parquet as pq
-    try:
-        pf = pq.ParquetFile(path)
-        md = pf.metadata
-        pf_schema = pf.schema
-    except Exception:  # noqa: BLE001
-        return None
+    buffer protocol object) to avoid a second disk read; ``meta`` /
+    ``pf_schema`` supply already-parsed footer metadata to avoid a
+    second footer parse."""
+    if meta is None or pf_schema is None:
+        import pyarrow.parquet as pq
+        try:
+            pf = pq.ParquetFile(path)
+            md = pf.metadata
+            pf_schema = pf.schema
+        except Exception:  # noqa: BLE001
+            return None
+    else:
+        md = meta
     if data is None:
         with open(path, "rb") as f:
             data = f.read()

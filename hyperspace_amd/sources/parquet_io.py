@@ -176,7 +176,7 @@ def read_files_batch_device(paths: List[str], device,
     import torch as _torch
     from ..execution.columnar import ColumnBatch
     from ..ops import native as native_ext
-    from .native_parquet import read_native_layout
+    from .native_parquet import layout_cache_get, read_native_layout
 
     # Fully-overlapped pipeline: footer metadata (cheap) sizes the output
     # tensors up front; worker threads then read each file into a pinned
@@ -194,50 +194,77 @@ def read_files_batch_device(paths: List[str], device,
         batch, rc = read_files_batch(paths, columns)
         return batch.to(device), rc
 
+    want = {c.lower() for c in columns} if columns is not None else None
+
+    # write-time layout cache: every file our native writer produced in
+    # this process skips the footer + page-header parse (the dominant
+    # GIL-bound cost of a ~200-file cold load)
+    cached = []
+    for p in paths:
+        ent = layout_cache_get(p)
+        if ent is None:
+            cached = None
+            break
+        cached.append(ent)
+
     metas = []
     schemas = []
-    for p in paths:
-        try:
-            pf = pq.ParquetFile(p)
-        except Exception:  # noqa: BLE001
-            return fallback()
-        metas.append(pf.metadata)
-        schemas.append(pf.schema)
+    per_file_layouts = None
+    if cached is not None:
+        row_counts = [ent[0] for ent in cached]
+        lay0 = cached[0][1]
+        names = [c.name for c in lay0
+                 if want is None or c.name.lower() in want]
+        dtypes = {c.name: c.np_dtype for c in lay0}
+        nullable_cols = {c.name for ent in cached for c in ent[1]
+                         if c.has_nulls}
+        per_file_layouts = [
+            [c for c in ent[1]
+             if want is None or c.name.lower() in want]
+            for ent in cached]
+    else:
+        for p in paths:
+            try:
+                pf = pq.ParquetFile(p)
+            except Exception:  # noqa: BLE001
+                return fallback()
+            metas.append(pf.metadata)
+            schemas.append(pf.schema)
 
-    # column nullability from chunk statistics (sizes the preallocated
-    # masks); an OPTIONAL chunk without statistics could hide nulls ->
-    # the host path decides instead
-    nullable_cols = set()
-    for md, sch in zip(metas, schemas):
-        for rg_i in range(md.num_row_groups):
-            rg = md.row_group(rg_i)
-            for i in range(rg.num_columns):
-                col = rg.column(i)
-                st = col.statistics
-                if st is None or st.null_count is None:
-                    if sch.column(i).max_definition_level > 0:
-                        return fallback()
-                elif st.null_count > 0:
-                    nullable_cols.add(col.path_in_schema)
+        # column nullability from chunk statistics (sizes the
+        # preallocated masks); an OPTIONAL chunk without statistics
+        # could hide nulls -> the host path decides instead
+        nullable_cols = set()
+        for md, sch in zip(metas, schemas):
+            for rg_i in range(md.num_row_groups):
+                rg = md.row_group(rg_i)
+                for i in range(rg.num_columns):
+                    col = rg.column(i)
+                    st = col.statistics
+                    if st is None or st.null_count is None:
+                        if sch.column(i).max_definition_level > 0:
+                            return fallback()
+                    elif st.null_count > 0:
+                        nullable_cols.add(col.path_in_schema)
 
-    want = {c.lower() for c in columns} if columns is not None else None
-    rg0 = metas[0].row_group(0)
-    names = []
-    dtypes = {}
-    for i in range(rg0.num_columns):
-        col = rg0.column(i)
-        if want is not None and col.path_in_schema.lower() not in want:
-            continue
-        npd = _PHYS_TO_NP.get(col.physical_type)
-        if npd is None:
-            return fallback()
-        names.append(col.path_in_schema)
-        dtypes[col.path_in_schema] = npd
+        rg0 = metas[0].row_group(0)
+        names = []
+        dtypes = {}
+        for i in range(rg0.num_columns):
+            col = rg0.column(i)
+            if want is not None and \
+                    col.path_in_schema.lower() not in want:
+                continue
+            npd = _PHYS_TO_NP.get(col.physical_type)
+            if npd is None:
+                return fallback()
+            names.append(col.path_in_schema)
+            dtypes[col.path_in_schema] = npd
+        row_counts = [md.num_rows for md in metas]
     if columns is not None:
         order = {c.lower(): i for i, c in enumerate(columns)}
         names.sort(key=lambda n: order.get(n.lower(), 99))
 
-    row_counts = [md.num_rows for md in metas]
     file_base = np.concatenate([[0], np.cumsum(row_counts)[:-1]])
     total_rows = int(sum(row_counts))
 
@@ -265,20 +292,25 @@ def read_files_batch_device(paths: List[str], device,
         view = memoryview(buf.numpy())
         with open(p, "rb", buffering=0) as f:
             f.readinto(view[:size])
-        lay = read_native_layout(p, columns, data=view[:size])
-        if lay is None:
-            _pinned_put(buf)
-            return None
+        if per_file_layouts is not None:
+            chunks = per_file_layouts[i]
+        else:
+            lay = read_native_layout(p, columns, data=view[:size],
+                                     meta=metas[i], pf_schema=schemas[i])
+            if lay is None:
+                _pinned_put(buf)
+                return None
+            chunks = lay[1]
         stream = streams[i % n_streams]
         with _torch.cuda.stream(stream):
-            return _decode_on_stream(i, buf, size, lay)
+            return _decode_on_stream(i, buf, size, chunks)
 
-    def _decode_on_stream(i, buf, size, lay):
+    def _decode_on_stream(i, buf, size, chunks):
         # upload only the file's bytes (+4B decode slack), not the whole
         # pooled size class
         dev_bytes = buf[:size + 4].to(device, non_blocking=True)
         cursors = {n: int(file_base[i]) for n in names}
-        for c in lay[1]:
+        for c in chunks:
             itemsize = c.np_dtype.itemsize
             written = cursors[c.name]
             dict_vals = None
@@ -341,7 +373,7 @@ def read_files_batch_device(paths: List[str], device,
         return buf
 
     if len(paths) > 2:
-        with ThreadPoolExecutor(max_workers=8) as pool:
+        with ThreadPoolExecutor(max_workers=16) as pool:
             bufs = list(pool.map(load_decode, range(len(paths))))
     else:
         bufs = [load_decode(i) for i in range(len(paths))]
