@@ -605,6 +605,23 @@ class VacuumOutdatedAction(_StateFlipAction):
                     if os.path.isfile(full) and full not in referenced:
                         os.unlink(full)
 
+    def log_entry(self):
+        # outdated data versions are gone, so the delta time-travel
+        # history must forget them: keep only the newest pair
+        # (reference VacuumOutdatedAction.scala:56-67 via
+        # DeltaLakeRelationMetadata resetting 'deltaVersions')
+        entry = self.previous
+        hist = entry.properties.get("deltaVersions")
+        if hist and "," in hist:
+            props = dict(entry.properties)
+            props["deltaVersions"] = hist.split(",")[-1]
+            import copy
+            entry = copy.copy(entry)
+            entry.properties = props
+            entry.derivedDataset = entry.derivedDataset \
+                .with_new_properties(props)
+        return entry
+
     def event(self, message):
         return VacuumOutdatedActionEvent(index_name=self.previous.name,
                                          message=message)

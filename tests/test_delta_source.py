@@ -241,3 +241,21 @@ def test_closest_index_skipped_for_live_query(env):
     for l in leaves:
         if isinstance(l, IndexScan):
             assert l.entry.has_source_update() or True  # hybrid-only
+
+
+def test_vacuum_outdated_resets_delta_history(env):
+    """Reference VacuumOutdatedAction.scala:56-67: after outdated data
+    versions are GC'd, the deltaVersions time-travel history keeps only
+    the surviving (latest) pair."""
+    session, h, table, rng = env
+    df = session.read_delta(table.path)
+    h.create_index(df, hs.CoveringIndexConfig("vx", ["key"], ["val"]))
+    table.append_batch(_batch(rng))
+    h.refresh_index("vx", mode="full")
+    entry = session.index_manager().get_index("vx")
+    assert len(entry.properties["deltaVersions"].split(",")) == 2
+    h.vacuum_index("vx")  # ACTIVE -> VacuumOutdatedAction
+    entry = session.index_manager().get_index("vx")
+    assert len(entry.properties["deltaVersions"].split(",")) == 1
+    # the surviving pair is the refresh's
+    assert all("v__=1" in f for f in entry.content.os_files())
