@@ -156,3 +156,30 @@ def test_zorder_quantile_mode_on_skewed_data(tmp_path, monkeypatch):
         return ex.stats.scanned_files
 
     assert scanned(True) <= scanned(False)
+
+
+def test_zorder_with_string_column(tmp_path, monkeypatch):
+    """String z-order columns: dictionary codes are order-preserving, so
+    the normalize-then-interleave path covers them (reference
+    ZOrderField string encoding); file pruning stays numeric-only."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "idx"))
+    rng = np.random.default_rng(13)
+    d = tmp_path / "zs"
+    d.mkdir()
+    cats = np.array(["aa", "bb", "cc", "dd"])[rng.integers(0, 4, 8000)]
+    key = rng.integers(0, 1000, 8000)
+    pq.write_table(pa.table({"cat": cats.tolist(), "key": key,
+                             "val": rng.random(8000)}),
+                   str(d / "part-0.parquet"))
+    session = hs.HyperspaceSession(device="cpu")
+    h = hs.Hyperspace(session)
+    df = session.read_parquet(str(d))
+    h.create_index(df, hs.ZOrderCoveringIndexConfig(
+        "zs", ["cat", "key"], ["val"]))
+    session.enable_hyperspace()
+    out = df.filter("cat = 'bb'").select("cat", "key", "val").collect()
+    assert out.num_rows == int((cats == "bb").sum())
+    out2 = df.filter("key >= 900").select("key", "val").collect()
+    assert out2.num_rows == int((key >= 900).sum())
