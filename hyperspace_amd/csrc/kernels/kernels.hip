@@ -616,17 +616,26 @@ __device__ __forceinline__ void mj_tile_narrow(
   }
 }
 
+// LDS staging capacity for the narrowed right window (u64 keys): most
+// tiles map into a tiny window (avg = n_right * MJ_TILE / n_left), so
+// per-row probes hit LDS instead of chasing L2/HBM latency
+#define MJ_LDS_WIN 2048
+
 // fill this thread's MJ_RPT consecutive rows: one binary search for the
 // first row, forward-monotone searches for the rest (keys ascend, so
-// each row's range starts at or after the previous row's)
+// each row's range starts at or after the previous row's).
+// ``rbase``/``r_base_off``: the searched array and the global offset of
+// its element 0 — (lwin, tile_lo) when the window is LDS-staged, else
+// (rkeys, 0); a_out entries are global right indices either way.
 __device__ __forceinline__ int64_t mj_thread_rows(
-    const uint64_t* __restrict__ lkeys, const uint64_t* __restrict__ rkeys,
-    const int64_t* __restrict__ lseg, const int64_t* __restrict__ rseg,
-    int64_t n_left, int64_t n_seg, int64_t first, int64_t tile_seg,
-    int64_t tile_lo, int64_t tile_hi, int64_t* a_out, int64_t* c_out) {
+    const uint64_t* __restrict__ lkeys, const uint64_t* __restrict__ rbase,
+    int64_t r_base_off, const int64_t* __restrict__ lseg,
+    const int64_t* __restrict__ rseg, int64_t n_left, int64_t n_seg,
+    int64_t first, int64_t tile_seg, int64_t tile_lo, int64_t tile_hi,
+    int64_t* a_out, int64_t* c_out) {
   int64_t total = 0;
-  int64_t a_prev = tile_lo;
-  int64_t r1 = tile_hi;
+  int64_t a_prev = tile_lo - r_base_off;
+  int64_t r1 = tile_hi - r_base_off;
   for (int k = 0; k < MJ_RPT; k++) {
     int64_t i = first + k;
     if (i >= n_left || (tile_seg >= 0 && i >= lseg[tile_seg + 1])) {
@@ -642,14 +651,29 @@ __device__ __forceinline__ int64_t mj_thread_rows(
       r1 = rseg[s + 1];
     }
     uint64_t key = lkeys[i];
-    int64_t a = lower_bound_u64(rkeys, a_prev, r1, key);
-    int64_t b = upper_bound_u64(rkeys, a, r1, key);
-    a_out[k] = a;
+    int64_t a = lower_bound_u64(rbase, a_prev, r1, key);
+    int64_t b = upper_bound_u64(rbase, a, r1, key);
+    a_out[k] = a + r_base_off;
     c_out[k] = b - a;
     total += b - a;
     a_prev = a;
   }
   return total;
+}
+
+// cooperative LDS stage of the narrowed window; returns true when the
+// per-row searches should probe lwin (tile fits one segment and the
+// window fits LDS)
+__device__ __forceinline__ bool mj_stage_window(
+    const uint64_t* __restrict__ rkeys, uint64_t* lwin, int64_t tile_seg,
+    int64_t tile_lo, int64_t tile_hi) {
+  bool use_lds = tile_seg >= 0 && tile_hi - tile_lo <= MJ_LDS_WIN;
+  if (use_lds) {
+    for (int64_t j = threadIdx.x; j < tile_hi - tile_lo; j += blockDim.x)
+      lwin[j] = rkeys[tile_lo + j];
+  }
+  __syncthreads();
+  return use_lds;
 }
 
 __global__ void k_mj_tile_count(const uint64_t* __restrict__ lkeys,
@@ -660,6 +684,7 @@ __global__ void k_mj_tile_count(const uint64_t* __restrict__ lkeys,
                                 int64_t* __restrict__ tile_counts) {
   __shared__ int64_t tile_seg, tile_lo, tile_hi;
   __shared__ int64_t red[THREADS];
+  __shared__ uint64_t lwin[MJ_LDS_WIN];
   int64_t n_tiles = cdiv(n_left, MJ_TILE);
   for (int64_t tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
     int64_t base = tile * MJ_TILE;
@@ -667,11 +692,13 @@ __global__ void k_mj_tile_count(const uint64_t* __restrict__ lkeys,
       mj_tile_narrow(lkeys, rkeys, lseg, rseg, n_left, n_seg, base,
                      &tile_seg, &tile_lo, &tile_hi);
     __syncthreads();
+    bool use_lds = mj_stage_window(rkeys, lwin, tile_seg, tile_lo,
+                                   tile_hi);
     int64_t a_arr[MJ_RPT], c_arr[MJ_RPT];
     red[threadIdx.x] = mj_thread_rows(
-        lkeys, rkeys, lseg, rseg, n_left, n_seg,
-        base + (int64_t)threadIdx.x * MJ_RPT, tile_seg, tile_lo, tile_hi,
-        a_arr, c_arr);
+        lkeys, use_lds ? lwin : rkeys, use_lds ? tile_lo : 0, lseg, rseg,
+        n_left, n_seg, base + (int64_t)threadIdx.x * MJ_RPT, tile_seg,
+        tile_lo, tile_hi, a_arr, c_arr);
     __syncthreads();
     for (int off = blockDim.x / 2; off > 0; off >>= 1) {
       if ((int)threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
@@ -692,6 +719,7 @@ __global__ void k_mj_tile_emit(const uint64_t* __restrict__ lkeys,
                                int64_t* __restrict__ out_r) {
   __shared__ int64_t tile_seg, tile_lo, tile_hi;
   __shared__ int64_t pfx[THREADS];
+  __shared__ uint64_t lwin[MJ_LDS_WIN];
   int64_t n_tiles = cdiv(n_left, MJ_TILE);
   for (int64_t tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
     int64_t base = tile * MJ_TILE;
@@ -699,11 +727,13 @@ __global__ void k_mj_tile_emit(const uint64_t* __restrict__ lkeys,
       mj_tile_narrow(lkeys, rkeys, lseg, rseg, n_left, n_seg, base,
                      &tile_seg, &tile_lo, &tile_hi);
     __syncthreads();
+    bool use_lds = mj_stage_window(rkeys, lwin, tile_seg, tile_lo,
+                                   tile_hi);
     int64_t a_arr[MJ_RPT], c_arr[MJ_RPT];
     int64_t my_total = mj_thread_rows(
-        lkeys, rkeys, lseg, rseg, n_left, n_seg,
-        base + (int64_t)threadIdx.x * MJ_RPT, tile_seg, tile_lo, tile_hi,
-        a_arr, c_arr);
+        lkeys, use_lds ? lwin : rkeys, use_lds ? tile_lo : 0, lseg, rseg,
+        n_left, n_seg, base + (int64_t)threadIdx.x * MJ_RPT, tile_seg,
+        tile_lo, tile_hi, a_arr, c_arr);
     // exclusive intra-tile prefix over per-thread totals (Hillis-Steele)
     pfx[threadIdx.x] = my_total;
     __syncthreads();
