@@ -28,7 +28,7 @@ _BUCKET_RE = re.compile(r".*_(\d+)(?:\.\w+)*\.parquet$")
 # through a size-classed pool (power-of-two classes, bounded total).
 _PINNED_POOL: Dict[int, List["torch.Tensor"]] = {}
 _PINNED_POOL_BYTES = 0
-_PINNED_POOL_MAX = 16 << 30
+_PINNED_POOL_MAX = 8 << 30  # per process; 8 ranks per node share host RAM
 import threading as _threading
 
 _pool_lock = _threading.Lock()
