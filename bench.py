@@ -35,7 +35,7 @@ def main():
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--gb-per-gpu", type=float,
-                    default=float(os.environ.get("BENCH_GB_PER_GPU", 4.0)))
+                    default=float(os.environ.get("BENCH_GB_PER_GPU", 8.0)))
     ap.add_argument("--num-buckets", type=int, default=200)
     ap.add_argument("--device", default=None)
     ap.add_argument("--workdir", default=None)
