@@ -7,11 +7,10 @@ createIndex, explain and whyNot.
 
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional, Union
+from typing import List, Optional, Union
 
 from .plan.expr import Expr, col as _col, parse_predicate
 from .plan.nodes import Filter, Join, LogicalPlan, Project
-from .exceptions import HyperspaceException
 
 
 class DataFrame:

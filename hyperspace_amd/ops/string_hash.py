@@ -11,7 +11,7 @@ per-batch code assignment.
 
 from __future__ import annotations
 
-from typing import List, Union
+from typing import List
 
 import torch
 

@@ -21,7 +21,6 @@ import torch.distributed as dist
 
 from .. import ops
 from ..execution.columnar import ColumnBatch, StringColumn
-from ..exceptions import HyperspaceException
 
 
 def exchange_by_bucket(batch: ColumnBatch, bucket_ids: torch.Tensor,
