@@ -98,6 +98,9 @@ def main():
 
     session = hs.HyperspaceSession(device=device)
     session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, args.num_buckets)
+    if os.environ.get("BENCH_GROUP_GB"):
+        session.conf.set("spark.hyperspace.index.build.groupBytes",
+                         int(float(os.environ["BENCH_GROUP_GB"]) * (1 << 30)))
     session.conf.set(hs.IndexConstants.INDEX_FILTER_RULE_USE_BUCKET_SPEC,
                      True)
     h = hs.Hyperspace(session)

@@ -15,6 +15,7 @@ Operator -> data-plane mapping:
 
 from __future__ import annotations
 
+import os
 from typing import Dict, List, Optional, Tuple
 
 import torch
@@ -234,6 +235,8 @@ class Executor:
                 files_per_bucket.append((b, len(paths)))
                 ordered_paths.extend(paths)
             self.stats.scanned_files += len(ordered_paths)
+            import time as _time
+            _t0 = _time.perf_counter()
             if not ordered_paths:
                 batch = ColumnBatch({c: torch.empty(0) for c in read_cols})
                 row_counts = []
@@ -243,6 +246,11 @@ class Executor:
             else:
                 batch, row_counts = read_files_batch(
                     ordered_paths, columns=read_cols)
+            if os.environ.get("HS_TIMING"):
+                import sys as _sys
+                print(f"[hs-timing] index scan read: {len(ordered_paths)}"
+                      f" files {_time.perf_counter() - _t0:.3f}s",
+                      file=_sys.stderr)
             seg_counts = torch.zeros(num_buckets + 1, dtype=torch.int64)
             fi = 0
             for b, nfiles in files_per_bucket:

@@ -25,23 +25,39 @@ _BUCKET_RE = re.compile(r".*_(\d+)(?:\.\w+)*\.parquet$")
 
 # Pinned host staging buffers are expensive to allocate (~60 ms/GB —
 # comparable to the copy itself), so the device read path recycles them
-# through a size-classed pool (power-of-two classes, bounded total).
+# through a size-classed pool.  Classes are 8 MiB-granular above 8 MiB
+# (pow2 wastes up to 2x of the budget on ~40-60 MB parquet files) and
+# power-of-two below; total bounded per process (8 ranks share a node's
+# host RAM).
 _PINNED_POOL: Dict[int, List["torch.Tensor"]] = {}
 _PINNED_POOL_BYTES = 0
-_PINNED_POOL_MAX = 8 << 30  # per process; 8 ranks per node share host RAM
+_PINNED_POOL_MAX = 16 << 30
 import threading as _threading
 
 _pool_lock = _threading.Lock()
 
 
+def _pinned_size_class(nbytes: int) -> int:
+    if nbytes > (8 << 20):
+        gran = 8 << 20
+        return (nbytes + gran - 1) // gran * gran
+    return 1 << max(12, (nbytes - 1).bit_length())
+
+
 def _pinned_get(nbytes: int) -> "torch.Tensor":
     global _PINNED_POOL_BYTES
-    size = 1 << max(12, (nbytes - 1).bit_length())
+    size = _pinned_size_class(nbytes)
     with _pool_lock:
-        lst = _PINNED_POOL.get(size)
-        if lst:
-            _PINNED_POOL_BYTES -= size
-            return lst.pop()
+        # exact class first, then up to 2x oversized (a 48 MB index-file
+        # read can reuse a 64 MB source-staging buffer instead of paying
+        # a fresh pinned allocation)
+        cls = size
+        while cls <= 2 * size:
+            lst = _PINNED_POOL.get(cls)
+            if lst:
+                _PINNED_POOL_BYTES -= cls
+                return lst.pop()
+            cls += (8 << 20) if cls >= (8 << 20) else cls
     return torch.empty(size, dtype=torch.uint8, pin_memory=True)
 
 
