@@ -367,6 +367,15 @@ class Executor:
                 batch, seg = self._exec_index_scan(child, eq_prune=eq)
                 self.stats.record("Filter")
                 return self._apply_predicate(batch, plan.condition), None
+        # hive partition pruning: partition-only conjuncts drop files on
+        # metadata alone (Spark's PartitioningAwareFileIndex pruning)
+        if isinstance(child, Scan) and child.file_subset is None:
+            prune = getattr(child.relation, "prune_partitions", None)
+            kept = prune(plan.condition) if prune is not None else None
+            if kept is not None:
+                batch = self._exec_scan(child, file_subset=kept)
+                self.stats.record("Filter(partition-pruned)")
+                return self._apply_predicate(batch, plan.condition), None
         batch, seg = self._exec(child)
         self.stats.record("Filter")
         return self._apply_predicate(batch, plan.condition), None

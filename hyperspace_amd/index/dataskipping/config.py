@@ -2,8 +2,9 @@
 
 Reference: index/dataskipping/DataSkippingIndexConfig.scala —
 (name, sketches…); rejects duplicate sketches (:86-94); auto-adds a
-PartitionSketch when autoPartitionSketch (not applicable here: the
-default parquet source has no hive partitioning in v0).
+PartitionSketch over each hive partition column when
+autoPartitionSketch is on (:56-84, default true) so disjunctions like
+``A = 1 OR part = 1`` stay convertible.
 """
 
 from __future__ import annotations
@@ -14,7 +15,7 @@ import torch
 
 from ..base import IndexConfigTrait, IndexerContext
 from .index import DataSkippingIndex
-from .sketches import Sketch
+from .sketches import PartitionSketch, Sketch
 from ...exceptions import HyperspaceException
 from ...log.entry import Schema
 from ...utils.resolver import resolve_all
@@ -51,7 +52,25 @@ class DataSkippingIndexConfig(IndexConfigTrait):
                 "createIndex requires a single file-source relation plan")
         scan: Scan = leaves[0]
         schema = scan.relation.schema
-        cols = resolve_all(schema.field_names(), self.referenced_columns())
+
+        # autoPartitionSketch: cover every hive partition column
+        sketches = list(self.sketches)
+        auto = str(ctx.session.conf.get(
+            "spark.hyperspace.index.dataSkipping.autoPartitionSketch",
+            True)).lower() != "false"
+        pschema_fn = getattr(scan.relation, "partition_schema", None)
+        if auto and pschema_fn is not None:
+            covered = {s.expr.lower() for s in sketches
+                       if isinstance(s, PartitionSketch)}
+            for f in pschema_fn().fields:
+                # numeric partition columns only: the first-value sketch
+                # compares tensors at query time (string partitions still
+                # prune via the executor's metadata partition pruning)
+                if f.name.lower() not in covered and \
+                        f.type in ("long", "double"):
+                    sketches.append(PartitionSketch(f.name))
+        cols = resolve_all(schema.field_names(),
+                           sorted({s.expr for s in sketches}))
 
         # pre-assign ids deterministically (shared with covering build)
         files = sorted(scan.relation.all_files(), key=lambda f: f.name)
@@ -60,9 +79,10 @@ class DataSkippingIndexConfig(IndexConfigTrait):
 
         ex = Executor(ctx.session)
         # per-file segmentation: read each column + per-file row counts
-        from ...sources.parquet_io import read_files_batch
+        # through the relation seam (partition columns materialize there)
         paths = [f.name for f in files]
-        batch, row_counts = read_files_batch(paths, columns=cols)
+        batch, row_counts = scan.relation.read_files(
+            paths, cols, ctx.session.device)
         if ctx.session.device.type == "cuda":
             batch = batch.to(ctx.session.device)
         seg = torch.zeros(len(paths) + 1, dtype=torch.int64)
@@ -75,7 +95,7 @@ class DataSkippingIndexConfig(IndexConfigTrait):
              for f in files], dtype=torch.int64)
         index_data: Dict[str, torch.Tensor] = {
             IndexConstants.DATA_FILE_NAME_ID_COLUMN: file_ids}
-        for sketch in self.sketches:
+        for sketch in sketches:
             resolved = resolve_all(schema.field_names(), [sketch.expr])[0]
             sketch.expr = resolved
             values = batch.tensor(resolved)
@@ -86,7 +106,7 @@ class DataSkippingIndexConfig(IndexConfigTrait):
         index_schema = Schema([f for f in schema.fields
                                if f.name.lower() in
                                {c.lower() for c in cols}])
-        index = DataSkippingIndex(self.sketches, index_schema,
+        index = DataSkippingIndex(sketches, index_schema,
                                   dict(properties))
         return index, index_data
 

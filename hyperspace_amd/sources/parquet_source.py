@@ -2,18 +2,65 @@
 
 Reference: index/sources/default/ (DefaultFileBasedRelation: signature =
 md5 chain over sorted (path,size,mtime), default/DefaultFileBasedRelation.scala:45-53;
-DataPathFilter excludes files starting with '_' or '.', util/PathUtils.scala).
+DataPathFilter excludes files starting with '_' or '.', util/PathUtils.scala;
+hive-style partition discovery + partition base path,
+default/DefaultFileBasedRelation.scala:75-89).
+
+Hive partitioning: ``root/key=value/...`` directory segments become
+table columns (appended after the file schema).  Partition values
+attach as constant per-file columns at read time, and partition-only
+predicates prune files before any IO (Spark's partition pruning).
 """
 
 from __future__ import annotations
 
 import os
-from typing import Dict, List, Optional
+from typing import Any, Dict, List, Optional, Tuple
 
 from .interfaces import FileBasedRelation, FileBasedSourceProvider
-from ..log.entry import FileInfo, Relation, Schema
+from ..log.entry import FileInfo, Relation, Schema, SchemaField
 from ..utils.hashing import md5_hex
 from ..exceptions import HyperspaceException
+
+
+def partition_values_of(root_paths: List[str], path: str
+                        ) -> "Dict[str, str]":
+    """key=value directory segments between the matching root and the
+    file, in order."""
+    for root in root_paths:
+        r = os.path.abspath(root)
+        if path.startswith(r + os.sep):
+            rel = os.path.relpath(os.path.dirname(path), r)
+            out: Dict[str, str] = {}
+            for seg in rel.split(os.sep):
+                if "=" in seg:
+                    k, v = seg.split("=", 1)
+                    out[k] = v
+            return out
+    return {}
+
+
+def _infer_partition_type(values: List[str]) -> str:
+    try:
+        for v in values:
+            int(v)
+        return "long"
+    except ValueError:
+        pass
+    try:
+        for v in values:
+            float(v)
+        return "double"
+    except ValueError:
+        return "string"
+
+
+def _cast_partition_value(v: str, spark_type: str) -> Any:
+    if spark_type == "long":
+        return int(v)
+    if spark_type == "double":
+        return float(v)
+    return v
 
 
 def _is_data_file(name: str) -> bool:
@@ -67,9 +114,108 @@ class ParquetRelation(FileBasedRelation):
                 raise HyperspaceException(
                     f"No parquet files under {self._root_paths}")
             import pyarrow.parquet as pq
-            self._schema = Schema.from_arrow(
-                pq.read_schema(files[0].name))
+            sch = Schema.from_arrow(pq.read_schema(files[0].name))
+            pschema = self.partition_schema()
+            existing = {f.name.lower() for f in sch.fields}
+            sch.fields.extend(f for f in pschema.fields
+                              if f.name.lower() not in existing)
+            self._schema = sch
         return self._schema
+
+    # -- hive partitioning -------------------------------------------------
+    def partition_schema(self) -> Schema:
+        """Partition columns inferred from key=value path segments."""
+        if getattr(self, "_pschema", None) is None:
+            per_key: Dict[str, List[str]] = {}
+            order: List[str] = []
+            for f in self.all_files():
+                for k, v in partition_values_of(self._root_paths,
+                                                f.name).items():
+                    if k not in per_key:
+                        per_key[k] = []
+                        order.append(k)
+                    per_key[k].append(v)
+            self._pschema = Schema([
+                SchemaField(k, _infer_partition_type(per_key[k]), False)
+                for k in order])
+        return self._pschema
+
+    def partition_values(self, path: str) -> Dict[str, Any]:
+        """Typed partition values for one data file."""
+        pschema = self.partition_schema()
+        raw = partition_values_of(self._root_paths, path)
+        return {f.name: _cast_partition_value(raw[f.name], f.type)
+                for f in pschema.fields if f.name in raw}
+
+    def read_files(self, paths: List[str], columns, device):
+        pschema = self.partition_schema()
+        if not pschema.fields:
+            return super().read_files(paths, columns, device)
+        import numpy as np
+        import torch
+        from ..execution.columnar import ColumnBatch, StringColumn
+        pnames = {f.name.lower(): f for f in pschema.fields}
+        file_cols = None
+        part_wanted = [f for f in pschema.fields]
+        if columns is not None:
+            file_cols = [c for c in columns if c.lower() not in pnames]
+            part_wanted = [pnames[c.lower()] for c in columns
+                           if c.lower() in pnames]
+        if file_cols == []:
+            # partition-only projection: row counts from footers, no IO
+            import pyarrow.parquet as pq
+            row_counts = [pq.ParquetFile(p).metadata.num_rows
+                          for p in paths]
+            batch = ColumnBatch({})
+        else:
+            batch, row_counts = super().read_files(paths, file_cols,
+                                                   device)
+        dev = batch.device if batch.columns else (
+            device if getattr(device, "type", "cpu") == "cuda"
+            else torch.device("cpu"))
+        cols = dict(batch.columns)
+        for f in part_wanted:
+            vals = [self.partition_values(p).get(f.name) for p in paths]
+            if f.type == "string":
+                uniq = sorted(set(vals))
+                code_of = {v: i for i, v in enumerate(uniq)}
+                codes = torch.cat([
+                    torch.full((rc,), code_of[v], dtype=torch.int32)
+                    for v, rc in zip(vals, row_counts)]) \
+                    if row_counts else torch.empty(0, dtype=torch.int32)
+                cols[f.name] = StringColumn(codes.to(dev), uniq)
+            else:
+                dt = torch.int64 if f.type == "long" else torch.float64
+                cols[f.name] = torch.cat([
+                    torch.full((rc,), v, dtype=dt)
+                    for v, rc in zip(vals, row_counts)]).to(dev) \
+                    if row_counts else torch.empty(0, dtype=dt)
+        out = ColumnBatch(cols, dict(batch.masks))
+        if columns is not None:
+            out = out.select(columns)
+        return out, row_counts
+
+    def prune_partitions(self, cond) -> Optional[List[str]]:
+        """Files surviving the partition-column conjuncts of ``cond``,
+        or None when the predicate has no partition conjunct (Spark's
+        partition pruning: evaluated on metadata, before any IO)."""
+        pschema = self.partition_schema()
+        if not pschema.fields:
+            return None
+        from ..plan.expr import split_conjunctive
+        pnames = {f.name.lower() for f in pschema.fields}
+        conjuncts = [c for c in split_conjunctive(cond)
+                     if c.references() and
+                     {r.lower() for r in c.references()} <= pnames]
+        if not conjuncts:
+            return None
+        kept = []
+        for f in self.all_files():
+            vals = {k.lower(): v
+                    for k, v in self.partition_values(f.name).items()}
+            if all(_eval_on_values(c, vals) for c in conjuncts):
+                kept.append(f.name)
+        return kept
 
     def all_files(self) -> List[FileInfo]:
         # re-listed on every call (like Spark's InMemoryFileIndex refresh
@@ -90,6 +236,41 @@ class ParquetRelation(FileBasedRelation):
 
     def refreshed(self) -> "ParquetRelation":
         return ParquetRelation(self._root_paths, self._options, None)
+
+
+def _eval_on_values(e, vals: Dict[str, Any]) -> bool:
+    """Evaluate a predicate over one file's partition values (SQL null
+    semantics: a missing key never matches a comparison)."""
+    from ..plan.expr import (And, BinComp, Col, In, IsNotNull, IsNull,
+                             Lit, Not, Or)
+    if isinstance(e, And):
+        return _eval_on_values(e.left, vals) and \
+            _eval_on_values(e.right, vals)
+    if isinstance(e, Or):
+        return _eval_on_values(e.left, vals) or \
+            _eval_on_values(e.right, vals)
+    if isinstance(e, Not):
+        return not _eval_on_values(e.child, vals)
+    if isinstance(e, IsNull):
+        return vals.get(e.col.name.lower()) is None
+    if isinstance(e, IsNotNull):
+        return vals.get(e.col.name.lower()) is not None
+    if isinstance(e, In):
+        v = vals.get(e.col.name.lower())
+        return v is not None and v in set(e.values)
+    if isinstance(e, BinComp) and isinstance(e.left, Col) and \
+            isinstance(e.right, Lit):
+        v = vals.get(e.left.name.lower())
+        if v is None:
+            return False
+        r = e.right.value
+        if isinstance(v, (int, float)) and isinstance(r, str):
+            return False
+        if isinstance(v, str) and isinstance(r, (int, float)):
+            return False
+        return {"=": v == r, "!=": v != r, "<": v < r, "<=": v <= r,
+                ">": v > r, ">=": v >= r}[e.op]
+    return True  # unknown form: keep the file (safe)
 
 
 class ParquetSourceProvider(FileBasedSourceProvider):
