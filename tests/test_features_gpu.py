@@ -107,3 +107,28 @@ def test_gpu_zorder_string(tmp_path, monkeypatch):
     session.enable_hyperspace()
     out = df.filter("cat = 'bb'").select("cat", "key", "val").collect()
     assert out.num_rows == int((cats == "bb").sum())
+
+
+def test_gpu_partitioned_source(tmp_path, monkeypatch):
+    from hyperspace_amd.execution.executor import Executor
+    session, h = _session(tmp_path, monkeypatch, buckets=4)
+    rng = np.random.default_rng(75)
+    root = tmp_path / "pt"
+    for day in (1, 2):
+        d = root / f"day={day}"
+        d.mkdir(parents=True)
+        pq.write_table(
+            pa.table({"key": rng.integers(0, 300, 100_000),
+                      "val": rng.random(100_000)}),
+            str(d / "part-0.parquet"))
+    df = session.read_parquet(str(root))
+    ex = Executor(session)
+    out = ex.execute(df.filter("day = 2").select("key", "day")
+                     .optimized_plan())
+    assert out.num_rows == 100_000 and ex.stats.scanned_files == 1
+    assert out.tensor("day").is_cuda
+    h.create_index(df, hs.CoveringIndexConfig("gpx", ["key"],
+                                              ["val", "day"]))
+    session.enable_hyperspace()
+    out2 = df.filter("key = 5").select("key", "val", "day").collect()
+    assert (out2.tensor("day") > 0).all()
