@@ -137,8 +137,11 @@ class CoveringIndex(Index):
 
     def _write_group(self, ctx, batch: ColumnBatch, task_id: int,
                      distributed: bool) -> List[str]:
+        import time as _time
+        timing = os.environ.get("HS_TIMING")
         n = self.num_buckets
         from ...ops.string_hash import bucket_hash_keys
+        t0 = _time.perf_counter()
         keys = bucket_hash_keys(batch, self.indexed_columns)
         key_masks = [batch.mask(c) for c in self.indexed_columns]
         bucket_ids = ops.murmur3_bucket(
@@ -149,9 +152,21 @@ class CoveringIndex(Index):
             batch, bucket_ids = exchange_by_bucket(batch, bucket_ids, n)
         if batch.num_rows == 0:
             return []
+        t1 = _time.perf_counter()
         batch, seg = sort_by_bucket_and_keys(
             batch, bucket_ids, self.indexed_columns, n)
-        return write_bucketed(batch, seg, ctx.index_data_path, n, task_id)
+        if timing:
+            import torch as _torch
+            if batch.device.type == "cuda":
+                _torch.cuda.synchronize()
+        t2 = _time.perf_counter()
+        out = write_bucketed(batch, seg, ctx.index_data_path, n, task_id)
+        if timing:
+            import sys as _sys
+            print(f"[hs-timing] build group: hash+exchange {t1-t0:.3f}s "
+                  f"sort {t2-t1:.3f}s write "
+                  f"{_time.perf_counter()-t2:.3f}s", file=_sys.stderr)
+        return out
 
     def _empty_batch(self, ctx) -> ColumnBatch:
         import torch as _torch
