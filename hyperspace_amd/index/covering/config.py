@@ -65,6 +65,16 @@ class CoveringIndexConfig(IndexConfigTrait):
                               self.indexed_columns)
         included = resolve_all(source_schema.field_names(),
                                self.included_columns)
+        # hive partition columns not named in the config join the
+        # included set, so queries projecting them stay covered
+        # (reference CreateActionBase adds missing partition columns to
+        # the covering slice)
+        pschema_fn = getattr(scan.relation, "partition_schema", None)
+        if pschema_fn is not None:
+            named = {c.lower() for c in indexed + included}
+            included = included + [
+                f.name for f in pschema_fn().fields
+                if f.name.lower() not in named]
 
         lineage = (properties.get(IndexConstants.LINEAGE_PROPERTY, "false")
                    .lower() == "true")

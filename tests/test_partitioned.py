@@ -120,3 +120,22 @@ def test_auto_partition_sketch(env):
     leaf = plan.collect_leaves()[0]
     assert leaf.file_subset is not None
     assert len(leaf.file_subset) == 2  # only day=1 files can match
+
+
+def test_covering_auto_includes_partition_columns(env):
+    """Reference CreateActionBase: partition columns missing from the
+    config join the covering slice, so partition projections stay
+    covered."""
+    session, h, root, frames = env
+    df = session.read_parquet(root)
+    h.create_index(df, hs.CoveringIndexConfig("pauto", ["key"], ["val"]))
+    entry = session.index_manager().get_index("pauto")
+    inc = [c.lower() for c in entry.derivedDataset.included_columns]
+    assert "day" in inc and "region" in inc
+    session.enable_hyperspace()
+    q = df.filter("key = 3").select("key", "val", "day", "region")
+    plan = q.optimized_plan()
+    assert any(isinstance(l, IndexScan) for l in plan.collect_leaves())
+    out = q.collect()
+    expected = sum(int((k == 3).sum()) for k, _ in frames.values())
+    assert out.num_rows == expected
