@@ -1,0 +1,105 @@
+"""Randomized maintenance-lifecycle fuzz: a sequence of source appends/
+deletes, refreshes (all modes), optimize, delete/restore/vacuum and
+indexed queries, checked against a pandas oracle after every query.
+Models the reference's long-lived index maintenance flows
+(IndexManagerTests / HybridScanSuite behaviors) under arbitrary
+interleavings."""
+
+import os
+
+import numpy as np
+import pandas as pd
+import pyarrow as pa
+import pyarrow.parquet as pq
+import pytest
+from hypothesis import HealthCheck, given, settings, strategies as st
+
+import hyperspace_amd as hs
+
+N_FILE = 1500
+
+
+def _write_file(d, rng, i):
+    key = rng.integers(0, 200, N_FILE)
+    val = rng.random(N_FILE)
+    p = str(d / f"part-{i:04d}.parquet")
+    pq.write_table(pa.table({"key": key, "val": val}), p)
+    return p, pd.DataFrame({"key": key, "val": val})
+
+
+OPS = st.lists(
+    st.sampled_from(["append", "delete_file", "refresh_inc",
+                     "refresh_full", "refresh_quick", "optimize",
+                     "query", "soft_delete_restore", "query",
+                     "vacuum_outdated"]),
+    min_size=4, max_size=12)
+
+
+@settings(max_examples=15, deadline=None,
+          suppress_health_check=[HealthCheck.function_scoped_fixture])
+@given(ops=OPS, seed=st.integers(0, 10_000))
+def test_lifecycle_fuzz(tmp_path_factory, ops, seed):
+    tmp_path = tmp_path_factory.mktemp("lcf")
+    os.environ["HYPERSPACE_SYSTEM_PATH"] = str(tmp_path / "idx")
+    rng = np.random.default_rng(seed)
+    d = tmp_path / "data"
+    d.mkdir()
+    live = {}
+    for i in range(2):
+        p, frame = _write_file(d, rng, i)
+        live[p] = frame
+    next_i = 2
+
+    session = hs.HyperspaceSession(device="cpu")
+    session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 4)
+    session.conf.set(hs.IndexConstants.INDEX_HYBRID_SCAN_ENABLED, True)
+    session.conf.set(
+        hs.IndexConstants.INDEX_HYBRID_SCAN_APPENDED_RATIO_THRESHOLD, 0.99)
+    session.conf.set(
+        hs.IndexConstants.INDEX_HYBRID_SCAN_DELETED_RATIO_THRESHOLD, 0.99)
+    session.conf.set(hs.IndexConstants.INDEX_LINEAGE_ENABLED, True)
+    h = hs.Hyperspace(session)
+    df = session.read_parquet(str(d))
+    h.create_index(df, hs.CoveringIndexConfig("fz", ["key"], ["val"]))
+    session.enable_hyperspace()
+    deleted_state = False
+
+    def oracle():
+        return pd.concat(live.values(), ignore_index=True)
+
+    for op in ops:
+        if op == "append":
+            p, frame = _write_file(d, rng, next_i)
+            live[p] = frame
+            next_i += 1
+        elif op == "delete_file" and len(live) > 1:
+            p = sorted(live)[0]
+            os.unlink(p)
+            del live[p]
+        elif op == "refresh_inc" and not deleted_state:
+            h.refresh_index("fz", mode="incremental")
+        elif op == "refresh_full" and not deleted_state:
+            h.refresh_index("fz", mode="full")
+        elif op == "refresh_quick" and not deleted_state:
+            h.refresh_index("fz", mode="quick")
+        elif op == "optimize" and not deleted_state:
+            h.optimize_index("fz")
+        elif op == "soft_delete_restore" and not deleted_state:
+            h.delete_index("fz")
+            h.restore_index("fz")
+        elif op == "vacuum_outdated" and not deleted_state:
+            h.vacuum_index("fz")  # ACTIVE -> GC outdated versions
+        elif op == "query":
+            want = oracle()
+            target = int(want.key.iloc[0]) if len(want) else 0
+            got = df.filter(f"key = {target}").select("key", "val") \
+                .collect()
+            exp = want[want.key == target]
+            assert got.num_rows == len(exp), (op, ops)
+            got_rows = sorted(np.round(got.tensor("val").numpy(), 9))
+            exp_rows = sorted(np.round(exp.val.to_numpy(), 9))
+            assert got_rows == exp_rows, (op, ops)
+    # final consistency check
+    want = oracle()
+    got = df.collect()
+    assert got.num_rows == len(want)
