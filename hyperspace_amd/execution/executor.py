@@ -368,11 +368,15 @@ class Executor:
                 self.stats.record("Filter")
                 return self._apply_predicate(batch, plan.condition), None
         # hive partition pruning: partition-only conjuncts drop files on
-        # metadata alone (Spark's PartitioningAwareFileIndex pruning)
-        if isinstance(child, Scan) and child.file_subset is None:
+        # metadata alone (Spark's PartitioningAwareFileIndex pruning);
+        # composes with a data-skipping file_subset by intersection
+        if isinstance(child, Scan):
             prune = getattr(child.relation, "prune_partitions", None)
             kept = prune(plan.condition) if prune is not None else None
             if kept is not None:
+                if child.file_subset is not None:
+                    subset = set(child.file_subset)
+                    kept = [p for p in kept if p in subset]
                 batch = self._exec_scan(child, file_subset=kept)
                 self.stats.record("Filter(partition-pruned)")
                 return self._apply_predicate(batch, plan.condition), None
