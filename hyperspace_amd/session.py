@@ -86,6 +86,15 @@ class HyperspaceSession:
         return DataFrame(self, Scan(TextFormatRelation("json",
                                                        list(paths))))
 
+    def read_text(self, *paths: str):
+        """Line-oriented text files: one string column 'value' (the
+        reference default source's text format)."""
+        from .dataframe import DataFrame
+        from .plan.nodes import Scan
+        from .sources.text_source import TextFormatRelation
+        return DataFrame(self, Scan(TextFormatRelation("text",
+                                                       list(paths))))
+
     def read_avro(self, *paths: str):
         """Avro container files (decoded by the built-in reader,
         sources/avro_io.py)."""

@@ -69,8 +69,22 @@ def _is_data_file(name: str) -> bool:
 
 
 def list_data_files(root_paths: List[str], suffix: str = "") -> List[str]:
-    out: List[str] = []
+    # globbing paths (reference: GLOBBING_PATTERN_KEY /
+    # SparkHadoopUtil.globPathIfNecessary): patterns expand before the
+    # walk and must match at least one path
+    import glob as _glob
+    expanded: List[str] = []
     for root in root_paths:
+        if any(ch in root for ch in "*?["):
+            matches = sorted(_glob.glob(root))
+            if not matches:
+                raise HyperspaceException(
+                    f"Glob pattern matched nothing: {root}")
+            expanded.extend(matches)
+        else:
+            expanded.append(root)
+    out: List[str] = []
+    for root in expanded:
         if os.path.isfile(root):
             if _is_data_file(root):
                 out.append(os.path.abspath(root))

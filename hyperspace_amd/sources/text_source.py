@@ -24,7 +24,7 @@ class TextFormatRelation(FileBasedRelation):
 
     def __init__(self, fmt: str, root_paths: List[str],
                  options: Optional[Dict[str, str]] = None):
-        assert fmt in ("csv", "json", "orc", "avro")
+        assert fmt in ("csv", "json", "orc", "avro", "text")
         self._fmt = fmt
         self._root_paths = [os.path.abspath(p) for p in root_paths]
         self._options = dict(options or {})
@@ -43,7 +43,7 @@ class TextFormatRelation(FileBasedRelation):
         return self._options
 
     def _suffix(self):
-        return "." + self._fmt
+        return ".txt" if self._fmt == "text" else "." + self._fmt
 
     def all_files(self) -> List[FileInfo]:
         infos = []
@@ -62,6 +62,14 @@ class TextFormatRelation(FileBasedRelation):
         if self._fmt == "avro":
             from .avro_io import read_avro
             return read_avro(path)
+        if self._fmt == "text":
+            # one string column "value" per line (Spark text source)
+            import pyarrow as pa
+            with open(path, "r", errors="replace") as f:
+                lines = f.read().split("\n")
+            if lines and lines[-1] == "":
+                lines.pop()
+            return pa.table({"value": lines})
         from pyarrow import json as pa_json
         return pa_json.read_json(path)
 
@@ -112,7 +120,7 @@ class TextFormatSourceProvider(FileBasedSourceProvider):
 
     def from_metadata(self, metadata: Relation):
         if metadata.fileFormat not in ("csv", "json", "orc",
-                                       "avro"):
+                                       "avro", "text"):
             return None
         return TextFormatRelation(metadata.fileFormat, metadata.rootPaths,
                                   metadata.options)

@@ -215,3 +215,39 @@ def test_index_on_avro_source(tmp_path, monkeypatch):
                for l in q.optimized_plan().collect_leaves())
     out = q.collect()
     assert out.num_rows == int((key == 42).sum())
+
+
+def test_text_format_lines(tmp_path, monkeypatch):
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "idx"))
+    import hyperspace_amd as hs
+    d = tmp_path / "logs"
+    d.mkdir()
+    (d / "a.txt").write_text("alpha\nbeta\ngamma\n")
+    (d / "b.txt").write_text("beta\ndelta\n")
+    session = hs.HyperspaceSession(device="cpu")
+    df = session.read_text(str(d))
+    out = df.collect()
+    assert out.num_rows == 5
+    vals = sorted(out.column("value").to_numpy().tolist())
+    assert vals == ["alpha", "beta", "beta", "delta", "gamma"]
+    got = df.filter("value = 'beta'").collect()
+    assert got.num_rows == 2
+
+
+def test_glob_paths(tmp_path):
+    import numpy as np
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    import hyperspace_amd as hs
+    from hyperspace_amd.exceptions import HyperspaceException
+    rng = np.random.default_rng(5)
+    for sub in ("d1", "d2", "other"):
+        d = tmp_path / sub
+        d.mkdir()
+        pq.write_table(pa.table({"key": rng.integers(0, 10, 100)}),
+                       str(d / "part-0.parquet"))
+    session = hs.HyperspaceSession(device="cpu")
+    df = session.read_parquet(str(tmp_path / "d*"))
+    assert df.collect().num_rows == 200  # d1 + d2, not "other"
+    with pytest.raises(HyperspaceException, match="matched nothing"):
+        session.read_parquet(str(tmp_path / "zz*")).collect()
