@@ -69,7 +69,12 @@ class IndexLogManager:
         if latest is None:
             return None
         for log_id in range(latest, -1, -1):
-            entry = self.get_log(log_id)
+            try:
+                entry = self.get_log(log_id)
+            except (json.JSONDecodeError, KeyError, ValueError):
+                # a torn/corrupt entry never wins; keep scanning back —
+                # earlier stable entries are immutable once written
+                continue
             if entry is not None and entry.state in States.STABLE_STATES:
                 return entry
         return None

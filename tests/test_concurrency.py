@@ -109,3 +109,31 @@ def test_minmax_analysis_output(env, tmp_path):
     out = analyze(df, ["key"])
     assert "column: key" in out
     assert "avg files per point lookup" in out
+
+
+def test_corrupt_latest_stable_scan_back(env, tmp_path):
+    """A torn latestStable copy falls back to the scan-back path
+    (reference getLatestStableLog semantics)."""
+    session, h, df = env
+    h.create_index(df, hs.CoveringIndexConfig("cx", ["key"], ["val"]))
+    log_dir = os.path.join(str(tmp_path / "indexes"), "cx",
+                           "_hyperspace_log")
+    with open(os.path.join(log_dir, "latestStable"), "w") as f:
+        f.write("{ torn json")
+    entry = session.index_manager().get_index("cx")
+    assert entry is not None and entry.state == States.ACTIVE
+    session.enable_hyperspace()
+    assert df.filter("key = 5").collect().num_rows >= 0
+
+
+def test_corrupt_tip_entry_scan_back(env, tmp_path):
+    session, h, df = env
+    h.create_index(df, hs.CoveringIndexConfig("cy", ["key"], ["val"]))
+    log_dir = os.path.join(str(tmp_path / "indexes"), "cy",
+                           "_hyperspace_log")
+    # fake a torn in-flight entry above the stable tip + torn stable copy
+    with open(os.path.join(log_dir, "7"), "w") as f:
+        f.write("not json at all")
+    os.unlink(os.path.join(log_dir, "latestStable"))
+    entry = session.index_manager().get_index("cy")
+    assert entry is not None and entry.state == States.ACTIVE
