@@ -137,3 +137,38 @@ def test_corrupt_tip_entry_scan_back(env, tmp_path):
     os.unlink(os.path.join(log_dir, "latestStable"))
     entry = session.index_manager().get_index("cy")
     assert entry is not None and entry.state == States.ACTIVE
+
+
+def _race_create(rank, tmpdir, results):
+    # spawned fresh process (torch mp.spawn): real cross-process race
+    os.environ["HYPERSPACE_SYSTEM_PATH"] = os.path.join(tmpdir, "indexes")
+    import hyperspace_amd as hs2
+    session = hs2.HyperspaceSession(device="cpu")
+    session.conf.set(hs2.IndexConstants.INDEX_NUM_BUCKETS, 4)
+    h2 = hs2.Hyperspace(session)
+    df2 = session.read_parquet(os.path.join(tmpdir, "data"))
+    try:
+        h2.create_index(df2, hs2.CoveringIndexConfig(
+            "race", ["key"], ["val"]))
+        results[rank] = "ok"
+    except Exception as e:  # noqa: BLE001
+        results[rank] = f"lost: {type(e).__name__}"
+
+
+def test_cross_process_create_race(env, tmp_path):
+    """Two PROCESSES race to create the same index: the hard-link log
+    claim arbitrates — at least one wins, the log ends consistent, and
+    the surviving index serves queries."""
+    import torch.multiprocessing as mp
+    session, h, df = env
+    with mp.Manager() as mgr:
+        results = mgr.dict()
+        mp.spawn(_race_create, args=(str(tmp_path), results), nprocs=2,
+                 join=True)
+        res = dict(results)
+    assert sum(1 for v in res.values() if v == "ok") >= 1, res
+    entry = session.index_manager().get_index("race")
+    assert entry is not None and entry.state == States.ACTIVE
+    session.enable_hyperspace()
+    out = df.filter("key = 5").select("key", "val").collect()
+    assert out.num_rows >= 0
