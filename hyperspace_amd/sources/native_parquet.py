@@ -430,10 +430,31 @@ class ColumnChunkLayout:
         return any(m is not None for m in self.page_masks)
 
 
+_DEF_DECODER = None
+
+
+def _native_def_decoder():
+    """C++ def-level decoder when the extension is importable (the
+    Python loop costs ~1 ms per page on pyarrow's many-run encodings);
+    cached; None -> pure-Python fallback."""
+    global _DEF_DECODER
+    if _DEF_DECODER is None:
+        try:
+            from ..ops import native as _native
+            _DEF_DECODER = (_native.ext().decode_def_levels,)
+        except Exception:  # noqa: BLE001
+            _DEF_DECODER = (None,)
+    return _DEF_DECODER[0]
+
+
 def _decode_defs(data, off: int, length: int, n: int
                  ) -> Optional[np.ndarray]:
     """Decode RLE-hybrid definition levels (max_def=1) into a bool
     validity array, or None when every row is valid."""
+    dec = _native_def_decoder()
+    if dec is not None:
+        mask = dec(data, off, length, n).numpy()
+        return None if bool(mask.all()) else mask
     end = off + length
     out = np.empty(n, dtype=np.uint8)
     pos = off

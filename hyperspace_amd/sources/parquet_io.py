@@ -342,9 +342,9 @@ def read_files_batch_device(paths: List[str], device,
                 ext.copy_unaligned(dev_bytes, dict_off, dict_vals, 0,
                                    dict_n * itemsize)
                 dict_vals = dict_vals[:dict_n].contiguous()
-            for page, pmask in zip(c.pages, c.page_masks):
-                nv = page[3] if page[0] == "dict" else page[2]
-                if pmask is None:
+            if not any(m is not None for m in c.page_masks):
+                for page in c.pages:
+                    nv = page[3] if page[0] == "dict" else page[2]
                     if page[0] == "dict":
                         _, p_start, p_end, _, bw = page
                         runs = ext.parse_rle_runs(buf, p_start, p_end, bw,
@@ -358,31 +358,48 @@ def read_files_batch_device(paths: List[str], device,
                         ext.copy_unaligned(dev_bytes, off, out[c.name],
                                            written * itemsize,
                                            nv * itemsize)
-                else:
-                    # nullable page: compacted values scatter to the
-                    # valid slots; null slots hold 0
-                    n_valid = int(pmask.sum())
-                    mask_dev = _torch.from_numpy(pmask).to(device)
+                    written += nv
+            else:
+                # nullable chunk: decode every page's compacted values
+                # into ONE temp, then a single chunk-wide scatter (per-
+                # page boolean indexing costs a kernel pair per page —
+                # thousands of launches on multi-page files)
+                n_rows = sum(p[3] if p[0] == "dict" else p[2]
+                             for p in c.pages)
+                valid_counts = [
+                    (int(m.sum()) if m is not None
+                     else (p[3] if p[0] == "dict" else p[2]))
+                    for p, m in zip(c.pages, c.page_masks)]
+                total_valid = sum(valid_counts)
+                tmp = _torch.empty(total_valid + 1,
+                                   dtype=out[c.name].dtype, device=device)
+                cur = 0
+                for page, n_valid in zip(c.pages, valid_counts):
                     if page[0] == "dict":
                         _, p_start, p_end, _, bw = page
                         runs = ext.parse_rle_runs(buf, p_start, p_end, bw,
                                                   n_valid)
                         idx = ext.rle_decode(dev_bytes, *runs, bw,
                                              n_valid).to(_torch.int64)
-                        vals = ext.gather_rows(dict_vals, idx)
+                        tmp[cur:cur + n_valid] = ext.gather_rows(
+                            dict_vals, idx)
                     else:
                         _, off, _ = page
-                        vals = _torch.empty(
-                            n_valid + 1, dtype=out[c.name].dtype,
-                            device=device)
-                        ext.copy_unaligned(dev_bytes, off, vals, 0,
+                        ext.copy_unaligned(dev_bytes, off, tmp,
+                                           cur * itemsize,
                                            n_valid * itemsize)
-                        vals = vals[:n_valid]
-                    dst = out[c.name][written:written + nv]
-                    dst.zero_()
-                    dst[mask_dev] = vals
-                    out_masks[c.name][written:written + nv] = mask_dev
-                written += nv
+                    cur += n_valid
+                chunk_mask = np.concatenate([
+                    (m if m is not None
+                     else np.ones(p[3] if p[0] == "dict" else p[2],
+                                  dtype=bool))
+                    for p, m in zip(c.pages, c.page_masks)])
+                mask_dev = _torch.from_numpy(chunk_mask).to(device)
+                dst = out[c.name][written:written + n_rows]
+                dst.zero_()
+                dst[mask_dev] = tmp[:total_valid]
+                out_masks[c.name][written:written + n_rows] = mask_dev
+                written += n_rows
             cursors[c.name] = written
         # pinned buffer must stay alive until the stream drains; the
         # caller-side synchronize below holds them via `bufs`
