@@ -269,6 +269,33 @@ torch::Tensor zorder_key(std::vector<torch::Tensor> cols,
   return out;
 }
 
+torch::Tensor snappy_decompress(torch::Tensor dev_bytes,
+                                torch::Tensor src_off,
+                                torch::Tensor src_end, torch::Tensor dst,
+                                torch::Tensor dst_off,
+                                torch::Tensor dst_len) {
+  check_cuda(dev_bytes, "dev_bytes");
+  check_cuda(dst, "dst");
+  TORCH_CHECK(dev_bytes.scalar_type() == torch::kUInt8 &&
+                  dst.scalar_type() == torch::kUInt8,
+              "byte tensors required");
+  int64_t n_pages = src_off.numel();
+  auto dev = dev_bytes.device();
+  auto i64 = torch::dtype(torch::kInt64).device(dev);
+  auto so = src_off.to(i64.device(), torch::kInt64).contiguous();
+  auto se = src_end.to(i64.device(), torch::kInt64).contiguous();
+  auto dofs = dst_off.to(i64.device(), torch::kInt64).contiguous();
+  auto dl = dst_len.to(i64.device(), torch::kInt64).contiguous();
+  auto status = torch::zeros(
+      {n_pages}, torch::dtype(torch::kInt32).device(dev));
+  hsk::snappy_decompress_pages(
+      dev_bytes.data_ptr<uint8_t>(), so.data_ptr<int64_t>(),
+      se.data_ptr<int64_t>(), dst.data_ptr<uint8_t>(),
+      dofs.data_ptr<int64_t>(), dl.data_ptr<int64_t>(),
+      status.data_ptr<int32_t>(), (int)n_pages, current_stream());
+  return status;
+}
+
 // Host decode of RLE-hybrid definition levels (max_def = 1) into a
 // validity byte array: the per-page Python decoder costs ~1 ms/page on
 // pyarrow's many-run encodings; this tight loop is ~20 us.
@@ -464,6 +491,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gather_rows", &gather_rows, "row gather by index");
   m.def("copy_unaligned", &copy_unaligned,
         "device parquet page decode (unaligned copy)");
+  m.def("snappy_decompress", &snappy_decompress,
+        "device snappy raw-block page decompression; returns status");
   m.def("decode_def_levels", &decode_def_levels,
         "host RLE-hybrid def-level decode -> bool validity");
   m.def("parse_rle_runs", &parse_rle_runs,

@@ -109,6 +109,14 @@ void rle_decode_indices(const uint8_t* src, const int64_t* run_kind,
 // alignment — to 4-byte-aligned dst+dst_off.  nbytes % 4 == 0.  The src
 // buffer must extend >= 4 bytes past src_off+nbytes (parquet footers
 // guarantee this for page payloads).
+// K1 snappy raw-block page decompression: one wave per page; status[p]
+// = 0 on success, nonzero on malformed input (caller falls back).
+void snappy_decompress_pages(const uint8_t* src, const int64_t* src_off,
+                             const int64_t* src_end, uint8_t* dst,
+                             const int64_t* dst_off,
+                             const int64_t* dst_len, int32_t* status,
+                             int n_pages, hipStream_t stream);
+
 void copy_unaligned(const uint8_t* src, int64_t src_off, uint8_t* dst,
                     int64_t dst_off, int64_t nbytes, hipStream_t stream);
 

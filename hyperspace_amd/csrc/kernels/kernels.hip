@@ -1192,4 +1192,116 @@ void copy_unaligned(const uint8_t* src, int64_t src_off, uint8_t* dst,
   }
 }
 
+// ---------------------------------------------------------------------------
+// K1: snappy raw-block decompression (parquet page codec).
+// One 64-lane wave per page: every lane walks the tag stream in
+// lockstep (identical control flow; the redundant tag loads hit L1) and
+// each op's byte movement distributes across the lanes.  Overlapping
+// back-references (off < len) replicate the off-length pattern via an
+// index modulo, which is byte-for-byte what serial snappy produces.
+// Before a copy op reads earlier OUTPUT bytes, __syncthreads() drains
+// the wave's pending vector writes (vmcnt) so other lanes' bytes are
+// visible.  Pages are independent streams, so the scan's parallelism
+// is pages x files.  Format: github.com/google/snappy format_description.
+// ---------------------------------------------------------------------------
+
+__global__ void k_snappy_decomp(const uint8_t* __restrict__ src,
+                                const int64_t* __restrict__ src_off,
+                                const int64_t* __restrict__ src_end,
+                                uint8_t* __restrict__ dst,
+                                const int64_t* __restrict__ dst_off,
+                                const int64_t* __restrict__ dst_len,
+                                int32_t* __restrict__ status) {
+  int p = blockIdx.x;
+  const uint8_t* s = src + src_off[p];
+  const uint8_t* send = src + src_end[p];
+  uint8_t* d = dst + dst_off[p];
+  int64_t expected = dst_len[p];
+  int lane = threadIdx.x;
+
+  // varint uncompressed length (lockstep on every lane)
+  uint64_t ulen = 0;
+  int shift = 0;
+  while (s < send) {
+    uint8_t b = *s++;
+    ulen |= (uint64_t)(b & 0x7F) << shift;
+    if (!(b & 0x80)) break;
+    shift += 7;
+  }
+  if ((int64_t)ulen != expected) {
+    if (lane == 0) status[p] = 1;
+    return;
+  }
+
+  int64_t out = 0;
+  while (out < expected && s < send) {
+    uint8_t tag = *s++;
+    int k = tag & 3;
+    int64_t len;
+    int64_t off = 0;
+    const uint8_t* lit = nullptr;
+    if (k == 0) {  // literal
+      int64_t l = tag >> 2;
+      if (l < 60) {
+        len = l + 1;
+      } else {
+        int nb = (int)(l - 59);  // 1..4 length bytes, little-endian
+        if (s + nb > send) { if (lane == 0) status[p] = 2; return; }
+        uint32_t v = 0;
+        for (int i = 0; i < nb; i++) v |= (uint32_t)s[i] << (8 * i);
+        s += nb;
+        len = (int64_t)v + 1;
+      }
+      lit = s;
+      s += len;
+      if (s > send) { if (lane == 0) status[p] = 2; return; }
+    } else if (k == 1) {  // copy, 1-byte offset
+      len = ((tag >> 2) & 7) + 4;
+      if (s + 1 > send) { if (lane == 0) status[p] = 2; return; }
+      off = ((int64_t)(tag >> 5) << 8) | s[0];
+      s += 1;
+    } else if (k == 2) {  // copy, 2-byte offset
+      len = (tag >> 2) + 1;
+      if (s + 2 > send) { if (lane == 0) status[p] = 2; return; }
+      off = (int64_t)s[0] | ((int64_t)s[1] << 8);
+      s += 2;
+    } else {  // copy, 4-byte offset
+      len = (tag >> 2) + 1;
+      if (s + 4 > send) { if (lane == 0) status[p] = 2; return; }
+      off = (int64_t)s[0] | ((int64_t)s[1] << 8) |
+            ((int64_t)s[2] << 16) | ((int64_t)s[3] << 24);
+      s += 4;
+    }
+    if (out + len > expected || (k != 0 && (off <= 0 || off > out))) {
+      if (lane == 0) status[p] = 3;
+      return;
+    }
+    if (k == 0) {
+      for (int64_t i = lane; i < len; i += 64) d[out + i] = lit[i];
+    } else {
+      __syncthreads();  // drain writes: the source may be recent output
+      if (off >= len) {
+        for (int64_t i = lane; i < len; i += 64)
+          d[out + i] = d[out - off + i];
+      } else {
+        for (int64_t i = lane; i < len; i += 64)
+          d[out + i] = d[out - off + (i % off)];
+      }
+    }
+    out += len;
+  }
+  if (lane == 0) status[p] = out == expected ? 0 : 4;
+}
+
+void snappy_decompress_pages(const uint8_t* src, const int64_t* src_off,
+                             const int64_t* src_end, uint8_t* dst,
+                             const int64_t* dst_off,
+                             const int64_t* dst_len, int32_t* status,
+                             int n_pages, hipStream_t stream) {
+  if (n_pages == 0) return;
+  hipLaunchKernelGGL(k_snappy_decomp, dim3(n_pages), dim3(64), 0, stream,
+                     src, src_off, src_end, dst, dst_off, dst_len,
+                     status);
+}
+
 }  // namespace hsk

@@ -538,10 +538,15 @@ def _walk_row_group(rg, pf_schema, data, want
         name = col.path_in_schema
         if want is not None and name.lower() not in want:
             continue
-        if col.compression.upper() != "UNCOMPRESSED":
+        codec = col.compression.upper()
+        if codec not in ("UNCOMPRESSED", "SNAPPY"):
             return None
         encs = set(col.encodings)
         is_dict = bool(encs & {"PLAIN_DICTIONARY", "RLE_DICTIONARY"})
+        if codec == "SNAPPY" and is_dict:
+            # the dictionary-index RLE stream is parsed on host; a
+            # compressed payload lives device-side only -> pyarrow
+            return None
         if encs - {"PLAIN", "RLE", "BIT_PACKED", "PLAIN_DICTIONARY",
                    "RLE_DICTIONARY"}:
             return None
@@ -562,6 +567,10 @@ def _walk_row_group(rg, pf_schema, data, want
         has_levels = col_schema.max_definition_level > 0
         st = col.statistics
         chunk_all_valid = st is not None and st.null_count == 0
+        if codec == "SNAPPY" and has_levels and not chunk_all_valid:
+            # def-levels sit inside the compressed payload; masked
+            # snappy pages decode via pyarrow
+            return None
 
         dict_page = None
         pos = col.data_page_offset
@@ -606,6 +615,22 @@ def _walk_row_group(rg, pf_schema, data, want
             values_off = r.pos
             page_end = r.pos + page_bytes
             mask = None
+            if codec == "SNAPPY":
+                # compressed PLAIN page: record the compressed extent +
+                # uncompressed size; the device decompresses and (for
+                # OPTIONAL all-valid chunks) skips the level prefix
+                # after decompression
+                if dph.get(2) != ENC_PLAIN:
+                    return None
+                unc = hdr.get(2)
+                if unc is None:
+                    return None
+                pages.append(("plain_z", values_off, page_end,
+                              num_values, unc, has_levels))
+                page_masks.append(None)
+                seen += num_values
+                pos = page_end
+                continue
             if has_levels:
                 lvl_len = struct.unpack_from("<I", data, values_off)[0]
                 if not chunk_all_valid:
@@ -633,7 +658,9 @@ def _walk_row_group(rg, pf_schema, data, want
         if seen != col.num_values:
             return None
         enc_kind = ("dict" if any(pg[0] == "dict" for pg in pages)
-                    else "plain")
+                    else ("plain_z"
+                          if any(pg[0] == "plain_z" for pg in pages)
+                          else "plain"))
         out.append(ColumnChunkLayout(name, np_dtype, pages,
                                      col.num_values, enc_kind, dict_page,
                                      page_masks))

@@ -294,6 +294,7 @@ def read_files_batch_device(paths: List[str], device,
                                 device=device)
                  for n in names if n in nullable_cols}
     ext = native_ext.ext()
+    statuses: List["_torch.Tensor"] = []  # snappy per-page status words
 
     # per-worker HIP streams: each file's H2D copy and decode kernels run
     # on their own stream, so copies overlap other files' decodes instead
@@ -329,6 +330,43 @@ def read_files_batch_device(paths: List[str], device,
         for c in chunks:
             itemsize = c.np_dtype.itemsize
             written = cursors[c.name]
+            if c.encoding == "plain_z":
+                # snappy chunk (K1): one wave per page decompresses into
+                # a scratch buffer; PLAIN decode then reads the scratch.
+                # OPTIONAL all-valid chunks carry a level prefix inside
+                # the decompressed stream — one small D2H reads the
+                # 4-byte lengths to skip it.
+                uncs = [p[4] for p in c.pages]
+                d_off = np.concatenate([[0], np.cumsum(uncs)])
+                scratch = _torch.empty(int(d_off[-1]) + 4,
+                                       dtype=_torch.uint8, device=device)
+                st = ext.snappy_decompress(
+                    dev_bytes,
+                    _torch.tensor([p[1] for p in c.pages],
+                                  dtype=_torch.int64),
+                    _torch.tensor([p[2] for p in c.pages],
+                                  dtype=_torch.int64),
+                    scratch,
+                    _torch.tensor(d_off[:-1], dtype=_torch.int64),
+                    _torch.tensor(uncs, dtype=_torch.int64))
+                statuses.append(st)
+                lvl_skips = [0] * len(c.pages)
+                with_lvl = [j for j, p in enumerate(c.pages) if p[5]]
+                if with_lvl:
+                    pref = _torch.stack([
+                        scratch[int(d_off[j]):int(d_off[j]) + 4]
+                        for j in with_lvl]).cpu()
+                    lens = pref.numpy().view("<u4").ravel()
+                    for j, ln in zip(with_lvl, lens):
+                        lvl_skips[j] = 4 + int(ln)
+                for j, page in enumerate(c.pages):
+                    nv = page[3]
+                    voff = int(d_off[j]) + lvl_skips[j]
+                    ext.copy_unaligned(scratch, voff, out[c.name],
+                                       written * itemsize, nv * itemsize)
+                    written += nv
+                cursors[c.name] = written
+                continue
             dict_vals = None
             if c.encoding == "dict":
                 # K1 dictionary path: decode the dictionary page once;
@@ -416,7 +454,9 @@ def read_files_batch_device(paths: List[str], device,
     for s in streams:
         cur.wait_stream(s)
     cur.synchronize()
-    if not all(b is not None and b is not False for b in bufs):
+    snappy_bad = any(bool((s != 0).any()) for s in statuses)
+    if snappy_bad or \
+            not all(b is not None and b is not False for b in bufs):
         for b in bufs:
             if b is not None and b is not False:
                 _pinned_put(b)
