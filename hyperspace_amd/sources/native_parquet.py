@@ -543,10 +543,6 @@ def _walk_row_group(rg, pf_schema, data, want
             return None
         encs = set(col.encodings)
         is_dict = bool(encs & {"PLAIN_DICTIONARY", "RLE_DICTIONARY"})
-        if codec == "SNAPPY" and is_dict:
-            # the dictionary-index RLE stream is parsed on host; a
-            # compressed payload lives device-side only -> pyarrow
-            return None
         if encs - {"PLAIN", "RLE", "BIT_PACKED", "PLAIN_DICTIONARY",
                    "RLE_DICTIONARY"}:
             return None
@@ -588,7 +584,13 @@ def _walk_row_group(rg, pf_schema, data, want
             dict_n = hdr.get(7, {}).get(1)
             if dict_n is None:
                 return None
-            dict_page = (r.pos, dict_n)
+            if codec == "SNAPPY":
+                if hdr.get(2) is None:
+                    return None
+                dict_page = ("z", r.pos, r.pos + hdr.get(3), dict_n,
+                             hdr.get(2))
+            else:
+                dict_page = (r.pos, dict_n)
             if dict_off >= pos:
                 # dictionary physically precedes data pages
                 pos = r.pos + hdr.get(3)
@@ -616,17 +618,23 @@ def _walk_row_group(rg, pf_schema, data, want
             page_end = r.pos + page_bytes
             mask = None
             if codec == "SNAPPY":
-                # compressed PLAIN page: record the compressed extent +
-                # uncompressed size; the device decompresses and (for
-                # OPTIONAL all-valid chunks) skips the level prefix
-                # after decompression
-                if dph.get(2) != ENC_PLAIN:
-                    return None
+                # compressed page: record the compressed extent +
+                # uncompressed size; the device decompresses, then (for
+                # OPTIONAL all-valid chunks) the level prefix is skipped
+                # post-decompression.  Dictionary-index pages D2H their
+                # small decompressed payload for the host run parser.
                 unc = hdr.get(2)
                 if unc is None:
                     return None
-                pages.append(("plain_z", values_off, page_end,
-                              num_values, unc, has_levels))
+                pe = dph.get(2)
+                if is_dict and pe in (2, 8):
+                    pages.append(("dict_z", values_off, page_end,
+                                  num_values, unc, has_levels))
+                elif pe == ENC_PLAIN:
+                    pages.append(("plain_z", values_off, page_end,
+                                  num_values, unc, has_levels))
+                else:
+                    return None
                 page_masks.append(None)
                 seen += num_values
                 pos = page_end
@@ -657,10 +665,14 @@ def _walk_row_group(rg, pf_schema, data, want
             pos = page_end
         if seen != col.num_values:
             return None
-        enc_kind = ("dict" if any(pg[0] == "dict" for pg in pages)
-                    else ("plain_z"
-                          if any(pg[0] == "plain_z" for pg in pages)
-                          else "plain"))
+        if any(pg[0] == "dict" for pg in pages):
+            enc_kind = "dict"
+        elif any(pg[0] == "dict_z" for pg in pages):
+            enc_kind = "dict_z"
+        elif any(pg[0] == "plain_z" for pg in pages):
+            enc_kind = "plain_z"
+        else:
+            enc_kind = "plain"
         out.append(ColumnChunkLayout(name, np_dtype, pages,
                                      col.num_values, enc_kind, dict_page,
                                      page_masks))

@@ -330,40 +330,85 @@ def read_files_batch_device(paths: List[str], device,
         for c in chunks:
             itemsize = c.np_dtype.itemsize
             written = cursors[c.name]
-            if c.encoding == "plain_z":
+            if c.encoding in ("plain_z", "dict_z"):
                 # snappy chunk (K1): one wave per page decompresses into
-                # a scratch buffer; PLAIN decode then reads the scratch.
-                # OPTIONAL all-valid chunks carry a level prefix inside
-                # the decompressed stream — one small D2H reads the
-                # 4-byte lengths to skip it.
-                uncs = [p[4] for p in c.pages]
-                d_off = np.concatenate([[0], np.cumsum(uncs)])
-                scratch = _torch.empty(int(d_off[-1]) + 4,
+                # a scratch buffer; PLAIN pages copy straight from
+                # scratch, dictionary-index pages D2H their (small)
+                # decompressed payload for the host run parser and
+                # gather through the decompressed dictionary on device.
+                has_zdict = c.encoding == "dict_z"
+                segs = []
+                if has_zdict:
+                    _, dz_off, dz_end, dict_n, dict_unc = c.dict_page
+                    segs.append((dz_off, dz_end, dict_unc))
+                for p in c.pages:
+                    segs.append((p[1], p[2], p[4]))
+                uncs_all = [s[2] for s in segs]
+                doff_all = np.concatenate([[0], np.cumsum(uncs_all)])
+                scratch = _torch.empty(int(doff_all[-1]) + 4,
                                        dtype=_torch.uint8, device=device)
                 st = ext.snappy_decompress(
                     dev_bytes,
-                    _torch.tensor([p[1] for p in c.pages],
+                    _torch.tensor([s[0] for s in segs],
                                   dtype=_torch.int64),
-                    _torch.tensor([p[2] for p in c.pages],
+                    _torch.tensor([s[1] for s in segs],
                                   dtype=_torch.int64),
                     scratch,
-                    _torch.tensor(d_off[:-1], dtype=_torch.int64),
-                    _torch.tensor(uncs, dtype=_torch.int64))
+                    _torch.tensor(doff_all[:-1], dtype=_torch.int64),
+                    _torch.tensor(uncs_all, dtype=_torch.int64))
                 statuses.append(st)
-                lvl_skips = [0] * len(c.pages)
-                with_lvl = [j for j, p in enumerate(c.pages) if p[5]]
-                if with_lvl:
-                    pref = _torch.stack([
-                        scratch[int(d_off[j]):int(d_off[j]) + 4]
-                        for j in with_lvl]).cpu()
-                    lens = pref.numpy().view("<u4").ravel()
-                    for j, ln in zip(with_lvl, lens):
-                        lvl_skips[j] = 4 + int(ln)
+                page_base = doff_all[1:-1] if has_zdict \
+                    else doff_all[:-1]
+                dict_vals = None
+                host_all = None
+                if has_zdict:
+                    dict_vals = _torch.empty(
+                        dict_n + 1, dtype=out[c.name].dtype,
+                        device=device)
+                    ext.copy_unaligned(scratch, 0, dict_vals, 0,
+                                       dict_n * itemsize)
+                    dict_vals = dict_vals[:dict_n].contiguous()
+                    # dict chunks are small: one D2H yields every page's
+                    # level prefix + RLE index stream (positions align
+                    # with scratch, so run bit-offsets need no fixup)
+                    host_all = scratch[:int(doff_all[-1])].cpu()
+                else:
+                    lvl_skips = [0] * len(c.pages)
+                    with_lvl = [j for j, p in enumerate(c.pages)
+                                if p[5]]
+                    if with_lvl:
+                        pref = _torch.stack([
+                            scratch[int(page_base[j]):
+                                    int(page_base[j]) + 4]
+                            for j in with_lvl]).cpu()
+                        lens = pref.numpy().view("<u4").ravel()
+                        for j, ln in zip(with_lvl, lens):
+                            lvl_skips[j] = 4 + int(ln)
                 for j, page in enumerate(c.pages):
                     nv = page[3]
-                    voff = int(d_off[j]) + lvl_skips[j]
-                    ext.copy_unaligned(scratch, voff, out[c.name],
-                                       written * itemsize, nv * itemsize)
+                    base = int(page_base[j])
+                    if page[0] == "dict_z":
+                        hb = host_all
+                        skip = 0
+                        if page[5]:
+                            skip = 4 + int(hb[base:base + 4].numpy()
+                                           .view("<u4")[0])
+                        bw = int(hb[base + skip])
+                        runs = ext.parse_rle_runs(
+                            hb, base + skip + 1, base + page[4], bw, nv)
+                        idx = ext.rle_decode(scratch, *runs, bw,
+                                             nv).to(_torch.int64)
+                        out[c.name][written:written + nv] = \
+                            ext.gather_rows(dict_vals, idx)
+                    else:
+                        skip = (4 + int(host_all[base:base + 4].numpy()
+                                        .view("<u4")[0])) \
+                            if (page[5] and has_zdict) else \
+                            (lvl_skips[j] if not has_zdict else 0)
+                        ext.copy_unaligned(scratch, base + skip,
+                                           out[c.name],
+                                           written * itemsize,
+                                           nv * itemsize)
                     written += nv
                 cursors[c.name] = written
                 continue

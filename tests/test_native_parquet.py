@@ -65,12 +65,22 @@ def test_native_reads_pyarrow_files_multipage(tmp_path, cols):
 
 
 def test_fallback_on_compressed(tmp_path, cols):
+    # snappy now yields a device-decodable layout; the HOST decode path
+    # still defers to pyarrow (device decompression is GPU-only), and
+    # truly unsupported codecs yield no layout at all
     p = str(tmp_path / "t.parquet")
     pq.write_table(pa.table(cols), p, compression="SNAPPY")
-    assert read_native_layout(p) is None
-    # read_files_batch still works via the pyarrow fallback
+    lay = read_native_layout(p)
+    assert lay is not None
+    assert all(c.encoding in ("plain_z", "dict_z") for c in lay[1])
+    assert read_native_host(p) is None
     batch, counts = read_files_batch([p])
     assert batch.num_rows == 50_000 and counts == [50_000]
+    p2 = str(tmp_path / "t2.parquet")
+    pq.write_table(pa.table(cols), p2, compression="GZIP")
+    assert read_native_layout(p2) is None
+    batch2, counts2 = read_files_batch([p2])
+    assert batch2.num_rows == 50_000 and counts2 == [50_000]
 
 
 def test_write_batch_parquet_uses_native(tmp_path, cols):

@@ -84,3 +84,24 @@ def test_snappy_nullable_falls_back(tmp_path):
     assert counts == [10_000]
     m = batch.mask("k")
     assert m is not None and int((~m.cpu()).sum()) == int((~mask).sum())
+
+
+def test_snappy_dictionary_device_read(tmp_path):
+    # Spark's default output shape: SNAPPY + dictionary encoding
+    rng = np.random.default_rng(6)
+    key = rng.integers(0, 1000, 400_000)   # dict-encodable
+    val = rng.random(400_000)              # dict overflows -> mixed
+    paths = []
+    for i in range(2):
+        p = str(tmp_path / f"sd{i}.parquet")
+        pq.write_table(pa.table({"key": key, "val": val}), p,
+                       compression="SNAPPY", use_dictionary=True,
+                       data_page_version="1.0")
+        paths.append(p)
+    dev_batch, counts = read_files_batch_device(
+        paths, torch.device("cuda:0"))
+    host_batch, hcounts = read_files_batch(paths)
+    assert counts == hcounts == [400_000, 400_000]
+    for cname in ("key", "val"):
+        assert torch.equal(dev_batch.tensor(cname).cpu(),
+                           host_batch.tensor(cname)), cname
