@@ -563,10 +563,6 @@ def _walk_row_group(rg, pf_schema, data, want
         has_levels = col_schema.max_definition_level > 0
         st = col.statistics
         chunk_all_valid = st is not None and st.null_count == 0
-        if codec == "SNAPPY" and has_levels and not chunk_all_valid:
-            # def-levels sit inside the compressed payload; masked
-            # snappy pages decode via pyarrow
-            return None
 
         dict_page = None
         pos = col.data_page_offset

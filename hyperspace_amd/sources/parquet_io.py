@@ -384,31 +384,66 @@ def read_files_batch_device(paths: List[str], device,
                         lens = pref.numpy().view("<u4").ravel()
                         for j, ln in zip(with_lvl, lens):
                             lvl_skips[j] = 4 + int(ln)
+                nullable_chunk = c.name in out_masks
                 for j, page in enumerate(c.pages):
                     nv = page[3]
                     base = int(page_base[j])
+                    # level prefix + (for nullable chunks) the decoded
+                    # validity mask; level bytes are small, so a D2H of
+                    # just that region feeds the host def decoder
+                    skip = 0
+                    pmask = None
+                    if page[5]:
+                        if has_zdict:
+                            ln = int(host_all[base:base + 4].numpy()
+                                     .view("<u4")[0])
+                            lvl_bytes = host_all[base + 4:
+                                                 base + 4 + ln].numpy()
+                        else:
+                            ln = lvl_skips[j] - 4
+                            lvl_bytes = None
+                        skip = 4 + ln
+                        if nullable_chunk:
+                            if lvl_bytes is None:
+                                lvl_bytes = scratch[
+                                    base + 4:base + 4 + ln].cpu().numpy()
+                            from .native_parquet import _decode_defs
+                            pmask = _decode_defs(
+                                lvl_bytes.tobytes(), 0, ln, nv)
+                    n_valid = int(pmask.sum()) if pmask is not None \
+                        else nv
                     if page[0] == "dict_z":
                         hb = host_all
-                        skip = 0
-                        if page[5]:
-                            skip = 4 + int(hb[base:base + 4].numpy()
-                                           .view("<u4")[0])
                         bw = int(hb[base + skip])
                         runs = ext.parse_rle_runs(
-                            hb, base + skip + 1, base + page[4], bw, nv)
+                            hb, base + skip + 1, base + page[4], bw,
+                            n_valid)
                         idx = ext.rle_decode(scratch, *runs, bw,
-                                             nv).to(_torch.int64)
-                        out[c.name][written:written + nv] = \
-                            ext.gather_rows(dict_vals, idx)
+                                             n_valid).to(_torch.int64)
+                        vals = ext.gather_rows(dict_vals, idx)
                     else:
-                        skip = (4 + int(host_all[base:base + 4].numpy()
-                                        .view("<u4")[0])) \
-                            if (page[5] and has_zdict) else \
-                            (lvl_skips[j] if not has_zdict else 0)
-                        ext.copy_unaligned(scratch, base + skip,
-                                           out[c.name],
-                                           written * itemsize,
-                                           nv * itemsize)
+                        if pmask is None:
+                            ext.copy_unaligned(scratch, base + skip,
+                                               out[c.name],
+                                               written * itemsize,
+                                               nv * itemsize)
+                            written += nv
+                            continue
+                        vals = _torch.empty(
+                            n_valid + 1, dtype=out[c.name].dtype,
+                            device=device)
+                        ext.copy_unaligned(scratch, base + skip, vals,
+                                           0, n_valid * itemsize)
+                        vals = vals[:n_valid]
+                    if pmask is None:
+                        out[c.name][written:written + nv] = vals
+                    else:
+                        mask_dev = _torch.from_numpy(pmask).to(device)
+                        dst = out[c.name][written:written + nv]
+                        dst.zero_()
+                        dst[mask_dev] = vals
+                        out_masks[c.name][written:written + nv] = \
+                            mask_dev
                     written += nv
                 cursors[c.name] = written
                 continue
