@@ -385,6 +385,11 @@ def read_files_batch_device(paths: List[str], device,
                         for j, ln in zip(with_lvl, lens):
                             lvl_skips[j] = 4 + int(ln)
                 nullable_chunk = c.name in out_masks
+                # batch unmasked dict_z pages: one rle_decode + gather
+                # per (chunk, bit-width) instead of per page — the
+                # per-page host->device run uploads dominate otherwise
+                zbatch: Dict[int, list] = {}
+                zbatch_rows = 0
                 for j, page in enumerate(c.pages):
                     nv = page[3]
                     base = int(page_base[j])
@@ -415,9 +420,19 @@ def read_files_batch_device(paths: List[str], device,
                     if page[0] == "dict_z":
                         hb = host_all
                         bw = int(hb[base + skip])
-                        runs = ext.parse_rle_runs(
+                        runs = list(ext.parse_rle_runs(
                             hb, base + skip + 1, base + page[4], bw,
-                            n_valid)
+                            n_valid))
+                        if pmask is None:
+                            lst = zbatch.setdefault(
+                                bw, {"runs": [], "rows": 0, "pages": []})
+                            runs[1] = runs[1] + lst["rows"]
+                            lst["runs"].append(runs)
+                            lst["pages"].append(
+                                (written, nv, lst["rows"]))
+                            lst["rows"] += nv
+                            written += nv
+                            continue
                         idx = ext.rle_decode(scratch, *runs, bw,
                                              n_valid).to(_torch.int64)
                         vals = ext.gather_rows(dict_vals, idx)
@@ -445,6 +460,15 @@ def read_files_batch_device(paths: List[str], device,
                         out_masks[c.name][written:written + nv] = \
                             mask_dev
                     written += nv
+                for bw, lst in zbatch.items():
+                    cat = [_torch.cat([r[k] for r in lst["runs"]])
+                           for k in range(5)]
+                    idx = ext.rle_decode(scratch, *cat, bw,
+                                         lst["rows"]).to(_torch.int64)
+                    vals = ext.gather_rows(dict_vals, idx)
+                    for woff, nv, rbase in lst["pages"]:
+                        out[c.name][woff:woff + nv] = \
+                            vals[rbase:rbase + nv]
                 cursors[c.name] = written
                 continue
             dict_vals = None
