@@ -46,6 +46,53 @@ class CoveringIndexConfig(IndexConfigTrait):
     def index_name(self) -> str:
         return self._name
 
+    class Builder:
+        """Builder pattern (reference CoveringIndexConfig.Builder,
+        index/covering/CoveringIndexConfig.scala:60-151): each setter
+        may be called once; create() validates like the constructor."""
+
+        def __init__(self):
+            self._name = ""
+            self._indexed: List[str] = []
+            self._included: List[str] = []
+
+        def index_name(self, name: str) -> "CoveringIndexConfig.Builder":
+            if self._name:
+                raise HyperspaceException("Index name is already set.")
+            if not name:
+                raise HyperspaceException(
+                    "Empty index name is not allowed.")
+            self._name = name
+            return self
+
+        def index_by(self, column: str, *more: str
+                     ) -> "CoveringIndexConfig.Builder":
+            if self._indexed:
+                raise HyperspaceException(
+                    "Indexed columns are already set.")
+            self._indexed = [column, *more]
+            return self
+
+        def include(self, column: str, *more: str
+                    ) -> "CoveringIndexConfig.Builder":
+            if self._included:
+                raise HyperspaceException(
+                    "Included columns are already set.")
+            self._included = [column, *more]
+            return self
+
+        # camelCase aliases mirroring the reference API
+        indexName = index_name
+        indexBy = index_by
+
+        def create(self) -> "CoveringIndexConfig":
+            return CoveringIndexConfig(self._name, self._indexed,
+                                       self._included)
+
+    @staticmethod
+    def builder() -> "CoveringIndexConfig.Builder":
+        return CoveringIndexConfig.Builder()
+
     def referenced_columns(self) -> List[str]:
         return self.indexed_columns + self.included_columns
 

@@ -139,3 +139,23 @@ def test_cache_with_transform():
     state["k"] = 2
     assert c.load() == 20     # conf change invalidates
     assert calls == [1, 2]
+
+
+def test_covering_config_builder():
+    """Builder pattern parity (reference CoveringIndexConfig.Builder)."""
+    import pytest as _pytest
+    import hyperspace_amd as hs
+    from hyperspace_amd.exceptions import HyperspaceException
+    cfg = (hs.CoveringIndexConfig.builder()
+           .index_name("bix").index_by("k1", "k2").include("v").create())
+    assert cfg.index_name == "bix"
+    assert cfg.indexed_columns == ["k1", "k2"]
+    assert cfg.included_columns == ["v"]
+    with _pytest.raises(HyperspaceException):
+        hs.CoveringIndexConfig.builder().index_name("a").index_name("b")
+    with _pytest.raises(HyperspaceException):
+        hs.CoveringIndexConfig.builder().index_by("x").create()  # no name
+    # camelCase aliases
+    cfg2 = (hs.CoveringIndexConfig.builder()
+            .indexName("c").indexBy("k").create())
+    assert cfg2.indexed_columns == ["k"]
