@@ -22,6 +22,10 @@ from .dataframe import DataFrame
 from .exceptions import HyperspaceException, KernelUnavailableError
 from .hyperspace import Hyperspace
 from .index.covering import CoveringIndex, CoveringIndexConfig
+
+# backward-compat alias (reference index/package.scala:24-33:
+# IndexConfig = CoveringIndexConfig)
+IndexConfig = CoveringIndexConfig
 from .index.dataskipping import (BloomFilterSketch, DataSkippingIndex,
                                  DataSkippingIndexConfig, MinMaxSketch,
                                  PartitionSketch)
@@ -34,7 +38,7 @@ __version__ = "0.1.0"
 __all__ = [
     "Conf", "IndexConstants", "DataFrame", "Hyperspace",
     "HyperspaceSession", "HyperspaceException", "KernelUnavailableError",
-    "CoveringIndex", "CoveringIndexConfig", "col", "lit",
+    "CoveringIndex", "CoveringIndexConfig", "IndexConfig", "col", "lit",
     "DataSkippingIndex", "DataSkippingIndexConfig", "MinMaxSketch",
     "BloomFilterSketch", "PartitionSketch",
     "ZOrderCoveringIndex", "ZOrderCoveringIndexConfig",
