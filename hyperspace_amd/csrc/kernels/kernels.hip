@@ -1234,6 +1234,7 @@ __global__ void k_snappy_decomp(const uint8_t* __restrict__ src,
   }
 
   int64_t out = 0;
+  int64_t synced = 0;  // output below this point is visible to all lanes
   while (out < expected && s < send) {
     uint8_t tag = *s++;
     int k = tag & 3;
@@ -1279,7 +1280,13 @@ __global__ void k_snappy_decomp(const uint8_t* __restrict__ src,
     if (k == 0) {
       for (int64_t i = lane; i < len; i += 64) d[out + i] = lit[i];
     } else {
-      __syncthreads();  // drain writes: the source may be recent output
+      // drain pending writes only when the source range reaches past
+      // the last sync point (i.e. could include other lanes' recent
+      // bytes); far-back references skip the barrier entirely
+      if (out - off + len > synced) {
+        __syncthreads();
+        synced = out;
+      }
       if (off >= len) {
         for (int64_t i = lane; i < len; i += 64)
           d[out + i] = d[out - off + i];
