@@ -172,3 +172,53 @@ def test_cross_process_create_race(env, tmp_path):
     session.enable_hyperspace()
     out = df.filter("key = 5").select("key", "val").collect()
     assert out.num_rows >= 0
+
+
+def test_query_during_refresh_sees_consistent_snapshot(env, tmp_path):
+    """Readers resolve indexes through latestStable only: while a full
+    refresh rewrites the index, concurrent queries must return either
+    the old or the new answer — never a torn state."""
+    import threading
+
+    session, h, df = env
+    h.create_index(df, hs.CoveringIndexConfig("qr", ["key"], ["val"]))
+    session.enable_hyperspace()
+    baseline = df.filter("key = 5").select("key", "val").collect().num_rows
+
+    data_dir = df.plan.collect_leaves()[0].relation.root_paths[0]
+    import numpy as np
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    rng = np.random.default_rng(123)
+    extra_key = rng.integers(0, 100, 2000)
+    pq.write_table(pa.table({"key": extra_key,
+                             "val": rng.random(2000)}),
+                   data_dir + "/part-9.parquet")
+    new_expected = baseline + int((extra_key == 5).sum())
+
+    results = []
+    errors = []
+    stop = threading.Event()
+
+    def reader():
+        while not stop.is_set():
+            try:
+                n = df.filter("key = 5").select("key", "val") \
+                    .collect().num_rows
+                results.append(n)
+            except Exception as e:  # noqa: BLE001
+                errors.append(repr(e))
+
+    t = threading.Thread(target=reader)
+    t.start()
+    try:
+        h.refresh_index("qr", mode="full")
+    finally:
+        stop.set()
+        t.join(30)
+    assert not errors, errors[:3]
+    # every observation is one of the two legal snapshots
+    assert set(results) <= {baseline, new_expected}, set(results)
+    # and the post-refresh answer is the new one
+    final = df.filter("key = 5").select("key", "val").collect().num_rows
+    assert final == new_expected
