@@ -62,16 +62,31 @@ def main():
 
     device = args.device or ("cuda" if on_gpu else "cpu")
 
+    def _mem_available_bytes():
+        # tmpfs "free" reports the mount SIZE limit, not physical RAM:
+        # filling /dev/shm past MemAvailable OOM-kills the host.  Gate
+        # every shm decision on actual available memory.
+        try:
+            with open("/proc/meminfo") as f:
+                for line in f:
+                    if line.startswith("MemAvailable:"):
+                        return int(line.split()[1]) * 1024
+        except OSError:
+            pass
+        return 64 << 30
+
     workdir = args.workdir or os.environ.get("BENCH_WORKDIR")
     if workdir is None:
         # prefer RAM-backed storage: GPU-box /tmp overlays are small
-        # (~79G) while hosts carry TBs of RAM — /dev/shm holds the
-        # synthetic source + index versions without disk limits
+        # (~79G); /dev/shm holds the synthetic source + index versions
+        # — but only up to what physical RAM actually allows
         needed_est = int(args.gb_per_gpu * (1 << 30)) * world * 4
         shm_free = shutil.disk_usage("/dev/shm").free \
             if os.path.isdir("/dev/shm") else 0
+        shm_capacity = min(shm_free,
+                           _mem_available_bytes() - (48 << 30))
         workdir = ("/dev/shm/hyperspace_bench"
-                   if shm_free > needed_est + (8 << 30)
+                   if shm_capacity > needed_est + (8 << 30)
                    else "/tmp/hyperspace_bench")
     data_dir = os.path.join(workdir, "fact")
     dim_dir = os.path.join(workdir, "dim")
@@ -133,6 +148,9 @@ def main():
     pending_cleanup = []
     needed = int((args.steps + args.warmup + 1) * total_bytes * 1.3)
     free = shutil.disk_usage(workdir).free
+    if workdir.startswith("/dev/shm"):
+        # tmpfs free space is bounded by RAM, not the mount size
+        free = min(free, _mem_available_bytes() - (48 << 30))
     cleanup_inline = free < needed + (50 << 30)
     if rank == 0 and os.environ.get("BENCH_DEBUG"):
         print(f"[dbg] disk free={free/2**30:.0f}G needed={needed/2**30:.0f}G"
