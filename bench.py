@@ -88,6 +88,17 @@ def main():
         workdir = ("/dev/shm/hyperspace_bench"
                    if shm_capacity > needed_est + (8 << 30)
                    else "/tmp/hyperspace_bench")
+    if os.environ.get("BENCH_PROBE"):
+        shm_free = shutil.disk_usage("/dev/shm").free \
+            if os.path.isdir("/dev/shm") else 0
+        print(json.dumps({
+            "probe": True, "workdir": workdir,
+            "mem_available_gb": round(_mem_available_bytes() / 2**30, 1),
+            "shm_free_gb": round(shm_free / 2**30, 1),
+            "tmp_free_gb": round(
+                shutil.disk_usage("/tmp").free / 2**30, 1)}))
+        return
+
     data_dir = os.path.join(workdir, "fact")
     dim_dir = os.path.join(workdir, "dim")
     index_root = os.path.join(workdir, "indexes")
