@@ -269,6 +269,25 @@ torch::Tensor zorder_key(std::vector<torch::Tensor> cols,
   return out;
 }
 
+torch::Tensor run_merge_perm(torch::Tensor keys, torch::Tensor seg,
+                             torch::Tensor split) {
+  check_cuda(keys, "keys");
+  TORCH_CHECK(keys.scalar_type() == torch::kInt64, "keys must be int64");
+  auto dev = keys.device();
+  auto i64 = torch::dtype(torch::kInt64);
+  auto seg_d = seg.to(dev, torch::kInt64).contiguous();
+  auto split_d = split.to(dev, torch::kInt64).contiguous();
+  int64_t n = keys.numel();
+  int64_t n_seg = seg_d.numel() - 1;
+  TORCH_CHECK(split_d.numel() == n_seg, "split must have n_seg entries");
+  auto perm = torch::empty({n}, i64.device(dev));
+  hsk::run_merge_perm((const uint64_t*)keys.data_ptr<int64_t>(),
+                      seg_d.data_ptr<int64_t>(),
+                      split_d.data_ptr<int64_t>(), n, n_seg,
+                      perm.data_ptr<int64_t>(), current_stream());
+  return perm;
+}
+
 torch::Tensor snappy_decompress(torch::Tensor dev_bytes,
                                 torch::Tensor src_off,
                                 torch::Tensor src_end, torch::Tensor dst,
@@ -491,6 +510,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gather_rows", &gather_rows, "row gather by index");
   m.def("copy_unaligned", &copy_unaligned,
         "device parquet page decode (unaligned copy)");
+  m.def("run_merge_perm", &run_merge_perm,
+        "segmented two-sorted-run merge permutation");
   m.def("snappy_decompress", &snappy_decompress,
         "device snappy raw-block page decompression; returns status");
   m.def("decode_def_levels", &decode_def_levels,

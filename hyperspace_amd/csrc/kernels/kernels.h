@@ -40,6 +40,11 @@ int64_t radix_sort_hist_size(int64_t n);
 // emits pairs at tile_offset + intra-tile LDS prefix (no per-row
 // metadata arrays touch HBM).
 int64_t merge_join_tile_size();
+// K4b: segmented two-sorted-run merge permutation (split[b] = A/B
+// boundary per bucket; == seg[b+1] for single-run buckets).
+void run_merge_perm(const uint64_t* keys, const int64_t* seg,
+                    const int64_t* split, int64_t n, int64_t n_seg,
+                    int64_t* perm, hipStream_t stream);
 void merge_join_tile_count(const uint64_t* lkeys, const uint64_t* rkeys,
                            const int64_t* lseg, const int64_t* rseg,
                            int64_t n_left, int64_t n_seg,

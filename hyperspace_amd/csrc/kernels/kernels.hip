@@ -758,6 +758,42 @@ __global__ void k_mj_tile_emit(const uint64_t* __restrict__ lkeys,
 
 int64_t merge_join_tile_size() { return MJ_TILE; }
 
+// K4b: segmented two-sorted-run merge permutation.  Each bucket b's
+// rows lie in [seg[b], seg[b+1]) as [A-run | B-run] with the boundary
+// at split[b] (split[b] == seg[b+1] for single-run buckets).  Each
+// row's merged position is its local rank plus its rank in the
+// opposite run; A-rows stably precede equal B-rows.  One launch
+// replaces a per-bucket sort loop when a scan merges the two sorted
+// files an incremental refresh / two-group build leaves per bucket.
+__global__ void k_run_merge_perm(const uint64_t* __restrict__ keys,
+                                 const int64_t* __restrict__ seg,
+                                 const int64_t* __restrict__ split,
+                                 int64_t n, int64_t n_seg,
+                                 int64_t* __restrict__ perm) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t s = mj_seg_of(seg, n_seg, i);
+    int64_t a0 = seg[s], b1 = seg[s + 1], sp = split[s];
+    uint64_t k = keys[i];
+    int64_t pos;
+    if (i < sp) {  // A row: count B elements strictly below it
+      pos = (i - a0) + (lower_bound_u64(keys, sp, b1, k) - sp);
+    } else {  // B row: count A elements at or below it
+      pos = (i - sp) + (upper_bound_u64(keys, a0, sp, k) - a0);
+    }
+    perm[a0 + pos] = i;
+  }
+}
+
+void run_merge_perm(const uint64_t* keys, const int64_t* seg,
+                    const int64_t* split, int64_t n, int64_t n_seg,
+                    int64_t* perm, hipStream_t stream) {
+  if (n == 0) return;
+  hipLaunchKernelGGL(k_run_merge_perm, dim3(grid_for(n)), dim3(THREADS),
+                     0, stream, keys, seg, split, n, n_seg, perm);
+}
+
 void merge_join_tile_count(const uint64_t* lkeys, const uint64_t* rkeys,
                            const int64_t* lseg, const int64_t* rseg,
                            int64_t n_left, int64_t n_seg,

@@ -271,23 +271,33 @@ class Executor:
             if plan.use_bucket_spec and sort_col is not None and \
                     any(n > 1 for _, n in files_per_bucket) and \
                     batch.num_rows:
-                merged_ok = not batch.has_nulls(sort_col)
-                pieces: List[ColumnBatch] = []
-                for (b, nfiles), (fi0, _) in zip(files_per_bucket,
-                                                 file_ix_per_bucket):
-                    lo, hi = int(seg[b]), int(seg[b + 1])
-                    sub = batch.slice(lo, hi)
-                    if nfiles == 2 and sub.num_rows and merged_ok:
-                        # two sorted runs: a searchsorted merge beats a
-                        # full re-sort (no radix passes, no histograms)
-                        n_a = row_counts[fi0]
-                        sub = _merge_two_sorted_runs(sub, sort_col, n_a)
-                    elif nfiles > 1 and sub.num_rows:
-                        perm = ops.sort_perm(
-                            ops.normalize_key(sub.tensor(sort_col)))
-                        sub = sub.gather(perm)
-                    pieces.append(sub)
-                batch = ColumnBatch.concat(pieces)
+                merged_ok = not batch.has_nulls(sort_col) and \
+                    all(n <= 2 for _, n in files_per_bucket)
+                if merged_ok:
+                    # every merged bucket holds exactly two sorted runs
+                    # (incremental refresh / two-group builds): ONE
+                    # segmented merge-rank launch (K4b) replaces the
+                    # per-bucket sort loop
+                    split = seg[1:].clone()
+                    for (b, nfiles), (fi0, _) in zip(
+                            files_per_bucket, file_ix_per_bucket):
+                        if nfiles == 2:
+                            split[b] = int(seg[b]) + row_counts[fi0]
+                    keys = ops.normalize_key(batch.tensor(sort_col))
+                    perm = ops.run_merge_perm(keys, seg, split)
+                    batch = batch.gather(perm)
+                else:
+                    pieces: List[ColumnBatch] = []
+                    for (b, nfiles), (fi0, _) in zip(files_per_bucket,
+                                                     file_ix_per_bucket):
+                        lo, hi = int(seg[b]), int(seg[b + 1])
+                        sub = batch.slice(lo, hi)
+                        if nfiles > 1 and sub.num_rows:
+                            perm = ops.sort_perm(
+                                ops.normalize_key(sub.tensor(sort_col)))
+                            sub = sub.gather(perm)
+                        pieces.append(sub)
+                    batch = ColumnBatch.concat(pieces)
             if cache:
                 cache.put(key, batch, seg)
 
@@ -659,27 +669,6 @@ def _join_key_tensors(lbatch: ColumnBatch, rbatch: ColumnBatch,
     rk = rlut[rcol.codes.long()] if len(rcol.values) else \
         rcol.codes.long()
     return lk, rk
-
-
-def _merge_two_sorted_runs(sub: ColumnBatch, sort_col: str,
-                           n_a: int) -> ColumnBatch:
-    """Merge a bucket whose rows are two sorted runs (rows [0, n_a) and
-    [n_a, n)) into one sorted run: each element's merged position is its
-    local rank plus its rank in the other run (two searchsorted calls),
-    with run-A elements stably preceding equal run-B elements."""
-    keys = ops.normalize_key(sub.tensor(sort_col))
-    sortable = keys ^ (-0x8000000000000000)
-    ka, kb = sortable[:n_a], sortable[n_a:]
-    dev = keys.device
-    pos_a = torch.arange(n_a, dtype=torch.int64, device=dev) + \
-        torch.searchsorted(kb, ka, right=False)
-    pos_b = torch.arange(kb.numel(), dtype=torch.int64, device=dev) + \
-        torch.searchsorted(ka, kb, right=True)
-    perm = torch.empty(sub.num_rows, dtype=torch.int64, device=dev)
-    perm[pos_a] = torch.arange(n_a, dtype=torch.int64, device=dev)
-    perm[pos_b] = torch.arange(n_a, sub.num_rows, dtype=torch.int64,
-                               device=dev)
-    return sub.gather(perm)
 
 
 def _drop_null_keys(batch: ColumnBatch, key_name: str,

@@ -27,7 +27,7 @@ from .cpu_ref import MASK32, SPARK_HASH_SEED  # re-export
 
 __all__ = [
     "murmur3_bucket", "normalize_key", "sort_pairs", "sort_perm",
-    "merge_join", "select_range_u64", "isin_sorted", "segmented_minmax",
+    "merge_join", "run_merge_perm", "select_range_u64", "isin_sorted", "segmented_minmax",
     "bloom_build", "bloom_probe", "zorder_key", "gather_rows", "native",
 ]
 
@@ -80,6 +80,16 @@ def sort_perm(keys_u64: torch.Tensor) -> torch.Tensor:
     payload = torch.arange(n, dtype=torch.int64, device=keys_u64.device)
     _, perm = sort_pairs(keys_u64, payload)
     return perm
+
+
+def run_merge_perm(keys_u64: torch.Tensor, seg: torch.Tensor,
+                   split: torch.Tensor) -> torch.Tensor:
+    """Segmented two-sorted-run merge permutation (K4b): one launch
+    merges every bucket's [A-run | B-run] layout into sorted order."""
+    if keys_u64.is_cuda:
+        return native.ext().run_merge_perm(keys_u64.contiguous(), seg,
+                                           split)
+    return cpu_ref.run_merge_perm(keys_u64, seg, split)
 
 
 def merge_join(lkeys: torch.Tensor, rkeys: torch.Tensor,

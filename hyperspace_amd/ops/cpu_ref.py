@@ -283,3 +283,26 @@ def zorder_key(cols_u64: List[torch.Tensor], bits_per_col: int
             out_pos = 63 - (b * n_cols + c)
             z = z | (bit << out_pos)
     return z
+
+
+def run_merge_perm(keys_u64: torch.Tensor, seg: torch.Tensor,
+                   split: torch.Tensor) -> torch.Tensor:
+    """Segmented two-sorted-run merge permutation (K4b reference).
+
+    Each segment b's rows [seg[b], seg[b+1]) hold two sorted runs split
+    at split[b] (== seg[b+1] when the segment is a single run); returns
+    perm such that gathering by it sorts each segment, with A-rows
+    stably preceding equal B-rows."""
+    sortable = _as_unsigned_sortable(keys_u64)
+    n = keys_u64.numel()
+    perm = torch.empty(n, dtype=torch.int64)
+    for b in range(seg.numel() - 1):
+        a0, b1, sp = int(seg[b]), int(seg[b + 1]), int(split[b])
+        ka, kb = sortable[a0:sp], sortable[sp:b1]
+        pos_a = torch.arange(sp - a0) + torch.searchsorted(
+            kb, ka, right=False)
+        pos_b = torch.arange(b1 - sp) + torch.searchsorted(
+            ka, kb, right=True)
+        perm[a0 + pos_a] = torch.arange(a0, sp)
+        perm[a0 + pos_b] = torch.arange(sp, b1)
+    return perm

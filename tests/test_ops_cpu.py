@@ -166,3 +166,40 @@ def test_zorder_locality():
     d01 = abs(int(z[1]) - int(z[0]))
     d02 = abs(int(z[2]) - int(z[0]))
     assert d01 < d02
+
+
+def test_run_merge_perm_matches_sort():
+    """K4b segmented two-run merge == stable per-segment sort."""
+    import numpy as np
+    from hyperspace_amd import ops
+    rng = np.random.default_rng(17)
+    seg_sizes = [0, 10, 1, 177, 64]
+    seg = torch.tensor(np.concatenate([[0], np.cumsum(
+        [a + b for a, b in zip(seg_sizes, seg_sizes[::-1])])]))
+    keys_parts, split = [], []
+    pos = 0
+    for na, nb in zip(seg_sizes, seg_sizes[::-1]):
+        keys_parts.append(np.sort(rng.integers(-50, 50, na)))
+        keys_parts.append(np.sort(rng.integers(-50, 50, nb)))
+        split.append(pos + na)
+        pos += na + nb
+    raw = torch.tensor(np.concatenate(keys_parts), dtype=torch.int64)
+    keys = ops.normalize_key(raw)
+    perm = ops.run_merge_perm(keys, seg, torch.tensor(split))
+    merged = raw[perm]
+    for s_i in range(seg.numel() - 1):
+        a, b = int(seg[s_i]), int(seg[s_i + 1])
+        part = merged[a:b]
+        assert (part[1:] >= part[:-1]).all()
+        # same multiset
+        assert sorted(part.tolist()) == sorted(raw[a:b].tolist())
+    # stability: A rows (idx < split) precede equal B rows
+    for s_i in range(seg.numel() - 1):
+        a, b = int(seg[s_i]), int(seg[s_i + 1])
+        sp = split[s_i]
+        p = perm[a:b]
+        k = raw[p]
+        for j in range(1, b - a):
+            if k[j] == k[j - 1]:
+                # equal keys: A-origin must not follow B-origin
+                assert not (p[j] < sp <= p[j - 1]), (s_i, j)

@@ -282,3 +282,22 @@ def test_device_decode_multi_rowgroup_dict(tmp_path):
     for name in cols:
         assert np.array_equal(dev.tensor(name).cpu().numpy(),
                               ref.column(name).to_numpy()), name
+
+
+def test_run_merge_perm_device_matches_cpu():
+    rng = np.random.default_rng(23)
+    seg_sizes = [(1000, 1000), (0, 500), (700, 0), (1, 2999)]
+    seg = [0]
+    split = []
+    parts = []
+    for na, nb in seg_sizes:
+        parts.append(np.sort(rng.integers(-(10**9), 10**9, na)))
+        parts.append(np.sort(rng.integers(-(10**9), 10**9, nb)))
+        split.append(seg[-1] + na)
+        seg.append(seg[-1] + na + nb)
+    raw = torch.tensor(np.concatenate(parts), dtype=torch.int64)
+    keys = ops.cpu_ref.normalize_key(raw)
+    seg_t, split_t = torch.tensor(seg), torch.tensor(split)
+    cpu_perm = ops.cpu_ref.run_merge_perm(keys, seg_t, split_t)
+    dev_perm = ops.run_merge_perm(keys.cuda(), seg_t, split_t).cpu()
+    assert torch.equal(cpu_perm, dev_perm)
