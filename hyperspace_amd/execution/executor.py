@@ -706,9 +706,44 @@ def _empty_batch(schema) -> ColumnBatch:
     return ColumnBatch(cols)
 
 
+def _eval_arith(batch: ColumnBatch, e) -> torch.Tensor:
+    """Evaluate a scalar arithmetic expression per row (Arith trees over
+    Col/Lit; '%' uses Java/Spark remainder = fmod)."""
+    from ..plan.expr import Arith
+    if isinstance(e, Col):
+        col = batch.column(e.name)
+        if isinstance(col, StringColumn):
+            raise HyperspaceException(
+                "arithmetic over string columns is not supported")
+        return col
+    if isinstance(e, Lit):
+        return e.value
+    if isinstance(e, Arith):
+        lv = _eval_arith(batch, e.left)
+        rv = _eval_arith(batch, e.right)
+        if e.op == "+":
+            return lv + rv
+        if e.op == "-":
+            return lv - rv
+        if e.op == "*":
+            return lv * rv
+        if e.op == "%":
+            return torch.fmod(lv, rv)
+        lt = lv.to(torch.float64) if torch.is_tensor(lv) else lv
+        return lt / rv
+    raise HyperspaceException(f"Cannot evaluate expression {e!r}")
+
+
 def _compare(batch: ColumnBatch, e: BinComp) -> torch.Tensor:
     """General comparison mask (non-hot path; hot single comparisons lower
     to select_range_u64 in _predicate_indices)."""
+    from ..plan.expr import Arith
+    if isinstance(e.left, Arith):
+        lv = _eval_arith(batch, e.left)
+        rv = (_eval_arith(batch, e.right)
+              if not isinstance(e.right, Lit) else e.right.value)
+        return {"=": lv == rv, "!=": lv != rv, "<": lv < rv,
+                "<=": lv <= rv, ">": lv > rv, ">=": lv >= rv}[e.op]
     if not isinstance(e.left, Col):
         raise HyperspaceException(f"Unsupported comparison {e!r}")
     col = batch.column(e.left.name)

@@ -38,7 +38,7 @@ class DataSkippingIndexConfig(IndexConfigTrait):
         return self._name
 
     def referenced_columns(self) -> List[str]:
-        return sorted({s.expr for s in self.sketches})
+        return sorted({s.base_column for s in self.sketches})
 
     def create_index(self, ctx: IndexerContext, df,
                      properties: Dict[str, str]
@@ -70,7 +70,7 @@ class DataSkippingIndexConfig(IndexConfigTrait):
                         f.type in ("long", "double"):
                     sketches.append(PartitionSketch(f.name))
         cols = resolve_all(schema.field_names(),
-                           sorted({s.expr for s in sketches}))
+                           sorted({s.base_column for s in sketches}))
 
         # pre-assign ids deterministically (shared with covering build)
         files = sorted(scan.relation.all_files(), key=lambda f: f.name)
@@ -96,10 +96,12 @@ class DataSkippingIndexConfig(IndexConfigTrait):
         index_data: Dict[str, torch.Tensor] = {
             IndexConstants.DATA_FILE_NAME_ID_COLUMN: file_ids}
         for sketch in sketches:
-            resolved = resolve_all(schema.field_names(), [sketch.expr])[0]
-            sketch.expr = resolved
-            values = batch.tensor(resolved)
-            dtype_name = schema.field_type(resolved) or "long"
+            resolved = resolve_all(schema.field_names(),
+                                   [sketch.base_column])[0]
+            sketch.rebind(resolved)
+            values = sketch.compute_values(batch.tensor(resolved))
+            dtype_name = sketch.value_type(
+                schema.field_type(resolved) or "long")
             index_data.update(sketch.aggregate(values, seg, dtype_name,
                                                batch.mask(resolved)))
 

@@ -39,7 +39,7 @@ class DataSkippingIndex(Index):
         return "DS"
 
     def indexed_columns_list(self) -> List[str]:
-        return sorted({s.expr for s in self.sketches})
+        return sorted({s.base_column for s in self.sketches})
 
     def referenced_columns(self) -> List[str]:
         return self.indexed_columns_list()
@@ -115,10 +115,12 @@ class DataSkippingIndex(Index):
             data: Dict[str, torch.Tensor] = {
                 IC.DATA_FILE_NAME_ID_COLUMN: file_ids}
             for sketch in self.sketches:
-                values = batch.tensor(sketch.expr)
-                dtype_name = self.schema.field_type(sketch.expr) or "long"
+                base = sketch.base_column
+                values = sketch.compute_values(batch.tensor(base))
+                dtype_name = sketch.value_type(
+                    self.schema.field_type(base) or "long")
                 data.update(sketch.aggregate(values, seg, dtype_name,
-                                             batch.mask(sketch.expr)))
+                                             batch.mask(base)))
             out = os.path.join(
                 ctx.index_data_path,
                 f"part-00001-{_uuid.uuid4().hex[:8]}_00000.c000.parquet")
@@ -203,8 +205,10 @@ class DataSkippingIndex(Index):
             return None
         refs = pred.references()
         for sketch in self.sketches:
-            if {r.lower() for r in refs} == {sketch.expr.lower()}:
-                dtype_name = self.schema.field_type(sketch.expr) or "long"
+            if {r.lower() for r in refs} == \
+                    {sketch.base_column.lower()}:
+                dtype_name = self.schema.field_type(
+                    sketch.base_column) or "long"
                 mask = sketch.convert_predicate(pred, sketch_data,
                                                 dtype_name)
                 if mask is not None:

@@ -17,9 +17,46 @@ from typing import Any, Dict, List, Optional
 
 import torch
 
+import re
+
 from ... import ops
 from ...exceptions import HyperspaceException
-from ...plan.expr import BinComp, Col, Expr, In, IsNotNull, Lit
+from ...plan.expr import (Arith, BinComp, Col, Expr, In, IsNotNull, Lit,
+                          _expr_eq)
+
+# single binary arithmetic over a column and a numeric literal, e.g.
+# "a % 10" — the supported scalar-expression subset (the reference
+# accepts arbitrary deterministic scalar expressions via Catalyst;
+# here the predicate language is comparison-based, so sketches bind to
+# the same single-op arithmetic the Expr API can express)
+_ARITH_RE = re.compile(
+    r"^\s*([A-Za-z_][\w.]*)\s*([+\-*/%])\s*(-?\d+(?:\.\d+)?)\s*$")
+
+
+def parse_sketch_expr(s: str):
+    """'col' or 'col OP literal' -> (col_name, op|None, literal|None)."""
+    m = _ARITH_RE.match(s)
+    if m:
+        col_s, op, lit = m.groups()
+        value = float(lit) if "." in lit else int(lit)
+        return col_s, op, value
+    return s.strip(), None, None
+
+
+def eval_sketch_values(values: torch.Tensor, op, lit) -> torch.Tensor:
+    """Apply the sketch's arithmetic to a column tensor.  ``%`` uses
+    Java/Spark remainder semantics (sign of the dividend: fmod)."""
+    if op is None:
+        return values
+    if op == "+":
+        return values + lit
+    if op == "-":
+        return values - lit
+    if op == "*":
+        return values * lit
+    if op == "%":
+        return torch.fmod(values, lit)
+    return values.to(torch.float64) / lit  # '/' is double division
 
 MINMAX_SKETCH_TYPE = (
     "com.microsoft.hyperspace.index.dataskipping.sketches.MinMaxSketch")
@@ -31,10 +68,44 @@ PARTITION_SKETCH_TYPE = (
 
 
 class Sketch:
-    """Base sketch over a single source column expression."""
+    """Base sketch over a single source column expression (a column
+    name or ``col OP literal`` arithmetic)."""
 
     def __init__(self, expr: str):
         self.expr = expr
+
+    @property
+    def base_column(self) -> str:
+        return parse_sketch_expr(self.expr)[0]
+
+    def _parsed(self):
+        return parse_sketch_expr(self.expr)
+
+    def rebind(self, resolved_col: str) -> None:
+        """Rewrite the expression with the schema-resolved column."""
+        _, op, lit = self._parsed()
+        self.expr = (f"{resolved_col} {op} {lit}" if op is not None
+                     else resolved_col)
+
+    def compute_values(self, col_values: torch.Tensor) -> torch.Tensor:
+        _, op, lit = self._parsed()
+        return eval_sketch_values(col_values, op, lit)
+
+    def value_type(self, col_type: str) -> str:
+        _, op, _ = self._parsed()
+        if op == "/":
+            return "double"
+        return col_type
+
+    def matches_lhs(self, lhs: Expr) -> bool:
+        """True when a predicate's left-hand side is structurally this
+        sketch's expression."""
+        name, op, lit = self._parsed()
+        if op is None:
+            return isinstance(lhs, Col) and \
+                lhs.name.lower() == name.lower()
+        return isinstance(lhs, Arith) and _expr_eq(
+            lhs, Arith(op, Col(name), Lit(lit)))
 
     @property
     def kind(self) -> str:
@@ -111,10 +182,9 @@ class MinMaxSketch(Sketch):
     def convert_predicate(self, pred, sketch_data, dtype_name):
         mn = sketch_data.tensor(self.out_columns()[0])
         mx = sketch_data.tensor(self.out_columns()[1])
-        col_l = self.expr.lower()
-        if isinstance(pred, BinComp) and isinstance(pred.left, Col) and \
-                pred.left.name.lower() == col_l and \
-                isinstance(pred.right, Lit):
+        dtype_name = self.value_type(dtype_name)
+        if isinstance(pred, BinComp) and isinstance(pred.right, Lit) and \
+                self.matches_lhs(pred.left):
             v = _norm_scalar(pred.right.value, dtype_name)
             if pred.op == "=":
                 return (mn <= v) & (mx >= v)
@@ -127,15 +197,13 @@ class MinMaxSketch(Sketch):
             if pred.op == ">=":
                 return mx >= v
             return None  # != not convertible
-        if isinstance(pred, In) and isinstance(pred.col, Col) and \
-                pred.col.name.lower() == col_l:
+        if isinstance(pred, In) and self.matches_lhs(pred.col):
             out = torch.zeros(mn.numel(), dtype=torch.bool)
             for value in pred.values:
                 v = _norm_scalar(value, dtype_name)
                 out |= (mn <= v) & (mx >= v)
             return out
-        if isinstance(pred, IsNotNull) and isinstance(pred.col, Col) and \
-                pred.col.name.lower() == col_l:
+        if isinstance(pred, IsNotNull) and self.matches_lhs(pred.col):
             return torch.ones(mn.numel(), dtype=torch.bool)
         return None
 
@@ -195,14 +263,13 @@ class BloomFilterSketch(Sketch):
                                  dtype=torch.int64)}
 
     def convert_predicate(self, pred, sketch_data, dtype_name):
-        col_l = self.expr.lower()
+        dtype_name = self.value_type(dtype_name)
         values = None
         if isinstance(pred, BinComp) and pred.op == "=" and \
-                isinstance(pred.left, Col) and \
-                pred.left.name.lower() == col_l and \
-                isinstance(pred.right, Lit):
+                isinstance(pred.right, Lit) and \
+                self.matches_lhs(pred.left):
             values = [pred.right.value]
-        elif isinstance(pred, In) and pred.col.name.lower() == col_l:
+        elif isinstance(pred, In) and self.matches_lhs(pred.col):
             values = pred.values
         if values is None:
             return None
@@ -267,11 +334,9 @@ class PartitionSketch(Sketch):
 
     def convert_predicate(self, pred, sketch_data, dtype_name):
         first = sketch_data.tensor(self.out_columns()[0])
-        col_l = self.expr.lower()
         if isinstance(pred, BinComp) and pred.op == "=" and \
-                isinstance(pred.left, Col) and \
-                pred.left.name.lower() == col_l and \
-                isinstance(pred.right, Lit):
+                isinstance(pred.right, Lit) and \
+                self.matches_lhs(pred.left):
             v = torch.tensor(pred.right.value, dtype=first.dtype)
             return first == v
         return None

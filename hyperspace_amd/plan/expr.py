@@ -26,6 +26,24 @@ class Expr:
     def __invert__(self):
         return Not(self)
 
+    # arithmetic sugar: (col("a") % 10) == 3 — scalar expressions the
+    # data-skipping sketches can bind to (reference ExpressionUtils
+    # accepts arbitrary deterministic scalar expressions)
+    def __add__(self, other):
+        return Arith("+", self, _wrap(other))
+
+    def __sub__(self, other):
+        return Arith("-", self, _wrap(other))
+
+    def __mul__(self, other):
+        return Arith("*", self, _wrap(other))
+
+    def __mod__(self, other):
+        return Arith("%", self, _wrap(other))
+
+    def __truediv__(self, other):
+        return Arith("/", self, _wrap(other))
+
 
 class Col(Expr):
     def __init__(self, name: str):
@@ -96,6 +114,70 @@ class BinComp(Expr):
 
     def __repr__(self):
         return f"({self.left!r} {self.op} {self.right!r})"
+
+
+class Arith(Expr):
+    """Scalar arithmetic over columns/literals: +, -, *, %, /.
+
+    ``%`` follows Java/Spark remainder semantics (sign of the dividend).
+    Comparisons on an Arith left-hand side evaluate it per row; the
+    data-skipping sketches match predicates whose LHS is structurally
+    equal to the sketch's expression."""
+
+    OPS = ("+", "-", "*", "%", "/")
+
+    def __init__(self, op: str, left: Expr, right: Expr):
+        if op not in self.OPS:
+            raise HyperspaceException(f"Bad arithmetic op {op}")
+        self.op = op
+        self.left = left
+        self.right = right
+
+    def references(self):
+        return self.left.references() | self.right.references()
+
+    def __eq__(self, other):  # type: ignore[override]
+        if isinstance(other, Arith):
+            # structural equality when both sides are expression trees
+            # (sketch matching); comparison against anything else builds
+            # a predicate like Col.__eq__ does
+            return (self.op == other.op and
+                    _expr_eq(self.left, other.left) and
+                    _expr_eq(self.right, other.right))
+        return BinComp("=", self, _wrap(other))
+
+    def __ne__(self, other):  # type: ignore[override]
+        return BinComp("!=", self, _wrap(other))
+
+    def __lt__(self, other):
+        return BinComp("<", self, _wrap(other))
+
+    def __le__(self, other):
+        return BinComp("<=", self, _wrap(other))
+
+    def __gt__(self, other):
+        return BinComp(">", self, _wrap(other))
+
+    def __ge__(self, other):
+        return BinComp(">=", self, _wrap(other))
+
+    def __hash__(self):
+        return hash(("Arith", self.op, repr(self)))
+
+    def __repr__(self):
+        return f"({self.left!r} {self.op} {self.right!r})"
+
+
+def _expr_eq(a: "Expr", b: "Expr") -> bool:
+    """Structural expression equality (case-insensitive column names)."""
+    if isinstance(a, Col) and isinstance(b, Col):
+        return a.name.lower() == b.name.lower()
+    if isinstance(a, Lit) and isinstance(b, Lit):
+        return a.value == b.value
+    if isinstance(a, Arith) and isinstance(b, Arith):
+        return (a.op == b.op and _expr_eq(a.left, b.left) and
+                _expr_eq(a.right, b.right))
+    return False
 
 
 class And(Expr):

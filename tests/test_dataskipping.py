@@ -204,3 +204,44 @@ def test_zorder_refresh_full(env, tmp_path):
     entry = session.index_manager().get_index("zr4")
     assert entry.derivedDataset.kind == "ZOrderCoveringIndex"
     assert len(entry.source_file_infos()) == 9
+
+
+def test_sketch_over_arithmetic_expression(tmp_path, monkeypatch):
+    """Sketches over scalar expressions (reference ExpressionUtils
+    accepts deterministic scalar exprs): MinMaxSketch('key % 100')
+    converts predicates whose LHS is structurally the same expression."""
+    from hyperspace_amd.plan.expr import col
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "idx"))
+    d = tmp_path / "data"
+    d.mkdir()
+    rng = np.random.default_rng(91)
+    # file i holds keys whose (key % 100) lies in [i*10, i*10+10)
+    all_keys = []
+    for i in range(8):
+        base = rng.integers(0, 50, 4000) * 100
+        key = base + rng.integers(i * 10, i * 10 + 10, 4000)
+        all_keys.append(key)
+        pq.write_table(pa.table({"key": key, "val": rng.random(4000)}),
+                       str(d / f"part-{i}.parquet"))
+    session = hs.HyperspaceSession(device="cpu")
+    h = hs.Hyperspace(session)
+    df = session.read_parquet(str(d))
+    h.create_index(df, hs.DataSkippingIndexConfig(
+        "dse", hs.MinMaxSketch("key % 100")))
+    session.enable_hyperspace()
+    q = df.filter((col("key") % 100) == 42).select("key", "val")
+    plan = q.optimized_plan()
+    leaf = plan.collect_leaves()[0]
+    assert leaf.file_subset is not None and len(leaf.file_subset) == 1
+    out = Executor(session).execute(plan)
+    expected = sum(int((k % 100 == 42).sum()) for k in all_keys)
+    assert out.num_rows == expected
+    assert (out.tensor("key") % 100 == 42).all()
+    # range over the expression
+    q2 = df.filter((col("key") % 100) >= 55)
+    leaf2 = q2.optimized_plan().collect_leaves()[0]
+    # files 5,6,7 cover remainders [50,80)
+    assert leaf2.file_subset is not None and len(leaf2.file_subset) == 3
+    out2 = Executor(session).execute(q2.optimized_plan())
+    expected2 = sum(int((k % 100 >= 55).sum()) for k in all_keys)
+    assert out2.num_rows == expected2
