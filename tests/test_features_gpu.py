@@ -132,3 +132,28 @@ def test_gpu_partitioned_source(tmp_path, monkeypatch):
     session.enable_hyperspace()
     out2 = df.filter("key = 5").select("key", "val", "day").collect()
     assert (out2.tensor("day") > 0).all()
+
+
+def test_gpu_arithmetic_sketch_filter(tmp_path, monkeypatch):
+    from hyperspace_amd.plan.expr import col
+    session, h = _session(tmp_path, monkeypatch, buckets=4)
+    rng = np.random.default_rng(76)
+    d = tmp_path / "ar"
+    d.mkdir()
+    keys = []
+    for i in range(4):
+        k = rng.integers(0, 1000, 50_000) * 10 + i
+        keys.append(k)
+        pq.write_table(pa.table({"key": k, "val": rng.random(50_000)}),
+                       str(d / f"part-{i}.parquet"))
+    df = session.read_parquet(str(d))
+    h.create_index(df, hs.DataSkippingIndexConfig(
+        "gar", hs.MinMaxSketch("key % 10")))
+    session.enable_hyperspace()
+    q = df.filter((col("key") % 10) == 2).select("key", "val")
+    plan = q.optimized_plan()
+    leaf = plan.collect_leaves()[0]
+    assert leaf.file_subset is not None and len(leaf.file_subset) == 1
+    out = Executor(session).execute(plan)
+    assert out.num_rows == 50_000
+    assert bool((out.tensor("key") % 10 == 2).all())
