@@ -108,3 +108,30 @@ def test_wide_refresh_and_optimize(env, tmp_path):
     t9 = pq.read_table(src + "/part-9.parquet")
     want += sum(1 for k in t9.column("key").to_pylist() if k == 7)
     assert out.num_rows == want
+
+
+def test_timestamp_column_index(tmp_path, monkeypatch):
+    """Timestamp columns map to int64 epochs in the columnar model and
+    index/sort/filter like any numeric column (reference indexes Spark
+    TimestampType natively)."""
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "idx"))
+    rng = np.random.default_rng(81)
+    d = tmp_path / "ts"
+    d.mkdir()
+    base = np.datetime64("2025-01-01T00:00:00", "us")
+    offs = rng.integers(0, 365 * 24 * 3600, 10_000).astype("timedelta64[s]")
+    ts = base + offs
+    v = rng.random(10_000)
+    pq.write_table(pa.table({"ts": ts, "v": v}),
+                   str(d / "part-0.parquet"))
+    session = hs.HyperspaceSession(device="cpu")
+    session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 4)
+    h = hs.Hyperspace(session)
+    df = session.read_parquet(str(d))
+    h.create_index(df, hs.CoveringIndexConfig("tsx", ["ts"], ["v"]))
+    session.enable_hyperspace()
+    target = int(ts[0].astype("datetime64[us]").astype(np.int64))
+    out = df.filter(f"ts = {target}").select("ts", "v").collect()
+    want = int((ts.astype("datetime64[us]").astype(np.int64)
+                == target).sum())
+    assert out.num_rows == want and want >= 1
