@@ -61,6 +61,8 @@ def test_lifecycle_fuzz(tmp_path_factory, ops, seed):
     h = hs.Hyperspace(session)
     df = session.read_parquet(str(d))
     h.create_index(df, hs.CoveringIndexConfig("fz", ["key"], ["val"]))
+    h.create_index(df, hs.DataSkippingIndexConfig(
+        "fzds", hs.MinMaxSketch("key")))
     session.enable_hyperspace()
     deleted_state = False
 
@@ -78,6 +80,7 @@ def test_lifecycle_fuzz(tmp_path_factory, ops, seed):
             del live[p]
         elif op == "refresh_inc" and not deleted_state:
             h.refresh_index("fz", mode="incremental")
+            h.refresh_index("fzds", mode="incremental")
         elif op == "refresh_full" and not deleted_state:
             h.refresh_index("fz", mode="full")
         elif op == "refresh_quick" and not deleted_state:
