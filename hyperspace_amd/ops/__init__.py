@@ -6,14 +6,18 @@ torch reference implementation (cpu_ref.py).  On CUDA the native extension
 is REQUIRED — there is no silent PyTorch fallback (KernelUnavailableError).
 
 Op -> reference data-plane mapping:
-  murmur3_bucket   K2  hash partitioning (Spark HashPartitioning-compatible)
+  murmur3_bucket   K2  hash partitioning (Spark HashPartitioning-compatible,
+                       null keys pass the seed through)
   sort_pairs       K3  per-bucket sort (stable LSD radix on device)
-  merge_join       K4  co-bucketed sort-merge join
+  merge_join       K4  co-bucketed sort-merge join (two-phase per-tile)
+  run_merge_perm   K4b segmented two-sorted-run merge (multi-file buckets)
   select_range     K1/filter scan predicate + compaction
   isin_sorted      K7  lineage delete filter
   segmented_minmax K8  data-skipping MinMax sketch
   bloom_build/probe K8 data-skipping BloomFilter sketch
   zorder_key       K10 z-address bit interleave
+  (extension-only: copy_unaligned/rle_decode/snappy_decompress — the K1
+  parquet page decode surface used by sources/parquet_io.py)
 """
 
 from __future__ import annotations
