@@ -301,3 +301,22 @@ def test_join_correct_after_incremental_refresh(env, tmp_path):
     assert accel.num_rows == baseline.num_rows
     cols = ["orderkey", "qty", "status"]
     assert _sorted_rows(accel, cols) == _sorted_rows(baseline, cols)
+
+
+def test_covering_index_serves_arithmetic_filter(env):
+    """An arithmetic predicate over the first indexed column still
+    rewrites to the index-only scan; the executor evaluates the
+    expression over index data."""
+    from hyperspace_amd.plan.expr import col as _col
+    session, h, left, _ = env
+    h.create_index(left, hs.CoveringIndexConfig("arx", ["qty"],
+                                                ["price"]))
+    session.enable_hyperspace()
+    q = left.filter((_col("qty") % 7) == 3).select("qty", "price")
+    plan = q.optimized_plan()
+    assert any(isinstance(l, IndexScan) for l in plan.collect_leaves())
+    out = q.collect()
+    session.disable_hyperspace()
+    base = q.collect()
+    assert out.num_rows == base.num_rows
+    assert bool((out.tensor("qty") % 7 == 3).all())
