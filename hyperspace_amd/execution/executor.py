@@ -468,6 +468,14 @@ class Executor:
             return torch.zeros(batch.num_rows, dtype=torch.bool,
                                device=batch.device)
         if isinstance(e, In):
+            from ..plan.expr import Arith
+            if isinstance(e.col, Arith):
+                t = _eval_arith(batch, e.col).to(torch.int64)
+                vs = torch.tensor(sorted(int(v) for v in e.values),
+                                  dtype=torch.int64, device=t.device)
+                m = ops.isin_sorted(t, vs)
+                v = self._valid_of(batch, e)
+                return m & v if v is not None else m
             col = batch.column(e.col.name)
             vals = e.values
             if isinstance(col, StringColumn):

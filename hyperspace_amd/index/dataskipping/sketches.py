@@ -15,9 +15,9 @@ from __future__ import annotations
 
 from typing import Any, Dict, List, Optional
 
-import torch
-
 import re
+
+import torch
 
 from ... import ops
 from ...exceptions import HyperspaceException

@@ -44,6 +44,9 @@ class Expr:
     def __truediv__(self, other):
         return Arith("/", self, _wrap(other))
 
+    def isin(self, values):
+        return In(self, list(values))
+
 
 class Col(Expr):
     def __init__(self, name: str):
@@ -69,9 +72,6 @@ class Col(Expr):
 
     def __ge__(self, other):
         return BinComp(">=", self, _wrap(other))
-
-    def isin(self, values: Sequence[Any]):
-        return In(self, list(values))
 
     def is_not_null(self):
         return IsNotNull(self)

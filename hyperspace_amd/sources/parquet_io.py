@@ -389,7 +389,6 @@ def read_files_batch_device(paths: List[str], device,
                 # per (chunk, bit-width) instead of per page — the
                 # per-page host->device run uploads dominate otherwise
                 zbatch: Dict[int, list] = {}
-                zbatch_rows = 0
                 for j, page in enumerate(c.pages):
                     nv = page[3]
                     base = int(page_base[j])
