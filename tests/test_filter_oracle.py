@@ -76,3 +76,40 @@ def test_filter_values_match_numpy(env):
         .collect().to_numpy()
     got = sorted(zip(out["key"].tolist(), out["qty"].tolist()))
     assert got == expected
+
+
+def test_arith_filter_oracle_with_nulls(tmp_path, monkeypatch):
+    """Arithmetic predicates vs a pandas oracle over nullable data:
+    null inputs never match, Java-style remainder on negatives."""
+    import pandas as pd
+    from hyperspace_amd.plan.expr import col
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "idx"))
+    rng = np.random.default_rng(37)
+    d = tmp_path / "ar"
+    d.mkdir()
+    k = rng.integers(-1000, 1000, 20000)
+    mask = rng.random(20000) > 0.1
+    pq.write_table(pa.table({"k": pa.array(k, mask=~mask),
+                             "v": rng.random(20000)}),
+                   str(d / "p.parquet"), compression="NONE",
+                   use_dictionary=False, data_page_version="1.0")
+    session = hs.HyperspaceSession(device="cpu")
+    df = session.read_parquet(str(d))
+    kv = pd.Series(k).where(mask)  # NaN at nulls
+
+    def java_mod(s, m):
+        # sign of the dividend (np.fmod), unlike python %
+        return np.fmod(s, m)
+
+    cases = [
+        ((col("k") % 7) == 3, java_mod(kv, 7) == 3),
+        ((col("k") * 2) >= 500, kv * 2 >= 500),
+        ((col("k") + 10) < -200, kv + 10 < -200),
+        ((col("k") - 1) != 0, (kv - 1).notna() & ((kv - 1) != 0)),
+        ((col("k") / 4) <= -100.0, kv / 4 <= -100.0),
+        ((col("k") % 9).isin([0, 4]), java_mod(kv, 9).isin([0, 4])),
+    ]
+    for pred, oracle in cases:
+        got = df.filter(pred).collect().num_rows
+        want = int(oracle.fillna(False).sum())
+        assert got == want, repr(pred)
