@@ -119,16 +119,15 @@ class Executor:
         else:
             batch, row_counts = _empty_batch(plan.relation.schema), []
         if lineage_tracker is not None:
-            tracker = lineage_tracker
-            ids = []
-            for f, n in zip(files, row_counts):
-                fid = tracker.add_file(f.name, f.size, f.modifiedTime)
-                ids.append(torch.full((n,), fid, dtype=torch.int64))
-            lineage = (torch.cat(ids) if ids
-                       else torch.empty(0, dtype=torch.int64))
+            # device-side lineage materialization (see ScanStream)
+            fids = torch.tensor(
+                [lineage_tracker.add_file(f.name, f.size, f.modifiedTime)
+                 for f in files], dtype=torch.int64, device=batch.device)
+            counts = torch.tensor(row_counts, dtype=torch.int64,
+                                  device=batch.device)
             batch = batch.with_column(
                 IndexConstants.DATA_FILE_NAME_ID_COLUMN,
-                lineage.to(batch.device))
+                torch.repeat_interleave(fids, counts))
         if self.device.type == "cuda":
             batch = batch.to(self.device)
         return batch

@@ -67,15 +67,17 @@ class ScanStream:
                 batch, row_counts = read_files_batch(
                     paths, columns=self.columns or None)
         if self.tracker is not None:
-            ids = []
-            for f, n in zip(group, row_counts):
-                fid = self.tracker.add_file(f.name, f.size, f.modifiedTime)
-                ids.append(torch.full((n,), fid, dtype=torch.int64))
-            lineage = (torch.cat(ids) if ids
-                       else torch.empty(0, dtype=torch.int64))
+            # materialize the lineage column ON DEVICE (two tiny H2D
+            # uploads + repeat_interleave) — a host torch.full/cat of
+            # the full column costs GBs of pageable traffic
+            fids = torch.tensor(
+                [self.tracker.add_file(f.name, f.size, f.modifiedTime)
+                 for f in group], dtype=torch.int64, device=batch.device)
+            counts = torch.tensor(row_counts, dtype=torch.int64,
+                                  device=batch.device)
             batch = batch.with_column(
                 IndexConstants.DATA_FILE_NAME_ID_COLUMN,
-                lineage.to(batch.device))
+                torch.repeat_interleave(fids, counts))
         return batch
 
     def batches(self) -> Iterator[ColumnBatch]:
