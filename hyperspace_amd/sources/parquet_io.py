@@ -360,7 +360,6 @@ def read_files_batch_device(paths: List[str], device,
                 page_base = doff_all[1:-1] if has_zdict \
                     else doff_all[:-1]
                 dict_vals = None
-                host_all = None
                 if has_zdict:
                     dict_vals = _torch.empty(
                         dict_n + 1, dtype=out[c.name].dtype,
@@ -368,60 +367,69 @@ def read_files_batch_device(paths: List[str], device,
                     ext.copy_unaligned(scratch, 0, dict_vals, 0,
                                        dict_n * itemsize)
                     dict_vals = dict_vals[:dict_n].contiguous()
-                    # dict chunks are small: one D2H yields every page's
-                    # level prefix + RLE index stream (positions align
-                    # with scratch, so run bit-offsets need no fixup)
-                    host_all = scratch[:int(doff_all[-1])].cpu()
-                else:
-                    lvl_skips = [0] * len(c.pages)
-                    with_lvl = [j for j, p in enumerate(c.pages)
-                                if p[5]]
-                    if with_lvl:
-                        pref = _torch.stack([
-                            scratch[int(page_base[j]):
-                                    int(page_base[j]) + 4]
-                            for j in with_lvl]).cpu()
-                        lens = pref.numpy().view("<u4").ravel()
-                        for j, ln in zip(with_lvl, lens):
-                            lvl_skips[j] = 4 + int(ln)
+                # level-length prefixes for every leveled page in one
+                # small D2H (4 bytes each); the index streams and level
+                # bytes D2H per page below — NEVER the whole chunk
+                # (plain page payloads can be hundreds of MB)
+                lvl_skips = [0] * len(c.pages)
+                with_lvl = [j for j, p in enumerate(c.pages) if p[5]]
+                if with_lvl:
+                    pref = _torch.stack([
+                        scratch[int(page_base[j]):
+                                int(page_base[j]) + 4]
+                        for j in with_lvl]).cpu()
+                    lens = pref.numpy().view("<u4").ravel()
+                    for j, ln in zip(with_lvl, lens):
+                        lvl_skips[j] = 4 + int(ln)
                 nullable_chunk = c.name in out_masks
+                # ONE D2H for every byte the host needs (index streams +
+                # nullable level bytes): per-page .cpu() calls each sync
+                # the stream and serialize the whole decode
+                regions = {}  # j -> (start in cat, length, abs base)
+                cat_parts = []
+                cur = 0
+                for j, page in enumerate(c.pages):
+                    base = int(page_base[j])
+                    spans = []
+                    if page[5] and nullable_chunk:
+                        spans.append(("lvl", base + 4,
+                                      base + lvl_skips[j]))
+                    if page[0] == "dict_z":
+                        skip_j = lvl_skips[j] if page[5] else 0
+                        spans.append(("idx", base + skip_j,
+                                      base + int(page[4])))
+                    for tag, a, b in spans:
+                        regions[(j, tag)] = (cur, b - a, a)
+                        cat_parts.append(scratch[a:b])
+                        cur += b - a
+                hb_all = (_torch.cat(cat_parts).cpu() if cat_parts
+                          else None)
                 # batch unmasked dict_z pages: one rle_decode + gather
-                # per (chunk, bit-width) instead of per page — the
-                # per-page host->device run uploads dominate otherwise
+                # per (chunk, bit-width) instead of per page
                 zbatch: Dict[int, list] = {}
                 for j, page in enumerate(c.pages):
                     nv = page[3]
                     base = int(page_base[j])
-                    # level prefix + (for nullable chunks) the decoded
-                    # validity mask; level bytes are small, so a D2H of
-                    # just that region feeds the host def decoder
                     skip = 0
                     pmask = None
                     if page[5]:
-                        if has_zdict:
-                            ln = int(host_all[base:base + 4].numpy()
-                                     .view("<u4")[0])
-                            lvl_bytes = host_all[base + 4:
-                                                 base + 4 + ln].numpy()
-                        else:
-                            ln = lvl_skips[j] - 4
-                            lvl_bytes = None
+                        ln = lvl_skips[j] - 4
                         skip = 4 + ln
                         if nullable_chunk:
-                            if lvl_bytes is None:
-                                lvl_bytes = scratch[
-                                    base + 4:base + 4 + ln].cpu().numpy()
+                            s0, rln, _ = regions[(j, "lvl")]
+                            lvl_bytes = hb_all[s0:s0 + rln].numpy()
                             from .native_parquet import _decode_defs
                             pmask = _decode_defs(
                                 lvl_bytes.tobytes(), 0, ln, nv)
                     n_valid = int(pmask.sum()) if pmask is not None \
                         else nv
                     if page[0] == "dict_z":
-                        hb = host_all
-                        bw = int(hb[base + skip])
+                        s0, rln, absbase = regions[(j, "idx")]
+                        bw = int(hb_all[s0])
                         runs = list(ext.parse_rle_runs(
-                            hb, base + skip + 1, base + page[4], bw,
-                            n_valid))
+                            hb_all, s0 + 1, s0 + rln, bw, n_valid))
+                        # bit-offsets: cat-local -> scratch coordinates
+                        runs[3] = runs[3] + (absbase - s0) * 8
                         if pmask is None:
                             lst = zbatch.setdefault(
                                 bw, {"runs": [], "rows": 0, "pages": []})

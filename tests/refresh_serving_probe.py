@@ -30,9 +30,16 @@ n = 27_000_000
 pq.write_table(pa.table({"key": rng.integers(0, 1 << 25, n),
                          "val": rng.random(n)}),
                work + "/fact/part-append.parquet")
+import cProfile, pstats, io as _io
+pr = cProfile.Profile()
 t0 = time.time()
+pr.enable()
 h.refresh_index("f", mode="incremental")
+pr.disable()
 torch.cuda.synchronize(); print(f"refresh incremental (+5%): {time.time()-t0:.2f}s")
+sbuf = _io.StringIO()
+pstats.Stats(pr, stream=sbuf).sort_stats("cumulative").print_stats(14)
+print("\n".join(sbuf.getvalue().splitlines()[4:22]))
 session.enable_hyperspace()
 q = fact.select("key", "val").join(dim.select("key", "status"), on="key")
 plan = q.optimized_plan()
