@@ -517,6 +517,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("decode_def_levels", &decode_def_levels,
         "host RLE-hybrid def-level decode -> bool validity");
   m.def("parse_rle_runs", &parse_rle_runs,
+        py::call_guard<py::gil_scoped_release>(),
         "host parse of an RLE/bit-packed hybrid run table");
   m.def("rle_decode", &rle_decode,
         "device RLE/bit-packed dictionary-index decode");
