@@ -218,3 +218,69 @@ def test_distributed_string_key_build(tmp_path):
     # rows for one sku live in exactly one bucket -> one owning rank
     assert sorted((res[0]["local"], res[1]["local"]))[0] == 0 or \
         res[0]["local"] + res[1]["local"] == total_42
+
+
+# ---------------------------------------------------------------------------
+# nullable-keyed distributed build (masks through the exchange)
+# ---------------------------------------------------------------------------
+
+def _nullable_worker(rank, world, tmpdir, rdv_file, results):
+    import torch.distributed as dist
+    dist.init_process_group(
+        backend="gloo", init_method=f"file://{rdv_file}",
+        rank=rank, world_size=world)
+    try:
+        os.environ["HYPERSPACE_SYSTEM_PATH"] = os.path.join(
+            tmpdir, "nindexes")
+        import hyperspace_amd as hs
+        from hyperspace_amd.plan.expr import col
+
+        session = hs.HyperspaceSession(device="cpu")
+        session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 4)
+        h = hs.Hyperspace(session)
+        df = session.read_parquet(os.path.join(tmpdir, "nfact"))
+        h.create_index(df, hs.CoveringIndexConfig("ndx", ["key"],
+                                                  ["val"]))
+        session.enable_hyperspace()
+        session.conf.set(
+            hs.IndexConstants.INDEX_FILTER_RULE_USE_BUCKET_SPEC, True)
+        from hyperspace_amd.execution.executor import Executor
+        n_null = Executor(session).execute(
+            df.filter(col("key").is_null()).optimized_plan()).num_rows
+        n_eq = Executor(session).execute(
+            df.filter("key = 0").select("key", "val")
+            .optimized_plan()).num_rows
+        t = torch.tensor([n_null, n_eq])
+        dist.all_reduce(t)
+        results[rank] = {"null_total": int(t[0]), "eq_total": int(t[1])}
+    finally:
+        dist.destroy_process_group()
+
+
+def test_distributed_nullable_build(tmp_path):
+    rng = np.random.default_rng(71)
+    d = tmp_path / "nfact"
+    d.mkdir()
+    total_null = 0
+    total_eq0 = 0
+    for i in range(4):
+        key = rng.integers(0, 50, 2000)
+        key[::41] = 0
+        mask = rng.random(2000) > 0.2
+        total_null += int((~mask).sum())
+        total_eq0 += int((mask & (key == 0)).sum())
+        pq.write_table(pa.table({"key": pa.array(key, mask=~mask),
+                                 "val": rng.random(2000)}),
+                       str(d / f"part-{i}.parquet"), compression="NONE",
+                       use_dictionary=False, data_page_version="1.0")
+    rdv = str(tmp_path / "nrdv")
+    with mp.Manager() as mgr:
+        results = mgr.dict()
+        mp.spawn(_nullable_worker,
+                 args=(WORLD, str(tmp_path), rdv, results),
+                 nprocs=WORLD, join=True)
+        res = dict(results)
+    # nulls live in ONE owned bucket (pmod(42, nb)); literal-0 rows in
+    # another: all-reduced per-rank counts must equal the generators'
+    assert res[0]["null_total"] == total_null, (res, total_null)
+    assert res[0]["eq_total"] == total_eq0, (res, total_eq0)
