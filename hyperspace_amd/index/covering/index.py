@@ -245,6 +245,11 @@ class CoveringIndex(Index):
                     continue
                 sub = sub.gather(
                     torch.nonzero(keep_mask, as_tuple=False).flatten())
+                if sub.num_rows == 0:
+                    # every row of this bucket file came from deleted
+                    # sources: the file simply disappears (Spark writes
+                    # no empty bucket files either)
+                    continue
                 out = os.path.join(ctx.index_data_path,
                                    bucket_file_name(1, b or 0))
                 write_batch_parquet(sub, out)

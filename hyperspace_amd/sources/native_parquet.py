@@ -691,6 +691,11 @@ def read_native_host(path: str, columns: Optional[List[str]] = None
     macc: Dict[str, list] = {}
     any_null: Dict[str, bool] = {}
     for c in chunks:  # row-group-major order
+        if not c.pages:  # zero-row chunk still contributes its column
+            acc.setdefault(c.name, []).append(
+                np.empty(0, dtype=c.np_dtype))
+            macc.setdefault(c.name, []).append(np.ones(0, dtype=bool))
+            continue
         for (_, off, nv), mask in zip(c.pages, c.page_masks):
             if mask is None:
                 part = np.frombuffer(data, dtype=c.np_dtype, count=nv,

@@ -254,3 +254,40 @@ def test_streaming_vs_materialized_same_rows(env, tmp_path):
     t2 = pq2.read_table(e2.content.os_files()).sort_by(
         [("key", "ascending"), ("val", "ascending")])
     assert t1.equals(t2)
+
+
+def test_optimize_after_full_bucket_deletion(tmp_path, monkeypatch):
+    """Regression (found by the lifecycle fuzzer): deleting every source
+    file that fed a bucket file used to leave a 0-row index file whose
+    empty native read broke optimize's compaction."""
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "idx"))
+    import pyarrow as pa2
+    import pyarrow.parquet as pq2
+    rng = np.random.default_rng(0)
+    d = tmp_path / "data"
+    d.mkdir()
+
+    def wf(i):
+        pq2.write_table(
+            pa2.table({"key": rng.integers(0, 200, 1500),
+                       "val": rng.random(1500)}),
+            str(d / f"part-{i:04d}.parquet"))
+
+    wf(0)
+    wf(1)
+    session = hs.HyperspaceSession(device="cpu")
+    session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 4)
+    session.conf.set(hs.IndexConstants.INDEX_LINEAGE_ENABLED, True)
+    h = hs.Hyperspace(session)
+    df = session.read_parquet(str(d))
+    h.create_index(df, hs.CoveringIndexConfig("opt0", ["key"], ["val"]))
+    wf(2)
+    os.unlink(str(d / "part-0000.parquet"))
+    os.unlink(str(d / "part-0001.parquet"))
+    h.refresh_index("opt0", mode="incremental")
+    h.optimize_index("opt0")  # used to raise KeyError('key')
+    session.enable_hyperspace()
+    t2 = pq2.read_table(str(d / "part-0002.parquet"))
+    k2 = t2.column("key").to_numpy()
+    out = df.filter("key = 5").select("key", "val").collect()
+    assert out.num_rows == int((k2 == 5).sum())
