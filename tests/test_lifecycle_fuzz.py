@@ -30,8 +30,8 @@ def _write_file(d, rng, i):
 OPS = st.lists(
     st.sampled_from(["append", "delete_file", "refresh_inc",
                      "refresh_full", "refresh_quick", "optimize",
-                     "query", "soft_delete_restore", "query",
-                     "vacuum_outdated"]),
+                     "optimize_full", "query", "soft_delete_restore",
+                     "query", "vacuum_outdated"]),
     min_size=4, max_size=12)
 
 
@@ -87,6 +87,8 @@ def test_lifecycle_fuzz(tmp_path_factory, ops, seed):
             h.refresh_index("fz", mode="quick")
         elif op == "optimize" and not deleted_state:
             h.optimize_index("fz")
+        elif op == "optimize_full" and not deleted_state:
+            h.optimize_index("fz", mode="full")
         elif op == "soft_delete_restore" and not deleted_state:
             h.delete_index("fz")
             h.restore_index("fz")
