@@ -254,10 +254,15 @@ def test_concurrent_maintenance_fuzz(env, tmp_path):
             op = r.choice(["append_refresh", "optimize", "del_restore"])
             try:
                 if op == "append_refresh":
+                    # atomic publish (write temp, rename): readers must
+                    # never see half-written source files — the same
+                    # contract Spark ingestion relies on
+                    tmp_p = data_dir + f"/.tmp-{seed}-{i}"
                     pq.write_table(
                         pa.table({"key": rng.integers(0, 100, 500),
-                                  "val": rng.random(500)}),
-                        data_dir + f"/part-{seed}-{i}.parquet")
+                                  "val": rng.random(500)}), tmp_p)
+                    os.rename(tmp_p,
+                              data_dir + f"/part-{seed}-{i}.parquet")
                     i += 1
                     h.refresh_index("cfz", mode="incremental")
                 elif op == "optimize":
