@@ -7,7 +7,7 @@ needs: column refs, literals, comparisons, boolean connectives, IN.
 from __future__ import annotations
 
 import re
-from typing import Any, List, Sequence, Set, Union
+from typing import Any, List, Set, Union
 
 from ..exceptions import HyperspaceException
 

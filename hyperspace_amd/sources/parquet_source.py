@@ -15,7 +15,7 @@ predicates prune files before any IO (Spark's partition pruning).
 from __future__ import annotations
 
 import os
-from typing import Any, Dict, List, Optional, Tuple
+from typing import Any, Dict, List, Optional
 
 from .interfaces import FileBasedRelation, FileBasedSourceProvider
 from ..log.entry import FileInfo, Relation, Schema, SchemaField
@@ -165,7 +165,6 @@ class ParquetRelation(FileBasedRelation):
         pschema = self.partition_schema()
         if not pschema.fields:
             return super().read_files(paths, columns, device)
-        import numpy as np
         import torch
         from ..execution.columnar import ColumnBatch, StringColumn
         pnames = {f.name.lower(): f for f in pschema.fields}
