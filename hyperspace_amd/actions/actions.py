@@ -150,27 +150,34 @@ class CreateAction(Action):
         if self.session.conf.lineage_enabled:
             properties[IndexConstants.LINEAGE_PROPERTY] = "true"
 
-        with self.session.with_rule_disabled():
-            index, batch = self.config.create_index(ctx, self.df, properties)
-            written = index.write(ctx, batch)
-        written = _all_written_files(data_path, written)
+        # freeze the source listing for the whole build: the scanned
+        # files, the relation metadata and the signature must describe
+        # one consistent snapshot even if the source is appended to
+        # concurrently (unfrozen after — the user's DataFrame keeps its
+        # dynamic listing)
+        frozen = self.df.plan.collect_leaves()[0].relation.freeze()
+        try:
+            with self.session.with_rule_disabled():
+                index, batch = self.config.create_index(
+                    ctx, self.df, properties)
+                written = index.write(ctx, batch)
+            written = _all_written_files(data_path, written)
 
-        # provider property enrichment (delta tables record the
-        # index->table version history, reference CreateActionBase ->
-        # FileBasedRelationMetadata.enrichIndexProperties)
-        relation0 = self.df.plan.collect_leaves()[0].relation
-        if hasattr(relation0, "enrich_index_properties"):
-            index = index.with_new_properties(
-                relation0.enrich_index_properties(
-                    index.properties, (self.base_id or 0) + 2))
+            # provider property enrichment (delta tables record the
+            # index->table version history, reference CreateActionBase ->
+            # FileBasedRelationMetadata.enrichIndexProperties)
+            if hasattr(frozen, "enrich_index_properties"):
+                index = index.with_new_properties(
+                    frozen.enrich_index_properties(
+                        index.properties, (self.base_id or 0) + 2))
 
-        from ..plan.nodes import Scan
-        relation = self.df.plan.collect_leaves()[0].relation
-        rel_meta = relation.create_relation_metadata(tracker)
-        self._entry = IndexLogEntry.create(
-            self.config.index_name, index, _content_from_paths(written),
-            Source(SourcePlan([rel_meta], _fingerprint(relation))),
-            index.properties)
+            rel_meta = frozen.create_relation_metadata(tracker)
+            self._entry = IndexLogEntry.create(
+                self.config.index_name, index, _content_from_paths(written),
+                Source(SourcePlan([rel_meta], _fingerprint(frozen))),
+                index.properties)
+        finally:
+            frozen.unfreeze()
 
     def log_entry(self) -> IndexLogEntry:
         assert self._entry is not None
@@ -221,9 +228,15 @@ class RefreshActionBase(Action):
                 f"Refresh requires ACTIVE index (is {self.previous.state})")
 
     def source_relation(self):
-        rel_meta = self.previous.relations[0]
-        return self.session.provider_manager.from_metadata(rel_meta) \
-            .refreshed()
+        # one frozen relation per action: the diff, the built index, the
+        # relation metadata and the signature must all describe the SAME
+        # file set — a concurrent append between any two of those would
+        # otherwise commit a signature covering files the index lacks
+        if getattr(self, "_src_relation", None) is None:
+            rel_meta = self.previous.relations[0]
+            self._src_relation = self.session.provider_manager \
+                .from_metadata(rel_meta).refreshed().freeze()
+        return self._src_relation
 
     def compute_diff(self) -> Tuple[List[FileInfo], List[FileInfo]]:
         """(appended, deleted) via FileInfo set-diff on (path,size,mtime)."""

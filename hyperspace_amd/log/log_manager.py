@@ -38,10 +38,13 @@ class IndexLogManager:
 
     def get_log(self, log_id: int) -> Optional[IndexLogEntry]:
         path = self._log_path(log_id)
-        if not os.path.exists(path):
+        try:
+            with open(path, "r") as f:
+                return IndexLogEntry.from_json_str(f.read())
+        except FileNotFoundError:
+            # a concurrent vacuum may trim entries between listing and
+            # read — absent is absent, whoever observed it
             return None
-        with open(path, "r") as f:
-            return IndexLogEntry.from_json_str(f.read())
 
     def get_latest_id(self) -> Optional[int]:
         if not os.path.isdir(self.log_dir):
@@ -57,14 +60,18 @@ class IndexLogManager:
         """latestStable copy if valid, else scan back for a stable state
         (index/IndexLogManager.scala:102-127)."""
         stable_path = os.path.join(self.log_dir, self.LATEST_STABLE)
-        if os.path.exists(stable_path):
-            try:
-                with open(stable_path, "r") as f:
-                    entry = IndexLogEntry.from_json_str(f.read())
-                if entry.state in States.STABLE_STATES:
-                    return entry
-            except (json.JSONDecodeError, KeyError):
-                pass
+        try:
+            with open(stable_path, "r") as f:
+                entry = IndexLogEntry.from_json_str(f.read())
+            if entry.state in States.STABLE_STATES:
+                return entry
+        except (FileNotFoundError, json.JSONDecodeError, KeyError):
+            # FileNotFoundError: a concurrent writer's
+            # delete_latest_stable_log between our exists-check and
+            # open — fall through to the scan-back, same as a corrupt
+            # copy (reference: IndexLogManager.scala:102-127 tolerates
+            # an unreadable latestStable the same way)
+            pass
         latest = self.get_latest_id()
         if latest is None:
             return None

@@ -69,6 +69,26 @@ class FileBasedRelation(ABC):
         """Re-list the current state of the source (for refresh actions)."""
         return self
 
+    def freeze(self) -> "FileBasedRelation":
+        """Pin the file listing at this instant: until :meth:`unfreeze`,
+        ``all_files()`` — and everything derived from it (signature,
+        relation metadata, partition schema) — returns one consistent
+        snapshot.  Lifecycle actions freeze their source so a concurrent
+        append between the diff and the commit cannot produce a log
+        entry whose signature covers files the index does not.  (Spark
+        gets this for free: the InMemoryFileIndex inside the plan is
+        listed once, reference index/sources/interfaces.scala:43-120
+        derives everything from that plan snapshot.)"""
+        snapshot = self.all_files()
+        self.all_files = (  # type: ignore[method-assign]
+            lambda: snapshot)
+        return self
+
+    def unfreeze(self) -> "FileBasedRelation":
+        """Restore dynamic listing (drop the :meth:`freeze` snapshot)."""
+        self.__dict__.pop("all_files", None)
+        return self
+
 
 class FileBasedSourceProvider(ABC):
     """Claims relations it supports and reconstructs them from metadata."""

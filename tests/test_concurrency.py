@@ -272,8 +272,9 @@ def test_concurrent_maintenance_fuzz(env, tmp_path):
                     h.restore_index("cfz")
             except HyperspaceException:
                 pass  # lost race / wrong-state: the designed outcome
-            except Exception as e:  # noqa: BLE001
-                hard_errors.append(repr(e))
+            except Exception:  # noqa: BLE001
+                import traceback as _tb
+                hard_errors.append(_tb.format_exc())
 
     def reader():
         while not stop.is_set():
