@@ -31,7 +31,14 @@ _BUCKET_RE = re.compile(r".*_(\d+)(?:\.\w+)*\.parquet$")
 # host RAM).
 _PINNED_POOL: Dict[int, List["torch.Tensor"]] = {}
 _PINNED_POOL_BYTES = 0
-_PINNED_POOL_MAX = 16 << 30
+# Default 32 GiB: a build-then-serve step keeps BOTH the source-staging
+# classes (~few large buffers) and the index-read classes (~hundreds of
+# bucket files) resident; a 16 GiB cap evicted one set every step and
+# re-paid ~190 ms/GB of pinned allocation on the cold query read
+# (profiles/cold_read_probe: full path 8.3 GB/s pool-cold vs 36 GB/s
+# pool-warm, H2D-bound).  HS_PINNED_POOL_GB overrides.
+_PINNED_POOL_MAX = int(
+    float(os.environ.get("HS_PINNED_POOL_GB", "32")) * (1 << 30))
 import threading as _threading
 
 _pool_lock = _threading.Lock()
