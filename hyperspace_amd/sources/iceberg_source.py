@@ -47,10 +47,11 @@ class IcebergTable:
     # -- metadata ----------------------------------------------------------
     def current_version(self) -> Optional[int]:
         hint = os.path.join(self.meta_dir, "version-hint.text")
-        if not os.path.exists(hint):
+        try:
+            with open(hint) as f:
+                return int(f.read().strip())
+        except FileNotFoundError:
             return None
-        with open(hint) as f:
-            return int(f.read().strip())
 
     def _snapshot(self, version: int) -> Dict:
         with open(os.path.join(self.meta_dir,

@@ -307,3 +307,63 @@ def test_concurrent_maintenance_fuzz(env, tmp_path):
     t_all = pq2.read_table(data_dir)
     want = int((t_all.column("key").to_numpy() == 5).sum())
     assert got.num_rows == want
+
+
+def test_relation_freeze_pins_listing(env, tmp_path):
+    """freeze() pins all_files/signature; unfreeze restores dynamic
+    listing (the invariant behind the action-layer snapshot fix)."""
+    session, h, df = env
+    rel = df.plan.collect_leaves()[0].relation
+    frozen = rel.freeze()
+    sig0 = frozen.signature()
+    n0 = len(frozen.all_files())
+    pq.write_table(pa.table({"key": np.arange(10),
+                             "val": np.arange(10.0)}),
+                   str(tmp_path / "data" / "part-new.parquet"))
+    # frozen view: the concurrent append is invisible
+    assert len(frozen.all_files()) == n0
+    assert frozen.signature() == sig0
+    frozen.unfreeze()
+    assert len(rel.all_files()) == n0 + 1
+    assert rel.signature() != sig0
+
+
+def test_refresh_commits_signature_of_built_snapshot(env, tmp_path):
+    """A file appended AFTER a refresh computed its diff must not be
+    covered by the committed signature: the post-refresh query must see
+    a signature mismatch and fall back to a full scan (deterministic
+    replay of the fuzz-found stale-result bug)."""
+    from hyperspace_amd.actions.actions import RefreshIncrementalAction
+
+    session, h, df = env
+    session.conf.set(hs.IndexConstants.INDEX_LINEAGE_ENABLED, True)
+    h.create_index(df, hs.CoveringIndexConfig("frz", ["key"], ["val"]))
+    data = tmp_path / "data"
+    pq.write_table(pa.table({"key": np.full(100, 5),
+                             "val": np.zeros(100)}),
+                   str(data / "part-1.parquet"))
+
+    mgr = session.index_manager()
+    _, log_mgr, data_mgr = mgr._managers("frz")
+    action = RefreshIncrementalAction(session, log_mgr, data_mgr)
+    # diff computed (freezes the listing) ...
+    appended, deleted = action.compute_diff()
+    assert [os.path.basename(f.name) for f in appended] == \
+        ["part-1.parquet"]
+    # ... then a racing writer appends another file mid-action
+    pq.write_table(pa.table({"key": np.full(50, 5),
+                             "val": np.ones(50)}),
+                   str(data / "part-2.parquet"))
+    action.run()
+
+    # committed signature describes the frozen snapshot, NOT the current
+    # source -> the index is stale and must NOT serve this query
+    entry = mgr.get_index("frz")
+    live_sig = df.plan.collect_leaves()[0].relation.signature()
+    assert entry.signature.value != live_sig
+
+    session.enable_hyperspace()
+    got = df.filter("key = 5").select("key", "val").collect()
+    t_all = pq.read_table(str(data))
+    want = int((t_all.column("key").to_numpy() == 5).sum())
+    assert got.num_rows == want
