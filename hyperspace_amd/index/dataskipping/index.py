@@ -68,21 +68,45 @@ class DataSkippingIndex(Index):
     # -- build -------------------------------------------------------------
     def write(self, ctx: IndexerContext, index_data) -> List[str]:
         """index_data: dict with '_data_file_id' per-file ids and the
-        sketch aggregate tensors (built by config.create_index)."""
+        sketch aggregate tensors (built by config.create_index).
+
+        Output is split to the targetIndexDataFileSize (256 MiB default)
+        capped at maxIndexDataFileCount files — the reference's writeImpl
+        sizing, index/dataskipping/DataSkippingIndex.scala:187-206."""
         os.makedirs(ctx.index_data_path, exist_ok=True)
-        out = os.path.join(ctx.index_data_path,
-                           "part-00000-sketches_00000.c000.parquet")
         import pyarrow as pa
         import pyarrow.parquet as pq
+        from ...config import IndexConstants as IC
         arrays = {}
+        row_bytes = 0
+        n_rows = 0
         for name, t in index_data.items():
+            n_rows = t.shape[0]
+            row_bytes += t.element_size() * (t.shape[1] if t.dim() == 2
+                                             else 1)
             if t.dim() == 2:  # bloom word matrix -> fixed-size list column
                 arrays[name] = pa.array(list(t.numpy()))
             else:
                 arrays[name] = pa.array(t.numpy())
-        pq.write_table(pa.table(arrays), out, compression="NONE",
-                       use_dictionary=False)
-        return [out]
+        target = int(ctx.session.conf.get(
+            IC.DATASKIPPING_TARGET_INDEX_DATA_FILE_SIZE))
+        max_files = int(ctx.session.conf.get(
+            IC.DATASKIPPING_MAX_INDEX_DATA_FILE_COUNT,
+            IC.DATASKIPPING_MAX_INDEX_DATA_FILE_COUNT_DEFAULT))
+        per_file = max(1, target // max(1, row_bytes))
+        n_files = min(max_files,
+                      max(1, -(-n_rows // per_file))) if n_rows else 1
+        per_file = -(-n_rows // n_files) if n_rows else 1
+        table = pa.table(arrays)
+        written = []
+        for i in range(n_files):
+            out = os.path.join(
+                ctx.index_data_path,
+                f"part-{i:05d}-sketches_00000.c000.parquet")
+            pq.write_table(table.slice(i * per_file, per_file), out,
+                           compression="NONE", use_dictionary=False)
+            written.append(out)
+        return written
 
     def refresh_full(self, ctx, df):
         raise NotImplementedError

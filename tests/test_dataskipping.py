@@ -245,3 +245,40 @@ def test_sketch_over_arithmetic_expression(tmp_path, monkeypatch):
     out2 = Executor(session).execute(q2.optimized_plan())
     expected2 = sum(int((k % 100 >= 55).sum()) for k in all_keys)
     assert out2.num_rows == expected2
+
+
+def test_sketch_file_sizing(tmp_path, monkeypatch):
+    """write() splits sketch rows to targetIndexDataFileSize (reference
+    DataSkippingIndex writeImpl sizing, DataSkippingIndex.scala:187-206)
+    and the split index still prunes correctly."""
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "idx"))
+    data = tmp_path / "data"
+    data.mkdir()
+    rng = np.random.default_rng(3)
+    for i in range(40):
+        lo = i * 100
+        pq.write_table(
+            pa.table({"key": rng.integers(lo, lo + 100, 500),
+                      "val": rng.random(500)}),
+            str(data / f"part-{i:02d}.parquet"))
+    session = hs.HyperspaceSession(device="cpu")
+    # ~40 sketch rows; force a tiny target so the write must split
+    session.conf.set(
+        hs.IndexConstants.DATASKIPPING_TARGET_INDEX_DATA_FILE_SIZE, 200)
+    h = hs.Hyperspace(session)
+    df = session.read_parquet(str(data))
+    from hyperspace_amd.index.dataskipping.sketches import MinMaxSketch
+    h.create_index(df, hs.DataSkippingIndexConfig(
+        "dsz", MinMaxSketch("key")))
+    entry = session.index_manager().get_index("dsz")
+    files = [p for p in entry.content.os_files()
+             if p.endswith(".parquet")]
+    assert len(files) > 1, "expected the sketch data to split"
+    session.enable_hyperspace()
+    from hyperspace_amd.execution.executor import Executor
+    ex = Executor(session)
+    out = ex.execute(df.filter("key >= 3900").optimized_plan())
+    t = pq.read_table(str(data))
+    assert out.num_rows == int(
+        (t.column("key").to_numpy() >= 3900).sum())
+    assert ex.stats.scanned_files == 1  # only the last file can match
