@@ -162,6 +162,10 @@ class CreateAction(Action):
                     ctx, self.df, properties)
                 written = index.write(ctx, batch)
             written = _all_written_files(data_path, written)
+            # covering builds leave the sorted bucket-major batch on ctx
+            # for the post-commit HBM cache write-through (local rank's
+            # files only — matching the per-rank query-time file set)
+            self.built_for_cache = getattr(ctx, "built_for_cache", None)
 
             # provider property enrichment (delta tables record the
             # index->table version history, reference CreateActionBase ->

@@ -189,9 +189,10 @@ class Executor:
                                                  num_buckets)
             full_files = [p for b in wanted_buckets
                           for p in sorted(by_bucket[b])]
-            full_key = (cache.key(entry, full_files, read_cols)
+            full_key = (cache.key(entry, full_files)
                         if cache else None)
-            full_cached = cache.get(full_key) if cache else None
+            full_cached = (cache.get(full_key, read_cols)
+                           if cache else None)
             if full_cached is not None:
                 batch, seg = full_cached
                 self.stats.record("IndexScan(cached)")
@@ -216,8 +217,8 @@ class Executor:
 
         cache_files = [p for b in wanted_buckets
                        for p in sorted(by_bucket[b])]
-        key = cache.key(entry, cache_files, read_cols) if cache else None
-        cached = cache.get(key) if cache else None
+        key = cache.key(entry, cache_files) if cache else None
+        cached = cache.get(key, read_cols) if cache else None
         if cached is not None:
             batch, seg = cached
             self.stats.record("IndexScan(cached)")

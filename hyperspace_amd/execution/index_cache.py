@@ -32,28 +32,39 @@ class IndexDataCache:
         self.misses = 0
 
     @staticmethod
-    def key(entry, files: List[str], columns: List[str],
-            extra: tuple = ()) -> tuple:
-        return (entry.name, entry.id, tuple(sorted(files)),
-                tuple(c.lower() for c in columns), extra)
+    def key(entry, files: List[str], extra: tuple = ()) -> tuple:
+        # columns are NOT part of the key: one stored batch serves every
+        # query whose columns are a subset (the build write-through
+        # stores the full slice; filter/join queries select from it)
+        return (entry.name, entry.id, tuple(sorted(files)), extra)
 
-    def get(self, key: tuple):
+    def get(self, key: tuple, columns: Optional[List[str]] = None):
         item = self._entries.get(key)
+        if item is not None and columns is not None:
+            have = {c.lower() for c in item[0].columns}
+            if not all(c.lower() in have for c in columns):
+                item = None  # stored batch lacks a needed column
         if item is None:
             self.misses += 1
             return None
         self._entries.move_to_end(key)
         self.hits += 1
-        return item
+        batch, seg = item
+        if columns is not None:
+            batch = batch.select(columns)
+        return batch, seg
 
     def put(self, key: tuple, batch: ColumnBatch,
             seg: Optional[torch.Tensor]) -> None:
+        old = self._entries.pop(key, None)
+        if old is not None:
+            self._bytes -= old[0].nbytes()
         nbytes = batch.nbytes()
         if nbytes > self.budget:
             return
         while self._bytes + nbytes > self.budget and self._entries:
-            _, (old, _seg) = self._entries.popitem(last=False)
-            self._bytes -= old.nbytes()
+            _, (evicted, _seg) = self._entries.popitem(last=False)
+            self._bytes -= evicted.nbytes()
         self._entries[key] = (batch, seg)
         self._bytes += nbytes
 

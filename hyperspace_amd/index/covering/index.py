@@ -133,6 +133,10 @@ class CoveringIndex(Index):
                 batch = self._empty_batch(ctx)
             written.extend(self._write_group(ctx, batch, task_id,
                                              distributed))
+        if n_groups > 1:
+            # multi-group (out-of-core) builds: the per-group batches
+            # are partial slices of each bucket — not cacheable as-is
+            ctx.built_for_cache = None
         return written
 
     def _write_group(self, ctx, batch: ColumnBatch, task_id: int,
@@ -161,6 +165,12 @@ class CoveringIndex(Index):
                 _torch.cuda.synchronize()
         t2 = _time.perf_counter()
         out = write_bucketed(batch, seg, ctx.index_data_path, n, task_id)
+        # build write-through for the HBM index-data cache: the sorted
+        # bucket-major batch + segment offsets ARE the index-scan layout,
+        # so the first post-build query need not re-read what is already
+        # resident (288 GB HBM3E; the put happens after the action
+        # commits, keyed on the final log id — see IndexManagement)
+        ctx.built_for_cache = (batch, seg, list(out))
         if timing:
             import sys as _sys
             print(f"[hs-timing] build group: hash+exchange {t1-t0:.3f}s "

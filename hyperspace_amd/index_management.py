@@ -43,7 +43,18 @@ class IndexCollectionManager:
     def create(self, df, config) -> None:
         path, log_mgr, data_mgr = self._managers(config.index_name)
         os.makedirs(path, exist_ok=True)
-        CreateAction(self.session, df, config, log_mgr, data_mgr).run()
+        action = CreateAction(self.session, df, config, log_mgr, data_mgr)
+        action.run()
+        # build write-through: the just-sorted index batch is already in
+        # HBM in index-scan layout; key it on the committed entry so the
+        # first query serves from residency instead of re-reading files
+        built = getattr(action, "built_for_cache", None)
+        cache = self.session.index_data_cache()
+        if built is not None and cache is not None:
+            entry = log_mgr.get_latest_stable_log()
+            if entry is not None:
+                batch, seg, files = built
+                cache.put(cache.key(entry, files), batch, seg)
 
     def delete(self, name: str) -> None:
         _, log_mgr, _ = self._managers(name)

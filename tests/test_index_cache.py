@@ -85,3 +85,21 @@ def test_cache_eviction_budget():
     assert cache.get(("k1",)) is None
     assert cache.get(("k2",)) is not None
     assert cache.get(("k3",)) is not None
+
+
+def test_build_write_through(env):
+    """The first query after create must hit the cache (the build's
+    sorted bucket-major batch is already HBM/host-resident in index-scan
+    layout) and scan zero files."""
+    session, h, df, _, _ = env
+    from hyperspace_amd.execution.executor import Executor
+    cache = session.index_data_cache()
+    h.create_index(df, hs.CoveringIndexConfig("wt", ["key"], ["val"]))
+    session.enable_hyperspace()
+    hits0, misses0 = cache.hits, cache.misses
+    ex = Executor(session)
+    out = ex.execute(df.filter("key >= 0").select("key", "val")
+                     .optimized_plan())
+    assert out.num_rows == df.collect().num_rows
+    assert cache.hits == hits0 + 1, (cache.hits, cache.misses)
+    assert ex.stats.scanned_files == 0

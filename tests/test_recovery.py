@@ -135,6 +135,12 @@ def test_missing_index_files_fail_loudly(env):
     entry = session.index_manager().get_index("mx")
     os.unlink(entry.content.os_files()[0])
     session.enable_hyperspace()
+    # drop the build write-through residency: this test targets the
+    # file READ path (a resident cache may legitimately keep serving
+    # after out-of-band deletion; lifecycle ops invalidate via log id)
+    cache = session.index_data_cache()
+    if cache is not None:
+        cache.clear()
     q = df.filter("key = 7").select("key", "val")
     with pytest.raises(Exception):
         q.collect()
