@@ -89,9 +89,9 @@ class Action(ABC):
             dc.barrier()
         else:
             # non-collective: broadcast the no-op outcome from rank 0
-            import torch
+            # (device tensor under NCCL/RCCL — a CPU tensor crashes there)
             import torch.distributed as dist
-            t = torch.tensor([1 if no_changes else 0])
+            t = dc.collective_tensor([1 if no_changes else 0])
             dist.broadcast(t, src=0)
             no_changes = bool(t[0])
         dc.barrier()

@@ -160,6 +160,12 @@ class ColumnBatch:
         for col in self.columns.values():
             t = col.codes if isinstance(col, StringColumn) else col
             total += t.numel() * t.element_size()
+            if isinstance(col, StringColumn):
+                # dictionary payload estimate (host-side, but budgeted so
+                # string-heavy cached batches don't under-account)
+                total += sum(len(v) for v in col.values)
+        for m in self.masks.values():
+            total += m.numel() * m.element_size()
         return total
 
     # -- transforms -------------------------------------------------------

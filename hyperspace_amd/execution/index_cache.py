@@ -56,12 +56,21 @@ class IndexDataCache:
 
     def put(self, key: tuple, batch: ColumnBatch,
             seg: Optional[torch.Tensor]) -> None:
-        old = self._entries.pop(key, None)
-        if old is not None:
-            self._bytes -= old[0].nbytes()
         nbytes = batch.nbytes()
         if nbytes > self.budget:
+            # oversized: keep whatever incumbent we have under this key
             return
+        old = self._entries.get(key)
+        if old is not None:
+            # keep the incumbent if it already serves every column of the
+            # new batch (e.g. the full-slice build write-through) — a
+            # narrow-column query's put must not evict the wide batch
+            have = {c.lower() for c in old[0].columns}
+            if all(c.lower() in have for c in batch.columns):
+                self._entries.move_to_end(key)
+                return
+            self._entries.pop(key)
+            self._bytes -= old[0].nbytes()
         while self._bytes + nbytes > self.budget and self._entries:
             _, (evicted, _seg) = self._entries.popitem(last=False)
             self._bytes -= evicted.nbytes()

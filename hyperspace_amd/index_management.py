@@ -51,8 +51,12 @@ class IndexCollectionManager:
         built = getattr(action, "built_for_cache", None)
         cache = self.session.index_data_cache()
         if built is not None and cache is not None:
-            entry = log_mgr.get_latest_stable_log()
-            if entry is not None:
+            # key on the entry THIS action committed (log id base+2), not
+            # latestStable — a concurrent action landing between end() and
+            # here would otherwise key the batch under a mismatched entry
+            entry = (log_mgr.get_log(action.base_id + 2)
+                     if action.base_id is not None else None)
+            if entry is not None and entry.state == States.ACTIVE:
                 batch, seg, files = built
                 cache.put(cache.key(entry, files), batch, seg)
 

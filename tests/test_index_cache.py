@@ -103,3 +103,47 @@ def test_build_write_through(env):
     assert out.num_rows == df.collect().num_rows
     assert cache.hits == hits0 + 1, (cache.hits, cache.misses)
     assert ex.stats.scanned_files == 0
+
+
+def test_put_keeps_superset_incumbent():
+    """A narrow-column put must not evict a wide incumbent under the same
+    key (the build write-through stores the full slice; a later query's
+    narrow put would otherwise thrash it)."""
+    from hyperspace_amd.execution.index_cache import IndexDataCache
+    from hyperspace_amd.execution.columnar import ColumnBatch
+    import torch
+    cache = IndexDataCache(budget_bytes=1 << 20)
+    wide = ColumnBatch({"key": torch.arange(10), "val": torch.arange(10)})
+    narrow = ColumnBatch({"key": torch.arange(10)})
+    cache.put(("k",), wide, None)
+    cache.put(("k",), narrow, None)
+    got, _ = cache.get(("k",), columns=["key", "val"])
+    assert set(got.names) == {"key", "val"}
+    # a genuinely different-column put does replace
+    other = ColumnBatch({"other": torch.arange(10)})
+    cache.put(("k",), other, None)
+    assert cache.get(("k",), columns=["key", "val"]) is None
+
+
+def test_oversized_put_keeps_incumbent():
+    """An oversized replacement must not drop a valid cached entry."""
+    from hyperspace_amd.execution.index_cache import IndexDataCache
+    from hyperspace_amd.execution.columnar import ColumnBatch
+    import torch
+    cache = IndexDataCache(budget_bytes=2000)
+    small = ColumnBatch({"a": torch.zeros(100, dtype=torch.int64)})
+    big = ColumnBatch({"b": torch.zeros(10000, dtype=torch.int64)})
+    cache.put(("k",), small, None)
+    cache.put(("k",), big, None)  # over budget: refused
+    assert cache.get(("k",), columns=["a"]) is not None
+
+
+def test_nbytes_counts_masks_and_strings():
+    from hyperspace_amd.execution.columnar import ColumnBatch, StringColumn
+    import torch
+    b = ColumnBatch(
+        {"s": StringColumn(torch.zeros(4, dtype=torch.int32), ["aa", "bb"]),
+         "x": torch.zeros(4, dtype=torch.int64)},
+        masks={"x": torch.ones(4, dtype=torch.bool)})
+    # codes 16 + dict 4 + x 32 + mask 4
+    assert b.nbytes() == 16 + 4 + 32 + 4
