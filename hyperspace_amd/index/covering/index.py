@@ -124,36 +124,104 @@ class CoveringIndex(Index):
             t = dc.collective_tensor([n_groups])
             dist.all_reduce(t, op=dist.ReduceOp.MAX)
             n_groups = int(t[0])
-
-        written: List[str] = []
-        for g in range(n_groups):
-            try:
-                batch = next(batches)
-            except StopIteration:
-                batch = self._empty_batch(ctx)
-            written.extend(self._write_group(ctx, batch, task_id,
-                                             distributed))
+            written = self._write_distributed(ctx, batches, n_groups,
+                                              task_id)
+        else:
+            written = []
+            for g in range(n_groups):
+                try:
+                    batch = next(batches)
+                except StopIteration:
+                    batch = self._empty_batch(ctx)
+                written.extend(self._write_group(ctx, batch, task_id))
         if n_groups > 1:
             # multi-group (out-of-core) builds: the per-group batches
             # are partial slices of each bucket — not cacheable as-is
             ctx.built_for_cache = None
         return written
 
-    def _write_group(self, ctx, batch: ColumnBatch, task_id: int,
-                     distributed: bool) -> List[str]:
+    def _bucket_ids(self, batch: ColumnBatch) -> torch.Tensor:
+        from ...ops.string_hash import bucket_hash_keys
+        keys = bucket_hash_keys(batch, self.indexed_columns)
+        key_masks = [batch.mask(c) for c in self.indexed_columns]
+        return ops.murmur3_bucket(
+            keys, self.num_buckets,
+            key_masks if any(m is not None for m in key_masks) else None)
+
+    def _write_distributed(self, ctx, batches, n_groups: int,
+                           task_id: int) -> List[str]:
+        """Pipelined multi-GPU build: each group is split into chunks;
+        chunk k+1's hash/pack and the previous group's sort/write overlap
+        chunk k's RCCL all-to-all in flight on the xGMI links
+        (SURVEY §7 hard part b; C1)."""
+        import torch.distributed as dist
+        from ...parallel import dist_context as dc
+        from ...parallel.exchange import BucketExchange
+        n = self.num_buckets
+        ex = BucketExchange(n)
+        ctx.exchange_stats = ex
+        chunk_bytes = int(ctx.session.conf.get(
+            "spark.hyperspace.exchange.chunkBytes", 1 << 30))
+        written: List[str] = []
+        recv_acc: Dict[int, list] = {}
+
+        def flush_group(g: int) -> None:
+            parts = recv_acc.pop(g, [])
+            parts = [(b, k) for b, k in parts if b.num_rows]
+            if not parts:
+                return
+            if len(parts) == 1:
+                rb, rbk = parts[0]
+            else:
+                rb = ColumnBatch.concat([p[0] for p in parts])
+                rbk = torch.cat([p[1] for p in parts])
+            batch, seg = sort_by_bucket_and_keys(
+                rb, rbk, self.indexed_columns, n)
+            out = write_bucketed(batch, seg, ctx.index_data_path, n,
+                                 task_id)
+            ctx.built_for_cache = (batch, seg, list(out))
+            written.extend(out)
+
+        prev = None  # (pending, group, last-chunk-of-group)
+        for g in range(n_groups):
+            try:
+                batch = next(batches)
+            except StopIteration:
+                batch = self._empty_batch(ctx)
+            # every rank must post the same number of chunk exchanges:
+            # agree on max chunk count, short ranks send empty chunks
+            local_chunks = 1
+            if chunk_bytes > 0 and batch.num_rows:
+                local_chunks = max(
+                    1, -(-batch.nbytes() // chunk_bytes))
+            t = dc.collective_tensor([local_chunks])
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            nchunks = int(t[0])
+            rows = batch.num_rows
+            for c in range(nchunks):
+                lo = rows * c // nchunks
+                hi = rows * (c + 1) // nchunks
+                sub = batch.slice(lo, hi) if nchunks > 1 else batch
+                pend = ex.start(sub, self._bucket_ids(sub))
+                if prev is not None:
+                    p, pg, plast = prev
+                    recv_acc.setdefault(pg, []).append(ex.finish(p))
+                    if plast:
+                        flush_group(pg)
+                prev = (pend, g, c == nchunks - 1)
+        if prev is not None:
+            p, pg, _ = prev
+            recv_acc.setdefault(pg, []).append(ex.finish(p))
+            flush_group(pg)
+        return written
+
+    def _write_group(self, ctx, batch: ColumnBatch,
+                     task_id: int) -> List[str]:
         import time as _time
         timing = os.environ.get("HS_TIMING")
         n = self.num_buckets
-        from ...ops.string_hash import bucket_hash_keys
         t0 = _time.perf_counter()
-        keys = bucket_hash_keys(batch, self.indexed_columns)
-        key_masks = [batch.mask(c) for c in self.indexed_columns]
-        bucket_ids = ops.murmur3_bucket(
-            keys, n,
-            key_masks if any(m is not None for m in key_masks) else None)
-        if distributed:
-            from ...parallel.exchange import exchange_by_bucket
-            batch, bucket_ids = exchange_by_bucket(batch, bucket_ids, n)
+        bucket_ids = self._bucket_ids(batch)
         if batch.num_rows == 0:
             return []
         t1 = _time.perf_counter()
@@ -173,20 +241,30 @@ class CoveringIndex(Index):
         ctx.built_for_cache = (batch, seg, list(out))
         if timing:
             import sys as _sys
-            print(f"[hs-timing] build group: hash+exchange {t1-t0:.3f}s "
+            print(f"[hs-timing] build group: hash {t1-t0:.3f}s "
                   f"sort {t2-t1:.3f}s write "
                   f"{_time.perf_counter()-t2:.3f}s", file=_sys.stderr)
         return out
 
     def _empty_batch(self, ctx) -> ColumnBatch:
+        """Zero-row batch matching the index schema — ranks that run out
+        of file groups still join every collective with the exact same
+        packed layout (string fields must be StringColumns, not i64)."""
         import torch as _torch
         dts = {"long": _torch.int64, "integer": _torch.int32,
-               "double": _torch.float64, "float": _torch.float32}
-        cols = {}
+               "short": _torch.int16, "byte": _torch.int8,
+               "double": _torch.float64, "float": _torch.float32,
+               "boolean": _torch.bool}
+        cols: Dict[str, Any] = {}
         for f in self.schema.fields:
-            cols[f.name] = _torch.empty(
-                0, dtype=dts.get(f.type, _torch.int64),
-                device=ctx.session.device)
+            if f.type == "string":
+                cols[f.name] = StringColumn(
+                    _torch.empty(0, dtype=_torch.int32,
+                                 device=ctx.session.device), [])
+            else:
+                cols[f.name] = _torch.empty(
+                    0, dtype=dts.get(f.type, _torch.int64),
+                    device=ctx.session.device)
         return ColumnBatch(cols)
 
     def optimize(self, ctx: IndexerContext,
