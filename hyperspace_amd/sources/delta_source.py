@@ -1,51 +1,83 @@
-"""Delta-style transactional table source.
+"""Delta Lake transactional table source — REAL Delta protocol.
 
 Reference: index/sources/delta/ — DeltaLakeRelation (signature = table
 version + path, delta/DeltaLakeRelation.scala:40-44; versionAsOf time
-travel; files from the transaction log) and DeltaLakeRelationMetadata
-(deltaVersions index-to-table version history property,
-delta/DeltaLakeRelationMetadata.scala:39-59).
+travel; files from the transaction log via TahoeLogFileIndex) and
+DeltaLakeRelationMetadata (deltaVersions index-to-table version history
+property, delta/DeltaLakeRelationMetadata.scala:39-59).
 
-The table format here is a minimal native transaction log (the engine is
-Spark-free, so it defines its own):
+This module reads and writes the actual open Delta Lake transaction-log
+format (the one Spark's delta-core and delta-rs produce), not a
+lookalike:
 
-    <table>/_delta_log/<version>.json   {"version", "timestamp",
-                                         "add": [{path,size,mtime}...],
-                                         "remove": [path...]}
-    <table>/*.parquet                   data files
+    <table>/_delta_log/<version:020d>.json      newline-delimited JSON
+        actions, each a single-key object:
+        {"commitInfo": {...}}            informational
+        {"protocol": {"minReaderVersion": 1, "minWriterVersion": 2}}
+        {"metaData": {"id", "format": {"provider": "parquet", ...},
+                      "schemaString": <Spark StructType JSON>,
+                      "partitionColumns": [...], "configuration", ...}}
+        {"add": {"path": <url-encoded relative path>, "partitionValues",
+                 "size", "modificationTime", "dataChange", ...}}
+        {"remove": {"path", "deletionTimestamp", "dataChange"}}
+    <table>/_delta_log/<version:020d>.checkpoint.parquet
+        full state snapshot (one action per row in columns
+        txn/add/remove/metaData/protocol); multi-part checkpoints
+        (<v>.checkpoint.<i:010d>.<n:010d>.parquet) are read too
+    <table>/_delta_log/_last_checkpoint   {"version": N, "size": M}
 
-Appends and deletes commit new log versions with atomic hard-link claims
-(same optimistic protocol as the index op log).
+Snapshot reconstruction starts from the newest checkpoint at or below
+the requested version and replays the JSON commits after it.  Writes
+commit with atomic hard-link claims (optimistic concurrency, same
+protocol as the index op log) and roll a checkpoint every
+``CHECKPOINT_INTERVAL`` commits like delta-core does.
 """
 
 from __future__ import annotations
 
 import json
 import os
+import re
 import tempfile
 import time
-from typing import Dict, List, Optional
+import uuid
+from typing import Any, Dict, List, Optional, Tuple
+from urllib.parse import quote, unquote
 
 from .interfaces import FileBasedRelation, FileBasedSourceProvider
 from ..exceptions import HyperspaceException
 from ..log.entry import FileInfo, Relation, Schema
 
 LOG_DIR = "_delta_log"
+LAST_CHECKPOINT = "_last_checkpoint"
+CHECKPOINT_INTERVAL = 10
+
+_COMMIT_RE = re.compile(r"^(\d{20})\.json$")
+_CHECKPOINT_RE = re.compile(
+    r"^(\d{20})\.checkpoint(?:\.\d{10}\.\d{10})?\.parquet$")
 
 
 class DeltaTable:
-    """Minimal append/delete table with a versioned commit log."""
+    """Reader/writer for a Delta Lake table directory."""
 
     def __init__(self, path: str):
         self.path = os.path.abspath(path)
         self.log_dir = os.path.join(self.path, LOG_DIR)
 
-    # -- log --------------------------------------------------------------
+    # -- log listing -------------------------------------------------------
     def versions(self) -> List[int]:
+        """All versions visible via JSON commits or checkpoints."""
         if not os.path.isdir(self.log_dir):
             return []
-        return sorted(int(f[:-5]) for f in os.listdir(self.log_dir)
-                      if f.endswith(".json") and f[:-5].isdigit())
+        vs = set()
+        for f in os.listdir(self.log_dir):
+            m = _COMMIT_RE.match(f)
+            if m:
+                vs.add(int(m.group(1)))
+            m = _CHECKPOINT_RE.match(f)
+            if m:
+                vs.add(int(m.group(1)))
+        return sorted(vs)
 
     @property
     def version(self) -> int:
@@ -54,15 +86,123 @@ class DeltaTable:
             raise HyperspaceException(f"Not a delta table: {self.path}")
         return vs[-1]
 
-    def _commit(self, version: int, add: List[Dict], remove: List[str]):
+    def _checkpoint_files(self, version: int) -> List[str]:
+        out = []
+        for f in os.listdir(self.log_dir):
+            m = _CHECKPOINT_RE.match(f)
+            if m and int(m.group(1)) == version:
+                out.append(os.path.join(self.log_dir, f))
+        return sorted(out)
+
+    def _latest_checkpoint_at_or_below(self, target: int) -> Optional[int]:
+        # fast path: _last_checkpoint hint (may point above target)
+        best: Optional[int] = None
+        hint = os.path.join(self.log_dir, LAST_CHECKPOINT)
+        if os.path.exists(hint):
+            try:
+                with open(hint) as f:
+                    v = int(json.load(f)["version"])
+                if v <= target and self._checkpoint_files(v):
+                    best = v
+            except (ValueError, KeyError, json.JSONDecodeError):
+                pass
+        if best == target:
+            return best
+        for f in os.listdir(self.log_dir):
+            m = _CHECKPOINT_RE.match(f)
+            if m:
+                v = int(m.group(1))
+                if v <= target and (best is None or v > best):
+                    best = v
+        return best
+
+    # -- action replay -----------------------------------------------------
+    def _apply_action(self, state: Dict[str, Any], action: Dict[str, Any]):
+        if "add" in action and action["add"]:
+            a = action["add"]
+            state["files"][unquote(a["path"])] = (
+                int(a["size"]), int(a.get("modificationTime", 0)),
+                a.get("partitionValues") or {})
+        elif "remove" in action and action["remove"]:
+            r = action["remove"]
+            state["files"].pop(unquote(r["path"]), None)
+        elif "metaData" in action and action["metaData"]:
+            state["metaData"] = action["metaData"]
+        elif "protocol" in action and action["protocol"]:
+            state["protocol"] = action["protocol"]
+
+    def _replay(self, target: int) -> Dict[str, Any]:
+        state: Dict[str, Any] = {"files": {}, "metaData": None,
+                                 "protocol": None}
+        start = 0
+        cp = self._latest_checkpoint_at_or_below(target)
+        if cp is not None:
+            import pyarrow.parquet as pq
+            for part in self._checkpoint_files(cp):
+                table = pq.read_table(part)
+                for row in table.to_pylist():
+                    for key in ("protocol", "metaData", "add", "remove"):
+                        if row.get(key) is not None:
+                            act = {key: _denull(row[key])}
+                            if key in ("add", "remove") and isinstance(
+                                    act[key].get("partitionValues"), list):
+                                # pyarrow map type -> list of
+                                # (key, value) pairs
+                                act[key]["partitionValues"] = dict(
+                                    (p["key"], p["value"]) if
+                                    isinstance(p, dict) else tuple(p)
+                                    for p in act[key]["partitionValues"])
+                            self._apply_action(state, act)
+            start = cp + 1
+        for v in range(start, target + 1):
+            p = os.path.join(self.log_dir, f"{v:020d}.json")
+            if not os.path.exists(p):
+                if v == 0 and cp is None:
+                    raise HyperspaceException(
+                        f"Not a delta table: {self.path}")
+                continue
+            with open(p) as f:
+                for line in f:
+                    line = line.strip()
+                    if line:
+                        self._apply_action(state, json.loads(line))
+        return state
+
+    # -- public snapshot API ------------------------------------------------
+    def files_at(self, version: Optional[int] = None) -> List[FileInfo]:
+        target = self.version if version is None else version
+        state = self._replay(target)
+        out = []
+        for rel, (size, mtime, _pv) in state["files"].items():
+            p = rel if os.path.isabs(rel) else os.path.join(self.path, rel)
+            out.append(FileInfo(p, size, mtime))
+        return sorted(out, key=lambda f: f.name)
+
+    def partition_values_at(self, version: Optional[int] = None
+                            ) -> Dict[str, Dict[str, str]]:
+        """abs path -> partitionValues for the snapshot."""
+        target = self.version if version is None else version
+        state = self._replay(target)
+        out = {}
+        for rel, (_s, _m, pv) in state["files"].items():
+            p = rel if os.path.isabs(rel) else os.path.join(self.path, rel)
+            out[p] = pv
+        return out
+
+    def metadata_at(self, version: Optional[int] = None
+                    ) -> Optional[Dict[str, Any]]:
+        target = self.version if version is None else version
+        return self._replay(target)["metaData"]
+
+    # -- write side ---------------------------------------------------------
+    def _commit(self, version: int, actions: List[Dict[str, Any]]):
         os.makedirs(self.log_dir, exist_ok=True)
-        entry = {"version": version, "timestamp": int(time.time() * 1000),
-                 "add": add, "remove": remove}
-        target = os.path.join(self.log_dir, f"{version:010d}.json")
+        target = os.path.join(self.log_dir, f"{version:020d}.json")
         fd, tmp = tempfile.mkstemp(dir=self.log_dir, prefix=".tmp_")
         try:
             with os.fdopen(fd, "w") as f:
-                json.dump(entry, f)
+                for a in actions:
+                    f.write(json.dumps(a, separators=(",", ":")) + "\n")
             try:
                 os.link(tmp, target)
             except FileExistsError:
@@ -70,58 +210,193 @@ class DeltaTable:
                     f"Concurrent commit lost race for version {version}")
         finally:
             os.unlink(tmp)
+        if version > 0 and version % CHECKPOINT_INTERVAL == 0:
+            try:
+                self.checkpoint(version)
+            except Exception:
+                pass  # checkpoints are an optimization, never required
 
-    # -- operations -------------------------------------------------------
     @staticmethod
-    def create(path: str) -> "DeltaTable":
+    def create(path: str, schema: Optional[Schema] = None,
+               partition_columns: Optional[List[str]] = None
+               ) -> "DeltaTable":
         t = DeltaTable(path)
         os.makedirs(path, exist_ok=True)
         if not t.versions():
-            t._commit(0, [], [])
+            t._commit(0, [
+                {"commitInfo": {"timestamp": _now_ms(),
+                                "operation": "CREATE TABLE",
+                                "operationParameters": {}}},
+                {"protocol": {"minReaderVersion": 1,
+                              "minWriterVersion": 2}},
+                {"metaData": _metadata_action(schema,
+                                              partition_columns or [])},
+            ])
         return t
 
+    def _ensure_schema_action(self, first_file: str) -> List[Dict]:
+        """If the table was created schema-less, record the real schema
+        from the first data file (delta requires metaData.schemaString)."""
+        md = self.metadata_at()
+        if md and md.get("schemaString"):
+            return []
+        import pyarrow.parquet as pq
+        schema = Schema.from_arrow(pq.read_schema(first_file))
+        new_md = md or _metadata_action(None, [])
+        new_md = dict(new_md)
+        new_md["schemaString"] = json.dumps(schema.to_json())
+        return [{"metaData": new_md}]
+
     def append_files(self, paths: List[str]) -> int:
-        add = []
+        actions: List[Dict[str, Any]] = [
+            {"commitInfo": {"timestamp": _now_ms(),
+                            "operation": "WRITE",
+                            "operationParameters": {"mode": "Append"}}}]
+        if paths:
+            actions.extend(self._ensure_schema_action(paths[0]))
         for p in paths:
             st = os.stat(p)
-            add.append({"path": os.path.abspath(p), "size": st.st_size,
-                        "mtime": int(st.st_mtime * 1000)})
+            rel = os.path.relpath(os.path.abspath(p), self.path)
+            actions.append({"add": {
+                "path": quote(rel),
+                "partitionValues": {},
+                "size": st.st_size,
+                "modificationTime": int(st.st_mtime * 1000),
+                "dataChange": True}})
         v = self.version + 1
-        self._commit(v, add, [])
+        self._commit(v, actions)
         return v
 
     def append_batch(self, batch, name_hint: str = "part") -> int:
         """Write a ColumnBatch as a new data file + commit."""
         from .parquet_io import write_batch_parquet
-        import uuid
         p = os.path.join(self.path,
                          f"{name_hint}-{uuid.uuid4().hex[:12]}.parquet")
         write_batch_parquet(batch, p)
         return self.append_files([p])
 
     def remove_files(self, paths: List[str]) -> int:
+        actions: List[Dict[str, Any]] = [
+            {"commitInfo": {"timestamp": _now_ms(),
+                            "operation": "DELETE",
+                            "operationParameters": {}}}]
+        for p in paths:
+            rel = os.path.relpath(os.path.abspath(p), self.path)
+            actions.append({"remove": {
+                "path": quote(rel),
+                "deletionTimestamp": _now_ms(),
+                "dataChange": True}})
         v = self.version + 1
-        self._commit(v, [], [os.path.abspath(p) for p in paths])
+        self._commit(v, actions)
         return v
 
-    # -- snapshots ---------------------------------------------------------
-    def files_at(self, version: Optional[int] = None) -> List[FileInfo]:
+    def checkpoint(self, version: Optional[int] = None) -> str:
+        """Write a single-part checkpoint parquet + _last_checkpoint."""
+        import pyarrow as pa
+        import pyarrow.parquet as pq
         target = self.version if version is None else version
-        live: Dict[str, FileInfo] = {}
-        for v in self.versions():
-            if v > target:
-                break
-            with open(os.path.join(self.log_dir, f"{v:010d}.json")) as f:
-                entry = json.load(f)
-            for a in entry.get("add", []):
-                live[a["path"]] = FileInfo(a["path"], a["size"], a["mtime"])
-            for r in entry.get("remove", []):
-                live.pop(r, None)
-        return sorted(live.values(), key=lambda f: f.name)
+        state = self._replay(target)
+        rows = []
+        if state["protocol"]:
+            rows.append({"protocol": state["protocol"], "metaData": None,
+                         "add": None, "remove": None})
+        if state["metaData"]:
+            rows.append({"protocol": None, "metaData": state["metaData"],
+                         "add": None, "remove": None})
+        for rel, (size, mtime, pv) in sorted(state["files"].items()):
+            rows.append({"protocol": None, "metaData": None,
+                         "add": {"path": quote(rel),
+                                 "partitionValues": pv or {},
+                                 "size": size,
+                                 "modificationTime": mtime,
+                                 "dataChange": True},
+                         "remove": None})
+        pv_type = pa.map_(pa.string(), pa.string())
+        schema = pa.schema([
+            ("protocol", pa.struct([("minReaderVersion", pa.int32()),
+                                    ("minWriterVersion", pa.int32())])),
+            ("metaData", pa.struct([
+                ("id", pa.string()), ("name", pa.string()),
+                ("description", pa.string()),
+                ("format", pa.struct([("provider", pa.string())])),
+                ("schemaString", pa.string()),
+                ("partitionColumns", pa.list_(pa.string())),
+                ("createdTime", pa.int64())])),
+            ("add", pa.struct([
+                ("path", pa.string()), ("partitionValues", pv_type),
+                ("size", pa.int64()), ("modificationTime", pa.int64()),
+                ("dataChange", pa.bool_())])),
+            ("remove", pa.struct([
+                ("path", pa.string()), ("deletionTimestamp", pa.int64()),
+                ("dataChange", pa.bool_())])),
+        ])
+        norm = []
+        for r in rows:
+            md = r["metaData"]
+            if md is not None:
+                md = {"id": md.get("id"), "name": md.get("name"),
+                      "description": md.get("description"),
+                      "format": {"provider":
+                                 (md.get("format") or {}).get(
+                                     "provider", "parquet")},
+                      "schemaString": md.get("schemaString"),
+                      "partitionColumns": md.get("partitionColumns", []),
+                      "createdTime": md.get("createdTime")}
+            a = r["add"]
+            if a is not None:
+                a = {"path": a["path"],
+                     "partitionValues": list(
+                         (a.get("partitionValues") or {}).items()),
+                     "size": a["size"],
+                     "modificationTime": a["modificationTime"],
+                     "dataChange": bool(a.get("dataChange", True))}
+            norm.append({"protocol": r["protocol"], "metaData": md,
+                         "add": a, "remove": r["remove"]})
+        table = pa.Table.from_pylist(norm, schema=schema)
+        out = os.path.join(self.log_dir,
+                           f"{target:020d}.checkpoint.parquet")
+        pq.write_table(table, out)
+        with open(os.path.join(self.log_dir, LAST_CHECKPOINT), "w") as f:
+            json.dump({"version": target, "size": len(rows)}, f)
+        return out
+
+    def clean_commits_before(self, version: int) -> None:
+        """Delete JSON commits below ``version`` (a checkpoint at or
+        above it must exist) — mirrors delta log retention cleanup and
+        proves snapshots reconstruct from the checkpoint alone."""
+        for f in os.listdir(self.log_dir):
+            m = _COMMIT_RE.match(f)
+            if m and int(m.group(1)) < version:
+                os.unlink(os.path.join(self.log_dir, f))
+
+
+def _now_ms() -> int:
+    return int(time.time() * 1000)
+
+
+def _metadata_action(schema: Optional[Schema],
+                     partition_columns: List[str]) -> Dict[str, Any]:
+    return {
+        "id": str(uuid.uuid4()),
+        "name": None,
+        "description": None,
+        "format": {"provider": "parquet", "options": {}},
+        "schemaString": json.dumps(schema.to_json()) if schema else "",
+        "partitionColumns": partition_columns,
+        "configuration": {},
+        "createdTime": _now_ms(),
+    }
+
+
+def _denull(d: Any) -> Any:
+    """Strip None-valued keys pyarrow fills into struct rows."""
+    if isinstance(d, dict):
+        return {k: v for k, v in d.items() if v is not None}
+    return d
 
 
 class DeltaTableRelation(FileBasedRelation):
-    """Relation over a DeltaTable snapshot.
+    """Relation over a Delta Lake snapshot.
 
     Signature = table version + path (reference
     delta/DeltaLakeRelation.scala:40-44): any commit changes the
@@ -157,12 +432,17 @@ class DeltaTableRelation(FileBasedRelation):
     @property
     def schema(self) -> Schema:
         if self._schema is None:
-            files = self.all_files()
-            if not files:
-                raise HyperspaceException(
-                    f"Empty delta table: {self.table.path}")
-            import pyarrow.parquet as pq
-            self._schema = Schema.from_arrow(pq.read_schema(files[0].name))
+            md = self.table.metadata_at(self.version_as_of)
+            if md and md.get("schemaString"):
+                self._schema = Schema.from_json(md["schemaString"])
+            else:
+                files = self.all_files()
+                if not files:
+                    raise HyperspaceException(
+                        f"Empty delta table: {self.table.path}")
+                import pyarrow.parquet as pq
+                self._schema = Schema.from_arrow(
+                    pq.read_schema(files[0].name))
         return self._schema
 
     def all_files(self) -> List[FileInfo]:

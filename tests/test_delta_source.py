@@ -116,9 +116,9 @@ def test_time_travel_query_uses_pinned_snapshot(env):
 def test_concurrent_commit_collision(env, tmp_path):
     session, h, table, rng = env
     v = table.version
-    table._commit(v + 1, [], [])
+    table._commit(v + 1, [{"commitInfo": {"timestamp": 0}}])
     with pytest.raises(HyperspaceException, match="lost race"):
-        table._commit(v + 1, [], [])
+        table._commit(v + 1, [{"commitInfo": {"timestamp": 0}}])
 
 
 # ---------------------------------------------------------------------------
