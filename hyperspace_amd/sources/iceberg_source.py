@@ -1,18 +1,31 @@
-"""Iceberg-style snapshot table source.
+"""Apache Iceberg table source — REAL Iceberg (HadoopTables) format.
 
 Reference: index/sources/iceberg/ — IcebergRelation (signature =
 snapshotId + table location, iceberg/IcebergRelation.scala:63-68; files
 from file-scan tasks; snapshot time travel).
 
-The table format here mirrors Iceberg's HadoopTables layout minimally:
+This reads and writes the actual Iceberg v1 table layout rather than a
+lookalike:
 
-    <table>/metadata/v<N>.metadata.json   {"snapshot-id", "timestamp-ms",
-                                           "manifest": [{path,size,mtime}]}
-    <table>/metadata/version-hint.text    current N
-    <table>/*.parquet                     data files
+    <table>/metadata/v<N>.metadata.json   table metadata: format-version,
+        table-uuid, schemas (field-ids), partition-specs, snapshots
+        (each with a "manifest-list" avro path), current-snapshot-id,
+        snapshot-log
+    <table>/metadata/version-hint.text    current metadata version N
+    <table>/metadata/snap-<id>-1-<uuid>.avro   manifest list: one
+        manifest_file record per manifest (manifest_path,
+        manifest_length, partition_spec_id, added_snapshot_id)
+    <table>/metadata/<uuid>-m0.avro       manifest: manifest_entry
+        records (status 0=EXISTING 1=ADDED 2=DELETED, snapshot_id,
+        data_file{file_path, file_format, partition, record_count,
+        file_size_in_bytes})
+    <table>/data/*.parquet                data files
 
-Each commit writes a full manifest (Iceberg snapshots are self-contained
-file listings, unlike the delta source's log of deltas).
+Snapshot reconstruction walks current (or pinned) snapshot ->
+manifest-list avro -> manifest avro -> live data_file entries, exactly
+the path the iceberg library's planFiles takes.  Avro decode is the
+in-repo container reader (avro_io.read_avro_records — nested records,
+unions, maps).
 """
 
 from __future__ import annotations
@@ -22,13 +35,59 @@ import os
 import tempfile
 import time
 import uuid
-from typing import Dict, List, Optional
+from typing import Any, Dict, List, Optional
 
+from .avro_io import read_avro_records, write_avro_records
 from .interfaces import FileBasedRelation, FileBasedSourceProvider
 from ..exceptions import HyperspaceException
-from ..log.entry import FileInfo, Relation, Schema
+from ..log.entry import FileInfo, Relation, Schema, SchemaField
 
 META_DIR = "metadata"
+
+_ICEBERG_TO_SPARK = {
+    "long": "long", "int": "integer", "double": "double",
+    "float": "float", "string": "string", "boolean": "boolean",
+    "date": "date", "timestamp": "timestamp", "timestamptz": "timestamp",
+    "binary": "binary", "uuid": "string",
+}
+_SPARK_TO_ICEBERG = {
+    "long": "long", "integer": "int", "double": "double",
+    "float": "float", "string": "string", "boolean": "boolean",
+    "date": "date", "timestamp": "timestamp", "binary": "binary",
+    "short": "int", "byte": "int",
+}
+
+_MANIFEST_ENTRY_SCHEMA = {
+    "type": "record", "name": "manifest_entry", "fields": [
+        {"name": "status", "type": "int", "field-id": 0},
+        {"name": "snapshot_id", "type": ["null", "long"], "field-id": 1},
+        {"name": "data_file", "field-id": 2, "type": {
+            "type": "record", "name": "r2", "fields": [
+                {"name": "file_path", "type": "string", "field-id": 100},
+                {"name": "file_format", "type": "string",
+                 "field-id": 101},
+                {"name": "partition", "field-id": 102, "type": {
+                    "type": "record", "name": "r102", "fields": []}},
+                {"name": "record_count", "type": "long", "field-id": 103},
+                {"name": "file_size_in_bytes", "type": "long",
+                 "field-id": 104},
+            ]}},
+    ]}
+
+_MANIFEST_LIST_SCHEMA = {
+    "type": "record", "name": "manifest_file", "fields": [
+        {"name": "manifest_path", "type": "string", "field-id": 500},
+        {"name": "manifest_length", "type": "long", "field-id": 501},
+        {"name": "partition_spec_id", "type": "int", "field-id": 502},
+        {"name": "added_snapshot_id", "type": ["null", "long"],
+         "field-id": 503},
+    ]}
+
+
+def _strip_uri(p: str) -> str:
+    if p.startswith("file://"):
+        return p[len("file://"):]
+    return p
 
 
 class IcebergTable:
@@ -37,11 +96,31 @@ class IcebergTable:
         self.meta_dir = os.path.join(self.path, META_DIR)
 
     @staticmethod
-    def create(path: str) -> "IcebergTable":
+    def create(path: str, schema: Optional[Schema] = None
+               ) -> "IcebergTable":
         t = IcebergTable(path)
         os.makedirs(t.meta_dir, exist_ok=True)
         if t.current_version() is None:
-            t._commit([], snapshot_id=0)
+            meta = {
+                "format-version": 1,
+                "table-uuid": str(uuid.uuid4()),
+                "location": "file://" + t.path,
+                "last-updated-ms": _now_ms(),
+                "last-column-id": len(schema.fields) if schema else 0,
+                "schema": _iceberg_schema(schema),
+                "schemas": [_iceberg_schema(schema)],
+                "current-schema-id": 0,
+                "partition-spec": [],
+                "partition-specs": [{"spec-id": 0, "fields": []}],
+                "default-spec-id": 0,
+                "last-partition-id": 999,
+                "properties": {},
+                "current-snapshot-id": -1,
+                "snapshots": [],
+                "snapshot-log": [],
+                "metadata-log": [],
+            }
+            t._write_metadata(1, meta)
         return t
 
     # -- metadata ----------------------------------------------------------
@@ -50,97 +129,226 @@ class IcebergTable:
         try:
             with open(hint) as f:
                 return int(f.read().strip())
-        except FileNotFoundError:
-            return None
+        except (FileNotFoundError, ValueError):
+            if not os.path.isdir(self.meta_dir):
+                return None
+            vs = [int(n[1:-len(".metadata.json")])
+                  for n in os.listdir(self.meta_dir)
+                  if n.startswith("v") and n.endswith(".metadata.json")
+                  and n[1:-len(".metadata.json")].isdigit()]
+            return max(vs) if vs else None
 
-    def _snapshot(self, version: int) -> Dict:
+    def metadata(self, version: Optional[int] = None) -> Dict[str, Any]:
+        v = self.current_version() if version is None else version
+        if v is None:
+            raise HyperspaceException(
+                f"Not an iceberg table: {self.path}")
         with open(os.path.join(self.meta_dir,
-                               f"v{version}.metadata.json")) as f:
+                               f"v{v}.metadata.json")) as f:
             return json.load(f)
 
     def snapshots(self) -> List[Dict]:
-        if not os.path.isdir(self.meta_dir):
-            return []
-        out = []
-        for name in sorted(os.listdir(self.meta_dir)):
-            if name.startswith("v") and name.endswith(".metadata.json"):
-                with open(os.path.join(self.meta_dir, name)) as f:
-                    out.append(json.load(f))
-        return out
+        return list(self.metadata().get("snapshots", []))
 
     @property
     def snapshot_id(self) -> int:
-        v = self.current_version()
-        if v is None:
-            raise HyperspaceException(f"Not an iceberg table: {self.path}")
-        return self._snapshot(v)["snapshot-id"]
+        sid = self.metadata().get("current-snapshot-id", -1)
+        if sid in (None, -1):
+            raise HyperspaceException(
+                f"Iceberg table has no snapshot: {self.path}")
+        return sid
 
-    def _commit(self, manifest: List[Dict], snapshot_id: Optional[int]
-                = None):
-        v = (self.current_version() or 0) + 1 \
-            if self.current_version() is not None else 0
-        if snapshot_id is None:
-            snapshot_id = uuid.uuid4().int & 0x7FFFFFFFFFFFFFFF
-        entry = {"snapshot-id": snapshot_id,
-                 "timestamp-ms": int(time.time() * 1000),
-                 "manifest": manifest}
-        target = os.path.join(self.meta_dir, f"v{v}.metadata.json")
+    def _write_metadata(self, version: int, meta: Dict[str, Any]):
+        target = os.path.join(self.meta_dir,
+                              f"v{version}.metadata.json")
         fd, tmp = tempfile.mkstemp(dir=self.meta_dir, prefix=".tmp_")
         try:
             with os.fdopen(fd, "w") as f:
-                json.dump(entry, f)
+                json.dump(meta, f)
             try:
                 os.link(tmp, target)
             except FileExistsError:
                 raise HyperspaceException(
-                    f"Concurrent commit lost race for v{v}")
+                    f"Concurrent commit lost race for v{version}")
         finally:
             os.unlink(tmp)
         with open(os.path.join(self.meta_dir, "version-hint.text"),
                   "w") as f:
-            f.write(str(v))
+            f.write(str(version))
+
+    # -- manifest plumbing --------------------------------------------------
+    def _write_manifest(self, entries: List[Dict], snapshot_id: int
+                        ) -> Dict[str, Any]:
+        name = f"{uuid.uuid4().hex}-m0.avro"
+        p = os.path.join(self.meta_dir, name)
+        write_avro_records(p, _MANIFEST_ENTRY_SCHEMA, entries)
+        return {"manifest_path": "file://" + p,
+                "manifest_length": os.stat(p).st_size,
+                "partition_spec_id": 0,
+                "added_snapshot_id": snapshot_id}
+
+    def _commit_snapshot(self, manifests: List[Dict[str, Any]],
+                         operation: str):
+        v = self.current_version()
+        meta = self.metadata(v)
+        sid = uuid.uuid4().int & 0x7FFFFFFFFFFFFFFF
+        list_name = f"snap-{sid}-1-{uuid.uuid4().hex}.avro"
+        list_path = os.path.join(self.meta_dir, list_name)
+        write_avro_records(list_path, _MANIFEST_LIST_SCHEMA, manifests)
+        snap = {"snapshot-id": sid,
+                "timestamp-ms": _now_ms(),
+                "summary": {"operation": operation},
+                "manifest-list": "file://" + list_path,
+                "schema-id": meta.get("current-schema-id", 0)}
+        parent = meta.get("current-snapshot-id", -1)
+        if parent not in (None, -1):
+            snap["parent-snapshot-id"] = parent
+        meta = dict(meta)
+        meta["snapshots"] = list(meta.get("snapshots", [])) + [snap]
+        meta["current-snapshot-id"] = sid
+        meta["last-updated-ms"] = snap["timestamp-ms"]
+        meta["snapshot-log"] = list(meta.get("snapshot-log", [])) + [
+            {"timestamp-ms": snap["timestamp-ms"], "snapshot-id": sid}]
+        self._write_metadata(v + 1, meta)
+
+    def _snapshot_manifests(self, snap: Dict[str, Any]) -> List[Dict]:
+        ml = _strip_uri(snap["manifest-list"])
+        return read_avro_records(ml)
+
+    def _live_entries(self, snapshot_id: Optional[int] = None
+                      ) -> List[Dict]:
+        if snapshot_id is None:
+            snapshot_id = self.snapshot_id
+        snap = next((s for s in self.snapshots()
+                     if s["snapshot-id"] == snapshot_id), None)
+        if snap is None:
+            raise HyperspaceException(
+                f"No snapshot {snapshot_id} in {self.path}")
+        out = []
+        for m in self._snapshot_manifests(snap):
+            for e in read_avro_records(_strip_uri(m["manifest_path"])):
+                if e.get("status") != 2:  # not DELETED
+                    out.append(e)
+        return out
 
     # -- operations --------------------------------------------------------
-    def _manifest_now(self) -> List[Dict]:
+    def _ensure_schema(self, first_file: str):
+        meta = self.metadata()
+        if meta.get("schema", {}).get("fields"):
+            return
+        import pyarrow.parquet as pq
+        schema = Schema.from_arrow(pq.read_schema(first_file))
         v = self.current_version()
-        return list(self._snapshot(v)["manifest"]) if v is not None else []
+        meta = dict(meta)
+        meta["schema"] = _iceberg_schema(schema)
+        meta["schemas"] = [meta["schema"]]
+        meta["last-column-id"] = len(schema.fields)
+        self._write_metadata(v + 1, meta)
 
     def append_files(self, paths: List[str]):
-        manifest = self._manifest_now()
+        if paths:
+            self._ensure_schema(paths[0])
+        sid_placeholder = 0
+        entries = []
+        carried = []
+        meta = self.metadata()
+        if meta.get("current-snapshot-id", -1) not in (None, -1):
+            snap = next(s for s in self.snapshots()
+                        if s["snapshot-id"] == meta["current-snapshot-id"])
+            carried = self._snapshot_manifests(snap)
         for p in paths:
             st = os.stat(p)
-            manifest.append({"path": os.path.abspath(p),
-                             "size": st.st_size,
-                             "mtime": int(st.st_mtime * 1000)})
-        self._commit(manifest)
+            try:
+                import pyarrow.parquet as pq
+                nrec = pq.read_metadata(p).num_rows
+            except Exception:
+                nrec = 0
+            entries.append({"status": 1, "snapshot_id": sid_placeholder,
+                            "data_file": {
+                                "file_path": "file://" + os.path.abspath(p),
+                                "file_format": "PARQUET",
+                                "partition": {},
+                                "record_count": nrec,
+                                "file_size_in_bytes": st.st_size}})
+        new_manifest = self._write_manifest(entries, sid_placeholder)
+        self._commit_snapshot(carried + [new_manifest], "append")
 
     def append_batch(self, batch, name_hint: str = "data"):
         from .parquet_io import write_batch_parquet
-        p = os.path.join(self.path,
+        data_dir = os.path.join(self.path, "data")
+        os.makedirs(data_dir, exist_ok=True)
+        p = os.path.join(data_dir,
                          f"{name_hint}-{uuid.uuid4().hex[:12]}.parquet")
         write_batch_parquet(batch, p)
         self.append_files([p])
 
     def remove_files(self, paths: List[str]):
+        """Rewrite manifests without the removed files (the iceberg
+        'delete' path via manifest rewrite)."""
         gone = {os.path.abspath(p) for p in paths}
-        manifest = [m for m in self._manifest_now()
-                    if m["path"] not in gone]
-        self._commit(manifest)
+        live = [e for e in self._live_entries()
+                if os.path.abspath(_strip_uri(
+                    e["data_file"]["file_path"])) not in gone]
+        for e in live:
+            e["status"] = 0  # EXISTING
+        manifest = self._write_manifest(live, 0)
+        self._commit_snapshot([manifest], "delete")
 
     def files_for_snapshot(self, snapshot_id: Optional[int] = None
                            ) -> List[FileInfo]:
-        if snapshot_id is None:
-            v = self.current_version()
-            snap = self._snapshot(v)
-        else:
-            snap = next((s for s in self.snapshots()
-                         if s["snapshot-id"] == snapshot_id), None)
-            if snap is None:
-                raise HyperspaceException(
-                    f"No snapshot {snapshot_id} in {self.path}")
-        return sorted((FileInfo(m["path"], m["size"], m["mtime"])
-                       for m in snap["manifest"]),
-                      key=lambda f: f.name)
+        out = []
+        for e in self._live_entries(snapshot_id):
+            df = e["data_file"]
+            p = _strip_uri(df["file_path"])
+            if not os.path.isabs(p):
+                p = os.path.join(self.path, p)
+            try:
+                mtime = int(os.stat(p).st_mtime * 1000)
+            except OSError:
+                mtime = 0
+            out.append(FileInfo(p, int(df["file_size_in_bytes"]), mtime))
+        return sorted(out, key=lambda f: f.name)
+
+    def schema(self) -> Optional[Schema]:
+        meta = self.metadata()
+        sid = meta.get("current-schema-id", 0)
+        struct = None
+        for s in meta.get("schemas", []):
+            if s.get("schema-id", 0) == sid:
+                struct = s
+                break
+        struct = struct or meta.get("schema")
+        if not struct or not struct.get("fields"):
+            return None
+        fields = []
+
+        def walk(fs, prefix=""):
+            for f in fs:
+                t = f["type"]
+                if isinstance(t, dict) and t.get("type") == "struct":
+                    walk(t["fields"], prefix + f["name"] + ".")
+                elif isinstance(t, str):
+                    spark = _ICEBERG_TO_SPARK.get(
+                        t, "decimal" if t.startswith("decimal") else t)
+                    fields.append(SchemaField(prefix + f["name"], spark,
+                                              not f.get("required",
+                                                        False)))
+        walk(struct["fields"])
+        return Schema(fields)
+
+
+def _now_ms() -> int:
+    return int(time.time() * 1000)
+
+
+def _iceberg_schema(schema: Optional[Schema]) -> Dict[str, Any]:
+    fields = []
+    if schema is not None:
+        for i, f in enumerate(schema.fields):
+            fields.append({"id": i + 1, "name": f.name,
+                           "required": not f.nullable,
+                           "type": _SPARK_TO_ICEBERG.get(f.type, f.type)})
+    return {"type": "struct", "schema-id": 0, "fields": fields}
 
 
 class IcebergTableRelation(FileBasedRelation):
@@ -176,12 +384,15 @@ class IcebergTableRelation(FileBasedRelation):
     @property
     def schema(self) -> Schema:
         if self._schema is None:
-            files = self.all_files()
-            if not files:
-                raise HyperspaceException(
-                    f"Empty iceberg table: {self.table.path}")
-            import pyarrow.parquet as pq
-            self._schema = Schema.from_arrow(pq.read_schema(files[0].name))
+            self._schema = self.table.schema()
+            if self._schema is None:
+                files = self.all_files()
+                if not files:
+                    raise HyperspaceException(
+                        f"Empty iceberg table: {self.table.path}")
+                import pyarrow.parquet as pq
+                self._schema = Schema.from_arrow(
+                    pq.read_schema(files[0].name))
         return self._schema
 
     def all_files(self) -> List[FileInfo]:
