@@ -50,9 +50,13 @@ def main():
     distributed = world > 1
     on_gpu = torch.cuda.is_available() and args.device != "cpu"
     if distributed:
-        backend = "nccl" if on_gpu else "gloo"
+        # RCCL refuses two ranks on one device, so ranks > GPUs (a
+        # validation shape on a 1-GPU box) exchange via gloo with host
+        # staging; on the 8-GPU node this is nccl (=RCCL) end to end
+        backend = ("nccl" if on_gpu
+                   and world <= torch.cuda.device_count() else "gloo")
         if on_gpu:
-            torch.cuda.set_device(local_rank)
+            torch.cuda.set_device(local_rank % torch.cuda.device_count())
         dist.init_process_group(backend=backend)
 
     import hyperspace_amd as hs

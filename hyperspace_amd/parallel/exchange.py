@@ -49,7 +49,12 @@ from ..execution.columnar import ColumnBatch, StringColumn
 
 
 def _collective_device(device: torch.device) -> torch.device:
-    """Device collectives must live on for the active backend."""
+    """Device collectives must live on for the active backend.
+
+    nccl/RCCL moves device tensors directly over xGMI.  gloo stages via
+    host — that path exists so multi-process GPU builds can be validated
+    on a single GPU (RCCL refuses two ranks on one device); on a real
+    multi-GPU node the backend is nccl and nothing touches the host."""
     if dist.get_backend() == "nccl":
         return device if device.type == "cuda" else torch.device("cuda")
     return torch.device("cpu")
@@ -222,7 +227,7 @@ class BucketExchange:
         recv_bytes = [c * row_bytes for c in recv_rows]
         n_recv = sum(recv_rows)
 
-        if cdev.type == "cuda" and device.type != "cuda":
+        if packed.device != cdev:
             packed = packed.to(cdev)
         recv_buf = torch.empty(sum(recv_bytes), dtype=torch.uint8,
                                device=packed.device)

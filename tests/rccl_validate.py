@@ -33,8 +33,14 @@ def main():
     assert torch.cuda.is_available(), "needs a GPU"
     dev_idx = local_rank % torch.cuda.device_count()
     torch.cuda.set_device(dev_idx)
-    dist.init_process_group(backend="nccl")
+    # RCCL refuses >1 rank per device; with more ranks than GPUs the
+    # exchange stages via gloo/host while all compute stays on device
+    backend = ("nccl" if world <= torch.cuda.device_count() else "gloo")
+    dist.init_process_group(backend=backend)
     device = torch.device(f"cuda:{dev_idx}")
+    if rank == 0:
+        print(f"backend={backend} world={world} "
+              f"gpus={torch.cuda.device_count()}", flush=True)
 
     from hyperspace_amd.execution.columnar import ColumnBatch, StringColumn
     from hyperspace_amd.parallel.exchange import BucketExchange
