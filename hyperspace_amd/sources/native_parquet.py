@@ -30,7 +30,7 @@ import numpy as np
 MAGIC = b"PAR1"
 
 # parquet physical types
-T_INT32, T_INT64, T_FLOAT, T_DOUBLE = 1, 2, 4, 5
+T_INT32, T_INT64, T_FLOAT, T_DOUBLE, T_BYTE_ARRAY = 1, 2, 4, 5, 6
 _NP_TO_PARQUET = {
     np.dtype("int64"): T_INT64,
     np.dtype("int32"): T_INT32,
@@ -40,11 +40,30 @@ _NP_TO_PARQUET = {
 _PARQUET_TO_NP = {v: k for k, v in _NP_TO_PARQUET.items()}
 
 ENC_PLAIN = 0
+ENC_PLAIN_DICTIONARY = 2
 ENC_RLE = 3
+ENC_RLE_DICTIONARY = 8
 CODEC_UNCOMPRESSED = 0
 PAGE_DATA = 0
+PAGE_DICTIONARY = 2
 REP_REQUIRED = 0
 REP_OPTIONAL = 1
+CONV_UTF8 = 0
+
+
+class StrCol:
+    """Dictionary-encoded string column for the native writer/reader:
+    int32 codes into a sorted list of unique utf-8 values (the in-memory
+    layout of execution.columnar.StringColumn, minus torch)."""
+
+    __slots__ = ("codes", "values")
+
+    def __init__(self, codes: np.ndarray, values: List[str]):
+        self.codes = np.ascontiguousarray(codes, dtype=np.int32)
+        self.values = values
+
+    def __len__(self):
+        return len(self.codes)
 
 
 # ---------------------------------------------------------------------------
@@ -204,7 +223,8 @@ class TReader:
 # ---------------------------------------------------------------------------
 
 def _page_header(num_values: int, nbytes: int,
-                 has_def_levels: bool = False) -> bytes:
+                 has_def_levels: bool = False,
+                 encoding: int = ENC_PLAIN) -> bytes:
     w = TWriter()
     w.struct_begin()
     w.field_i32(1, PAGE_DATA)
@@ -212,13 +232,59 @@ def _page_header(num_values: int, nbytes: int,
     w.field_i32(3, nbytes)
     w.field_struct_begin(5)  # DataPageHeader
     w.field_i32(1, num_values)
-    w.field_i32(2, ENC_PLAIN)
+    w.field_i32(2, encoding)
     # definition_level_encoding: RLE for OPTIONAL columns
     w.field_i32(3, ENC_RLE if has_def_levels else ENC_PLAIN)
     w.field_i32(4, ENC_PLAIN)  # repetition_level_encoding
     w.struct_end()
     w.struct_end()
     return bytes(w.buf)
+
+
+def _dict_page_header(num_values: int, nbytes: int) -> bytes:
+    w = TWriter()
+    w.struct_begin()
+    w.field_i32(1, PAGE_DICTIONARY)
+    w.field_i32(2, nbytes)
+    w.field_i32(3, nbytes)
+    w.field_struct_begin(7)  # DictionaryPageHeader
+    w.field_i32(1, num_values)
+    w.field_i32(2, ENC_PLAIN_DICTIONARY)
+    w.struct_end()
+    w.struct_end()
+    return bytes(w.buf)
+
+
+def _bitpack(vals: np.ndarray, width: int) -> bytes:
+    """Bit-pack values LSB-first at ``width`` bits each (the RLE-hybrid
+    bit-packed run layout; values padded to a multiple of 8)."""
+    n = len(vals)
+    pad = (-n) % 8
+    if pad:
+        vals = np.concatenate(
+            [vals, np.zeros(pad, dtype=vals.dtype)])
+    bits = ((vals[:, None].astype(np.uint64)
+             >> np.arange(width, dtype=np.uint64)) & 1).astype(np.uint8)
+    return np.packbits(bits.reshape(-1), bitorder="little").tobytes()
+
+
+def _dict_indices_payload(codes: np.ndarray, num_dict: int) -> Tuple[
+        bytes, int]:
+    """[u8 bit-width][one bit-packed RLE-hybrid run] + the bit width."""
+    bw = max(1, (num_dict - 1).bit_length()) if num_dict > 1 else 1
+    groups = (len(codes) + 7) // 8
+    body = _varint((groups << 1) | 1) + _bitpack(codes, bw)
+    return bytes([bw]) + body, bw
+
+
+def _plain_byte_array_dict(values: List[str]) -> bytes:
+    """PLAIN-encoded BYTE_ARRAY dictionary page payload."""
+    out = bytearray()
+    for v in values:
+        b = v.encode("utf-8")
+        out += struct.pack("<I", len(b))
+        out += b
+    return bytes(out)
 
 
 def _statistics(arr: np.ndarray) -> Tuple[bytes, bytes]:
@@ -277,11 +343,15 @@ def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str,
     written OPTIONAL with RLE def-levels and compacted values.
     Returns (size, mtime_ms)."""
     names = list(columns.keys())
-    arrays = [np.ascontiguousarray(columns[n]) for n in names]
+    arrays: List[Any] = []
+    for n in names:
+        c = columns[n]
+        arrays.append(c if isinstance(c, StrCol)
+                      else np.ascontiguousarray(c))
     num_rows = len(arrays[0]) if arrays else 0
     col_masks: List[Optional[np.ndarray]] = []
     for n, a in zip(names, arrays):
-        if a.dtype not in _NP_TO_PARQUET:
+        if not isinstance(a, StrCol) and a.dtype not in _NP_TO_PARQUET:
             raise ValueError(f"dtype {a.dtype} not supported natively")
         assert len(a) == num_rows
         m = (masks or {}).get(n)
@@ -294,6 +364,47 @@ def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str,
     layouts: List[ColumnChunkLayout] = []
     offset = 4  # after magic
     for name, arr, mask in zip(names, arrays, col_masks):
+        if isinstance(arr, StrCol):
+            # dictionary-encoded BYTE_ARRAY chunk: PLAIN dictionary page
+            # + one bit-packed RLE_DICTIONARY-style data page — the
+            # on-disk twin of the in-memory StringColumn (codes travel,
+            # values parse once)
+            dict_payload = _plain_byte_array_dict(arr.values)
+            dict_header = _dict_page_header(len(arr.values),
+                                            len(dict_payload))
+            dict_off = offset
+            offset += len(dict_header) + len(dict_payload)
+            chunks.append(dict_header)
+            chunks.append(dict_payload)
+
+            codes = arr.codes if mask is None else arr.codes[mask]
+            null_count = int(num_rows - len(codes))
+            idx_payload, bw = _dict_indices_payload(codes,
+                                                    len(arr.values))
+            lvl_len = 0
+            if mask is None:
+                payload: Any = idx_payload
+            else:
+                levels = _def_levels_payload(mask)
+                lvl_len = len(levels)
+                payload = levels + idx_payload
+            header = _page_header(num_rows, len(payload), mask is not None,
+                                  encoding=ENC_PLAIN_DICTIONARY)
+            col_meta.append((name, T_BYTE_ARRAY, offset,
+                             offset - dict_off + len(header) + len(payload),
+                             num_rows, b"", b"", null_count,
+                             mask is not None, dict_off))
+            layouts.append(ColumnChunkLayout(
+                name, np.dtype("int32"),
+                [("dict", offset + len(header) + lvl_len + 1,
+                  offset + len(header) + len(payload), num_rows, bw)],
+                num_rows, "dict",
+                (dict_off + len(dict_header), len(arr.values)), [mask],
+                is_string=True, str_values=list(arr.values)))
+            chunks.append(header)
+            chunks.append(payload)
+            offset += len(header) + len(payload)
+            continue
         lvl_len = 0
         if mask is None:
             # zero-copy: the PLAIN payload IS the little-endian column
@@ -312,7 +423,7 @@ def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str,
         mn, mx = _statistics(valid) if len(valid) else (b"", b"")
         col_meta.append((name, _NP_TO_PARQUET[arr.dtype], offset,
                          len(header) + nbytes, num_rows, mn, mx,
-                         null_count, mask is not None))
+                         null_count, mask is not None, None))
         layouts.append(ColumnChunkLayout(
             name, arr.dtype,
             [("plain", offset + len(header) + lvl_len, num_rows)],
@@ -331,12 +442,14 @@ def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str,
     w.field_string(4, "schema")
     w.field_i32(5, len(names))
     w.struct_end()
-    for name, ptype, *rest in col_meta:
-        optional = rest[-1]
+    for name, ptype, off, nbytes, nvals, mn, mx, null_count, optional, \
+            dict_off in col_meta:
         w.struct_elem_begin()
         w.field_i32(1, ptype)
         w.field_i32(3, REP_OPTIONAL if optional else REP_REQUIRED)
         w.field_string(4, name)
+        if ptype == T_BYTE_ARRAY:
+            w.field_i32(6, CONV_UTF8)  # converted_type: UTF8
         w.struct_end()
     w.field_i64(3, num_rows)
     # row_groups
@@ -344,19 +457,19 @@ def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str,
     w.struct_elem_begin()
     total_bytes = sum(m[3] for m in col_meta)
     w.field_list_begin(1, CT_STRUCT, len(col_meta))
-    for name, ptype, off, nbytes, nvals, mn, mx, null_count, optional \
-            in col_meta:
+    for name, ptype, off, nbytes, nvals, mn, mx, null_count, optional, \
+            dict_off in col_meta:
         w.struct_elem_begin()  # ColumnChunk
-        w.field_i64(2, off)  # file_offset
+        w.field_i64(2, dict_off if dict_off is not None else off)
         w.field_struct_begin(3)  # ColumnMetaData
         w.field_i32(1, ptype)
-        if optional:
-            w.field_list_begin(2, CT_I32, 2)
-            w.i32_elem(ENC_PLAIN)
-            w.i32_elem(ENC_RLE)
-        else:
-            w.field_list_begin(2, CT_I32, 1)
-            w.i32_elem(ENC_PLAIN)
+        encs = ([ENC_PLAIN_DICTIONARY, ENC_RLE]
+                if dict_off is not None
+                else ([ENC_PLAIN, ENC_RLE] if optional
+                      else [ENC_PLAIN]))
+        w.field_list_begin(2, CT_I32, len(encs))
+        for e in encs:
+            w.i32_elem(e)
         w.field_list_begin(3, CT_BINARY, 1)
         w.buf += _varint(len(name.encode()))
         w.buf += name.encode()
@@ -365,6 +478,8 @@ def write_parquet_native(columns: "Dict[str, np.ndarray]", path: str,
         w.field_i64(6, nbytes)
         w.field_i64(7, nbytes)
         w.field_i64(9, off)  # data_page_offset
+        if dict_off is not None:
+            w.field_i64(11, dict_off)  # dictionary_page_offset
         if mn or null_count:
             w.field_struct_begin(12)  # Statistics
             if mn:
@@ -411,12 +526,17 @@ class ColumnChunkLayout:
     ``page_masks`` parallels ``pages``: a bool validity array per page for
     OPTIONAL columns with nulls in that page, else None.  A page's stored
     values are compacted (non-null only); num values counts rows.
+
+    ``is_string``: BYTE_ARRAY dictionary chunk — decoded values are int32
+    codes into ``str_values`` (parsed lazily from the dictionary page
+    when not supplied by the writer's layout cache).
     """
     __slots__ = ("name", "np_dtype", "pages", "num_values", "encoding",
-                 "dict_page", "page_masks")
+                 "dict_page", "page_masks", "is_string", "str_values")
 
     def __init__(self, name, np_dtype, pages, num_values,
-                 encoding="plain", dict_page=None, page_masks=None):
+                 encoding="plain", dict_page=None, page_masks=None,
+                 is_string=False, str_values=None):
         self.name = name
         self.np_dtype = np_dtype
         self.pages = pages
@@ -424,6 +544,23 @@ class ColumnChunkLayout:
         self.encoding = encoding
         self.dict_page = dict_page
         self.page_masks = page_masks or [None] * len(pages)
+        self.is_string = is_string
+        self.str_values = str_values
+
+    def dict_values(self, data) -> List[str]:
+        """The string dictionary: cached from the writer, else parsed
+        from the PLAIN dictionary page payload in ``data``."""
+        if self.str_values is None:
+            off, n = self.dict_page
+            vals = []
+            pos = off
+            for _ in range(n):
+                ln = struct.unpack_from("<I", data, pos)[0]
+                pos += 4
+                vals.append(bytes(data[pos:pos + ln]).decode("utf-8"))
+                pos += ln
+            self.str_values = vals
+        return self.str_values
 
     @property
     def has_nulls(self) -> bool:
@@ -546,11 +683,20 @@ def _walk_row_group(rg, pf_schema, data, want
         if encs - {"PLAIN", "RLE", "BIT_PACKED", "PLAIN_DICTIONARY",
                    "RLE_DICTIONARY"}:
             return None
-        ptype = {"INT64": T_INT64, "INT32": T_INT32, "FLOAT": T_FLOAT,
-                 "DOUBLE": T_DOUBLE}.get(col.physical_type)
-        if ptype is None:
-            return None
-        np_dtype = _PARQUET_TO_NP[ptype]
+        is_string = False
+        if col.physical_type == "BYTE_ARRAY":
+            # dictionary-encoded strings decode natively (codes + one
+            # dictionary parse); PLAIN/compressed strings -> pyarrow
+            if not is_dict or codec != "UNCOMPRESSED":
+                return None
+            is_string = True
+            np_dtype = np.dtype("int32")
+        else:
+            ptype = {"INT64": T_INT64, "INT32": T_INT32, "FLOAT": T_FLOAT,
+                     "DOUBLE": T_DOUBLE}.get(col.physical_type)
+            if ptype is None:
+                return None
+            np_dtype = _PARQUET_TO_NP[ptype]
         # OPTIONAL columns carry a def-level prefix per page; when the
         # chunk statistics prove null_count == 0 the levels are skipped
         # without decoding, else each page's levels become a validity
@@ -647,6 +793,8 @@ def _walk_row_group(rg, pf_schema, data, want
                 bit_width = data[values_off]
                 pages.append(("dict", values_off + 1, page_end,
                               num_values, bit_width))
+            elif is_string:
+                return None  # PLAIN byte-array page (dict overflow)
             elif page_enc == ENC_PLAIN:
                 # also reached as the writer's mid-chunk fallback when a
                 # dictionary overflows: later pages switch to PLAIN
@@ -671,7 +819,49 @@ def _walk_row_group(rg, pf_schema, data, want
             enc_kind = "plain"
         out.append(ColumnChunkLayout(name, np_dtype, pages,
                                      col.num_values, enc_kind, dict_page,
-                                     page_masks))
+                                     page_masks, is_string=is_string))
+    return out
+
+
+def _decode_rle_indices(data, off: int, end: int, n: int, bw: int
+                        ) -> np.ndarray:
+    """Host RLE/bit-packed hybrid decode of dictionary indices at
+    ``bw`` bits (testing / CPU path; the device twin is k_rle_decode)."""
+    out = np.empty(n, dtype=np.int32)
+    pos = off
+    filled = 0
+    vbytes = (bw + 7) // 8
+    while filled < n and pos < end:
+        header = 0
+        shift = 0
+        while True:
+            b = data[pos]
+            pos += 1
+            header |= (b & 0x7F) << shift
+            if not b & 0x80:
+                break
+            shift += 7
+        if header & 1:  # bit-packed run of (header>>1) groups of 8
+            groups = header >> 1
+            nbytes = groups * bw
+            packed = np.frombuffer(data, dtype=np.uint8, count=nbytes,
+                                   offset=pos)
+            bits = np.unpackbits(packed, bitorder="little")
+            vals = bits.reshape(-1, bw).astype(np.uint32)
+            vals = (vals << np.arange(bw, dtype=np.uint32)).sum(
+                axis=1)
+            cnt = min(groups * 8, n - filled)
+            out[filled:filled + cnt] = vals[:cnt]
+            pos += nbytes
+            filled += cnt
+        else:  # RLE run: repeated value of ceil(bw/8) bytes LE
+            run = min(header >> 1, n - filled)
+            v = int.from_bytes(bytes(data[pos:pos + vbytes]), "little")
+            pos += vbytes
+            out[filled:filled + run] = v
+            filled += run
+    if filled < n:
+        out[filled:] = 0
     return out
 
 
@@ -680,21 +870,44 @@ def read_native_host(path: str, columns: Optional[List[str]] = None
                                          Dict[str, np.ndarray]]]:
     """Host-side decode of a native-layout file (testing / CPU path).
     Returns (columns, validity masks); masks holds entries only for
-    columns containing nulls (null slots in the value array are 0)."""
+    columns containing nulls (null slots in the value array are 0).
+    String columns come back as StrCol (codes + dictionary)."""
     layout = read_native_layout(path, columns)
     if layout is None:
         return None
     data, chunks = layout
-    if any(c.encoding != "plain" for c in chunks):
-        return None  # dictionary decode is the device path; host->pyarrow
+    if any(c.encoding != "plain" and not c.is_string for c in chunks):
+        return None  # numeric dict decode is the device path
     acc: Dict[str, list] = {}
     macc: Dict[str, list] = {}
     any_null: Dict[str, bool] = {}
+    str_dicts: Dict[str, list] = {}
     for c in chunks:  # row-group-major order
         if not c.pages:  # zero-row chunk still contributes its column
             acc.setdefault(c.name, []).append(
                 np.empty(0, dtype=c.np_dtype))
             macc.setdefault(c.name, []).append(np.ones(0, dtype=bool))
+            if c.is_string:
+                str_dicts.setdefault(c.name, []).append([])
+            continue
+        if c.is_string:
+            values = c.dict_values(data)
+            for page, mask in zip(c.pages, c.page_masks):
+                _, p_start, p_end, nv, bw = page
+                n_valid = int(mask.sum()) if mask is not None else nv
+                codes = _decode_rle_indices(data, p_start, p_end,
+                                            n_valid, bw)
+                if mask is None:
+                    part = codes
+                    pm = np.ones(nv, dtype=bool)
+                else:
+                    part = np.zeros(nv, dtype=np.int32)
+                    part[mask] = codes
+                    pm = mask
+                    any_null[c.name] = True
+                acc.setdefault(c.name, []).append(part)
+                macc.setdefault(c.name, []).append(pm)
+                str_dicts.setdefault(c.name, []).append(values)
             continue
         for (_, off, nv), mask in zip(c.pages, c.page_masks):
             if mask is None:
@@ -710,10 +923,29 @@ def read_native_host(path: str, columns: Optional[List[str]] = None
                 any_null[c.name] = True
             acc.setdefault(c.name, []).append(part)
             macc.setdefault(c.name, []).append(pm)
-    cols = {name: (np.concatenate(parts) if len(parts) > 1
-                   else (parts[0] if parts[0].flags.writeable
-                         else parts[0].copy()))
-            for name, parts in acc.items()}
+    cols: Dict[str, Any] = {}
+    for name, parts in acc.items():
+        if name in str_dicts:
+            dicts = str_dicts[name]
+            if all(d == dicts[0] for d in dicts[1:]):
+                merged_vals = dicts[0]
+                codes = (np.concatenate(parts) if len(parts) > 1
+                         else parts[0].copy())
+            else:  # per-row-group dictionaries: merge + remap
+                merged_vals = sorted(set().union(*map(set, dicts)))
+                vi = {v: i for i, v in enumerate(merged_vals)}
+                remapped = []
+                for part, d in zip(parts, dicts):
+                    lut = np.array([vi[v] for v in d] or [0],
+                                   dtype=np.int32)
+                    remapped.append(lut[part])
+                codes = (np.concatenate(remapped) if len(remapped) > 1
+                         else remapped[0])
+            cols[name] = StrCol(codes, list(merged_vals))
+        else:
+            cols[name] = (np.concatenate(parts) if len(parts) > 1
+                          else (parts[0] if parts[0].flags.writeable
+                                else parts[0].copy()))
     masks = {name: np.concatenate(macc[name]) if len(macc[name]) > 1
              else macc[name][0]
              for name in acc if any_null.get(name)}

@@ -96,13 +96,18 @@ def write_batch_parquet(batch, path: str, compression: Optional[str] = None,
     columns fall back to pyarrow.
     """
     from ..execution.columnar import StringColumn
-    if compression is None and not any(
-            isinstance(c, StringColumn) for c in batch.columns.values()):
-        from .native_parquet import write_parquet_native
-        cols = {name: t.numpy() for name, t in batch.columns.items()}
-        masks = {name: m.numpy() for name, m in batch.masks.items()} \
-            if batch.masks else None
-        return write_parquet_native(cols, path, masks)
+    if compression is None:
+        from .native_parquet import (write_parquet_native, StrCol,
+                                     _NP_TO_PARQUET)
+        cols = {name: (StrCol(c.codes.numpy(), c.values)
+                       if isinstance(c, StringColumn) else c.numpy())
+                for name, c in batch.columns.items()}
+        if all(isinstance(a, StrCol) or a.dtype in _NP_TO_PARQUET
+               for a in cols.values()):
+            masks = {name: m.numpy()
+                     for name, m in batch.masks.items()} \
+                if batch.masks else None
+            return write_parquet_native(cols, path, masks)
     import pyarrow.parquet as pq
     table = batch.to_arrow()
     pq.write_table(table, path, compression=compression or "NONE",
@@ -140,12 +145,34 @@ def read_files_batch(paths: List[str], columns: Optional[List[str]] = None):
         if columns is not None:
             order = {c.lower(): i for i, c in enumerate(columns)}
             names.sort(key=lambda n: order.get(n.lower(), 99))
+        from .native_parquet import StrCol
+        from ..execution.columnar import StringColumn
         merged = {}
         merged_masks = {}
         for name in names:
             arrs = [f[name] for f in per_file]
-            merged[name] = torch.from_numpy(
-                np.concatenate(arrs) if len(arrs) > 1 else arrs[0])
+            if any(isinstance(a, StrCol) for a in arrs):
+                # merge per-file dictionaries, remap codes
+                dicts = [a.values for a in arrs]
+                if all(d == dicts[0] for d in dicts[1:]):
+                    mvals = list(dicts[0])
+                    codes = (np.concatenate([a.codes for a in arrs])
+                             if len(arrs) > 1 else arrs[0].codes)
+                else:
+                    mvals = sorted(set().union(*map(set, dicts)))
+                    vi = {v: i for i, v in enumerate(mvals)}
+                    parts = []
+                    for a in arrs:
+                        lut = np.array([vi[v] for v in a.values] or [0],
+                                       dtype=np.int32)
+                        parts.append(lut[a.codes])
+                    codes = (np.concatenate(parts) if len(parts) > 1
+                             else parts[0])
+                merged[name] = StringColumn(
+                    torch.from_numpy(np.ascontiguousarray(codes)), mvals)
+            else:
+                merged[name] = torch.from_numpy(
+                    np.concatenate(arrs) if len(arrs) > 1 else arrs[0])
             if any(name in fm for fm in per_file_masks):
                 mparts = [fm.get(name, np.ones(rc, dtype=bool))
                           for fm, rc in zip(per_file_masks, row_counts)]
@@ -233,6 +260,7 @@ def read_files_batch_device(paths: List[str], device,
     metas = []
     schemas = []
     per_file_layouts = None
+    string_cols = set()
     if cached is not None:
         row_counts = [ent[0] for ent in cached]
         lay0 = cached[0][1]
@@ -241,6 +269,8 @@ def read_files_batch_device(paths: List[str], device,
         dtypes = {c.name: c.np_dtype for c in lay0}
         nullable_cols = {c.name for ent in cached for c in ent[1]
                          if c.has_nulls}
+        string_cols = {c.name for ent in cached for c in ent[1]
+                       if c.is_string}
         per_file_layouts = [
             [c for c in ent[1]
              if want is None or c.name.lower() in want]
@@ -280,7 +310,14 @@ def read_files_batch_device(paths: List[str], device,
                 continue
             npd = _PHYS_TO_NP.get(col.physical_type)
             if npd is None:
-                return fallback()
+                if col.physical_type == "BYTE_ARRAY":
+                    # dictionary strings decode to int32 codes; the
+                    # layout parse validates the encoding (else the
+                    # worker falls back)
+                    npd = np.dtype("int32")
+                    string_cols.add(col.path_in_schema)
+                else:
+                    return fallback()
             names.append(col.path_in_schema)
             dtypes[col.path_in_schema] = npd
         row_counts = [md.num_rows for md in metas]
@@ -302,6 +339,9 @@ def read_files_batch_device(paths: List[str], device,
                  for n in names if n in nullable_cols}
     ext = native_ext.ext()
     statuses: List["_torch.Tensor"] = []  # snappy per-page status words
+    # (name, abs row start, abs row end, dictionary values) per string
+    # chunk — merged into one dictionary per column after the sync
+    str_chunks: List[Tuple[str, int, int, List[str]]] = []
 
     # per-worker HIP streams: each file's H2D copy and decode kernels run
     # on their own stream, so copies overlap other files' decodes instead
@@ -486,7 +526,16 @@ def read_files_batch_device(paths: List[str], device,
                 cursors[c.name] = written
                 continue
             dict_vals = None
-            if c.encoding == "dict":
+            is_str = c.is_string
+            if c.encoding == "dict" and is_str:
+                # string chunk (K1): indices ARE the column (int32 codes
+                # into the host-parsed dictionary) — no value gather;
+                # per-chunk dictionaries merge after the sync below
+                str_chunks.append(
+                    (c.name, written,
+                     written + sum(p[3] for p in c.pages),
+                     c.dict_values(buf.numpy())))
+            elif c.encoding == "dict":
                 # K1 dictionary path: decode the dictionary page once;
                 # pages gather through it (a chunk can also carry PLAIN
                 # pages — the writer's mid-chunk dictionary-overflow
@@ -505,10 +554,13 @@ def read_files_batch_device(paths: List[str], device,
                         _, p_start, p_end, _, bw = page
                         runs = ext.parse_rle_runs(buf, p_start, p_end, bw,
                                                   nv)
-                        idx = ext.rle_decode(dev_bytes, *runs, bw,
-                                             nv).to(_torch.int64)
-                        out[c.name][written:written + nv] = \
-                            ext.gather_rows(dict_vals, idx)
+                        idx = ext.rle_decode(dev_bytes, *runs, bw, nv)
+                        if is_str:
+                            out[c.name][written:written + nv] = idx
+                        else:
+                            out[c.name][written:written + nv] = \
+                                ext.gather_rows(dict_vals,
+                                                idx.to(_torch.int64))
                     else:
                         _, off, _ = page
                         ext.copy_unaligned(dev_bytes, off, out[c.name],
@@ -536,9 +588,10 @@ def read_files_batch_device(paths: List[str], device,
                         runs = ext.parse_rle_runs(buf, p_start, p_end, bw,
                                                   n_valid)
                         idx = ext.rle_decode(dev_bytes, *runs, bw,
-                                             n_valid).to(_torch.int64)
-                        tmp[cur:cur + n_valid] = ext.gather_rows(
-                            dict_vals, idx)
+                                             n_valid)
+                        tmp[cur:cur + n_valid] = idx if is_str else \
+                            ext.gather_rows(dict_vals,
+                                            idx.to(_torch.int64))
                     else:
                         _, off, _ = page
                         ext.copy_unaligned(dev_bytes, off, tmp,
@@ -581,4 +634,27 @@ def read_files_batch_device(paths: List[str], device,
         return fallback()
     for b in bufs:
         _pinned_put(b)
+    if string_cols:
+        from ..execution.columnar import StringColumn
+        by_name: Dict[str, list] = {}
+        for name, lo, hi, vals in str_chunks:
+            by_name.setdefault(name, []).append((lo, hi, vals))
+        for name in string_cols:
+            chunks_n = by_name.get(name, [])
+            dicts = [v for _, _, v in chunks_n]
+            if dicts and all(d == dicts[0] for d in dicts[1:]):
+                merged = list(dicts[0])
+            else:
+                merged = sorted(set().union(*map(set, dicts))) \
+                    if dicts else []
+                vi = {v: i for i, v in enumerate(merged)}
+                for lo, hi, vals in chunks_n:
+                    if vals == merged:
+                        continue
+                    lut = _torch.tensor(
+                        [vi[v] for v in vals] or [0],
+                        dtype=_torch.int32, device=device)
+                    seg = out[name][lo:hi]
+                    out[name][lo:hi] = lut[seg.long()]
+            out[name] = StringColumn(out[name], merged)
     return ColumnBatch(out, out_masks), row_counts

@@ -91,7 +91,78 @@ def test_gpu_cobucketed_join(env, tmp_path):
     assert ex.stats.shuffles == 0
     gpu.disable_hyperspace()
     base = q.collect()
-    assert out.num_rows == base.num_rows
+    # full (key, val, status) multiset equality, not just row count
+    assert _rows(out, ["key", "val", "status"]) == \
+        _rows(base, ["key", "val", "status"])
+
+
+def test_gpu_string_native_decode_and_join(tmp_path, monkeypatch):
+    """Device string path: dictionary BYTE_ARRAY files
+    decode on device (codes stay in HBM), string-keyed index build +
+    co-bucketed join return content equal to the CPU engine."""
+    from hyperspace_amd.execution.columnar import (ColumnBatch,
+                                                   StringColumn)
+    from hyperspace_amd.sources.parquet_io import (
+        read_files_batch_device, write_batch_parquet)
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "sidx"))
+    rng = np.random.default_rng(33)
+    src = tmp_path / "ssrc"
+    src.mkdir()
+    vocab = sorted(f"sku-{i:05d}" for i in range(5000))
+    paths = []
+    for i in range(3):
+        codes = rng.integers(0, 5000, 300_000).astype(np.int32)
+        mask = rng.random(300_000) > 0.05
+        b = ColumnBatch(
+            {"sku": StringColumn(torch.from_numpy(codes), list(vocab)),
+             "v": torch.from_numpy(rng.random(300_000))},
+            masks={"sku": torch.from_numpy(mask)})
+        p = str(src / f"part-{i}.parquet")
+        write_batch_parquet(b, p)
+        paths.append(p)
+
+    # 1. device decode == host decode, codes resident on device
+    from hyperspace_amd.sources.native_parquet import _LAYOUT_CACHE
+    _LAYOUT_CACHE.clear()  # force the footer/page parse path
+    dev_batch, rc_dev = read_files_batch_device(
+        paths, torch.device("cuda"))
+    from hyperspace_amd.sources.parquet_io import read_files_batch
+    host_batch, rc_host = read_files_batch(paths)
+    assert rc_dev == rc_host
+    s_dev = dev_batch.column("sku")
+    assert isinstance(s_dev, StringColumn)
+    assert s_dev.codes.device.type == "cuda"
+    s_host = host_batch.column("sku")
+    assert s_dev.values == s_host.values
+    assert torch.equal(s_dev.codes.cpu(), s_host.codes)
+    assert torch.equal(dev_batch.mask("sku").cpu(), host_batch.mask("sku"))
+
+    # 2. string-keyed index + join: content equality vs CPU baseline
+    rdir = tmp_path / "sdim"
+    rdir.mkdir()
+    write_batch_parquet(ColumnBatch(
+        {"sku": StringColumn(
+            torch.arange(5000, dtype=torch.int32), list(vocab)),
+         "status": torch.from_numpy(rng.integers(0, 5, 5000))}),
+        str(rdir / "part-0.parquet"))
+    gpu = hs.HyperspaceSession(device="cuda")
+    gpu.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 32)
+    h = hs.Hyperspace(gpu)
+    left = gpu.read_parquet(str(src))
+    right = gpu.read_parquet(str(rdir))
+    h.create_index(left, hs.CoveringIndexConfig("gsl", ["sku"], ["v"]))
+    h.create_index(right, hs.CoveringIndexConfig("gsr", ["sku"],
+                                                 ["status"]))
+    gpu.enable_hyperspace()
+    q = left.select("sku", "v").join(right.select("sku", "status"),
+                                     on="sku")
+    ex = Executor(gpu)
+    out = ex.execute(q.optimized_plan())
+    assert ex.stats.merge_joins == 1 and ex.stats.shuffles == 0
+    gpu.disable_hyperspace()
+    base = q.collect()
+    assert _rows(out, ["sku", "v", "status"]) == \
+        _rows(base, ["sku", "v", "status"])
 
 
 def test_gpu_sorted_within_buckets(env, tmp_path):
