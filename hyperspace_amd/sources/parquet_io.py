@@ -346,10 +346,11 @@ def read_files_batch_device(paths: List[str], device,
     # per-worker HIP streams: each file's H2D copy and decode kernels run
     # on their own stream, so copies overlap other files' decodes instead
     # of serializing on the default stream (xfers are the cold-load bound)
-    n_streams = min(8, max(1, len(paths)))
+    n_streams = 8  # row-group units of a single file fan out too
     streams = [_torch.cuda.Stream(device=device) for _ in range(n_streams)]
 
-    def load_decode(i):
+    def load_file(i):
+        """Read + layout-parse one file (no decode)."""
         p = paths[i]
         size = os.path.getsize(p)
         buf = _pinned_get(size + 4)
@@ -365,15 +366,29 @@ def read_files_batch_device(paths: List[str], device,
                 _pinned_put(buf)
                 return None
             chunks = lay[1]
-        stream = streams[i % n_streams]
-        with _torch.cuda.stream(stream):
-            return _decode_on_stream(i, buf, size, chunks)
+        return buf, size, chunks
 
-    def _decode_on_stream(i, buf, size, chunks):
-        # upload only the file's bytes (+4B decode slack), not the whole
-        # pooled size class
-        dev_bytes = buf[:size + 4].to(device, non_blocking=True)
-        cursors = {n: int(file_base[i]) for n in names}
+    def split_row_groups(chunks):
+        """[(row_offset, chunks-of-one-row-group)] — layout order is
+        row-group-major, so a repeated column name starts a new group."""
+        groups = []
+        cur: list = []
+        seen = set()
+        row_off = 0
+        for c in chunks:
+            if c.name in seen:
+                groups.append((row_off, cur))
+                row_off += cur[0].num_values
+                cur = []
+                seen = set()
+            cur.append(c)
+            seen.add(c.name)
+        if cur:
+            groups.append((row_off, cur))
+        return groups
+
+    def _decode_on_stream(i, buf, size, chunks, dev_bytes, row_off):
+        cursors = {n: int(file_base[i]) + row_off for n in names}
         for c in chunks:
             itemsize = c.np_dtype.itemsize
             written = cursors[c.name]
@@ -614,11 +629,64 @@ def read_files_batch_device(paths: List[str], device,
         # caller-side synchronize below holds them via `bufs`
         return buf
 
+    # phase 1: read + layout-parse every file (parallel, GIL-released IO)
     if len(paths) > 2:
         with ThreadPoolExecutor(max_workers=16) as pool:
-            bufs = list(pool.map(load_decode, range(len(paths))))
+            infos = list(pool.map(load_file, range(len(paths))))
     else:
-        bufs = [load_decode(i) for i in range(len(paths))]
+        infos = [load_file(i) for i in range(len(paths))]
+    bufs = [inf[0] if inf is not None else None for inf in infos]
+    if not all(inf is not None for inf in infos):
+        for b in bufs:
+            if b is not None:
+                _pinned_put(b)
+        return fallback()
+
+    # phase 2: decode units.  Few big files split at ROW-GROUP
+    # granularity so a single appended file's row groups decode on
+    # parallel workers/streams (the round-1 incremental-refresh and
+    # single-file cold-load bottleneck); many files keep one unit per
+    # file.  Uploads happen once per file; other units of the same file
+    # wait on the upload event.
+    units = []  # (file_i, row_off, chunks)
+    if len(paths) < n_streams:
+        for i, inf in enumerate(infos):
+            for row_off, rg_chunks in split_row_groups(inf[2]):
+                units.append((i, row_off, rg_chunks))
+    else:
+        units = [(i, 0, inf[2]) for i, inf in enumerate(infos)]
+
+    import threading
+    dev_bufs: List[Optional["_torch.Tensor"]] = [None] * len(paths)
+    upload_events: List[Optional["_torch.cuda.Event"]] = \
+        [None] * len(paths)
+    upload_lock = threading.Lock()
+
+    def decode_unit(u):
+        uid, (i, row_off, chunks) = u
+        buf, size, _ = infos[i]
+        stream = streams[uid % n_streams]
+        with _torch.cuda.stream(stream):
+            with upload_lock:
+                if dev_bufs[i] is None:
+                    # upload only the file's bytes (+4B decode slack),
+                    # not the whole pooled size class
+                    dev_bufs[i] = buf[:size + 4].to(device,
+                                                    non_blocking=True)
+                    ev = _torch.cuda.Event()
+                    ev.record(stream)
+                    upload_events[i] = ev
+                else:
+                    stream.wait_event(upload_events[i])
+            _decode_on_stream(i, buf, size, chunks, dev_bufs[i], row_off)
+        return True
+
+    if len(units) > 2:
+        with ThreadPoolExecutor(max_workers=16) as pool:
+            list(pool.map(decode_unit, enumerate(units)))
+    else:
+        for u in enumerate(units):
+            decode_unit(u)
     # order the default stream after every worker stream, then host-sync
     # so the pinned buffers can be recycled
     cur = _torch.cuda.current_stream()
