@@ -319,10 +319,19 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
                              P* __restrict__ opayload, int64_t n,
                              int shift, const uint32_t* __restrict__ hist,
                              int nb, int64_t tiles_per_block) {
+  // LDS-staged scatter: elements are first counting-sorted by digit
+  // WITHIN the tile (stage[] in digit order), then written out with
+  // threadIdx-contiguous reads — global stores become per-digit-run
+  // coalesced bursts instead of 256-way scattered singles (the round-1
+  // k_rs_scatter showed 32.6% issue-stall on exactly those stores).
   __shared__ uint32_t cur[RS_RADIX];
   __shared__ uint32_t wave_cnt[WAVES][RS_RADIX];
   __shared__ uint32_t wave_pref[WAVES][RS_RADIX];
   __shared__ uint32_t tile_total[RS_RADIX];
+  __shared__ uint32_t digit_start[RS_RADIX];  // excl scan of tile_total
+  __shared__ uint32_t wsum[WAVES];
+  __shared__ uint64_t stage[THREADS];  // keys, then payload (reused)
+  static_assert(RS_RADIX == THREADS, "one digit per thread");
   for (int d = threadIdx.x; d < RS_RADIX; d += blockDim.x)
     cur[d] = hist[(int64_t)d * nb + blockIdx.x];
   int lane = threadIdx.x & 63;
@@ -331,6 +340,8 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
   int64_t t1 = t0 + tiles_per_block;
   for (int64_t tile = t0; tile < t1 && tile * THREADS < n; tile++) {
     int64_t i = tile * (int64_t)THREADS + threadIdx.x;
+    int tile_n = (int)((n - tile * THREADS) < THREADS
+                           ? (n - tile * THREADS) : THREADS);
     bool valid = i < n;
     uint64_t key = valid ? keys[i] : 0;
     P pl = valid ? payload[i] : P(0);
@@ -348,24 +359,56 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
     int rank = __popcll(eq & ((1ull << lane) - 1ull));
     if (valid && rank == 0) wave_cnt[wave][d] = (uint32_t)__popcll(eq);
     __syncthreads();
-    for (int dd = threadIdx.x; dd < RS_RADIX; dd += blockDim.x) {
+    // one digit per thread: cross-wave prefix + wave-level shfl scan of
+    // the 256 tile totals into digit_start (no extra barrier rounds)
+    {
+      int dd = threadIdx.x;
       uint32_t p = 0;
+#pragma unroll
       for (int w = 0; w < WAVES; w++) {
         wave_pref[w][dd] = p;
         p += wave_cnt[w][dd];
       }
       tile_total[dd] = p;
+      uint32_t v = p;  // inclusive scan over this wave's 64 digits
+#pragma unroll
+      for (int off = 1; off < 64; off <<= 1) {
+        uint32_t t = __shfl_up(v, off);
+        if (lane >= off) v += t;
+      }
+      if (lane == 63) wsum[wave] = v;
+      __syncthreads();
+      uint32_t wbase = 0;
+#pragma unroll
+      for (int w = 0; w < WAVES; w++)
+        if (w < wave) wbase += wsum[w];
+      digit_start[dd] = wbase + v - p;  // exclusive scan
     }
     __syncthreads();
+    // counting-sort the tile into LDS (digit-major, stable)
+    uint32_t tile_pos = 0;
     if (valid) {
-      uint32_t pos = cur[d] + wave_pref[wave][d] + rank;
-      okeys[pos] = key;
-      opayload[pos] = pl;
+      tile_pos = digit_start[d] + wave_pref[wave][d] + rank;
+      stage[tile_pos] = key;
     }
     __syncthreads();
-    for (int dd = threadIdx.x; dd < RS_RADIX; dd += blockDim.x)
-      cur[dd] += tile_total[dd];
+    // coalesced write-out: thread j emits the tile's j-th digit-ordered
+    // key; destination = digit's global cursor + offset within run
+    uint64_t k2 = 0;
+    uint32_t dst = 0;
+    if (threadIdx.x < tile_n) {
+      k2 = stage[threadIdx.x];
+      int d2 = (int)((k2 >> shift) & 255);
+      dst = cur[d2] + threadIdx.x - digit_start[d2];
+      okeys[dst] = k2;
+    }
     __syncthreads();
+    if (valid) stage[tile_pos] = (uint64_t)pl;
+    __syncthreads();
+    if (threadIdx.x < tile_n) opayload[dst] = (P)stage[threadIdx.x];
+    __syncthreads();
+    cur[threadIdx.x] += tile_total[threadIdx.x];
+    // next tile's reads of cur happen after its own barriers
   }
 }
 
@@ -1241,6 +1284,65 @@ void copy_unaligned(const uint8_t* src, int64_t src_off, uint8_t* dst,
 // is pages x files.  Format: github.com/google/snappy format_description.
 // ---------------------------------------------------------------------------
 
+// Register-window stream cursor: the tag walk is inherently serial, so
+// its cost is per-op DEPENDENT loads.  A 16-byte aligned window turns
+// ~3-6 dependent byte loads per op into one aligned u64 load per 8
+// stream bytes (the round-1 kernel was latency-bound on exactly those
+// byte loads at ~1.8 GB/s).
+struct SnapCursor {
+  const uint8_t* s0;    // aligned-down stream base
+  const uint8_t* send;  // one past last valid byte
+  int64_t pos;          // absolute byte position (s0-relative)
+  int64_t base;         // window start (8-aligned, s0-relative)
+  uint64_t lo, hi;      // 16-byte window [base, base+16)
+
+  __device__ __forceinline__ uint64_t load8(int64_t b) {
+    // aligned u64 within bounds -> single load; tail -> clamped bytes
+    if (s0 + b + 8 <= send) return *(const uint64_t*)(s0 + b);
+    uint64_t v = 0;
+    for (int i = 0; i < 8; i++)
+      if (s0 + b + i < send) v |= (uint64_t)s0[b + i] << (8 * i);
+    return v;
+  }
+  __device__ __forceinline__ void init(const uint8_t* start,
+                                       const uint8_t* end) {
+    s0 = (const uint8_t*)((uintptr_t)start & ~7ull);
+    send = end;
+    pos = start - s0;
+    base = 0;
+    lo = load8(0);
+    hi = load8(8);
+  }
+  __device__ __forceinline__ void slide() {
+    while (pos - base >= 8) {
+      if (pos - base >= 16) {  // long jump (literal): reinit window
+        base = pos & ~7ll;
+        lo = load8(base);
+        hi = load8(base + 8);
+        return;
+      }
+      lo = hi;
+      base += 8;
+      hi = load8(base + 8);
+    }
+  }
+  // peek up to 8 bytes at pos (little-endian); window invariant
+  // pos-base <= 7 guarantees 9+ valid bytes ahead
+  __device__ __forceinline__ uint64_t peek() {
+    int sh = (int)(pos - base) * 8;
+    uint64_t v = lo >> sh;
+    if (sh) v |= hi << (64 - sh);
+    return v;
+  }
+  __device__ __forceinline__ void advance(int64_t nbytes) {
+    pos += nbytes;
+    slide();
+  }
+  __device__ __forceinline__ bool exhausted() {
+    return s0 + pos >= send;
+  }
+};
+
 __global__ void k_snappy_decomp(const uint8_t* __restrict__ src,
                                 const int64_t* __restrict__ src_off,
                                 const int64_t* __restrict__ src_end,
@@ -1249,17 +1351,21 @@ __global__ void k_snappy_decomp(const uint8_t* __restrict__ src,
                                 const int64_t* __restrict__ dst_len,
                                 int32_t* __restrict__ status) {
   int p = blockIdx.x;
-  const uint8_t* s = src + src_off[p];
+  const uint8_t* sbeg = src + src_off[p];
   const uint8_t* send = src + src_end[p];
   uint8_t* d = dst + dst_off[p];
   int64_t expected = dst_len[p];
   int lane = threadIdx.x;
 
+  SnapCursor c;
+  c.init(sbeg, send);
+
   // varint uncompressed length (lockstep on every lane)
   uint64_t ulen = 0;
   int shift = 0;
-  while (s < send) {
-    uint8_t b = *s++;
+  while (!c.exhausted()) {
+    uint8_t b = (uint8_t)c.peek();
+    c.advance(1);
     ulen |= (uint64_t)(b & 0x7F) << shift;
     if (!(b & 0x80)) break;
     shift += 7;
@@ -1271,8 +1377,9 @@ __global__ void k_snappy_decomp(const uint8_t* __restrict__ src,
 
   int64_t out = 0;
   int64_t synced = 0;  // output below this point is visible to all lanes
-  while (out < expected && s < send) {
-    uint8_t tag = *s++;
+  while (out < expected && !c.exhausted()) {
+    uint64_t w = c.peek();
+    uint8_t tag = (uint8_t)w;
     int k = tag & 3;
     int64_t len;
     int64_t off = 0;
@@ -1281,40 +1388,56 @@ __global__ void k_snappy_decomp(const uint8_t* __restrict__ src,
       int64_t l = tag >> 2;
       if (l < 60) {
         len = l + 1;
+        c.advance(1);
       } else {
         int nb = (int)(l - 59);  // 1..4 length bytes, little-endian
-        if (s + nb > send) { if (lane == 0) status[p] = 2; return; }
-        uint32_t v = 0;
-        for (int i = 0; i < nb; i++) v |= (uint32_t)s[i] << (8 * i);
-        s += nb;
+        uint32_t v = (uint32_t)((w >> 8) &
+                                (0xFFFFFFFFu >> (8 * (4 - nb))));
+        c.advance(1 + nb);
         len = (int64_t)v + 1;
       }
-      lit = s;
-      s += len;
-      if (s > send) { if (lane == 0) status[p] = 2; return; }
+      lit = c.s0 + c.pos;
+      if (lit + len > send) { if (lane == 0) status[p] = 2; return; }
+      c.advance(len);
     } else if (k == 1) {  // copy, 1-byte offset
       len = ((tag >> 2) & 7) + 4;
-      if (s + 1 > send) { if (lane == 0) status[p] = 2; return; }
-      off = ((int64_t)(tag >> 5) << 8) | s[0];
-      s += 1;
+      off = ((int64_t)(tag >> 5) << 8) | ((w >> 8) & 0xFF);
+      c.advance(2);
     } else if (k == 2) {  // copy, 2-byte offset
       len = (tag >> 2) + 1;
-      if (s + 2 > send) { if (lane == 0) status[p] = 2; return; }
-      off = (int64_t)s[0] | ((int64_t)s[1] << 8);
-      s += 2;
+      off = (int64_t)((w >> 8) & 0xFFFF);
+      c.advance(3);
     } else {  // copy, 4-byte offset
       len = (tag >> 2) + 1;
-      if (s + 4 > send) { if (lane == 0) status[p] = 2; return; }
-      off = (int64_t)s[0] | ((int64_t)s[1] << 8) |
-            ((int64_t)s[2] << 16) | ((int64_t)s[3] << 24);
-      s += 4;
+      off = (int64_t)((w >> 8) & 0xFFFFFFFFull);
+      c.advance(5);
     }
     if (out + len > expected || (k != 0 && (off <= 0 || off > out))) {
       if (lane == 0) status[p] = 3;
       return;
     }
     if (k == 0) {
-      for (int64_t i = lane; i < len; i += 64) d[out + i] = lit[i];
+      // wide literal copy: 8B lanes over the aligned middle when long
+      if (len >= 128) {
+        int64_t head = (8 - ((uintptr_t)(d + out) & 7)) & 7;
+        int64_t body = (len - head) & ~7ll;
+        for (int64_t i = lane; i < head; i += 64) d[out + i] = lit[i];
+        const uint8_t* lsrc = lit + head;
+        uint8_t* ldst = d + out + head;
+        if (lsrc + body + 8 <= send) {
+          for (int64_t i = lane; i * 8 < body; i += 64) {
+            uint64_t v;
+            __builtin_memcpy(&v, lsrc + i * 8, 8);
+            *(uint64_t*)(ldst + i * 8) = v;
+          }
+        } else {
+          for (int64_t i = lane; i < body; i += 64) ldst[i] = lsrc[i];
+        }
+        for (int64_t i = head + body + lane; i < len; i += 64)
+          d[out + i] = lit[i];
+      } else {
+        for (int64_t i = lane; i < len; i += 64) d[out + i] = lit[i];
+      }
     } else {
       // drain pending writes only when the source range reaches past
       // the last sync point (i.e. could include other lanes' recent
