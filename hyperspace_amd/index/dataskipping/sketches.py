@@ -24,39 +24,138 @@ from ...exceptions import HyperspaceException
 from ...plan.expr import (Arith, BinComp, Col, Expr, In, IsNotNull, Lit,
                           _expr_eq)
 
-# single binary arithmetic over a column and a numeric literal, e.g.
-# "a % 10" — the supported scalar-expression subset (the reference
-# accepts arbitrary deterministic scalar expressions via Catalyst;
-# here the predicate language is comparison-based, so sketches bind to
-# the same single-op arithmetic the Expr API can express)
-_ARITH_RE = re.compile(
-    r"^\s*([A-Za-z_][\w.]*)\s*([+\-*/%])\s*(-?\d+(?:\.\d+)?)\s*$")
+# Sketch expressions are arbitrary composed deterministic scalar
+# arithmetic over ONE column and numeric literals — e.g. "a % 10",
+# "a * 2 + 1", "(a + 3) % 7" — parsed into the plan layer's Expr trees
+# and matched structurally against predicate left-hand sides (the
+# reference resolves arbitrary deterministic scalar expressions through
+# Catalyst and rejects aggregates/windows/subqueries,
+# dataskipping/expressions/ExpressionUtils.scala:38-95; this grammar has
+# no aggregate forms, and multi-column / column-free expressions are
+# rejected below).
+
+_TOKEN_RE = re.compile(
+    r"\s*(\d+\.\d+|\d+|[A-Za-z_][\w.]*|[()+\-*/%])")
 
 
-def parse_sketch_expr(s: str):
-    """'col' or 'col OP literal' -> (col_name, op|None, literal|None)."""
-    m = _ARITH_RE.match(s)
-    if m:
-        col_s, op, lit = m.groups()
-        value = float(lit) if "." in lit else int(lit)
-        return col_s, op, value
-    return s.strip(), None, None
+def parse_expr_string(s: str) -> Expr:
+    """Parse '+ - * / %' arithmetic with parentheses over column names
+    and numeric literals into an Expr tree (standard precedence)."""
+    tokens: List[str] = []
+    pos = 0
+    while pos < len(s):
+        m = _TOKEN_RE.match(s, pos)
+        if m is None:
+            if s[pos:].strip():
+                raise HyperspaceException(
+                    f"Bad sketch expression near {s[pos:]!r}")
+            break
+        tokens.append(m.group(1))
+        pos = m.end()
+    it = {"i": 0}
+
+    def peek():
+        return tokens[it["i"]] if it["i"] < len(tokens) else None
+
+    def take():
+        t = peek()
+        it["i"] += 1
+        return t
+
+    def atom() -> Expr:
+        t = take()
+        if t is None:
+            raise HyperspaceException(f"Truncated sketch expression {s!r}")
+        if t == "(":
+            e = add()
+            if take() != ")":
+                raise HyperspaceException(f"Unbalanced parens in {s!r}")
+            return e
+        if t == "-":
+            inner = atom()
+            if isinstance(inner, Lit):
+                return Lit(-inner.value)
+            return Arith("-", Lit(0), inner)
+        if re.fullmatch(r"\d+\.\d+", t):
+            return Lit(float(t))
+        if re.fullmatch(r"\d+", t):
+            return Lit(int(t))
+        if re.fullmatch(r"[A-Za-z_][\w.]*", t):
+            return Col(t)
+        raise HyperspaceException(f"Bad token {t!r} in {s!r}")
+
+    def mul() -> Expr:
+        e = atom()
+        while peek() in ("*", "/", "%"):
+            e = Arith(take(), e, atom())
+        return e
+
+    def add() -> Expr:
+        e = mul()
+        while peek() in ("+", "-"):
+            e = Arith(take(), e, mul())
+        return e
+
+    out = add()
+    if peek() is not None:
+        raise HyperspaceException(f"Trailing tokens in {s!r}")
+    return out
 
 
-def eval_sketch_values(values: torch.Tensor, op, lit) -> torch.Tensor:
-    """Apply the sketch's arithmetic to a column tensor.  ``%`` uses
-    Java/Spark remainder semantics (sign of the dividend: fmod)."""
-    if op is None:
-        return values
-    if op == "+":
-        return values + lit
-    if op == "-":
-        return values - lit
-    if op == "*":
-        return values * lit
-    if op == "%":
-        return torch.fmod(values, lit)
-    return values.to(torch.float64) / lit  # '/' is double division
+def expr_to_string(e: Expr) -> str:
+    """Canonical form: flat for a single binary op ('a % 10' — the
+    round-1 naming), parenthesized sub-expressions when nested."""
+    def fmt(x: Expr, nested: bool) -> str:
+        if isinstance(x, Col):
+            return x.name
+        if isinstance(x, Lit):
+            return repr(x.value)
+        if isinstance(x, Arith):
+            inner = f"{fmt(x.left, True)} {x.op} {fmt(x.right, True)}"
+            return f"({inner})" if nested else inner
+        raise HyperspaceException(f"Not a sketch expression: {x!r}")
+    return fmt(e, False)
+
+
+def eval_expr_tree(e: Expr, columns: Dict[str, torch.Tensor]):
+    """Evaluate an arithmetic Expr over tensors.  ``%`` uses Java/Spark
+    remainder semantics (sign of the dividend: fmod); ``/`` is double
+    division."""
+    if isinstance(e, Col):
+        for k, v in columns.items():
+            if k.lower() == e.name.lower():
+                return v
+        raise HyperspaceException(f"No column {e.name}")
+    if isinstance(e, Lit):
+        return e.value
+    if isinstance(e, Arith):
+        lv = eval_expr_tree(e.left, columns)
+        rv = eval_expr_tree(e.right, columns)
+        if e.op == "+":
+            return lv + rv
+        if e.op == "-":
+            return lv - rv
+        if e.op == "*":
+            return lv * rv
+        if e.op == "%":
+            return torch.fmod(lv, rv) if torch.is_tensor(lv) else lv % rv
+        lt = lv.to(torch.float64) if torch.is_tensor(lv) else float(lv)
+        return lt / rv
+    raise HyperspaceException(f"Cannot evaluate {e!r}")
+
+
+def _has_division(e: Expr) -> bool:
+    return isinstance(e, Arith) and (
+        e.op == "/" or _has_division(e.left) or _has_division(e.right))
+
+
+def _rename_cols(e: Expr, new_name: str) -> Expr:
+    if isinstance(e, Col):
+        return Col(new_name)
+    if isinstance(e, Arith):
+        return Arith(e.op, _rename_cols(e.left, new_name),
+                     _rename_cols(e.right, new_name))
+    return e
 
 MINMAX_SKETCH_TYPE = (
     "com.microsoft.hyperspace.index.dataskipping.sketches.MinMaxSketch")
@@ -68,44 +167,48 @@ PARTITION_SKETCH_TYPE = (
 
 
 class Sketch:
-    """Base sketch over a single source column expression (a column
-    name or ``col OP literal`` arithmetic)."""
+    """Base sketch over a single source column expression: a column
+    name or any composed arithmetic over that column and literals
+    (reference SingleExprSketch + ExpressionUtils resolve/normalize)."""
 
     def __init__(self, expr: str):
-        self.expr = expr
+        self.expr = expr.strip()
+        self._tree: Optional[Expr] = None
+
+    @property
+    def tree(self) -> Expr:
+        if self._tree is None:
+            self._tree = parse_expr_string(self.expr)
+        return self._tree
 
     @property
     def base_column(self) -> str:
-        return parse_sketch_expr(self.expr)[0]
-
-    def _parsed(self):
-        return parse_sketch_expr(self.expr)
+        cols = sorted(self.tree.references())
+        if len(cols) != 1:
+            raise HyperspaceException(
+                "A sketch expression must reference exactly one column; "
+                f"{self.expr!r} references {cols or 'none'}")
+        return cols[0]
 
     def rebind(self, resolved_col: str) -> None:
         """Rewrite the expression with the schema-resolved column."""
-        _, op, lit = self._parsed()
-        self.expr = (f"{resolved_col} {op} {lit}" if op is not None
-                     else resolved_col)
+        self._tree = _rename_cols(self.tree, resolved_col)
+        self.expr = expr_to_string(self._tree)
 
     def compute_values(self, col_values: torch.Tensor) -> torch.Tensor:
-        _, op, lit = self._parsed()
-        return eval_sketch_values(col_values, op, lit)
+        out = eval_expr_tree(self.tree, {self.base_column: col_values})
+        if not torch.is_tensor(out):
+            raise HyperspaceException(
+                f"Sketch expression {self.expr!r} is constant")
+        return out
 
     def value_type(self, col_type: str) -> str:
-        _, op, _ = self._parsed()
-        if op == "/":
-            return "double"
-        return col_type
+        return "double" if _has_division(self.tree) else col_type
 
     def matches_lhs(self, lhs: Expr) -> bool:
         """True when a predicate's left-hand side is structurally this
         sketch's expression."""
-        name, op, lit = self._parsed()
-        if op is None:
-            return isinstance(lhs, Col) and \
-                lhs.name.lower() == name.lower()
-        return isinstance(lhs, Arith) and _expr_eq(
-            lhs, Arith(op, Col(name), Lit(lit)))
+        return _expr_eq(lhs, self.tree)
 
     @property
     def kind(self) -> str:
