@@ -253,6 +253,21 @@ torch::Tensor bloom_probe(torch::Tensor vals, torch::Tensor words,
   return out;
 }
 
+torch::Tensor bloom_probe_many(torch::Tensor vals, torch::Tensor words,
+                               int64_t m_bits, int64_t k) {
+  check_cuda(vals, "vals");
+  TORCH_CHECK(words.dim() == 2, "words must be [n_filters, words]");
+  auto words_d = words.to(vals.device(), torch::kInt64).contiguous();
+  int64_t n_filters = words_d.size(0);
+  auto out = torch::zeros(
+      {n_filters}, torch::dtype(torch::kBool).device(vals.device()));
+  hsk::bloom_probe_many(vals.data_ptr<int64_t>(), vals.numel(),
+                        (const uint64_t*)words_d.data_ptr<int64_t>(),
+                        words_d.size(1), n_filters, m_bits, (int)k,
+                        out.data_ptr<bool>(), current_stream());
+  return out;
+}
+
 torch::Tensor zorder_key(std::vector<torch::Tensor> cols,
                          int64_t bits_per_col) {
   TORCH_CHECK(!cols.empty() && cols.size() <= 8, "1..8 zorder columns");
@@ -506,6 +521,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("segmented_minmax", &segmented_minmax, "per-segment min/max");
   m.def("bloom_build", &bloom_build, "bloom filter build");
   m.def("bloom_probe", &bloom_probe, "bloom filter probe");
+  m.def("bloom_probe_many", &bloom_probe_many,
+        "K9 device sketch-predicate probe over per-file filters");
   m.def("zorder_key", &zorder_key, "z-order bit interleave");
   m.def("gather_rows", &gather_rows, "row gather by index");
   m.def("copy_unaligned", &copy_unaligned,

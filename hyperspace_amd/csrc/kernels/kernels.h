@@ -87,6 +87,12 @@ void bloom_build(const int64_t* vals, int64_t n, uint64_t* words,
                  int64_t m_bits, int k, hipStream_t stream);
 void bloom_probe(const int64_t* vals, int64_t n, const uint64_t* words,
                  int64_t m_bits, int k, bool* out, hipStream_t stream);
+// K9: batched device sketch-predicate probe — every value against every
+// per-file filter; out[f] |= filter f may contain any value
+void bloom_probe_many(const int64_t* vals, int64_t n_vals,
+                      const uint64_t* words, int64_t words_per_filter,
+                      int64_t n_filters, int64_t m_bits, int k, bool* out,
+                      hipStream_t stream);
 
 // K10: z-order bit interleave of up to 8 normalized u64 columns
 void zorder_key(const uint64_t* const* cols, int n_cols, int bits_per_col,

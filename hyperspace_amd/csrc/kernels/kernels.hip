@@ -232,12 +232,10 @@ void normalize_key(const void* vals, int dtype, uint64_t* out, int64_t n,
 // ---------------------------------------------------------------------------
 
 constexpr int RS_RADIX = 256;      // 8-bit digits
-constexpr int RS_V = 4;            // keys per thread per tile
-constexpr int RS_TILE = THREADS * RS_V;  // 1024 keys per tile
 constexpr int RS_MAX_BLOCKS = 1024;
 
 static inline int rs_num_blocks(int64_t n) {
-  int64_t tiles = cdiv(n, RS_TILE);
+  int64_t tiles = cdiv(n, THREADS);
   return (int)(tiles < RS_MAX_BLOCKS ? (tiles < 1 ? 1 : tiles)
                                      : RS_MAX_BLOCKS);
 }
@@ -274,12 +272,14 @@ __global__ void k_rs_hist(const uint64_t* __restrict__ keys, int64_t n,
   __shared__ uint32_t lh[RS_RADIX];
   for (int d = threadIdx.x; d < RS_RADIX; d += blockDim.x) lh[d] = 0;
   __syncthreads();
-  int64_t e0 = (int64_t)blockIdx.x * tiles_per_block * RS_TILE;
-  int64_t e1 = e0 + tiles_per_block * RS_TILE;
-  if (e1 > n) e1 = n;
-  for (int64_t i = e0 + threadIdx.x; i < e1; i += THREADS) {
-    int d = (int)((keys[i] >> shift) & 255);
-    atomicAdd(&lh[d], 1u);
+  int64_t t0 = (int64_t)blockIdx.x * tiles_per_block;
+  int64_t t1 = t0 + tiles_per_block;
+  for (int64_t tile = t0; tile < t1 && tile * THREADS < n; tile++) {
+    int64_t i = tile * (int64_t)THREADS + threadIdx.x;
+    if (i < n) {
+      int d = (int)((keys[i] >> shift) & 255);
+      atomicAdd(&lh[d], 1u);
+    }
   }
   __syncthreads();
   for (int d = threadIdx.x; d < RS_RADIX; d += blockDim.x)
@@ -319,20 +319,18 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
                              P* __restrict__ opayload, int64_t n,
                              int shift, const uint32_t* __restrict__ hist,
                              int nb, int64_t tiles_per_block) {
-  // LDS-staged multi-item scatter: each thread carries RS_V keys
-  // (striped rounds, preserving element order), the 1024-key tile is
-  // counting-sorted into LDS by digit, then written out with
+  // LDS-staged scatter: elements are first counting-sorted by digit
+  // WITHIN the tile (stage[] in digit order), then written out with
   // threadIdx-contiguous reads — global stores become per-digit-run
-  // coalesced bursts instead of 256-way scattered singles, and the
-  // per-tile barrier cost amortizes over 4x the keys (the round-1
-  // k_rs_scatter showed 32.6% issue-stall on the scattered stores).
+  // coalesced bursts instead of 256-way scattered singles (the round-1
+  // k_rs_scatter showed 32.6% issue-stall on exactly those stores).
   __shared__ uint32_t cur[RS_RADIX];
-  // per (round, wave, digit) counts, scanned IN PLACE into prefixes
-  __shared__ uint32_t cnt[RS_V][WAVES][RS_RADIX];
+  __shared__ uint32_t wave_cnt[WAVES][RS_RADIX];
+  __shared__ uint32_t wave_pref[WAVES][RS_RADIX];
   __shared__ uint32_t tile_total[RS_RADIX];
   __shared__ uint32_t digit_start[RS_RADIX];  // excl scan of tile_total
   __shared__ uint32_t wsum[WAVES];
-  __shared__ uint64_t stage[RS_TILE];  // keys, then payload (reused)
+  __shared__ uint64_t stage[THREADS];  // keys, then payload (reused)
   static_assert(RS_RADIX == THREADS, "one digit per thread");
   for (int d = threadIdx.x; d < RS_RADIX; d += blockDim.x)
     cur[d] = hist[(int64_t)d * nb + blockIdx.x];
@@ -340,52 +338,37 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
   int wave = threadIdx.x >> 6;
   int64_t t0 = (int64_t)blockIdx.x * tiles_per_block;
   int64_t t1 = t0 + tiles_per_block;
-  for (int64_t tile = t0; tile < t1 && tile * RS_TILE < n; tile++) {
-    int64_t ebase = tile * (int64_t)RS_TILE;
-    int tile_n = (int)((n - ebase) < RS_TILE ? (n - ebase) : RS_TILE);
-    uint64_t key[RS_V];
-    P pl[RS_V];
-    int dig[RS_V];
-    int rank[RS_V];
-    for (int x = threadIdx.x; x < RS_V * WAVES * RS_RADIX;
-         x += blockDim.x)
-      ((uint32_t*)cnt)[x] = 0;
+  for (int64_t tile = t0; tile < t1 && tile * THREADS < n; tile++) {
+    int64_t i = tile * (int64_t)THREADS + threadIdx.x;
+    int tile_n = (int)((n - tile * THREADS) < THREADS
+                           ? (n - tile * THREADS) : THREADS);
+    bool valid = i < n;
+    uint64_t key = valid ? keys[i] : 0;
+    P pl = valid ? payload[i] : P(0);
+    int d = valid ? (int)((key >> shift) & 255) : RS_RADIX;  // sentinel
+    for (int x = threadIdx.x; x < WAVES * RS_RADIX; x += blockDim.x)
+      ((uint32_t*)wave_cnt)[x] = 0;
     __syncthreads();
+    // 9-bit ballot multi-split: lanes with equal digit (incl. sentinel)
+    unsigned long long eq = ~0ull;
 #pragma unroll
-    for (int v = 0; v < RS_V; v++) {
-      // striped round v covers elements [v*256, (v+1)*256): round-major
-      // order equals element order, so cross-round ranks stay stable
-      int64_t i = ebase + (int64_t)v * THREADS + threadIdx.x;
-      bool valid = i < n;
-      key[v] = valid ? keys[i] : 0;
-      pl[v] = valid ? payload[i] : P(0);
-      int d = valid ? (int)((key[v] >> shift) & 255) : RS_RADIX;
-      dig[v] = d;
-      // 9-bit ballot multi-split (sentinel included, never counted)
-      unsigned long long eq = ~0ull;
-#pragma unroll
-      for (int b = 0; b < 9; b++) {
-        unsigned long long m = __ballot((d >> b) & 1);
-        eq &= ((d >> b) & 1) ? m : ~m;
-      }
-      rank[v] = __popcll(eq & ((1ull << lane) - 1ull));
-      if (valid && rank[v] == 0)
-        cnt[v][wave][d] = (uint32_t)__popcll(eq);
+    for (int b = 0; b < 9; b++) {
+      unsigned long long m = __ballot((d >> b) & 1);
+      eq &= ((d >> b) & 1) ? m : ~m;
     }
+    int rank = __popcll(eq & ((1ull << lane) - 1ull));
+    if (valid && rank == 0) wave_cnt[wave][d] = (uint32_t)__popcll(eq);
     __syncthreads();
-    // one digit per thread: in-place (round, wave) prefix + wave-level
-    // shfl scan of the 256 tile totals into digit_start
+    // one digit per thread: cross-wave prefix + wave-level shfl scan of
+    // the 256 tile totals into digit_start (no extra barrier rounds)
     {
       int dd = threadIdx.x;
       uint32_t p = 0;
 #pragma unroll
-      for (int v = 0; v < RS_V; v++)
-#pragma unroll
-        for (int w = 0; w < WAVES; w++) {
-          uint32_t c = cnt[v][w][dd];
-          cnt[v][w][dd] = p;
-          p += c;
-        }
+      for (int w = 0; w < WAVES; w++) {
+        wave_pref[w][dd] = p;
+        p += wave_cnt[w][dd];
+      }
       tile_total[dd] = p;
       uint32_t v = p;  // inclusive scan over this wave's 64 digits
 #pragma unroll
@@ -403,38 +386,26 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
     }
     __syncthreads();
     // counting-sort the tile into LDS (digit-major, stable)
-    uint32_t tile_pos[RS_V];
-#pragma unroll
-    for (int v = 0; v < RS_V; v++)
-      if (dig[v] < RS_RADIX) {
-        tile_pos[v] = digit_start[dig[v]] + cnt[v][wave][dig[v]]
-                      + rank[v];
-        stage[tile_pos[v]] = key[v];
-      }
+    uint32_t tile_pos = 0;
+    if (valid) {
+      tile_pos = digit_start[d] + wave_pref[wave][d] + rank;
+      stage[tile_pos] = key;
+    }
     __syncthreads();
-    // coalesced write-out: slot j holds the tile's j-th digit-ordered
+    // coalesced write-out: thread j emits the tile's j-th digit-ordered
     // key; destination = digit's global cursor + offset within run
-    uint32_t dst[RS_V];
-#pragma unroll
-    for (int v = 0; v < RS_V; v++) {
-      int j = v * THREADS + threadIdx.x;
-      if (j < tile_n) {
-        uint64_t k2 = stage[j];
-        int d2 = (int)((k2 >> shift) & 255);
-        dst[v] = cur[d2] + j - digit_start[d2];
-        okeys[dst[v]] = k2;
-      }
+    uint64_t k2 = 0;
+    uint32_t dst = 0;
+    if (threadIdx.x < tile_n) {
+      k2 = stage[threadIdx.x];
+      int d2 = (int)((k2 >> shift) & 255);
+      dst = cur[d2] + threadIdx.x - digit_start[d2];
+      okeys[dst] = k2;
     }
     __syncthreads();
-#pragma unroll
-    for (int v = 0; v < RS_V; v++)
-      if (dig[v] < RS_RADIX) stage[tile_pos[v]] = (uint64_t)pl[v];
+    if (valid) stage[tile_pos] = (uint64_t)pl;
     __syncthreads();
-#pragma unroll
-    for (int v = 0; v < RS_V; v++) {
-      int j = v * THREADS + threadIdx.x;
-      if (j < tile_n) opayload[dst[v]] = (P)stage[j];
-    }
+    if (threadIdx.x < tile_n) opayload[dst] = (P)stage[threadIdx.x];
     __syncthreads();
     cur[threadIdx.x] += tile_total[threadIdx.x];
     // next tile's reads of cur happen after its own barriers
@@ -1125,6 +1096,45 @@ void bloom_probe(const int64_t* vals, int64_t n, const uint64_t* words,
                      stream, vals, n, words, m_bits, k, out);
 }
 
+// K9 device sketch-predicate eval: probe every value against every
+// file's bloom filter in one launch; out[f] = any value may be in f.
+__global__ void k_bloom_probe_many(const int64_t* __restrict__ vals,
+                                   int64_t n_vals,
+                                   const uint64_t* __restrict__ words,
+                                   int64_t words_per_filter,
+                                   int64_t n_filters, int64_t m_bits,
+                                   int k, bool* __restrict__ out) {
+  int64_t total = n_filters * n_vals;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       t < total; t += stride) {
+    int64_t f = t / n_vals;
+    int64_t v = t - f * n_vals;
+    const uint64_t* w = words + f * words_per_filter;
+    uint32_t h1, h2;
+    bloom_h12(vals[v], h1, h2);
+    bool all = true;
+    for (int j = 0; j < k && all; j++) {
+      uint64_t combined = ((uint64_t)h1 + (uint64_t)j * (uint64_t)h2) &
+                          0x7FFFFFFFFFFFFFFFull;
+      uint64_t pos = combined % (uint64_t)m_bits;
+      all = (w[pos >> 6] >> (pos & 63)) & 1;
+    }
+    if (all) out[f] = true;  // benign same-value race
+  }
+}
+
+void bloom_probe_many(const int64_t* vals, int64_t n_vals,
+                      const uint64_t* words, int64_t words_per_filter,
+                      int64_t n_filters, int64_t m_bits, int k, bool* out,
+                      hipStream_t stream) {
+  if (n_vals == 0 || n_filters == 0) return;
+  hipLaunchKernelGGL(k_bloom_probe_many,
+                     dim3(grid_for(n_filters * n_vals)), dim3(THREADS), 0,
+                     stream, vals, n_vals, words, words_per_filter,
+                     n_filters, m_bits, k, out);
+}
+
 // ---------------------------------------------------------------------------
 // Z-order bit interleave (K10)
 // ---------------------------------------------------------------------------
@@ -1480,8 +1490,10 @@ __global__ void k_snappy_decomp(const uint8_t* __restrict__ src,
         synced = out;
       }
       if (off >= len) {
-        if (len >= 8 && off >= 8) {
-          // disjoint ranges: 8B-granule copy, dest-aligned middle
+        if (len >= 256 && off >= 8) {
+          // long disjoint copy: 8B-granule, dest-aligned middle (short
+          // copies — the common case — stay on the 1-instruction
+          // byte path; the wide path only pays off past ~4 B/lane)
           int64_t head = (8 - ((uintptr_t)(d + out) & 7)) & 7;
           if (head > len) head = len;
           int64_t body = (len - head) & ~7ll;

@@ -184,7 +184,7 @@ class DataSkippingIndex(Index):
         return written, kept
 
     # -- query -------------------------------------------------------------
-    def load_sketch_data(self, entry):
+    def load_sketch_data(self, entry, device=None):
         import pyarrow.parquet as pq
         import numpy as np
         from ...execution.columnar import ColumnBatch
@@ -203,6 +203,8 @@ class DataSkippingIndex(Index):
                 if not arr.flags.writeable:
                     arr = arr.copy()
                 cols[name] = torch.from_numpy(arr)
+        if device is not None and str(device) != "cpu":
+            cols = {k: v.to(device) for k, v in cols.items()}
         return _SketchData(cols)
 
     def translate_filter(self, pred: Expr, sketch_data
@@ -239,15 +241,18 @@ class DataSkippingIndex(Index):
                     return mask
         return None
 
-    def prune_files(self, entry, pred: Expr, source_files: List[str]
-                    ) -> Tuple[List[str], int]:
+    def prune_files(self, entry, pred: Expr, source_files: List[str],
+                    device=None) -> Tuple[List[str], int]:
         """Returns (files to scan, number skipped).  Files absent from the
-        sketch table (e.g. appended after build) are kept."""
-        data = self.load_sketch_data(entry)
+        sketch table (e.g. appended after build) are kept.  With a CUDA
+        ``device`` the sketch predicate evaluates on device (K9) and only
+        the per-file boolean mask crosses back to the host."""
+        data = self.load_sketch_data(entry, device)
         mask = self.translate_filter(pred, data)
         if mask is None:
             return source_files, 0
-        file_ids = data.tensor("_data_file_id")
+        mask = mask.cpu()
+        file_ids = data.tensor("_data_file_id").cpu()
         id_to_keep = {int(fid): bool(m)
                       for fid, m in zip(file_ids, mask)}
         # map file path -> logged id

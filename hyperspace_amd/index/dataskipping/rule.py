@@ -43,7 +43,8 @@ class ApplyDataSkippingIndex(HyperspaceRule):
         for cand in cands:
             kept, skipped = cand.index.prune_files(
                 cand.entry, filt.condition,
-                [f.name for f in scan.relation.all_files()])
+                [f.name for f in scan.relation.all_files()],
+                device=self.session.device)
             if skipped == 0:
                 self.reasons.add(cand.name, plan, FilterReason(
                     FilterReasons.NO_FILTER_ON_INDEXED_COL,

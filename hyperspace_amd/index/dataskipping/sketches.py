@@ -301,13 +301,15 @@ class MinMaxSketch(Sketch):
                 return mx >= v
             return None  # != not convertible
         if isinstance(pred, In) and self.matches_lhs(pred.col):
-            out = torch.zeros(mn.numel(), dtype=torch.bool)
+            out = torch.zeros(mn.numel(), dtype=torch.bool,
+                              device=mn.device)
             for value in pred.values:
                 v = _norm_scalar(value, dtype_name)
                 out |= (mn <= v) & (mx >= v)
             return out
         if isinstance(pred, IsNotNull) and self.matches_lhs(pred.col):
-            return torch.ones(mn.numel(), dtype=torch.bool)
+            return torch.ones(mn.numel(), dtype=torch.bool,
+                              device=mn.device)
         return None
 
     def to_json(self):
@@ -377,23 +379,19 @@ class BloomFilterSketch(Sketch):
         if values is None:
             return None
         words = sketch_data.tensor(self.out_columns()[0])  # [files, words]
-        n_files = words.shape[0]
-        out = torch.zeros(n_files, dtype=torch.bool)
-        for v in values:
-            dt = {"double": torch.float64,
-                  "float": torch.float32}.get(dtype_name, torch.int64)
-            t = torch.tensor([v], dtype=dt)
-            if dt == torch.float64:
-                t = t.view(torch.int64)
-            elif dt == torch.float32:
-                t = t.to(torch.float64).view(torch.int64)
-            else:
-                t = t.to(torch.int64)
-            for f in range(n_files):
-                hit = ops.cpu_ref.bloom_probe(t, words[f], self.m_bits,
-                                              self.k)
-                out[f] |= bool(hit[0])
-        return out
+        dt = {"double": torch.float64,
+              "float": torch.float32}.get(dtype_name, torch.int64)
+        t = torch.tensor(list(values), dtype=dt)
+        if dt == torch.float64:
+            t = t.view(torch.int64)
+        elif dt == torch.float32:
+            t = t.to(torch.float64).view(torch.int64)
+        else:
+            t = t.to(torch.int64)
+        # K9 on device: one batched kernel probes every value against
+        # every file's filter (host loop fallback on CPU)
+        return ops.bloom_probe_many(t.to(words.device), words,
+                                    self.m_bits, self.k)
 
     def to_json(self):
         return {"type": BLOOM_SKETCH_TYPE, "expr": self.expr,
@@ -440,8 +438,7 @@ class PartitionSketch(Sketch):
         if isinstance(pred, BinComp) and pred.op == "=" and \
                 isinstance(pred.right, Lit) and \
                 self.matches_lhs(pred.left):
-            v = torch.tensor(pred.right.value, dtype=first.dtype)
-            return first == v
+            return first == pred.right.value  # scalar broadcast, any dev
         return None
 
     def to_json(self):

@@ -32,7 +32,7 @@ from .cpu_ref import MASK32, SPARK_HASH_SEED  # re-export
 __all__ = [
     "murmur3_bucket", "normalize_key", "sort_pairs", "sort_perm",
     "merge_join", "run_merge_perm", "select_range_u64", "isin_sorted", "segmented_minmax",
-    "bloom_build", "bloom_probe", "zorder_key", "gather_rows", "native",
+    "bloom_build", "bloom_probe", "bloom_probe_many", "zorder_key", "gather_rows", "native",
 ]
 
 
@@ -138,6 +138,20 @@ def bloom_probe(vals: torch.Tensor, words: torch.Tensor, m_bits: int, k: int
     if _is_cuda(vals):
         return native.ext().bloom_probe(vals, words, m_bits, k)
     return cpu_ref.bloom_probe(vals, words, m_bits, k)
+
+
+def bloom_probe_many(vals: torch.Tensor, words: torch.Tensor, m_bits: int,
+                     k: int) -> torch.Tensor:
+    """K9 batched sketch-predicate probe: ``words`` is [n_filters,
+    words_per_filter]; returns bool[n_filters] = any value may be in
+    that filter (device kernel when the tensors are on GPU)."""
+    if _is_cuda(vals):
+        return native.ext().bloom_probe_many(vals, words, m_bits, k)
+    out = torch.zeros(words.shape[0], dtype=torch.bool)
+    for f in range(words.shape[0]):
+        out[f] = bool(cpu_ref.bloom_probe(vals, words[f], m_bits, k)
+                      .any())
+    return out
 
 
 def zorder_key(cols_u64: List[torch.Tensor], bits_per_col: int
