@@ -22,6 +22,7 @@ class DataFrame:
     def filter(self, condition: Union[str, Expr]) -> "DataFrame":
         if isinstance(condition, str):
             condition = parse_predicate(condition)
+        condition = _bind_decimal_literals(condition, self.plan)
         return DataFrame(self.session, Filter(condition, self.plan))
 
     where = filter
@@ -73,3 +74,66 @@ class DataFrame:
 
     def __repr__(self):
         return f"DataFrame:\n{self.plan.pretty()}"
+
+
+def _bind_decimal_literals(cond: Expr, plan: LogicalPlan) -> Expr:
+    """Scale numeric literals compared against decimal(p,s) columns to
+    the column's unscaled-int64 representation (decimals ingest as
+    unscaled integers with the scale recorded in the schema; Spark's
+    analyzer performs the same literal cast)."""
+    import decimal as _dec
+    from .plan.expr import And, Arith, BinComp, Col, In, Lit, Not, Or
+    scales = {}
+    for leaf in plan.collect_leaves():
+        rel = getattr(leaf, "relation", None)
+        if rel is None:
+            continue
+        try:
+            fields = rel.schema.fields
+        except Exception:  # noqa: BLE001
+            continue
+        for f in fields:
+            t = f.type or ""
+            if t.startswith("decimal("):
+                try:
+                    scales[f.name.lower()] = int(
+                        t[len("decimal("):-1].split(",")[1])
+                except (IndexError, ValueError):
+                    pass
+    if not scales:
+        return cond
+
+    def col_scale(e):
+        if isinstance(e, Col):
+            return scales.get(e.name.lower())
+        if isinstance(e, Arith):
+            return col_scale(e.left) if col_scale(e.left) is not None \
+                else col_scale(e.right)
+        return None
+
+    def scale_value(v, s):
+        return int(_dec.Decimal(str(v)).scaleb(s))
+
+    def walk(e):
+        if isinstance(e, And):
+            return And(walk(e.left), walk(e.right))
+        if isinstance(e, Or):
+            return Or(walk(e.left), walk(e.right))
+        if isinstance(e, Not):
+            return Not(walk(e.child))
+        if isinstance(e, BinComp) and isinstance(e.right, Lit):
+            s = col_scale(e.left)
+            if s is not None and isinstance(
+                    e.right.value, (int, float, _dec.Decimal)):
+                return BinComp(e.op, e.left,
+                               Lit(scale_value(e.right.value, s)))
+        if isinstance(e, In):
+            s = col_scale(e.col)
+            if s is not None:
+                return In(e.col, [
+                    scale_value(v, s)
+                    if isinstance(v, (int, float, _dec.Decimal)) else v
+                    for v in e.values])
+        return e
+
+    return walk(cond)

@@ -295,6 +295,18 @@ class ColumnBatch:
             if pa.types.is_string(col.type) or pa.types.is_large_string(
                     col.type):
                 cols[name] = StringColumn.from_strings(col.to_pylist())
+            elif pa.types.is_decimal(col.type):
+                # decimal(p<=18, s): unscaled int64 representation —
+                # order-correct for a fixed scale; the scale rides in
+                # the schema type string ("decimal(p,s)") and literals
+                # are scaled at predicate binding (dataframe.py)
+                if col.type.precision > 18:
+                    raise HyperspaceException(
+                        f"decimal precision > 18 unsupported: {name}")
+                s = col.type.scale
+                vals = [0 if d is None else int(d.scaleb(s))
+                        for d in col.to_pylist()]
+                cols[name] = torch.tensor(vals, dtype=torch.int64)
             else:
                 np_arr = col.to_numpy(zero_copy_only=False)
                 if np_arr.dtype == np.dtype("datetime64[us]") or \

@@ -95,7 +95,13 @@ class Schema:
                     walk(child, f"{prefix}{f.name}.",
                          nullable or f.nullable)
             else:
-                t = _PYARROW_TO_SPARK.get(str(f.type), str(f.type))
+                ts = str(f.type)
+                if ts.startswith("decimal128(") or \
+                        ts.startswith("decimal256("):
+                    inner = ts[ts.index("(") + 1:-1].replace(" ", "")
+                    t = f"decimal({inner})"
+                else:
+                    t = _PYARROW_TO_SPARK.get(ts, ts)
                 fields.append(SchemaField(prefix + f.name, t,
                                           nullable or f.nullable))
 
