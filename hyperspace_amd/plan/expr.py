@@ -287,27 +287,48 @@ def extract_equi_join_keys(e: Expr) -> List[tuple]:
 
 _PRED_RE = re.compile(
     r"^\s*([A-Za-z_][A-Za-z0-9_.]*)\s*(=|==|!=|<=|>=|<|>)\s*(.+?)\s*$")
+_IN_RE = re.compile(
+    r"^\s*([A-Za-z_][A-Za-z0-9_.]*)\s+(?:in|IN)\s*\(([^)]*)\)\s*$")
+_NULL_RE = re.compile(
+    r"^\s*([A-Za-z_][A-Za-z0-9_.]*)\s+(?:is|IS)\s+"
+    r"(?:(not|NOT)\s+)?(?:null|NULL)\s*$")
+
+
+def _parse_literal(val: str) -> Any:
+    if val.startswith("'") or val.startswith('"'):
+        return val.strip("'\"")
+    try:
+        return int(val)
+    except ValueError:
+        return float(val)
 
 
 def parse_predicate(s: str) -> Expr:
-    """Tiny predicate parser: 'col <op> literal' [AND ...]."""
-    parts = re.split(r"\s+(?:AND|and)\s+", s)
+    """Tiny predicate parser: 'col <op> literal', 'col IN (v, ...)',
+    'col IS [NOT] NULL', joined by AND."""
+    parts = re.split(r"\s+(?:AND|and)\s+(?![^(]*\))", s)
     exprs: List[Expr] = []
     for part in parts:
+        m = _IN_RE.match(part)
+        if m:
+            name, vals = m.groups()
+            exprs.append(In(Col(name),
+                            [_parse_literal(v.strip())
+                             for v in vals.split(",") if v.strip()]))
+            continue
+        m = _NULL_RE.match(part)
+        if m:
+            name, neg = m.groups()
+            exprs.append(IsNotNull(Col(name)) if neg
+                         else IsNull(Col(name)))
+            continue
         m = _PRED_RE.match(part)
         if not m:
             raise HyperspaceException(f"Cannot parse predicate: {part}")
         name, op, val = m.groups()
         if op == "==":
             op = "="
-        if val.startswith("'") or val.startswith('"'):
-            value: Any = val.strip("'\"")
-        else:
-            try:
-                value = int(val)
-            except ValueError:
-                value = float(val)
-        exprs.append(BinComp(op, Col(name), Lit(value)))
+        exprs.append(BinComp(op, Col(name), Lit(_parse_literal(val))))
     out = exprs[0]
     for e in exprs[1:]:
         out = And(out, e)
