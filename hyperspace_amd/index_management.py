@@ -25,14 +25,24 @@ from .log.path_resolver import PathResolver
 
 
 class IndexCollectionManager:
-    def __init__(self, session):
+    def __init__(self, session, log_manager_factory=None,
+                 data_manager_factory=None):
+        """``*_factory``: DI seams (reference index/factories.scala) —
+        action/unit tests inject mocked managers through them; the
+        defaults build the real filesystem-backed managers."""
+        from .log import (IndexDataManagerFactory, IndexLogManagerFactory)
         self.session = session
         self.path_resolver = PathResolver(session.conf)
+        self.log_manager_factory = (log_manager_factory
+                                    or IndexLogManagerFactory())
+        self.data_manager_factory = (data_manager_factory
+                                     or IndexDataManagerFactory())
 
     # -- helpers -----------------------------------------------------------
     def _managers(self, name: str):
         path = self.path_resolver.get_index_path(name)
-        return path, IndexLogManager(path), IndexDataManager(path)
+        return (path, self.log_manager_factory.create(path),
+                self.data_manager_factory.create(path))
 
     def log_manager(self, name: str) -> IndexLogManager:
         """Log manager for one index (closestIndex time travel reads
@@ -73,7 +83,8 @@ class IndexCollectionManager:
         latest = log_mgr.get_latest_stable_log()
         if latest is not None and latest.state == States.ACTIVE:
             VacuumOutdatedAction(self.session, log_mgr,
-                                 IndexDataManager(path)).run()
+                                 self.data_manager_factory
+                                 .create(path)).run()
         else:
             VacuumAction(self.session, log_mgr, path).run()
 
