@@ -225,25 +225,81 @@ class CoveringIndex(Index):
         if batch.num_rows == 0:
             return []
         t1 = _time.perf_counter()
-        batch, seg = sort_by_bucket_and_keys(
-            batch, bucket_ids, self.indexed_columns, n)
-        if timing:
-            import torch as _torch
-            if batch.device.type == "cuda":
-                _torch.cuda.synchronize()
+
+        pipeline = (batch.device.type == "cuda" and n >= 8
+                    and batch.num_rows >= 1 << 22)
+        if not pipeline:
+            batch, seg = sort_by_bucket_and_keys(
+                batch, bucket_ids, self.indexed_columns, n)
+            if timing:
+                import torch as _torch
+                if batch.device.type == "cuda":
+                    _torch.cuda.synchronize()
+            t2 = _time.perf_counter()
+            out = write_bucketed(batch, seg, ctx.index_data_path, n,
+                                 task_id)
+            # build write-through for the HBM index-data cache: the
+            # sorted bucket-major batch + segment offsets ARE the
+            # index-scan layout, so the first post-build query serves
+            # from residency (288 GB HBM3E; the put happens after the
+            # action commits — see IndexManagement)
+            ctx.built_for_cache = (batch, seg, list(out))
+            if timing:
+                import sys as _sys
+                print(f"[hs-timing] build group: hash {t1-t0:.3f}s "
+                      f"sort {t2-t1:.3f}s write "
+                      f"{_time.perf_counter()-t2:.3f}s",
+                      file=_sys.stderr)
+            return out
+
+        # Pipelined sort+write: one stable bucket-partition pass, then
+        # bucket BATCHES key-sort on the GPU while earlier batches'
+        # D2H + parquet writes drain on worker threads/streams — the
+        # write wall-time (the largest build phase) overlaps the sort
+        # instead of following it.
+        import torch as _torch
+        from concurrent.futures import ThreadPoolExecutor
+        b64 = bucket_ids.to(_torch.int64)
+        perm = ops.sort_perm(ops.normalize_key(b64))
+        batch = batch.gather(perm)
+        sorted_b = ops.gather_rows(b64, perm)
+        seg = _bucket_seg_offsets(sorted_b, n)
         t2 = _time.perf_counter()
-        out = write_bucketed(batch, seg, ctx.index_data_path, n, task_id)
-        # build write-through for the HBM index-data cache: the sorted
-        # bucket-major batch + segment offsets ARE the index-scan layout,
-        # so the first post-build query need not re-read what is already
-        # resident (288 GB HBM3E; the put happens after the action
-        # commits, keyed on the final log id — see IndexManagement)
-        ctx.built_for_cache = (batch, seg, list(out))
+
+        nb_batches = min(8, n)
+        bounds = [round(i * n / nb_batches) for i in range(nb_batches + 1)]
+        pool = ThreadPoolExecutor(max_workers=2)
+        futures = []
+        pieces: List[ColumnBatch] = []
+        for i in range(nb_batches):
+            b_lo, b_hi = bounds[i], bounds[i + 1]
+            r_lo, r_hi = int(seg[b_lo]), int(seg[b_hi])
+            if r_hi <= r_lo:
+                continue
+            sub = batch.slice(r_lo, r_hi)
+            sub_sorted, lseg = sort_by_bucket_and_keys(
+                sub, sorted_b[r_lo:r_hi], self.indexed_columns, n)
+            ev = _torch.cuda.Event()
+            ev.record()
+            pieces.append(sub_sorted)
+            futures.append(pool.submit(
+                write_bucketed, sub_sorted, lseg, ctx.index_data_path,
+                n, task_id, ev))
+        out = []
+        for f in futures:
+            out.extend(f.result())
+        pool.shutdown()
+        if timing:
+            _torch.cuda.synchronize()
+        t3 = _time.perf_counter()
+        full = (ColumnBatch.concat(pieces) if len(pieces) > 1
+                else pieces[0]) if pieces else batch
+        ctx.built_for_cache = (full, seg, list(out))
         if timing:
             import sys as _sys
             print(f"[hs-timing] build group: hash {t1-t0:.3f}s "
-                  f"sort {t2-t1:.3f}s write "
-                  f"{_time.perf_counter()-t2:.3f}s", file=_sys.stderr)
+                  f"partition {t2-t1:.3f}s sort+write(pipe) "
+                  f"{t3-t2:.3f}s", file=_sys.stderr)
         return out
 
     def _empty_batch(self, ctx) -> ColumnBatch:
@@ -376,6 +432,26 @@ def _multi_key_sort_perm(batch: ColumnBatch, key_cols: List[str]
     return perm
 
 
+def _bucket_seg_offsets(sorted_bkeys_u64: torch.Tensor, num_buckets: int
+                        ) -> torch.Tensor:
+    """Per-bucket segment offsets (num_buckets+1) of a bucket-sorted
+    normalized-or-raw int64 bucket column, via device searchsorted."""
+    probes = ops.cpu_ref.normalize_key(
+        torch.arange(num_buckets + 1, dtype=torch.int64)).to(
+            sorted_bkeys_u64.device)
+    if sorted_bkeys_u64.numel() and int(sorted_bkeys_u64.max()) >= 0 \
+            and int(sorted_bkeys_u64.min()) >= 0:
+        # raw (un-normalized) bucket ids sort in plain int64 order
+        seg = torch.searchsorted(
+            sorted_bkeys_u64,
+            torch.arange(num_buckets + 1, dtype=torch.int64,
+                         device=sorted_bkeys_u64.device))
+        return seg.cpu()
+    sortable = sorted_bkeys_u64 ^ (-0x8000000000000000)
+    sortable_probes = probes ^ (-0x8000000000000000)
+    return torch.searchsorted(sortable, sortable_probes).cpu()
+
+
 def sort_by_bucket_and_keys(batch: ColumnBatch, bucket_ids: torch.Tensor,
                             key_cols: List[str], num_buckets: int
                             ) -> Tuple[ColumnBatch, torch.Tensor]:
@@ -397,13 +473,17 @@ def sort_by_bucket_and_keys(batch: ColumnBatch, bucket_ids: torch.Tensor,
 
 
 def write_bucketed(batch: ColumnBatch, seg: torch.Tensor, out_dir: str,
-                   num_buckets: int, task_id: int = 0) -> List[str]:
+                   num_buckets: int, task_id: int = 0,
+                   after_event: Optional["torch.cuda.Event"] = None
+                   ) -> List[str]:
     """Write per-bucket parquet files honoring the bucket-id filename
     contract.  Empty buckets produce no file (as in Spark).
 
     Files are written by a thread pool: the native encoder's byte
     assembly (numpy tobytes) and os.write both release the GIL, so the
-    200-file write overlaps to page-cache speed.
+    200-file write overlaps to page-cache speed.  ``after_event``: D2H
+    copies wait on it (orders them after the producing sort when the
+    caller pipelines sort and write on different streams).
     """
     from concurrent.futures import ThreadPoolExecutor
     from ...execution.columnar import StringColumn
@@ -430,6 +510,8 @@ def write_bucketed(batch: ColumnBatch, seg: torch.Tensor, out_dir: str,
         held = []
         if on_device:
             s = streams[job_i % len(streams)]
+            if after_event is not None:
+                s.wait_event(after_event)
             cols = {}
             masks = {}
             with torch.cuda.stream(s):
