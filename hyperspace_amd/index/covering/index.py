@@ -227,7 +227,8 @@ class CoveringIndex(Index):
         t1 = _time.perf_counter()
 
         pipeline = (batch.device.type == "cuda" and n >= 8
-                    and batch.num_rows >= 1 << 22)
+                    and batch.num_rows >= 1 << 22
+                    and not os.environ.get("HS_NO_PIPE"))
         if not pipeline:
             batch, seg = sort_by_bucket_and_keys(
                 batch, bucket_ids, self.indexed_columns, n)

@@ -49,4 +49,41 @@ class CandidateIndexAnalyzer:
             for r in rs:
                 detail = f" {r.args}" if (extended and r.args) else ""
                 lines.append(f"  {r.code}{detail}")
+
+        # per-(index, subplan) matrix (reference
+        # CandidateIndexAnalyzer.scala:29-346 renders the reason table
+        # against each sub-plan the rules considered)
+        matrix = reasons.by_subplan()
+        if matrix:
+            lines.append(bar)
+            lines.append("Reasons per sub-plan:")
+            lines.append(bar)
+            for sp_i, (nid, per_index) in enumerate(matrix.items(), 1):
+                node = reasons.nodes.get(nid)
+                lines.append(
+                    f"SubPlan #{sp_i}: {_one_line(node)}")
+                width = max([len(n) for n in per_index if n] + [5])
+                lines.append(f"  {'index'.ljust(width)} | reason")
+                lines.append(f"  {'-' * width}-+-{'-' * 30}")
+                for iname in sorted(per_index):
+                    if index_name and iname and iname != index_name:
+                        continue
+                    for r in per_index[iname]:
+                        detail = f" {r.args}" if (extended and r.args) \
+                            else ""
+                        lines.append(
+                            f"  {(iname or '*').ljust(width)} | "
+                            f"{r.code}{detail}")
         return "\n".join(lines)
+
+
+def _one_line(node) -> str:
+    """Compress a plan subtree into one line for the matrix header."""
+    if node is None:
+        return "<plan>"
+    try:
+        flat = " / ".join(part.strip()
+                          for part in node.pretty().splitlines())
+    except Exception:  # noqa: BLE001
+        flat = repr(node)
+    return flat[:110] + ("..." if len(flat) > 110 else "")

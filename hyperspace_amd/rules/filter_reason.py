@@ -46,11 +46,13 @@ class ReasonCollector:
         self.enabled = enabled
         self.reasons: Dict[Tuple[str, int], List[FilterReason]] = {}
         self.applied: Dict[str, List[str]] = {}
+        self.nodes: Dict[int, object] = {}  # id -> plan node (matrix)
 
     def add(self, index_name: str, plan_node, reason: FilterReason):
         if not self.enabled:
             return
         key = (index_name, id(plan_node))
+        self.nodes[id(plan_node)] = plan_node
         self.reasons.setdefault(key, []).append(reason)
 
     def all_for_index(self, index_name: str) -> List[FilterReason]:
@@ -58,4 +60,12 @@ class ReasonCollector:
         for (name, _), rs in self.reasons.items():
             if name == index_name:
                 out.extend(rs)
+        return out
+
+    def by_subplan(self) -> "Dict[int, Dict[str, List[FilterReason]]]":
+        """plan-node id -> index name -> reasons (the reference's
+        per-(index, subplan) whyNot matrix)."""
+        out: Dict[int, Dict[str, List[FilterReason]]] = {}
+        for (name, nid), rs in self.reasons.items():
+            out.setdefault(nid, {}).setdefault(name, []).extend(rs)
         return out

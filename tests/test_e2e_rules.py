@@ -320,3 +320,27 @@ def test_covering_index_serves_arithmetic_filter(env):
     base = q.collect()
     assert out.num_rows == base.num_rows
     assert bool((out.tensor("qty") % 7 == 3).all())
+
+
+def test_why_not_per_subplan_matrix(env):
+    """whyNot renders the per-(index, subplan) reason matrix (reference
+    CandidateIndexAnalyzer renders reasons against each sub-plan)."""
+    session, h, left, right = env
+    h.create_index(left, hs.CoveringIndexConfig("mfix", ["qty"],
+                                                ["price"]))
+    q = left.filter("price >= 50.0").select("price")
+    out = h.why_not(q, extended=True)
+    assert "Reasons per sub-plan:" in out
+    assert "SubPlan #1:" in out
+    assert "mfix" in out
+    # the matrix names the sub-plan shape
+    assert "Filter" in out or "Scan" in out
+
+
+def test_minmax_html_report(env, tmp_path):
+    session, h, left, right = env
+    from hyperspace_amd.utils.minmax_analysis import analyze_html
+    html = analyze_html(left, ["qty"])
+    assert html.startswith("<!DOCTYPE html>")
+    assert "<svg" in html and "rect" in html
+    assert "qty" in html
