@@ -564,16 +564,20 @@ void exclusive_scan_i64(const int64_t* in, int64_t* out, int64_t n,
   }
   int nb = scan_num_blocks(n);
   int64_t per_block = cdiv(n, nb);
-  // temp block totals: caller-invisible; small static buffer via hipMalloc
-  static thread_local int64_t* tots = nullptr;
-  if (tots == nullptr)
-    HIP_CHECK(hipMalloc(&tots, SCAN_MAX_BLOCKS * sizeof(int64_t)));
+  // temp block totals: stream-ordered allocation — pool-backed (no
+  // steady-state allocation cost), freed on the same stream, and safe
+  // under multi-stream use from one host thread (a shared static here
+  // would race between concurrent streams)
+  int64_t* tots = nullptr;
+  HIP_CHECK(hipMallocAsync(&tots, SCAN_MAX_BLOCKS * sizeof(int64_t),
+                           stream));
   hipLaunchKernelGGL(k_scan_partial, dim3(nb), dim3(THREADS), 0, stream, in,
                      out, n, per_block, tots);
   hipLaunchKernelGGL(k_scan_block_tots, dim3(1), dim3(1024), 0, stream, tots,
                      nb, total);
   hipLaunchKernelGGL(k_scan_add_base, dim3(nb), dim3(THREADS), 0, stream,
                      out, n, per_block, tots);
+  HIP_CHECK(hipFreeAsync(tots, stream));
 }
 
 // ---------------------------------------------------------------------------
