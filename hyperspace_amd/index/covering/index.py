@@ -226,9 +226,15 @@ class CoveringIndex(Index):
             return []
         t1 = _time.perf_counter()
 
+        # Bucket-batch pipelined sort+write, OPT-IN: a same-box A/B
+        # (profiles/README round 2) measured it 45% SLOWER than the
+        # monolithic sort+write at 8 GiB/200 buckets — the extra
+        # partition pass + per-batch sort fixed costs outweigh the
+        # write overlap.  Kept behind HS_PIPE_WRITE for re-evaluation
+        # at other shapes (more buckets, slower storage).
         pipeline = (batch.device.type == "cuda" and n >= 8
                     and batch.num_rows >= 1 << 22
-                    and not os.environ.get("HS_NO_PIPE"))
+                    and bool(os.environ.get("HS_PIPE_WRITE")))
         if not pipeline:
             batch, seg = sort_by_bucket_and_keys(
                 batch, bucket_ids, self.indexed_columns, n)
