@@ -15,6 +15,7 @@ from __future__ import annotations
 
 import os
 import re
+import sys
 import uuid
 from typing import Dict, List, Optional, Tuple
 
@@ -630,6 +631,9 @@ def read_files_batch_device(paths: List[str], device,
         return buf
 
     # phase 1: read + layout-parse every file (parallel, GIL-released IO)
+    import time as _time
+    _dt = os.environ.get("HS_DECODE_TIMING")
+    _t0 = _time.perf_counter()
     if len(paths) > 2:
         with ThreadPoolExecutor(max_workers=16) as pool:
             infos = list(pool.map(load_file, range(len(paths))))
@@ -681,18 +685,24 @@ def read_files_batch_device(paths: List[str], device,
             _decode_on_stream(i, buf, size, chunks, dev_bufs[i], row_off)
         return True
 
+    _t1 = _time.perf_counter()
     if len(units) > 2:
         with ThreadPoolExecutor(max_workers=16) as pool:
             list(pool.map(decode_unit, enumerate(units)))
     else:
         for u in enumerate(units):
             decode_unit(u)
+    _t2 = _time.perf_counter()
     # order the default stream after every worker stream, then host-sync
     # so the pinned buffers can be recycled
     cur = _torch.cuda.current_stream()
     for s in streams:
         cur.wait_stream(s)
     cur.synchronize()
+    if _dt:
+        print(f"[hs-decode] files={len(paths)} units={len(units)} "
+              f"read+layout {_t1-_t0:.3f}s decode-host {_t2-_t1:.3f}s "
+              f"sync {_time.perf_counter()-_t2:.3f}s", file=sys.stderr)
     snappy_bad = any(bool((s != 0).any()) for s in statuses)
     if snappy_bad or \
             not all(b is not None and b is not False for b in bufs):
