@@ -686,8 +686,9 @@ def _walk_row_group(rg, pf_schema, data, want
         is_string = False
         if col.physical_type == "BYTE_ARRAY":
             # dictionary-encoded strings decode natively (codes + one
-            # dictionary parse); PLAIN/compressed strings -> pyarrow
-            if not is_dict or codec != "UNCOMPRESSED":
+            # dictionary parse), uncompressed or snappy; PLAIN-encoded
+            # strings -> pyarrow
+            if not is_dict:
                 return None
             is_string = True
             np_dtype = np.dtype("int32")
@@ -876,8 +877,10 @@ def read_native_host(path: str, columns: Optional[List[str]] = None
     if layout is None:
         return None
     data, chunks = layout
-    if any(c.encoding != "plain" and not c.is_string for c in chunks):
-        return None  # numeric dict decode is the device path
+    if any(not (c.encoding == "plain"
+                or (c.is_string and c.encoding == "dict"))
+           for c in chunks):
+        return None  # numeric dict + snappy decode is the device path
     acc: Dict[str, list] = {}
     macc: Dict[str, list] = {}
     any_null: Dict[str, bool] = {}

@@ -347,7 +347,7 @@ def read_files_batch_device(paths: List[str], device,
     # per-worker HIP streams: each file's H2D copy and decode kernels run
     # on their own stream, so copies overlap other files' decodes instead
     # of serializing on the default stream (xfers are the cold-load bound)
-    n_streams = 8  # row-group units of a single file fan out too
+    n_streams = 16  # row-group units of a single file fan out too
     streams = [_torch.cuda.Stream(device=device) for _ in range(n_streams)]
 
     def load_file(i):
@@ -423,7 +423,25 @@ def read_files_batch_device(paths: List[str], device,
                 page_base = doff_all[1:-1] if has_zdict \
                     else doff_all[:-1]
                 dict_vals = None
-                if has_zdict:
+                z_is_str = c.is_string
+                if has_zdict and z_is_str:
+                    # string dictionary: small D2H of the decompressed
+                    # dict page, parsed to values on host; indices stay
+                    # on device as the column's codes
+                    draw = scratch[:dict_unc].cpu().numpy()
+                    vals_list = []
+                    pos = 0
+                    for _ in range(dict_n):
+                        ln = int(np.frombuffer(draw, "<u4", 1, pos)[0])
+                        pos += 4
+                        vals_list.append(
+                            draw[pos:pos + ln].tobytes().decode("utf-8"))
+                        pos += ln
+                    str_chunks.append(
+                        (c.name, written,
+                         written + sum(p[3] for p in c.pages),
+                         vals_list))
+                elif has_zdict:
                     dict_vals = _torch.empty(
                         dict_n + 1, dtype=out[c.name].dtype,
                         device=device)
@@ -467,6 +485,46 @@ def read_files_batch_device(paths: List[str], device,
                         cur += b - a
                 hb_all = (_torch.cat(cat_parts).cpu() if cat_parts
                           else None)
+                # fast path: all-dict_z chunk with no nulls — ONE
+                # GIL-released C++ run parse over every page and (for
+                # the common uniform bit width) ONE rle_decode, instead
+                # of per-page python work (the host cost that dominated
+                # multi-row-group snappy decode)
+                if c.pages and not nullable_chunk and \
+                        all(p[0] == "dict_z" for p in c.pages):
+                    bws = [int(hb_all[regions[(j, "idx")][0]])
+                           for j in range(len(c.pages))]
+                    if len(set(bws)) == 1 and bws[0] > 0:
+                        starts, ends, nvs, outs, bshift = \
+                            [], [], [], [], []
+                        row = 0
+                        for j, page in enumerate(c.pages):
+                            s0, rln, absbase = regions[(j, "idx")]
+                            starts.append(s0 + 1)
+                            ends.append(s0 + rln)
+                            nvs.append(page[3])
+                            outs.append(row)
+                            bshift.append((absbase - s0) * 8)
+                            row += page[3]
+
+                        def t64(x):
+                            return _torch.tensor(x, dtype=_torch.int64)
+                        kind, ooff, ln, boff, val, _cnt = \
+                            ext.parse_rle_runs_batch(
+                                hb_all, t64(starts), t64(ends),
+                                t64(bws), t64(nvs), t64(outs),
+                                t64(bshift))
+                        idx = ext.rle_decode(scratch, kind, ooff, ln,
+                                             boff, val, bws[0], row)
+                        if z_is_str:
+                            out[c.name][written:written + row] = idx
+                        else:
+                            out[c.name][written:written + row] = \
+                                ext.gather_rows(dict_vals,
+                                                idx.to(_torch.int64))
+                        written += row
+                        cursors[c.name] = written
+                        continue
                 # batch unmasked dict_z pages: one rle_decode + gather
                 # per (chunk, bit-width) instead of per page
                 zbatch: Dict[int, list] = {}
@@ -504,8 +562,9 @@ def read_files_batch_device(paths: List[str], device,
                             written += nv
                             continue
                         idx = ext.rle_decode(scratch, *runs, bw,
-                                             n_valid).to(_torch.int64)
-                        vals = ext.gather_rows(dict_vals, idx)
+                                             n_valid)
+                        vals = idx if z_is_str else ext.gather_rows(
+                            dict_vals, idx.to(_torch.int64))
                     else:
                         if pmask is None:
                             ext.copy_unaligned(scratch, base + skip,
@@ -534,8 +593,9 @@ def read_files_batch_device(paths: List[str], device,
                     cat = [_torch.cat([r[k] for r in lst["runs"]])
                            for k in range(5)]
                     idx = ext.rle_decode(scratch, *cat, bw,
-                                         lst["rows"]).to(_torch.int64)
-                    vals = ext.gather_rows(dict_vals, idx)
+                                         lst["rows"])
+                    vals = idx if z_is_str else ext.gather_rows(
+                        dict_vals, idx.to(_torch.int64))
                     for woff, nv, rbase in lst["pages"]:
                         out[c.name][woff:woff + nv] = \
                             vals[rbase:rbase + nv]
