@@ -152,3 +152,26 @@ def test_corrupt_snappy_file_falls_back(tmp_path):
         # if pyarrow tolerated it, fine — just require no crash
     except Exception:
         pass  # host fallback raising on corrupt data is acceptable
+
+
+def test_snappy_string_dictionary_device_read(tmp_path):
+    """Snappy-compressed dictionary STRING chunks decode on device: the
+    dictionary page D2Hs once for value parsing, the RLE index pages
+    become codes in HBM; content equals the pyarrow read."""
+    rng = np.random.default_rng(17)
+    words = sorted(f"w{i:04d}" for i in range(800))
+    data = [words[i] for i in rng.integers(0, 800, 400_000)]
+    p = str(tmp_path / "sdz.parquet")
+    pq.write_table(pa.table({"s": pa.array(data).dictionary_encode(),
+                             "k": rng.integers(0, 99, 400_000)}),
+                   p, compression="SNAPPY", use_dictionary=True,
+                   data_page_version="1.0")
+    from hyperspace_amd.execution.columnar import StringColumn
+    batch, rc = read_files_batch_device([p], torch.device("cuda:0"))
+    s = batch.column("s")
+    assert isinstance(s, StringColumn)
+    assert s.codes.device.type == "cuda"
+    got = s.to_numpy()
+    assert (got == np.array(data, dtype=object)).all()
+    assert int(batch.tensor("k").sum()) == int(
+        pq.read_table(p).column("k").to_numpy().sum())
