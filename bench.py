@@ -59,6 +59,13 @@ def main():
             torch.cuda.set_device(local_rank % torch.cuda.device_count())
         dist.init_process_group(backend=backend)
 
+    # pinned-host staging pool: the per-process default (32 GiB) is
+    # sized for 1 rank; 8 ranks on one node would pin 256 GiB of host
+    # RAM — scale it down before the package reads the env
+    if world > 1:
+        os.environ.setdefault("HS_PINNED_POOL_GB",
+                              str(max(4, 32 // world)))
+
     import hyperspace_amd as hs
     from hyperspace_amd import bench_utils
     from hyperspace_amd.execution.executor import Executor
