@@ -38,8 +38,8 @@ h.refresh_index("f", mode="incremental")
 pr.disable()
 torch.cuda.synchronize(); print(f"refresh incremental (+5%): {time.time()-t0:.2f}s")
 sbuf = _io.StringIO()
-pstats.Stats(pr, stream=sbuf).sort_stats("cumulative").print_stats(14)
-print("\n".join(sbuf.getvalue().splitlines()[4:22]))
+pstats.Stats(pr, stream=sbuf).sort_stats("cumulative").print_stats(30)
+print("\n".join(sbuf.getvalue().splitlines()[4:40]))
 session.enable_hyperspace()
 q = fact.select("key", "val").join(dim.select("key", "status"), on="key")
 plan = q.optimized_plan()
