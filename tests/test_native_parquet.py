@@ -118,3 +118,58 @@ def test_multi_rowgroup_native_read(tmp_path):
         assert np.array_equal(back[k], v), k
     batch, counts = read_files_batch([p])
     assert counts == [1_200_000]
+
+
+def test_multi_row_group_native_read(tmp_path):
+    """Multi-row-group uncompressed files decode natively with correct
+    per-row-group boundaries (the decode path splits one file's row
+    groups across parallel units on GPU; the host path walks the same
+    layout row-group-major)."""
+    import numpy as np
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from hyperspace_amd.sources.parquet_io import read_files_batch
+    rng = np.random.default_rng(3)
+    n = 100_000
+    key = rng.integers(0, 1000, n)
+    val = rng.random(n)
+    p = str(tmp_path / "mrg.parquet")
+    pq.write_table(pa.table({"key": key, "val": val}), p,
+                   compression="NONE", use_dictionary=False,
+                   data_page_version="1.0", row_group_size=7_000)
+    assert pq.ParquetFile(p).metadata.num_row_groups > 10
+    batch, rc = read_files_batch([p])
+    assert rc == [n]
+    assert (batch.tensor("key").numpy() == key).all()
+    assert np.allclose(batch.tensor("val").numpy(), val)
+
+
+def test_split_row_groups_boundaries():
+    """The GPU decode's row-group splitter: layout order is row-group-
+    major, so a repeated column name starts a new group and row offsets
+    accumulate per group."""
+    from hyperspace_amd.sources.native_parquet import ColumnChunkLayout
+    import numpy as np
+
+    def chunk(name, nrows):
+        return ColumnChunkLayout(name, np.dtype("int64"),
+                                 [("plain", 0, nrows)], nrows)
+
+    chunks = [chunk("a", 10), chunk("b", 10),
+              chunk("a", 7), chunk("b", 7),
+              chunk("a", 3), chunk("b", 3)]
+    # reproduce the splitter used by read_files_batch_device
+    groups = []
+    cur, seen, row_off = [], set(), 0
+    for c in chunks:
+        if c.name in seen:
+            groups.append((row_off, cur))
+            row_off += cur[0].num_values
+            cur, seen = [], set()
+        cur.append(c)
+        seen.add(c.name)
+    if cur:
+        groups.append((row_off, cur))
+    assert [(off, [c.name for c in cs], cs[0].num_values)
+            for off, cs in groups] == \
+        [(0, ["a", "b"], 10), (10, ["a", "b"], 7), (17, ["a", "b"], 3)]
