@@ -485,46 +485,59 @@ def read_files_batch_device(paths: List[str], device,
                         cur += b - a
                 hb_all = (_torch.cat(cat_parts).cpu() if cat_parts
                           else None)
-                # fast path: all-dict_z chunk with no nulls — ONE
-                # GIL-released C++ run parse over every page and (for
-                # the common uniform bit width) ONE rle_decode, instead
-                # of per-page python work (the host cost that dominated
-                # multi-row-group snappy decode)
-                if c.pages and not nullable_chunk and \
-                        all(p[0] == "dict_z" for p in c.pages):
-                    bws = [int(hb_all[regions[(j, "idx")][0]])
-                           for j in range(len(c.pages))]
-                    if len(set(bws)) == 1 and bws[0] > 0:
-                        starts, ends, nvs, outs, bshift = \
-                            [], [], [], [], []
-                        row = 0
-                        for j, page in enumerate(c.pages):
+                # fast path for null-free chunks: ONE GIL-released C++
+                # run parse over every dictionary page (uniform bit
+                # width) + ONE rle_decode, with PLAIN pages streaming
+                # through copy_unaligned — instead of per-page python
+                # work (the host cost that dominated multi-row-group
+                # snappy decode; mixed dict+plain chunks appear when
+                # pyarrow's dictionary overflows mid-chunk)
+                dz_js = [j for j, p in enumerate(c.pages)
+                         if p[0] == "dict_z"]
+                bws = sorted({int(hb_all[regions[(j, "idx")][0]])
+                              for j in dz_js})
+                if c.pages and not nullable_chunk and len(bws) <= 1 \
+                        and (not bws or bws[0] > 0):
+                    starts, ends, nvs, outs, bshift = [], [], [], [], []
+                    dplaces = []  # (abs row, dict-compact row, nv)
+                    row = written
+                    drows = 0
+                    for j, page in enumerate(c.pages):
+                        nv = page[3]
+                        if page[0] == "dict_z":
                             s0, rln, absbase = regions[(j, "idx")]
                             starts.append(s0 + 1)
                             ends.append(s0 + rln)
-                            nvs.append(page[3])
-                            outs.append(row)
+                            nvs.append(nv)
+                            outs.append(drows)
                             bshift.append((absbase - s0) * 8)
-                            row += page[3]
-
+                            dplaces.append((row, drows, nv))
+                            drows += nv
+                        else:  # plain_z
+                            skip = lvl_skips[j] if page[5] else 0
+                            ext.copy_unaligned(
+                                scratch, int(page_base[j]) + skip,
+                                out[c.name], row * itemsize,
+                                nv * itemsize)
+                        row += nv
+                    if dz_js:
                         def t64(x):
                             return _torch.tensor(x, dtype=_torch.int64)
                         kind, ooff, ln, boff, val, _cnt = \
                             ext.parse_rle_runs_batch(
                                 hb_all, t64(starts), t64(ends),
-                                t64(bws), t64(nvs), t64(outs),
-                                t64(bshift))
+                                t64([bws[0]] * len(starts)), t64(nvs),
+                                t64(outs), t64(bshift))
                         idx = ext.rle_decode(scratch, kind, ooff, ln,
-                                             boff, val, bws[0], row)
-                        if z_is_str:
-                            out[c.name][written:written + row] = idx
-                        else:
-                            out[c.name][written:written + row] = \
-                                ext.gather_rows(dict_vals,
-                                                idx.to(_torch.int64))
-                        written += row
-                        cursors[c.name] = written
-                        continue
+                                             boff, val, bws[0], drows)
+                        vals = idx if z_is_str else ext.gather_rows(
+                            dict_vals, idx.to(_torch.int64))
+                        for abs_row, drow, nv in dplaces:
+                            out[c.name][abs_row:abs_row + nv] = \
+                                vals[drow:drow + nv]
+                    written = row
+                    cursors[c.name] = written
+                    continue
                 # batch unmasked dict_z pages: one rle_decode + gather
                 # per (chunk, bit-width) instead of per page
                 zbatch: Dict[int, list] = {}

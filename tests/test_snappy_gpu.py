@@ -175,3 +175,26 @@ def test_snappy_string_dictionary_device_read(tmp_path):
     assert (got == np.array(data, dtype=object)).all()
     assert int(batch.tensor("k").sum()) == int(
         pq.read_table(p).column("k").to_numpy().sum())
+
+
+def test_snappy_mixed_dict_plain_chunk(tmp_path):
+    """pyarrow's dictionary overflow mid-chunk produces MIXED dict+PLAIN
+    snappy pages in one column chunk; the batched fast path must decode
+    both kinds with exact content (the incremental-refresh file shape)."""
+    rng = np.random.default_rng(23)
+    n = 2_000_000
+    key = rng.integers(0, 1 << 24, n)  # huge keyspace -> dict overflow
+    val = rng.random(n)
+    p = str(tmp_path / "mix.parquet")
+    pq.write_table(pa.table({"key": key, "val": val}), p,
+                   compression="SNAPPY", use_dictionary=True,
+                   dictionary_pagesize_limit=64 * 1024,
+                   data_page_version="1.0")
+    encs = set(pq.ParquetFile(p).metadata.row_group(0).column(0)
+               .encodings)
+    assert "PLAIN" in encs and encs & {"PLAIN_DICTIONARY",
+                                       "RLE_DICTIONARY"}, encs
+    batch, rc = read_files_batch_device([p], torch.device("cuda:0"))
+    assert rc == [n]
+    assert (batch.tensor("key").cpu().numpy() == key).all()
+    assert np.allclose(batch.tensor("val").cpu().numpy(), val)
