@@ -247,7 +247,12 @@ class DeltaTable:
         new_md["schemaString"] = json.dumps(schema.to_json())
         return [{"metaData": new_md}]
 
-    def append_files(self, paths: List[str]) -> int:
+    def append_files(self, paths: List[str],
+                     partition_values: Optional[
+                         Dict[str, Dict[str, str]]] = None) -> int:
+        """``partition_values``: optional abs-path -> partitionValues
+        map for partitioned tables (values are strings, per the Delta
+        protocol)."""
         actions: List[Dict[str, Any]] = [
             {"commitInfo": {"timestamp": _now_ms(),
                             "operation": "WRITE",
@@ -256,10 +261,11 @@ class DeltaTable:
             actions.extend(self._ensure_schema_action(paths[0]))
         for p in paths:
             st = os.stat(p)
-            rel = os.path.relpath(os.path.abspath(p), self.path)
+            ap = os.path.abspath(p)
+            rel = os.path.relpath(ap, self.path)
             actions.append({"add": {
                 "path": quote(rel),
-                "partitionValues": {},
+                "partitionValues": (partition_values or {}).get(ap, {}),
                 "size": st.st_size,
                 "modificationTime": int(st.st_mtime * 1000),
                 "dataChange": True}})
@@ -447,6 +453,50 @@ class DeltaTableRelation(FileBasedRelation):
 
     def all_files(self) -> List[FileInfo]:
         return self.table.files_at(self.version_as_of)
+
+    # -- partitioned tables -------------------------------------------------
+    def partition_schema(self) -> Schema:
+        """Partition columns from metaData.partitionColumns; their types
+        come from schemaString (Delta includes partition columns in the
+        table schema but NOT in the data files)."""
+        if getattr(self, "_pschema", None) is None:
+            md = self.table.metadata_at(self.version_as_of) or {}
+            pcols = md.get("partitionColumns") or []
+            if not pcols:
+                self._pschema = Schema([])
+            else:
+                from ..log.entry import SchemaField
+                full = self.schema
+                self._pschema = Schema([
+                    SchemaField(c, full.field_type(c) or "string", False)
+                    for c in pcols])
+        return self._pschema
+
+    def partition_values(self, path: str) -> Dict[str, Any]:
+        """Typed partition values for one data file (from the add
+        actions' partitionValues map)."""
+        if getattr(self, "_pvalues", None) is None:
+            self._pvalues = self.table.partition_values_at(
+                self.version_as_of)
+        from .parquet_source import _cast_partition_value
+        raw = self._pvalues.get(path, {})
+        out: Dict[str, Any] = {}
+        for f in self.partition_schema().fields:
+            v = raw.get(f.name)
+            out[f.name] = (None if v is None
+                           else _cast_partition_value(v, f.type))
+        return out
+
+    def read_files(self, paths: List[str], columns, device):
+        from .parquet_source import partitioned_read_files
+        return partitioned_read_files(
+            self, paths, columns, device,
+            lambda p, c, d: super(DeltaTableRelation, self).read_files(
+                p, c, d))
+
+    def prune_partitions(self, cond):
+        from .parquet_source import prune_partitions_generic
+        return prune_partitions_generic(self, cond)
 
     def signature(self) -> str:
         return f"{self.snapshot_version}.{self.table.path}"

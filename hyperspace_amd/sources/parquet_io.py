@@ -492,27 +492,22 @@ def read_files_batch_device(paths: List[str], device,
                 # work (the host cost that dominated multi-row-group
                 # snappy decode; mixed dict+plain chunks appear when
                 # pyarrow's dictionary overflows mid-chunk)
-                dz_js = [j for j, p in enumerate(c.pages)
-                         if p[0] == "dict_z"]
-                bws = sorted({int(hb_all[regions[(j, "idx")][0]])
-                              for j in dz_js})
-                if c.pages and not nullable_chunk and len(bws) <= 1 \
-                        and (not bws or bws[0] > 0):
-                    starts, ends, nvs, outs, bshift = [], [], [], [], []
-                    dplaces = []  # (abs row, dict-compact row, nv)
+                page_bw = {j: int(hb_all[regions[(j, "idx")][0]])
+                           for j, p in enumerate(c.pages)
+                           if p[0] == "dict_z"}
+                if c.pages and not nullable_chunk and \
+                        all(bw > 0 for bw in page_bw.values()):
+                    # per-bw batches: pyarrow grows the index bit width
+                    # as the dictionary fills, so one chunk carries a
+                    # handful of widths — one C++ parse + one
+                    # rle_decode per width
+                    groups: Dict[int, list] = {}
                     row = written
-                    drows = 0
                     for j, page in enumerate(c.pages):
                         nv = page[3]
                         if page[0] == "dict_z":
-                            s0, rln, absbase = regions[(j, "idx")]
-                            starts.append(s0 + 1)
-                            ends.append(s0 + rln)
-                            nvs.append(nv)
-                            outs.append(drows)
-                            bshift.append((absbase - s0) * 8)
-                            dplaces.append((row, drows, nv))
-                            drows += nv
+                            groups.setdefault(page_bw[j], []).append(
+                                (j, row, nv))
                         else:  # plain_z
                             skip = lvl_skips[j] if page[5] else 0
                             ext.copy_unaligned(
@@ -520,16 +515,30 @@ def read_files_batch_device(paths: List[str], device,
                                 out[c.name], row * itemsize,
                                 nv * itemsize)
                         row += nv
-                    if dz_js:
-                        def t64(x):
-                            return _torch.tensor(x, dtype=_torch.int64)
+
+                    def t64(x):
+                        return _torch.tensor(x, dtype=_torch.int64)
+                    for bw, pages_g in groups.items():
+                        starts, ends, nvs, outs, bshift = \
+                            [], [], [], [], []
+                        dplaces = []
+                        drows = 0
+                        for j, abs_row, nv in pages_g:
+                            s0, rln, absbase = regions[(j, "idx")]
+                            starts.append(s0 + 1)
+                            ends.append(s0 + rln)
+                            nvs.append(nv)
+                            outs.append(drows)
+                            bshift.append((absbase - s0) * 8)
+                            dplaces.append((abs_row, drows, nv))
+                            drows += nv
                         kind, ooff, ln, boff, val, _cnt = \
                             ext.parse_rle_runs_batch(
                                 hb_all, t64(starts), t64(ends),
-                                t64([bws[0]] * len(starts)), t64(nvs),
+                                t64([bw] * len(starts)), t64(nvs),
                                 t64(outs), t64(bshift))
                         idx = ext.rle_decode(scratch, kind, ooff, ln,
-                                             boff, val, bws[0], drows)
+                                             boff, val, bw, drows)
                         vals = idx if z_is_str else ext.gather_rows(
                             dict_vals, idx.to(_torch.int64))
                         for abs_row, drow, nv in dplaces:

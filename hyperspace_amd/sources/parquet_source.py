@@ -162,73 +162,16 @@ class ParquetRelation(FileBasedRelation):
                 for f in pschema.fields if f.name in raw}
 
     def read_files(self, paths: List[str], columns, device):
-        pschema = self.partition_schema()
-        if not pschema.fields:
-            return super().read_files(paths, columns, device)
-        import torch
-        from ..execution.columnar import ColumnBatch, StringColumn
-        pnames = {f.name.lower(): f for f in pschema.fields}
-        file_cols = None
-        part_wanted = [f for f in pschema.fields]
-        if columns is not None:
-            file_cols = [c for c in columns if c.lower() not in pnames]
-            part_wanted = [pnames[c.lower()] for c in columns
-                           if c.lower() in pnames]
-        if file_cols == []:
-            # partition-only projection: row counts from footers, no IO
-            import pyarrow.parquet as pq
-            row_counts = [pq.ParquetFile(p).metadata.num_rows
-                          for p in paths]
-            batch = ColumnBatch({})
-        else:
-            batch, row_counts = super().read_files(paths, file_cols,
-                                                   device)
-        dev = batch.device if batch.columns else (
-            device if getattr(device, "type", "cpu") == "cuda"
-            else torch.device("cpu"))
-        cols = dict(batch.columns)
-        for f in part_wanted:
-            vals = [self.partition_values(p).get(f.name) for p in paths]
-            if f.type == "string":
-                uniq = sorted(set(vals))
-                code_of = {v: i for i, v in enumerate(uniq)}
-                codes = torch.cat([
-                    torch.full((rc,), code_of[v], dtype=torch.int32)
-                    for v, rc in zip(vals, row_counts)]) \
-                    if row_counts else torch.empty(0, dtype=torch.int32)
-                cols[f.name] = StringColumn(codes.to(dev), uniq)
-            else:
-                dt = torch.int64 if f.type == "long" else torch.float64
-                cols[f.name] = torch.cat([
-                    torch.full((rc,), v, dtype=dt)
-                    for v, rc in zip(vals, row_counts)]).to(dev) \
-                    if row_counts else torch.empty(0, dtype=dt)
-        out = ColumnBatch(cols, dict(batch.masks))
-        if columns is not None:
-            out = out.select(columns)
-        return out, row_counts
+        return partitioned_read_files(
+            self, paths, columns, device,
+            lambda p, c, d: super(ParquetRelation, self).read_files(
+                p, c, d))
 
     def prune_partitions(self, cond) -> Optional[List[str]]:
         """Files surviving the partition-column conjuncts of ``cond``,
         or None when the predicate has no partition conjunct (Spark's
         partition pruning: evaluated on metadata, before any IO)."""
-        pschema = self.partition_schema()
-        if not pschema.fields:
-            return None
-        from ..plan.expr import split_conjunctive
-        pnames = {f.name.lower() for f in pschema.fields}
-        conjuncts = [c for c in split_conjunctive(cond)
-                     if c.references() and
-                     {r.lower() for r in c.references()} <= pnames]
-        if not conjuncts:
-            return None
-        kept = []
-        for f in self.all_files():
-            vals = {k.lower(): v
-                    for k, v in self.partition_values(f.name).items()}
-            if all(_eval_on_values(c, vals) for c in conjuncts):
-                kept.append(f.name)
-        return kept
+        return prune_partitions_generic(self, cond)
 
     def all_files(self) -> List[FileInfo]:
         # re-listed on every call (like Spark's InMemoryFileIndex refresh
@@ -249,6 +192,81 @@ class ParquetRelation(FileBasedRelation):
 
     def refreshed(self) -> "ParquetRelation":
         return ParquetRelation(self._root_paths, self._options, None)
+
+
+def partitioned_read_files(relation, paths: List[str], columns, device,
+                           base_read):
+    """Generic partitioned read: file columns come from ``base_read``,
+    partition columns materialize as per-file constants (Spark attaches
+    partition values the same way).  Shared by the hive-style default
+    source and the Delta source (partitionValues from the action log)."""
+    pschema = relation.partition_schema()
+    if not pschema.fields:
+        return base_read(paths, columns, device)
+    import torch
+    from ..execution.columnar import ColumnBatch, StringColumn
+    pnames = {f.name.lower(): f for f in pschema.fields}
+    file_cols = None
+    part_wanted = [f for f in pschema.fields]
+    if columns is not None:
+        file_cols = [c for c in columns if c.lower() not in pnames]
+        part_wanted = [pnames[c.lower()] for c in columns
+                       if c.lower() in pnames]
+    if file_cols == []:
+        # partition-only projection: row counts from footers, no IO
+        import pyarrow.parquet as pq
+        row_counts = [pq.ParquetFile(p).metadata.num_rows
+                      for p in paths]
+        batch = ColumnBatch({})
+    else:
+        batch, row_counts = base_read(paths, file_cols, device)
+    dev = batch.device if batch.columns else (
+        device if getattr(device, "type", "cpu") == "cuda"
+        else torch.device("cpu"))
+    cols = dict(batch.columns)
+    for f in part_wanted:
+        vals = [relation.partition_values(p).get(f.name) for p in paths]
+        if f.type == "string":
+            uniq = sorted(set(vals))
+            code_of = {v: i for i, v in enumerate(uniq)}
+            codes = torch.cat([
+                torch.full((rc,), code_of[v], dtype=torch.int32)
+                for v, rc in zip(vals, row_counts)]) \
+                if row_counts else torch.empty(0, dtype=torch.int32)
+            cols[f.name] = StringColumn(codes.to(dev), uniq)
+        else:
+            dt = torch.int64 if f.type in ("long", "integer", "date") \
+                else torch.float64
+            cols[f.name] = torch.cat([
+                torch.full((rc,), v, dtype=dt)
+                for v, rc in zip(vals, row_counts)]).to(dev) \
+                if row_counts else torch.empty(0, dtype=dt)
+    out = ColumnBatch(cols, dict(batch.masks))
+    if columns is not None:
+        out = out.select(columns)
+    return out, row_counts
+
+
+def prune_partitions_generic(relation, cond) -> Optional[List[str]]:
+    """Partition pruning over any relation exposing partition_schema()
+    and partition_values(path)."""
+    pschema = relation.partition_schema()
+    if not pschema.fields:
+        return None
+    from ..plan.expr import split_conjunctive
+    pnames = {f.name.lower() for f in pschema.fields}
+    conjuncts = [c for c in split_conjunctive(cond)
+                 if c.references() and
+                 {r.lower() for r in c.references()} <= pnames]
+    if not conjuncts:
+        return None
+    kept = []
+    for f in relation.all_files():
+        vals = {k.lower(): v
+                for k, v in relation.partition_values(f.name).items()}
+        if all(_eval_on_values(c, vals) for c in conjuncts):
+            kept.append(f.name)
+    return kept
 
 
 def _eval_on_values(e, vals: Dict[str, Any]) -> bool:
