@@ -751,6 +751,8 @@ def read_files_batch_device(paths: List[str], device,
     def decode_unit(u):
         uid, (i, row_off, chunks) = u
         buf, size, _ = infos[i]
+        if _dt == "2":
+            t0u = _time.perf_counter()
         stream = streams[uid % n_streams]
         with _torch.cuda.stream(stream):
             with upload_lock:
@@ -765,6 +767,10 @@ def read_files_batch_device(paths: List[str], device,
                 else:
                     stream.wait_event(upload_events[i])
             _decode_on_stream(i, buf, size, chunks, dev_bufs[i], row_off)
+        if _dt == "2":
+            print(f"[hs-unit] u{uid} rows={chunks[0].num_values} "
+                  f"chunks={[(c.name, c.encoding, len(c.pages)) for c in chunks]} "
+                  f"{_time.perf_counter()-t0u:.3f}s", file=sys.stderr)
         return True
 
     _t1 = _time.perf_counter()

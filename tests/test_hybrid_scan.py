@@ -140,3 +140,97 @@ def test_too_much_appended_blocks_hybrid(env):
     plan = df.filter("key = 77").select("key", "val").optimized_plan()
     assert not any(isinstance(l, IndexScan)
                    for l in plan.collect_leaves())
+
+
+# ---------------------------------------------------------------------------
+# Hybrid Scan over transactional table formats (reference:
+# HybridScanForDeltaLakeTest / HybridScanForIcebergTest — append and
+# delete through the TABLE LOG, then assert plan shape and results)
+# ---------------------------------------------------------------------------
+
+def _tbatch(rng, k_hi=1000, n=N):
+    import torch
+    return __import__("hyperspace_amd").execution.columnar.ColumnBatch({
+        "key": torch.from_numpy(rng.integers(0, k_hi, n)),
+        "val": torch.from_numpy(rng.random(n))})
+
+
+def test_hybrid_scan_delta_append_and_delete(tmp_path, monkeypatch):
+    from hyperspace_amd.execution.columnar import ColumnBatch
+    import torch
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "idx"))
+    from hyperspace_amd.sources.delta_source import DeltaTable
+    rng = np.random.default_rng(41)
+    t = DeltaTable.create(str(tmp_path / "t"))
+    for _ in range(4):
+        t.append_batch(ColumnBatch({
+            "key": torch.from_numpy(rng.integers(0, 1000, N)),
+            "val": torch.from_numpy(rng.random(N))}))
+    session = hs.HyperspaceSession(device="cpu")
+    session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 8)
+    session.conf.set(hs.IndexConstants.INDEX_LINEAGE_ENABLED, True)
+    h = hs.Hyperspace(session)
+    df = session.read_delta(str(tmp_path / "t"))
+    h.create_index(df, hs.CoveringIndexConfig("hdx", ["key"], ["val"]))
+
+    # append one more file + delete one original THROUGH THE DELTA LOG
+    t.append_batch(ColumnBatch({
+        "key": torch.from_numpy(rng.integers(0, 1000, 1000)),
+        "val": torch.from_numpy(rng.random(1000))}))
+    victim = t.files_at()[0].name
+    t.remove_files([victim])
+
+    session.enable_hyperspace()
+    session.conf.set(hs.IndexConstants.INDEX_HYBRID_SCAN_ENABLED, True)
+    live = session.read_delta(str(tmp_path / "t"))
+    q = live.filter("key = 77").select("key", "val")
+    plan = q.optimized_plan()
+    # hybrid shape: index scan + appended-file scan under a Union
+    assert any(isinstance(n, UnionNode)
+               for n in _walk(plan)), plan.pretty()
+    assert any(isinstance(l, IndexScan) for l in plan.collect_leaves())
+    out = q.collect()
+    session.disable_hyperspace()
+    assert _rows(out, ["key", "val"]) == _rows(q.collect(),
+                                               ["key", "val"])
+
+
+def test_hybrid_scan_iceberg_append(tmp_path, monkeypatch):
+    from hyperspace_amd.execution.columnar import ColumnBatch
+    import torch
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "idx"))
+    from hyperspace_amd.sources.iceberg_source import IcebergTable
+    rng = np.random.default_rng(43)
+    t = IcebergTable.create(str(tmp_path / "it"))
+    for _ in range(4):
+        t.append_batch(ColumnBatch({
+            "key": torch.from_numpy(rng.integers(0, 1000, N)),
+            "val": torch.from_numpy(rng.random(N))}))
+    session = hs.HyperspaceSession(device="cpu")
+    session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 8)
+    session.conf.set(hs.IndexConstants.INDEX_LINEAGE_ENABLED, True)
+    h = hs.Hyperspace(session)
+    df = session.read_iceberg(str(tmp_path / "it"))
+    h.create_index(df, hs.CoveringIndexConfig("hix", ["key"], ["val"]))
+
+    t.append_batch(ColumnBatch({
+        "key": torch.from_numpy(rng.integers(0, 1000, 1000)),
+        "val": torch.from_numpy(rng.random(1000))}))
+    session.enable_hyperspace()
+    session.conf.set(hs.IndexConstants.INDEX_HYBRID_SCAN_ENABLED, True)
+    live = session.read_iceberg(str(tmp_path / "it"))
+    q = live.filter("key = 77").select("key", "val")
+    plan = q.optimized_plan()
+    assert any(isinstance(n, UnionNode)
+               for n in _walk(plan)), plan.pretty()
+    out = q.collect()
+    session.disable_hyperspace()
+    assert _rows(out, ["key", "val"]) == _rows(q.collect(),
+                                               ["key", "val"])
+
+
+def _walk(plan):
+    out = [plan]
+    for c in getattr(plan, "children", []):
+        out.extend(_walk(c))
+    return out
