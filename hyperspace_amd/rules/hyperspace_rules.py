@@ -92,8 +92,11 @@ def _hybrid_appended_scan(cand: Candidate, scan: Scan,
     appended = list(cand.tags.get(TAG_APPENDED_FILES, []))
     if not appended:
         return None
-    sub_relation = scan.relation.__class__(appended, scan.relation.options)
-    return Project(columns, Scan(sub_relation))
+    # scan the SAME relation restricted to the appended files: partition
+    # materialization and table-format read paths stay intact for
+    # hive/delta/iceberg sources alike
+    return Project(columns, Scan(scan.relation, scan.options,
+                                 file_subset=appended))
 
 
 class FilterIndexRule(HyperspaceRule):

@@ -48,6 +48,17 @@ class FileBasedRelation(ABC):
             return read_files_batch_device(paths, device, columns)
         return read_files_batch(paths, columns)
 
+    def file_subset_relation(self, paths: List[str]
+                             ) -> "FileBasedRelation":
+        """Relation over a SUBSET of this relation's data files — the
+        Hybrid Scan appended-file scan (reference reads appended files
+        via the provider's internalFileFormatName; for Delta/Iceberg
+        the underlying files are plain parquet).  Sources with
+        partition metadata override to keep partition-column
+        materialization working on the subset."""
+        from .parquet_source import ParquetRelation
+        return ParquetRelation(paths, dict(self.options))
+
     def describe(self) -> str:
         return f"{self.file_format}:{','.join(self.root_paths)}"
 

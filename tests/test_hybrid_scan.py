@@ -162,7 +162,9 @@ def test_hybrid_scan_delta_append_and_delete(tmp_path, monkeypatch):
     from hyperspace_amd.sources.delta_source import DeltaTable
     rng = np.random.default_rng(41)
     t = DeltaTable.create(str(tmp_path / "t"))
-    for _ in range(4):
+    # 8 base files so deleting one stays under the 0.2 deleted-ratio
+    # ceiling (reference maxDeletedRatio semantics)
+    for _ in range(8):
         t.append_batch(ColumnBatch({
             "key": torch.from_numpy(rng.integers(0, 1000, N)),
             "val": torch.from_numpy(rng.random(N))}))
