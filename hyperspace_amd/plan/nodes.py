@@ -80,8 +80,12 @@ class Scan(LogicalPlan):
         return self
 
     def _node_str(self):
-        extra = (f", dataskipping:-{self.skipped_files}files"
-                 if self.file_subset is not None else "")
+        if self.file_subset is None:
+            extra = ""
+        elif self.skipped_files:
+            extra = f", dataskipping:-{self.skipped_files}files"
+        else:
+            extra = f", files:{len(self.file_subset)}"
         return f"Scan({self.relation.describe()}{extra})"
 
 
