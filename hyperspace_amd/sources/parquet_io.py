@@ -393,6 +393,8 @@ def read_files_batch_device(paths: List[str], device,
         for c in chunks:
             itemsize = c.np_dtype.itemsize
             written = cursors[c.name]
+            if _dt == "3":
+                t_c0 = _time.perf_counter()
             if c.encoding in ("plain_z", "dict_z"):
                 # snappy chunk (K1): one wave per page decompresses into
                 # a scratch buffer; PLAIN pages copy straight from
@@ -483,8 +485,12 @@ def read_files_batch_device(paths: List[str], device,
                         regions[(j, tag)] = (cur, b - a, a)
                         cat_parts.append(scratch[a:b])
                         cur += b - a
+                if _dt == "3":
+                    tA = _time.perf_counter()
                 hb_all = (_torch.cat(cat_parts).cpu() if cat_parts
                           else None)
+                if _dt == "3":
+                    tB = _time.perf_counter()
                 # fast path for null-free chunks: ONE GIL-released C++
                 # run parse over every dictionary page (uniform bit
                 # width) + ONE rle_decode, with PLAIN pages streaming
@@ -546,6 +552,11 @@ def read_files_batch_device(paths: List[str], device,
                                 vals[drow:drow + nv]
                     written = row
                     cursors[c.name] = written
+                    if _dt == "3":
+                        print(f"[hs-chunk] {c.name} pages={len(c.pages)}"
+                              f" pre={tA-t_c0:.3f} d2h={tB-tA:.3f}"
+                              f" fast={_time.perf_counter()-tB:.3f}",
+                              file=sys.stderr)
                     continue
                 # batch unmasked dict_z pages: one rle_decode + gather
                 # per (chunk, bit-width) instead of per page
