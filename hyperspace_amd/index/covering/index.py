@@ -496,6 +496,33 @@ def write_bucketed(batch: ColumnBatch, seg: torch.Tensor, out_dir: str,
     from ...execution.columnar import StringColumn
     from ...sources.parquet_io import _pinned_get, _pinned_put
     on_device = batch.device.type == "cuda"
+    if on_device and os.environ.get("HS_WRITE_V2"):
+        # A/B variant: ONE bulk pinned D2H per column, then the bucket
+        # files slice host memory (no per-bucket device copies)
+        if after_event is not None:
+            torch.cuda.current_stream().wait_event(after_event)
+        host_cols = {}
+        host_masks = {}
+        held_bufs = []
+        for name, col in batch.columns.items():
+            if isinstance(col, StringColumn):
+                host_cols[name] = col.to("cpu")
+                continue
+            nb = col.numel() * col.element_size()
+            bufp = _pinned_get(nb)
+            host = bufp[:nb].view(col.dtype)
+            host.copy_(col, non_blocking=True)
+            host_cols[name] = host
+            held_bufs.append(bufp)
+        for name, m in batch.masks.items():
+            host_masks[name] = m.to("cpu")
+        torch.cuda.current_stream().synchronize()
+        host_batch = ColumnBatch(host_cols, host_masks)
+        out = write_bucketed(host_batch, seg, out_dir, num_buckets,
+                             task_id)
+        for bufp in held_bufs:
+            _pinned_put(bufp)
+        return out
     jobs = []
     for b in range(num_buckets):
         lo, hi = int(seg[b]), int(seg[b + 1])

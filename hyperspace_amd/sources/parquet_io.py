@@ -450,6 +450,104 @@ def read_files_batch_device(paths: List[str], device,
                     ext.copy_unaligned(scratch, 0, dict_vals, 0,
                                        dict_n * itemsize)
                     dict_vals = dict_vals[:dict_n].contiguous()
+
+                nullable_chunk = c.name in out_masks
+                if not nullable_chunk:
+                    # FAST PATH (no nulls): one cat + ONE D2H sync per
+                    # chunk; all page headers (level-length prefixes,
+                    # bit widths) are then read from HOST memory and the
+                    # dictionary-index runs of every page parse in one
+                    # GIL-released C++ call per bit width.  The round-1
+                    # shape (per-page torch slicing + a level-prefix
+                    # stack().cpu() sync per chunk) serialized 26-unit
+                    # decodes on the GIL.
+                    parts = []
+                    roff = []  # page -> offset of its span in hb
+                    cur = 0
+                    for j, page in enumerate(c.pages):
+                        base = int(page_base[j])
+                        if page[0] == "dict_z":
+                            ln_span = int(page[4])
+                        elif page[5]:
+                            ln_span = 4  # level-length prefix only
+                        else:
+                            roff.append(-1)
+                            continue
+                        roff.append(cur)
+                        parts.append(scratch[base:base + ln_span])
+                        cur += ln_span
+                    hb = (_torch.cat(parts).cpu().numpy() if parts
+                          else np.empty(0, dtype=np.uint8))
+                    if _dt == "3":
+                        tA = tB = _time.perf_counter()
+
+                    groups: Dict[int, list] = {}
+                    row = written
+                    ok = True
+                    for j, page in enumerate(c.pages):
+                        nv = page[3]
+                        base = int(page_base[j])
+                        r0 = roff[j]
+                        skip = 0
+                        if page[5]:
+                            ln = int(np.frombuffer(hb, "<u4", 1, r0)[0])
+                            skip = 4 + ln
+                        if page[0] == "dict_z":
+                            bw = int(hb[r0 + skip])
+                            if bw <= 0 or skip + 1 >= int(page[4]):
+                                ok = False
+                                break
+                            groups.setdefault(bw, []).append(
+                                (j, row, nv, r0 + skip + 1,
+                                 r0 + int(page[4]), (base - r0) * 8))
+                        else:
+                            ext.copy_unaligned(
+                                scratch, base + skip, out[c.name],
+                                row * itemsize, nv * itemsize)
+                        row += nv
+                    if ok:
+                        hb_t = _torch.from_numpy(hb)
+
+                        def t64(x):
+                            return _torch.tensor(x, dtype=_torch.int64)
+                        for bw, pages_g in groups.items():
+                            starts = [g[3] for g in pages_g]
+                            ends = [g[4] for g in pages_g]
+                            nvs = [g[2] for g in pages_g]
+                            bshift = [g[5] for g in pages_g]
+                            outs = []
+                            drows = 0
+                            dplaces = []
+                            for _, abs_row, nv, _, _, _ in pages_g:
+                                outs.append(drows)
+                                dplaces.append((abs_row, drows, nv))
+                                drows += nv
+                            kind, ooff, ln_t, boff, val, _cnt = \
+                                ext.parse_rle_runs_batch(
+                                    hb_t, t64(starts), t64(ends),
+                                    t64([bw] * len(starts)), t64(nvs),
+                                    t64(outs), t64(bshift))
+                            idx = ext.rle_decode(scratch, kind, ooff,
+                                                 ln_t, boff, val, bw,
+                                                 drows)
+                            vals = idx if z_is_str else \
+                                ext.gather_rows(dict_vals,
+                                                idx.to(_torch.int64))
+                            for abs_row, drow, nv in dplaces:
+                                out[c.name][abs_row:abs_row + nv] = \
+                                    vals[drow:drow + nv]
+                        written = row
+                        cursors[c.name] = written
+                        if _dt == "3":
+                            print(f"[hs-chunk] {c.name} "
+                                  f"pages={len(c.pages)} "
+                                  f"pre={tA-t_c0:.3f} "
+                                  f"fast={_time.perf_counter()-tB:.3f}",
+                                  file=sys.stderr)
+                        continue
+                    # rare malformed/empty-dict page: fall through to
+                    # the general per-page path below
+
                 # level-length prefixes for every leveled page in one
                 # small D2H (4 bytes each); the index streams and level
                 # bytes D2H per page below — NEVER the whole chunk
@@ -464,11 +562,7 @@ def read_files_batch_device(paths: List[str], device,
                     lens = pref.numpy().view("<u4").ravel()
                     for j, ln in zip(with_lvl, lens):
                         lvl_skips[j] = 4 + int(ln)
-                nullable_chunk = c.name in out_masks
-                # ONE D2H for every byte the host needs (index streams +
-                # nullable level bytes): per-page .cpu() calls each sync
-                # the stream and serialize the whole decode
-                regions = {}  # j -> (start in cat, length, abs base)
+                regions = {}  # (j, tag) -> (start in cat, length, abs)
                 cat_parts = []
                 cur = 0
                 for j, page in enumerate(c.pages):
@@ -481,83 +575,12 @@ def read_files_batch_device(paths: List[str], device,
                         skip_j = lvl_skips[j] if page[5] else 0
                         spans.append(("idx", base + skip_j,
                                       base + int(page[4])))
-                    for tag, a, b in spans:
-                        regions[(j, tag)] = (cur, b - a, a)
-                        cat_parts.append(scratch[a:b])
-                        cur += b - a
-                if _dt == "3":
-                    tA = _time.perf_counter()
+                    for tag, a2, b2 in spans:
+                        regions[(j, tag)] = (cur, b2 - a2, a2)
+                        cat_parts.append(scratch[a2:b2])
+                        cur += b2 - a2
                 hb_all = (_torch.cat(cat_parts).cpu() if cat_parts
                           else None)
-                if _dt == "3":
-                    tB = _time.perf_counter()
-                # fast path for null-free chunks: ONE GIL-released C++
-                # run parse over every dictionary page (uniform bit
-                # width) + ONE rle_decode, with PLAIN pages streaming
-                # through copy_unaligned — instead of per-page python
-                # work (the host cost that dominated multi-row-group
-                # snappy decode; mixed dict+plain chunks appear when
-                # pyarrow's dictionary overflows mid-chunk)
-                page_bw = {j: int(hb_all[regions[(j, "idx")][0]])
-                           for j, p in enumerate(c.pages)
-                           if p[0] == "dict_z"}
-                if c.pages and not nullable_chunk and \
-                        all(bw > 0 for bw in page_bw.values()):
-                    # per-bw batches: pyarrow grows the index bit width
-                    # as the dictionary fills, so one chunk carries a
-                    # handful of widths — one C++ parse + one
-                    # rle_decode per width
-                    groups: Dict[int, list] = {}
-                    row = written
-                    for j, page in enumerate(c.pages):
-                        nv = page[3]
-                        if page[0] == "dict_z":
-                            groups.setdefault(page_bw[j], []).append(
-                                (j, row, nv))
-                        else:  # plain_z
-                            skip = lvl_skips[j] if page[5] else 0
-                            ext.copy_unaligned(
-                                scratch, int(page_base[j]) + skip,
-                                out[c.name], row * itemsize,
-                                nv * itemsize)
-                        row += nv
-
-                    def t64(x):
-                        return _torch.tensor(x, dtype=_torch.int64)
-                    for bw, pages_g in groups.items():
-                        starts, ends, nvs, outs, bshift = \
-                            [], [], [], [], []
-                        dplaces = []
-                        drows = 0
-                        for j, abs_row, nv in pages_g:
-                            s0, rln, absbase = regions[(j, "idx")]
-                            starts.append(s0 + 1)
-                            ends.append(s0 + rln)
-                            nvs.append(nv)
-                            outs.append(drows)
-                            bshift.append((absbase - s0) * 8)
-                            dplaces.append((abs_row, drows, nv))
-                            drows += nv
-                        kind, ooff, ln, boff, val, _cnt = \
-                            ext.parse_rle_runs_batch(
-                                hb_all, t64(starts), t64(ends),
-                                t64([bw] * len(starts)), t64(nvs),
-                                t64(outs), t64(bshift))
-                        idx = ext.rle_decode(scratch, kind, ooff, ln,
-                                             boff, val, bw, drows)
-                        vals = idx if z_is_str else ext.gather_rows(
-                            dict_vals, idx.to(_torch.int64))
-                        for abs_row, drow, nv in dplaces:
-                            out[c.name][abs_row:abs_row + nv] = \
-                                vals[drow:drow + nv]
-                    written = row
-                    cursors[c.name] = written
-                    if _dt == "3":
-                        print(f"[hs-chunk] {c.name} pages={len(c.pages)}"
-                              f" pre={tA-t_c0:.3f} d2h={tB-tA:.3f}"
-                              f" fast={_time.perf_counter()-tB:.3f}",
-                              file=sys.stderr)
-                    continue
                 # batch unmasked dict_z pages: one rle_decode + gather
                 # per (chunk, bit-width) instead of per page
                 zbatch: Dict[int, list] = {}
@@ -802,7 +825,10 @@ def read_files_batch_device(paths: List[str], device,
         print(f"[hs-decode] files={len(paths)} units={len(units)} "
               f"read+layout {_t1-_t0:.3f}s decode-host {_t2-_t1:.3f}s "
               f"sync {_time.perf_counter()-_t2:.3f}s", file=sys.stderr)
-    snappy_bad = any(bool((s != 0).any()) for s in statuses)
+    # one device reduction + sync for every chunk's status words (a
+    # per-tensor .any() paid ~1.4 ms of sync each across ~50 chunks)
+    snappy_bad = bool(_torch.cat(statuses).ne(0).any().item()) \
+        if statuses else False
     if snappy_bad or \
             not all(b is not None and b is not False for b in bufs):
         for b in bufs:
