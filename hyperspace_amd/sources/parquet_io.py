@@ -808,8 +808,9 @@ def read_files_batch_device(paths: List[str], device,
         return True
 
     _t1 = _time.perf_counter()
-    if len(units) > 2:
-        with ThreadPoolExecutor(max_workers=16) as pool:
+    _dw = int(os.environ.get("HS_DECODE_WORKERS", "16"))
+    if len(units) > 2 and _dw > 1:
+        with ThreadPoolExecutor(max_workers=_dw) as pool:
             list(pool.map(decode_unit, enumerate(units)))
     else:
         for u in enumerate(units):
