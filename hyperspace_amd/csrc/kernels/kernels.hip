@@ -1330,8 +1330,7 @@ void copy_unaligned(const uint8_t* src, int64_t src_off, uint8_t* dst,
 // Register-window stream cursor: the tag walk is inherently serial, so
 // its cost is per-op DEPENDENT loads.  A 16-byte aligned window turns
 // ~3-6 dependent byte loads per op into one aligned u64 load per 8
-// stream bytes (the round-1 kernel was latency-bound on exactly those
-// byte loads at ~1.8 GB/s).
+// stream bytes.
 struct SnapCursor {
   const uint8_t* s0;    // aligned-down stream base
   const uint8_t* send;  // one past last valid byte
@@ -1386,17 +1385,6 @@ struct SnapCursor {
   }
 };
 
-// LDS output ring: snappy back-references reach at most 64 KiB (the
-// reference compressor fragments input at 64 KiB), so the whole
-// copy-resolution working set fits a 72 KiB LDS ring.  Copies become
-// ds ops gated by cheap lgkmcnt waits instead of global-memory L1
-// round-trips gated by vmcnt(0) drains, and output reaches HBM as
-// 8 KiB coalesced segment flushes.  Ring size is a multiple of 8 so a
-// ring position is ≡ out (mod 8) — aligning `out` aligns the LDS slot.
-constexpr int SNAP_RING = 72 * 1024;
-constexpr int SNAP_SEG = 8 * 1024;
-constexpr int64_t SNAP_MAX_OFF = 64 * 1024;
-
 __global__ void k_snappy_decomp(const uint8_t* __restrict__ src,
                                 const int64_t* __restrict__ src_off,
                                 const int64_t* __restrict__ src_end,
@@ -1404,7 +1392,6 @@ __global__ void k_snappy_decomp(const uint8_t* __restrict__ src,
                                 const int64_t* __restrict__ dst_off,
                                 const int64_t* __restrict__ dst_len,
                                 int32_t* __restrict__ status) {
-  __shared__ uint8_t ring[SNAP_RING];
   int p = blockIdx.x;
   const uint8_t* sbeg = src + src_off[p];
   const uint8_t* send = src + src_end[p];
@@ -1431,29 +1418,7 @@ __global__ void k_snappy_decomp(const uint8_t* __restrict__ src,
   }
 
   int64_t out = 0;
-  int64_t flushed = 0;  // bytes of d[] already written to global
-
-  // flush whole 8 KiB segments [flushed, target) ring -> global.
-  // Segment ring ranges are contiguous (SNAP_SEG divides SNAP_RING).
-  auto flush_to = [&](int64_t target) {
-    while (flushed + SNAP_SEG <= target) {
-      int rbase = (int)(flushed % SNAP_RING);
-      uint8_t* g = d + flushed;
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      if (((uintptr_t)g & 7) == 0) {
-        // 128 B per lane, u64 granules (both sides 8-aligned)
-        for (int i = lane; i * 8 < SNAP_SEG; i += 64)
-          *(uint64_t*)(g + i * 8) = *(const uint64_t*)(ring + rbase + i * 8);
-      } else {
-        for (int i = lane; i * 8 < SNAP_SEG; i += 64) {
-          uint64_t v = *(const uint64_t*)(ring + rbase + i * 8);
-          __builtin_memcpy(g + i * 8, &v, 8);
-        }
-      }
-      flushed += SNAP_SEG;
-    }
-  };
-
+  int64_t synced = 0;  // output below this point is visible to all lanes
   while (out < expected && !c.exhausted()) {
     uint64_t w = c.peek();
     uint8_t tag = (uint8_t)w;
@@ -1493,107 +1458,71 @@ __global__ void k_snappy_decomp(const uint8_t* __restrict__ src,
       if (lane == 0) status[p] = 3;
       return;
     }
-    // ring-overwrite guard: writing len bytes at slots [out, out+len)
-    // recycles the slots of bytes [out+len-RING, ...) — those must be
-    // durable in global first (target <= out whenever len <= 64 KiB)
-    if (len <= SNAP_MAX_OFF) {
-      int64_t need = out + len - SNAP_RING + SNAP_SEG;
-      if (flushed < need) flush_to(need);
-    }
     if (k == 0) {
-      if (len > SNAP_MAX_OFF) {
-        // giant literal (incompressible page): only its last 64 KiB
-        // can ever be a copy source — stream the middle straight to
-        // global, route the tail through the ring
-        int64_t direct = len - SNAP_MAX_OFF;
-        flush_to(out);  // drain ring below out first
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        // partial head of the current segment (out may not be on a
-        // segment boundary): bytes [flushed, out) are still ring-only;
-        // write them out too so `flushed` can jump forward
-        for (int64_t i = flushed + lane; i < out; i += 64)
-          d[i] = ring[i % SNAP_RING];
-        uint8_t* g = d + out;
-        int64_t head = (16 - ((uintptr_t)g & 15)) & 15;
-        if (head > direct) head = direct;
-        int64_t body = (direct - head) & ~15ll;
-        for (int64_t i = lane; i < head; i += 64) g[i] = lit[i];
-        if (lit + head + body + 16 <= send) {
+      // wide literal copy: 16B lanes over the dest-aligned middle
+      // (incompressible parquet pages — e.g. random doubles — arrive
+      // as ONE giant literal, so this path carries ~half the bytes)
+      if (len >= 256) {
+        int64_t head = (16 - ((uintptr_t)(d + out) & 15)) & 15;
+        int64_t body = (len - head) & ~15ll;
+        for (int64_t i = lane; i < head; i += 64) d[out + i] = lit[i];
+        const uint8_t* lsrc = lit + head;
+        uint8_t* ldst = d + out + head;
+        if (lsrc + body + 16 <= send) {
           for (int64_t i = lane; i * 16 < body; i += 64) {
             uint4 v;
-            __builtin_memcpy(&v, lit + head + i * 16, 16);
-            *(uint4*)(g + head + i * 16) = v;
+            __builtin_memcpy(&v, lsrc + i * 16, 16);
+            *(uint4*)(ldst + i * 16) = v;
           }
         } else {
-          for (int64_t i = lane; i < body; i += 64)
-            g[head + i] = lit[head + i];
-        }
-        for (int64_t i = head + body + lane; i < direct; i += 64)
-          g[i] = lit[i];
-        // tail through the ring
-        for (int64_t i = lane; i < SNAP_MAX_OFF; i += 64)
-          ring[(out + direct + i) % SNAP_RING] = lit[direct + i];
-        out += len;
-        flushed = out - SNAP_MAX_OFF;
-        continue;
-      }
-      // normal literal: global -> ring; align `out` so the 8-byte
-      // middle hits 8-aligned ring slots (ring pos ≡ out mod 8)
-      if (len >= 64) {
-        int64_t head = (8 - (out & 7)) & 7;
-        int64_t body = (len - head) & ~7ll;
-        for (int64_t i = lane; i < head; i += 64)
-          ring[(out + i) % SNAP_RING] = lit[i];
-        if (lit + head + body + 8 <= send) {
-          for (int64_t i = lane; i * 8 < body; i += 64) {
-            uint64_t v;
-            __builtin_memcpy(&v, lit + head + i * 8, 8);
-            *(uint64_t*)(ring + (out + head + i * 8) % SNAP_RING) = v;
-          }
-        } else {
-          for (int64_t i = lane; i < body; i += 64)
-            ring[(out + head + i) % SNAP_RING] = lit[head + i];
+          for (int64_t i = lane; i < body; i += 64) ldst[i] = lsrc[i];
         }
         for (int64_t i = head + body + lane; i < len; i += 64)
-          ring[(out + i) % SNAP_RING] = lit[i];
+          d[out + i] = lit[i];
       } else {
-        for (int64_t i = lane; i < len; i += 64)
-          ring[(out + i) % SNAP_RING] = lit[i];
-      }
-    } else if (off <= SNAP_MAX_OFF) {
-      // in-ring copy: wait for pending ds writes once, then move bytes
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      if (off >= len) {
-        for (int64_t i = lane; i < len; i += 64)
-          ring[(out + i) % SNAP_RING] = ring[(out - off + i) % SNAP_RING];
-      } else {
-        for (int64_t i = lane; i < len; i += 64)
-          ring[(out + i) % SNAP_RING] =
-              ring[(out - off + (i % off)) % SNAP_RING];
+        for (int64_t i = lane; i < len; i += 64) d[out + i] = lit[i];
       }
     } else {
-      // off > 64 KiB (nonstandard stream): source is already flushed
-      // to global (flush policy keeps flushed >= out - 64K)
-      flush_to(out);
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      for (int64_t i = flushed + lane; i < out; i += 64)
-        d[i] = ring[i % SNAP_RING];
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      for (int64_t i = lane; i < len; i += 64) {
-        int64_t sidx = out - off + (off >= len ? i : (i % off));
-        uint8_t v = d[sidx];
-        ring[(out + i) % SNAP_RING] = v;
+      // drain this wave's pending stores only when the source range
+      // reaches past the last drain point (could include other lanes'
+      // recent bytes).  One wave per block: a full barrier is not
+      // needed — s_waitcnt vmcnt(0) makes every lane's stores visible
+      // through the CU's L1 to the wave's subsequent loads.
+      if (out - off + len > synced) {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        synced = out;
+      }
+      if (off >= len) {
+        if (len >= 256 && off >= 8) {
+          // long disjoint copy: 8B-granule, dest-aligned middle (short
+          // copies — the common case — stay on the 1-instruction
+          // byte path; the wide path only pays off past ~4 B/lane)
+          int64_t head = (8 - ((uintptr_t)(d + out) & 7)) & 7;
+          if (head > len) head = len;
+          int64_t body = (len - head) & ~7ll;
+          for (int64_t i = lane; i < head; i += 64)
+            d[out + i] = d[out - off + i];
+          const uint8_t* csrc = d + out - off + head;
+          uint8_t* cdst = d + out + head;
+          for (int64_t i = lane; i * 8 < body; i += 64) {
+            uint64_t v;
+            __builtin_memcpy(&v, csrc + i * 8, 8);
+            *(uint64_t*)(cdst + i * 8) = v;
+          }
+          for (int64_t i = head + body + lane; i < len; i += 64)
+            d[out + i] = d[out - off + i];
+        } else {
+          for (int64_t i = lane; i < len; i += 64)
+            d[out + i] = d[out - off + i];
+        }
+      } else {
+        for (int64_t i = lane; i < len; i += 64)
+          d[out + i] = d[out - off + (i % off)];
       }
     }
     out += len;
   }
   if (lane == 0) status[p] = out == expected ? 0 : 4;
-  if (out != expected) return;
-  // final flush: whole segments, then the partial tail
-  flush_to(out);
-  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-  for (int64_t i = flushed + lane; i < out; i += 64)
-    d[i] = ring[i % SNAP_RING];
 }
 
 void snappy_decompress_pages(const uint8_t* src, const int64_t* src_off,

@@ -412,16 +412,54 @@ def read_files_batch_device(paths: List[str], device,
                 doff_all = np.concatenate([[0], np.cumsum(uncs_all)])
                 scratch = _torch.empty(int(doff_all[-1]) + 4,
                                        dtype=_torch.uint8, device=device)
-                st = ext.snappy_decompress(
-                    dev_bytes,
-                    _torch.tensor([s[0] for s in segs],
-                                  dtype=_torch.int64),
-                    _torch.tensor([s[1] for s in segs],
-                                  dtype=_torch.int64),
-                    scratch,
-                    _torch.tensor(doff_all[:-1], dtype=_torch.int64),
-                    _torch.tensor(uncs_all, dtype=_torch.int64))
-                statuses.append(st)
+                # latency/bandwidth split: a page that actually
+                # COMPRESSED is copy-dense — a serial ~150-500 ns/op
+                # chain on a single wave (a 1 MB dictionary page of
+                # small-valued int64s measured 137 ms as 264k tiny ops,
+                # profiles/r2_summary.md) — so it decodes on the HOST
+                # C++ codec (~GB/s per worker thread) and uploads;
+                # near-incompressible pages (giant literals) stay on
+                # device where decode is a bandwidth-bound copy.
+                dev_idx = [i for i, s in enumerate(segs)
+                           if (s[1] - s[0]) >= 0.93 * s[2]]
+                host_idx = [i for i in range(len(segs))
+                            if i not in dev_idx]
+                if dev_idx:
+                    st = ext.snappy_decompress(
+                        dev_bytes,
+                        _torch.tensor([segs[i][0] for i in dev_idx],
+                                      dtype=_torch.int64),
+                        _torch.tensor([segs[i][1] for i in dev_idx],
+                                      dtype=_torch.int64),
+                        scratch,
+                        _torch.tensor([int(doff_all[i])
+                                       for i in dev_idx],
+                                      dtype=_torch.int64),
+                        _torch.tensor([segs[i][2] for i in dev_idx],
+                                      dtype=_torch.int64))
+                    statuses.append(st)
+                if host_idx:
+                    import pyarrow as _pa
+                    codec = _pa.Codec("snappy")
+                    hview = buf.numpy()
+                    for i in host_idx:
+                        a2, b2, unc2 = segs[i]
+                        try:
+                            dec = codec.decompress(
+                                hview[a2:b2].tobytes())
+                        except Exception:  # noqa: BLE001
+                            statuses.append(_torch.ones(
+                                1, dtype=_torch.int32, device=device))
+                            continue
+                        if len(dec) != unc2:
+                            statuses.append(_torch.ones(
+                                1, dtype=_torch.int32, device=device))
+                            continue
+                        hb_np = np.frombuffer(dec, dtype=np.uint8)
+                        scratch[int(doff_all[i]):
+                                int(doff_all[i]) + unc2].copy_(
+                            _torch.from_numpy(hb_np.copy()),
+                            non_blocking=True)
                 page_base = doff_all[1:-1] if has_zdict \
                     else doff_all[:-1]
                 dict_vals = None
