@@ -446,7 +446,7 @@ def read_files_batch_device(paths: List[str], device,
                         a2, b2, unc2 = segs[i]
                         try:
                             dec = codec.decompress(
-                                hview[a2:b2].tobytes())
+                                hview[a2:b2].tobytes(), unc2)
                         except Exception:  # noqa: BLE001
                             statuses.append(_torch.ones(
                                 1, dtype=_torch.int32, device=device))
