@@ -930,7 +930,9 @@ def read_native_host(path: str, columns: Optional[List[str]] = None
     for name, parts in acc.items():
         if name in str_dicts:
             dicts = str_dicts[name]
-            if all(d == dicts[0] for d in dicts[1:]):
+            if all(d == dicts[0] for d in dicts[1:]) and \
+                    all(dicts[0][i] <= dicts[0][i + 1]
+                        for i in range(len(dicts[0]) - 1)):
                 merged_vals = dicts[0]
                 codes = (np.concatenate(parts) if len(parts) > 1
                          else parts[0].copy())

@@ -153,9 +153,13 @@ def read_files_batch(paths: List[str], columns: Optional[List[str]] = None):
         for name in names:
             arrs = [f[name] for f in per_file]
             if any(isinstance(a, StrCol) for a in arrs):
-                # merge per-file dictionaries, remap codes
+                # merge per-file dictionaries, remap codes (unsorted
+                # parquet insertion-order dictionaries force the remap:
+                # StringColumn codes must follow lex order)
                 dicts = [a.values for a in arrs]
-                if all(d == dicts[0] for d in dicts[1:]):
+                if all(d == dicts[0] for d in dicts[1:]) and \
+                        all(dicts[0][i] <= dicts[0][i + 1]
+                            for i in range(len(dicts[0]) - 1)):
                     mvals = list(dicts[0])
                     codes = (np.concatenate([a.codes for a in arrs])
                              if len(arrs) > 1 else arrs[0].codes)
@@ -884,7 +888,12 @@ def read_files_batch_device(paths: List[str], device,
         for name in string_cols:
             chunks_n = by_name.get(name, [])
             dicts = [v for _, _, v in chunks_n]
-            if dicts and all(d == dicts[0] for d in dicts[1:]):
+            # parquet dictionaries are in INSERTION order; StringColumn
+            # requires SORTED values (code order = lex order), so any
+            # unsorted dictionary forces the remap path
+            if dicts and all(d == dicts[0] for d in dicts[1:]) and \
+                    all(dicts[0][i] <= dicts[0][i + 1]
+                        for i in range(len(dicts[0]) - 1)):
                 merged = list(dicts[0])
             else:
                 merged = sorted(set().union(*map(set, dicts))) \

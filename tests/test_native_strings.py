@@ -79,6 +79,13 @@ def test_reads_pyarrow_dict_string_natively(tmp_path):
     s = rb.column("s")
     assert isinstance(s, StringColumn)
     assert (s.to_numpy() == np.array(data, dtype=object)).all()
+    # parquet dictionaries are insertion-ordered ("alpha, beta, gamma,
+    # delta" here); StringColumn codes MUST follow lex order or every
+    # order-dependent op (join merge, z-order, range compare) breaks
+    assert s.values == sorted(s.values)
+    import torch as _torch
+    order_ok = s.codes[np.array(data, dtype=object) == "delta"]
+    assert (order_ok == s.values.index("delta")).all()
 
 
 def test_string_index_build_without_pyarrow_data_path(tmp_path,
