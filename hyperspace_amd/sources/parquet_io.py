@@ -788,35 +788,14 @@ def read_files_batch_device(paths: List[str], device,
         # caller-side synchronize below holds them via `bufs`
         return buf
 
-    # phase 1: read + layout-parse every file (parallel, GIL-released IO)
+    # phase 1+2 pipeline: file reads stream into decode units as each
+    # read completes (a strict all-reads-then-all-decodes barrier left
+    # the H2D + decode tail serialized after ~0.46 s of source reads)
     import time as _time
     _dt = os.environ.get("HS_DECODE_TIMING")
     _t0 = _time.perf_counter()
-    if len(paths) > 2:
-        with ThreadPoolExecutor(max_workers=16) as pool:
-            infos = list(pool.map(load_file, range(len(paths))))
-    else:
-        infos = [load_file(i) for i in range(len(paths))]
-    bufs = [inf[0] if inf is not None else None for inf in infos]
-    if not all(inf is not None for inf in infos):
-        for b in bufs:
-            if b is not None:
-                _pinned_put(b)
-        return fallback()
-
-    # phase 2: decode units.  Few big files split at ROW-GROUP
-    # granularity so a single appended file's row groups decode on
-    # parallel workers/streams (the round-1 incremental-refresh and
-    # single-file cold-load bottleneck); many files keep one unit per
-    # file.  Uploads happen once per file; other units of the same file
-    # wait on the upload event.
-    units = []  # (file_i, row_off, chunks)
-    if len(paths) < n_streams:
-        for i, inf in enumerate(infos):
-            for row_off, rg_chunks in split_row_groups(inf[2]):
-                units.append((i, row_off, rg_chunks))
-    else:
-        units = [(i, 0, inf[2]) for i, inf in enumerate(infos)]
+    infos: List[Optional[tuple]] = [None] * len(paths)
+    split_rgs = len(paths) < n_streams
 
     import threading
     dev_bufs: List[Optional["_torch.Tensor"]] = [None] * len(paths)
@@ -851,22 +830,69 @@ def read_files_batch_device(paths: List[str], device,
 
     _t1 = _time.perf_counter()
     _dw = int(os.environ.get("HS_DECODE_WORKERS", "16"))
-    if len(units) > 2 and _dw > 1:
-        with ThreadPoolExecutor(max_workers=_dw) as pool:
-            list(pool.map(decode_unit, enumerate(units)))
+    n_units = 0
+    failed = False
+    if len(paths) > 2 and _dw > 1:
+        from concurrent.futures import as_completed
+        import itertools
+        uid_gen = itertools.count()
+        dec_futs = []
+        with ThreadPoolExecutor(max_workers=16) as read_pool, \
+                ThreadPoolExecutor(max_workers=_dw) as dec_pool:
+            read_futs = {read_pool.submit(load_file, i): i
+                         for i in range(len(paths))}
+            for fut in as_completed(read_futs):
+                i = read_futs[fut]
+                inf = fut.result()
+                infos[i] = inf
+                if inf is None:
+                    failed = True
+                    continue
+                if failed:
+                    continue  # drain remaining reads; fallback below
+                if split_rgs:
+                    for row_off, rg_chunks in split_row_groups(inf[2]):
+                        dec_futs.append(dec_pool.submit(
+                            decode_unit,
+                            (next(uid_gen), (i, row_off, rg_chunks))))
+                        n_units += 1
+                else:
+                    dec_futs.append(dec_pool.submit(
+                        decode_unit, (next(uid_gen), (i, 0, inf[2]))))
+                    n_units += 1
+            if not failed:
+                for f in dec_futs:
+                    f.result()
     else:
-        for u in enumerate(units):
-            decode_unit(u)
-    _t2 = _time.perf_counter()
-    # order the default stream after every worker stream, then host-sync
+        for i in range(len(paths)):
+            infos[i] = load_file(i)
+            if infos[i] is None:
+                failed = True
+                break
+        if not failed:
+            uid = 0
+            for i in range(len(paths)):
+                unit_list = (split_row_groups(infos[i][2]) if split_rgs
+                             else [(0, infos[i][2])])
+                for row_off, rg_chunks in unit_list:
+                    decode_unit((uid, (i, row_off, rg_chunks)))
+                    uid += 1
+                    n_units += 1
+    bufs = [inf[0] if inf is not None else None for inf in infos]
+    if failed:
+        for bq in bufs:
+            if bq is not None:
+                _pinned_put(bq)
+        return fallback()
+    _t2 = _time.perf_counter()    # order the default stream after every worker stream, then host-sync
     # so the pinned buffers can be recycled
     cur = _torch.cuda.current_stream()
     for s in streams:
         cur.wait_stream(s)
     cur.synchronize()
     if _dt:
-        print(f"[hs-decode] files={len(paths)} units={len(units)} "
-              f"read+layout {_t1-_t0:.3f}s decode-host {_t2-_t1:.3f}s "
+        print(f"[hs-decode] files={len(paths)} units={n_units} "
+              f"read+decode(pipe) {_t2-_t1:.3f}s "
               f"sync {_time.perf_counter()-_t2:.3f}s", file=sys.stderr)
     # one device reduction + sync for every chunk's status words (a
     # per-tensor .any() paid ~1.4 ms of sync each across ~50 chunks)
