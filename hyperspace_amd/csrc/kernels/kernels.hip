@@ -232,7 +232,7 @@ void normalize_key(const void* vals, int dtype, uint64_t* out, int64_t n,
 // ---------------------------------------------------------------------------
 
 constexpr int RS_RADIX = 256;      // 8-bit digits
-constexpr int RS_MAX_BLOCKS = 1024;
+constexpr int RS_MAX_BLOCKS = 2048;  // 256 CUs x 8 blocks/CU
 
 static inline int rs_num_blocks(int64_t n) {
   int64_t tiles = cdiv(n, THREADS);
