@@ -398,6 +398,70 @@ class IcebergTableRelation(FileBasedRelation):
     def all_files(self) -> List[FileInfo]:
         return self.table.files_for_snapshot(self.snapshot_id_opt)
 
+    # -- partitioned tables (identity transforms) --------------------------
+    def partition_schema(self) -> Schema:
+        """Identity-partition fields from the default partition spec;
+        names/types resolve through the iceberg schema field ids
+        (non-identity transforms — bucket/truncate/days — contribute no
+        queryable column and are skipped)."""
+        if getattr(self, "_pschema", None) is None:
+            from ..log.entry import SchemaField
+            meta = self.table.metadata()
+            specs = meta.get("partition-specs") or []
+            spec = next((sp for sp in specs
+                         if sp.get("spec-id") == meta.get(
+                             "default-spec-id", 0)), None)
+            fields = []
+            if spec:
+                by_id = {}
+                for s in meta.get("schemas", []):
+                    for f in s.get("fields", []):
+                        by_id[f["id"]] = f
+                for pf in spec.get("fields", []):
+                    if pf.get("transform", "identity") != "identity":
+                        continue
+                    src = by_id.get(pf.get("source-id"))
+                    if src is None:
+                        continue
+                    t = src["type"] if isinstance(src["type"], str) \
+                        else "string"
+                    fields.append(SchemaField(
+                        pf.get("name", src["name"]),
+                        _ICEBERG_TO_SPARK.get(t, t), False))
+            self._pschema = Schema(fields)
+        return self._pschema
+
+    def partition_values(self, path: str) -> Dict[str, Any]:
+        """Per-file partition values from the manifest entries'
+        data_file.partition record."""
+        if getattr(self, "_pvalues", None) is None:
+            vals: Dict[str, Dict[str, Any]] = {}
+            for e in self.table._live_entries(self.snapshot_id_opt):
+                df = e["data_file"]
+                p = _strip_uri(df["file_path"])
+                if not os.path.isabs(p):
+                    p = os.path.join(self.table.path, p)
+                part = df.get("partition") or {}
+                vals[p] = part if isinstance(part, dict) else {}
+            self._pvalues = vals
+        raw = self._pvalues.get(path, {})
+        out: Dict[str, Any] = {}
+        for f in self.partition_schema().fields:
+            v = raw.get(f.name)
+            out[f.name] = v
+        return out
+
+    def read_files(self, paths: List[str], columns, device):
+        from .parquet_source import partitioned_read_files
+        return partitioned_read_files(
+            self, paths, columns, device,
+            lambda p, c, d: super(IcebergTableRelation, self).read_files(
+                p, c, d))
+
+    def prune_partitions(self, cond):
+        from .parquet_source import prune_partitions_generic
+        return prune_partitions_generic(self, cond)
+
     def signature(self) -> str:
         return f"{self.effective_snapshot_id}.{self.table.path}"
 

@@ -202,3 +202,100 @@ def test_index_over_real_iceberg_end_to_end(tmp_path, monkeypatch):
     df_old = session.read_iceberg(str(tmp_path / "t"),
                                   snapshot_id=snap_before)
     assert df_old.filter("key = 1").collect().num_rows == 400
+
+
+def test_partitioned_iceberg_identity_spec(tmp_path):
+    """Identity-partitioned iceberg: partition values come from the
+    manifest entries' data_file.partition record and materialize as
+    constant columns; partition predicates prune before IO."""
+    root = tmp_path / "pext"
+    meta = root / "metadata"
+    meta.mkdir(parents=True)
+    rng = np.random.default_rng(3)
+    rows = {}
+    entries = []
+    entry_schema = {
+        "type": "record", "name": "manifest_entry", "fields": [
+            {"name": "status", "type": "int"},
+            {"name": "snapshot_id", "type": ["null", "long"]},
+            {"name": "data_file", "type": {
+                "type": "record", "name": "r2", "fields": [
+                    {"name": "file_path", "type": "string"},
+                    {"name": "file_format", "type": "string"},
+                    {"name": "partition", "type": {
+                        "type": "record", "name": "r102", "fields": [
+                            {"name": "region",
+                             "type": ["null", "string"]}]}},
+                    {"name": "record_count", "type": "long"},
+                    {"name": "file_size_in_bytes", "type": "long"},
+                ]}},
+        ]}
+    for region in ("eu", "us"):
+        d = root / "data" / f"region={region}"
+        d.mkdir(parents=True)
+        n = int(rng.integers(400, 900))
+        rows[region] = n
+        p = d / "00000-0.parquet"
+        pq.write_table(pa.table({"key": rng.integers(0, 50, n),
+                                 "v": rng.random(n)}), str(p))
+        entries.append({"status": 1, "snapshot_id": 7,
+                        "data_file": {
+                            "file_path": "file://" + str(p),
+                            "file_format": "PARQUET",
+                            "partition": {"region": region},
+                            "record_count": n,
+                            "file_size_in_bytes": os.stat(p).st_size}})
+    manifest = meta / "m0.avro"
+    write_avro_records(str(manifest), entry_schema, entries)
+    list_schema = {
+        "type": "record", "name": "manifest_file", "fields": [
+            {"name": "manifest_path", "type": "string"},
+            {"name": "manifest_length", "type": "long"},
+            {"name": "partition_spec_id", "type": "int"},
+            {"name": "added_snapshot_id", "type": ["null", "long"]}]}
+    ml = meta / "snap-7-1-x.avro"
+    write_avro_records(str(ml), list_schema, [{
+        "manifest_path": "file://" + str(manifest),
+        "manifest_length": os.stat(manifest).st_size,
+        "partition_spec_id": 0, "added_snapshot_id": 7}])
+    md = {
+        "format-version": 2, "table-uuid": "p1",
+        "location": "file://" + str(root),
+        "last-updated-ms": 1, "last-column-id": 3,
+        "schemas": [{"type": "struct", "schema-id": 0, "fields": [
+            {"id": 1, "name": "key", "required": False, "type": "long"},
+            {"id": 2, "name": "v", "required": False, "type": "double"},
+            {"id": 3, "name": "region", "required": False,
+             "type": "string"}]}],
+        "current-schema-id": 0,
+        "partition-specs": [{"spec-id": 0, "fields": [
+            {"name": "region", "transform": "identity",
+             "source-id": 3, "field-id": 1000}]}],
+        "default-spec-id": 0,
+        "current-snapshot-id": 7,
+        "snapshots": [{"snapshot-id": 7, "timestamp-ms": 1,
+                       "summary": {"operation": "append"},
+                       "manifest-list": "file://" + str(ml),
+                       "schema-id": 0}],
+    }
+    with open(meta / "v1.metadata.json", "w") as f:
+        json.dump(md, f)
+    with open(meta / "version-hint.text", "w") as f:
+        f.write("1")
+
+    import hyperspace_amd as hs
+    session = hs.HyperspaceSession(device="cpu")
+    df = session.read_iceberg(str(root))
+    rel = df.plan.collect_leaves()[0].relation
+    assert rel.partition_schema().field_names() == ["region"]
+    out = df.select("key", "region").collect()
+    assert out.num_rows == sum(rows.values())
+    import collections
+    counts = collections.Counter(out.column("region").to_numpy().tolist())
+    assert counts == rows
+    from hyperspace_amd.execution.executor import Executor
+    ex = Executor(session)
+    got = ex.execute(df.filter("region = 'us'").select("key", "v")
+                     .optimized_plan())
+    assert got.num_rows == rows["us"]
+    assert ex.stats.scanned_files == 1
