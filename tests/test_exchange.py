@@ -90,7 +90,7 @@ def _exchange_worker(rank, world, rdv_file, results, n, num_buckets):
         dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("world", [2, 4])
+@pytest.mark.parametrize("world", [2, 4, 8])
 def test_exchange_contract(tmp_path, world):
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     rdv = str(tmp_path / f"rdv{world}")
