@@ -84,3 +84,23 @@ def test_plan_stability(env, name):
     assert plan == expected, (
         f"plan drifted for {name}:\n--- got ---\n{plan}\n"
         f"--- golden ---\n{expected}")
+
+
+def test_explain_golden(env):
+    """Explain output golden (reference plananalysis/ExplainTest against
+    resources/expected/) — paths and version dirs normalized."""
+    session, fact, dim = env
+    import hyperspace_amd as hs
+    h = hs.Hyperspace(session)
+    q = fact.filter("qty = 7").select("qty", "val")
+    out = h.explain(q, verbose=True)
+    out = re.sub(r"(parquet|delta|iceberg):[^,)\s@]*", r"\1:<PATH>", out)
+    out = re.sub(r":/[^\s]*/indexes/", ":<SYS>/", out)
+    golden_path = os.path.join(GOLDEN_DIR, "explain_filter.txt")
+    if GENERATE:
+        with open(golden_path, "w") as f:
+            f.write(out + "\n")
+        pytest.skip("golden regenerated")
+    with open(golden_path) as f:
+        expected = f.read().rstrip("\n")
+    assert out == expected
