@@ -344,3 +344,87 @@ def test_minmax_html_report(env, tmp_path):
     assert html.startswith("<!DOCTYPE html>")
     assert "<svg" in html and "rect" in html
     assert "qty" in html
+
+
+def test_deleted_index_not_used_restored_is(env):
+    """Soft-deleted indexes are invisible to the optimizer; restore
+    brings them back (reference E2EHyperspaceRulesTest behavior)."""
+    session, h, left, right = env
+    h.create_index(left, hs.CoveringIndexConfig("del1", ["qty"],
+                                                ["price"]))
+    session.enable_hyperspace()
+    q = left.filter("qty = 7").select("qty", "price")
+    assert any(isinstance(l, IndexScan)
+               for l in q.optimized_plan().collect_leaves())
+    h.delete_index("del1")
+    assert not any(isinstance(l, IndexScan)
+                   for l in q.optimized_plan().collect_leaves())
+    h.restore_index("del1")
+    assert any(isinstance(l, IndexScan)
+               for l in q.optimized_plan().collect_leaves())
+
+
+def test_disable_enable_toggle(env):
+    session, h, left, right = env
+    h.create_index(left, hs.CoveringIndexConfig("tog", ["qty"],
+                                                ["price"]))
+    q = left.filter("qty = 7").select("qty", "price")
+    session.disable_hyperspace()
+    assert not any(isinstance(l, IndexScan)
+                   for l in q.optimized_plan().collect_leaves())
+    session.enable_hyperspace()
+    assert any(isinstance(l, IndexScan)
+               for l in q.optimized_plan().collect_leaves())
+
+
+def test_uncovered_projection_blocks_rewrite(env):
+    """A filter query projecting a column the index does not include
+    cannot rewrite (MISSING_REQUIRED_COL)."""
+    session, h, left, right = env
+    h.create_index(left, hs.CoveringIndexConfig("cov1", ["qty"], []))
+    session.enable_hyperspace()
+    q = left.filter("qty = 7").select("qty", "price")
+    assert not any(isinstance(l, IndexScan)
+                   for l in q.optimized_plan().collect_leaves())
+    out = h.why_not(q, index_name="cov1")
+    assert "MISSING_REQUIRED_COL" in out
+
+
+def test_projection_only_no_rewrite(env):
+    """Project-without-filter plans are not FilterIndexRule targets
+    (reference requires the Filter node)."""
+    session, h, left, right = env
+    h.create_index(left, hs.CoveringIndexConfig("pj1", ["qty"],
+                                                ["price"]))
+    session.enable_hyperspace()
+    q = left.select("qty", "price")
+    assert not any(isinstance(l, IndexScan)
+                   for l in q.optimized_plan().collect_leaves())
+
+
+def test_self_join_uses_same_index_both_sides(env):
+    session, h, left, right = env
+    h.create_index(left, hs.CoveringIndexConfig("sj", ["orderkey"],
+                                                ["qty"]))
+    session.enable_hyperspace()
+    from hyperspace_amd.plan.expr import col
+    q = left.select("orderkey", "qty").join(
+        left.select("orderkey", "qty"), on="orderkey")
+    plan = q.optimized_plan()
+    scans = [l for l in plan.collect_leaves()
+             if isinstance(l, IndexScan)]
+    assert len(scans) == 2
+    assert all(s.entry.name == "sj" for s in scans)
+    from hyperspace_amd.execution.executor import Executor
+    ex = Executor(session)
+    out = ex.execute(plan)
+    assert ex.stats.shuffles == 0
+    session.disable_hyperspace()
+    assert out.num_rows == q.collect().num_rows
+
+
+def test_invalid_refresh_mode(env):
+    session, h, left, right = env
+    h.create_index(left, hs.CoveringIndexConfig("rm1", ["qty"], []))
+    with pytest.raises(Exception, match="Unsupported refresh mode"):
+        h.refresh_index("rm1", "bogus")
