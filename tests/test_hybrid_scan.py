@@ -236,3 +236,39 @@ def _walk(plan):
     for c in getattr(plan, "children", []):
         out.extend(_walk(c))
     return out
+
+
+def test_hybrid_scan_partitioned_source(tmp_path, monkeypatch):
+    """Hybrid scan over hive-partitioned data (reference
+    HybridScanForPartitionedDataTest): the appended file lives in a
+    partition directory; the appended-scan keeps partition columns."""
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "idx"))
+    rng = np.random.default_rng(53)
+    root = tmp_path / "pt"
+    for day in (1, 2):
+        d = root / f"day={day}"
+        d.mkdir(parents=True)
+        for i in range(3):
+            _write(rng, d / f"part-{i}.parquet", n=4000)
+    session = hs.HyperspaceSession(device="cpu")
+    session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 8)
+    session.conf.set(hs.IndexConstants.INDEX_LINEAGE_ENABLED, True)
+    h = hs.Hyperspace(session)
+    df = session.read_parquet(str(root))
+    h.create_index(df, hs.CoveringIndexConfig(
+        "hpx", ["key"], ["val", "day"]))
+
+    # append INTO a partition after the build
+    _write(rng, root / "day=2" / "part-new.parquet", n=1500)
+    session.enable_hyperspace()
+    session.conf.set(hs.IndexConstants.INDEX_HYBRID_SCAN_ENABLED, True)
+    live = session.read_parquet(str(root))
+    q = live.filter("key = 77").select("key", "val", "day")
+    plan = q.optimized_plan()
+    assert any(isinstance(n, UnionNode) for n in _walk(plan)), \
+        plan.pretty()
+    out = q.collect()
+    session.disable_hyperspace()
+    base = q.collect()
+    assert _rows(out, ["key", "val", "day"]) == \
+        _rows(base, ["key", "val", "day"])
