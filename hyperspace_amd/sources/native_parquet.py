@@ -532,11 +532,13 @@ class ColumnChunkLayout:
     when not supplied by the writer's layout cache).
     """
     __slots__ = ("name", "np_dtype", "pages", "num_values", "encoding",
-                 "dict_page", "page_masks", "is_string", "str_values")
+                 "dict_page", "page_masks", "is_string", "str_values",
+                 "codec")
 
     def __init__(self, name, np_dtype, pages, num_values,
                  encoding="plain", dict_page=None, page_masks=None,
-                 is_string=False, str_values=None):
+                 is_string=False, str_values=None,
+                 codec="UNCOMPRESSED"):
         self.name = name
         self.np_dtype = np_dtype
         self.pages = pages
@@ -546,6 +548,7 @@ class ColumnChunkLayout:
         self.page_masks = page_masks or [None] * len(pages)
         self.is_string = is_string
         self.str_values = str_values
+        self.codec = codec
 
     def dict_values(self, data) -> List[str]:
         """The string dictionary: cached from the writer, else parsed
@@ -676,7 +679,11 @@ def _walk_row_group(rg, pf_schema, data, want
         if want is not None and name.lower() not in want:
             continue
         codec = col.compression.upper()
-        if codec not in ("UNCOMPRESSED", "SNAPPY"):
+        # SNAPPY decodes on device (or host for copy-dense pages);
+        # GZIP/ZSTD/BROTLI pages decode on host codec threads but keep
+        # the native page-assembly path (no pyarrow table fallback)
+        if codec not in ("UNCOMPRESSED", "SNAPPY", "GZIP", "ZSTD",
+                         "BROTLI"):
             return None
         encs = set(col.encodings)
         is_dict = bool(encs & {"PLAIN_DICTIONARY", "RLE_DICTIONARY"})
@@ -727,7 +734,7 @@ def _walk_row_group(rg, pf_schema, data, want
             dict_n = hdr.get(7, {}).get(1)
             if dict_n is None:
                 return None
-            if codec == "SNAPPY":
+            if codec != "UNCOMPRESSED":
                 if hdr.get(2) is None:
                     return None
                 dict_page = ("z", r.pos, r.pos + hdr.get(3), dict_n,
@@ -760,7 +767,7 @@ def _walk_row_group(rg, pf_schema, data, want
             values_off = r.pos
             page_end = r.pos + page_bytes
             mask = None
-            if codec == "SNAPPY":
+            if codec != "UNCOMPRESSED":
                 # compressed page: record the compressed extent +
                 # uncompressed size; the device decompresses, then (for
                 # OPTIONAL all-valid chunks) the level prefix is skipped
@@ -820,7 +827,8 @@ def _walk_row_group(rg, pf_schema, data, want
             enc_kind = "plain"
         out.append(ColumnChunkLayout(name, np_dtype, pages,
                                      col.num_values, enc_kind, dict_page,
-                                     page_masks, is_string=is_string))
+                                     page_masks, is_string=is_string,
+                                     codec=codec))
     return out
 
 

@@ -424,8 +424,11 @@ def read_files_batch_device(paths: List[str], device,
                 # C++ codec (~GB/s per worker thread) and uploads;
                 # near-incompressible pages (giant literals) stay on
                 # device where decode is a bandwidth-bound copy.
-                dev_idx = [i for i, s in enumerate(segs)
-                           if (s[1] - s[0]) >= 0.93 * s[2]]
+                if c.codec == "SNAPPY":
+                    dev_idx = [i for i, s in enumerate(segs)
+                               if (s[1] - s[0]) >= 0.93 * s[2]]
+                else:
+                    dev_idx = []  # gzip/zstd/brotli: host codec only
                 host_idx = [i for i in range(len(segs))
                             if i not in dev_idx]
                 if dev_idx:
@@ -444,7 +447,7 @@ def read_files_batch_device(paths: List[str], device,
                     statuses.append(st)
                 if host_idx:
                     import pyarrow as _pa
-                    codec = _pa.Codec("snappy")
+                    codec = _pa.Codec(c.codec.lower())
                     hview = buf.numpy()
                     for i in host_idx:
                         a2, b2, unc2 = segs[i]

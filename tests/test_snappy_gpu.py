@@ -198,3 +198,34 @@ def test_snappy_mixed_dict_plain_chunk(tmp_path):
     assert rc == [n]
     assert (batch.tensor("key").cpu().numpy() == key).all()
     assert np.allclose(batch.tensor("val").cpu().numpy(), val)
+
+
+@pytest.mark.parametrize("codec", ["zstd", "gzip"])
+def test_zstd_gzip_pages_host_codec_path(tmp_path, codec):
+    """ZSTD/GZIP pages keep the native page-assembly path: host codec
+    threads decompress into device scratch, everything downstream
+    (copy_unaligned, dict decode) is unchanged."""
+    rng = np.random.default_rng(29)
+    n = 1_500_000
+    key = rng.integers(0, 10_000, n)
+    val = rng.random(n)
+    p = str(tmp_path / f"{codec}.parquet")
+    pq.write_table(pa.table({"key": key, "val": val}), p,
+                   compression=codec.upper(), use_dictionary=True,
+                   data_page_version="1.0")
+    from hyperspace_amd.sources import parquet_io as pio
+    orig = pio.read_files_batch
+    called = []
+
+    def spy(*a, **k):
+        called.append(1)
+        return orig(*a, **k)
+    pio.read_files_batch = spy
+    try:
+        batch, rc = read_files_batch_device([p], torch.device("cuda:0"))
+    finally:
+        pio.read_files_batch = orig
+    assert not called, "fell back to the host table path"
+    assert rc == [n]
+    assert (batch.tensor("key").cpu().numpy() == key).all()
+    assert np.allclose(batch.tensor("val").cpu().numpy(), val)
