@@ -77,9 +77,16 @@ def test_fallback_on_compressed(tmp_path, cols):
     batch, counts = read_files_batch([p])
     assert batch.num_rows == 50_000 and counts == [50_000]
     p2 = str(tmp_path / "t2.parquet")
+    # GZIP/ZSTD now yield layouts too (host-codec pages on the native
+    # assembly path); LZ4 remains unsupported -> no layout
     pq.write_table(pa.table(cols), p2, compression="GZIP")
-    assert read_native_layout(p2) is None
-    batch2, counts2 = read_files_batch([p2])
+    lay2 = read_native_layout(p2)
+    assert lay2 is not None
+    assert all(c.codec == "GZIP" for c in lay2[1])
+    p3 = str(tmp_path / "t3.parquet")
+    pq.write_table(pa.table(cols), p3, compression="LZ4")
+    assert read_native_layout(p3) is None
+    batch2, counts2 = read_files_batch([p3])
     assert batch2.num_rows == 50_000 and counts2 == [50_000]
 
 
