@@ -64,12 +64,21 @@ class HyperspaceSession:
         from .sources.parquet_source import ParquetRelation
         return DataFrame(self, Scan(ParquetRelation(list(paths))))
 
-    def read_delta(self, path: str, version_as_of=None):
-        """Read a delta-style transactional table (time travel via
-        ``version_as_of``)."""
+    def read_delta(self, path: str, version_as_of=None,
+                   timestamp_as_of=None):
+        """Read a Delta Lake table (time travel via ``version_as_of``
+        or ``timestamp_as_of`` — the newest version committed at or
+        before the timestamp, Delta's timestampAsOf semantics)."""
         from .dataframe import DataFrame
         from .plan.nodes import Scan
-        from .sources.delta_source import DeltaTableRelation
+        from .sources.delta_source import DeltaTable, DeltaTableRelation
+        if timestamp_as_of is not None:
+            if version_as_of is not None:
+                from .exceptions import HyperspaceException
+                raise HyperspaceException(
+                    "Specify only one of version_as_of/timestamp_as_of")
+            version_as_of = DeltaTable(path).version_at_timestamp(
+                timestamp_as_of)
         return DataFrame(self, Scan(DeltaTableRelation(path,
                                                        version_as_of)))
 

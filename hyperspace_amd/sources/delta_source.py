@@ -366,6 +366,43 @@ class DeltaTable:
             json.dump({"version": target, "size": len(rows)}, f)
         return out
 
+    def version_at_timestamp(self, ts_ms) -> int:
+        """Newest version whose commit timestamp is <= the given time
+        (Delta's timestampAsOf).  Accepts epoch millis, seconds, or an
+        ISO date/datetime string."""
+        import datetime as _dt
+        if isinstance(ts_ms, str):
+            dt = _dt.datetime.fromisoformat(ts_ms)
+            if dt.tzinfo is None:
+                dt = dt.replace(tzinfo=_dt.timezone.utc)
+            ts_ms = int(dt.timestamp() * 1000)
+        elif isinstance(ts_ms, (int, float)) and ts_ms < 10**12:
+            ts_ms = int(ts_ms * 1000)  # epoch seconds
+        best = None
+        for v in self.versions():
+            p = os.path.join(self.log_dir, f"{v:020d}.json")
+            t = None
+            if os.path.exists(p):
+                with open(p) as f:
+                    for line in f:
+                        line = line.strip()
+                        if not line:
+                            continue
+                        a = json.loads(line)
+                        ci = a.get("commitInfo")
+                        if ci and "timestamp" in ci:
+                            t = int(ci["timestamp"])
+                            break
+            if t is None:
+                t = int(os.path.getmtime(p) * 1000) \
+                    if os.path.exists(p) else None
+            if t is not None and t <= ts_ms:
+                best = v
+        if best is None:
+            raise HyperspaceException(
+                f"No delta version at or before timestamp {ts_ms}")
+        return best
+
     def clean_commits_before(self, version: int) -> None:
         """Delete JSON commits below ``version`` (a checkpoint at or
         above it must exist) — mirrors delta log retention cleanup and

@@ -258,3 +258,44 @@ def test_partitioned_delta_table(tmp_path, monkeypatch):
     n7 = df.filter("key = 7").select("key", "val").collect().num_rows
     session.disable_hyperspace()
     assert n7 == df.filter("key = 7").collect().num_rows
+
+
+def test_timestamp_as_of_time_travel(tmp_path, monkeypatch):
+    """timestampAsOf resolves the newest version committed at or before
+    the given time (Delta semantics; commitInfo timestamps)."""
+    import json as _json
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "idx"))
+    rng = np.random.default_rng(3)
+    t = DeltaTable.create(str(tmp_path / "t"))
+    p1 = _write_part(str(tmp_path / "t"), rng, name="a.parquet")
+    t.append_files([p1])  # v1
+    p2 = _write_part(str(tmp_path / "t"), rng, name="b.parquet")
+    t.append_files([p2])  # v2
+    # rewrite commit timestamps to known values
+    log = tmp_path / "t" / "_delta_log"
+    for v, ts in ((0, 1_000_000), (1, 2_000_000), (2, 3_000_000)):
+        pth = log / f"{v:020d}.json"
+        lines = [ _json.loads(l) for l in open(pth) ]
+        for a in lines:
+            if "commitInfo" in a:
+                a["commitInfo"]["timestamp"] = ts
+        with open(pth, "w") as f:
+            for a in lines:
+                f.write(_json.dumps(a) + "\n")
+
+    assert t.version_at_timestamp(2_500_000) == 1
+    assert t.version_at_timestamp(3_000_000) == 2
+    with pytest.raises(Exception, match="No delta version"):
+        t.version_at_timestamp(500_000)
+
+    import hyperspace_amd as hs
+    session = hs.HyperspaceSession(device="cpu")
+    df = session.read_delta(str(tmp_path / "t"),
+                            timestamp_as_of=2_500_000)
+    assert df.collect().num_rows == 1000  # only a.parquet at v1
+    df2 = session.read_delta(str(tmp_path / "t"),
+                             timestamp_as_of=3_500_000)
+    assert df2.collect().num_rows == 2000
+    with pytest.raises(Exception, match="only one of"):
+        session.read_delta(str(tmp_path / "t"), version_as_of=1,
+                           timestamp_as_of=2_500_000)
