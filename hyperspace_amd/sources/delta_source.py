@@ -368,16 +368,16 @@ class DeltaTable:
 
     def version_at_timestamp(self, ts_ms) -> int:
         """Newest version whose commit timestamp is <= the given time
-        (Delta's timestampAsOf).  Accepts epoch millis, seconds, or an
-        ISO date/datetime string."""
+        (Delta's timestampAsOf).  Accepts epoch millis or an ISO
+        date/datetime string."""
         import datetime as _dt
         if isinstance(ts_ms, str):
             dt = _dt.datetime.fromisoformat(ts_ms)
             if dt.tzinfo is None:
                 dt = dt.replace(tzinfo=_dt.timezone.utc)
             ts_ms = int(dt.timestamp() * 1000)
-        elif isinstance(ts_ms, (int, float)) and ts_ms < 10**12:
-            ts_ms = int(ts_ms * 1000)  # epoch seconds
+        else:
+            ts_ms = int(ts_ms)  # epoch millis (Delta convention)
         best = None
         for v in self.versions():
             p = os.path.join(self.log_dir, f"{v:020d}.json")
