@@ -390,3 +390,34 @@ def test_composed_expr_oracle_vs_numpy(env):
         vn = cpu_ref.normalize_key(torch.from_numpy(v))
         assert int(mn[i]) == int(vn.min())
         assert int(mx[i]) == int(vn.max())
+
+
+def test_dataskipping_over_delta_table(tmp_path, monkeypatch):
+    """Sketch index over a Delta table: per-file pruning composes with
+    the transaction-log file listing (reference DataSkipping +
+    DeltaLakeIntegration crossover)."""
+    import torch
+    from hyperspace_amd.sources.delta_source import DeltaTable
+    from hyperspace_amd.execution.columnar import ColumnBatch
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "idx"))
+    rng = np.random.default_rng(13)
+    t = DeltaTable.create(str(tmp_path / "t"))
+    for i in range(6):  # disjoint ranges -> minmax skipping bites
+        t.append_batch(ColumnBatch({
+            "key": torch.from_numpy(
+                rng.integers(i * 100, (i + 1) * 100, 2000)),
+            "val": torch.from_numpy(rng.random(2000))}))
+    session = hs.HyperspaceSession(device="cpu")
+    h = hs.Hyperspace(session)
+    df = session.read_delta(str(tmp_path / "t"))
+    h.create_index(df, hs.DataSkippingIndexConfig(
+        "dsd", hs.MinMaxSketch("key")))
+    session.enable_hyperspace()
+    q = df.filter("key = 250").select("key", "val")
+    plan = q.optimized_plan()
+    scans = [l for l in plan.collect_leaves() if isinstance(l, Scan)]
+    assert scans and scans[0].file_subset is not None
+    assert scans[0].skipped_files == 5
+    out = Executor(session).execute(plan)
+    session.disable_hyperspace()
+    assert out.num_rows == q.collect().num_rows
