@@ -232,10 +232,12 @@ void normalize_key(const void* vals, int dtype, uint64_t* out, int64_t n,
 // ---------------------------------------------------------------------------
 
 constexpr int RS_RADIX = 256;      // 8-bit digits
+constexpr int RS_V = 2;            // keys per thread per tile
+constexpr int RS_TILE = THREADS * RS_V;  // 1024 keys per tile
 constexpr int RS_MAX_BLOCKS = 2048;  // 256 CUs x 8 blocks/CU
 
 static inline int rs_num_blocks(int64_t n) {
-  int64_t tiles = cdiv(n, THREADS);
+  int64_t tiles = cdiv(n, RS_TILE);
   return (int)(tiles < RS_MAX_BLOCKS ? (tiles < 1 ? 1 : tiles)
                                      : RS_MAX_BLOCKS);
 }
@@ -272,14 +274,12 @@ __global__ void k_rs_hist(const uint64_t* __restrict__ keys, int64_t n,
   __shared__ uint32_t lh[RS_RADIX];
   for (int d = threadIdx.x; d < RS_RADIX; d += blockDim.x) lh[d] = 0;
   __syncthreads();
-  int64_t t0 = (int64_t)blockIdx.x * tiles_per_block;
-  int64_t t1 = t0 + tiles_per_block;
-  for (int64_t tile = t0; tile < t1 && tile * THREADS < n; tile++) {
-    int64_t i = tile * (int64_t)THREADS + threadIdx.x;
-    if (i < n) {
-      int d = (int)((keys[i] >> shift) & 255);
-      atomicAdd(&lh[d], 1u);
-    }
+  int64_t e0 = (int64_t)blockIdx.x * tiles_per_block * RS_TILE;
+  int64_t e1 = e0 + tiles_per_block * RS_TILE;
+  if (e1 > n) e1 = n;
+  for (int64_t i = e0 + threadIdx.x; i < e1; i += THREADS) {
+    int d = (int)((keys[i] >> shift) & 255);
+    atomicAdd(&lh[d], 1u);
   }
   __syncthreads();
   for (int d = threadIdx.x; d < RS_RADIX; d += blockDim.x)
@@ -319,18 +319,20 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
                              P* __restrict__ opayload, int64_t n,
                              int shift, const uint32_t* __restrict__ hist,
                              int nb, int64_t tiles_per_block) {
-  // LDS-staged scatter: elements are first counting-sorted by digit
-  // WITHIN the tile (stage[] in digit order), then written out with
+  // LDS-staged multi-item scatter: each thread carries RS_V keys
+  // (striped rounds, preserving element order), the 1024-key tile is
+  // counting-sorted into LDS by digit, then written out with
   // threadIdx-contiguous reads — global stores become per-digit-run
-  // coalesced bursts instead of 256-way scattered singles (the round-1
-  // k_rs_scatter showed 32.6% issue-stall on exactly those stores).
+  // coalesced bursts instead of 256-way scattered singles, and the
+  // per-tile barrier cost amortizes over 4x the keys (the round-1
+  // k_rs_scatter showed 32.6% issue-stall on the scattered stores).
   __shared__ uint32_t cur[RS_RADIX];
-  __shared__ uint32_t wave_cnt[WAVES][RS_RADIX];
-  __shared__ uint32_t wave_pref[WAVES][RS_RADIX];
+  // per (round, wave, digit) counts, scanned IN PLACE into prefixes
+  __shared__ uint32_t cnt[RS_V][WAVES][RS_RADIX];
   __shared__ uint32_t tile_total[RS_RADIX];
   __shared__ uint32_t digit_start[RS_RADIX];  // excl scan of tile_total
   __shared__ uint32_t wsum[WAVES];
-  __shared__ uint64_t stage[THREADS];  // keys, then payload (reused)
+  __shared__ uint64_t stage[RS_TILE];  // keys, then payload (reused)
   static_assert(RS_RADIX == THREADS, "one digit per thread");
   for (int d = threadIdx.x; d < RS_RADIX; d += blockDim.x)
     cur[d] = hist[(int64_t)d * nb + blockIdx.x];
@@ -338,37 +340,52 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
   int wave = threadIdx.x >> 6;
   int64_t t0 = (int64_t)blockIdx.x * tiles_per_block;
   int64_t t1 = t0 + tiles_per_block;
-  for (int64_t tile = t0; tile < t1 && tile * THREADS < n; tile++) {
-    int64_t i = tile * (int64_t)THREADS + threadIdx.x;
-    int tile_n = (int)((n - tile * THREADS) < THREADS
-                           ? (n - tile * THREADS) : THREADS);
-    bool valid = i < n;
-    uint64_t key = valid ? keys[i] : 0;
-    P pl = valid ? payload[i] : P(0);
-    int d = valid ? (int)((key >> shift) & 255) : RS_RADIX;  // sentinel
-    for (int x = threadIdx.x; x < WAVES * RS_RADIX; x += blockDim.x)
-      ((uint32_t*)wave_cnt)[x] = 0;
+  for (int64_t tile = t0; tile < t1 && tile * RS_TILE < n; tile++) {
+    int64_t ebase = tile * (int64_t)RS_TILE;
+    int tile_n = (int)((n - ebase) < RS_TILE ? (n - ebase) : RS_TILE);
+    uint64_t key[RS_V];
+    P pl[RS_V];
+    int dig[RS_V];
+    int rank[RS_V];
+    for (int x = threadIdx.x; x < RS_V * WAVES * RS_RADIX;
+         x += blockDim.x)
+      ((uint32_t*)cnt)[x] = 0;
     __syncthreads();
-    // 9-bit ballot multi-split: lanes with equal digit (incl. sentinel)
-    unsigned long long eq = ~0ull;
 #pragma unroll
-    for (int b = 0; b < 9; b++) {
-      unsigned long long m = __ballot((d >> b) & 1);
-      eq &= ((d >> b) & 1) ? m : ~m;
+    for (int v = 0; v < RS_V; v++) {
+      // striped round v covers elements [v*256, (v+1)*256): round-major
+      // order equals element order, so cross-round ranks stay stable
+      int64_t i = ebase + (int64_t)v * THREADS + threadIdx.x;
+      bool valid = i < n;
+      key[v] = valid ? keys[i] : 0;
+      pl[v] = valid ? payload[i] : P(0);
+      int d = valid ? (int)((key[v] >> shift) & 255) : RS_RADIX;
+      dig[v] = d;
+      // 9-bit ballot multi-split (sentinel included, never counted)
+      unsigned long long eq = ~0ull;
+#pragma unroll
+      for (int b = 0; b < 9; b++) {
+        unsigned long long m = __ballot((d >> b) & 1);
+        eq &= ((d >> b) & 1) ? m : ~m;
+      }
+      rank[v] = __popcll(eq & ((1ull << lane) - 1ull));
+      if (valid && rank[v] == 0)
+        cnt[v][wave][d] = (uint32_t)__popcll(eq);
     }
-    int rank = __popcll(eq & ((1ull << lane) - 1ull));
-    if (valid && rank == 0) wave_cnt[wave][d] = (uint32_t)__popcll(eq);
     __syncthreads();
-    // one digit per thread: cross-wave prefix + wave-level shfl scan of
-    // the 256 tile totals into digit_start (no extra barrier rounds)
+    // one digit per thread: in-place (round, wave) prefix + wave-level
+    // shfl scan of the 256 tile totals into digit_start
     {
       int dd = threadIdx.x;
       uint32_t p = 0;
 #pragma unroll
-      for (int w = 0; w < WAVES; w++) {
-        wave_pref[w][dd] = p;
-        p += wave_cnt[w][dd];
-      }
+      for (int v = 0; v < RS_V; v++)
+#pragma unroll
+        for (int w = 0; w < WAVES; w++) {
+          uint32_t c = cnt[v][w][dd];
+          cnt[v][w][dd] = p;
+          p += c;
+        }
       tile_total[dd] = p;
       uint32_t v = p;  // inclusive scan over this wave's 64 digits
 #pragma unroll
@@ -386,26 +403,38 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
     }
     __syncthreads();
     // counting-sort the tile into LDS (digit-major, stable)
-    uint32_t tile_pos = 0;
-    if (valid) {
-      tile_pos = digit_start[d] + wave_pref[wave][d] + rank;
-      stage[tile_pos] = key;
-    }
+    uint32_t tile_pos[RS_V];
+#pragma unroll
+    for (int v = 0; v < RS_V; v++)
+      if (dig[v] < RS_RADIX) {
+        tile_pos[v] = digit_start[dig[v]] + cnt[v][wave][dig[v]]
+                      + rank[v];
+        stage[tile_pos[v]] = key[v];
+      }
     __syncthreads();
-    // coalesced write-out: thread j emits the tile's j-th digit-ordered
+    // coalesced write-out: slot j holds the tile's j-th digit-ordered
     // key; destination = digit's global cursor + offset within run
-    uint64_t k2 = 0;
-    uint32_t dst = 0;
-    if (threadIdx.x < tile_n) {
-      k2 = stage[threadIdx.x];
-      int d2 = (int)((k2 >> shift) & 255);
-      dst = cur[d2] + threadIdx.x - digit_start[d2];
-      okeys[dst] = k2;
+    uint32_t dst[RS_V];
+#pragma unroll
+    for (int v = 0; v < RS_V; v++) {
+      int j = v * THREADS + threadIdx.x;
+      if (j < tile_n) {
+        uint64_t k2 = stage[j];
+        int d2 = (int)((k2 >> shift) & 255);
+        dst[v] = cur[d2] + j - digit_start[d2];
+        okeys[dst[v]] = k2;
+      }
     }
     __syncthreads();
-    if (valid) stage[tile_pos] = (uint64_t)pl;
+#pragma unroll
+    for (int v = 0; v < RS_V; v++)
+      if (dig[v] < RS_RADIX) stage[tile_pos[v]] = (uint64_t)pl[v];
     __syncthreads();
-    if (threadIdx.x < tile_n) opayload[dst] = (P)stage[threadIdx.x];
+#pragma unroll
+    for (int v = 0; v < RS_V; v++) {
+      int j = v * THREADS + threadIdx.x;
+      if (j < tile_n) opayload[dst[v]] = (P)stage[j];
+    }
     __syncthreads();
     cur[threadIdx.x] += tile_total[threadIdx.x];
     // next tile's reads of cur happen after its own barriers
