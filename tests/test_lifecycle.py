@@ -291,3 +291,43 @@ def test_optimize_after_full_bucket_deletion(tmp_path, monkeypatch):
     k2 = t2.column("key").to_numpy()
     out = df.filter("key = 5").select("key", "val").collect()
     assert out.num_rows == int((k2 == 5).sum())
+
+
+def test_optimize_full_vs_quick_thresholds(env, tmp_path):
+    """quick mode only compacts buckets whose files sit under the
+    fileSizeThreshold; full mode compacts every multi-file bucket
+    (reference OptimizeAction.scala:57-148)."""
+    session, h, df, data, rng = env
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    h.create_index(df, hs.CoveringIndexConfig("opt2", ["key"], ["val"]))
+    # two refreshes -> up to 3 files per bucket
+    for i in range(2):
+        t = pa.table({"key": rng.integers(0, 100, 2000),
+                      "val": rng.random(2000)})
+        pq.write_table(t, str(data / f"part-opt{i}.parquet"))
+        h.refresh_index("opt2", "incremental")
+    entry = session.index_manager().get_index("opt2")
+    from collections import Counter
+    from hyperspace_amd.sources.parquet_io import bucket_id_of_file
+    per_bucket = Counter(bucket_id_of_file(p)
+                         for p in entry.content.os_files())
+    assert max(per_bucket.values()) >= 2
+    # quick with a tiny threshold: nothing qualifies -> NoChanges no-op
+    session.conf.set(hs.IndexConstants.OPTIMIZE_FILE_SIZE_THRESHOLD, 1)
+    h.optimize_index("opt2", "quick")
+    entry_q = session.index_manager().get_index("opt2")
+    per_bucket_q = Counter(bucket_id_of_file(p)
+                           for p in entry_q.content.os_files())
+    assert per_bucket_q == per_bucket  # unchanged
+    # full ignores the threshold: every multi-file bucket compacts to 1
+    h.optimize_index("opt2", "full")
+    entry_f = session.index_manager().get_index("opt2")
+    per_bucket_f = Counter(bucket_id_of_file(p)
+                           for p in entry_f.content.os_files())
+    assert max(per_bucket_f.values()) == 1
+    # results unchanged
+    session.enable_hyperspace()
+    n = df.filter("key = 7").collect().num_rows
+    session.disable_hyperspace()
+    assert n == df.filter("key = 7").collect().num_rows
