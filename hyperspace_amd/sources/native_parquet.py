@@ -680,10 +680,12 @@ def _walk_row_group(rg, pf_schema, data, want
             continue
         codec = col.compression.upper()
         # SNAPPY decodes on device (or host for copy-dense pages);
-        # GZIP/ZSTD/BROTLI pages decode on host codec threads but keep
-        # the native page-assembly path (no pyarrow table fallback)
+        # GZIP/ZSTD/BROTLI/LZ4 pages decode on host codec threads but
+        # keep the native page-assembly path (no pyarrow table
+        # fallback).  Parquet "LZ4"/"LZ4_RAW" pages are raw LZ4 blocks
+        # (arrow writes block format for both) -> pa.Codec("lz4_raw").
         if codec not in ("UNCOMPRESSED", "SNAPPY", "GZIP", "ZSTD",
-                         "BROTLI"):
+                         "BROTLI", "LZ4", "LZ4_RAW"):
             return None
         encs = set(col.encodings)
         is_dict = bool(encs & {"PLAIN_DICTIONARY", "RLE_DICTIONARY"})
@@ -781,6 +783,12 @@ def _walk_row_group(rg, pf_schema, data, want
                     pages.append(("dict_z", values_off, page_end,
                                   num_values, unc, has_levels))
                 elif pe == ENC_PLAIN:
+                    if is_string:
+                        # compressed PLAIN byte-array page (dictionary
+                        # overflowed mid-chunk): treating the variable-
+                        # length payload as int32 codes would corrupt —
+                        # pyarrow fallback
+                        return None
                     pages.append(("plain_z", values_off, page_end,
                                   num_values, unc, has_levels))
                 else:

@@ -447,7 +447,10 @@ def read_files_batch_device(paths: List[str], device,
                     statuses.append(st)
                 if host_idx:
                     import pyarrow as _pa
-                    codec = _pa.Codec(c.codec.lower())
+                    # parquet LZ4 / LZ4_RAW pages are raw LZ4 blocks
+                    codec = _pa.Codec(
+                        "lz4_raw" if c.codec in ("LZ4", "LZ4_RAW")
+                        else c.codec.lower())
                     hview = buf.numpy()
                     for i in host_idx:
                         a2, b2, unc2 = segs[i]

@@ -84,10 +84,19 @@ def test_fallback_on_compressed(tmp_path, cols):
     assert lay2 is not None
     assert all(c.codec == "GZIP" for c in lay2[1])
     p3 = str(tmp_path / "t3.parquet")
+    # LZ4 (raw-block pages) parses natively too
     pq.write_table(pa.table(cols), p3, compression="LZ4")
-    assert read_native_layout(p3) is None
+    lay3 = read_native_layout(p3)
+    assert lay3 is not None
+    assert all(c.codec == "LZ4" for c in lay3[1])
     batch2, counts2 = read_files_batch([p3])
     assert batch2.num_rows == 50_000 and counts2 == [50_000]
+    # DataPageV2 files stay out of the native layout (pyarrow fallback)
+    p4 = str(tmp_path / "t4.parquet")
+    pq.write_table(pa.table(cols), p4, data_page_version="2.0")
+    assert read_native_layout(p4) is None
+    batch3, counts3 = read_files_batch([p4])
+    assert batch3.num_rows == 50_000 and counts3 == [50_000]
 
 
 def test_write_batch_parquet_uses_native(tmp_path, cols):

@@ -154,3 +154,28 @@ def test_empty_string_column_roundtrip(tmp_path):
     rb, _ = read_files_batch([p])
     assert rb.num_rows == 0
     assert isinstance(rb.column("s"), StringColumn)
+
+
+def test_compressed_string_dict_overflow_falls_back(tmp_path):
+    """A SNAPPY string chunk whose dictionary overflows mid-chunk mixes
+    dict-index and PLAIN byte-array pages; the compressed PLAIN string
+    payload must NOT be claimed by the native layout (it would be
+    decoded as int32 codes) — pyarrow fallback, content exact."""
+    from hyperspace_amd.sources.native_parquet import read_native_layout
+    vals = [f"s{i:07d}" for i in range(300_000)]
+    p = str(tmp_path / "ovf.parquet")
+    pq.write_table(pa.table({"s": vals}), p, compression="SNAPPY",
+                   use_dictionary=True,
+                   dictionary_pagesize_limit=64 * 1024,
+                   data_page_version="1.0")
+    encs = set(pq.ParquetFile(p).metadata.row_group(0).column(0)
+               .encodings)
+    assert "PLAIN" in encs and encs & {"PLAIN_DICTIONARY",
+                                       "RLE_DICTIONARY"}, encs
+    assert read_native_layout(p) is None
+    rb, counts = read_files_batch([p])
+    assert counts == [300_000]
+    got = rb.column("s")
+    assert isinstance(got, StringColumn)
+    assert got.to_numpy()[:3].tolist() == ["s0000000", "s0000001",
+                                           "s0000002"]

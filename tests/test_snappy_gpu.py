@@ -200,9 +200,9 @@ def test_snappy_mixed_dict_plain_chunk(tmp_path):
     assert np.allclose(batch.tensor("val").cpu().numpy(), val)
 
 
-@pytest.mark.parametrize("codec", ["zstd", "gzip"])
+@pytest.mark.parametrize("codec", ["zstd", "gzip", "lz4"])
 def test_zstd_gzip_pages_host_codec_path(tmp_path, codec):
-    """ZSTD/GZIP pages keep the native page-assembly path: host codec
+    """ZSTD/GZIP/LZ4 pages keep the native page-assembly path: host codec
     threads decompress into device scratch, everything downstream
     (copy_unaligned, dict decode) is unchanged."""
     rng = np.random.default_rng(29)
