@@ -449,6 +449,53 @@ std::vector<torch::Tensor> parse_rle_runs_batch(
           mk(counts)};
 }
 
+// PLAIN BYTE_ARRAY page parse (K1 string path): each value is a 4-byte
+// little-endian length prefix + payload.  Concatenates the values of
+// every listed page into one arrow-style (int32 offsets, bytes) pair so
+// the caller can dictionary-encode in one pass (parquet-format.md
+// Encodings: PLAIN for BYTE_ARRAY; reference uses parquet-mr's reader).
+std::vector<torch::Tensor> parse_byte_arrays(
+    torch::Tensor bytes, torch::Tensor starts, torch::Tensor ends,
+    torch::Tensor counts) {
+  TORCH_CHECK(!bytes.is_cuda() && bytes.scalar_type() == torch::kUInt8,
+              "bytes must be a cpu u8 tensor");
+  const uint8_t* d = bytes.data_ptr<uint8_t>();
+  int64_t n_pages = starts.numel();
+  auto sa = starts.data_ptr<int64_t>();
+  auto ea = ends.data_ptr<int64_t>();
+  auto ca = counts.data_ptr<int64_t>();
+  int64_t total_n = 0, cap = 0;
+  for (int64_t p = 0; p < n_pages; ++p) {
+    total_n += ca[p];
+    cap += ea[p] - sa[p];
+  }
+  TORCH_CHECK(cap <= INT32_MAX, "byte-array payload exceeds int32 offsets");
+  auto offs = torch::empty({total_n + 1}, torch::dtype(torch::kInt32));
+  auto out = torch::empty({cap}, torch::dtype(torch::kUInt8));
+  int32_t* op = offs.data_ptr<int32_t>();
+  uint8_t* bp = out.data_ptr<uint8_t>();
+  int64_t written = 0, vi = 0;
+  for (int64_t p = 0; p < n_pages; ++p) {
+    int64_t pos = sa[p], end = ea[p];
+    for (int64_t k = 0; k < ca[p]; ++k) {
+      TORCH_CHECK(pos + 4 <= end, "byte-array page ", p,
+                  ": truncated length prefix");
+      uint32_t len = (uint32_t)d[pos] | ((uint32_t)d[pos + 1] << 8) |
+                     ((uint32_t)d[pos + 2] << 16) |
+                     ((uint32_t)d[pos + 3] << 24);
+      pos += 4;
+      TORCH_CHECK(pos + (int64_t)len <= end, "byte-array page ", p,
+                  ": value overruns page");
+      op[vi++] = (int32_t)written;
+      memcpy(bp + written, d + pos, len);
+      written += len;
+      pos += len;
+    }
+  }
+  op[vi] = (int32_t)written;
+  return {offs, out.narrow(0, 0, written)};
+}
+
 std::vector<torch::Tensor> parse_rle_runs(torch::Tensor bytes,
                                            int64_t start, int64_t end,
                                            int64_t bit_width,
@@ -612,6 +659,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("parse_rle_runs", &parse_rle_runs,
         py::call_guard<py::gil_scoped_release>(),
         "host parse of an RLE/bit-packed hybrid run table");
+  m.def("parse_byte_arrays", &parse_byte_arrays,
+        py::call_guard<py::gil_scoped_release>(),
+        "host PLAIN byte-array page parse -> (offsets, bytes)");
   m.def("rle_decode", &rle_decode,
         "device RLE/bit-packed dictionary-index decode");
 }

@@ -695,9 +695,10 @@ def _walk_row_group(rg, pf_schema, data, want
         is_string = False
         if col.physical_type == "BYTE_ARRAY":
             # dictionary-encoded strings decode natively (codes + one
-            # dictionary parse), uncompressed or snappy; PLAIN-encoded
-            # strings -> pyarrow
-            if not is_dict:
+            # dictionary parse), uncompressed or snappy; uncompressed
+            # PLAIN-encoded string pages parse host-side into codes
+            # ("splain"); compressed PLAIN strings -> pyarrow
+            if not is_dict and codec != "UNCOMPRESSED":
                 return None
             is_string = True
             np_dtype = np.dtype("int32")
@@ -809,8 +810,15 @@ def _walk_row_group(rg, pf_schema, data, want
                 bit_width = data[values_off]
                 pages.append(("dict", values_off + 1, page_end,
                               num_values, bit_width))
+            elif is_string and page_enc == ENC_PLAIN:
+                # PLAIN byte-array page: a whole-chunk PLAIN string
+                # column, or pyarrow's mid-chunk dictionary overflow.
+                # The (compacted) length-prefixed values parse on host
+                # (parse_byte_arrays) and dictionary-encode per chunk.
+                pages.append(("splain", values_off, page_end,
+                              num_values))
             elif is_string:
-                return None  # PLAIN byte-array page (dict overflow)
+                return None
             elif page_enc == ENC_PLAIN:
                 # also reached as the writer's mid-chunk fallback when a
                 # dictionary overflows: later pages switch to PLAIN
@@ -827,6 +835,8 @@ def _walk_row_group(rg, pf_schema, data, want
             return None
         if any(pg[0] == "dict" for pg in pages):
             enc_kind = "dict"
+        elif any(pg[0] == "splain" for pg in pages):
+            enc_kind = "splain"
         elif any(pg[0] == "dict_z" for pg in pages):
             enc_kind = "dict_z"
         elif any(pg[0] == "plain_z" for pg in pages):
@@ -882,6 +892,61 @@ def _decode_rle_indices(data, off: int, end: int, n: int, bw: int
     return out
 
 
+def decode_splain_pages(buf, starts: List[int], ends: List[int],
+                        counts: List[int]
+                        ) -> Tuple[np.ndarray, List[str]]:
+    """PLAIN byte-array pages -> (int32 codes over all pages' values
+    concatenated, SORTED dictionary).  ``buf`` is a cpu uint8 torch
+    tensor or a bytes-like.  The length-prefixed values parse in one
+    GIL-released C++ call (parse_byte_arrays) when the extension is
+    available, else a python walk; dictionary-encoding runs in arrow
+    C++ and the insertion-order dictionary is canonicalized to the
+    sorted StringColumn contract."""
+    import torch
+    from ..ops import native as _native
+    n = int(sum(counts))
+    if _native.available():
+        if not isinstance(buf, torch.Tensor):
+            import warnings
+            with warnings.catch_warnings():
+                warnings.simplefilter("ignore")
+                buf = torch.frombuffer(buf, dtype=torch.uint8)
+        offs, bts = _native.ext().parse_byte_arrays(
+            buf, torch.tensor(starts, dtype=torch.int64),
+            torch.tensor(ends, dtype=torch.int64),
+            torch.tensor(counts, dtype=torch.int64))
+        import pyarrow as pa
+        arr = pa.Array.from_buffers(
+            pa.utf8(), n,
+            [None, pa.py_buffer(offs.numpy()), pa.py_buffer(bts.numpy())])
+        denc = arr.dictionary_encode()
+        vals = denc.dictionary.to_pylist()
+        codes = denc.indices.to_numpy(zero_copy_only=False).astype(
+            np.int32, copy=False)
+    else:  # pure-python fallback (CPU testing without the extension)
+        data = buf.numpy().tobytes() if hasattr(buf, "numpy") else buf
+        raw: List[str] = []
+        for st, end, cnt in zip(starts, ends, counts):
+            pos = st
+            for _ in range(cnt):
+                ln = struct.unpack_from("<I", data, pos)[0]
+                pos += 4
+                raw.append(data[pos:pos + ln].decode("utf-8"))
+                pos += ln
+        vals = list(dict.fromkeys(raw))
+        vi0 = {v: i for i, v in enumerate(vals)}
+        codes = np.fromiter((vi0[v] for v in raw), np.int32, n)
+    if any(vals[i] > vals[i + 1] for i in range(len(vals) - 1)):
+        order = sorted(range(len(vals)), key=vals.__getitem__)
+        svals = [vals[i] for i in order]
+        lut = np.empty(len(vals), np.int32)
+        for rank, i in enumerate(order):
+            lut[i] = rank
+        codes = lut[codes]
+        return codes, svals
+    return np.ascontiguousarray(codes), list(vals)
+
+
 def read_native_host(path: str, columns: Optional[List[str]] = None
                      ) -> Optional[Tuple[Dict[str, np.ndarray],
                                          Dict[str, np.ndarray]]]:
@@ -894,7 +959,7 @@ def read_native_host(path: str, columns: Optional[List[str]] = None
         return None
     data, chunks = layout
     if any(not (c.encoding == "plain"
-                or (c.is_string and c.encoding == "dict"))
+                or (c.is_string and c.encoding in ("dict", "splain")))
            for c in chunks):
         return None  # numeric dict + snappy decode is the device path
     acc: Dict[str, list] = {}
@@ -910,12 +975,19 @@ def read_native_host(path: str, columns: Optional[List[str]] = None
                 str_dicts.setdefault(c.name, []).append([])
             continue
         if c.is_string:
-            values = c.dict_values(data)
+            values = c.dict_values(data) if c.dict_page else []
             for page, mask in zip(c.pages, c.page_masks):
-                _, p_start, p_end, nv, bw = page
-                n_valid = int(mask.sum()) if mask is not None else nv
-                codes = _decode_rle_indices(data, p_start, p_end,
-                                            n_valid, bw)
+                if page[0] == "splain":
+                    _, p_start, p_end, nv = page
+                    n_valid = int(mask.sum()) if mask is not None else nv
+                    codes, page_vals = decode_splain_pages(
+                        data, [p_start], [p_end], [n_valid])
+                else:
+                    _, p_start, p_end, nv, bw = page
+                    n_valid = int(mask.sum()) if mask is not None else nv
+                    codes = _decode_rle_indices(data, p_start, p_end,
+                                                n_valid, bw)
+                    page_vals = values
                 if mask is None:
                     part = codes
                     pm = np.ones(nv, dtype=bool)
@@ -926,7 +998,7 @@ def read_native_host(path: str, columns: Optional[List[str]] = None
                     any_null[c.name] = True
                 acc.setdefault(c.name, []).append(part)
                 macc.setdefault(c.name, []).append(pm)
-                str_dicts.setdefault(c.name, []).append(values)
+                str_dicts.setdefault(c.name, []).append(page_vals)
             continue
         for (_, off, nv), mask in zip(c.pages, c.page_masks):
             if mask is None:

@@ -707,14 +707,34 @@ def read_files_batch_device(paths: List[str], device,
                 continue
             dict_vals = None
             is_str = c.is_string
-            if c.encoding == "dict" and is_str:
+            chunk_dict = None
+            sp_codes: Dict[int, np.ndarray] = {}
+            sp_vals = None
+            if is_str:
                 # string chunk (K1): indices ARE the column (int32 codes
                 # into the host-parsed dictionary) — no value gather;
-                # per-chunk dictionaries merge after the sync below
-                str_chunks.append(
-                    (c.name, written,
-                     written + sum(p[3] for p in c.pages),
-                     c.dict_values(buf.numpy())))
+                # per-run str_chunks spans merge after the sync below.
+                # PLAIN byte-array pages (whole-chunk PLAIN, or the
+                # writer's mid-chunk dictionary overflow) parse host-
+                # side in one GIL-released call and dictionary-encode
+                # per chunk; their codes upload directly.
+                chunk_dict = (c.dict_values(buf.numpy())
+                              if c.dict_page else None)
+                sp = [(j, p) for j, p in enumerate(c.pages)
+                      if p[0] == "splain"]
+                if sp:
+                    from .native_parquet import decode_splain_pages
+                    sp_counts = [
+                        (int(c.page_masks[j].sum())
+                         if c.page_masks[j] is not None else p[3])
+                        for j, p in sp]
+                    all_codes, sp_vals = decode_splain_pages(
+                        buf, [p[1] for _, p in sp],
+                        [p[2] for _, p in sp], sp_counts)
+                    cur0 = 0
+                    for (j, _), nv2 in zip(sp, sp_counts):
+                        sp_codes[j] = all_codes[cur0:cur0 + nv2]
+                        cur0 += nv2
             elif c.encoding == "dict":
                 # K1 dictionary path: decode the dictionary page once;
                 # pages gather through it (a chunk can also carry PLAIN
@@ -728,8 +748,10 @@ def read_files_batch_device(paths: List[str], device,
                                    dict_n * itemsize)
                 dict_vals = dict_vals[:dict_n].contiguous()
             if not any(m is not None for m in c.page_masks):
-                for page in c.pages:
-                    nv = page[3] if page[0] == "dict" else page[2]
+                run_lo = written  # current dict-page run (string chunks)
+                for j, page in enumerate(c.pages):
+                    nv = page[3] if page[0] in ("dict", "splain") \
+                        else page[2]
                     if page[0] == "dict":
                         _, p_start, p_end, _, bw = page
                         runs = ext.parse_rle_runs(buf, p_start, p_end, bw,
@@ -741,28 +763,46 @@ def read_files_batch_device(paths: List[str], device,
                             out[c.name][written:written + nv] = \
                                 ext.gather_rows(dict_vals,
                                                 idx.to(_torch.int64))
+                    elif page[0] == "splain":
+                        if written > run_lo:
+                            str_chunks.append((c.name, run_lo, written,
+                                               chunk_dict))
+                        out[c.name][written:written + nv] = \
+                            _torch.from_numpy(
+                                np.ascontiguousarray(sp_codes[j])).to(
+                                device, non_blocking=True)
+                        str_chunks.append((c.name, written,
+                                           written + nv, sp_vals))
+                        run_lo = written + nv
                     else:
                         _, off, _ = page
                         ext.copy_unaligned(dev_bytes, off, out[c.name],
                                            written * itemsize,
                                            nv * itemsize)
                     written += nv
+                if is_str and written > run_lo:
+                    str_chunks.append((c.name, run_lo, written,
+                                       chunk_dict))
             else:
                 # nullable chunk: decode every page's compacted values
                 # into ONE temp, then a single chunk-wide scatter (per-
                 # page boolean indexing costs a kernel pair per page —
                 # thousands of launches on multi-page files)
-                n_rows = sum(p[3] if p[0] == "dict" else p[2]
-                             for p in c.pages)
+                n_rows = sum(p[3] if p[0] in ("dict", "splain")
+                             else p[2] for p in c.pages)
                 valid_counts = [
                     (int(m.sum()) if m is not None
-                     else (p[3] if p[0] == "dict" else p[2]))
+                     else (p[3] if p[0] in ("dict", "splain")
+                           else p[2]))
                     for p, m in zip(c.pages, c.page_masks)]
                 total_valid = sum(valid_counts)
                 tmp = _torch.empty(total_valid + 1,
                                    dtype=out[c.name].dtype, device=device)
                 cur = 0
-                for page, n_valid in zip(c.pages, valid_counts):
+                row_pos = written
+                run_lo = written
+                for j, (page, n_valid) in enumerate(
+                        zip(c.pages, valid_counts)):
                     if page[0] == "dict":
                         _, p_start, p_end, _, bw = page
                         runs = ext.parse_rle_runs(buf, p_start, p_end, bw,
@@ -772,16 +812,32 @@ def read_files_batch_device(paths: List[str], device,
                         tmp[cur:cur + n_valid] = idx if is_str else \
                             ext.gather_rows(dict_vals,
                                             idx.to(_torch.int64))
+                        row_pos += page[3]
+                    elif page[0] == "splain":
+                        if row_pos > run_lo:
+                            str_chunks.append((c.name, run_lo, row_pos,
+                                               chunk_dict))
+                        tmp[cur:cur + n_valid] = _torch.from_numpy(
+                            np.ascontiguousarray(sp_codes[j])).to(
+                            device, non_blocking=True)
+                        str_chunks.append((c.name, row_pos,
+                                           row_pos + page[3], sp_vals))
+                        row_pos += page[3]
+                        run_lo = row_pos
                     else:
                         _, off, _ = page
                         ext.copy_unaligned(dev_bytes, off, tmp,
                                            cur * itemsize,
                                            n_valid * itemsize)
+                        row_pos += page[2]
                     cur += n_valid
+                if is_str and row_pos > run_lo:
+                    str_chunks.append((c.name, run_lo, row_pos,
+                                       chunk_dict))
                 chunk_mask = np.concatenate([
                     (m if m is not None
-                     else np.ones(p[3] if p[0] == "dict" else p[2],
-                                  dtype=bool))
+                     else np.ones(p[3] if p[0] in ("dict", "splain")
+                                  else p[2], dtype=bool))
                     for p, m in zip(c.pages, c.page_masks)])
                 mask_dev = _torch.from_numpy(chunk_mask).to(device)
                 dst = out[c.name][written:written + n_rows]

@@ -179,3 +179,69 @@ def test_compressed_string_dict_overflow_falls_back(tmp_path):
     assert isinstance(got, StringColumn)
     assert got.to_numpy()[:3].tolist() == ["s0000000", "s0000001",
                                            "s0000002"]
+
+
+def _plain_string_file(tmp_path, name, vals, **kw):
+    p = str(tmp_path / name)
+    pq.write_table(pa.table({"s": vals}), p, compression="NONE",
+                   data_page_version="1.0", **kw)
+    return p
+
+
+def test_plain_string_pages_host_native(tmp_path):
+    """PLAIN (non-dictionary) byte-array pages parse natively: host
+    byte-array walk (parse_byte_arrays) + arrow dictionary-encode to
+    the sorted StringColumn contract (reference reads strings through
+    parquet-mr; here K1's splain path)."""
+    from hyperspace_amd.sources.native_parquet import (read_native_host,
+                                                       read_native_layout)
+    vals = [f"v{i % 977:04d}" for i in range(120_000)]
+    p = _plain_string_file(tmp_path, "sp.parquet", vals,
+                           use_dictionary=False)
+    lay = read_native_layout(p)
+    assert lay is not None
+    assert lay[1][0].encoding == "splain"
+    cols, masks = read_native_host(p)
+    s = cols["s"]
+    got = [s.values[c] for c in s.codes[::997]]
+    assert got == vals[::997]
+    assert all(s.values[i] <= s.values[i + 1]
+               for i in range(len(s.values) - 1))
+    rb, counts = read_files_batch([p])
+    assert counts == [120_000]
+    assert rb.column("s").to_numpy()[::997].tolist() == vals[::997]
+
+
+def test_mixed_dict_plain_string_chunk_host(tmp_path):
+    """Uncompressed dictionary-overflow chunks (dict pages then PLAIN
+    byte-array pages in ONE chunk) decode natively with per-run
+    dictionary merge."""
+    from hyperspace_amd.sources.native_parquet import (read_native_host,
+                                                       read_native_layout)
+    vals = [f"s{i:07d}" for i in range(300_000)]
+    p = _plain_string_file(tmp_path, "mx.parquet", vals,
+                           use_dictionary=True,
+                           dictionary_pagesize_limit=64 * 1024)
+    lay = read_native_layout(p)
+    assert lay is not None
+    kinds = {pg[0] for c in lay[1] for pg in c.pages}
+    assert kinds == {"dict", "splain"}, kinds
+    cols, _ = read_native_host(p)
+    s = cols["s"]
+    for i in (0, 1, 99_999, 100_000, 299_999):
+        assert s.values[s.codes[i]] == vals[i]
+    assert all(s.values[i] <= s.values[i + 1]
+               for i in range(len(s.values) - 1))
+
+
+def test_plain_string_pages_nullable_host(tmp_path):
+    from hyperspace_amd.sources.native_parquet import read_native_host
+    vals = [None if i % 7 == 0 else f"x{i % 50}" for i in range(50_000)]
+    p = _plain_string_file(tmp_path, "nl.parquet", vals,
+                           use_dictionary=False)
+    cols, masks = read_native_host(p)
+    s, m = cols["s"], masks["s"]
+    assert not m[0] and m[1]
+    for i in range(0, 50_000, 317):
+        assert (vals[i] is None and not m[i]) or \
+            s.values[s.codes[i]] == vals[i]
