@@ -421,3 +421,27 @@ def test_dataskipping_over_delta_table(tmp_path, monkeypatch):
     out = Executor(session).execute(plan)
     session.disable_hyperspace()
     assert out.num_rows == q.collect().num_rows
+
+
+def test_covering_index_outscores_dataskipping(env):
+    """When a covering filter index AND a data-skipping index both apply
+    to the same Filter(Scan), the score-based optimizer must pick the
+    covering rewrite (50 x coverage vs the DS flat score of 1;
+    reference: ApplyDataSkippingIndex.scala score + ScoreBased
+    optimizer)."""
+    session, h, df, _, _ = env
+    h.create_index(df, hs.DataSkippingIndexConfig(
+        "ds_both", hs.MinMaxSketch("key")))
+    h.create_index(df, hs.CoveringIndexConfig(
+        "ci_both", ["key"], ["val"]))
+    session.enable_hyperspace()
+    q = df.filter("key = 1500").select("key", "val")
+    plan = q.optimized_plan().pretty()
+    assert "IndexScan(ci_both" in plan, plan
+    assert "dataskipping" not in plan, plan
+    # with the covering index gone, the DS index takes over
+    h.delete_index("ci_both")
+    h.vacuum_index("ci_both")
+    q2 = df.filter("key = 1500").select("key", "val")
+    plan2 = q2.optimized_plan().pretty()
+    assert "dataskipping:-7files" in plan2, plan2
