@@ -4,8 +4,8 @@
 // no CUDA-compat paths.  These implement the reference's Spark data-plane
 // operators:
 //   K2  murmur3 bucket hash (Spark Murmur3_x86_32-compatible, seed 42)
-//   K3  stable LSD radix sort (4-bit digits, ballot multi-split ranking,
-//       constant-nibble pass skipping)
+//   K3  stable LSD radix sort (8/9-bit digits, ballot multi-split
+//       ranking, constant-byte pass skipping / 9-bit pass planning)
 //   K4  segmented sorted merge join (count + emit)
 //   K7  sorted-set membership (lineage delete filter)
 //   K8  segmented min/max + bloom filter build/probe
@@ -23,6 +23,7 @@
 #include <cassert>
 #include <cstdint>
 #include <cstdio>
+#include <cstdlib>
 
 #include "kernels.h"
 
@@ -228,7 +229,7 @@ void normalize_key(const void* vals, int dtype, uint64_t* out, int64_t n,
 }
 
 // ---------------------------------------------------------------------------
-// Stable LSD radix sort (4-bit digits, wave64 ballot multi-split)
+// Stable LSD radix sort (8- or 9-bit digits, wave64 ballot multi-split)
 // ---------------------------------------------------------------------------
 
 constexpr int RS_RADIX = 256;      // 8-bit digits
@@ -244,7 +245,8 @@ static inline int rs_num_blocks(int64_t n) {
 
 int64_t radix_sort_hist_size(int64_t n) {
   (void)n;
-  return (int64_t)RS_RADIX * RS_MAX_BLOCKS;
+  // sized for the widest digit (9-bit / 512-bin passes)
+  return (int64_t)512 * RS_MAX_BLOCKS;
 }
 
 __global__ void k_xor_or_reduce(const uint64_t* __restrict__ keys, int64_t n,
@@ -266,23 +268,24 @@ __global__ void k_xor_or_reduce(const uint64_t* __restrict__ keys, int64_t n,
     atomicOr((unsigned long long*)out_mask, (unsigned long long)lds[0]);
 }
 
+template <int RADIX>
 __global__ void k_rs_hist(const uint64_t* __restrict__ keys, int64_t n,
                           int shift, uint32_t* __restrict__ hist, int nb,
                           int64_t tiles_per_block) {
   // CONTIGUOUS tile->block ranges: block-major order must equal element
   // order or cross-pass LSD stability breaks.
-  __shared__ uint32_t lh[RS_RADIX];
-  for (int d = threadIdx.x; d < RS_RADIX; d += blockDim.x) lh[d] = 0;
+  __shared__ uint32_t lh[RADIX];
+  for (int d = threadIdx.x; d < RADIX; d += blockDim.x) lh[d] = 0;
   __syncthreads();
   int64_t e0 = (int64_t)blockIdx.x * tiles_per_block * RS_TILE;
   int64_t e1 = e0 + tiles_per_block * RS_TILE;
   if (e1 > n) e1 = n;
   for (int64_t i = e0 + threadIdx.x; i < e1; i += THREADS) {
-    int d = (int)((keys[i] >> shift) & 255);
+    int d = (int)((keys[i] >> shift) & (RADIX - 1));
     atomicAdd(&lh[d], 1u);
   }
   __syncthreads();
-  for (int d = threadIdx.x; d < RS_RADIX; d += blockDim.x)
+  for (int d = threadIdx.x; d < RADIX; d += blockDim.x)
     hist[(int64_t)d * nb + blockIdx.x] = lh[d];
 }
 
@@ -312,7 +315,7 @@ __global__ void k_scan_u32(uint32_t* __restrict__ data, int64_t n) {
   }
 }
 
-template <typename P>
+template <typename P, int RADIX>
 __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
                              const P* __restrict__ payload,
                              uint64_t* __restrict__ okeys,
@@ -323,18 +326,22 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
   // (striped rounds, preserving element order), the 1024-key tile is
   // counting-sorted into LDS by digit, then written out with
   // threadIdx-contiguous reads — global stores become per-digit-run
-  // coalesced bursts instead of 256-way scattered singles, and the
-  // per-tile barrier cost amortizes over 4x the keys (the round-1
+  // coalesced bursts instead of RADIX-way scattered singles, and the
+  // per-tile barrier cost amortizes over RS_V x the keys (the round-1
   // k_rs_scatter showed 32.6% issue-stall on the scattered stores).
-  __shared__ uint32_t cur[RS_RADIX];
+  // RADIX 256 = byte passes; RADIX 512 = 9-bit passes (one fewer pass
+  // for <=27 varying bits at ~1.5x the LDS per block).
+  static_assert(RADIX % THREADS == 0, "digits split evenly over threads");
+  constexpr int D = RADIX / THREADS;      // digits owned per thread
+  constexpr int DBITS = 32 - __builtin_clz((unsigned)RADIX);  // sentinel
+  __shared__ uint32_t cur[RADIX];
   // per (round, wave, digit) counts, scanned IN PLACE into prefixes
-  __shared__ uint32_t cnt[RS_V][WAVES][RS_RADIX];
-  __shared__ uint32_t tile_total[RS_RADIX];
-  __shared__ uint32_t digit_start[RS_RADIX];  // excl scan of tile_total
+  __shared__ uint32_t cnt[RS_V][WAVES][RADIX];
+  __shared__ uint32_t tile_total[RADIX];
+  __shared__ uint32_t digit_start[RADIX];  // excl scan of tile_total
   __shared__ uint32_t wsum[WAVES];
   __shared__ uint64_t stage[RS_TILE];  // keys, then payload (reused)
-  static_assert(RS_RADIX == THREADS, "one digit per thread");
-  for (int d = threadIdx.x; d < RS_RADIX; d += blockDim.x)
+  for (int d = threadIdx.x; d < RADIX; d += blockDim.x)
     cur[d] = hist[(int64_t)d * nb + blockIdx.x];
   int lane = threadIdx.x & 63;
   int wave = threadIdx.x >> 6;
@@ -347,7 +354,7 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
     P pl[RS_V];
     int dig[RS_V];
     int rank[RS_V];
-    for (int x = threadIdx.x; x < RS_V * WAVES * RS_RADIX;
+    for (int x = threadIdx.x; x < RS_V * WAVES * RADIX;
          x += blockDim.x)
       ((uint32_t*)cnt)[x] = 0;
     __syncthreads();
@@ -359,12 +366,12 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
       bool valid = i < n;
       key[v] = valid ? keys[i] : 0;
       pl[v] = valid ? payload[i] : P(0);
-      int d = valid ? (int)((key[v] >> shift) & 255) : RS_RADIX;
+      int d = valid ? (int)((key[v] >> shift) & (RADIX - 1)) : RADIX;
       dig[v] = d;
-      // 9-bit ballot multi-split (sentinel included, never counted)
+      // DBITS-bit ballot multi-split (sentinel included, never counted)
       unsigned long long eq = ~0ull;
 #pragma unroll
-      for (int b = 0; b < 9; b++) {
+      for (int b = 0; b < DBITS; b++) {
         unsigned long long m = __ballot((d >> b) & 1);
         eq &= ((d >> b) & 1) ? m : ~m;
       }
@@ -373,21 +380,29 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
         cnt[v][wave][d] = (uint32_t)__popcll(eq);
     }
     __syncthreads();
-    // one digit per thread: in-place (round, wave) prefix + wave-level
-    // shfl scan of the 256 tile totals into digit_start
+    // D consecutive digits per thread: in-place (round, wave) prefix +
+    // wave-level shfl scan of the per-thread sums into digit_start
+    // (thread-major consecutive ownership keeps digit order)
     {
-      int dd = threadIdx.x;
-      uint32_t p = 0;
+      uint32_t tot[D];
+      uint32_t tsum = 0;
 #pragma unroll
-      for (int v = 0; v < RS_V; v++)
+      for (int q = 0; q < D; q++) {
+        int dd = threadIdx.x * D + q;
+        uint32_t p = 0;
 #pragma unroll
-        for (int w = 0; w < WAVES; w++) {
-          uint32_t c = cnt[v][w][dd];
-          cnt[v][w][dd] = p;
-          p += c;
-        }
-      tile_total[dd] = p;
-      uint32_t v = p;  // inclusive scan over this wave's 64 digits
+        for (int v = 0; v < RS_V; v++)
+#pragma unroll
+          for (int w = 0; w < WAVES; w++) {
+            uint32_t c = cnt[v][w][dd];
+            cnt[v][w][dd] = p;
+            p += c;
+          }
+        tile_total[dd] = p;
+        tot[q] = p;
+        tsum += p;
+      }
+      uint32_t v = tsum;  // inclusive scan over this wave's 64 threads
 #pragma unroll
       for (int off = 1; off < 64; off <<= 1) {
         uint32_t t = __shfl_up(v, off);
@@ -399,14 +414,20 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
 #pragma unroll
       for (int w = 0; w < WAVES; w++)
         if (w < wave) wbase += wsum[w];
-      digit_start[dd] = wbase + v - p;  // exclusive scan
+      uint32_t base = wbase + v - tsum;  // exclusive over threads
+#pragma unroll
+      for (int q = 0; q < D; q++) {
+        int dd = threadIdx.x * D + q;
+        digit_start[dd] = base;
+        base += tot[q];
+      }
     }
     __syncthreads();
     // counting-sort the tile into LDS (digit-major, stable)
     uint32_t tile_pos[RS_V];
 #pragma unroll
     for (int v = 0; v < RS_V; v++)
-      if (dig[v] < RS_RADIX) {
+      if (dig[v] < RADIX) {
         tile_pos[v] = digit_start[dig[v]] + cnt[v][wave][dig[v]]
                       + rank[v];
         stage[tile_pos[v]] = key[v];
@@ -420,7 +441,7 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
       int j = v * THREADS + threadIdx.x;
       if (j < tile_n) {
         uint64_t k2 = stage[j];
-        int d2 = (int)((k2 >> shift) & 255);
+        int d2 = (int)((k2 >> shift) & (RADIX - 1));
         dst[v] = cur[d2] + j - digit_start[d2];
         okeys[dst[v]] = k2;
       }
@@ -428,7 +449,7 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
     __syncthreads();
 #pragma unroll
     for (int v = 0; v < RS_V; v++)
-      if (dig[v] < RS_RADIX) stage[tile_pos[v]] = (uint64_t)pl[v];
+      if (dig[v] < RADIX) stage[tile_pos[v]] = (uint64_t)pl[v];
     __syncthreads();
 #pragma unroll
     for (int v = 0; v < RS_V; v++) {
@@ -436,7 +457,11 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
       if (j < tile_n) opayload[dst[v]] = (P)stage[j];
     }
     __syncthreads();
-    cur[threadIdx.x] += tile_total[threadIdx.x];
+#pragma unroll
+    for (int q = 0; q < D; q++) {
+      int dd = threadIdx.x * D + q;
+      cur[dd] += tile_total[dd];
+    }
     // next tile's reads of cur happen after its own barriers
   }
 }
@@ -465,17 +490,49 @@ static void radix_sort_impl(uint64_t* keys, P* payload, uint64_t* tmp_keys,
   uint64_t* kb = tmp_keys;
   P* pa = payload;
   P* pb = tmp_payload;
-  for (int shift = 0; shift < 64; shift += 8) {
-    if (((mask >> shift) & 255ull) == 0) continue;  // constant byte
+  // pass planning: byte-aligned digits with constant-byte skipping vs
+  // 9-bit digits anchored at the lowest varying bit — take whichever
+  // needs fewer passes (26-bit bench keys: 4 byte passes -> 3 nine-bit
+  // passes).  Both cover every varying bit exactly once in LSB->MSB
+  // order, so LSD stability is preserved.
+  int shifts8[8], n8 = 0;
+  for (int s = 0; s < 64; s += 8)
+    if ((mask >> s) & 255ull) shifts8[n8++] = s;
+  int shifts9[8], n9 = 0;
+  if (mask) {
+    for (int s = __builtin_ctzll(mask); s < 64; s += 9)
+      if ((mask >> s) & 511ull) shifts9[n9++] = s;
+  }
+  static const bool force8 = [] {  // same-box A/B knob
+    const char* e = getenv("HS_RS_FORCE8");
+    return e && e[0] == '1';
+  }();
+  bool use9 = n9 < n8 && !force8;
+  const int* shifts = use9 ? shifts9 : shifts8;
+  int npass = use9 ? n9 : n8;
+  int radix = use9 ? 512 : 256;
+  for (int pi = 0; pi < npass; pi++) {
+    int shift = shifts[pi];
     HIP_CHECK(hipMemsetAsync(hist, 0,
-                             (size_t)RS_RADIX * nb * sizeof(uint32_t),
+                             (size_t)radix * nb * sizeof(uint32_t),
                              stream));
-    hipLaunchKernelGGL(k_rs_hist, dim3(nb), dim3(THREADS), 0, stream, ka, n,
-                       shift, hist, nb, tpb);
-    hipLaunchKernelGGL(k_scan_u32, dim3(1), dim3(1024), 0, stream, hist,
-                       (int64_t)RS_RADIX * nb);
-    hipLaunchKernelGGL(k_rs_scatter<P>, dim3(nb), dim3(THREADS), 0, stream,
-                       ka, pa, kb, pb, n, shift, hist, nb, tpb);
+    if (use9) {
+      hipLaunchKernelGGL((k_rs_hist<512>), dim3(nb), dim3(THREADS), 0,
+                         stream, ka, n, shift, hist, nb, tpb);
+      hipLaunchKernelGGL(k_scan_u32, dim3(1), dim3(1024), 0, stream, hist,
+                         (int64_t)512 * nb);
+      hipLaunchKernelGGL((k_rs_scatter<P, 512>), dim3(nb), dim3(THREADS),
+                         0, stream, ka, pa, kb, pb, n, shift, hist, nb,
+                         tpb);
+    } else {
+      hipLaunchKernelGGL((k_rs_hist<256>), dim3(nb), dim3(THREADS), 0,
+                         stream, ka, n, shift, hist, nb, tpb);
+      hipLaunchKernelGGL(k_scan_u32, dim3(1), dim3(1024), 0, stream, hist,
+                         (int64_t)256 * nb);
+      hipLaunchKernelGGL((k_rs_scatter<P, 256>), dim3(nb), dim3(THREADS),
+                         0, stream, ka, pa, kb, pb, n, shift, hist, nb,
+                         tpb);
+    }
     uint64_t* tk = ka; ka = kb; kb = tk;
     P* tp = pa; pa = pb; pb = tp;
   }
