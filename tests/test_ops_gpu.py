@@ -91,6 +91,24 @@ def test_radix_sort_small_range_pass_skipping():
     assert torch.equal(cp, gp.cpu())
 
 
+@pytest.mark.parametrize("lo,hi", [
+    (0, 1 << 26),            # bench shape: 26 bits -> three 9-bit passes
+    (1 << 8, 1 << 17),       # bits 8..16 -> one 9-bit pass
+    (0, 1 << 18),            # 18 bits -> two 9-bit passes
+])
+def test_radix_sort_nine_bit_planner(lo, hi):
+    """Key shapes where the 9-bit pass planner beats byte passes must
+    still sort stably and bit-identically to the CPU oracle."""
+    rng = np.random.default_rng(lo % 13 + hi % 7)
+    keys = cpu_ref.normalize_key(torch.from_numpy(
+        rng.integers(lo, hi, 2_000_000, dtype=np.int64)))
+    payload = torch.arange(2_000_000, dtype=torch.int64)
+    ck, cp = cpu_ref.stable_sort_u64(keys, payload)
+    gk, gp = ops.sort_pairs(keys.cuda(), payload.cuda())
+    assert torch.equal(ck, gk.cpu())
+    assert torch.equal(cp, gp.cpu())
+
+
 def test_merge_join_matches_cpu():
     rng = np.random.default_rng(7)
     nseg = 16
