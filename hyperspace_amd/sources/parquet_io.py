@@ -452,24 +452,34 @@ def read_files_batch_device(paths: List[str], device,
                         "lz4_raw" if c.codec in ("LZ4", "LZ4_RAW")
                         else c.codec.lower())
                     hview = buf.numpy()
+                    # stage all host-decoded pages in ONE pinned buffer:
+                    # the codec reads the compressed page zero-copy from
+                    # the read buffer, one host memcpy lands it in
+                    # pinned memory, and the H2D is a true async copy
+                    # (the old per-page pageable copy_ serialized the
+                    # decode stream on every page)
+                    total_host = sum(segs[i][2] for i in host_idx)
+                    hstage = _pinned_get(total_host)
+                    keepalive.append(hstage)
+                    hs_np = hstage.numpy()
+                    hoff = 0
                     for i in host_idx:
                         a2, b2, unc2 = segs[i]
                         try:
-                            dec = codec.decompress(
-                                hview[a2:b2].tobytes(), unc2)
+                            dec = codec.decompress(hview[a2:b2], unc2)
                         except Exception:  # noqa: BLE001
+                            dec = None
+                        if dec is None or len(dec) != unc2:
                             statuses.append(_torch.ones(
                                 1, dtype=_torch.int32, device=device))
+                            hoff += unc2
                             continue
-                        if len(dec) != unc2:
-                            statuses.append(_torch.ones(
-                                1, dtype=_torch.int32, device=device))
-                            continue
-                        hb_np = np.frombuffer(dec, dtype=np.uint8)
+                        hs_np[hoff:hoff + unc2] = np.frombuffer(
+                            dec, dtype=np.uint8)
                         scratch[int(doff_all[i]):
                                 int(doff_all[i]) + unc2].copy_(
-                            _torch.from_numpy(hb_np.copy()),
-                            non_blocking=True)
+                            hstage[hoff:hoff + unc2], non_blocking=True)
+                        hoff += unc2
                 page_base = doff_all[1:-1] if has_zdict \
                     else doff_all[:-1]
                 dict_vals = None
@@ -861,6 +871,7 @@ def read_files_batch_device(paths: List[str], device,
 
     import threading
     dev_bufs: List[Optional["_torch.Tensor"]] = [None] * len(paths)
+    keepalive: List["_torch.Tensor"] = []  # pinned codec staging buffers
     upload_events: List[Optional["_torch.cuda.Event"]] = \
         [None] * len(paths)
     upload_lock = threading.Lock()
@@ -945,6 +956,8 @@ def read_files_batch_device(paths: List[str], device,
         for bq in bufs:
             if bq is not None:
                 _pinned_put(bq)
+        for bq in keepalive:
+            _pinned_put(bq)
         return fallback()
     _t2 = _time.perf_counter()    # order the default stream after every worker stream, then host-sync
     # so the pinned buffers can be recycled
@@ -965,8 +978,12 @@ def read_files_batch_device(paths: List[str], device,
         for b in bufs:
             if b is not None and b is not False:
                 _pinned_put(b)
+        for b in keepalive:
+            _pinned_put(b)
         return fallback()
     for b in bufs:
+        _pinned_put(b)
+    for b in keepalive:
         _pinned_put(b)
     if string_cols:
         from ..execution.columnar import StringColumn
