@@ -24,6 +24,7 @@
 #include <cstdint>
 #include <cstdio>
 #include <cstdlib>
+#include <type_traits>
 
 #include "kernels.h"
 
@@ -336,12 +337,15 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
   constexpr int DBITS = 32 - __builtin_clz((unsigned)RADIX);  // sentinel
   __shared__ uint32_t cur[RADIX];
   // per (round, wave, digit) counts, scanned IN PLACE into prefixes.
-  // u16: every value here is bounded by the tile size (1024), and the
-  // 512-bin variant at u32 cost 30 KB LDS -> 5 blocks/CU (measured 10%
-  // SLOWER than 4 byte passes); u16 restores 8 blocks/CU.
-  __shared__ uint16_t cnt[RS_V][WAVES][RADIX];
-  __shared__ uint16_t tile_total[RADIX];
-  __shared__ uint16_t digit_start[RADIX];  // excl scan of tile_total
+  // Counter width: every value is bounded by the tile size (1024) so
+  // u16 suffices; the 512-bin variant uses it to halve its LDS (30 KB
+  // at u32 -> 5 blocks/CU), the proven 256-bin path keeps u32 (u16
+  // measured ~noise-level slower there; profiles/r2_summary.md).
+  using cnt_t =
+      typename std::conditional<(RADIX > 256), uint16_t, uint32_t>::type;
+  __shared__ cnt_t cnt[RS_V][WAVES][RADIX];
+  __shared__ cnt_t tile_total[RADIX];
+  __shared__ cnt_t digit_start[RADIX];  // excl scan of tile_total
   __shared__ uint32_t wsum[WAVES];
   __shared__ uint64_t stage[RS_TILE];  // keys, then payload (reused)
   for (int d = threadIdx.x; d < RADIX; d += blockDim.x)
@@ -357,9 +361,10 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
     P pl[RS_V];
     int dig[RS_V];
     int rank[RS_V];
-    for (int x = threadIdx.x; x < RS_V * WAVES * RADIX / 2;
+    for (int x = threadIdx.x;
+         x < RS_V * WAVES * RADIX * (int)sizeof(cnt_t) / 4;
          x += blockDim.x)
-      ((uint32_t*)cnt)[x] = 0;  // paired u16 zeroing
+      ((uint32_t*)cnt)[x] = 0;  // word-packed zeroing
     __syncthreads();
 #pragma unroll
     for (int v = 0; v < RS_V; v++) {
@@ -380,7 +385,7 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
       }
       rank[v] = __popcll(eq & ((1ull << lane) - 1ull));
       if (valid && rank[v] == 0)
-        cnt[v][wave][d] = (uint16_t)__popcll(eq);
+        cnt[v][wave][d] = (cnt_t)__popcll(eq);
     }
     __syncthreads();
     // D consecutive digits per thread: in-place (round, wave) prefix +
@@ -392,14 +397,14 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
 #pragma unroll
       for (int q = 0; q < D; q++) {
         int dd = threadIdx.x * D + q;
-        uint16_t p = 0;
+        cnt_t p = 0;
 #pragma unroll
         for (int v = 0; v < RS_V; v++)
 #pragma unroll
           for (int w = 0; w < WAVES; w++) {
-            uint16_t c = cnt[v][w][dd];
+            cnt_t c = cnt[v][w][dd];
             cnt[v][w][dd] = p;
-            p = (uint16_t)(p + c);
+            p = (cnt_t)(p + c);
           }
         tile_total[dd] = p;
         tot[q] = p;
@@ -421,7 +426,7 @@ __global__ void k_rs_scatter(const uint64_t* __restrict__ keys,
 #pragma unroll
       for (int q = 0; q < D; q++) {
         int dd = threadIdx.x * D + q;
-        digit_start[dd] = (uint16_t)base;
+        digit_start[dd] = (cnt_t)base;
         base += tot[q];
       }
     }
@@ -506,11 +511,16 @@ static void radix_sort_impl(uint64_t* keys, P* payload, uint64_t* tmp_keys,
     for (int s = __builtin_ctzll(mask); s < 64; s += 9)
       if ((mask >> s) & 511ull) shifts9[n9++] = s;
   }
-  static const bool force8 = [] {  // same-box A/B knob
-    const char* e = getenv("HS_RS_FORCE8");
+  // 9-bit passes saved one pass for 26-bit keys but measured ~10%
+  // slower end-to-end on MI355X (the scatter is wave-cap bound, and the
+  // wider digit costs an extra ballot + 2-digit scan per thread) —
+  // byte passes stay the default; HS_RS_9BIT=1 re-enables the wide
+  // planner for future tuning (profiles/r2_summary.md).
+  static const bool want9 = [] {
+    const char* e = getenv("HS_RS_9BIT");
     return e && e[0] == '1';
   }();
-  bool use9 = n9 < n8 && !force8;
+  bool use9 = n9 < n8 && want9;
   const int* shifts = use9 ? shifts9 : shifts8;
   int npass = use9 ? n9 : n8;
   int radix = use9 ? 512 : 256;
