@@ -78,6 +78,21 @@ def _pinned_put(buf: "torch.Tensor") -> None:
             _PINNED_POOL_BYTES += size
 
 
+_CODEC_POOL = None
+
+
+def _codec_pool():
+    """Shared host-codec thread pool (gzip/zstd/lz4/copy-dense-snappy
+    page decompression; the codecs release the GIL)."""
+    global _CODEC_POOL
+    if _CODEC_POOL is None:
+        from concurrent.futures import ThreadPoolExecutor
+        _CODEC_POOL = ThreadPoolExecutor(
+            max_workers=int(os.environ.get("HS_CODEC_WORKERS", "16")),
+            thread_name_prefix="hs-codec")
+    return _CODEC_POOL
+
+
 def bucket_id_of_file(path: str) -> Optional[int]:
     m = _BUCKET_RE.match(os.path.basename(path))
     return int(m.group(1)) if m else None
@@ -462,11 +477,19 @@ def read_files_batch_device(paths: List[str], device,
                     hstage = _pinned_get(total_host)
                     keepalive.append(hstage)
                     hs_np = hstage.numpy()
+                    # fan the per-page codec calls over the shared codec
+                    # pool — pyarrow codecs release the GIL, and a large
+                    # column chunk's copy-dense pages otherwise
+                    # decompress serially on this one decode thread
+                    futs = [_codec_pool().submit(
+                        codec.decompress, hview[s0:s1], u)
+                        for i in host_idx
+                        for s0, s1, u in (segs[i],)]
                     hoff = 0
-                    for i in host_idx:
-                        a2, b2, unc2 = segs[i]
+                    for i, fut in zip(host_idx, futs):
+                        unc2 = segs[i][2]
                         try:
-                            dec = codec.decompress(hview[a2:b2], unc2)
+                            dec = fut.result()
                         except Exception:  # noqa: BLE001
                             dec = None
                         if dec is None or len(dec) != unc2:
