@@ -784,13 +784,12 @@ def _walk_row_group(rg, pf_schema, data, want
                     pages.append(("dict_z", values_off, page_end,
                                   num_values, unc, has_levels))
                 elif pe == ENC_PLAIN:
-                    if is_string:
-                        # compressed PLAIN byte-array page (dictionary
-                        # overflowed mid-chunk): treating the variable-
-                        # length payload as int32 codes would corrupt —
-                        # pyarrow fallback
-                        return None
-                    pages.append(("plain_z", values_off, page_end,
+                    # compressed PLAIN byte-array pages (whole-chunk
+                    # PLAIN strings, or a mid-chunk dictionary
+                    # overflow) decompress like any page, then the
+                    # payload D2Hs for the host byte-array parse
+                    kind = "splain_z" if is_string else "plain_z"
+                    pages.append((kind, values_off, page_end,
                                   num_values, unc, has_levels))
                 else:
                     return None
@@ -839,6 +838,8 @@ def _walk_row_group(rg, pf_schema, data, want
             enc_kind = "splain"
         elif any(pg[0] == "dict_z" for pg in pages):
             enc_kind = "dict_z"
+        elif any(pg[0] == "splain_z" for pg in pages):
+            enc_kind = "splain_z"
         elif any(pg[0] == "plain_z" for pg in pages):
             enc_kind = "plain_z"
         else:

@@ -414,7 +414,7 @@ def read_files_batch_device(paths: List[str], device,
             written = cursors[c.name]
             if _dt == "3":
                 t_c0 = _time.perf_counter()
-            if c.encoding in ("plain_z", "dict_z"):
+            if c.encoding in ("plain_z", "dict_z", "splain_z"):
                 # snappy chunk (K1): one wave per page decompresses into
                 # a scratch buffer; PLAIN pages copy straight from
                 # scratch, dictionary-index pages D2H their (small)
@@ -507,6 +507,8 @@ def read_files_batch_device(paths: List[str], device,
                     else doff_all[:-1]
                 dict_vals = None
                 z_is_str = c.is_string
+                has_spz = any(p[0] == "splain_z" for p in c.pages)
+                vals_list = None
                 if has_zdict and z_is_str:
                     # string dictionary: small D2H of the decompressed
                     # dict page, parsed to values on host; indices stay
@@ -520,10 +522,13 @@ def read_files_batch_device(paths: List[str], device,
                         vals_list.append(
                             draw[pos:pos + ln].tobytes().decode("utf-8"))
                         pos += ln
-                    str_chunks.append(
-                        (c.name, written,
-                         written + sum(p[3] for p in c.pages),
-                         vals_list))
+                    if not has_spz:
+                        str_chunks.append(
+                            (c.name, written,
+                             written + sum(p[3] for p in c.pages),
+                             vals_list))
+                    # with dictionary-overflow PLAIN pages in the chunk
+                    # the dictionary spans become per-run entries below
                 elif has_zdict:
                     dict_vals = _torch.empty(
                         dict_n + 1, dtype=out[c.name].dtype,
@@ -563,7 +568,9 @@ def read_files_batch_device(paths: List[str], device,
                         tA = tB = _time.perf_counter()
 
                     groups: Dict[int, list] = {}
+                    sp_zp = []  # splain_z: (j, abs_row, nv, lo, hi)
                     row = written
+                    run_lo = written  # dict-page run (string chunks)
                     ok = True
                     for j, page in enumerate(c.pages):
                         nv = page[3]
@@ -581,11 +588,23 @@ def read_files_batch_device(paths: List[str], device,
                             groups.setdefault(bw, []).append(
                                 (j, row, nv, r0 + skip + 1,
                                  r0 + int(page[4]), (base - r0) * 8))
+                        elif page[0] == "splain_z":
+                            if has_spz and vals_list is not None \
+                                    and row > run_lo:
+                                str_chunks.append(
+                                    (c.name, run_lo, row, vals_list))
+                            sp_zp.append((j, row, nv, base + skip,
+                                          base + int(page[4])))
+                            run_lo = row + nv
                         else:
                             ext.copy_unaligned(
                                 scratch, base + skip, out[c.name],
                                 row * itemsize, nv * itemsize)
                         row += nv
+                    if ok and has_spz and vals_list is not None \
+                            and row > run_lo:
+                        str_chunks.append(
+                            (c.name, run_lo, row, vals_list))
                     if ok:
                         hb_t = _torch.from_numpy(hb)
 
@@ -617,6 +636,34 @@ def read_files_batch_device(paths: List[str], device,
                             for abs_row, drow, nv in dplaces:
                                 out[c.name][abs_row:abs_row + nv] = \
                                     vals[drow:drow + nv]
+                        if sp_zp:
+                            # one D2H of the decompressed byte-array
+                            # payloads, one host parse + dict-encode
+                            from .native_parquet import \
+                                decode_splain_pages
+                            pb = _torch.cat(
+                                [scratch[a:b]
+                                 for _, _, _, a, b in sp_zp]).cpu()
+                            offs = []
+                            cur0 = 0
+                            for _, _, nv2, a, b in sp_zp:
+                                offs.append((cur0, cur0 + (b - a), nv2))
+                                cur0 += b - a
+                            codes_all, spz_vals = decode_splain_pages(
+                                pb, [o[0] for o in offs],
+                                [o[1] for o in offs],
+                                [o[2] for o in offs])
+                            cpos = 0
+                            for (j2, abs_row, nv2, _, _) in sp_zp:
+                                seg2 = codes_all[cpos:cpos + nv2]
+                                out[c.name][abs_row:abs_row + nv2] = \
+                                    _torch.from_numpy(
+                                        np.ascontiguousarray(seg2)).to(
+                                        device, non_blocking=True)
+                                str_chunks.append(
+                                    (c.name, abs_row, abs_row + nv2,
+                                     spz_vals))
+                                cpos += nv2
                         written = row
                         cursors[c.name] = written
                         if _dt == "3":
@@ -702,6 +749,22 @@ def read_files_batch_device(paths: List[str], device,
                                              n_valid)
                         vals = idx if z_is_str else ext.gather_rows(
                             dict_vals, idx.to(_torch.int64))
+                        if z_is_str and has_spz and \
+                                vals_list is not None:
+                            str_chunks.append((c.name, written,
+                                               written + nv, vals_list))
+                    elif page[0] == "splain_z":
+                        # decompressed PLAIN byte-array payload: D2H,
+                        # host parse + dict-encode, codes back up
+                        from .native_parquet import decode_splain_pages
+                        pb = scratch[base + skip:
+                                     base + int(page[4])].cpu()
+                        codes_np, pvals = decode_splain_pages(
+                            pb, [0], [int(pb.numel())], [n_valid])
+                        vals = _torch.from_numpy(
+                            np.ascontiguousarray(codes_np)).to(device)
+                        str_chunks.append((c.name, written,
+                                           written + nv, pvals))
                     else:
                         if pmask is None:
                             ext.copy_unaligned(scratch, base + skip,
@@ -736,6 +799,10 @@ def read_files_batch_device(paths: List[str], device,
                     for woff, nv, rbase in lst["pages"]:
                         out[c.name][woff:woff + nv] = \
                             vals[rbase:rbase + nv]
+                        if z_is_str and has_spz and \
+                                vals_list is not None:
+                            str_chunks.append((c.name, woff,
+                                               woff + nv, vals_list))
                 cursors[c.name] = written
                 continue
             dict_vals = None

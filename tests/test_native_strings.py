@@ -156,12 +156,14 @@ def test_empty_string_column_roundtrip(tmp_path):
     assert isinstance(rb.column("s"), StringColumn)
 
 
-def test_compressed_string_dict_overflow_falls_back(tmp_path):
+def test_compressed_string_dict_overflow_layout(tmp_path):
     """A SNAPPY string chunk whose dictionary overflows mid-chunk mixes
-    dict-index and PLAIN byte-array pages; the compressed PLAIN string
-    payload must NOT be claimed by the native layout (it would be
-    decoded as int32 codes) — pyarrow fallback, content exact."""
-    from hyperspace_amd.sources.native_parquet import read_native_layout
+    dict-index and PLAIN byte-array pages; the layout records the PLAIN
+    pages as splain_z (device path: decompress + host byte-array
+    parse), never as numeric codes.  The HOST reader falls back to
+    pyarrow for compressed chunks, content exact."""
+    from hyperspace_amd.sources.native_parquet import (read_native_host,
+                                                       read_native_layout)
     vals = [f"s{i:07d}" for i in range(300_000)]
     p = str(tmp_path / "ovf.parquet")
     pq.write_table(pa.table({"s": vals}), p, compression="SNAPPY",
@@ -172,7 +174,11 @@ def test_compressed_string_dict_overflow_falls_back(tmp_path):
                .encodings)
     assert "PLAIN" in encs and encs & {"PLAIN_DICTIONARY",
                                        "RLE_DICTIONARY"}, encs
-    assert read_native_layout(p) is None
+    lay = read_native_layout(p)
+    assert lay is not None
+    kinds = {pg[0] for c in lay[1] for pg in c.pages}
+    assert kinds == {"dict_z", "splain_z"}, kinds
+    assert read_native_host(p) is None  # compressed: device-only
     rb, counts = read_files_batch([p])
     assert counts == [300_000]
     got = rb.column("s")

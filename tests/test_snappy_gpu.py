@@ -288,3 +288,61 @@ def test_plain_string_nullable_device_read(tmp_path):
     got = s.to_numpy()
     for i in range(0, 200_000, 317):
         assert (vals[i] is None and not m[i]) or got[i] == vals[i]
+
+
+def test_snappy_string_dict_overflow_device_read(tmp_path):
+    """SNAPPY string chunks with mid-chunk dictionary overflow (dict_z
+    + splain_z pages in one chunk) decode natively: dict-index pages
+    as codes, PLAIN byte-array pages decompressed then host-parsed."""
+    from hyperspace_amd.execution.columnar import StringColumn
+    vals = [f"s{i:07d}" for i in range(300_000)]
+    p = str(tmp_path / "ovfz.parquet")
+    pq.write_table(pa.table({"s": vals}), p, compression="SNAPPY",
+                   use_dictionary=True,
+                   dictionary_pagesize_limit=64 * 1024,
+                   data_page_version="1.0")
+    batch, rc = read_files_batch_device([p], torch.device("cuda:0"))
+    assert rc == [300_000]
+    s = batch.column("s")
+    assert isinstance(s, StringColumn)
+    assert (s.to_numpy() == np.array(vals, dtype=object)).all()
+
+
+def test_snappy_plain_string_chunk_device_read(tmp_path):
+    """Whole-chunk SNAPPY PLAIN (non-dictionary) strings: pure splain_z
+    chunks decode through decompression + host byte-array parse."""
+    from hyperspace_amd.execution.columnar import StringColumn
+    rng = np.random.default_rng(37)
+    vals = [f"w{i % 1251:05d}" for i in rng.integers(0, 1 << 30,
+                                                     400_000)]
+    p = str(tmp_path / "spz.parquet")
+    pq.write_table(pa.table({"s": vals, "k": np.arange(400_000)}), p,
+                   compression="SNAPPY", use_dictionary=False,
+                   data_page_version="1.0")
+    from hyperspace_amd.sources.native_parquet import read_native_layout
+    lay = read_native_layout(p)
+    assert lay is not None
+    assert any(c.encoding == "splain_z" for c in lay[1])
+    batch, rc = read_files_batch_device([p], torch.device("cuda:0"))
+    assert rc == [400_000]
+    s = batch.column("s")
+    assert isinstance(s, StringColumn)
+    assert (s.to_numpy() == np.array(vals, dtype=object)).all()
+    assert (batch.tensor("k").cpu().numpy() == np.arange(400_000)).all()
+
+
+def test_snappy_plain_string_nullable_device_read(tmp_path):
+    from hyperspace_amd.execution.columnar import StringColumn
+    vals = [None if i % 5 == 0 else f"x{i % 40}" for i in range(150_000)]
+    p = str(tmp_path / "spzn.parquet")
+    pq.write_table(pa.table({"s": vals}), p, compression="SNAPPY",
+                   use_dictionary=False, data_page_version="1.0")
+    batch, rc = read_files_batch_device([p], torch.device("cuda:0"))
+    assert rc == [150_000]
+    s = batch.column("s")
+    assert isinstance(s, StringColumn)
+    m = batch.mask("s").cpu().numpy()
+    assert not m[0] and m[1]
+    got = s.to_numpy()
+    for i in range(0, 150_000, 311):
+        assert (vals[i] is None and not m[i]) or got[i] == vals[i]
