@@ -79,6 +79,7 @@ def _pinned_put(buf: "torch.Tensor") -> None:
 
 
 _CODEC_POOL = None
+_codec_tl = _threading.local()
 
 
 def _codec_pool():
@@ -91,6 +92,18 @@ def _codec_pool():
             max_workers=int(os.environ.get("HS_CODEC_WORKERS", "16")),
             thread_name_prefix="hs-codec")
     return _CODEC_POOL
+
+
+def _tl_decompress(name: str, view, unc: int):
+    """Decompress with a THREAD-LOCAL pyarrow codec: a pa.Codec shared
+    across threads segfaults (reproduced with zstd at 16 threads), so
+    each pool worker keeps its own instance per codec name."""
+    c = getattr(_codec_tl, name, None)
+    if c is None:
+        import pyarrow as _pa
+        c = _pa.Codec(name)
+        setattr(_codec_tl, name, c)
+    return c.decompress(view, unc)
 
 
 def bucket_id_of_file(path: str) -> Optional[int]:
@@ -461,11 +474,10 @@ def read_files_batch_device(paths: List[str], device,
                                       dtype=_torch.int64))
                     statuses.append(st)
                 if host_idx:
-                    import pyarrow as _pa
                     # parquet LZ4 / LZ4_RAW pages are raw LZ4 blocks
-                    codec = _pa.Codec(
-                        "lz4_raw" if c.codec in ("LZ4", "LZ4_RAW")
-                        else c.codec.lower())
+                    codec_name = ("lz4_raw"
+                                  if c.codec in ("LZ4", "LZ4_RAW")
+                                  else c.codec.lower())
                     hview = buf.numpy()
                     # stage all host-decoded pages in ONE pinned buffer:
                     # the codec reads the compressed page zero-copy from
@@ -482,7 +494,7 @@ def read_files_batch_device(paths: List[str], device,
                     # column chunk's copy-dense pages otherwise
                     # decompress serially on this one decode thread
                     futs = [_codec_pool().submit(
-                        codec.decompress, hview[s0:s1], u)
+                        _tl_decompress, codec_name, hview[s0:s1], u)
                         for i in host_idx
                         for s0, s1, u in (segs[i],)]
                     hoff = 0
