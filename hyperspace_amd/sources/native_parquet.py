@@ -695,11 +695,9 @@ def _walk_row_group(rg, pf_schema, data, want
         is_string = False
         if col.physical_type == "BYTE_ARRAY":
             # dictionary-encoded strings decode natively (codes + one
-            # dictionary parse), uncompressed or snappy; uncompressed
-            # PLAIN-encoded string pages parse host-side into codes
-            # ("splain"); compressed PLAIN strings -> pyarrow
-            if not is_dict and codec != "UNCOMPRESSED":
-                return None
+            # dictionary parse), uncompressed or compressed; PLAIN
+            # byte-array pages parse host-side into codes ("splain" /
+            # "splain_z" after decompression)
             is_string = True
             np_dtype = np.dtype("int32")
         else:
