@@ -80,6 +80,11 @@ def _pinned_put(buf: "torch.Tensor") -> None:
 
 _CODEC_POOL = None
 _codec_tl = _threading.local()
+# snappy pages with comp >= ratio*unc (near-incompressible, giant
+# literals) decode on device; the rest are copy-dense -> host codec
+# pool.  0.93 measured best (see profiles/r2_summary.md); 0 = all
+# device, >1 = all host (sweep knob).
+_DEV_RATIO = float(os.environ.get("HS_SNAPPY_DEV_RATIO", "0.93"))
 
 
 def _codec_pool():
@@ -454,7 +459,7 @@ def read_files_batch_device(paths: List[str], device,
                 # device where decode is a bandwidth-bound copy.
                 if c.codec == "SNAPPY":
                     dev_idx = [i for i, s in enumerate(segs)
-                               if (s[1] - s[0]) >= 0.93 * s[2]]
+                               if (s[1] - s[0]) >= _DEV_RATIO * s[2]]
                 else:
                     dev_idx = []  # gzip/zstd/brotli: host codec only
                 host_idx = [i for i in range(len(segs))
