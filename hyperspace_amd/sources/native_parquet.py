@@ -1033,6 +1033,10 @@ def read_native_host(path: str, columns: Optional[List[str]] = None
                     remapped.append(lut[part])
                 codes = (np.concatenate(remapped) if len(remapped) > 1
                          else remapped[0])
+            if not merged_vals and len(codes):
+                # all-null column: masked slots carry code 0 — give it
+                # a value to dereference
+                merged_vals = [""]
             cols[name] = StrCol(codes, list(merged_vals))
         else:
             cols[name] = (np.concatenate(parts) if len(parts) > 1

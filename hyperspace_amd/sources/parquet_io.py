@@ -1110,6 +1110,8 @@ def read_files_batch_device(paths: List[str], device,
             else:
                 merged = sorted(set().union(*map(set, dicts))) \
                     if dicts else []
+                if not merged:
+                    merged = [""]  # all-null column: masked 0-codes
                 vi = {v: i for i, v in enumerate(merged)}
                 for lo, hi, vals in chunks_n:
                     if vals == merged:
