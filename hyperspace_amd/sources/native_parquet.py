@@ -759,6 +759,73 @@ def _walk_row_group(rg, pf_schema, data, want
             if hdr.get(1) == 2:  # stray dictionary page: skip
                 pos = r.pos + page_bytes
                 continue
+            if hdr.get(1) == 3:  # DataPageV2
+                # levels are stored UNCOMPRESSED at the payload start
+                # with a KNOWN byte length (no 4-byte prefix); only the
+                # data section after them is subject to the codec
+                # (parquet-format PageHeader.data_page_header_v2)
+                d2 = hdr.get(8, {})
+                num_values = d2.get(1)
+                dl_len = d2.get(5, 0) or 0
+                rl_len = d2.get(6, 0) or 0
+                if num_values is None or rl_len:
+                    return None  # repeated leaves: pyarrow
+                enc2 = d2.get(4)
+                is_comp = codec != "UNCOMPRESSED" and \
+                    d2.get(7, True) is not False
+                values_off = r.pos
+                page_end = r.pos + page_bytes
+                mask = None
+                if dl_len:
+                    if not chunk_all_valid:
+                        mask = _decode_defs(data, values_off, dl_len,
+                                            num_values)
+                    values_off += dl_len
+                n_valid = int(mask.sum()) if mask is not None \
+                    else num_values
+                if codec != "UNCOMPRESSED":
+                    # under a codec, every V2 page becomes a z page so
+                    # the chunk stays uniform (the dictionary page is
+                    # always compressed); per-page is_compressed=false
+                    # pages carry is_comp=False (7th element) and are
+                    # identity-copied into scratch at decode
+                    if is_comp:
+                        unc = hdr.get(2)
+                        if unc is None:
+                            return None
+                        unc_data = unc - dl_len - rl_len
+                    else:
+                        unc_data = page_end - values_off
+                    if is_dict and enc2 in (2, 8):
+                        pages.append(("dict_z", values_off, page_end,
+                                      num_values, unc_data, False,
+                                      is_comp))
+                    elif enc2 == ENC_PLAIN:
+                        pages.append((
+                            "splain_z" if is_string else "plain_z",
+                            values_off, page_end, num_values, unc_data,
+                            False, is_comp))
+                    else:
+                        return None
+                else:
+                    if is_dict and enc2 in (2, 8):
+                        bit_width = data[values_off]
+                        pages.append(("dict", values_off + 1, page_end,
+                                      num_values, bit_width))
+                    elif is_string and enc2 == ENC_PLAIN:
+                        pages.append(("splain", values_off, page_end,
+                                      num_values))
+                    elif enc2 == ENC_PLAIN:
+                        expected = n_valid * np_dtype.itemsize
+                        if values_off + expected > page_end:
+                            return None
+                        pages.append(("plain", values_off, num_values))
+                    else:
+                        return None
+                page_masks.append(mask)
+                seen += num_values
+                pos = page_end
+                continue
             if hdr.get(1) != PAGE_DATA:
                 return None
             dph = hdr.get(5, {})

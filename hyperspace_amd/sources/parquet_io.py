@@ -440,11 +440,14 @@ def read_files_batch_device(paths: List[str], device,
                 # gather through the decompressed dictionary on device.
                 has_zdict = c.encoding == "dict_z"
                 segs = []
+                idn = []  # V2 is_compressed=false page: identity copy
                 if has_zdict:
                     _, dz_off, dz_end, dict_n, dict_unc = c.dict_page
                     segs.append((dz_off, dz_end, dict_unc))
+                    idn.append(False)
                 for p in c.pages:
                     segs.append((p[1], p[2], p[4]))
+                    idn.append(len(p) > 6 and not p[6])
                 uncs_all = [s[2] for s in segs]
                 doff_all = np.concatenate([[0], np.cumsum(uncs_all)])
                 scratch = _torch.empty(int(doff_all[-1]) + 4,
@@ -459,11 +462,16 @@ def read_files_batch_device(paths: List[str], device,
                 # device where decode is a bandwidth-bound copy.
                 if c.codec == "SNAPPY":
                     dev_idx = [i for i, s in enumerate(segs)
-                               if (s[1] - s[0]) >= _DEV_RATIO * s[2]]
+                               if not idn[i]
+                               and (s[1] - s[0]) >= _DEV_RATIO * s[2]]
                 else:
                     dev_idx = []  # gzip/zstd/brotli: host codec only
                 host_idx = [i for i in range(len(segs))
-                            if i not in dev_idx]
+                            if i not in dev_idx and not idn[i]]
+                for i, s0 in enumerate(segs):
+                    if idn[i]:  # stored uncompressed: straight copy
+                        ext.copy_unaligned(dev_bytes, s0[0], scratch,
+                                           int(doff_all[i]), s0[2])
                 if dev_idx:
                     st = ext.snappy_decompress(
                         dev_bytes,
@@ -733,8 +741,12 @@ def read_files_batch_device(paths: List[str], device,
                     nv = page[3]
                     base = int(page_base[j])
                     skip = 0
-                    pmask = None
-                    if page[5]:
+                    # V2 pages carry their mask from the layout parse
+                    # (levels live uncompressed in the file); V1 pages
+                    # decode the level prefix from the decompressed
+                    # payload here
+                    pmask = c.page_masks[j]
+                    if pmask is None and page[5]:
                         ln = lvl_skips[j] - 4
                         skip = 4 + ln
                         if nullable_chunk:

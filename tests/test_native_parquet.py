@@ -77,8 +77,8 @@ def test_fallback_on_compressed(tmp_path, cols):
     batch, counts = read_files_batch([p])
     assert batch.num_rows == 50_000 and counts == [50_000]
     p2 = str(tmp_path / "t2.parquet")
-    # GZIP/ZSTD now yield layouts too (host-codec pages on the native
-    # assembly path); LZ4 remains unsupported -> no layout
+    # GZIP/ZSTD yield layouts too (host-codec pages on the native
+    # assembly path)
     pq.write_table(pa.table(cols), p2, compression="GZIP")
     lay2 = read_native_layout(p2)
     assert lay2 is not None
@@ -91,12 +91,21 @@ def test_fallback_on_compressed(tmp_path, cols):
     assert all(c.codec == "LZ4" for c in lay3[1])
     batch2, counts2 = read_files_batch([p3])
     assert batch2.num_rows == 50_000 and counts2 == [50_000]
-    # DataPageV2 files stay out of the native layout (pyarrow fallback)
+    # DataPageV2 parses natively too (levels uncompressed at payload
+    # start, per-page is_compressed honored)
     p4 = str(tmp_path / "t4.parquet")
     pq.write_table(pa.table(cols), p4, data_page_version="2.0")
-    assert read_native_layout(p4) is None
+    assert read_native_layout(p4) is not None
     batch3, counts3 = read_files_batch([p4])
     assert batch3.num_rows == 50_000 and counts3 == [50_000]
+    # a physical type outside the native set (FIXED_LEN_BYTE_ARRAY
+    # decimal) yields no layout at all
+    p5 = str(tmp_path / "t5.parquet")
+    import decimal
+    pq.write_table(pa.table({"d": pa.array(
+        [decimal.Decimal("1.23")] * 100,
+        type=pa.decimal128(20, 2))}), p5)
+    assert read_native_layout(p5) is None
 
 
 def test_write_batch_parquet_uses_native(tmp_path, cols):
@@ -189,3 +198,40 @@ def test_split_row_groups_boundaries():
     assert [(off, [c.name for c in cs], cs[0].num_values)
             for off, cs in groups] == \
         [(0, ["a", "b"], 10), (10, ["a", "b"], 7), (17, ["a", "b"], 3)]
+
+
+def test_data_page_v2_host_decode(tmp_path):
+    """DataPageV2: definition levels live uncompressed at the payload
+    start with a header-declared byte length (no 4-byte prefix); the
+    host decode must match pyarrow for numerics, nullables and both
+    string encodings (parquet-format.md DataPageHeaderV2)."""
+    from hyperspace_amd.sources.native_parquet import read_native_host
+    rng = np.random.default_rng(41)
+    n = 80_000
+    t = pa.table({
+        "k": rng.integers(0, 5000, n),
+        "v": rng.random(n),
+        "nn": pa.array([None if i % 11 == 0 else int(i)
+                        for i in range(n)], type=pa.int64()),
+    })
+    p = str(tmp_path / "v2.parquet")
+    pq.write_table(t, p, compression="NONE", use_dictionary=False,
+                   data_page_version="2.0")
+    cols, masks = read_native_host(p)
+    assert (np.asarray(cols["k"]) == t["k"].to_numpy()).all()
+    assert np.allclose(np.asarray(cols["v"]), t["v"].to_numpy())
+    m = masks["nn"]
+    assert not m[0] and m[1]
+    nn = np.asarray(cols["nn"])
+    assert nn[1] == 1 and nn[12] == 12
+    # dict + plain strings
+    for ud in (True, False):
+        ps = str(tmp_path / f"v2s{ud}.parquet")
+        pq.write_table(pa.table({"s": [f"s{i % 997:04d}"
+                                       for i in range(n)]}), ps,
+                       compression="NONE", use_dictionary=ud,
+                       data_page_version="2.0")
+        cs, _ = read_native_host(ps)
+        s = cs["s"]
+        assert all(s.values[s.codes[i]] == f"s{i % 997:04d}"
+                   for i in range(0, n, 997))
