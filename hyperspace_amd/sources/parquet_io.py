@@ -470,8 +470,11 @@ def read_files_batch_device(paths: List[str], device,
                             if i not in dev_idx and not idn[i]]
                 for i, s0 in enumerate(segs):
                     if idn[i]:  # stored uncompressed: straight copy
-                        ext.copy_unaligned(dev_bytes, s0[0], scratch,
-                                           int(doff_all[i]), s0[2])
+                        # (torch u8 slice copy: arbitrary alignment,
+                        # async on the decode stream)
+                        d0 = int(doff_all[i])
+                        scratch[d0:d0 + s0[2]] = \
+                            dev_bytes[s0[0]:s0[0] + s0[2]]
                 if dev_idx:
                     st = ext.snappy_decompress(
                         dev_bytes,
