@@ -235,3 +235,27 @@ def test_data_page_v2_host_decode(tmp_path):
         s = cs["s"]
         assert all(s.values[s.codes[i]] == f"s{i % 997:04d}"
                    for i in range(0, n, 997))
+
+
+def test_spark_shaped_decimal_int64_native(tmp_path):
+    """Spark writes decimal(p<=18) with INT64/INT32 physical type and a
+    DECIMAL logical annotation; the native layout decodes the unscaled
+    integers directly — exactly the engine's decimal representation
+    (execution/columnar.py from_arrow).  FLBA-physical decimals
+    (pyarrow default, p>18) decline to the pyarrow path."""
+    import decimal
+    vals = [decimal.Decimal(f"{i}.25") for i in range(10_000)]
+    t = pa.table({"d": pa.array(vals, type=pa.decimal128(12, 2))})
+    p = str(tmp_path / "dec64.parquet")
+    pq.write_table(t, p, compression="NONE", use_dictionary=False,
+                   store_decimal_as_integer=True,
+                   data_page_version="1.0")
+    md = pq.ParquetFile(p).metadata.row_group(0).column(0)
+    assert md.physical_type == "INT64"
+    cols, _ = read_native_host(p)
+    unscaled = np.asarray(cols["d"])
+    assert unscaled[5] == 525 and unscaled[9999] == 999925
+    # FLBA-physical (pyarrow default): no native layout
+    p2 = str(tmp_path / "decflba.parquet")
+    pq.write_table(t, p2)
+    assert read_native_layout(p2) is None
