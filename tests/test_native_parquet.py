@@ -259,3 +259,25 @@ def test_spark_shaped_decimal_int64_native(tmp_path):
     p2 = str(tmp_path / "decflba.parquet")
     pq.write_table(t, p2)
     assert read_native_layout(p2) is None
+
+
+def test_decimal_filter_over_native_read(tmp_path):
+    """End to end: a decimal filter binds its literal to the unscaled
+    representation and serves from the natively-decoded INT64-physical
+    column."""
+    import decimal
+    import hyperspace_amd as hs
+    os.environ["HYPERSPACE_SYSTEM_PATH"] = str(tmp_path / "ix")
+    d = tmp_path / "t"
+    d.mkdir()
+    vals = [decimal.Decimal(f"{i}.25") for i in range(10_000)]
+    pq.write_table(pa.table({"d": pa.array(vals,
+                                           type=pa.decimal128(12, 2)),
+                             "k": list(range(10_000))}),
+                   str(d / "p0.parquet"), compression="NONE",
+                   use_dictionary=False, store_decimal_as_integer=True,
+                   data_page_version="1.0")
+    s = hs.HyperspaceSession(device="cpu")
+    df = s.read_parquet(str(d))
+    out = df.filter("d > 9998.0").select("k", "d").collect()
+    assert out.num_rows == 2
