@@ -152,10 +152,53 @@ def write_batch_parquet(batch, path: str, compression: Optional[str] = None,
     return st.st_size, int(st.st_mtime * 1000)
 
 
+def _resolve_read_columns(p, columns):
+    """Map requested column names onto a file schema: a dotted request
+    ("a.b.c") may be a nested leaf (read the top-level struct and let
+    from_arrow flatten it); flat columns whose NAME contains dots (our
+    index data for nested leaves) match the file schema directly."""
+    import pyarrow.parquet as pq
+    if columns is None:
+        return None
+    top = {n.lower(): n for n in pq.ParquetFile(p).schema_arrow.names}
+    read_cols = []
+    for c in columns:
+        if c.lower() in top:
+            read_cols.append(top[c.lower()])
+        elif "." in c and c.split(".")[0].lower() in top:
+            root = top[c.split(".")[0].lower()]
+            if root not in read_cols:
+                read_cols.append(root)
+        else:
+            read_cols.append(c)
+    return read_cols
+
+
+def _pyarrow_file_dict(p, columns):
+    """One non-native file read through pyarrow into the per-file
+    (columns, masks) dict shape of read_native_host, so a batch can mix
+    native and fallback files."""
+    import pyarrow.parquet as pq
+    from ..execution.columnar import ColumnBatch, StringColumn
+    from .native_parquet import StrCol
+    t = pq.read_table(p, columns=_resolve_read_columns(p, columns))
+    b = ColumnBatch.from_arrow(t)
+    if columns is not None:
+        b = b.select(columns)
+    cols = {}
+    for name, c in b.columns.items():
+        cols[name] = (StrCol(c.codes.numpy(), c.values)
+                      if isinstance(c, StringColumn) else c.numpy())
+    masks = {name: m.numpy() for name, m in (b.masks or {}).items()}
+    return cols, masks, t.num_rows
+
+
 def read_files_batch(paths: List[str], columns: Optional[List[str]] = None):
     """Read parquet files into a single host ColumnBatch, plus per-file row
     counts (for lineage / per-file segmentation).  Native-layout files
-    (uncompressed PLAIN numeric) bypass pyarrow."""
+    (uncompressed PLAIN numeric) bypass pyarrow; non-native files decode
+    per-file through pyarrow and merge with the native ones (a single
+    legacy file no longer drops the whole batch to the slow path)."""
     import numpy as np
     from ..execution.columnar import ColumnBatch
     from .native_parquet import read_native_host
@@ -164,16 +207,20 @@ def read_files_batch(paths: List[str], columns: Optional[List[str]] = None):
     per_file_masks = []
     row_counts = []
     native_ok = True
-    for p in paths:
-        res = read_native_host(p, columns)
-        if res is None:
-            native_ok = False
-            break
-        cols, fmasks = res
-        per_file.append(cols)
-        per_file_masks.append(fmasks)
-        n = len(next(iter(cols.values()))) if cols else 0
-        row_counts.append(n)
+    try:
+        for p in paths:
+            res = read_native_host(p, columns)
+            if res is None:
+                cols, fmasks, n = _pyarrow_file_dict(p, columns)
+            else:
+                cols, fmasks = res
+                n = len(next(iter(cols.values()))) if cols else 0
+            per_file.append(cols)
+            per_file_masks.append(fmasks)
+            row_counts.append(n)
+    except Exception:  # exotic schema mix: whole-batch pyarrow path
+        native_ok = False
+        row_counts = []
     if native_ok and paths:
         names = list(per_file[0].keys())
         if columns is not None:
@@ -224,25 +271,7 @@ def read_files_batch(paths: List[str], columns: Optional[List[str]] = None):
     tables = []
     row_counts = []
     for p in paths:
-        read_cols = columns
-        if columns is not None:
-            # a dotted request ("a.b.c") may be a nested leaf: read the
-            # top-level struct and let from_arrow flatten it.  Flat
-            # columns whose NAME contains dots (our index data for
-            # nested leaves) match the file schema directly.
-            top = {n.lower(): n for n in pq.ParquetFile(p)
-                   .schema_arrow.names}
-            read_cols = []
-            for c in columns:
-                if c.lower() in top:
-                    read_cols.append(top[c.lower()])
-                elif "." in c and c.split(".")[0].lower() in top:
-                    root = top[c.split(".")[0].lower()]
-                    if root not in read_cols:
-                        read_cols.append(root)
-                else:
-                    read_cols.append(c)
-        t = pq.read_table(p, columns=read_cols)
+        t = pq.read_table(p, columns=_resolve_read_columns(p, columns))
         tables.append(t)
         row_counts.append(t.num_rows)
     if not tables:

@@ -281,3 +281,30 @@ def test_decimal_filter_over_native_read(tmp_path):
     df = s.read_parquet(str(d))
     out = df.filter("d > 9998.0").select("k", "d").collect()
     assert out.num_rows == 2
+
+
+def test_mixed_native_and_fallback_files_in_one_batch(tmp_path):
+    """A batch mixing native-layout files with a non-native one (FLBA
+    decimal physical type) merges per file — the legacy file reads via
+    pyarrow, the rest stay on the native path, content is exact."""
+    import decimal
+    rng = np.random.default_rng(13)
+    p1 = str(tmp_path / "native.parquet")
+    write_parquet_native({"k": np.arange(1000, dtype=np.int64),
+                          "d": np.arange(1000, dtype=np.int64) * 100},
+                         p1)
+    # non-native: FLBA decimal (pyarrow default decimal128 layout),
+    # same logical schema decimal(12,2) stored unscaled x100
+    p2 = str(tmp_path / "legacy.parquet")
+    pq.write_table(pa.table({
+        "k": np.arange(1000, 2000),
+        "d": pa.array([decimal.Decimal(f"{i}.00")
+                       for i in range(1000, 2000)],
+                      type=pa.decimal128(12, 2))}), p2)
+    assert read_native_layout(p2) is None
+    batch, counts = read_files_batch([p1, p2])
+    assert counts == [1000, 1000]
+    k = batch.tensor("k").numpy()
+    d = batch.tensor("d").numpy()
+    assert (k == np.arange(2000)).all()
+    assert (d == np.arange(2000) * 100).all()
