@@ -1027,6 +1027,39 @@ def read_files_batch_device(paths: List[str], device,
         [None] * len(paths)
     upload_lock = threading.Lock()
 
+    def host_unit(i):
+        """Non-native file inside a mixed batch: pyarrow-read on this
+        worker thread, upload into the preallocated slices (the native
+        files around it stay on the device decode path)."""
+        from .native_parquet import StrCol
+        cols, fmasks, n = _pyarrow_file_dict(paths[i], columns)
+        assert n == row_counts[i], (paths[i], n, row_counts[i])
+        low = {k.lower(): v for k, v in cols.items()}
+        mlow = {k.lower(): v for k, v in fmasks.items()}
+        base = int(file_base[i])
+        stream = streams[i % n_streams]
+        with _torch.cuda.stream(stream):
+            for name in names:
+                arr = low[name.lower()]
+                if isinstance(arr, StrCol):
+                    codes = _torch.from_numpy(
+                        np.ascontiguousarray(arr.codes))
+                    out[name][base:base + n] = codes.to(
+                        device, non_blocking=True)
+                    str_chunks.append((name, base, base + n,
+                                       arr.values))
+                else:
+                    t = _torch.from_numpy(np.ascontiguousarray(
+                        arr.astype(dtypes[name], copy=False)))
+                    out[name][base:base + n] = t.to(device,
+                                                    non_blocking=True)
+                if name in out_masks:
+                    m = mlow.get(name.lower())
+                    if m is not None:
+                        out_masks[name][base:base + n] =                             _torch.from_numpy(m).to(device,
+                                                    non_blocking=True)
+        return True
+
     def decode_unit(u):
         uid, (i, row_off, chunks) = u
         buf, size, _ = infos[i]
@@ -1068,10 +1101,13 @@ def read_files_batch_device(paths: List[str], device,
             for fut in as_completed(read_futs):
                 i = read_futs[fut]
                 inf = fut.result()
-                infos[i] = inf
                 if inf is None:
-                    failed = True
+                    # no native layout: per-file pyarrow host read
+                    infos[i] = ("host", 0, None)
+                    dec_futs.append(dec_pool.submit(host_unit, i))
+                    n_units += 1
                     continue
+                infos[i] = inf
                 if failed:
                     continue  # drain remaining reads; fallback below
                 if split_rgs:
@@ -1091,11 +1127,14 @@ def read_files_batch_device(paths: List[str], device,
         for i in range(len(paths)):
             infos[i] = load_file(i)
             if infos[i] is None:
-                failed = True
-                break
+                infos[i] = ("host", 0, None)
         if not failed:
             uid = 0
             for i in range(len(paths)):
+                if infos[i][0] == "host":
+                    host_unit(i)
+                    n_units += 1
+                    continue
                 unit_list = (split_row_groups(infos[i][2]) if split_rgs
                              else [(0, infos[i][2])])
                 for row_off, rg_chunks in unit_list:
@@ -1105,7 +1144,7 @@ def read_files_batch_device(paths: List[str], device,
     bufs = [inf[0] if inf is not None else None for inf in infos]
     if failed:
         for bq in bufs:
-            if bq is not None:
+            if isinstance(bq, _torch.Tensor):
                 _pinned_put(bq)
         for bq in keepalive:
             _pinned_put(bq)
@@ -1127,13 +1166,14 @@ def read_files_batch_device(paths: List[str], device,
     if snappy_bad or \
             not all(b is not None and b is not False for b in bufs):
         for b in bufs:
-            if b is not None and b is not False:
+            if isinstance(b, _torch.Tensor):
                 _pinned_put(b)
         for b in keepalive:
             _pinned_put(b)
         return fallback()
     for b in bufs:
-        _pinned_put(b)
+        if isinstance(b, _torch.Tensor):
+            _pinned_put(b)
     for b in keepalive:
         _pinned_put(b)
     if string_cols:

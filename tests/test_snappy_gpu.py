@@ -379,3 +379,38 @@ def test_data_page_v2_snappy_device_read(tmp_path):
     got = batch.tensor("nn").cpu().numpy()
     for i in range(0, n, 431):
         assert (nn[i] is None and not m[i]) or got[i] == nn[i]
+
+
+def test_mixed_native_and_host_files_device_read(tmp_path):
+    """A device batch mixing native files with a DELTA_BINARY_PACKED
+    file: the exotic file pyarrow-reads on a worker thread into the
+    shared preallocated tensors while its neighbors decode on device."""
+    rng = np.random.default_rng(53)
+    n = 200_000
+    paths = []
+    all_k, all_v = [], []
+    for i in range(3):
+        k = rng.integers(0, 10_000, n)
+        v = rng.random(n)
+        p = str(tmp_path / f"m{i}.parquet")
+        if i == 1:  # exotic encoding: no native layout
+            pq.write_table(pa.table({"k": k, "v": v}), p,
+                           use_dictionary=False, compression="NONE",
+                           column_encoding={"k": "DELTA_BINARY_PACKED",
+                                            "v": "PLAIN"})
+            from hyperspace_amd.sources.native_parquet import \
+                read_native_layout
+            assert read_native_layout(p) is None
+        else:
+            pq.write_table(pa.table({"k": k, "v": v}), p,
+                           compression="SNAPPY", use_dictionary=True,
+                           data_page_version="1.0")
+        paths.append(p)
+        all_k.append(k)
+        all_v.append(v)
+    batch, rc = read_files_batch_device(paths, torch.device("cuda:0"))
+    assert rc == [n, n, n]
+    assert (batch.tensor("k").cpu().numpy()
+            == np.concatenate(all_k)).all()
+    assert np.allclose(batch.tensor("v").cpu().numpy(),
+                       np.concatenate(all_v))
