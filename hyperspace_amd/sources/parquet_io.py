@@ -383,6 +383,14 @@ def read_files_batch_device(paths: List[str], device,
                     # worker falls back)
                     npd = np.dtype("int32")
                     string_cols.add(col.path_in_schema)
+                elif col.physical_type == "FIXED_LEN_BYTE_ARRAY" and                         schemas[0].column(i).logical_type.type ==                         "DECIMAL" and                         schemas[0].column(i).precision <= 18:
+                    # FLBA decimal(p<=18): engine representation is the
+                    # unscaled int64; files carrying it have no native
+                    # layout and read per file on host workers
+                    npd = np.dtype("int64")
+                elif col.physical_type == "INT96":
+                    # legacy timestamp: per-file host read -> int64
+                    npd = np.dtype("int64")
                 else:
                     return fallback()
             names.append(col.path_in_schema)

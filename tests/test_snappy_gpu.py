@@ -414,3 +414,29 @@ def test_mixed_native_and_host_files_device_read(tmp_path):
             == np.concatenate(all_k)).all()
     assert np.allclose(batch.tensor("v").cpu().numpy(),
                        np.concatenate(all_v))
+
+
+def test_flba_decimal_file_in_device_batch(tmp_path):
+    """FLBA-physical decimals (pyarrow's default decimal128 layout)
+    ride the per-file host fallback inside a device batch: unscaled
+    int64 in the shared tensor, native neighbors unaffected."""
+    import decimal
+    rng = np.random.default_rng(59)
+    n = 100_000
+    p1 = str(tmp_path / "native.parquet")
+    pq.write_table(pa.table({"k": np.arange(n),
+                             "d": np.arange(n) * 100}), p1,
+                   compression="SNAPPY", use_dictionary=True,
+                   data_page_version="1.0")
+    p2 = str(tmp_path / "flba.parquet")
+    pq.write_table(pa.table({
+        "k": np.arange(n, 2 * n),
+        "d": pa.array([decimal.Decimal(f"{i}.00")
+                       for i in range(n, 2 * n)],
+                      type=pa.decimal128(12, 2))}), p2)
+    batch, rc = read_files_batch_device([p1, p2],
+                                        torch.device("cuda:0"))
+    assert rc == [n, n]
+    assert (batch.tensor("k").cpu().numpy() == np.arange(2 * n)).all()
+    assert (batch.tensor("d").cpu().numpy()
+            == np.arange(2 * n) * 100).all()
