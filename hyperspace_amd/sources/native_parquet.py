@@ -533,12 +533,12 @@ class ColumnChunkLayout:
     """
     __slots__ = ("name", "np_dtype", "pages", "num_values", "encoding",
                  "dict_page", "page_masks", "is_string", "str_values",
-                 "codec")
+                 "codec", "max_def")
 
     def __init__(self, name, np_dtype, pages, num_values,
                  encoding="plain", dict_page=None, page_masks=None,
                  is_string=False, str_values=None,
-                 codec="UNCOMPRESSED"):
+                 codec="UNCOMPRESSED", max_def=1):
         self.name = name
         self.np_dtype = np_dtype
         self.pages = pages
@@ -549,6 +549,7 @@ class ColumnChunkLayout:
         self.is_string = is_string
         self.str_values = str_values
         self.codec = codec
+        self.max_def = max_def
 
     def dict_values(self, data) -> List[str]:
         """The string dictionary: cached from the writer, else parsed
@@ -587,10 +588,18 @@ def _native_def_decoder():
     return _DEF_DECODER[0]
 
 
-def _decode_defs(data, off: int, length: int, n: int
-                 ) -> Optional[np.ndarray]:
-    """Decode RLE-hybrid definition levels (max_def=1) into a bool
-    validity array, or None when every row is valid."""
+def _decode_defs(data, off: int, length: int, n: int,
+                 max_def: int = 1) -> Optional[np.ndarray]:
+    """Decode RLE-hybrid definition levels into a bool validity array
+    (level == max_def <=> leaf present; any smaller level is a null at
+    the leaf or an ancestor struct), or None when every row is valid."""
+    if max_def > 1:
+        # multi-bit levels (nullable-struct leaves): decode the level
+        # integers with the generic RLE/bit-packed reader
+        bw = max_def.bit_length()
+        levels = _decode_rle_indices(data, off, off + length, n, bw)
+        mask = levels == max_def
+        return None if bool(mask.all()) else mask
     dec = _native_def_decoder()
     if dec is not None:
         mask = dec(data, off, length, n).numpy()
@@ -712,10 +721,10 @@ def _walk_row_group(rg, pf_schema, data, want
         # mask.  Nested leaves (max_def > 1 or repeated) use wider level
         # encodings: pyarrow fallback.
         col_schema = pf_schema.column(col_index)
-        if col_schema.max_definition_level > 1 or \
-                col_schema.max_repetition_level > 0:
-            return None
-        has_levels = col_schema.max_definition_level > 0
+        if col_schema.max_repetition_level > 0:
+            return None  # repeated (list) leaves: pyarrow
+        max_def = col_schema.max_definition_level
+        has_levels = max_def > 0
         st = col.statistics
         chunk_all_valid = st is not None and st.null_count == 0
 
@@ -779,7 +788,7 @@ def _walk_row_group(rg, pf_schema, data, want
                 if dl_len:
                     if not chunk_all_valid:
                         mask = _decode_defs(data, values_off, dl_len,
-                                            num_values)
+                                            num_values, max_def)
                     values_off += dl_len
                 n_valid = int(mask.sum()) if mask is not None \
                     else num_values
@@ -866,7 +875,7 @@ def _walk_row_group(rg, pf_schema, data, want
                 lvl_len = struct.unpack_from("<I", data, values_off)[0]
                 if not chunk_all_valid:
                     mask = _decode_defs(data, values_off + 4, lvl_len,
-                                        num_values)
+                                        num_values, max_def)
                 values_off += 4 + lvl_len
             n_valid = int(mask.sum()) if mask is not None else num_values
             page_enc = dph.get(2)
@@ -912,7 +921,7 @@ def _walk_row_group(rg, pf_schema, data, want
         out.append(ColumnChunkLayout(name, np_dtype, pages,
                                      col.num_values, enc_kind, dict_page,
                                      page_masks, is_string=is_string,
-                                     codec=codec))
+                                     codec=codec, max_def=max_def))
     return out
 
 

@@ -794,7 +794,8 @@ def read_files_batch_device(paths: List[str], device,
                             lvl_bytes = hb_all[s0:s0 + rln].numpy()
                             from .native_parquet import _decode_defs
                             pmask = _decode_defs(
-                                lvl_bytes.tobytes(), 0, ln, nv)
+                                lvl_bytes.tobytes(), 0, ln, nv,
+                                getattr(c, "max_def", 1))
                     n_valid = int(pmask.sum()) if pmask is not None \
                         else nv
                     if page[0] == "dict_z":

@@ -164,3 +164,44 @@ def test_nested_leaf_incremental_refresh(tmp_path, monkeypatch):
     session.disable_hyperspace()
     assert got == df2.filter("nested.leaf = 7").collect().num_rows
     assert got >= before
+
+
+def test_nullable_struct_leaves_native_host(tmp_path):
+    """Nullable struct with nullable leaf (max_def=2): definition
+    levels are 2-bit; the native reader decodes level==max_def as the
+    leaf mask with parent-null propagation (parquet-format.md nested
+    encoding; reference reads via parquet-mr)."""
+    from hyperspace_amd.sources.native_parquet import (read_native_host,
+                                                       read_native_layout)
+    n = 50_000
+    structs = [None if i % 13 == 0 else
+               {"x": (None if i % 7 == 0 else i), "y": float(i)}
+               for i in range(n)]
+    t = pa.table({"s": pa.array(structs,
+                                type=pa.struct([("x", pa.int64()),
+                                                ("y", pa.float64())])),
+                  "k": np.arange(n)})
+    p = str(tmp_path / "nest.parquet")
+    pq.write_table(t, p, compression="NONE", use_dictionary=False,
+                   data_page_version="1.0")
+    lay = read_native_layout(p)
+    assert lay is not None
+    assert {c.name: c.max_def for c in lay[1]} == \
+        {"s.x": 2, "s.y": 2, "k": 1}
+    cols, masks = read_native_host(p)
+    mx, vx = masks["s.x"], np.asarray(cols["s.x"])
+    my, vy = masks["s.y"], np.asarray(cols["s.y"])
+    for i in range(0, n, 313):
+        if structs[i] is None or structs[i]["x"] is None:
+            assert not mx[i]
+        else:
+            assert mx[i] and vx[i] == i
+        if structs[i] is None:
+            assert not my[i]
+        else:
+            assert my[i] and vy[i] == float(i)
+    # repeated (list) leaves still decline
+    tl = pa.table({"l": pa.array([[1, 2], [3]] * 100)})
+    pl = str(tmp_path / "list.parquet")
+    pq.write_table(tl, pl, compression="NONE")
+    assert read_native_layout(pl) is None

@@ -440,3 +440,34 @@ def test_flba_decimal_file_in_device_batch(tmp_path):
     assert (batch.tensor("k").cpu().numpy() == np.arange(2 * n)).all()
     assert (batch.tensor("d").cpu().numpy()
             == np.arange(2 * n) * 100).all()
+
+
+def test_nullable_struct_leaves_snappy_device_read(tmp_path):
+    """Compressed nullable-struct leaves (2-bit def levels) decode on
+    device with parent-null propagation."""
+    n = 300_000
+    structs = [None if i % 13 == 0 else
+               {"x": (None if i % 7 == 0 else i), "y": float(i)}
+               for i in range(n)]
+    t = pa.table({"s": pa.array(structs,
+                                type=pa.struct([("x", pa.int64()),
+                                                ("y", pa.float64())])),
+                  "k": np.arange(n)})
+    p = str(tmp_path / "nestz.parquet")
+    pq.write_table(t, p, compression="SNAPPY", use_dictionary=False,
+                   data_page_version="1.0")
+    batch, rc = read_files_batch_device([p], torch.device("cuda:0"))
+    assert rc == [n]
+    mx = batch.mask("s.x").cpu().numpy()
+    vx = batch.tensor("s.x").cpu().numpy()
+    my = batch.mask("s.y").cpu().numpy()
+    vy = batch.tensor("s.y").cpu().numpy()
+    for i in range(0, n, 431):
+        if structs[i] is None or structs[i]["x"] is None:
+            assert not mx[i], i
+        else:
+            assert mx[i] and vx[i] == i, i
+        if structs[i] is None:
+            assert not my[i], i
+        else:
+            assert my[i] and vy[i] == float(i), i
