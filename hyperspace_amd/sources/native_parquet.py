@@ -1033,10 +1033,10 @@ def read_native_host(path: str, columns: Optional[List[str]] = None
     if layout is None:
         return None
     data, chunks = layout
-    if any(not (c.encoding == "plain"
-                or (c.is_string and c.encoding in ("dict", "splain")))
+    if any(not (c.encoding in ("plain", "dict")
+                or (c.is_string and c.encoding == "splain"))
            for c in chunks):
-        return None  # numeric dict + snappy decode is the device path
+        return None  # compressed chunks decode on the device path
     acc: Dict[str, list] = {}
     macc: Dict[str, list] = {}
     any_null: Dict[str, bool] = {}
@@ -1075,14 +1075,31 @@ def read_native_host(path: str, columns: Optional[List[str]] = None
                 macc.setdefault(c.name, []).append(pm)
                 str_dicts.setdefault(c.name, []).append(page_vals)
             continue
-        for (_, off, nv), mask in zip(c.pages, c.page_masks):
+        dvals = None
+        if c.dict_page is not None:
+            doff, dn = c.dict_page
+            dvals = np.frombuffer(data, dtype=c.np_dtype, count=dn,
+                                  offset=doff)
+        for page, mask in zip(c.pages, c.page_masks):
+            if page[0] == "dict":
+                # numeric dictionary page (Spark's default output
+                # shape): decode indices, gather through the PLAIN
+                # dictionary values
+                _, p_start, p_end, nv, bw = page
+                n_valid = int(mask.sum()) if mask is not None else nv
+                idx = _decode_rle_indices(data, p_start, p_end,
+                                          n_valid, bw)
+                vals = (dvals[idx] if n_valid
+                        else np.empty(0, dtype=c.np_dtype))
+            else:
+                _, off, nv = page
+                n_valid = int(mask.sum()) if mask is not None else nv
+                vals = np.frombuffer(data, dtype=c.np_dtype,
+                                     count=n_valid, offset=off)
             if mask is None:
-                part = np.frombuffer(data, dtype=c.np_dtype, count=nv,
-                                     offset=off)
+                part = vals
                 pm = np.ones(nv, dtype=bool)
             else:
-                vals = np.frombuffer(data, dtype=c.np_dtype,
-                                     count=int(mask.sum()), offset=off)
                 part = np.zeros(nv, dtype=c.np_dtype)
                 part[mask] = vals
                 pm = mask

@@ -308,3 +308,31 @@ def test_mixed_native_and_fallback_files_in_one_batch(tmp_path):
     d = batch.tensor("d").numpy()
     assert (k == np.arange(2000)).all()
     assert (d == np.arange(2000) * 100).all()
+
+
+def test_numeric_dictionary_host_decode(tmp_path):
+    """Spark's default shape (dictionary-encoded numerics) decodes on
+    the HOST path too: RLE indices gathered through the PLAIN
+    dictionary page, incl. mid-chunk overflow and nullable columns."""
+    rng = np.random.default_rng(2)
+    n = 300_000
+    key = rng.integers(0, 1000, n)
+    big = rng.integers(0, 1 << 40, n)  # dict overflows mid-chunk
+    nn = pa.array([None if i % 9 == 0 else int(i % 777)
+                   for i in range(n)], type=pa.int64())
+    p = str(tmp_path / "numdict.parquet")
+    pq.write_table(pa.table({"key": key, "big": big, "nn": nn}), p,
+                   compression="NONE", use_dictionary=True,
+                   data_page_version="1.0")
+    lay = read_native_layout(p)
+    kinds = {c.name: c.encoding for c in lay[1]}
+    assert kinds["key"] == "dict"
+    cols, masks = read_native_host(p)
+    assert (np.asarray(cols["key"]) == key).all()
+    assert (np.asarray(cols["big"]) == big).all()
+    m, v = masks["nn"], np.asarray(cols["nn"])
+    for i in range(0, n, 313):
+        assert (not m[i]) if i % 9 == 0 else (m[i] and v[i] == i % 777)
+    batch, rc = read_files_batch([p])
+    assert rc == [n]
+    assert (batch.tensor("big").numpy() == big).all()
