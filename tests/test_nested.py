@@ -205,3 +205,24 @@ def test_nullable_struct_leaves_native_host(tmp_path):
     pl = str(tmp_path / "list.parquet")
     pq.write_table(tl, pl, compression="NONE")
     assert read_native_layout(pl) is None
+
+
+def test_list_column_source_scalar_index(tmp_path, monkeypatch):
+    """A source containing a LIST column (unsupported as an engine
+    column type) must not break indexing/querying of its scalar
+    columns — the column-pruned reads never touch the list."""
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "ix"))
+    d = tmp_path / "t"
+    d.mkdir()
+    pq.write_table(pa.table({
+        "k": np.arange(5000),
+        "v": np.random.default_rng(0).random(5000),
+        "tags": pa.array([[1, 2], [3]] * 2500)}),
+        str(d / "p0.parquet"))
+    s = hs.HyperspaceSession(device="cpu")
+    h = hs.Hyperspace(s)
+    df = s.read_parquet(str(d))
+    h.create_index(df, hs.CoveringIndexConfig("lx", ["k"], ["v"]))
+    s.enable_hyperspace()
+    out = df.filter("k = 42").select("k", "v").collect()
+    assert out.num_rows == 1
