@@ -496,6 +496,25 @@ std::vector<torch::Tensor> parse_byte_arrays(
   return {offs, out.narrow(0, 0, written)};
 }
 
+// GIL-released memcpy of a buffer-protocol object (e.g. a pyarrow
+// Buffer from Codec::Decompress) into a cpu uint8 tensor slice — a
+// plain numpy slice assignment holds the GIL for the whole copy, which
+// serializes the 16-thread host-codec upload staging.
+void host_memcpy(torch::Tensor dst, int64_t off, py::buffer src) {
+  TORCH_CHECK(!dst.is_cuda() && dst.scalar_type() == torch::kUInt8,
+              "dst must be a cpu u8 tensor");
+  py::buffer_info info = src.request();
+  int64_t n = (int64_t)info.size * (int64_t)info.itemsize;
+  TORCH_CHECK(off >= 0 && off + n <= dst.numel(),
+              "host_memcpy out of range");
+  uint8_t* d = dst.data_ptr<uint8_t>() + off;
+  const void* s = info.ptr;
+  {
+    py::gil_scoped_release rel;
+    memcpy(d, s, (size_t)n);
+  }
+}
+
 std::vector<torch::Tensor> parse_rle_runs(torch::Tensor bytes,
                                            int64_t start, int64_t end,
                                            int64_t bit_width,
@@ -662,6 +681,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("parse_byte_arrays", &parse_byte_arrays,
         py::call_guard<py::gil_scoped_release>(),
         "host PLAIN byte-array page parse -> (offsets, bytes)");
+  m.def("host_memcpy", &host_memcpy,
+        "GIL-released buffer -> cpu-u8-tensor memcpy");
   m.def("rle_decode", &rle_decode,
         "device RLE/bit-packed dictionary-index decode");
 }

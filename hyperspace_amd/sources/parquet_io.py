@@ -541,33 +541,40 @@ def read_files_batch_device(paths: List[str], device,
                     total_host = sum(segs[i][2] for i in host_idx)
                     hstage = _pinned_get(total_host)
                     keepalive.append(hstage)
-                    hs_np = hstage.numpy()
-                    # fan the per-page codec calls over the shared codec
-                    # pool — pyarrow codecs release the GIL, and a large
-                    # column chunk's copy-dense pages otherwise
-                    # decompress serially on this one decode thread
-                    futs = [_codec_pool().submit(
-                        _tl_decompress, codec_name, hview[s0:s1], u)
-                        for i in host_idx
-                        for s0, s1, u in (segs[i],)]
+                    # fan the per-page codec calls AND the pinned
+                    # staging memcpy over the shared codec pool — the
+                    # pyarrow codecs release the GIL, and host_memcpy
+                    # releases it for the copy (a numpy slice
+                    # assignment held it for ~GB of memcpy per read,
+                    # serializing all 16 workers)
+                    hoffs = []
                     hoff = 0
-                    for i, fut in zip(host_idx, futs):
+                    for i in host_idx:
+                        hoffs.append(hoff)
+                        hoff += segs[i][2]
+
+                    def _stage(i2, off2):
+                        a2, b2, unc2 = segs[i2]
+                        dec = _tl_decompress(codec_name,
+                                             hview[a2:b2], unc2)
+                        if len(dec) != unc2:
+                            raise ValueError("short page")
+                        ext.host_memcpy(hstage, off2, dec)
+
+                    futs = [_codec_pool().submit(_stage, i2, off2)
+                            for i2, off2 in zip(host_idx, hoffs)]
+                    for i, fut, off2 in zip(host_idx, futs, hoffs):
                         unc2 = segs[i][2]
                         try:
-                            dec = fut.result()
+                            fut.result()
                         except Exception:  # noqa: BLE001
-                            dec = None
-                        if dec is None or len(dec) != unc2:
                             statuses.append(_torch.ones(
                                 1, dtype=_torch.int32, device=device))
-                            hoff += unc2
                             continue
-                        hs_np[hoff:hoff + unc2] = np.frombuffer(
-                            dec, dtype=np.uint8)
                         scratch[int(doff_all[i]):
                                 int(doff_all[i]) + unc2].copy_(
-                            hstage[hoff:hoff + unc2], non_blocking=True)
-                        hoff += unc2
+                            hstage[off2:off2 + unc2],
+                            non_blocking=True)
                 page_base = doff_all[1:-1] if has_zdict \
                     else doff_all[:-1]
                 dict_vals = None
