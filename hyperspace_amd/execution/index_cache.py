@@ -13,6 +13,7 @@ MI355X-first design choice).
 from __future__ import annotations
 
 import collections
+import threading
 from typing import Dict, List, Optional, Tuple
 
 import torch
@@ -30,6 +31,10 @@ class IndexDataCache:
         self._bytes = 0
         self.hits = 0
         self.misses = 0
+        # concurrent queries share the session cache: all structural
+        # mutation happens under this lock (an unlocked pop/popitem
+        # interleaving corrupts the byte accounting)
+        self._lock = threading.Lock()
 
     @staticmethod
     def key(entry, files: List[str], extra: tuple = ()) -> tuple:
@@ -39,17 +44,18 @@ class IndexDataCache:
         return (entry.name, entry.id, tuple(sorted(files)), extra)
 
     def get(self, key: tuple, columns: Optional[List[str]] = None):
-        item = self._entries.get(key)
-        if item is not None and columns is not None:
-            have = {c.lower() for c in item[0].columns}
-            if not all(c.lower() in have for c in columns):
-                item = None  # stored batch lacks a needed column
-        if item is None:
-            self.misses += 1
-            return None
-        self._entries.move_to_end(key)
-        self.hits += 1
-        batch, seg = item
+        with self._lock:
+            item = self._entries.get(key)
+            if item is not None and columns is not None:
+                have = {c.lower() for c in item[0].columns}
+                if not all(c.lower() in have for c in columns):
+                    item = None  # stored batch lacks a needed column
+            if item is None:
+                self.misses += 1
+                return None
+            self._entries.move_to_end(key)
+            self.hits += 1
+            batch, seg = item
         if columns is not None:
             batch = batch.select(columns)
         return batch, seg
@@ -60,6 +66,10 @@ class IndexDataCache:
         if nbytes > self.budget:
             # oversized: keep whatever incumbent we have under this key
             return
+        with self._lock:
+            self._put_locked(key, batch, seg, nbytes)
+
+    def _put_locked(self, key, batch, seg, nbytes):
         old = self._entries.get(key)
         if old is not None:
             # keep the incumbent if it already serves every column of the
@@ -78,5 +88,6 @@ class IndexDataCache:
         self._bytes += nbytes
 
     def clear(self):
-        self._entries.clear()
-        self._bytes = 0
+        with self._lock:
+            self._entries.clear()
+            self._bytes = 0

@@ -367,3 +367,56 @@ def test_refresh_commits_signature_of_built_snapshot(env, tmp_path):
     t_all = pq.read_table(str(data))
     want = int((t_all.column("key").to_numpy() == 5).sum())
     assert got.num_rows == want
+
+
+def test_concurrent_queries_during_refresh(tmp_path, monkeypatch):
+    """N query threads hammer an indexed filter while a refresh runs:
+    every result must be correct (pre- or post-refresh view), the
+    shared HBM/CPU index-data cache must stay consistent (locked LRU),
+    and the maintenance kill-switch must keep the refresh's own reads
+    index-free."""
+    import threading
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "ix"))
+    rng = np.random.default_rng(17)
+    d = tmp_path / "src"
+    d.mkdir()
+    key = rng.integers(0, 500, 20_000)
+    expected1 = int((key == 77).sum())
+    pq.write_table(pa.table({"key": key, "val": rng.random(20_000)}),
+                   str(d / "p0.parquet"))
+    session = hs.HyperspaceSession(device="cpu")
+    session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 8)
+    h = hs.Hyperspace(session)
+    df = session.read_parquet(str(d))
+    h.create_index(df, hs.CoveringIndexConfig("cqx", ["key"], ["val"]))
+    session.enable_hyperspace()
+    # append then refresh concurrently with queries
+    key2 = rng.integers(0, 500, 5_000)
+    expected2 = expected1 + int((key2 == 77).sum())
+    pq.write_table(pa.table({"key": key2, "val": rng.random(5_000)}),
+                   str(d / "p1.parquet"))
+    errors = []
+    counts = set()
+
+    def worker():
+        try:
+            for _ in range(8):
+                q = session.read_parquet(str(d)).filter("key = 77") \
+                    .select("key", "val")
+                counts.add(q.collect().num_rows)
+        except Exception as e:  # noqa: BLE001
+            errors.append(e)
+
+    threads = [threading.Thread(target=worker) for _ in range(6)]
+    for t in threads:
+        t.start()
+    h.refresh_index("cqx", "incremental")
+    for t in threads:
+        t.join()
+    assert not errors, errors
+    # every observed count is one of the two consistent views
+    assert counts <= {expected1, expected2}, (counts, expected1,
+                                              expected2)
+    # post-refresh queries see the merged view
+    q = session.read_parquet(str(d)).filter("key = 77").select("key")
+    assert q.collect().num_rows == expected2
