@@ -157,3 +157,55 @@ def test_gpu_arithmetic_sketch_filter(tmp_path, monkeypatch):
     out = Executor(session).execute(plan)
     assert out.num_rows == 50_000
     assert bool((out.tensor("key") % 10 == 2).all())
+
+
+def test_gpu_concurrent_indexed_queries(tmp_path, monkeypatch):
+    """Six threads run indexed filter + join queries concurrently on
+    one GPU session: results stay exact (shared HBM cache is locked,
+    kernels use per-thread current streams)."""
+    import threading
+    monkeypatch.setenv("HYPERSPACE_SYSTEM_PATH", str(tmp_path / "ix"))
+    rng = np.random.default_rng(19)
+    d = tmp_path / "src"
+    d.mkdir()
+    n = 400_000
+    key = rng.integers(0, 5000, n)
+    expected = int((key == 77).sum())
+    pq.write_table(pa.table({"key": key, "val": rng.random(n)}),
+                   str(d / "p0.parquet"))
+    dim = tmp_path / "dim"
+    dim.mkdir()
+    pq.write_table(pa.table({"key": np.arange(5000),
+                             "status": rng.integers(0, 3, 5000)}),
+                   str(dim / "p0.parquet"))
+    session = hs.HyperspaceSession(device="cuda")
+    session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 16)
+    h = hs.Hyperspace(session)
+    df = session.read_parquet(str(d))
+    dimdf = session.read_parquet(str(dim))
+    h.create_index(df, hs.CoveringIndexConfig("cgx", ["key"], ["val"]))
+    h.create_index(dimdf, hs.CoveringIndexConfig("cgd", ["key"],
+                                                 ["status"]))
+    session.enable_hyperspace()
+    errors = []
+
+    def worker(tid):
+        try:
+            for r in range(6):
+                if (tid + r) % 2:
+                    q = df.filter("key = 77").select("key", "val")
+                    assert q.collect().num_rows == expected
+                else:
+                    j = df.select("key", "val").join(
+                        dimdf.select("key", "status"), on="key")
+                    assert j.collect().num_rows == n
+        except Exception as e:  # noqa: BLE001
+            errors.append(e)
+
+    threads = [threading.Thread(target=worker, args=(t,))
+               for t in range(6)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert not errors, errors
