@@ -75,6 +75,14 @@ def env(tmp_path_factory):
     dt = DeltaTable.create(str(delta_dir))
     dt.append_files([str(delta_dir / "part-0.parquet")])
 
+    # iceberg copy of item
+    from hyperspace_amd.sources.iceberg_source import IcebergTable
+    ice_dir = tmp / "item_ice"
+    ice_dir.mkdir()
+    pq.write_table(pa.table(it), str(ice_dir / "part-0.parquet"))
+    ic = IcebergTable.create(str(ice_dir))
+    ic.append_files([str(ice_dir / "part-0.parquet")])
+
     session = hs.HyperspaceSession(device="cpu")
     session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 8)
     h = hs.Hyperspace(session)
@@ -83,6 +91,7 @@ def env(tmp_path_factory):
     itdf = session.read_parquet(str(tmp / "item"))
     sspart = session.read_parquet(str(part_root))
     dddelta = session.read_delta(str(delta_dir))
+    itice = session.read_iceberg(str(ice_dir))
     h.create_index(ssdf, hs.CoveringIndexConfig(
         "d_ss_date", ["ss_sold_date_sk"],
         ["ss_item_sk", "ss_quantity"]))
@@ -103,58 +112,68 @@ def env(tmp_path_factory):
         "d_ssp_qty", ["ss_quantity"], ["ss_sales_price"]))
     h.create_index(dddelta, hs.CoveringIndexConfig(
         "d_dd_delta", ["d_date_sk"], ["d_moy"]))
+    h.create_index(itice, hs.CoveringIndexConfig(
+        "d_it_ice", ["i_item_sk"], ["i_brand"]))
     session.enable_hyperspace()
-    return session, ssdf, dddf, itdf, sspart, dddelta
+    return session, ssdf, dddf, itdf, sspart, dddelta, itice
 
 
 QUERIES = {
     # q3-shaped: 3-way star join — both join pairs rewritten in ONE plan
-    "d01_star_join": lambda ss, dd, it, sp, dl:
+    "d01_star_join": lambda ss, dd, it, sp, dl, ii:
         ss.select("ss_sold_date_sk", "ss_item_sk", "ss_quantity")
         .join(dd.select("d_date_sk", "d_moy"),
               on=col("ss_sold_date_sk") == col("d_date_sk"))
         .join(it.select("i_item_sk", "i_category"),
               on=col("ss_item_sk") == col("i_item_sk")),
-    "d02_filter_qty_eq": lambda ss, dd, it, sp, dl:
+    "d02_filter_qty_eq": lambda ss, dd, it, sp, dl, ii:
         ss.filter("ss_quantity = 48")
         .select("ss_quantity", "ss_sales_price"),
-    "d03_join_subset_project": lambda ss, dd, it, sp, dl:
+    "d03_join_subset_project": lambda ss, dd, it, sp, dl, ii:
         ss.select("ss_item_sk", "ss_quantity")
         .join(it.select("i_item_sk", "i_category"),
               on=col("ss_item_sk") == col("i_item_sk")),
     # z-order: second indexed column alone still matches
-    "d04_zorder_customer": lambda ss, dd, it, sp, dl:
+    "d04_zorder_customer": lambda ss, dd, it, sp, dl, ii:
         ss.filter("ss_customer_sk = 777")
         .select("ss_customer_sk", "ss_sales_price"),
     # data-skipping: bloom eq on customer + minmax range on price
-    "d05_ds_bloom_eq": lambda ss, dd, it, sp, dl:
+    "d05_ds_bloom_eq": lambda ss, dd, it, sp, dl, ii:
         ss.filter("ss_customer_sk = 123")
         .select("ss_sold_date_sk", "ss_sales_price"),
-    "d06_ds_price_range": lambda ss, dd, it, sp, dl:
+    "d06_ds_price_range": lambda ss, dd, it, sp, dl, ii:
         ss.filter("ss_sales_price <= 0.25")
         .select("ss_sold_date_sk", "ss_sales_price"),
     # partitioned hive source: covering rewrite over partitioned files
-    "d07_partitioned_filter": lambda ss, dd, it, sp, dl:
+    "d07_partitioned_filter": lambda ss, dd, it, sp, dl, ii:
         sp.filter("ss_quantity = 11")
         .select("ss_quantity", "ss_sales_price"),
     # delta source: covering rewrite over a Delta relation
-    "d08_delta_filter": lambda ss, dd, it, sp, dl:
+    "d08_delta_filter": lambda ss, dd, it, sp, dl, ii:
         dl.filter("d_date_sk = 100").select("d_date_sk", "d_moy"),
     # ss_customer_sk is not covered by d_ss_date -> no join rewrite
-    "d09_join_not_covered": lambda ss, dd, it, sp, dl:
+    "d09_join_not_covered": lambda ss, dd, it, sp, dl, ii:
         ss.select("ss_sold_date_sk", "ss_customer_sk")
         .join(dd.select("d_date_sk", "d_moy"),
               on=col("ss_sold_date_sk") == col("d_date_sk")),
     # IN-list + range conjunction on the filter index
-    "d10_filter_in_range": lambda ss, dd, it, sp, dl:
+    "d10_filter_in_range": lambda ss, dd, it, sp, dl, ii:
         ss.filter("ss_quantity in (5, 6, 7)")
         .filter("ss_sales_price > 100")
         .select("ss_quantity", "ss_sales_price"),
+    # iceberg source: covering rewrite over an Iceberg relation
+    "d11_iceberg_filter": lambda ss, dd, it, sp, dl, ii:
+        ii.filter("i_item_sk = 1500").select("i_item_sk", "i_brand"),
+    # fact ⋈ iceberg-dim join with both sides indexed
+    "d12_join_iceberg_dim": lambda ss, dd, it, sp, dl, ii:
+        ss.select("ss_item_sk", "ss_quantity")
+        .join(ii.select("i_item_sk", "i_brand"),
+              on=col("ss_item_sk") == col("i_item_sk")),
 }
 
 
 @pytest.mark.parametrize("name", sorted(QUERIES))
 def test_tpcds_plan_stability(env, name):
-    session, ssdf, dddf, itdf, sspart, dddelta = env
-    q = QUERIES[name](ssdf, dddf, itdf, sspart, dddelta)
+    session, ssdf, dddf, itdf, sspart, dddelta, itice = env
+    q = QUERIES[name](ssdf, dddf, itdf, sspart, dddelta, itice)
     _check(name, q.optimized_plan().pretty())
