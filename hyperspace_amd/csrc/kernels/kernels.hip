@@ -1280,33 +1280,11 @@ template <typename T>
 __global__ void k_gather(const T* __restrict__ in,
                          const int64_t* __restrict__ idx,
                          T* __restrict__ out, int64_t n) {
-  // 4 independent random loads in flight per thread: the 1-elem
-  // grid-stride form measured 78% parked / 0.7% active-issue (PMC,
-  // profiles/r2_pmc_join.md) — memory-latency bound with no ILP
-  constexpr int V = 4;
-  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t nthreads = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t s = tid * V; s < n; s += nthreads * V) {
-    int64_t cnt = n - s < V ? n - s : V;
-    int64_t ix[V];
-    T val[V];
-#pragma unroll
-    for (int v = 0; v < V; v++)
-      if (v < cnt) ix[v] = idx[s + v];
-#pragma unroll
-    for (int v = 0; v < V; v++)
-      if (v < cnt) val[v] = in[ix[v]];
-#pragma unroll
-    for (int v = 0; v < V; v++)
-      if (v < cnt) out[s + v] = val[v];
-  }
-}
-
-template <typename T>
-__global__ void k_gather_s1(const T* __restrict__ in,
-                            const int64_t* __restrict__ idx,
-                            T* __restrict__ out, int64_t n) {
-  // A/B reference: the original 1-element grid-stride gather
+  // simple grid-stride gather: measured EQUAL to explicit 4-way ILP
+  // variants (37 Grows/s / ~0.9 TB/s effective on 268M random rows —
+  // the compiler pipelines the independent iterations; the
+  // 4-consecutive-per-thread form was 17% SLOWER from broken idx/out
+  // coalescing, profiles/r2_summary.md)
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride)
@@ -1316,34 +1294,24 @@ __global__ void k_gather_s1(const T* __restrict__ in,
 void gather(const void* in, const int64_t* idx, void* out, int64_t n,
             int elem_size, hipStream_t stream) {
   if (n == 0) return;
-  static const bool v1 = [] {  // same-box A/B knob
-    const char* e = getenv("HS_GATHER_V1");
-    return e && e[0] == '1';
-  }();
   int g = grid_for(n);
-#define HS_GATHER_LAUNCH(T)                                              \
-  do {                                                                   \
-    if (v1)                                                              \
-      hipLaunchKernelGGL(k_gather_s1<T>, dim3(g), dim3(THREADS), 0,      \
-                         stream, (const T*)in, idx, (T*)out, n);         \
-    else                                                                 \
-      hipLaunchKernelGGL(k_gather<T>, dim3(g), dim3(THREADS), 0,         \
-                         stream, (const T*)in, idx, (T*)out, n);         \
-  } while (0)
   switch (elem_size) {
     case 8:
-      HS_GATHER_LAUNCH(uint64_t);
+      hipLaunchKernelGGL(k_gather<uint64_t>, dim3(g), dim3(THREADS), 0,
+                         stream, (const uint64_t*)in, idx, (uint64_t*)out, n);
       break;
     case 4:
-      HS_GATHER_LAUNCH(uint32_t);
+      hipLaunchKernelGGL(k_gather<uint32_t>, dim3(g), dim3(THREADS), 0,
+                         stream, (const uint32_t*)in, idx, (uint32_t*)out, n);
       break;
     case 2:
-      HS_GATHER_LAUNCH(uint16_t);
+      hipLaunchKernelGGL(k_gather<uint16_t>, dim3(g), dim3(THREADS), 0,
+                         stream, (const uint16_t*)in, idx, (uint16_t*)out, n);
       break;
     default:
-      HS_GATHER_LAUNCH(uint8_t);
+      hipLaunchKernelGGL(k_gather<uint8_t>, dim3(g), dim3(THREADS), 0,
+                         stream, (const uint8_t*)in, idx, (uint8_t*)out, n);
   }
-#undef HS_GATHER_LAUNCH
 }
 
 // ---------------------------------------------------------------------------
