@@ -16,7 +16,7 @@ from __future__ import annotations
 import json
 import os
 import tempfile
-from typing import List, Optional
+from typing import Optional
 
 from .constants import States
 from .entry import IndexLogEntry

@@ -9,7 +9,6 @@ Root = conf ``spark.hyperspace.system.path`` (reference default:
 from __future__ import annotations
 
 import os
-from typing import Optional
 
 from ..config import Conf
 

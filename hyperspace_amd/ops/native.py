@@ -10,7 +10,6 @@ eager fallback.
 from __future__ import annotations
 
 import importlib
-import os
 
 _ext = None
 _load_error = None
