@@ -471,3 +471,24 @@ def test_nullable_struct_leaves_snappy_device_read(tmp_path):
             assert not my[i], i
         else:
             assert my[i] and vy[i] == float(i), i
+
+
+def test_column_projection_on_new_page_kinds(tmp_path):
+    """columns= projection must hold for splain/splain_z/V2 chunks: only
+    requested columns decode and the content stays exact."""
+    rng = np.random.default_rng(67)
+    n = 200_000
+    vals = [f"p{i % 301:03d}" for i in range(n)]
+    p1 = str(tmp_path / "a.parquet")
+    pq.write_table(pa.table({"s": vals, "k": np.arange(n),
+                             "drop": rng.random(n)}), p1,
+                   compression="SNAPPY", use_dictionary=False,
+                   data_page_version="2.0")
+    batch, rc = read_files_batch_device([p1], torch.device("cuda:0"),
+                                        columns=["k", "s"])
+    assert rc == [n]
+    assert set(batch.columns.keys()) == {"k", "s"}
+    assert (batch.tensor("k").cpu().numpy() == np.arange(n)).all()
+    s = batch.column("s")
+    assert (s.to_numpy()[::733] == np.array(vals[::733],
+                                            dtype=object)).all()
