@@ -58,6 +58,14 @@ def t_sync():
 def main():
     work = "/dev/shm/tpch_bench" if os.path.isdir("/dev/shm") \
         else "/tmp/tpch_bench"
+    # a previous run at a DIFFERENT scale factor leaves its files in
+    # the same dirs (mixed tables + shm exhaustion): reset on mismatch
+    marker = os.path.join(work, f".sf{SF:g}")
+    if os.path.isdir(work) and not os.path.exists(marker):
+        import shutil
+        shutil.rmtree(work, ignore_errors=True)
+    os.makedirs(work, exist_ok=True)
+    open(marker, "w").close()
     os.environ["HYPERSPACE_SYSTEM_PATH"] = os.path.join(work, "indexes")
     t0 = time.perf_counter()
     li_dir, o_dir, li_b, o_b, n_li, n_orders = gen(work)
