@@ -497,6 +497,7 @@ def read_files_batch_device(paths: List[str], device,
                 # C++ codec (~GB/s per worker thread) and uploads;
                 # near-incompressible pages (giant literals) stay on
                 # device where decode is a bandwidth-bound copy.
+                host_bytes: Dict[int, object] = {}
                 if c.codec == "SNAPPY":
                     dev_idx = [i for i, s in enumerate(segs)
                                if not idn[i]
@@ -560,13 +561,19 @@ def read_files_batch_device(paths: List[str], device,
                         if len(dec) != unc2:
                             raise ValueError("short page")
                         ext.host_memcpy(hstage, off2, dec)
+                        return dec
 
                     futs = [_codec_pool().submit(_stage, i2, off2)
                             for i2, off2 in zip(host_idx, hoffs)]
                     for i, fut, off2 in zip(host_idx, futs, hoffs):
                         unc2 = segs[i][2]
                         try:
-                            fut.result()
+                            # host-decoded payloads stay available for
+                            # the header/run parse below — it reads them
+                            # WITHOUT a device round-trip (the old
+                            # scratch D2H sync serialized every chunk
+                            # on its own decompression)
+                            host_bytes[i] = fut.result()
                         except Exception:  # noqa: BLE001
                             statuses.append(_torch.ones(
                                 1, dtype=_torch.int32, device=device))
@@ -585,7 +592,10 @@ def read_files_batch_device(paths: List[str], device,
                     # string dictionary: small D2H of the decompressed
                     # dict page, parsed to values on host; indices stay
                     # on device as the column's codes
-                    draw = scratch[:dict_unc].cpu().numpy()
+                    hb0 = host_bytes.get(0)
+                    draw = (np.frombuffer(hb0, dtype=np.uint8)
+                            if hb0 is not None
+                            else scratch[:dict_unc].cpu().numpy())
                     vals_list = []
                     pos = 0
                     for _ in range(dict_n):
@@ -619,7 +629,9 @@ def read_files_batch_device(paths: List[str], device,
                     # shape (per-page torch slicing + a level-prefix
                     # stack().cpu() sync per chunk) serialized 26-unit
                     # decodes on the GIL.
-                    parts = []
+                    pieces = []
+                    dev_parts = []   # scratch slices still needing D2H
+                    dev_marks = []   # their slots in pieces
                     roff = []  # page -> offset of its span in hb
                     cur = 0
                     for j, page in enumerate(c.pages):
@@ -632,9 +644,26 @@ def read_files_batch_device(paths: List[str], device,
                             roff.append(-1)
                             continue
                         roff.append(cur)
-                        parts.append(scratch[base:base + ln_span])
+                        hbuf = host_bytes.get(
+                            j + 1 if has_zdict else j)
+                        if hbuf is not None:
+                            pieces.append(np.frombuffer(
+                                hbuf, dtype=np.uint8,
+                                count=ln_span))
+                        else:
+                            dev_marks.append(len(pieces))
+                            pieces.append(None)
+                            dev_parts.append(
+                                scratch[base:base + ln_span])
                         cur += ln_span
-                    hb = (_torch.cat(parts).cpu().numpy() if parts
+                    if dev_parts:
+                        dv = _torch.cat(dev_parts).cpu().numpy()
+                        off3 = 0
+                        for m, t3 in zip(dev_marks, dev_parts):
+                            ln3 = t3.numel()
+                            pieces[m] = dv[off3:off3 + ln3]
+                            off3 += ln3
+                    hb = (np.concatenate(pieces) if pieces
                           else np.empty(0, dtype=np.uint8))
                     if _dt == "3":
                         tA = tB = _time.perf_counter()
