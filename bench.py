@@ -174,7 +174,11 @@ def main():
     if workdir.startswith("/dev/shm"):
         # tmpfs free space is bounded by RAM, not the mount size
         free = min(free, _mem_available_bytes() - (48 << 30))
-    cleanup_inline = free < needed + (50 << 30)
+    # long runs force inline cleanup regardless of headroom estimates:
+    # letting tens of per-step index versions accumulate in tmpfs is
+    # how a host runs out of RAM mid-benchmark
+    cleanup_inline = free < needed + (50 << 30) or \
+        (args.steps + args.warmup) > 25
     if rank == 0 and os.environ.get("BENCH_DEBUG"):
         print(f"[dbg] disk free={free/2**30:.0f}G needed={needed/2**30:.0f}G"
               f" inline_cleanup={cleanup_inline}", file=sys.stderr)
