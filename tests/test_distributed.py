@@ -11,6 +11,10 @@ import numpy as np
 import pyarrow as pa
 import pyarrow.parquet as pq
 import pytest
+
+# a hung rendezvous must fail, not stall the suite
+pytestmark = pytest.mark.timeout(600)
+
 import torch
 import torch.multiprocessing as mp
 

@@ -18,6 +18,10 @@ import pytest
 import torch
 import torch.multiprocessing as mp
 
+# a hung rendezvous must fail, not stall the suite
+pytestmark = pytest.mark.timeout(600)
+
+
 
 def _mk_batch(rank, n, with_strings=True, with_nulls=True, seed=0):
     from hyperspace_amd.execution.columnar import ColumnBatch, StringColumn
