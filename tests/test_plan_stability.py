@@ -104,3 +104,33 @@ def test_explain_golden(env):
     with open(golden_path) as f:
         expected = f.read().rstrip("\n")
     assert out == expected
+
+
+def test_hybrid_delete_plan_stability(tmp_path):
+    """Hybrid Scan over DELETED source files: the plan excludes the
+    deleted file's rows via the lineage filter, shown as the index
+    scan's -Nfiles annotation (reference:
+    CoveringIndexRuleUtils.scala:247-252 lineage NOT-IN filter)."""
+    os.environ["HYPERSPACE_SYSTEM_PATH"] = str(tmp_path / "ix")
+    rng = np.random.default_rng(5)
+    d = tmp_path / "src"
+    d.mkdir()
+    for i in range(8):
+        pq.write_table(pa.table({"key": rng.integers(0, 500, 4000),
+                                 "val": rng.random(4000)}),
+                       str(d / f"p{i}.parquet"))
+    session = hs.HyperspaceSession(device="cpu")
+    session.conf.set(hs.IndexConstants.INDEX_NUM_BUCKETS, 8)
+    session.conf.set(hs.IndexConstants.INDEX_LINEAGE_ENABLED, True)
+    h = hs.Hyperspace(session)
+    df = session.read_parquet(str(d))
+    h.create_index(df, hs.CoveringIndexConfig("hdel", ["key"], ["val"]))
+    (d / "p7.parquet").unlink()
+    session.conf.set(hs.IndexConstants.INDEX_HYBRID_SCAN_ENABLED, True)
+    session.enable_hyperspace()
+    q = session.read_parquet(str(d)).filter("key = 77") \
+        .select("key", "val")
+    plan = normalize(q.optimized_plan().pretty())
+    assert plan == ("Project(['key', 'val'])\n"
+                    "  Filter((key = 77))\n"
+                    "    IndexScan(hdel, -<N>files)"), plan
